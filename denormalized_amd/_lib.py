@@ -1,0 +1,326 @@
+"""ctypes binding over the C ABI (include/denormalized_amd.h).
+
+The HIP engine is REQUIRED for compute: load failures and missing-GPU errors
+raise immediately — there is no CPU fallback in the product path.
+"""
+import ctypes
+import os
+
+import numpy as np
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_SO = os.path.join(_DIR, "_dzengine.so")
+
+DZ_OK = 0
+
+WINDOW_TUMBLING = 0
+WINDOW_SLIDING = 1
+AGG_COUNT, AGG_MIN, AGG_MAX, AGG_SUM, AGG_AVG = 0, 1, 2, 3, 4
+KEY_UTF8, KEY_INT64, KEY_DENSE_INT64 = 0, 1, 2
+CMP = {"<": 0, "<=": 1, ">": 2, ">=": 3, "==": 4, "!=": 5}
+AGG_BY_NAME = {"count": AGG_COUNT, "min": AGG_MIN, "max": AGG_MAX,
+               "sum": AGG_SUM, "avg": AGG_AVG, "average": AGG_AVG}
+
+
+class DzAggDesc(ctypes.Structure):
+    _fields_ = [("op", ctypes.c_int), ("input_col", ctypes.c_int32)]
+
+
+class DzWindowDesc(ctypes.Structure):
+    _fields_ = [
+        ("window_type", ctypes.c_int),
+        ("length_ms", ctypes.c_int64),
+        ("slide_ms", ctypes.c_int64),
+        ("ts_col", ctypes.c_int32),
+        ("group_col", ctypes.c_int32),
+        ("key_kind", ctypes.c_int),
+        ("aggs", ctypes.POINTER(DzAggDesc)),
+        ("n_aggs", ctypes.c_int32),
+        ("n_keys_hint", ctypes.c_int64),
+        ("device", ctypes.c_int32),
+        ("max_open_windows", ctypes.c_int32),
+    ]
+
+
+class DzColumn(ctypes.Structure):
+    _fields_ = [
+        ("len", ctypes.c_int64),
+        ("validity", ctypes.c_void_p),
+        ("offsets", ctypes.c_void_p),
+        ("data", ctypes.c_void_p),
+    ]
+
+
+class DzBatch(ctypes.Structure):
+    _fields_ = [("n_rows", ctypes.c_int64), ("n_cols", ctypes.c_int32),
+                ("cols", ctypes.POINTER(DzColumn))]
+
+
+class DzOutBatch(ctypes.Structure):
+    _fields_ = [
+        ("n_rows", ctypes.c_int64),
+        ("key_i64", ctypes.POINTER(ctypes.c_int64)),
+        ("key_offsets", ctypes.POINTER(ctypes.c_int32)),
+        ("key_data", ctypes.c_char_p),
+        ("agg_cols", ctypes.POINTER(ctypes.c_void_p)),
+        ("agg_valid", ctypes.POINTER(ctypes.c_uint8)),
+        ("window_start_ms", ctypes.POINTER(ctypes.c_int64)),
+        ("window_end_ms", ctypes.POINTER(ctypes.c_int64)),
+    ]
+
+
+class DzKernelStat(ctypes.Structure):
+    _fields_ = [("name", ctypes.c_char * 32), ("launches", ctypes.c_uint64),
+                ("total_ms", ctypes.c_double), ("bytes_per_launch_alg", ctypes.c_double)]
+
+
+_lib = None
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        if not os.path.exists(_SO):
+            raise RuntimeError(
+                f"denormalized_amd HIP engine missing: {_SO}. "
+                "Run __graft_entry__.build() (hipcc --offload-arch=gfx950).")
+        L = ctypes.CDLL(_SO)
+        p = ctypes.c_void_p
+        i64 = ctypes.c_int64
+        L.dz_window_op_create.restype = p
+        L.dz_window_op_create.argtypes = [ctypes.POINTER(DzWindowDesc)]
+        L.dz_window_op_push.argtypes = [p, ctypes.POINTER(DzBatch)]
+        L.dz_window_op_push_device.argtypes = [p, i64, p, p, p]
+        L.dz_window_op_poll.argtypes = [p, ctypes.POINTER(ctypes.POINTER(DzOutBatch))]
+        L.dz_window_op_finish.argtypes = [p]
+        L.dz_window_op_destroy.argtypes = [p]
+        L.dz_last_error.restype = ctypes.c_char_p
+        L.dz_last_error.argtypes = [p]
+        L.dz_window_op_advance_watermark.argtypes = [p, i64]
+        L.dz_window_op_watermark.restype = i64
+        L.dz_window_op_watermark.argtypes = [p]
+        L.dz_window_op_open_windows.restype = i64
+        L.dz_window_op_open_windows.argtypes = [p]
+        L.dz_window_op_set_filter.argtypes = [p, ctypes.c_int32, ctypes.c_int32,
+                                              ctypes.c_double]
+        L.dz_window_op_kernel_stats.argtypes = [p, ctypes.POINTER(DzKernelStat),
+                                                ctypes.c_int32,
+                                                ctypes.POINTER(ctypes.c_int32)]
+        L.dz_generate.argtypes = [ctypes.c_int32, ctypes.c_uint64, i64, i64, i64,
+                                  i64, i64, p, p, p, p]
+        L.dz_device_malloc.argtypes = [ctypes.c_int32, ctypes.c_size_t,
+                                       ctypes.POINTER(p)]
+        L.dz_device_free.argtypes = [p]
+        L.dz_device_synchronize.argtypes = [ctypes.c_int32]
+        L.dz_memcpy_d2h.argtypes = [p, p, ctypes.c_size_t]
+        L.dz_memcpy_h2d.argtypes = [p, p, ctypes.c_size_t]
+        L.dz_debug_windows_for_range.restype = i64
+        L.dz_debug_windows_for_range.argtypes = [i64, i64, i64, i64, p, p, i64]
+        L.dz_version.restype = ctypes.c_char_p
+        _lib = L
+    return _lib
+
+
+def _err(L, h=None):
+    m = L.dz_last_error(h)
+    return m.decode() if m else "unknown error"
+
+
+def _np_ptr(a):
+    return a.ctypes.data_as(ctypes.c_void_p) if a is not None else None
+
+
+class WindowOp:
+    """Python handle over dz_window_op (one partition's stream)."""
+
+    def __init__(self, length_ms, slide_ms=0, aggs=(("count", 0), ("min", 0),
+                 ("max", 0), ("avg", 0)), key_kind=KEY_UTF8, n_keys_hint=1024,
+                 device=0, max_open_windows=0, ts_col=0, group_col=1,
+                 value_col=2):
+        self._L = lib()
+        n = len(aggs)
+        self._agg_arr = (DzAggDesc * n)()
+        self.agg_names = []
+        for i, (name, _col) in enumerate(aggs):
+            self._agg_arr[i].op = AGG_BY_NAME[name] if isinstance(name, str) else name
+            self._agg_arr[i].input_col = value_col
+            self.agg_names.append(name if isinstance(name, str) else str(name))
+        d = DzWindowDesc(
+            window_type=WINDOW_SLIDING if slide_ms else WINDOW_TUMBLING,
+            length_ms=length_ms, slide_ms=slide_ms or 0, ts_col=ts_col,
+            group_col=group_col, key_kind=key_kind,
+            aggs=ctypes.cast(self._agg_arr, ctypes.POINTER(DzAggDesc)),
+            n_aggs=n, n_keys_hint=n_keys_hint, device=device,
+            max_open_windows=max_open_windows)
+        self.key_kind = key_kind
+        self._h = self._L.dz_window_op_create(ctypes.byref(d))
+        if not self._h:
+            raise RuntimeError(f"dz_window_op_create failed: {_err(self._L)}")
+
+    def _check(self, st, what):
+        if st != DZ_OK:
+            raise RuntimeError(f"{what} failed: {_err(self._L, self._h)}")
+
+    def push(self, ts_ms, keys, vals, val_valid_bitmap=None):
+        """Host-batch push (the drop-in boundary). keys: np.int64 array, or a
+        list of str/bytes for utf8. val_valid_bitmap: LSB-first bitmap bytes."""
+        ts_ms = np.ascontiguousarray(ts_ms, np.int64)
+        vals = np.ascontiguousarray(vals, np.float64)
+        n = len(ts_ms)
+        cols = (DzColumn * 3)()
+        cols[0] = DzColumn(n, None, None, _np_ptr(ts_ms).value if n else None)
+        keep = []
+        if self.key_kind == KEY_UTF8:
+            data = b"".join(k.encode() if isinstance(k, str) else bytes(k) for k in keys)
+            offs = np.zeros(n + 1, np.int32)
+            pos = 0
+            for i, k in enumerate(keys):
+                pos += len(k.encode() if isinstance(k, str) else bytes(k))
+                offs[i + 1] = pos
+            buf = np.frombuffer(data, np.uint8) if data else np.zeros(1, np.uint8)
+            keep += [offs, buf]
+            cols[1] = DzColumn(n, None, _np_ptr(offs).value, _np_ptr(buf).value)
+        else:
+            karr = np.ascontiguousarray(keys, np.int64)
+            keep.append(karr)
+            cols[1] = DzColumn(n, None, None, _np_ptr(karr).value if n else None)
+        bm = None
+        if val_valid_bitmap is not None:
+            bm = np.ascontiguousarray(val_valid_bitmap, np.uint8)
+            keep.append(bm)
+        cols[2] = DzColumn(n, _np_ptr(bm).value if bm is not None else None, None,
+                           _np_ptr(vals).value if n else None)
+        batch = DzBatch(n, 3, ctypes.cast(cols, ctypes.POINTER(DzColumn)))
+        self._check(self._L.dz_window_op_push(self._h, ctypes.byref(batch)), "push")
+
+    def push_device(self, n, d_ts, d_kid32, d_vals):
+        self._check(self._L.dz_window_op_push_device(
+            self._h, n, d_ts, d_kid32, d_vals), "push_device")
+
+    def poll(self):
+        """Returns a dict of numpy copies of one emitted batch, or None."""
+        outp = ctypes.POINTER(DzOutBatch)()
+        self._check(self._L.dz_window_op_poll(self._h, ctypes.byref(outp)), "poll")
+        if not outp:
+            return None
+        ob = outp.contents
+        n = ob.n_rows
+        res = {"n_rows": n}
+        if self.key_kind == KEY_UTF8:
+            offs = np.ctypeslib.as_array(ob.key_offsets, (n + 1,)).copy() if n else np.zeros(1, np.int32)
+            total = int(offs[-1]) if n else 0
+            data = ctypes.string_at(ob.key_data, total) if total else b""
+            res["key"] = [data[offs[i]:offs[i + 1]].decode() for i in range(n)]
+        else:
+            res["key"] = np.ctypeslib.as_array(ob.key_i64, (n,)).copy() if n else np.zeros(0, np.int64)
+        for i, name in enumerate(self.agg_names):
+            pt = ob.agg_cols[i]
+            if AGG_BY_NAME.get(name, name) == AGG_COUNT or name == "count":
+                arr = np.ctypeslib.as_array(ctypes.cast(pt, ctypes.POINTER(ctypes.c_int64)), (n,)).copy() if n else np.zeros(0, np.int64)
+            else:
+                arr = np.ctypeslib.as_array(ctypes.cast(pt, ctypes.POINTER(ctypes.c_double)), (n,)).copy() if n else np.zeros(0, np.float64)
+            res[name] = arr
+        res["valid"] = np.ctypeslib.as_array(ob.agg_valid, (n,)).copy() if n else np.zeros(0, np.uint8)
+        res["window_start"] = np.ctypeslib.as_array(ob.window_start_ms, (n,)).copy() if n else np.zeros(0, np.int64)
+        res["window_end"] = np.ctypeslib.as_array(ob.window_end_ms, (n,)).copy() if n else np.zeros(0, np.int64)
+        return res
+
+    def poll_all(self):
+        out = []
+        while True:
+            b = self.poll()
+            if b is None:
+                return out
+            out.append(b)
+
+    def finish(self):
+        self._check(self._L.dz_window_op_finish(self._h), "finish")
+
+    def set_filter(self, agg_name_or_idx, cmp, literal):
+        idx = (self.agg_names.index(agg_name_or_idx)
+               if isinstance(agg_name_or_idx, str) else agg_name_or_idx)
+        self._check(self._L.dz_window_op_set_filter(
+            self._h, idx, CMP[cmp] if isinstance(cmp, str) else cmp,
+            float(literal)), "set_filter")
+
+    def advance_watermark(self, wm_ms):
+        self._check(self._L.dz_window_op_advance_watermark(self._h, wm_ms),
+                    "advance_watermark")
+
+    @property
+    def watermark(self):
+        return self._L.dz_window_op_watermark(self._h)
+
+    @property
+    def open_windows(self):
+        return self._L.dz_window_op_open_windows(self._h)
+
+    def kernel_stats(self):
+        arr = (DzKernelStat * 16)()
+        nout = ctypes.c_int32()
+        self._check(self._L.dz_window_op_kernel_stats(self._h, arr, 16,
+                                                      ctypes.byref(nout)), "stats")
+        return {arr[i].name.decode(): {
+            "launches": arr[i].launches, "total_ms": arr[i].total_ms,
+            "bytes_per_launch_alg": arr[i].bytes_per_launch_alg,
+        } for i in range(nout.value)}
+
+    def close(self):
+        if getattr(self, "_h", None):
+            self._L.dz_window_op_destroy(self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+class DeviceArray:
+    """Raw device allocation (no torch types cross the C ABI)."""
+
+    def __init__(self, device, nbytes):
+        self._L = lib()
+        self.device = device
+        self.nbytes = nbytes
+        p = ctypes.c_void_p()
+        if self._L.dz_device_malloc(device, nbytes, ctypes.byref(p)) != DZ_OK:
+            raise RuntimeError("device malloc failed")
+        self.ptr = p
+
+    def to_host(self, dtype, count):
+        out = np.empty(count, dtype)
+        if self._L.dz_memcpy_d2h(_np_ptr(out), self.ptr, out.nbytes) != DZ_OK:
+            raise RuntimeError("d2h failed")
+        return out
+
+    def from_host(self, arr):
+        arr = np.ascontiguousarray(arr)
+        if self._L.dz_memcpy_h2d(self.ptr, _np_ptr(arr), arr.nbytes) != DZ_OK:
+            raise RuntimeError("h2d failed")
+
+    def free(self):
+        if getattr(self, "ptr", None):
+            self._L.dz_device_free(self.ptr)
+            self.ptr = None
+
+    def __del__(self):
+        try:
+            self.free()
+        except Exception:
+            pass
+
+
+def generate(device, seed, t0_ms, start_row, n_rows, n_keys, rows_per_ms,
+             d_ts=None, d_keys=None, d_kid32=None, d_vals=None):
+    L = lib()
+    if L.dz_generate(device, seed, t0_ms, start_row, n_rows, n_keys, rows_per_ms,
+                     d_ts, d_keys, d_kid32, d_vals) != DZ_OK:
+        raise RuntimeError(f"dz_generate failed: {_err(L)}")
+
+
+def synchronize(device=0):
+    L = lib()
+    if L.dz_device_synchronize(device) != DZ_OK:
+        raise RuntimeError("device synchronize failed")

@@ -1,0 +1,71 @@
+/* dz_internal.h — shared internal declarations between the engine host code
+ * (window_op.cpp) and the CDNA4 kernels (kernels.hip). Not part of the C ABI.
+ */
+#pragma once
+#include <cstdint>
+#include <hip/hip_runtime.h>
+
+namespace dz {
+
+/* Partition geometry. Rows are bucketed by (key_id & (NB-1)): with dense
+ * dictionary ids this balances buckets to ±1 key. One wave folds one bucket,
+ * so NB also sets fold parallelism (4096 waves = 16 waves/CU on 256 CUs). */
+constexpr int NB = 4096;
+constexpr int LOG_NB = 12;
+constexpr int BLOCK = 256;          /* 4 waves */
+constexpr int WAVES_PER_BLOCK = BLOCK / 64;
+constexpr int MAX_RANGES = 4096;    /* window frames touched by one batch */
+
+/* meta word packed per record: kloc (8b) | widx (12b) | valid (1b) */
+constexpr uint32_t META_KLOC_MASK = 0xFFu;
+constexpr int META_WIDX_SHIFT = 8;
+constexpr uint32_t META_WIDX_MASK = 0xFFFu;
+constexpr int META_VALID_SHIFT = 20;
+
+struct WinParams {
+    int64_t s0;          /* first window start of this batch's range list */
+    int64_t len_ms;
+    int64_t slide_ms;    /* 0 => tumbling */
+    int32_t nw;          /* ranges in this batch */
+    int32_t is_sliding;
+};
+
+struct FoldChunk {
+    int32_t w_lo, w_hi;  /* widx range [w_lo, w_hi) this launch folds */
+    int32_t k_lo, k_hi;  /* kloc range [k_lo, k_hi) */
+    int64_t kcap;        /* state key capacity (multiple of NB) */
+    uint32_t batch_seq;
+};
+
+/* Launch wrappers implemented in kernels.hip. All run on `stream`;
+ * n = rows in batch, C = partition chunks (hist/scatter grid). */
+void launch_gen(hipStream_t stream, uint64_t seed, int64_t t0, int64_t start_row,
+                int64_t n, int64_t nkeys, int64_t rows_per_ms,
+                int64_t* d_ts, int64_t* d_keys, int32_t* d_kid, double* d_vals);
+
+/* d_scalars: [0]=min ts (order-mapped u64), [1]=max ts, [2]=max kid (u64).
+ * Host must pre-set [0]=UINT64_MAX, [1]=0, [2]=0. */
+void launch_minmax(hipStream_t stream, const int64_t* d_ts, const int32_t* d_kid,
+                   int64_t n, uint64_t* d_scalars);
+
+void launch_hist(hipStream_t stream, const int32_t* d_kid, const int64_t* d_ts,
+                 int64_t n, int64_t chunk, int C, const WinParams& wp,
+                 uint32_t* d_ghist);
+
+void launch_scan(hipStream_t stream, const uint32_t* d_ghist, int C,
+                 uint32_t* d_total, uint32_t* d_bucket_base /*NB+1*/,
+                 uint32_t* d_gofs);
+
+void launch_scatter(hipStream_t stream, const int32_t* d_kid, const int64_t* d_ts,
+                    const double* d_vals, const uint8_t* d_validity /*bitmap|null*/,
+                    int64_t n, int64_t chunk, int C, const WinParams& wp,
+                    const uint32_t* d_gofs, uint32_t* d_meta, uint32_t* d_ridx,
+                    double* d_rvals);
+
+void launch_fold(hipStream_t stream, const uint32_t* d_meta, const uint32_t* d_ridx,
+                 const double* d_rvals, const uint32_t* d_bucket_base,
+                 const FoldChunk& fc, const int32_t* d_slot_of_widx,
+                 uint64_t* s_cnt, double* s_min, double* s_max, double* s_sum,
+                 uint64_t* s_first);
+
+} // namespace dz

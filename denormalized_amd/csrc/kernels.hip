@@ -1,0 +1,402 @@
+/* kernels.hip — hand-written CDNA4 (gfx950) kernels for the streaming
+ * windowed aggregate. MI355X-first design, not a translation: the reference
+ * computes this path on CPU with arrow-rs compare/filter kernels plus a
+ * DataFusion hash table and accumulators over 32-row batches
+ * (grouped_window_agg_stream.rs:548-605, :501-537); here the same semantics
+ * run as a deterministic stable key-partition + per-group row-order fold over
+ * multi-million-row device-resident batches:
+ *
+ *   k_minmax   batch watermark bounds (time.rs:31-57) + max key id
+ *   k_hist     per-chunk bucket histogram (bucket = key_id & (NB-1)),
+ *              with per-row window multiplicity (sliding windows expand)
+ *   k_scan_*   bucket bases + per-chunk stable offsets
+ *   k_scatter  stable partition: records (meta,rowidx,value) land in bucket
+ *              regions IN GLOBAL ROW ORDER (weighted intra-wave ranks via
+ *              wave-64 shuffles + wave-serialized LDS cursors)
+ *   k_fold     one wave per bucket; each lane owns one (window,key) group and
+ *              folds its rows IN ROW ORDER — count/min/max and the f64 sum
+ *              are bit-identical to the reference's sequential accumulator
+ *              updates (update_batch call order, grouped_window_agg_stream.rs:533)
+ *
+ * No atomics on the data path (only LDS histogram counts); every kernel is
+ * deterministic. Roofline: HBM bandwidth (no contraction => no MFMA).
+ */
+#include "dz_internal.h"
+
+#include <algorithm>
+
+namespace dz {
+
+__device__ __forceinline__ int64_t i64min(int64_t a, int64_t b) { return a < b ? a : b; }
+
+/* ------------------------------------------------------------------ */
+/* helpers                                                             */
+/* ------------------------------------------------------------------ */
+
+__device__ __forceinline__ uint64_t splitmix64(uint64_t x) {
+    x += 0x9e3779b97f4a7c15ULL;
+    x = (x ^ (x >> 30)) * 0xbf58476d1ce4e5b9ULL;
+    x = (x ^ (x >> 27)) * 0x94d049bb133111ebULL;
+    return x ^ (x >> 31);
+}
+
+/* snap_to_window_start (streaming_window.rs:1088-1094): whole-second
+ * truncation; ms generalization for sub-second lengths. Assumes
+ * ts >= 0 (the reference's SystemTime arithmetic asserts the same). */
+__device__ __forceinline__ int64_t snap_ms(int64_t ts, int64_t len_ms) {
+    int64_t len_s = len_ms / 1000;
+    if (len_s == 0) return ts - (ts % len_ms);
+    return (ts / 1000) / len_s * len_s * 1000;
+}
+
+__device__ __forceinline__ int64_t floordiv(int64_t a, int64_t b) {
+    int64_t q = a / b, r = a % b;
+    return q - ((r != 0) & ((r < 0) != (b < 0)));
+}
+
+/* Window coverage of a row: first widx and count (sliding expands).
+ * Mirrors get_windows_for_watermark membership (streaming_window.rs:1053-1086):
+ * row t is in window j iff starts[j] <= t < starts[j]+len. */
+__device__ __forceinline__ void row_windows(int64_t t, const WinParams& wp,
+                                            int32_t* jmin, int32_t* m) {
+    if (wp.is_sliding) {
+        int64_t lo = floordiv(t - wp.len_ms - wp.s0, wp.slide_ms) + 1; /* start > t-len */
+        int64_t hi = floordiv(t - wp.s0, wp.slide_ms);                 /* start <= t    */
+        if (lo < 0) lo = 0;
+        if (hi > wp.nw - 1) hi = wp.nw - 1;
+        *jmin = (int32_t)lo;
+        *m = (hi >= lo) ? (int32_t)(hi - lo + 1) : 0;
+    } else {
+        /* windows are consecutive [s0 + i*len) (streaming_window.rs:1076-1082) */
+        *jmin = (int32_t)((t - wp.s0) / wp.len_ms);
+        *m = 1;
+    }
+}
+
+/* ------------------------------------------------------------------ */
+/* generator (spec shared with oracle orc_gen)   */
+/* ------------------------------------------------------------------ */
+
+__global__ void k_gen(uint64_t seed, int64_t t0, int64_t start_row, int64_t n,
+                      int64_t nkeys, int64_t rows_per_ms, int64_t* ts,
+                      int64_t* keys, int32_t* kid, double* vals) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n; i += stride) {
+        int64_t gi = start_row + i;
+        uint64_t r = splitmix64(seed ^ (0x9e3779b97f4a7c15ULL * (uint64_t)(gi + 1)));
+        uint64_t k = r % (uint64_t)nkeys;
+        if (ts) ts[i] = t0 + gi / rows_per_ms;
+        if (keys) keys[i] = (int64_t)k;
+        if (kid) kid[i] = (int32_t)k;
+        if (vals) vals[i] = (double)(splitmix64(r) >> 11) * (1.0 / 9007199254740992.0) * 115.0;
+    }
+}
+
+void launch_gen(hipStream_t s, uint64_t seed, int64_t t0, int64_t start_row,
+                int64_t n, int64_t nkeys, int64_t rows_per_ms,
+                int64_t* d_ts, int64_t* d_keys, int32_t* d_kid, double* d_vals) {
+    int blocks = (int)std::min<int64_t>((n + BLOCK - 1) / BLOCK, 2048);
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(k_gen, dim3(blocks), dim3(BLOCK), 0, s, seed, t0,
+                       start_row, n, nkeys, rows_per_ms, d_ts, d_keys, d_kid, d_vals);
+}
+
+/* ------------------------------------------------------------------ */
+/* batch min/max ts + max kid                                          */
+/* ------------------------------------------------------------------ */
+
+__device__ __forceinline__ uint64_t map_i64(int64_t x) {
+    return (uint64_t)x ^ 0x8000000000000000ULL; /* order-preserving i64->u64 */
+}
+
+__global__ void k_minmax(const int64_t* ts, const int32_t* kid, int64_t n,
+                         uint64_t* scalars) {
+    uint64_t mn = ~0ULL, mx = 0, km = 0;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint64_t t = map_i64(ts[i]);
+        mn = min(mn, t);
+        mx = max(mx, t);
+        if (kid) km = max(km, (uint64_t)(uint32_t)kid[i]);
+    }
+    /* wave reduce */
+    for (int o = 32; o > 0; o >>= 1) {
+        mn = min(mn, (uint64_t)__shfl_down((unsigned long long)mn, o));
+        mx = max(mx, (uint64_t)__shfl_down((unsigned long long)mx, o));
+        km = max(km, (uint64_t)__shfl_down((unsigned long long)km, o));
+    }
+    if ((threadIdx.x & 63) == 0) {
+        atomicMin((unsigned long long*)&scalars[0], (unsigned long long)mn);
+        atomicMax((unsigned long long*)&scalars[1], (unsigned long long)mx);
+        atomicMax((unsigned long long*)&scalars[2], (unsigned long long)km);
+    }
+}
+
+void launch_minmax(hipStream_t s, const int64_t* d_ts, const int32_t* d_kid,
+                   int64_t n, uint64_t* d_scalars) {
+    int blocks = (int)std::min<int64_t>((n + BLOCK - 1) / BLOCK, 2048);
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(k_minmax, dim3(blocks), dim3(BLOCK), 0, s, d_ts, d_kid, n,
+                       d_scalars);
+}
+
+/* ------------------------------------------------------------------ */
+/* histogram                                                           */
+/* ------------------------------------------------------------------ */
+
+__global__ __launch_bounds__(BLOCK) void k_hist(const int32_t* kid,
+        const int64_t* ts, int64_t n, int64_t chunk, WinParams wp,
+        uint32_t* ghist) {
+    __shared__ uint32_t h[NB];
+    for (int t = threadIdx.x; t < NB; t += BLOCK) h[t] = 0;
+    __syncthreads();
+    int64_t lo = blockIdx.x * chunk;
+    int64_t hi = i64min(n, lo + chunk);
+    for (int64_t i = lo + threadIdx.x; i < hi; i += BLOCK) {
+        int32_t jm, m;
+        row_windows(ts[i], wp, &jm, &m);
+        atomicAdd(&h[kid[i] & (NB - 1)], (uint32_t)m);
+    }
+    __syncthreads();
+    for (int t = threadIdx.x; t < NB; t += BLOCK)
+        ghist[(int64_t)blockIdx.x * NB + t] = h[t];
+}
+
+void launch_hist(hipStream_t s, const int32_t* d_kid, const int64_t* d_ts,
+                 int64_t n, int64_t chunk, int C, const WinParams& wp,
+                 uint32_t* d_ghist) {
+    hipLaunchKernelGGL(k_hist, dim3(C), dim3(BLOCK), 0, s, d_kid, d_ts, n, chunk,
+                       wp, d_ghist);
+}
+
+/* ------------------------------------------------------------------ */
+/* scans: bucket totals -> bases; per-chunk stable offsets             */
+/* ------------------------------------------------------------------ */
+
+/* one wave per 64 buckets, coalesced column sweep over chunks */
+__global__ void k_scan_totals(const uint32_t* ghist, int C, uint32_t* total) {
+    int bkt = blockIdx.x * blockDim.x + threadIdx.x;
+    if (bkt >= NB) return;
+    uint32_t s = 0;
+    for (int c = 0; c < C; c++) s += ghist[(int64_t)c * NB + bkt];
+    total[bkt] = s;
+}
+
+/* single block: exclusive scan of NB totals -> bucket_base[NB+1] */
+__global__ __launch_bounds__(1024) void k_scan_base(const uint32_t* total,
+                                                    uint32_t* base) {
+    __shared__ uint32_t part[1024];
+    constexpr int PER = NB / 1024;
+    uint32_t loc[PER];
+    uint32_t s = 0;
+    for (int j = 0; j < PER; j++) {
+        loc[j] = s;
+        s += total[threadIdx.x * PER + j];
+    }
+    part[threadIdx.x] = s;
+    __syncthreads();
+    /* Hillis-Steele inclusive scan over 1024 partials */
+    for (int o = 1; o < 1024; o <<= 1) {
+        uint32_t v = (threadIdx.x >= o) ? part[threadIdx.x - o] : 0;
+        __syncthreads();
+        part[threadIdx.x] += v;
+        __syncthreads();
+    }
+    uint32_t pre = (threadIdx.x > 0) ? part[threadIdx.x - 1] : 0;
+    for (int j = 0; j < PER; j++)
+        base[threadIdx.x * PER + j] = pre + loc[j];
+    if (threadIdx.x == 1023) base[NB] = part[1023];
+}
+
+/* per-chunk running offsets: gofs[c][bkt] = base[bkt] + sum_{c'<c} ghist[c'][bkt] */
+__global__ void k_scan_offsets(const uint32_t* ghist, int C,
+                               const uint32_t* base, uint32_t* gofs) {
+    int bkt = blockIdx.x * blockDim.x + threadIdx.x;
+    if (bkt >= NB) return;
+    uint32_t run = base[bkt];
+    for (int c = 0; c < C; c++) {
+        uint32_t t = ghist[(int64_t)c * NB + bkt];
+        gofs[(int64_t)c * NB + bkt] = run;
+        run += t;
+    }
+}
+
+void launch_scan(hipStream_t s, const uint32_t* d_ghist, int C,
+                 uint32_t* d_total, uint32_t* d_base, uint32_t* d_gofs) {
+    hipLaunchKernelGGL(k_scan_totals, dim3(NB / 256), dim3(256), 0, s, d_ghist, C,
+                       d_total);
+    hipLaunchKernelGGL(k_scan_base, dim3(1), dim3(1024), 0, s, d_total, d_base);
+    hipLaunchKernelGGL(k_scan_offsets, dim3(NB / 256), dim3(256), 0, s, d_ghist, C,
+                       d_base, d_gofs);
+}
+
+/* ------------------------------------------------------------------ */
+/* stable scatter                                                      */
+/* ------------------------------------------------------------------ */
+
+__global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
+        const int64_t* ts, const double* vals, const uint8_t* validity,
+        int64_t n, int64_t chunk, WinParams wp, const uint32_t* gofs,
+        uint32_t* rmeta, uint32_t* rridx, double* rvals) {
+    __shared__ uint32_t cur[NB];
+    for (int t = threadIdx.x; t < NB; t += BLOCK)
+        cur[t] = gofs[(int64_t)blockIdx.x * NB + t];
+    __syncthreads();
+
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    int64_t lo = blockIdx.x * chunk;
+    int64_t hi = i64min(n, lo + chunk);
+    constexpr uint32_t SENT = 0xFFFFFFFFu;
+
+    for (int64_t t0 = lo; t0 < hi; t0 += BLOCK) {
+        int64_t i = t0 + threadIdx.x;
+        uint32_t bkt = SENT;
+        int32_t jmin = 0, m = 0;
+        uint32_t kv = 0;
+        int64_t tsv = 0;
+        double v = 0.0;
+        uint32_t valid = 1;
+        if (i < hi) {
+            kv = (uint32_t)kid[i];
+            tsv = ts[i];
+            v = vals[i];
+            if (validity) valid = (validity[i >> 3] >> (i & 7)) & 1u;
+            row_windows(tsv, wp, &jmin, &m);
+            bkt = kv & (NB - 1);
+        }
+        /* weighted intra-wave rank (row order), per-bucket wave totals,
+         * first lane per bucket. 64-wide loop keeps ranks deterministic. */
+        uint32_t r = 0, wtot = 0;
+        int fl = lane;
+        for (int j = 0; j < 64; j++) {
+            uint32_t bj = (uint32_t)__shfl((int)bkt, j);
+            uint32_t mj = (uint32_t)__shfl(m, j);
+            if (bj == bkt) {
+                if (j < lane) r += mj;
+                wtot += mj;
+                if (j < fl) fl = j;
+            }
+        }
+        /* wave-serialized cursor allocation keeps cross-wave row order */
+        uint32_t base = 0;
+        for (int w = 0; w < WAVES_PER_BLOCK; w++) {
+            if (wave == w) {
+                uint32_t pre = 0;
+                if (lane == fl && bkt != SENT) {
+                    pre = cur[bkt];
+                    cur[bkt] = pre + wtot;
+                }
+                pre = (uint32_t)__shfl((int)pre, fl);
+                base = pre + r;
+            }
+            __syncthreads();
+        }
+        uint32_t kloc = kv >> LOG_NB;
+        for (int jj = 0; jj < m; jj++) {
+            uint32_t d = base + jj;
+            rmeta[d] = kloc | ((uint32_t)(jmin + jj) << META_WIDX_SHIFT)
+                       | (valid << META_VALID_SHIFT);
+            rridx[d] = (uint32_t)i;
+            rvals[d] = v;
+        }
+    }
+}
+
+void launch_scatter(hipStream_t s, const int32_t* d_kid, const int64_t* d_ts,
+                    const double* d_vals, const uint8_t* d_validity, int64_t n,
+                    int64_t chunk, int C, const WinParams& wp,
+                    const uint32_t* d_gofs, uint32_t* d_meta, uint32_t* d_ridx,
+                    double* d_rvals) {
+    hipLaunchKernelGGL(k_scatter, dim3(C), dim3(BLOCK), 0, s, d_kid, d_ts, d_vals,
+                       d_validity, n, chunk, wp, d_gofs, d_meta, d_ridx, d_rvals);
+}
+
+/* ------------------------------------------------------------------ */
+/* fold — one wave per bucket, one lane per (window,key) group,        */
+/* rows folded in row order (bit-exact vs the reference accumulators)  */
+/* ------------------------------------------------------------------ */
+
+__global__ __launch_bounds__(BLOCK) void k_fold(const uint32_t* rmeta,
+        const uint32_t* rridx, const double* rvals, const uint32_t* bucket_base,
+        FoldChunk fc, const int32_t* slot_of_widx, uint64_t* s_cnt,
+        double* s_min, double* s_max, double* s_sum, uint64_t* s_first) {
+    const int bkt = (int)((blockIdx.x * (int64_t)BLOCK + threadIdx.x) >> 6);
+    const int lane = threadIdx.x & 63;
+    if (bkt >= NB) return;
+    const uint32_t lo = bucket_base[bkt];
+    const uint32_t hi = bucket_base[bkt + 1];
+    if (lo == hi) return;
+
+    const int nk = fc.k_hi - fc.k_lo;
+    const int my_widx = fc.w_lo + lane / nk;
+    const int my_kloc = fc.k_lo + lane % nk;
+    const bool own = my_widx < fc.w_hi;
+    const uint32_t want = (uint32_t)my_kloc | ((uint32_t)my_widx << META_WIDX_SHIFT);
+
+    int64_t sidx = 0;
+    uint64_t cnt = 0, fst = ~0ULL;
+    double mn = 0.0, mx = 0.0, sm = 0.0;
+    if (own) {
+        int64_t slot = slot_of_widx[my_widx];
+        sidx = slot * fc.kcap + (((int64_t)my_kloc << LOG_NB) | bkt);
+        cnt = s_cnt[sidx];
+        fst = s_first[sidx];
+        if (cnt > 0) {
+            mn = s_min[sidx];
+            mx = s_max[sidx];
+            sm = s_sum[sidx];
+        }
+    }
+
+    for (uint32_t base = lo; base < hi; base += 64) {
+        const int nt = (int)min(64u, hi - base);
+        uint32_t m = 0, ri = 0;
+        double v = 0.0;
+        if (lane < nt) {
+            m = rmeta[base + lane];
+            ri = rridx[base + lane];
+            v = rvals[base + lane];
+        }
+        for (int s = 0; s < nt; s++) {
+            const uint32_t ms_ = (uint32_t)__shfl((int)m, s);
+            const double vs = __shfl(v, s);
+            const uint32_t rs = (uint32_t)__shfl((int)ri, s);
+            if ((ms_ & 0xFFFFFu) == want && own) {
+                if (fst == ~0ULL) fst = ((uint64_t)fc.batch_seq << 32) | rs;
+                if (ms_ >> META_VALID_SHIFT) {
+                    /* DF accumulator semantics: count non-null; min/max by
+                     * strict compare (first non-null initializes); sum in
+                     * row order (see oracle.c pinning note) */
+                    if (cnt == 0) { mn = vs; mx = vs; }
+                    else {
+                        if (vs < mn) mn = vs;
+                        if (vs > mx) mx = vs;
+                    }
+                    cnt++;
+                    sm += vs;
+                }
+            }
+        }
+    }
+
+    if (own) {
+        s_cnt[sidx] = cnt;
+        s_first[sidx] = fst;
+        s_min[sidx] = mn;
+        s_max[sidx] = mx;
+        s_sum[sidx] = sm;
+    }
+}
+
+void launch_fold(hipStream_t s, const uint32_t* d_meta, const uint32_t* d_ridx,
+                 const double* d_rvals, const uint32_t* d_bucket_base,
+                 const FoldChunk& fc, const int32_t* d_slot_of_widx,
+                 uint64_t* s_cnt, double* s_min, double* s_max, double* s_sum,
+                 uint64_t* s_first) {
+    hipLaunchKernelGGL(k_fold, dim3(NB / WAVES_PER_BLOCK), dim3(BLOCK), 0, s,
+                       d_meta, d_ridx, d_rvals, d_bucket_base, fc, d_slot_of_widx,
+                       s_cnt, s_min, s_max, s_sum, s_first);
+}
+
+} // namespace dz

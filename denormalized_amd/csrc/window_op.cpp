@@ -1,0 +1,892 @@
+/* window_op.cpp — MI355X-native engine behind the C ABI in
+ * include/denormalized_amd.h. Host-side mirror of the reference's
+ * GroupedWindowAggStream control flow (window frames, watermark, trigger:
+ * crates/core/src/physical_plan/continuous/grouped_window_agg_stream.rs) with
+ * all per-row work on the GPU (kernels.hip). The GPU is REQUIRED: every
+ * compute entry point fails loudly when no HIP device is present — there is
+ * no CPU fallback in the product path.
+ */
+#include "dz_internal.h"
+#include "denormalized_amd.h"
+
+#include <algorithm>
+#include <cstdio>
+#include <cstring>
+#include <deque>
+#include <map>
+#include <string>
+#include <unordered_map>
+#include <vector>
+
+namespace {
+
+std::string g_err; /* create-time errors (no handle yet) */
+
+int64_t snap_ms_host(int64_t ts, int64_t len_ms) {
+    int64_t len_s = len_ms / 1000;
+    if (len_s == 0) return ts - (ts % len_ms);
+    return (ts / 1000) / len_s * len_s * 1000;
+}
+
+} // namespace
+
+/* get_windows_for_watermark restated for the host (streaming_window.rs:
+ * 1053-1086; ms generalization for sub-second lengths per SURVEY §7). */
+extern "C" int64_t dz_debug_windows_for_range(int64_t min_ts, int64_t max_ts,
+                                              int64_t len_ms, int64_t slide_ms,
+                                              int64_t* starts, int64_t* ends,
+                                              int64_t cap) {
+    int64_t n = 0;
+    if (slide_ms > 0) {
+        int64_t cur = snap_ms_host(min_ts - len_ms, len_ms);
+        while (cur <= max_ts) {
+            int64_t end = cur + len_ms;
+            if (!(min_ts > end || max_ts < cur)) {
+                if (n < cap) { starts[n] = cur; ends[n] = end; }
+                n++;
+            }
+            cur += slide_ms;
+        }
+    } else {
+        int64_t cur = snap_ms_host(min_ts, len_ms);
+        while (cur <= max_ts) {
+            if (n < cap) { starts[n] = cur; ends[n] = cur + len_ms; }
+            n++;
+            cur += len_ms;
+        }
+    }
+    return n;
+}
+
+extern "C" const char* dz_version(void) { return "denormalized-amd 0.1 (gfx950)"; }
+
+/* ------------------------------------------------------------------ */
+
+struct PendingEvent {
+    std::string name;
+    hipEvent_t a, b;
+    double bytes;
+};
+
+struct KStatAcc {
+    uint64_t launches = 0;
+    double ms = 0.0;
+    double last_bytes = 0.0;
+};
+
+struct OutBuf {
+    std::vector<int64_t> key_i64;
+    std::vector<int32_t> key_offsets;
+    std::vector<char> key_data;
+    std::vector<std::vector<int64_t>> agg_i64; /* per agg col (COUNT) */
+    std::vector<std::vector<double>> agg_f64;  /* per agg col (others) */
+    std::vector<uint8_t> agg_valid;
+    std::vector<int64_t> wstart, wend;
+    /* view plumbing */
+    std::vector<const void*> agg_ptrs;
+    dz_out_batch view;
+};
+
+struct dz_window_op {
+    /* config */
+    dz_window_type wtype;
+    int64_t len_ms, slide_ms;
+    int32_t ts_col, group_col;
+    dz_key_kind key_kind;
+    std::vector<dz_agg_desc> aggs;
+    int device = 0;
+    int32_t max_open = 64;
+    std::string err;
+
+    hipStream_t stream = nullptr;
+
+    /* dictionary (host side; first-seen insertion order == emitted id order
+     * is NOT assumed — emission sorts by first row; the dict only maps
+     * key values <-> dense ids) */
+    std::unordered_map<std::string, int32_t> dict_utf8;
+    std::vector<std::string> dict_strs;
+    std::unordered_map<int64_t, int32_t> dict_i64;
+    std::vector<int64_t> dict_vals;
+    int64_t n_keys = 0; /* dense key count seen so far */
+
+    /* persistent group state [slot][kcap] per field */
+    int64_t kcap = 0; /* multiple of NB */
+    int32_t nslots = 0;
+    uint64_t* s_cnt = nullptr;
+    uint64_t* s_first = nullptr;
+    double* s_min = nullptr;
+    double* s_max = nullptr;
+    double* s_sum = nullptr;
+    std::vector<int32_t> free_slots;
+    struct OpenWin { int64_t end; int32_t slot; };
+    std::map<int64_t, OpenWin> open; /* by window start (BTreeMap order) */
+
+    int64_t watermark = INT64_MIN;
+    bool has_wm = false;
+    uint32_t batch_seq = 0;
+
+    /* scratch */
+    uint32_t* d_ghist = nullptr;
+    uint32_t* d_gofs = nullptr;
+    uint32_t* d_total = nullptr;
+    uint32_t* d_base = nullptr;
+    int C_cap = 0;
+    uint32_t* d_meta = nullptr;
+    uint32_t* d_ridx = nullptr;
+    double* d_rvals = nullptr;
+    int64_t rec_cap = 0;
+    uint64_t* d_scalars = nullptr;
+    uint64_t* h_scalars = nullptr; /* pinned, 3 */
+    int32_t* d_slotmap = nullptr;
+    int slotmap_cap = 0;
+
+    /* input staging (host-batch path) */
+    int64_t* d_ts = nullptr;
+    int32_t* d_kid = nullptr;
+    double* d_vals = nullptr;
+    uint8_t* d_valbm = nullptr;
+    int64_t in_cap = 0;
+    char* h_stage = nullptr; /* pinned */
+    size_t h_stage_cap = 0;
+
+    /* pinned emission staging */
+    uint64_t* e_cnt = nullptr;
+    uint64_t* e_first = nullptr;
+    double* e_min = nullptr;
+    double* e_max = nullptr;
+    double* e_sum = nullptr;
+    int64_t e_cap = 0;
+
+    /* filter pushdown */
+    bool has_filter = false;
+    int32_t f_idx = 0, f_cmp = 0;
+    double f_lit = 0.0;
+
+    /* output */
+    std::deque<OutBuf> outq;
+    OutBuf current;
+    bool has_current = false;
+
+    /* timing */
+    std::vector<hipEvent_t> ev_pool;
+    std::deque<PendingEvent> pending;
+    std::map<std::string, KStatAcc> stats;
+};
+
+#define CHK(op, call)                                                      \
+    do {                                                                   \
+        hipError_t e_ = (call);                                            \
+        if (e_ != hipSuccess) {                                            \
+            (op)->err = std::string(#call) + ": " + hipGetErrorString(e_); \
+            return DZ_ERR;                                                 \
+        }                                                                  \
+    } while (0)
+
+#define CHKV(op, call)                                                     \
+    do {                                                                   \
+        hipError_t e_ = (call);                                            \
+        if (e_ != hipSuccess) {                                            \
+            (op)->err = std::string(#call) + ": " + hipGetErrorString(e_); \
+            return;                                                        \
+        }                                                                  \
+    } while (0)
+
+static hipEvent_t get_event(dz_window_op* op) {
+    if (!op->ev_pool.empty()) {
+        hipEvent_t e = op->ev_pool.back();
+        op->ev_pool.pop_back();
+        return e;
+    }
+    hipEvent_t e;
+    hipEventCreate(&e);
+    return e;
+}
+
+static void drain_events(dz_window_op* op, bool wait) {
+    while (!op->pending.empty()) {
+        PendingEvent& p = op->pending.front();
+        if (!wait && hipEventQuery(p.b) != hipSuccess) break;
+        if (wait) hipEventSynchronize(p.b);
+        float ms = 0;
+        hipEventElapsedTime(&ms, p.a, p.b);
+        KStatAcc& s = op->stats[p.name];
+        s.launches++;
+        s.ms += ms;
+        s.last_bytes = p.bytes;
+        op->ev_pool.push_back(p.a);
+        op->ev_pool.push_back(p.b);
+        op->pending.pop_front();
+    }
+}
+
+template <typename F>
+static void timed(dz_window_op* op, const char* name, double bytes, F&& fn) {
+    hipEvent_t a = get_event(op), b = get_event(op);
+    hipEventRecord(a, op->stream);
+    fn();
+    hipEventRecord(b, op->stream);
+    op->pending.push_back({name, a, b, bytes});
+}
+
+/* ------------------------------------------------------------------ */
+/* state allocation / growth                                           */
+/* ------------------------------------------------------------------ */
+
+static dz_status state_alloc(dz_window_op* op, int64_t kcap_new, int32_t nslots_new) {
+    /* allocate fresh arrays, memset to empty, copy any open-slot regions */
+    kcap_new = std::max<int64_t>(kcap_new, dz::NB);
+    kcap_new = (kcap_new + dz::NB - 1) / dz::NB * dz::NB;
+    if (kcap_new / dz::NB > 255) {
+        op->err = "key capacity exceeds 255*NB (~1.04M) — not supported yet";
+        return DZ_ERR;
+    }
+    if (kcap_new == op->kcap && nslots_new <= op->nslots) return DZ_OK;
+    nslots_new = std::max(nslots_new, op->nslots);
+    size_t cells = (size_t)kcap_new * nslots_new;
+    uint64_t* n_cnt; uint64_t* n_first; double* n_min; double* n_max; double* n_sum;
+    CHK(op, hipMalloc(&n_cnt, cells * 8));
+    CHK(op, hipMalloc(&n_first, cells * 8));
+    CHK(op, hipMalloc(&n_min, cells * 8));
+    CHK(op, hipMalloc(&n_max, cells * 8));
+    CHK(op, hipMalloc(&n_sum, cells * 8));
+    CHK(op, hipMemsetAsync(n_cnt, 0, cells * 8, op->stream));
+    CHK(op, hipMemsetAsync(n_first, 0xFF, cells * 8, op->stream));
+    /* min/max/sum need no init: fold ignores them while cnt == 0 */
+    for (auto& kv : op->open) { /* preserve open-window state */
+        int32_t s = kv.second.slot;
+        int64_t span = std::min(op->kcap, kcap_new);
+        CHK(op, hipMemcpyAsync(n_cnt + (size_t)s * kcap_new, op->s_cnt + (size_t)s * op->kcap,
+                               span * 8, hipMemcpyDeviceToDevice, op->stream));
+        CHK(op, hipMemcpyAsync(n_first + (size_t)s * kcap_new, op->s_first + (size_t)s * op->kcap,
+                               span * 8, hipMemcpyDeviceToDevice, op->stream));
+        CHK(op, hipMemcpyAsync(n_min + (size_t)s * kcap_new, op->s_min + (size_t)s * op->kcap,
+                               span * 8, hipMemcpyDeviceToDevice, op->stream));
+        CHK(op, hipMemcpyAsync(n_max + (size_t)s * kcap_new, op->s_max + (size_t)s * op->kcap,
+                               span * 8, hipMemcpyDeviceToDevice, op->stream));
+        CHK(op, hipMemcpyAsync(n_sum + (size_t)s * kcap_new, op->s_sum + (size_t)s * op->kcap,
+                               span * 8, hipMemcpyDeviceToDevice, op->stream));
+    }
+    CHK(op, hipStreamSynchronize(op->stream));
+    hipFree(op->s_cnt); hipFree(op->s_first); hipFree(op->s_min);
+    hipFree(op->s_max); hipFree(op->s_sum);
+    op->s_cnt = n_cnt; op->s_first = n_first; op->s_min = n_min;
+    op->s_max = n_max; op->s_sum = n_sum;
+    for (int32_t s = op->nslots; s < nslots_new; s++) op->free_slots.push_back(s);
+    op->kcap = kcap_new;
+    op->nslots = nslots_new;
+    return DZ_OK;
+}
+
+static dz_status slot_reset(dz_window_op* op, int32_t slot) {
+    CHK(op, hipMemsetAsync(op->s_cnt + (size_t)slot * op->kcap, 0, op->kcap * 8, op->stream));
+    CHK(op, hipMemsetAsync(op->s_first + (size_t)slot * op->kcap, 0xFF, op->kcap * 8, op->stream));
+    return DZ_OK;
+}
+
+/* ------------------------------------------------------------------ */
+/* create / destroy                                                    */
+/* ------------------------------------------------------------------ */
+
+extern "C" dz_window_op* dz_window_op_create(const dz_window_desc* desc) {
+    g_err.clear();
+    if (!desc) { g_err = "null desc"; return nullptr; }
+    if (desc->length_ms <= 0) { g_err = "length_ms must be > 0"; return nullptr; }
+    if (desc->window_type == DZ_WINDOW_SLIDING && desc->slide_ms <= 0) {
+        g_err = "sliding window needs slide_ms > 0";
+        return nullptr;
+    }
+    if (desc->n_aggs <= 0 || !desc->aggs) { g_err = "need at least one aggregate"; return nullptr; }
+    int ndev = 0;
+    if (hipGetDeviceCount(&ndev) != hipSuccess || ndev <= desc->device) {
+        g_err = "no HIP device available (this operator has no CPU fallback)";
+        return nullptr;
+    }
+    auto* op = new dz_window_op();
+    op->wtype = desc->window_type;
+    op->len_ms = desc->length_ms;
+    op->slide_ms = desc->window_type == DZ_WINDOW_SLIDING ? desc->slide_ms : 0;
+    op->ts_col = desc->ts_col;
+    op->group_col = desc->group_col;
+    op->key_kind = desc->key_kind;
+    op->aggs.assign(desc->aggs, desc->aggs + desc->n_aggs);
+    op->device = desc->device;
+    op->max_open = desc->max_open_windows > 0 ? desc->max_open_windows : 64;
+    if (hipSetDevice(op->device) != hipSuccess ||
+        hipStreamCreate(&op->stream) != hipSuccess) {
+        g_err = "hip device/stream init failed";
+        delete op;
+        return nullptr;
+    }
+    hipMalloc(&op->d_scalars, 3 * 8);
+    hipHostMalloc((void**)&op->h_scalars, 3 * 8);
+    int64_t hint = std::max<int64_t>(desc->n_keys_hint, 1);
+    if (state_alloc(op, hint, 4) != DZ_OK) {
+        g_err = op->err;
+        hipStreamDestroy(op->stream);
+        delete op;
+        return nullptr;
+    }
+    return op;
+}
+
+extern "C" void dz_window_op_destroy(dz_window_op* op) {
+    if (!op) return;
+    hipSetDevice(op->device);
+    hipStreamSynchronize(op->stream);
+    drain_events(op, true);
+    for (auto e : op->ev_pool) hipEventDestroy(e);
+    hipFree(op->s_cnt); hipFree(op->s_first); hipFree(op->s_min);
+    hipFree(op->s_max); hipFree(op->s_sum);
+    hipFree(op->d_ghist); hipFree(op->d_gofs); hipFree(op->d_total); hipFree(op->d_base);
+    hipFree(op->d_meta); hipFree(op->d_ridx); hipFree(op->d_rvals);
+    hipFree(op->d_scalars); hipFree(op->d_slotmap);
+    hipFree(op->d_ts); hipFree(op->d_kid); hipFree(op->d_vals); hipFree(op->d_valbm);
+    if (op->h_scalars) hipHostFree(op->h_scalars);
+    if (op->h_stage) hipHostFree(op->h_stage);
+    if (op->e_cnt) hipHostFree(op->e_cnt);
+    if (op->e_first) hipHostFree(op->e_first);
+    if (op->e_min) hipHostFree(op->e_min);
+    if (op->e_max) hipHostFree(op->e_max);
+    if (op->e_sum) hipHostFree(op->e_sum);
+    hipStreamDestroy(op->stream);
+    delete op;
+}
+
+extern "C" const char* dz_last_error(dz_window_op* op) {
+    if (!op) return g_err.empty() ? nullptr : g_err.c_str();
+    return op->err.empty() ? nullptr : op->err.c_str();
+}
+
+/* ------------------------------------------------------------------ */
+/* emission (trigger_windows, grouped_window_agg_stream.rs:220-253)    */
+/* ------------------------------------------------------------------ */
+
+static bool filter_pass(dz_window_op* op, const OutBuf& ob, int64_t row_cnt,
+                        double vmin, double vmax, double vsum, bool valid) {
+    (void)ob;
+    if (!op->has_filter) return true;
+    double v;
+    dz_agg_op o = op->aggs[op->f_idx].op;
+    if (o == DZ_AGG_COUNT) {
+        v = (double)row_cnt;
+    } else {
+        if (!valid) return false; /* NULL never passes a comparison filter */
+        switch (o) {
+            case DZ_AGG_MIN: v = vmin; break;
+            case DZ_AGG_MAX: v = vmax; break;
+            case DZ_AGG_SUM: v = vsum; break;
+            case DZ_AGG_AVG: v = vsum / (double)row_cnt; break;
+            default: v = 0; break;
+        }
+    }
+    switch (op->f_cmp) {
+        case 0: return v < op->f_lit;
+        case 1: return v <= op->f_lit;
+        case 2: return v > op->f_lit;
+        case 3: return v >= op->f_lit;
+        case 4: return v == op->f_lit;
+        case 5: return v != op->f_lit;
+        default: return true;
+    }
+}
+
+static dz_status emit_window(dz_window_op* op, int64_t wstart, int64_t wend,
+                             int32_t slot) {
+    int64_t K = op->n_keys;
+    if (K > op->e_cap) {
+        int64_t nc = std::max<int64_t>(K, op->e_cap ? op->e_cap * 2 : 4096);
+        if (op->e_cnt) { hipHostFree(op->e_cnt); hipHostFree(op->e_first);
+                         hipHostFree(op->e_min); hipHostFree(op->e_max); hipHostFree(op->e_sum); }
+        CHK(op, hipHostMalloc((void**)&op->e_cnt, nc * 8));
+        CHK(op, hipHostMalloc((void**)&op->e_first, nc * 8));
+        CHK(op, hipHostMalloc((void**)&op->e_min, nc * 8));
+        CHK(op, hipHostMalloc((void**)&op->e_max, nc * 8));
+        CHK(op, hipHostMalloc((void**)&op->e_sum, nc * 8));
+        op->e_cap = nc;
+    }
+    if (K > 0) {
+        size_t base = (size_t)slot * op->kcap;
+        CHK(op, hipMemcpyAsync(op->e_cnt, op->s_cnt + base, K * 8, hipMemcpyDeviceToHost, op->stream));
+        CHK(op, hipMemcpyAsync(op->e_first, op->s_first + base, K * 8, hipMemcpyDeviceToHost, op->stream));
+        CHK(op, hipMemcpyAsync(op->e_min, op->s_min + base, K * 8, hipMemcpyDeviceToHost, op->stream));
+        CHK(op, hipMemcpyAsync(op->e_max, op->s_max + base, K * 8, hipMemcpyDeviceToHost, op->stream));
+        CHK(op, hipMemcpyAsync(op->e_sum, op->s_sum + base, K * 8, hipMemcpyDeviceToHost, op->stream));
+        CHK(op, hipStreamSynchronize(op->stream));
+        drain_events(op, false);
+    }
+    /* groups in first-seen (insertion) order: GroupValues emits insertion
+     * order; we sort touched keys by first-row sequence (stable, exact). */
+    std::vector<int32_t> touched;
+    touched.reserve(1024);
+    for (int64_t k = 0; k < K; k++)
+        if (op->e_first[k] != ~0ULL) touched.push_back((int32_t)k);
+    std::stable_sort(touched.begin(), touched.end(), [&](int32_t a, int32_t b) {
+        return op->e_first[a] < op->e_first[b];
+    });
+
+    OutBuf ob;
+    size_t na = op->aggs.size();
+    ob.agg_i64.resize(na);
+    ob.agg_f64.resize(na);
+    bool utf8 = op->key_kind == DZ_KEY_UTF8;
+    if (utf8) ob.key_offsets.push_back(0);
+    for (int32_t k : touched) {
+        int64_t cnt = (int64_t)op->e_cnt[k];
+        bool valid = cnt > 0;
+        double mn = op->e_min[k], mx = op->e_max[k], sm = op->e_sum[k];
+        if (!filter_pass(op, ob, cnt, mn, mx, sm, valid)) continue;
+        if (utf8) {
+            const std::string& s = op->dict_strs[k];
+            ob.key_data.insert(ob.key_data.end(), s.begin(), s.end());
+            ob.key_offsets.push_back((int32_t)ob.key_data.size());
+        } else if (op->key_kind == DZ_KEY_INT64) {
+            ob.key_i64.push_back(op->dict_vals[k]);
+        } else {
+            ob.key_i64.push_back(k);
+        }
+        for (size_t a = 0; a < na; a++) {
+            switch (op->aggs[a].op) {
+                case DZ_AGG_COUNT: ob.agg_i64[a].push_back(cnt); break;
+                case DZ_AGG_MIN: ob.agg_f64[a].push_back(valid ? mn : 0.0); break;
+                case DZ_AGG_MAX: ob.agg_f64[a].push_back(valid ? mx : 0.0); break;
+                case DZ_AGG_SUM: ob.agg_f64[a].push_back(valid ? sm : 0.0); break;
+                case DZ_AGG_AVG:
+                    ob.agg_f64[a].push_back(valid ? sm / (double)cnt : 0.0);
+                    break;
+            }
+        }
+        ob.agg_valid.push_back(valid ? 1 : 0);
+        ob.wstart.push_back(wstart);
+        ob.wend.push_back(wend);
+    }
+    ob.view.n_rows = (int64_t)ob.wstart.size();
+    op->outq.push_back(std::move(ob));
+    return DZ_OK;
+}
+
+static dz_status trigger_windows(dz_window_op* op) {
+    if (!op->has_wm) return DZ_OK;
+    for (auto it = op->open.begin(); it != op->open.end();) {
+        if (op->watermark >= it->second.end) {
+            if (emit_window(op, it->first, it->second.end, it->second.slot) != DZ_OK)
+                return DZ_ERR;
+            if (slot_reset(op, it->second.slot) != DZ_OK) return DZ_ERR;
+            op->free_slots.push_back(it->second.slot);
+            it = op->open.erase(it);
+        } else {
+            ++it;
+        }
+    }
+    return DZ_OK;
+}
+
+/* ------------------------------------------------------------------ */
+/* the push core (device-resident inputs)                              */
+/* ------------------------------------------------------------------ */
+
+static dz_status ensure_scratch(dz_window_op* op, int C, int64_t nrec) {
+    if (C > op->C_cap) {
+        hipFree(op->d_ghist); hipFree(op->d_gofs);
+        CHK(op, hipMalloc(&op->d_ghist, (size_t)C * dz::NB * 4));
+        CHK(op, hipMalloc(&op->d_gofs, (size_t)C * dz::NB * 4));
+        if (!op->d_total) {
+            CHK(op, hipMalloc(&op->d_total, dz::NB * 4));
+            CHK(op, hipMalloc(&op->d_base, (dz::NB + 1) * 4));
+        }
+        op->C_cap = C;
+    }
+    if (nrec > op->rec_cap) {
+        hipFree(op->d_meta); hipFree(op->d_ridx); hipFree(op->d_rvals);
+        CHK(op, hipMalloc(&op->d_meta, (size_t)nrec * 4));
+        CHK(op, hipMalloc(&op->d_ridx, (size_t)nrec * 4));
+        CHK(op, hipMalloc(&op->d_rvals, (size_t)nrec * 8));
+        op->rec_cap = nrec;
+    }
+    return DZ_OK;
+}
+
+static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
+                           const int32_t* d_kid, const double* d_vals,
+                           const uint8_t* d_valbm, bool keys_are_dense) {
+    if (n <= 0) return DZ_OK; /* empty batch: reference emits empty (no-op) */
+    CHK(op, hipSetDevice(op->device));
+
+    /* 1. batch watermark bounds + max key id (time.rs:31-57) */
+    CHK(op, hipMemsetAsync(op->d_scalars, 0xFF, 8, op->stream));
+    CHK(op, hipMemsetAsync(op->d_scalars + 1, 0, 16, op->stream));
+    timed(op, "minmax", (double)n * 12, [&] {
+        dz::launch_minmax(op->stream, d_ts, keys_are_dense ? d_kid : nullptr, n,
+                          op->d_scalars);
+    });
+    CHK(op, hipMemcpyAsync(op->h_scalars, op->d_scalars, 24, hipMemcpyDeviceToHost,
+                           op->stream));
+    CHK(op, hipStreamSynchronize(op->stream));
+    drain_events(op, false);
+    int64_t mn = (int64_t)(op->h_scalars[0] ^ 0x8000000000000000ULL);
+    int64_t mx = (int64_t)(op->h_scalars[1] ^ 0x8000000000000000ULL);
+    if (mn < 0) {
+        op->err = "negative timestamps not supported (reference SystemTime panics too)";
+        return DZ_ERR;
+    }
+    if (keys_are_dense) {
+        int64_t kmax = (int64_t)op->h_scalars[2];
+        op->n_keys = std::max(op->n_keys, kmax + 1);
+    }
+    if (op->n_keys > op->kcap) {
+        if (state_alloc(op, std::max(op->n_keys, op->kcap * 2), op->nslots) != DZ_OK)
+            return DZ_ERR;
+    }
+
+    /* 2. window ranges for this batch + frame slots */
+    std::vector<int64_t> ws(dz::MAX_RANGES), we(dz::MAX_RANGES);
+    int64_t nw = dz_debug_windows_for_range(mn, mx, op->len_ms, op->slide_ms,
+                                            ws.data(), we.data(), dz::MAX_RANGES);
+    if (nw > dz::MAX_RANGES) {
+        op->err = "batch spans more than " + std::to_string(dz::MAX_RANGES) +
+                  " windows";
+        return DZ_ERR;
+    }
+    std::vector<int32_t> slotmap(nw);
+    for (int64_t r = 0; r < nw; r++) {
+        auto it = op->open.find(ws[r]);
+        if (it == op->open.end()) {
+            if (op->free_slots.empty()) {
+                int32_t cap_slots = std::max(op->max_open, 64);
+                if (op->nslots >= cap_slots) {
+                    op->err = "too many open windows (cap " +
+                              std::to_string(cap_slots) + ")";
+                    return DZ_ERR;
+                }
+                int32_t grow = std::min(std::max(4, op->nslots),
+                                        cap_slots - op->nslots);
+                if (state_alloc(op, op->kcap, op->nslots + grow) != DZ_OK)
+                    return DZ_ERR;
+            }
+            int32_t slot = op->free_slots.back();
+            op->free_slots.pop_back();
+            if (slot_reset(op, slot) != DZ_OK) return DZ_ERR;
+            op->open[ws[r]] = {we[r], slot};
+            slotmap[r] = slot;
+        } else {
+            slotmap[r] = it->second.slot;
+        }
+    }
+    if ((int)nw > op->slotmap_cap) {
+        hipFree(op->d_slotmap);
+        CHK(op, hipMalloc(&op->d_slotmap, nw * 4));
+        op->slotmap_cap = (int)nw;
+    }
+    if (nw > 0)
+        CHK(op, hipMemcpyAsync(op->d_slotmap, slotmap.data(), nw * 4,
+                               hipMemcpyHostToDevice, op->stream));
+
+    dz::WinParams wp;
+    wp.s0 = nw > 0 ? ws[0] : 0;
+    wp.len_ms = op->len_ms;
+    wp.slide_ms = op->slide_ms;
+    wp.nw = (int32_t)nw;
+    wp.is_sliding = op->slide_ms > 0;
+
+    /* 3. partition + fold */
+    int C = (int)std::min<int64_t>(2048, std::max<int64_t>(1, (n + 1023) / 1024));
+    int64_t chunk = (n + C - 1) / C;
+    int64_t expand = wp.is_sliding
+        ? (op->len_ms + op->slide_ms - 1) / op->slide_ms + 1 : 1;
+    int64_t nrec_max = n * expand;
+    if (nrec_max >= (int64_t)UINT32_MAX) {
+        op->err = "batch expands past 2^32 records; push smaller batches";
+        return DZ_ERR;
+    }
+    if (ensure_scratch(op, C, nrec_max) != DZ_OK) return DZ_ERR;
+
+    timed(op, "hist", (double)n * 12, [&] {
+        dz::launch_hist(op->stream, d_kid, d_ts, n, chunk, C, wp, op->d_ghist);
+    });
+    timed(op, "scan", (double)C * dz::NB * 12, [&] {
+        dz::launch_scan(op->stream, op->d_ghist, C, op->d_total, op->d_base,
+                        op->d_gofs);
+    });
+    timed(op, "scatter", (double)n * 20 + (double)nrec_max * 16, [&] {
+        dz::launch_scatter(op->stream, d_kid, d_ts, d_vals, d_valbm, n, chunk, C,
+                           wp, op->d_gofs, op->d_meta, op->d_ridx, op->d_rvals);
+    });
+    int64_t klocs = op->kcap >> dz::LOG_NB;
+    for (int64_t k_lo = 0; k_lo < klocs; k_lo += 64) {
+        int32_t nk = (int32_t)std::min<int64_t>(64, klocs - k_lo);
+        int32_t wstep = 64 / nk;
+        for (int64_t w_lo = 0; w_lo < nw; w_lo += wstep) {
+            dz::FoldChunk fc;
+            fc.w_lo = (int32_t)w_lo;
+            fc.w_hi = (int32_t)std::min<int64_t>(nw, w_lo + wstep);
+            fc.k_lo = (int32_t)k_lo;
+            fc.k_hi = (int32_t)(k_lo + nk);
+            fc.kcap = op->kcap;
+            fc.batch_seq = op->batch_seq;
+            timed(op, "fold", (double)nrec_max * 16, [&] {
+                dz::launch_fold(op->stream, op->d_meta, op->d_ridx, op->d_rvals,
+                                op->d_base, fc, op->d_slotmap, op->s_cnt,
+                                op->s_min, op->s_max, op->s_sum, op->s_first);
+            });
+        }
+    }
+    op->batch_seq++;
+
+    /* 4. watermark (running max of batch minimums, :255-266) + trigger */
+    if (!op->has_wm || op->watermark <= mn) {
+        op->watermark = mn;
+        op->has_wm = true;
+    }
+    return trigger_windows(op);
+}
+
+extern "C" dz_status dz_window_op_push_device(dz_window_op* op, int64_t n_rows,
+                                              const int64_t* d_ts_ms,
+                                              const int32_t* d_key_ids,
+                                              const double* d_vals) {
+    if (!op) return DZ_ERR;
+    return push_core(op, n_rows, d_ts_ms, d_key_ids, d_vals, nullptr, true);
+}
+
+/* ------------------------------------------------------------------ */
+/* host-batch push: dictionary-encode + stage + H2D                    */
+/* ------------------------------------------------------------------ */
+
+extern "C" dz_status dz_window_op_push(dz_window_op* op, const dz_batch* batch) {
+    if (!op) return DZ_ERR;
+    if (!batch) { op->err = "null batch"; return DZ_ERR; }
+    int64_t n = batch->n_rows;
+    if (n == 0) return DZ_OK;
+    int32_t need = std::max({op->ts_col, op->group_col});
+    for (auto& a : op->aggs) need = std::max(need, a.input_col);
+    if (batch->n_cols <= need) { op->err = "batch has too few columns"; return DZ_ERR; }
+    const dz_column& tsc = batch->cols[op->ts_col];
+    const dz_column& kc = batch->cols[op->group_col];
+    const dz_column& vc = batch->cols[op->aggs[0].input_col];
+    for (auto& a : op->aggs)
+        if (a.input_col != op->aggs[0].input_col) {
+            op->err = "all aggregates must share one input column (hot-path shape)";
+            return DZ_ERR;
+        }
+
+    CHK(op, hipSetDevice(op->device));
+    /* staging buffers */
+    size_t bm_bytes = (size_t)((n + 7) / 8);
+    size_t stage = (size_t)n * (8 + 4 + 8) + bm_bytes + 64;
+    if (stage > op->h_stage_cap) {
+        if (op->h_stage) hipHostFree(op->h_stage);
+        CHK(op, hipHostMalloc((void**)&op->h_stage, stage));
+        op->h_stage_cap = stage;
+    }
+    if (n > op->in_cap) {
+        hipFree(op->d_ts); hipFree(op->d_kid); hipFree(op->d_vals); hipFree(op->d_valbm);
+        CHK(op, hipMalloc(&op->d_ts, (size_t)n * 8));
+        CHK(op, hipMalloc(&op->d_kid, (size_t)n * 4));
+        CHK(op, hipMalloc(&op->d_vals, (size_t)n * 8));
+        CHK(op, hipMalloc(&op->d_valbm, bm_bytes ? bm_bytes : 1));
+        op->in_cap = n;
+    }
+    int64_t* h_ts = (int64_t*)op->h_stage;
+    int32_t* h_kid = (int32_t*)(h_ts + n);
+    double* h_vals = (double*)(h_kid + n);
+    uint8_t* h_bm = (uint8_t*)(h_vals + n);
+
+    memcpy(h_ts, tsc.data, (size_t)n * 8);
+    memcpy(h_vals, vc.data, (size_t)n * 8);
+    bool have_bm = vc.validity != nullptr;
+    if (have_bm) memcpy(h_bm, vc.validity, bm_bytes);
+
+    /* dictionary encode keys (first-seen dense ids; the per-frame insertion
+     * order the reference emits is recovered at emission via first-row sort) */
+    if (op->key_kind == DZ_KEY_UTF8) {
+        const int32_t* offs = kc.offsets;
+        const char* data = (const char*)kc.data;
+        if (!offs || !data) { op->err = "utf8 key column needs offsets+data"; return DZ_ERR; }
+        for (int64_t i = 0; i < n; i++) {
+            std::string s(data + offs[i], data + offs[i + 1]);
+            auto it = op->dict_utf8.find(s);
+            int32_t id;
+            if (it == op->dict_utf8.end()) {
+                id = (int32_t)op->dict_strs.size();
+                op->dict_utf8.emplace(s, id);
+                op->dict_strs.push_back(std::move(s));
+            } else {
+                id = it->second;
+            }
+            h_kid[i] = id;
+        }
+        op->n_keys = (int64_t)op->dict_strs.size();
+    } else if (op->key_kind == DZ_KEY_INT64) {
+        const int64_t* kv = (const int64_t*)kc.data;
+        for (int64_t i = 0; i < n; i++) {
+            auto it = op->dict_i64.find(kv[i]);
+            int32_t id;
+            if (it == op->dict_i64.end()) {
+                id = (int32_t)op->dict_vals.size();
+                op->dict_i64.emplace(kv[i], id);
+                op->dict_vals.push_back(kv[i]);
+            } else {
+                id = it->second;
+            }
+            h_kid[i] = id;
+        }
+        op->n_keys = (int64_t)op->dict_vals.size();
+    } else { /* DENSE: caller promises ids in [0, n_keys) */
+        const int64_t* kv = (const int64_t*)kc.data;
+        for (int64_t i = 0; i < n; i++) h_kid[i] = (int32_t)kv[i];
+    }
+    if (op->n_keys > op->kcap) {
+        if (state_alloc(op, std::max(op->n_keys, op->kcap * 2), op->nslots) != DZ_OK)
+            return DZ_ERR;
+    }
+
+    CHK(op, hipMemcpyAsync(op->d_ts, h_ts, (size_t)n * 8, hipMemcpyHostToDevice, op->stream));
+    CHK(op, hipMemcpyAsync(op->d_kid, h_kid, (size_t)n * 4, hipMemcpyHostToDevice, op->stream));
+    CHK(op, hipMemcpyAsync(op->d_vals, h_vals, (size_t)n * 8, hipMemcpyHostToDevice, op->stream));
+    if (have_bm)
+        CHK(op, hipMemcpyAsync(op->d_valbm, h_bm, bm_bytes, hipMemcpyHostToDevice, op->stream));
+
+    return push_core(op, n, op->d_ts, op->d_kid, op->d_vals,
+                     have_bm ? op->d_valbm : nullptr,
+                     op->key_kind == DZ_KEY_DENSE_INT64);
+}
+
+/* ------------------------------------------------------------------ */
+/* poll / finish / watermark / filter / stats                          */
+/* ------------------------------------------------------------------ */
+
+extern "C" dz_status dz_window_op_poll(dz_window_op* op, const dz_out_batch** out) {
+    if (!op || !out) return DZ_ERR;
+    *out = nullptr;
+    if (op->outq.empty()) return DZ_OK;
+    op->current = std::move(op->outq.front());
+    op->outq.pop_front();
+    op->has_current = true;
+    OutBuf& ob = op->current;
+    ob.agg_ptrs.clear();
+    for (size_t a = 0; a < op->aggs.size(); a++) {
+        if (op->aggs[a].op == DZ_AGG_COUNT)
+            ob.agg_ptrs.push_back((const void*)ob.agg_i64[a].data());
+        else
+            ob.agg_ptrs.push_back((const void*)ob.agg_f64[a].data());
+    }
+    ob.view.key_i64 = ob.key_i64.empty() ? nullptr : ob.key_i64.data();
+    ob.view.key_offsets = ob.key_offsets.empty() ? nullptr : ob.key_offsets.data();
+    ob.view.key_data = ob.key_data.empty() ? nullptr : ob.key_data.data();
+    ob.view.agg_cols = ob.agg_ptrs.data();
+    ob.view.agg_valid = ob.agg_valid.data();
+    ob.view.window_start_ms = ob.wstart.data();
+    ob.view.window_end_ms = ob.wend.data();
+    *out = &ob.view;
+    return DZ_OK;
+}
+
+extern "C" dz_status dz_window_op_finish(dz_window_op* op) {
+    if (!op) return DZ_ERR;
+    CHK(op, hipSetDevice(op->device));
+    int64_t mx = op->has_wm ? op->watermark : INT64_MIN;
+    for (auto& kv : op->open) mx = std::max(mx, kv.second.end);
+    if (mx != INT64_MIN) {
+        op->watermark = mx;
+        op->has_wm = true;
+    }
+    if (trigger_windows(op) != DZ_OK) return DZ_ERR;
+    CHK(op, hipStreamSynchronize(op->stream));
+    drain_events(op, true);
+    return DZ_OK;
+}
+
+extern "C" dz_status dz_window_op_advance_watermark(dz_window_op* op, int64_t wm) {
+    if (!op) return DZ_ERR;
+    CHK(op, hipSetDevice(op->device));
+    if (!op->has_wm || op->watermark <= wm) {
+        op->watermark = wm;
+        op->has_wm = true;
+    }
+    return trigger_windows(op);
+}
+
+extern "C" int64_t dz_window_op_watermark(dz_window_op* op) {
+    return (op && op->has_wm) ? op->watermark : INT64_MIN;
+}
+
+extern "C" int64_t dz_window_op_open_windows(dz_window_op* op) {
+    return op ? (int64_t)op->open.size() : 0;
+}
+
+extern "C" dz_status dz_window_op_set_filter(dz_window_op* op, int32_t agg_idx,
+                                             int32_t cmp, double literal) {
+    if (!op) return DZ_ERR;
+    if (agg_idx < 0 || agg_idx >= (int32_t)op->aggs.size()) {
+        op->err = "filter agg index out of range";
+        return DZ_ERR;
+    }
+    op->has_filter = true;
+    op->f_idx = agg_idx;
+    op->f_cmp = cmp;
+    op->f_lit = literal;
+    return DZ_OK;
+}
+
+extern "C" dz_status dz_window_op_kernel_stats(dz_window_op* op,
+                                               dz_kernel_stat* out, int32_t cap,
+                                               int32_t* n_out) {
+    if (!op) return DZ_ERR;
+    CHK(op, hipStreamSynchronize(op->stream));
+    drain_events(op, true);
+    int32_t n = 0;
+    for (auto& kv : op->stats) {
+        if (n >= cap) break;
+        snprintf(out[n].name, sizeof(out[n].name), "%s", kv.first.c_str());
+        out[n].launches = kv.second.launches;
+        out[n].total_ms = kv.second.ms;
+        out[n].bytes_per_launch_alg = kv.second.last_bytes;
+        n++;
+    }
+    if (n_out) *n_out = n;
+    return DZ_OK;
+}
+
+/* ------------------------------------------------------------------ */
+/* generator + device helpers                                          */
+/* ------------------------------------------------------------------ */
+
+static std::string g_util_err;
+
+extern "C" dz_status dz_generate(int32_t device, uint64_t seed, int64_t t0_ms,
+                                 int64_t start_row, int64_t n_rows, int64_t n_keys,
+                                 int64_t rows_per_ms, int64_t* d_ts, int64_t* d_keys,
+                                 int32_t* d_key_ids, double* d_vals) {
+    if (hipSetDevice(device) != hipSuccess) { g_err = "hipSetDevice failed"; return DZ_ERR; }
+    dz::launch_gen(nullptr, seed, t0_ms, start_row, n_rows, n_keys, rows_per_ms,
+                   d_ts, d_keys, d_key_ids, d_vals);
+    if (hipGetLastError() != hipSuccess) { g_err = "dz_generate launch failed"; return DZ_ERR; }
+    return DZ_OK;
+}
+
+extern "C" dz_status dz_device_malloc(int32_t device, size_t bytes, void** out) {
+    if (hipSetDevice(device) != hipSuccess || hipMalloc(out, bytes) != hipSuccess) {
+        g_err = "hipMalloc failed";
+        return DZ_ERR;
+    }
+    return DZ_OK;
+}
+
+extern "C" dz_status dz_device_free(void* p) {
+    return hipFree(p) == hipSuccess ? DZ_OK : DZ_ERR;
+}
+
+extern "C" dz_status dz_device_synchronize(int32_t device) {
+    if (hipSetDevice(device) != hipSuccess || hipDeviceSynchronize() != hipSuccess) {
+        g_err = "hipDeviceSynchronize failed";
+        return DZ_ERR;
+    }
+    return DZ_OK;
+}
+
+extern "C" dz_status dz_memcpy_d2h(void* dst, const void* src, size_t bytes) {
+    return hipMemcpy(dst, src, bytes, hipMemcpyDeviceToHost) == hipSuccess ? DZ_OK : DZ_ERR;
+}
+
+extern "C" dz_status dz_memcpy_h2d(void* dst, const void* src, size_t bytes) {
+    return hipMemcpy(dst, src, bytes, hipMemcpyHostToDevice) == hipSuccess ? DZ_OK : DZ_ERR;
+}

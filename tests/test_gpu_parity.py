@@ -1,0 +1,392 @@
+"""GPU parity: the HIP path through the C ABI vs the CPU oracle on identical
+seeded inputs. Bar (BASELINE.json north_star): count/min/max bit-exact, avg
+within 1 ulp — this implementation preserves per-group row order, so ALL
+aggregates including avg/sum are asserted BIT-EXACT (tolerance 0; stricter
+than the stated 1-ulp bar)."""
+import numpy as np
+import pytest
+
+import __graft_entry__ as graft
+from oracle import pyoracle
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module", autouse=True)
+def built():
+    graft.build()
+
+
+def make_op(len_ms, slide_ms=0, key_kind=None, n_keys_hint=64, **kw):
+    from denormalized_amd import WindowOp, _lib
+    return WindowOp(length_ms=len_ms, slide_ms=slide_ms,
+                    key_kind=_lib.KEY_INT64 if key_kind is None else key_kind,
+                    n_keys_hint=n_keys_hint, **kw)
+
+
+def cat(outs, field):
+    arrs = [b[field] for b in outs]
+    if arrs and isinstance(arrs[0], list):
+        return [x for a in arrs for x in a]
+    return np.concatenate(arrs) if arrs else np.zeros(0)
+
+
+def assert_parity(outs, exp, utf8_keys=None):
+    """outs: list of emitted batches from the GPU op; exp: oracle fetch dict."""
+    n = len(exp["key"])
+    got_key = cat(outs, "key")
+    if utf8_keys is not None:
+        assert list(got_key) == [utf8_keys[k] for k in exp["key"]]
+    else:
+        assert np.array_equal(np.asarray(got_key), exp["key"]), "keys/order"
+    assert len(got_key) == n
+    assert np.array_equal(cat(outs, "count"), exp["count"]), "count"
+    assert np.array_equal(cat(outs, "valid"), exp["valid"]), "validity"
+    v = exp["valid"].astype(bool)
+    for f in ("min", "max", "avg"):
+        g = cat(outs, f)
+        assert np.array_equal(g[v], exp[f][v]), f"{f} not bit-exact"
+    assert np.array_equal(cat(outs, "window_start"), exp["window_start"])
+    assert np.array_equal(cat(outs, "window_end"), exp["window_end"])
+
+
+def run_both(len_ms, slide_ms, batches, key_kind=None, n_keys_hint=64,
+             valids=None, finish=True, **kw):
+    op = make_op(len_ms, slide_ms, key_kind=key_kind, n_keys_hint=n_keys_hint, **kw)
+    o = pyoracle.Oracle(len_ms, slide_ms)
+    outs = []
+    for bi, (ts, k, v) in enumerate(batches):
+        bm = None
+        vv = None
+        if valids is not None and valids[bi] is not None:
+            vv = valids[bi]
+            bm = np.packbits(vv.astype(np.uint8), bitorder="little")
+        op.push(ts, k, v, bm)
+        outs += op.poll_all()
+        o.push(ts, k, v, vv)
+    if finish:
+        op.finish()
+        outs += op.poll_all()
+        o.finish()
+    exp = o.fetch()
+    op.close()
+    o.close()
+    return outs, exp
+
+
+def gen_batches(seed, nbatches, rows, nkeys, rows_per_ms, t0=1_000_000):
+    out = []
+    for b in range(nbatches):
+        out.append(pyoracle.gen(seed, t0, b * rows, rows, nkeys, rows_per_ms))
+    return out
+
+
+# ----------------------------------------------------------------- basics
+
+def test_tumbling_small():
+    batches = gen_batches(1, 4, 10_000, 37, 10)
+    outs, exp = run_both(1000, 0, batches)
+    assert len(exp["key"]) > 0
+    assert_parity(outs, exp)
+
+
+def test_tumbling_5s_reference_example_shape():
+    # simple_aggregation.rs:53 uses a 5s tumbling window
+    batches = gen_batches(2, 3, 50_000, 10, 10)
+    outs, exp = run_both(5000, 0, batches)
+    assert_parity(outs, exp)
+
+
+def test_snap_quirk_1500ms():
+    batches = gen_batches(3, 2, 20_000, 11, 20)
+    outs, exp = run_both(1500, 0, batches)
+    assert_parity(outs, exp)
+
+
+def test_subsecond_window_500ms():
+    batches = gen_batches(4, 2, 20_000, 13, 20)
+    outs, exp = run_both(500, 0, batches)
+    assert_parity(outs, exp)
+
+
+def test_sliding_2000_500():
+    batches = gen_batches(5, 4, 15_000, 23, 15)
+    outs, exp = run_both(2000, 500, batches)
+    assert len(exp["key"]) > 0
+    assert_parity(outs, exp)
+
+
+def test_sliding_500_100_cfg3_shape():
+    batches = gen_batches(6, 3, 30_000, 41, 40)
+    outs, exp = run_both(500, 100, batches)
+    assert_parity(outs, exp)
+
+
+def test_sliding_with_gaps_slide_gt_len():
+    # slide > length leaves uncovered rows (legal; m == 0 for those rows)
+    batches = gen_batches(7, 2, 10_000, 7, 10)
+    outs, exp = run_both(500, 1500, batches)
+    assert_parity(outs, exp)
+
+
+# ----------------------------------------------------------- key handling
+
+def test_utf8_keys_insertion_order():
+    rng = np.random.default_rng(8)
+    names = [f"sensor_{i}" for i in range(25)]
+    batches = []
+    for b in range(3):
+        n = 5000
+        ts = 1_000_000 + np.arange(b * n, (b + 1) * n) // 10
+        kid = rng.integers(0, 25, n)
+        v = rng.uniform(0, 115, n)
+        batches.append((ts.astype(np.int64), kid, v))
+    from denormalized_amd import _lib
+    op = make_op(1000, key_kind=_lib.KEY_UTF8)
+    o = pyoracle.Oracle(1000, 0)
+    outs = []
+    for ts, kid, v in batches:
+        op.push(ts, [names[k] for k in kid], v)
+        outs += op.poll_all()
+        o.push(ts, kid, v)
+    op.finish()
+    outs += op.poll_all()
+    o.finish()
+    exp = o.fetch()
+    assert_parity(outs, exp, utf8_keys=names)
+    op.close()
+    o.close()
+
+
+def test_sparse_int64_keys_host_dict():
+    rng = np.random.default_rng(9)
+    keyspace = rng.integers(-2**62, 2**62, 50)  # sparse, negative included
+    batches = []
+    for b in range(3):
+        n = 4000
+        ts = 1_000_000 + np.arange(b * n, (b + 1) * n) // 8
+        k = keyspace[rng.integers(0, 50, n)]
+        v = rng.uniform(-5, 120, n)
+        batches.append((ts.astype(np.int64), k.astype(np.int64), v))
+    outs, exp = run_both(1000, 0, batches)
+    assert_parity(outs, exp)
+
+
+def test_key_growth_beyond_hint():
+    # dict grows past n_keys_hint (and past NB buckets' first kloc)
+    rng = np.random.default_rng(10)
+    n = 200_000
+    ts = 1_000_000 + np.arange(n) // 100
+    k = rng.integers(0, 9000, n)
+    v = rng.uniform(0, 115, n)
+    outs, exp = run_both(1000, 0, [(ts.astype(np.int64), k, v)],
+                         n_keys_hint=16)
+    assert_parity(outs, exp)
+
+
+# --------------------------------------------------------------- semantics
+
+def test_late_data_reemission():
+    ts1 = np.array([1_000_000, 1_000_500], np.int64)
+    ts2 = np.array([1_002_500], np.int64)
+    ts3 = np.array([1_000_700], np.int64)  # late: window re-created, re-emitted
+    k = np.array([1, 2], np.int64)
+    batches = [(ts1, k, np.array([1.0, 2.0])),
+               (ts2, k[:1], np.array([3.0])),
+               (ts3, k[:1], np.array([9.0]))]
+    outs, exp = run_both(1000, 0, batches)
+    assert_parity(outs, exp)
+    assert (np.asarray(exp["window_start"]) == 1_000_000).sum() >= 2  # re-emit
+
+
+def test_null_values_bitmap():
+    rng = np.random.default_rng(11)
+    n = 30_000
+    ts = 1_000_000 + np.arange(n) // 30
+    k = rng.integers(0, 19, n)
+    v = rng.uniform(0, 115, n)
+    valid = (rng.random(n) > 0.3)
+    outs, exp = run_both(1000, 0, [(ts.astype(np.int64), k, v)],
+                         valids=[valid])
+    assert_parity(outs, exp)
+    assert (exp["valid"] == 0).sum() >= 0  # may contain all-null groups
+
+
+def test_boundary_timestamps():
+    ts = np.array([999_999, 1_000_000, 1_000_999, 1_001_000, 1_001_001], np.int64)
+    k = np.zeros(5, np.int64)
+    v = np.arange(5, dtype=np.float64)
+    outs, exp = run_both(1000, 0, [(ts, k, v)])
+    assert_parity(outs, exp)
+
+
+def test_emission_timing_matches_oracle():
+    # windows must emit exactly when the watermark (max of batch minimums)
+    # passes their end — compare emitted counts batch by batch
+    from denormalized_amd import _lib
+    op = make_op(1000)
+    o = pyoracle.Oracle(1000, 0)
+    rng = np.random.default_rng(12)
+    t0 = 1_000_000
+    for b in range(6):
+        n = 2000
+        ts = t0 + np.sort(rng.integers(0, 900, n)) + b * 700
+        k = rng.integers(0, 9, n)
+        v = rng.uniform(0, 115, n)
+        op.push(ts.astype(np.int64), k, v)
+        o.push(ts.astype(np.int64), k, v)
+        got = op.poll_all()
+        exp = o.fetch()
+        assert_parity(got, exp)
+        assert op.watermark == o.watermark
+        assert op.open_windows == o.open_frames
+    op.close()
+    o.close()
+
+
+def test_multiple_aggs_of_same_kind_and_sum():
+    from denormalized_amd import WindowOp, _lib
+    rng = np.random.default_rng(13)
+    n = 10_000
+    ts = (1_000_000 + np.arange(n) // 10).astype(np.int64)
+    k = rng.integers(0, 21, n)
+    v = rng.uniform(0, 115, n)
+    op = WindowOp(length_ms=1000, aggs=[("sum", 0), ("count", 0), ("avg", 0)],
+                  key_kind=_lib.KEY_INT64, n_keys_hint=32)
+    op.push(ts, k, v)
+    op.finish()
+    outs = op.poll_all()
+    o = pyoracle.Oracle(1000, 0)
+    o.push(ts, k, v)
+    o.finish()
+    exp = o.fetch()
+    assert np.array_equal(cat(outs, "sum"), exp["sum"])
+    assert np.array_equal(cat(outs, "count"), exp["count"])
+    assert np.array_equal(cat(outs, "avg"), exp["avg"])
+    op.close()
+    o.close()
+
+
+def test_filter_pushdown_max_gt():
+    # the BASELINE pipeline's filter(col("max") > 113)
+    batches = gen_batches(14, 3, 40_000, 29, 40)
+    op = make_op(1000)
+    op.set_filter("max", ">", 113.0)
+    o = pyoracle.Oracle(1000, 0)
+    outs = []
+    for ts, k, v in batches:
+        op.push(ts, k, v)
+        outs += op.poll_all()
+        o.push(ts, k, v)
+    op.finish()
+    outs += op.poll_all()
+    o.finish()
+    exp = o.fetch()
+    keep = exp["max"] > 113.0
+    assert np.array_equal(np.asarray(cat(outs, "key")), exp["key"][keep])
+    assert np.array_equal(cat(outs, "count"), exp["count"][keep])
+    assert np.array_equal(cat(outs, "avg"), exp["avg"][keep])
+    op.close()
+    o.close()
+
+
+def test_empty_and_single_row_batches():
+    batches = [(np.array([1_000_000], np.int64), np.array([5], np.int64),
+                np.array([7.5])),
+               (np.array([], np.int64), np.array([], np.int64),
+                np.array([], np.float64)),
+               (np.array([1_002_000], np.int64), np.array([5], np.int64),
+                np.array([1.5]))]
+    outs, exp = run_both(1000, 0, batches)
+    assert_parity(outs, exp)
+
+
+def test_watermark_injection():
+    op = make_op(1000)
+    ts = np.array([1_000_000, 1_000_100], np.int64)
+    op.push(ts, np.array([1, 2], np.int64), np.array([1.0, 2.0]))
+    assert op.poll_all() == []
+    op.advance_watermark(1_001_000)  # external (multi-shard) watermark
+    outs = op.poll_all()
+    assert len(outs) == 1 and outs[0]["n_rows"] == 2
+    op.close()
+
+
+# --------------------------------------------------- device-resident path
+
+def test_device_generator_matches_oracle_gen():
+    from denormalized_amd import DeviceArray, generate, synchronize
+    n = 1_000_000
+    d_ts = DeviceArray(0, n * 8)
+    d_keys = DeviceArray(0, n * 8)
+    d_kid = DeviceArray(0, n * 4)
+    d_vals = DeviceArray(0, n * 8)
+    generate(0, 42, 1_000_000, 0, n, 10_000, 1000, d_ts.ptr, d_keys.ptr,
+             d_kid.ptr, d_vals.ptr)
+    synchronize(0)
+    ts, kid, val = pyoracle.gen(42, 1_000_000, 0, n, 10_000, 1000)
+    assert np.array_equal(d_ts.to_host(np.int64, n), ts)
+    assert np.array_equal(d_keys.to_host(np.int64, n), kid)
+    assert np.array_equal(d_kid.to_host(np.int32, n), kid.astype(np.int32))
+    assert np.array_equal(d_vals.to_host(np.float64, n), val)
+    for a in (d_ts, d_keys, d_kid, d_vals):
+        a.free()
+
+
+def test_push_device_dense_parity_1M():
+    from denormalized_amd import DeviceArray, WindowOp, _lib, generate, synchronize
+    n = 1_000_000
+    nkeys = 2000
+    d_ts = DeviceArray(0, n * 8)
+    d_kid = DeviceArray(0, n * 4)
+    d_vals = DeviceArray(0, n * 8)
+    generate(0, 99, 5_000_000, 0, n, nkeys, 1000, d_ts.ptr, None, d_kid.ptr,
+             d_vals.ptr)
+    synchronize(0)
+    op = WindowOp(length_ms=1000, key_kind=_lib.KEY_DENSE_INT64,
+                  n_keys_hint=nkeys)
+    half = n // 2
+    itemsz = {"ts": 8, "kid": 4, "vals": 8}
+    import ctypes
+    op.push_device(half, d_ts.ptr, d_kid.ptr, d_vals.ptr)
+    op.push_device(half,
+                   ctypes.c_void_p(d_ts.ptr.value + half * 8),
+                   ctypes.c_void_p(d_kid.ptr.value + half * 4),
+                   ctypes.c_void_p(d_vals.ptr.value + half * 8))
+    op.finish()
+    outs = op.poll_all()
+    stats = op.kernel_stats()
+    assert stats["fold"]["launches"] >= 2 and stats["scatter"]["total_ms"] > 0
+
+    ts, kid, val = pyoracle.gen(99, 5_000_000, 0, n, nkeys, 1000)
+    o = pyoracle.Oracle(1000, 0)
+    o.push(ts[:half], kid[:half], val[:half])
+    o.push(ts[half:], kid[half:], val[half:])
+    o.finish()
+    exp = o.fetch()
+    assert_parity(outs, exp)
+    op.close()
+    o.close()
+    for a in (d_ts, d_kid, d_vals):
+        a.free()
+
+
+def test_randomized_stress_matrix():
+    rng = np.random.default_rng(1234)
+    for case in range(6):
+        len_ms = int(rng.choice([500, 1000, 1500, 2000]))
+        slide = int(rng.choice([0, 0, 250, 500]))
+        if slide > len_ms:
+            slide = 0
+        nkeys = int(rng.choice([1, 3, 64, 500]))
+        batches = []
+        t0 = 1_000_000
+        for b in range(int(rng.integers(2, 5))):
+            n = int(rng.integers(100, 20_000))
+            ts = t0 + np.cumsum(rng.integers(0, 3, n))
+            t0 = int(ts.max())
+            k = rng.integers(0, nkeys, n)
+            v = np.round(rng.uniform(-10, 115, n), 6)
+            batches.append((ts.astype(np.int64), k, v))
+        outs, exp = run_both(len_ms, slide, batches)
+        assert_parity(outs, exp)
