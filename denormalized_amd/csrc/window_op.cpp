@@ -310,7 +310,7 @@ extern "C" dz_window_op* dz_window_op_create(const dz_window_desc* desc) {
     op->key_kind = desc->key_kind;
     op->aggs.assign(desc->aggs, desc->aggs + desc->n_aggs);
     op->device = desc->device;
-    op->max_open = desc->max_open_windows > 0 ? desc->max_open_windows : 64;
+    op->max_open = desc->max_open_windows > 0 ? desc->max_open_windows : 4096;
     if (hipSetDevice(op->device) != hipSuccess ||
         hipStreamCreate(&op->stream) != hipSuccess) {
         g_err = "hip device/stream init failed";
