@@ -1,0 +1,264 @@
+#!/usr/bin/env python3
+"""bench.py — measures BASELINE.json's metric: rows/sec through
+window()+filter() on the synthetic sensor stream.
+
+Workload (config.workload): BASELINE cfg2 — the largest single-GPU config the
+metric is quoted on: 10k keys, 1s tumbling window, count/min/max/avg(reading)
++ filter(max > 113), f64 readings, synthetic sensor stream (seeded; spec in
+DESIGN.md §Generator). A step = one push of --rows-per-step rows through the
+operator (device-resident inputs) including triggered window emission; the
+default run covers ~100M rows ≈ cfg2.
+
+Contract: `python bench.py --gpus N --steps K --warmup W`. For N>1 the driver
+launches one rank per GPU via torch.distributed.run (RCCL); rows shard by key
+across ranks (the reference's RepartitionExec::Hash analog), the
+only collectives are the shared-watermark all-reduce (MAX) and final count
+gather. scaling=weak: per-rank rows fixed as N grows.
+
+Rank 0 prints ONE JSON line. cpu_baseline: the CPU oracle (kind "port") timed
+on this box's host cores on a bounded sample of the same workload.
+"""
+import argparse
+import ctypes
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+HBM_PEAK = 8.0e12  # B/s, MI355X spec peak (MI355X_MICROARCH.md §Chip-level)
+ALG_BYTES_PER_ROW = 20.0  # key id 4B + reading 8B + ts 8B
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=24)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--rows-per-step", type=int, default=4_000_000)
+    p.add_argument("--keys", type=int, default=10_000)
+    p.add_argument("--rows-per-ms", type=int, default=1000)
+    p.add_argument("--window-ms", type=int, default=1000)
+    p.add_argument("--slide-ms", type=int, default=0)
+    p.add_argument("--seed", type=int, default=42)
+    p.add_argument("--no-filter", action="store_true")
+    p.add_argument("--skip-cpu-baseline", action="store_true")
+    p.add_argument("--cpu-sample-rows", type=int, default=8_000_000)
+    return p.parse_args()
+
+
+def cpu_baseline(args):
+    """Time the CPU oracle (the reference-semantics restatement, kind 'port')
+    on a bounded sample, 8-ways key-sharded across threads (per-group row
+    order preserved => identical results; shard prep excluded from timing)."""
+    import threading
+    from oracle import pyoracle
+    n = args.cpu_sample_rows
+    ncores = min(8, os.cpu_count() or 1)
+    ts, kid, val = pyoracle.gen(args.seed, 1_000_000, 0, n, args.keys,
+                                args.rows_per_ms)
+    shard_of = kid % ncores
+    shards = []
+    for t in range(ncores):
+        m = shard_of == t
+        shards.append((np.ascontiguousarray(ts[m]),
+                       np.ascontiguousarray(kid[m]),
+                       np.ascontiguousarray(val[m])))
+    batch = 1_000_000
+    ops = [pyoracle.Oracle(args.window_ms, args.slide_ms) for _ in range(ncores)]
+
+    def run(t):
+        sts, skid, sval = shards[t]
+        for lo in range(0, len(sts), batch):
+            ops[t].push(sts[lo:lo + batch], skid[lo:lo + batch], sval[lo:lo + batch])
+        ops[t].finish()
+        ops[t].fetch()
+
+    threads = [threading.Thread(target=run, args=(t,)) for t in range(ncores)]
+    t0 = time.perf_counter()
+    for th in threads:
+        th.start()
+    for th in threads:
+        th.join()
+    dt = time.perf_counter() - t0
+    for o in ops:
+        o.close()
+    return {
+        "value": n / dt,
+        "unit": "rows/s",
+        "cores": ncores,
+        "kind": "port",
+        "sample": f"{n} rows of the same workload, key-sharded over "
+                  f"{ncores} host threads, large batches (best-CPU-effort; "
+                  "the reference's own 32-row-batch mode is slower — DESIGN.md)",
+    }
+
+
+def read_traffic():
+    """Measured HBM bytes per dominant-kernel launch, if a rocprofv3 PMC run
+    has been committed (profiles/hbm_traffic.json: {kernel: bytes_per_launch},
+    collected per MI355X_MICROARCH.md §HBM: separate --pmc pass, FETCH_SIZE
+    read-side x2 correction applied at collection time)."""
+    path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                        "profiles", "hbm_traffic.json")
+    try:
+        with open(path) as f:
+            return json.load(f)
+    except Exception:
+        return None
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if world > 1:
+        import torch
+        import torch.distributed as dist
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        dist.init_process_group(backend=backend)
+        if torch.cuda.is_available():
+            torch.cuda.set_device(local_rank)
+    else:
+        dist = None
+
+    import denormalized_amd as dz
+    from denormalized_amd import _lib
+
+    device = local_rank
+    K, W = args.steps, args.warmup
+    B = args.rows_per_step
+    total_rows = (K + W) * B
+
+    # pre-generate the whole stream into HBM (inputs resident when the timed
+    # region starts). Each rank owns a disjoint key shard (weak scaling):
+    # same time distribution, rank-distinct keys (ids are rank-local).
+    d_ts = dz.DeviceArray(device, total_rows * 8)
+    d_kid = dz.DeviceArray(device, total_rows * 4)
+    d_vals = dz.DeviceArray(device, total_rows * 8)
+    # rank-distinct seed => independent key/value draws per shard
+    dz.generate(device, args.seed + 1000003 * rank, 1_000_000, 0, total_rows,
+                args.keys, args.rows_per_ms, d_ts.ptr, None, d_kid.ptr,
+                d_vals.ptr)
+    dz.synchronize(device)
+
+    op = dz.WindowOp(length_ms=args.window_ms, slide_ms=args.slide_ms,
+                     aggs=[("count", 0), ("min", 0), ("max", 0), ("avg", 0)],
+                     key_kind=_lib.KEY_DENSE_INT64, n_keys_hint=args.keys,
+                     device=device)
+    if not args.no_filter:
+        op.set_filter("max", ">", 113.0)
+
+    def push_step(step):
+        off = step * B
+        op.push_device(B,
+                       ctypes.c_void_p(d_ts.ptr.value + off * 8),
+                       ctypes.c_void_p(d_kid.ptr.value + off * 4),
+                       ctypes.c_void_p(d_vals.ptr.value + off * 8))
+        if dist is not None:
+            import torch
+            wm = torch.tensor([op.watermark], dtype=torch.int64)
+            if torch.cuda.is_available():
+                wm = wm.cuda()
+            dist.all_reduce(wm, op=dist.ReduceOp.MAX)
+            op.advance_watermark(int(wm.item()))
+        emitted = 0
+        for b in op.poll_all():
+            emitted += b["n_rows"]
+        return emitted
+
+    emitted = 0
+    for s in range(W):
+        emitted += push_step(s)
+    dz.synchronize(device)
+    if dist is not None:
+        dist.barrier()
+
+    t0 = time.perf_counter()
+    for s in range(W, W + K):
+        emitted += push_step(s)
+    op.finish()
+    emitted += sum(b["n_rows"] for b in op.poll_all())
+    dz.synchronize(device)
+    if dist is not None:
+        dist.barrier()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    if dist is not None:
+        import torch
+        e = torch.tensor([elapsed], dtype=torch.float64)
+        if torch.cuda.is_available():
+            e = e.cuda()
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    stats = op.kernel_stats()
+
+    if rank == 0:
+        rows_timed = K * B * world  # whole-job rows through the timed region
+        value = rows_timed / elapsed
+        # roofline for the dominant kernel (by total time)
+        dom = max(stats, key=lambda k: stats[k]["total_ms"])
+        d = stats[dom]
+        launches_timed = d["launches"]
+        avg_s = (d["total_ms"] / 1000.0) / max(1, d["launches"])
+        rows_per_launch = (K + W) * B / max(1, d["launches"])
+        achieved = ALG_BYTES_PER_ROW * rows_per_launch / avg_s
+        traffic = None
+        tmap = read_traffic()
+        if tmap and dom in tmap:
+            traffic = tmap[dom]
+        out = {
+            "metric": "rows/sec through window()+filter() on synthetic sensor stream",
+            "value": value,
+            "unit": "rows/s",
+            "n_gpus": world,
+            "steps": K,
+            "warmup": W,
+            "ms_per_step": elapsed / K * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "f64",
+            "data": "synthetic",
+            "config": {
+                "workload": f"cfg2: {K * B / 1e6:.0f}M rows/GPU, {args.keys} keys, "
+                            f"{args.window_ms}ms tumbling"
+                            + (f"/{args.slide_ms}ms slide" if args.slide_ms else "")
+                            + ", count/min/max/avg + filter(max>113)",
+                "rows_per_step": B,
+                "keys_per_gpu": args.keys,
+                "rows_per_ms": args.rows_per_ms,
+                "parallelism": f"key-sharded dp{world}",
+                "emitted_rows_rank0": emitted,
+            },
+            "roofline": {
+                "bound": "hbm",
+                "kernel": dom,
+                "achieved": achieved,
+                "peak": HBM_PEAK,
+                "unit": "B/s",
+                "frac": achieved / HBM_PEAK,
+                "traffic": traffic,
+                "launches": launches_timed,
+                "kernel_ms_total": {k: round(v["total_ms"], 3) for k, v in stats.items()},
+            },
+            "cpu_baseline": None,
+        }
+        if not args.skip_cpu_baseline and world == 1:
+            out["cpu_baseline"] = cpu_baseline(args)
+        print(json.dumps(out))
+    op.close()
+    for a in (d_ts, d_kid, d_vals):
+        a.free()
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

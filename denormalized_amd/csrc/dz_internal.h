@@ -52,8 +52,9 @@ void launch_hist(hipStream_t stream, const int32_t* d_kid, const int64_t* d_ts,
                  int64_t n, int64_t chunk, int C, const WinParams& wp,
                  uint32_t* d_ghist);
 
+constexpr int SCAN_SSPLIT = 32; /* d_psum is SCAN_SSPLIT * NB u32 */
 void launch_scan(hipStream_t stream, const uint32_t* d_ghist, int C,
-                 uint32_t* d_total, uint32_t* d_bucket_base /*NB+1*/,
+                 uint32_t* d_psum, uint32_t* d_bucket_base /*NB+1*/,
                  uint32_t* d_gofs);
 
 void launch_scatter(hipStream_t stream, const int32_t* d_kid, const int64_t* d_ts,

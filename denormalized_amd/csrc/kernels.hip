@@ -109,8 +109,9 @@ __device__ __forceinline__ uint64_t map_i64(int64_t x) {
     return (uint64_t)x ^ 0x8000000000000000ULL; /* order-preserving i64->u64 */
 }
 
-__global__ void k_minmax(const int64_t* ts, const int32_t* kid, int64_t n,
-                         uint64_t* scalars) {
+__global__ __launch_bounds__(BLOCK) void k_minmax(const int64_t* ts,
+        const int32_t* kid, int64_t n, uint64_t* scalars) {
+    __shared__ uint64_t red[3][WAVES_PER_BLOCK];
     uint64_t mn = ~0ULL, mx = 0, km = 0;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n; i += stride) {
@@ -125,7 +126,21 @@ __global__ void k_minmax(const int64_t* ts, const int32_t* kid, int64_t n,
         mx = max(mx, (uint64_t)__shfl_down((unsigned long long)mx, o));
         km = max(km, (uint64_t)__shfl_down((unsigned long long)km, o));
     }
+    /* block reduce: one atomic triple per block (device atomics are ~10ns
+     * each serialised on a word — keep their count per launch small) */
+    const int wave = threadIdx.x >> 6;
     if ((threadIdx.x & 63) == 0) {
+        red[0][wave] = mn;
+        red[1][wave] = mx;
+        red[2][wave] = km;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        for (int w = 1; w < WAVES_PER_BLOCK; w++) {
+            mn = min(mn, red[0][w]);
+            mx = max(mx, red[1][w]);
+            km = max(km, red[2][w]);
+        }
         atomicMin((unsigned long long*)&scalars[0], (unsigned long long)mn);
         atomicMax((unsigned long long*)&scalars[1], (unsigned long long)mx);
         atomicMax((unsigned long long*)&scalars[2], (unsigned long long)km);
@@ -134,7 +149,7 @@ __global__ void k_minmax(const int64_t* ts, const int32_t* kid, int64_t n,
 
 void launch_minmax(hipStream_t s, const int64_t* d_ts, const int32_t* d_kid,
                    int64_t n, uint64_t* d_scalars) {
-    int blocks = (int)std::min<int64_t>((n + BLOCK - 1) / BLOCK, 2048);
+    int blocks = (int)std::min<int64_t>((n + BLOCK - 1) / BLOCK, 512);
     if (blocks < 1) blocks = 1;
     hipLaunchKernelGGL(k_minmax, dim3(blocks), dim3(BLOCK), 0, s, d_ts, d_kid, n,
                        d_scalars);
@@ -173,25 +188,35 @@ void launch_hist(hipStream_t s, const int32_t* d_kid, const int64_t* d_ts,
 /* scans: bucket totals -> bases; per-chunk stable offsets             */
 /* ------------------------------------------------------------------ */
 
-/* one wave per 64 buckets, coalesced column sweep over chunks */
-__global__ void k_scan_totals(const uint32_t* ghist, int C, uint32_t* total) {
+/* Two-level scan over the [C chunks][NB buckets] histogram matrix:
+ * SSPLIT chunk-segments give 512+ blocks of parallelism at every stage. */
+constexpr int SSPLIT = SCAN_SSPLIT;
+
+/* partial per-segment bucket sums: psum[s][bkt] = sum of ghist over segment s */
+__global__ void k_scan_partial(const uint32_t* ghist, int C, int cs,
+                               uint32_t* psum) {
     int bkt = blockIdx.x * blockDim.x + threadIdx.x;
+    int seg = blockIdx.y;
     if (bkt >= NB) return;
+    int c0 = seg * cs, c1 = min(C, c0 + cs);
     uint32_t s = 0;
-    for (int c = 0; c < C; c++) s += ghist[(int64_t)c * NB + bkt];
-    total[bkt] = s;
+    for (int c = c0; c < c1; c++) s += ghist[(int64_t)c * NB + bkt];
+    psum[(int64_t)seg * NB + bkt] = s;
 }
 
-/* single block: exclusive scan of NB totals -> bucket_base[NB+1] */
-__global__ __launch_bounds__(1024) void k_scan_base(const uint32_t* total,
+/* single block: bucket totals from psum + exclusive scan -> base[NB+1] */
+__global__ __launch_bounds__(1024) void k_scan_base(const uint32_t* psum,
                                                     uint32_t* base) {
     __shared__ uint32_t part[1024];
     constexpr int PER = NB / 1024;
     uint32_t loc[PER];
     uint32_t s = 0;
     for (int j = 0; j < PER; j++) {
+        int bkt = threadIdx.x * PER + j;
+        uint32_t t = 0;
+        for (int g = 0; g < SSPLIT; g++) t += psum[(int64_t)g * NB + bkt];
         loc[j] = s;
-        s += total[threadIdx.x * PER + j];
+        s += t;
     }
     part[threadIdx.x] = s;
     __syncthreads();
@@ -208,13 +233,18 @@ __global__ __launch_bounds__(1024) void k_scan_base(const uint32_t* total,
     if (threadIdx.x == 1023) base[NB] = part[1023];
 }
 
-/* per-chunk running offsets: gofs[c][bkt] = base[bkt] + sum_{c'<c} ghist[c'][bkt] */
-__global__ void k_scan_offsets(const uint32_t* ghist, int C,
-                               const uint32_t* base, uint32_t* gofs) {
+/* per-chunk running offsets within each segment:
+ * gofs[c][bkt] = base[bkt] + psum[<seg][bkt] + ghist[[c0,c)][bkt] */
+__global__ void k_scan_offsets(const uint32_t* ghist, int C, int cs,
+                               const uint32_t* psum, const uint32_t* base,
+                               uint32_t* gofs) {
     int bkt = blockIdx.x * blockDim.x + threadIdx.x;
+    int seg = blockIdx.y;
     if (bkt >= NB) return;
     uint32_t run = base[bkt];
-    for (int c = 0; c < C; c++) {
+    for (int g = 0; g < seg; g++) run += psum[(int64_t)g * NB + bkt];
+    int c0 = seg * cs, c1 = min(C, c0 + cs);
+    for (int c = c0; c < c1; c++) {
         uint32_t t = ghist[(int64_t)c * NB + bkt];
         gofs[(int64_t)c * NB + bkt] = run;
         run += t;
@@ -222,12 +252,13 @@ __global__ void k_scan_offsets(const uint32_t* ghist, int C,
 }
 
 void launch_scan(hipStream_t s, const uint32_t* d_ghist, int C,
-                 uint32_t* d_total, uint32_t* d_base, uint32_t* d_gofs) {
-    hipLaunchKernelGGL(k_scan_totals, dim3(NB / 256), dim3(256), 0, s, d_ghist, C,
-                       d_total);
-    hipLaunchKernelGGL(k_scan_base, dim3(1), dim3(1024), 0, s, d_total, d_base);
-    hipLaunchKernelGGL(k_scan_offsets, dim3(NB / 256), dim3(256), 0, s, d_ghist, C,
-                       d_base, d_gofs);
+                 uint32_t* d_psum, uint32_t* d_base, uint32_t* d_gofs) {
+    int cs = (C + SSPLIT - 1) / SSPLIT;
+    hipLaunchKernelGGL(k_scan_partial, dim3(NB / 256, SSPLIT), dim3(256), 0, s,
+                       d_ghist, C, cs, d_psum);
+    hipLaunchKernelGGL(k_scan_base, dim3(1), dim3(1024), 0, s, d_psum, d_base);
+    hipLaunchKernelGGL(k_scan_offsets, dim3(NB / 256, SSPLIT), dim3(256), 0, s,
+                       d_ghist, C, cs, d_psum, d_base, d_gofs);
 }
 
 /* ------------------------------------------------------------------ */

@@ -219,6 +219,16 @@ static void drain_events(dz_window_op* op, bool wait) {
     }
 }
 
+#include <chrono>
+struct HostTimer {
+    dz_window_op* op;
+    const char* name;
+    std::chrono::steady_clock::time_point t0;
+    HostTimer(dz_window_op* o, const char* n)
+        : op(o), name(n), t0(std::chrono::steady_clock::now()) {}
+    ~HostTimer();
+};
+
 template <typename F>
 static void timed(dz_window_op* op, const char* name, double bytes, F&& fn) {
     hipEvent_t a = get_event(op), b = get_event(op);
@@ -226,6 +236,14 @@ static void timed(dz_window_op* op, const char* name, double bytes, F&& fn) {
     fn();
     hipEventRecord(b, op->stream);
     op->pending.push_back({name, a, b, bytes});
+}
+
+HostTimer::~HostTimer() {
+    double ms = std::chrono::duration<double, std::milli>(
+                    std::chrono::steady_clock::now() - t0).count();
+    KStatAcc& s = op->stats[name];
+    s.launches++;
+    s.ms += ms;
 }
 
 /* ------------------------------------------------------------------ */
@@ -405,6 +423,7 @@ static dz_status emit_window(dz_window_op* op, int64_t wstart, int64_t wend,
         op->e_cap = nc;
     }
     if (K > 0) {
+        HostTimer ht(op, "h_emit_d2h");
         size_t base = (size_t)slot * op->kcap;
         CHK(op, hipMemcpyAsync(op->e_cnt, op->s_cnt + base, K * 8, hipMemcpyDeviceToHost, op->stream));
         CHK(op, hipMemcpyAsync(op->e_first, op->s_first + base, K * 8, hipMemcpyDeviceToHost, op->stream));
@@ -414,15 +433,19 @@ static dz_status emit_window(dz_window_op* op, int64_t wstart, int64_t wend,
         CHK(op, hipStreamSynchronize(op->stream));
         drain_events(op, false);
     }
+    HostTimer ht(op, "h_emit_build");
     /* groups in first-seen (insertion) order: GroupValues emits insertion
-     * order; we sort touched keys by first-row sequence (stable, exact). */
-    std::vector<int32_t> touched;
-    touched.reserve(1024);
+     * order; we sort touched keys by first-row sequence (stable, exact —
+     * first values are distinct rows, so plain sort is stable here). */
+    std::vector<std::pair<uint64_t, int32_t>> touched_p;
+    touched_p.reserve(4096);
     for (int64_t k = 0; k < K; k++)
-        if (op->e_first[k] != ~0ULL) touched.push_back((int32_t)k);
-    std::stable_sort(touched.begin(), touched.end(), [&](int32_t a, int32_t b) {
-        return op->e_first[a] < op->e_first[b];
-    });
+        if (op->e_first[k] != ~0ULL)
+            touched_p.emplace_back(op->e_first[k], (int32_t)k);
+    std::sort(touched_p.begin(), touched_p.end());
+    std::vector<int32_t> touched;
+    touched.reserve(touched_p.size());
+    for (auto& p : touched_p) touched.push_back(p.second);
 
     OutBuf ob;
     size_t na = op->aggs.size();
@@ -490,7 +513,7 @@ static dz_status ensure_scratch(dz_window_op* op, int C, int64_t nrec) {
         CHK(op, hipMalloc(&op->d_ghist, (size_t)C * dz::NB * 4));
         CHK(op, hipMalloc(&op->d_gofs, (size_t)C * dz::NB * 4));
         if (!op->d_total) {
-            CHK(op, hipMalloc(&op->d_total, dz::NB * 4));
+            CHK(op, hipMalloc(&op->d_total, (size_t)dz::SCAN_SSPLIT * dz::NB * 4));
             CHK(op, hipMalloc(&op->d_base, (dz::NB + 1) * 4));
         }
         op->C_cap = C;
@@ -520,7 +543,10 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
     });
     CHK(op, hipMemcpyAsync(op->h_scalars, op->d_scalars, 24, hipMemcpyDeviceToHost,
                            op->stream));
-    CHK(op, hipStreamSynchronize(op->stream));
+    {
+        HostTimer ht(op, "h_minmax_sync");
+        CHK(op, hipStreamSynchronize(op->stream));
+    }
     drain_events(op, false);
     int64_t mn = (int64_t)(op->h_scalars[0] ^ 0x8000000000000000ULL);
     int64_t mx = (int64_t)(op->h_scalars[1] ^ 0x8000000000000000ULL);
