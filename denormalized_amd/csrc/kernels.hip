@@ -370,7 +370,8 @@ __global__ __launch_bounds__(BLOCK) void k_fold(const uint32_t* rmeta,
     double mn = 0.0, mx = 0.0, sm = 0.0;
     if (own) {
         int64_t slot = slot_of_widx[my_widx];
-        sidx = slot * fc.kcap + (((int64_t)my_kloc << LOG_NB) | bkt);
+        /* state slabs: [slot][5 fields][kcap] (window_op.cpp) */
+        sidx = slot * (5 * fc.kcap) + (((int64_t)my_kloc << LOG_NB) | bkt);
         cnt = s_cnt[sidx];
         fst = s_first[sidx];
         if (cnt > 0) {

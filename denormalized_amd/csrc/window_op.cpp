@@ -109,9 +109,12 @@ struct dz_window_op {
     std::vector<int64_t> dict_vals;
     int64_t n_keys = 0; /* dense key count seen so far */
 
-    /* persistent group state [slot][kcap] per field */
+    /* persistent group state: one slab per slot, 5 contiguous fields of kcap
+     * 8-byte entries each ([slot][{cnt,first,min,max,sum}][kcap]) so a window
+     * close is ONE D2H copy. Field base pointers stride 5*kcap per slot. */
     int64_t kcap = 0; /* multiple of NB */
     int32_t nslots = 0;
+    uint64_t* s_base = nullptr; /* the slab allocation */
     uint64_t* s_cnt = nullptr;
     uint64_t* s_first = nullptr;
     double* s_min = nullptr;
@@ -149,13 +152,11 @@ struct dz_window_op {
     char* h_stage = nullptr; /* pinned */
     size_t h_stage_cap = 0;
 
-    /* pinned emission staging */
-    uint64_t* e_cnt = nullptr;
-    uint64_t* e_first = nullptr;
-    double* e_min = nullptr;
-    double* e_max = nullptr;
-    double* e_sum = nullptr;
-    int64_t e_cap = 0;
+    /* pinned emission staging: a small ring of whole-slot slabs so several
+     * window closes share one stream sync */
+    static constexpr int E_RING = 4;
+    uint64_t* e_ring[E_RING] = {nullptr, nullptr, nullptr, nullptr};
+    int64_t e_ring_kcap = 0;
 
     /* filter pushdown */
     bool has_filter = false;
@@ -260,33 +261,31 @@ static dz_status state_alloc(dz_window_op* op, int64_t kcap_new, int32_t nslots_
     }
     if (kcap_new == op->kcap && nslots_new <= op->nslots) return DZ_OK;
     nslots_new = std::max(nslots_new, op->nslots);
-    size_t cells = (size_t)kcap_new * nslots_new;
-    uint64_t* n_cnt; uint64_t* n_first; double* n_min; double* n_max; double* n_sum;
-    CHK(op, hipMalloc(&n_cnt, cells * 8));
-    CHK(op, hipMalloc(&n_first, cells * 8));
-    CHK(op, hipMalloc(&n_min, cells * 8));
-    CHK(op, hipMalloc(&n_max, cells * 8));
-    CHK(op, hipMalloc(&n_sum, cells * 8));
-    CHK(op, hipMemsetAsync(n_cnt, 0, cells * 8, op->stream));
-    CHK(op, hipMemsetAsync(n_first, 0xFF, cells * 8, op->stream));
-    /* min/max/sum need no init: fold ignores them while cnt == 0 */
+    size_t cells = (size_t)kcap_new * nslots_new * 5;
+    uint64_t* n_base;
+    CHK(op, hipMalloc(&n_base, cells * 8));
+    uint64_t* n_cnt = n_base;
+    uint64_t* n_first = n_base + kcap_new;
+    double* n_min = (double*)(n_base + 2 * kcap_new);
+    double* n_max = (double*)(n_base + 3 * kcap_new);
+    double* n_sum = (double*)(n_base + 4 * kcap_new);
+    size_t stride_new = (size_t)kcap_new * 5, stride_old = (size_t)op->kcap * 5;
+    for (int32_t s = 0; s < nslots_new; s++) {
+        CHK(op, hipMemsetAsync(n_cnt + s * stride_new, 0, kcap_new * 8, op->stream));
+        CHK(op, hipMemsetAsync(n_first + s * stride_new, 0xFF, kcap_new * 8, op->stream));
+        /* min/max/sum need no init: fold ignores them while cnt == 0 */
+    }
     for (auto& kv : op->open) { /* preserve open-window state */
         int32_t s = kv.second.slot;
         int64_t span = std::min(op->kcap, kcap_new);
-        CHK(op, hipMemcpyAsync(n_cnt + (size_t)s * kcap_new, op->s_cnt + (size_t)s * op->kcap,
-                               span * 8, hipMemcpyDeviceToDevice, op->stream));
-        CHK(op, hipMemcpyAsync(n_first + (size_t)s * kcap_new, op->s_first + (size_t)s * op->kcap,
-                               span * 8, hipMemcpyDeviceToDevice, op->stream));
-        CHK(op, hipMemcpyAsync(n_min + (size_t)s * kcap_new, op->s_min + (size_t)s * op->kcap,
-                               span * 8, hipMemcpyDeviceToDevice, op->stream));
-        CHK(op, hipMemcpyAsync(n_max + (size_t)s * kcap_new, op->s_max + (size_t)s * op->kcap,
-                               span * 8, hipMemcpyDeviceToDevice, op->stream));
-        CHK(op, hipMemcpyAsync(n_sum + (size_t)s * kcap_new, op->s_sum + (size_t)s * op->kcap,
-                               span * 8, hipMemcpyDeviceToDevice, op->stream));
+        for (int f = 0; f < 5; f++)
+            CHK(op, hipMemcpyAsync(n_base + s * stride_new + (size_t)f * kcap_new,
+                                   op->s_base + s * stride_old + (size_t)f * op->kcap,
+                                   span * 8, hipMemcpyDeviceToDevice, op->stream));
     }
     CHK(op, hipStreamSynchronize(op->stream));
-    hipFree(op->s_cnt); hipFree(op->s_first); hipFree(op->s_min);
-    hipFree(op->s_max); hipFree(op->s_sum);
+    hipFree(op->s_base);
+    op->s_base = n_base;
     op->s_cnt = n_cnt; op->s_first = n_first; op->s_min = n_min;
     op->s_max = n_max; op->s_sum = n_sum;
     for (int32_t s = op->nslots; s < nslots_new; s++) op->free_slots.push_back(s);
@@ -296,8 +295,9 @@ static dz_status state_alloc(dz_window_op* op, int64_t kcap_new, int32_t nslots_
 }
 
 static dz_status slot_reset(dz_window_op* op, int32_t slot) {
-    CHK(op, hipMemsetAsync(op->s_cnt + (size_t)slot * op->kcap, 0, op->kcap * 8, op->stream));
-    CHK(op, hipMemsetAsync(op->s_first + (size_t)slot * op->kcap, 0xFF, op->kcap * 8, op->stream));
+    size_t stride = (size_t)op->kcap * 5;
+    CHK(op, hipMemsetAsync(op->s_cnt + slot * stride, 0, op->kcap * 8, op->stream));
+    CHK(op, hipMemsetAsync(op->s_first + slot * stride, 0xFF, op->kcap * 8, op->stream));
     return DZ_OK;
 }
 
@@ -353,19 +353,15 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipStreamSynchronize(op->stream);
     drain_events(op, true);
     for (auto e : op->ev_pool) hipEventDestroy(e);
-    hipFree(op->s_cnt); hipFree(op->s_first); hipFree(op->s_min);
-    hipFree(op->s_max); hipFree(op->s_sum);
+    hipFree(op->s_base);
     hipFree(op->d_ghist); hipFree(op->d_gofs); hipFree(op->d_total); hipFree(op->d_base);
     hipFree(op->d_meta); hipFree(op->d_ridx); hipFree(op->d_rvals);
     hipFree(op->d_scalars); hipFree(op->d_slotmap);
     hipFree(op->d_ts); hipFree(op->d_kid); hipFree(op->d_vals); hipFree(op->d_valbm);
     if (op->h_scalars) hipHostFree(op->h_scalars);
     if (op->h_stage) hipHostFree(op->h_stage);
-    if (op->e_cnt) hipHostFree(op->e_cnt);
-    if (op->e_first) hipHostFree(op->e_first);
-    if (op->e_min) hipHostFree(op->e_min);
-    if (op->e_max) hipHostFree(op->e_max);
-    if (op->e_sum) hipHostFree(op->e_sum);
+    for (auto& r : op->e_ring)
+        if (r) hipHostFree(r);
     hipStreamDestroy(op->stream);
     delete op;
 }
@@ -379,9 +375,8 @@ extern "C" const char* dz_last_error(dz_window_op* op) {
 /* emission (trigger_windows, grouped_window_agg_stream.rs:220-253)    */
 /* ------------------------------------------------------------------ */
 
-static bool filter_pass(dz_window_op* op, const OutBuf& ob, int64_t row_cnt,
+static bool filter_pass(dz_window_op* op, int64_t row_cnt,
                         double vmin, double vmax, double vsum, bool valid) {
-    (void)ob;
     if (!op->has_filter) return true;
     double v;
     dz_agg_op o = op->aggs[op->f_idx].op;
@@ -408,96 +403,150 @@ static bool filter_pass(dz_window_op* op, const OutBuf& ob, int64_t row_cnt,
     }
 }
 
-static dz_status emit_window(dz_window_op* op, int64_t wstart, int64_t wend,
-                             int32_t slot) {
-    int64_t K = op->n_keys;
-    if (K > op->e_cap) {
-        int64_t nc = std::max<int64_t>(K, op->e_cap ? op->e_cap * 2 : 4096);
-        if (op->e_cnt) { hipHostFree(op->e_cnt); hipHostFree(op->e_first);
-                         hipHostFree(op->e_min); hipHostFree(op->e_max); hipHostFree(op->e_sum); }
-        CHK(op, hipHostMalloc((void**)&op->e_cnt, nc * 8));
-        CHK(op, hipHostMalloc((void**)&op->e_first, nc * 8));
-        CHK(op, hipHostMalloc((void**)&op->e_min, nc * 8));
-        CHK(op, hipHostMalloc((void**)&op->e_max, nc * 8));
-        CHK(op, hipHostMalloc((void**)&op->e_sum, nc * 8));
-        op->e_cap = nc;
-    }
-    if (K > 0) {
-        HostTimer ht(op, "h_emit_d2h");
-        size_t base = (size_t)slot * op->kcap;
-        CHK(op, hipMemcpyAsync(op->e_cnt, op->s_cnt + base, K * 8, hipMemcpyDeviceToHost, op->stream));
-        CHK(op, hipMemcpyAsync(op->e_first, op->s_first + base, K * 8, hipMemcpyDeviceToHost, op->stream));
-        CHK(op, hipMemcpyAsync(op->e_min, op->s_min + base, K * 8, hipMemcpyDeviceToHost, op->stream));
-        CHK(op, hipMemcpyAsync(op->e_max, op->s_max + base, K * 8, hipMemcpyDeviceToHost, op->stream));
-        CHK(op, hipMemcpyAsync(op->e_sum, op->s_sum + base, K * 8, hipMemcpyDeviceToHost, op->stream));
-        CHK(op, hipStreamSynchronize(op->stream));
-        drain_events(op, false);
-    }
+/* Build one emitted batch from a pinned copy of a slot slab
+ * ([cnt][first][min][max][sum], each `kcap` 8-byte entries). */
+static dz_status build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
+                                const uint64_t* slab) {
     HostTimer ht(op, "h_emit_build");
+    int64_t K = op->n_keys;
+    int64_t kcap = op->e_ring_kcap;
+    const uint64_t* f_cnt = slab;
+    const uint64_t* f_first = slab + kcap;
+    const double* f_min = (const double*)(slab + 2 * kcap);
+    const double* f_max = (const double*)(slab + 3 * kcap);
+    const double* f_sum = (const double*)(slab + 4 * kcap);
+
     /* groups in first-seen (insertion) order: GroupValues emits insertion
-     * order; we sort touched keys by first-row sequence (stable, exact —
-     * first values are distinct rows, so plain sort is stable here). */
-    std::vector<std::pair<uint64_t, int32_t>> touched_p;
-    touched_p.reserve(4096);
+     * order; sort touched keys by first-row sequence (first values are
+     * distinct rows, so plain sort preserves the exact order). */
+    std::vector<std::pair<uint64_t, int32_t>> touched;
+    touched.reserve(4096);
     for (int64_t k = 0; k < K; k++)
-        if (op->e_first[k] != ~0ULL)
-            touched_p.emplace_back(op->e_first[k], (int32_t)k);
-    std::sort(touched_p.begin(), touched_p.end());
-    std::vector<int32_t> touched;
-    touched.reserve(touched_p.size());
-    for (auto& p : touched_p) touched.push_back(p.second);
+        if (f_first[k] != ~0ULL) touched.emplace_back(f_first[k], (int32_t)k);
+    std::sort(touched.begin(), touched.end());
+
+    /* filter pushdown (datastream.rs:94-105) — keep list */
+    std::vector<int32_t> rows;
+    rows.reserve(touched.size());
+    for (auto& p : touched) {
+        int32_t k = p.second;
+        int64_t cnt = (int64_t)f_cnt[k];
+        if (filter_pass(op, cnt, f_min[k], f_max[k], f_sum[k], cnt > 0))
+            rows.push_back(k);
+    }
+    size_t n = rows.size(), na = op->aggs.size();
 
     OutBuf ob;
-    size_t na = op->aggs.size();
     ob.agg_i64.resize(na);
     ob.agg_f64.resize(na);
-    bool utf8 = op->key_kind == DZ_KEY_UTF8;
-    if (utf8) ob.key_offsets.push_back(0);
-    for (int32_t k : touched) {
-        int64_t cnt = (int64_t)op->e_cnt[k];
-        bool valid = cnt > 0;
-        double mn = op->e_min[k], mx = op->e_max[k], sm = op->e_sum[k];
-        if (!filter_pass(op, ob, cnt, mn, mx, sm, valid)) continue;
-        if (utf8) {
-            const std::string& s = op->dict_strs[k];
-            ob.key_data.insert(ob.key_data.end(), s.begin(), s.end());
-            ob.key_offsets.push_back((int32_t)ob.key_data.size());
-        } else if (op->key_kind == DZ_KEY_INT64) {
-            ob.key_i64.push_back(op->dict_vals[k]);
-        } else {
-            ob.key_i64.push_back(k);
+    if (op->key_kind == DZ_KEY_UTF8) {
+        ob.key_offsets.resize(n + 1);
+        ob.key_offsets[0] = 0;
+        size_t total = 0;
+        for (size_t i = 0; i < n; i++) total += op->dict_strs[rows[i]].size();
+        ob.key_data.resize(total);
+        size_t pos = 0;
+        for (size_t i = 0; i < n; i++) {
+            const std::string& s = op->dict_strs[rows[i]];
+            memcpy(ob.key_data.data() + pos, s.data(), s.size());
+            pos += s.size();
+            ob.key_offsets[i + 1] = (int32_t)pos;
         }
-        for (size_t a = 0; a < na; a++) {
-            switch (op->aggs[a].op) {
-                case DZ_AGG_COUNT: ob.agg_i64[a].push_back(cnt); break;
-                case DZ_AGG_MIN: ob.agg_f64[a].push_back(valid ? mn : 0.0); break;
-                case DZ_AGG_MAX: ob.agg_f64[a].push_back(valid ? mx : 0.0); break;
-                case DZ_AGG_SUM: ob.agg_f64[a].push_back(valid ? sm : 0.0); break;
-                case DZ_AGG_AVG:
-                    ob.agg_f64[a].push_back(valid ? sm / (double)cnt : 0.0);
-                    break;
+    } else if (op->key_kind == DZ_KEY_INT64) {
+        ob.key_i64.resize(n);
+        for (size_t i = 0; i < n; i++) ob.key_i64[i] = op->dict_vals[rows[i]];
+    } else {
+        ob.key_i64.resize(n);
+        for (size_t i = 0; i < n; i++) ob.key_i64[i] = rows[i];
+    }
+    for (size_t a = 0; a < na; a++) {
+        switch (op->aggs[a].op) {
+            case DZ_AGG_COUNT: {
+                auto& col = ob.agg_i64[a];
+                col.resize(n);
+                for (size_t i = 0; i < n; i++) col[i] = (int64_t)f_cnt[rows[i]];
+                break;
+            }
+            case DZ_AGG_MIN: {
+                auto& col = ob.agg_f64[a];
+                col.resize(n);
+                for (size_t i = 0; i < n; i++)
+                    col[i] = f_cnt[rows[i]] ? f_min[rows[i]] : 0.0;
+                break;
+            }
+            case DZ_AGG_MAX: {
+                auto& col = ob.agg_f64[a];
+                col.resize(n);
+                for (size_t i = 0; i < n; i++)
+                    col[i] = f_cnt[rows[i]] ? f_max[rows[i]] : 0.0;
+                break;
+            }
+            case DZ_AGG_SUM: {
+                auto& col = ob.agg_f64[a];
+                col.resize(n);
+                for (size_t i = 0; i < n; i++)
+                    col[i] = f_cnt[rows[i]] ? f_sum[rows[i]] : 0.0;
+                break;
+            }
+            case DZ_AGG_AVG: {
+                auto& col = ob.agg_f64[a];
+                col.resize(n);
+                for (size_t i = 0; i < n; i++) {
+                    uint64_t c = f_cnt[rows[i]];
+                    col[i] = c ? f_sum[rows[i]] / (double)c : 0.0;
+                }
+                break;
             }
         }
-        ob.agg_valid.push_back(valid ? 1 : 0);
-        ob.wstart.push_back(wstart);
-        ob.wend.push_back(wend);
     }
-    ob.view.n_rows = (int64_t)ob.wstart.size();
+    ob.agg_valid.resize(n);
+    for (size_t i = 0; i < n; i++) ob.agg_valid[i] = f_cnt[rows[i]] > 0 ? 1 : 0;
+    ob.wstart.assign(n, wstart);
+    ob.wend.assign(n, wend);
+    ob.view.n_rows = (int64_t)n;
     op->outq.push_back(std::move(ob));
     return DZ_OK;
 }
 
 static dz_status trigger_windows(dz_window_op* op) {
     if (!op->has_wm) return DZ_OK;
+    struct Closed { int64_t start, end; int32_t slot; };
+    std::vector<Closed> closed;
     for (auto it = op->open.begin(); it != op->open.end();) {
         if (op->watermark >= it->second.end) {
-            if (emit_window(op, it->first, it->second.end, it->second.slot) != DZ_OK)
-                return DZ_ERR;
-            if (slot_reset(op, it->second.slot) != DZ_OK) return DZ_ERR;
-            op->free_slots.push_back(it->second.slot);
+            closed.push_back({it->first, it->second.end, it->second.slot});
             it = op->open.erase(it);
         } else {
             ++it;
+        }
+    }
+    if (closed.empty()) return DZ_OK;
+    /* (re)size the pinned ring */
+    if (op->e_ring_kcap != op->kcap) {
+        for (auto& r : op->e_ring)
+            if (r) { hipHostFree(r); r = nullptr; }
+        for (auto& r : op->e_ring)
+            CHK(op, hipHostMalloc((void**)&r, (size_t)op->kcap * 5 * 8));
+        op->e_ring_kcap = op->kcap;
+    }
+    size_t stride = (size_t)op->kcap * 5;
+    for (size_t g = 0; g < closed.size(); g += dz_window_op::E_RING) {
+        size_t gn = std::min<size_t>(dz_window_op::E_RING, closed.size() - g);
+        {
+            HostTimer ht(op, "h_emit_d2h");
+            for (size_t i = 0; i < gn; i++)
+                CHK(op, hipMemcpyAsync(op->e_ring[i],
+                                       op->s_base + (size_t)closed[g + i].slot * stride,
+                                       stride * 8, hipMemcpyDeviceToHost, op->stream));
+            CHK(op, hipStreamSynchronize(op->stream));
+            drain_events(op, false);
+        }
+        for (size_t i = 0; i < gn; i++) {
+            if (build_emission(op, closed[g + i].start, closed[g + i].end,
+                               op->e_ring[i]) != DZ_OK)
+                return DZ_ERR;
+            if (slot_reset(op, closed[g + i].slot) != DZ_OK) return DZ_ERR;
+            op->free_slots.push_back(closed[g + i].slot);
         }
     }
     return DZ_OK;
