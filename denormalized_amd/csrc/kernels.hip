@@ -301,8 +301,8 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
         uint32_t r = 0, wtot = 0;
         int fl = lane;
         for (int j = 0; j < 64; j++) {
-            uint32_t bj = (uint32_t)__shfl((int)bkt, j);
-            uint32_t mj = (uint32_t)__shfl(m, j);
+            uint32_t bj = (uint32_t)__builtin_amdgcn_readlane((int)bkt, j);
+            uint32_t mj = (uint32_t)__builtin_amdgcn_readlane(m, j);
             if (bj == bkt) {
                 if (j < lane) r += mj;
                 wtot += mj;
@@ -390,10 +390,17 @@ __global__ __launch_bounds__(BLOCK) void k_fold(const uint32_t* rmeta,
             ri = rridx[base + lane];
             v = rvals[base + lane];
         }
+        /* broadcast via readlane (uniform lane index -> scalar ops; __shfl
+         * would lower to ds_bpermute + lgkm waits, ~5x the issue cost) */
+        const uint64_t vb = (uint64_t)__double_as_longlong(v);
+        const uint32_t vlo = (uint32_t)vb, vhi = (uint32_t)(vb >> 32);
         for (int s = 0; s < nt; s++) {
-            const uint32_t ms_ = (uint32_t)__shfl((int)m, s);
-            const double vs = __shfl(v, s);
-            const uint32_t rs = (uint32_t)__shfl((int)ri, s);
+            const uint32_t ms_ = (uint32_t)__builtin_amdgcn_readlane((int)m, s);
+            const uint32_t rs = (uint32_t)__builtin_amdgcn_readlane((int)ri, s);
+            const uint32_t lo32 = (uint32_t)__builtin_amdgcn_readlane((int)vlo, s);
+            const uint32_t hi32 = (uint32_t)__builtin_amdgcn_readlane((int)vhi, s);
+            const double vs = __longlong_as_double(
+                (long long)(((uint64_t)hi32 << 32) | lo32));
             if ((ms_ & 0xFFFFFu) == want && own) {
                 if (fst == ~0ULL) fst = ((uint64_t)fc.batch_seq << 32) | rs;
                 if (ms_ >> META_VALID_SHIFT) {

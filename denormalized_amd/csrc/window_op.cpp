@@ -16,6 +16,7 @@
 #include <map>
 #include <string>
 #include <unordered_map>
+#include <thread>
 #include <vector>
 
 namespace {
@@ -405,9 +406,8 @@ static bool filter_pass(dz_window_op* op, int64_t row_cnt,
 
 /* Build one emitted batch from a pinned copy of a slot slab
  * ([cnt][first][min][max][sum], each `kcap` 8-byte entries). */
-static dz_status build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
-                                const uint64_t* slab) {
-    HostTimer ht(op, "h_emit_build");
+static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
+                           const uint64_t* slab, OutBuf* out) {
     int64_t K = op->n_keys;
     int64_t kcap = op->e_ring_kcap;
     const uint64_t* f_cnt = slab;
@@ -520,8 +520,7 @@ static dz_status build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
     ob.wstart.assign(n, wstart);
     ob.wend.assign(n, wend);
     ob.view.n_rows = (int64_t)n;
-    op->outq.push_back(std::move(ob));
-    return DZ_OK;
+    *out = std::move(ob);
 }
 
 static dz_status trigger_windows(dz_window_op* op) {
@@ -557,10 +556,21 @@ static dz_status trigger_windows(dz_window_op* op) {
             CHK(op, hipStreamSynchronize(op->stream));
             drain_events(op, false);
         }
+        {
+            /* builds are independent per window: run them on host threads */
+            HostTimer ht(op, "h_emit_build");
+            std::vector<OutBuf> obs(gn);
+            std::vector<std::thread> th;
+            for (size_t i = 1; i < gn; i++)
+                th.emplace_back(build_emission, op, closed[g + i].start,
+                                closed[g + i].end, op->e_ring[i], &obs[i]);
+            build_emission(op, closed[g].start, closed[g].end, op->e_ring[0],
+                           &obs[0]);
+            for (auto& t : th) t.join();
+            for (size_t i = 0; i < gn; i++)
+                op->outq.push_back(std::move(obs[i]));
+        }
         for (size_t i = 0; i < gn; i++) {
-            if (build_emission(op, closed[g + i].start, closed[g + i].end,
-                               op->e_ring[i]) != DZ_OK)
-                return DZ_ERR;
             if (slot_reset(op, closed[g + i].slot) != DZ_OK) return DZ_ERR;
             op->free_slots.push_back(closed[g + i].slot);
         }
