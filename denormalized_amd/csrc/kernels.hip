@@ -401,21 +401,22 @@ __global__ __launch_bounds__(BLOCK) void k_fold(const uint32_t* rmeta,
             const uint32_t hi32 = (uint32_t)__builtin_amdgcn_readlane((int)vhi, s);
             const double vs = __longlong_as_double(
                 (long long)(((uint64_t)hi32 << 32) | lo32));
-            if ((ms_ & 0xFFFFFu) == want && own) {
-                if (fst == ~0ULL) fst = ((uint64_t)fc.batch_seq << 32) | rs;
-                if (ms_ >> META_VALID_SHIFT) {
-                    /* DF accumulator semantics: count non-null; min/max by
-                     * strict compare (first non-null initializes); sum in
-                     * row order (see oracle.c pinning note) */
-                    if (cnt == 0) { mn = vs; mx = vs; }
-                    else {
-                        if (vs < mn) mn = vs;
-                        if (vs > mx) mx = vs;
-                    }
-                    cnt++;
-                    sm += vs;
-                }
-            }
+            /* branchless: at most ONE lane matches each record, so exec-mask
+             * branches would serialize per record — selects instead.
+             * DF accumulator semantics: count non-null; min/max by strict
+             * compare (first non-null initializes); f64 sum in row order
+             * (see oracle.c pinning note). Selects follow the adds so no
+             * value is perturbed (e.g. no -0.0 + 0.0 normalization). */
+            const bool match = ((ms_ & 0xFFFFFu) == want) & own;
+            const bool upd = match & (bool)(ms_ >> META_VALID_SHIFT);
+            fst = (match && fst == ~0ULL) ? (((uint64_t)fc.batch_seq << 32) | rs)
+                                          : fst;
+            const bool fresh = cnt == 0;
+            mn = (upd && (fresh || vs < mn)) ? vs : mn;
+            mx = (upd && (fresh || vs > mx)) ? vs : mx;
+            const double sm2 = sm + vs;
+            sm = upd ? sm2 : sm;
+            cnt += upd ? 1 : 0;
         }
     }
 
