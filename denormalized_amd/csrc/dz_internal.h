@@ -10,8 +10,8 @@ namespace dz {
 /* Partition geometry. Rows are bucketed by (key_id & (NB-1)): with dense
  * dictionary ids this balances buckets to ±1 key. One wave folds one bucket,
  * so NB also sets fold parallelism (4096 waves = 16 waves/CU on 256 CUs). */
-constexpr int NB = 4096;
-constexpr int LOG_NB = 12;
+constexpr int NB = 8192;
+constexpr int LOG_NB = 13;
 constexpr int BLOCK = 256;          /* 4 waves */
 constexpr int WAVES_PER_BLOCK = BLOCK / 64;
 constexpr int MAX_RANGES = 4096;    /* window frames touched by one batch */
