@@ -689,7 +689,7 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
     wp.is_sliding = op->slide_ms > 0;
 
     /* 3. partition + fold */
-    int C = (int)std::min<int64_t>(2048, std::max<int64_t>(1, (n + 1023) / 1024));
+    int C = (int)std::min<int64_t>(512, std::max<int64_t>(1, (n + 8191) / 8192));
     int64_t chunk = (n + C - 1) / C;
     int64_t expand = wp.is_sliding
         ? (op->len_ms + op->slide_ms - 1) / op->slide_ms + 1 : 1;
