@@ -63,10 +63,18 @@ void launch_scatter(hipStream_t stream, const int32_t* d_kid, const int64_t* d_t
                     const uint32_t* d_gofs, uint32_t* d_meta, uint32_t* d_ridx,
                     double* d_rvals);
 
-void launch_fold(hipStream_t stream, const uint32_t* d_meta, const uint32_t* d_ridx,
-                 const double* d_rvals, const uint32_t* d_bucket_base,
-                 const FoldChunk& fc, const int32_t* d_slot_of_widx,
-                 uint64_t* s_cnt, double* s_min, double* s_max, double* s_sum,
-                 uint64_t* s_first);
+constexpr int FOLD_GCAP = 128; /* groups per bucket per fold chunk */
+
+void launch_regroup(hipStream_t stream, const uint32_t* d_meta,
+                    const uint32_t* d_ridx, const double* d_rvals,
+                    const uint32_t* d_bucket_base, const FoldChunk& fc,
+                    uint32_t* d_binoffs, uint32_t* d_binlens, double* d_gval,
+                    uint32_t* d_gridx);
+
+void launch_fold3(hipStream_t stream, const double* d_gval, const uint32_t* d_gridx,
+                  const uint32_t* d_bucket_base, const uint32_t* d_binoffs,
+                  const uint32_t* d_binlens, const FoldChunk& fc,
+                  const int32_t* d_slot_of_widx, uint64_t* s_cnt, double* s_min,
+                  double* s_max, double* s_sum, uint64_t* s_first);
 
 } // namespace dz

@@ -344,33 +344,141 @@ void launch_scatter(hipStream_t s, const int32_t* d_kid, const int64_t* d_ts,
 }
 
 /* ------------------------------------------------------------------ */
-/* fold — one wave per bucket, one lane per (window,key) group,        */
-/* rows folded in row order (bit-exact vs the reference accumulators)  */
+/* regroup — one wave per bucket: stable wave-local split of the       */
+/* bucket's records into per-(window,key) GROUP segments               */
+/* (val 8B + rowidx/valid 4B), so the fold can walk each group's rows  */
+/* sequentially. Two passes over the bucket region (second is L2-hot): */
+/* count bins (LDS atomics), exclusive prefix, then ranked placement   */
+/* (bit-ballot same-group masks + LDS cursors; ranks are row-ordered). */
 /* ------------------------------------------------------------------ */
 
-__global__ __launch_bounds__(BLOCK) void k_fold(const uint32_t* rmeta,
+constexpr int GCAP = FOLD_GCAP; /* bins (groups) per bucket per chunk */
+
+__global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
         const uint32_t* rridx, const double* rvals, const uint32_t* bucket_base,
-        FoldChunk fc, const int32_t* slot_of_widx, uint64_t* s_cnt,
-        double* s_min, double* s_max, double* s_sum, uint64_t* s_first) {
+        FoldChunk fc, uint32_t* binoffs, uint32_t* binlens, double* gval,
+        uint32_t* gridx) {
+    __shared__ uint32_t lds[2 * GCAP * WAVES_PER_BLOCK];
     const int bkt = (int)((blockIdx.x * (int64_t)BLOCK + threadIdx.x) >> 6);
     const int lane = threadIdx.x & 63;
+    const int wave = (threadIdx.x >> 6);
+    uint32_t* cnt = lds + 2 * GCAP * wave;
+    uint32_t* cur = cnt + GCAP;
     if (bkt >= NB) return;
     const uint32_t lo = bucket_base[bkt];
     const uint32_t hi = bucket_base[bkt + 1];
-    if (lo == hi) return;
-
     const int nk = fc.k_hi - fc.k_lo;
-    const int my_widx = fc.w_lo + lane / nk;
-    const int my_kloc = fc.k_lo + lane % nk;
-    const bool own = my_widx < fc.w_hi;
-    const uint32_t want = (uint32_t)my_kloc | ((uint32_t)my_widx << META_WIDX_SHIFT);
+    const int glocal_n = (fc.w_hi - fc.w_lo) * nk;
+    for (int g = lane; g < GCAP; g += 64) cnt[g] = 0;
+    if (lo == hi) { /* still publish empty bins for the fold */
+        for (int g = lane; g < glocal_n; g += 64) {
+            binoffs[(int64_t)bkt * GCAP + g] = 0;
+            binlens[(int64_t)bkt * GCAP + g] = 0;
+        }
+        return;
+    }
+    /* pass 1: bin counts */
+    for (uint32_t base = lo + lane; base < hi; base += 64) {
+        const uint32_t ms_ = rmeta[base];
+        const int widx = (int)((ms_ >> META_WIDX_SHIFT) & META_WIDX_MASK);
+        const int kloc = (int)(ms_ & META_KLOC_MASK);
+        if (widx >= fc.w_lo && widx < fc.w_hi && kloc >= fc.k_lo && kloc < fc.k_hi)
+            atomicAdd(&cnt[(widx - fc.w_lo) * nk + (kloc - fc.k_lo)], 1u);
+    }
+    /* exclusive prefix over GCAP bins (wave-serial by lane 0 is fine: tiny) */
+    if (lane == 0) {
+        uint32_t run = 0;
+        for (int g = 0; g < GCAP; g++) {
+            uint32_t t = cnt[g];
+            cur[g] = run;
+            run += t;
+        }
+    }
+    /* publish segment layout for the fold */
+    for (int g = lane; g < glocal_n; g += 64) {
+        binoffs[(int64_t)bkt * GCAP + g] = cur[g];
+        binlens[(int64_t)bkt * GCAP + g] = cnt[g];
+    }
+    /* pass 2: ranked placement (row order within each group) */
+    for (uint32_t base = lo; base < hi; base += 64) {
+        const int nt = (int)min(64u, hi - base);
+        uint32_t g = 0xFFu; /* sentinel: record outside this chunk */
+        uint32_t ri = 0;
+        double v = 0.0;
+        if (lane < nt) {
+            const uint32_t ms_ = rmeta[base + lane];
+            const int widx = (int)((ms_ >> META_WIDX_SHIFT) & META_WIDX_MASK);
+            const int kloc = (int)(ms_ & META_KLOC_MASK);
+            if (widx >= fc.w_lo && widx < fc.w_hi && kloc >= fc.k_lo &&
+                kloc < fc.k_hi) {
+                g = (uint32_t)((widx - fc.w_lo) * nk + (kloc - fc.k_lo));
+                ri = rridx[base + lane] | ((ms_ >> META_VALID_SHIFT) << 31);
+                v = rvals[base + lane];
+            }
+        }
+        /* same-group lane mask via bit-ballots over the 8 bin-id bits */
+        uint64_t same = ~0ULL;
+        for (int b = 0; b < 8; b++) {
+            uint64_t bb = __ballot((g >> b) & 1);
+            same &= ((g >> b) & 1) ? bb : ~bb;
+        }
+        const uint64_t below = (lane == 63) ? ~0ULL : ((1ULL << (lane + 1)) - 1);
+        const int rank = (int)__popcll(same & below) - 1;
+        const int leader = __ffsll((unsigned long long)same) - 1;
+        uint32_t pos = 0;
+        if (g != 0xFFu) {
+            pos = cur[g] + (uint32_t)rank;
+            if (lane == leader) cur[g] += (uint32_t)__popcll(same);
+            gval[lo + pos] = v;
+            gridx[lo + pos] = ri;
+        }
+        __builtin_amdgcn_wave_barrier(); /* keep cur updates tile-ordered */
+    }
+}
 
+void launch_regroup(hipStream_t s, const uint32_t* d_meta, const uint32_t* d_ridx,
+                    const double* d_rvals, const uint32_t* d_bucket_base,
+                    const FoldChunk& fc, uint32_t* d_binoffs, uint32_t* d_binlens,
+                    double* d_gval, uint32_t* d_gridx) {
+    hipLaunchKernelGGL(k_regroup, dim3(NB / WAVES_PER_BLOCK), dim3(BLOCK), 0, s,
+                       d_meta, d_ridx, d_rvals, d_bucket_base, fc, d_binoffs,
+                       d_binlens, d_gval, d_gridx);
+}
+
+/* ------------------------------------------------------------------ */
+/* segmented fold — one LANE per (window,key) group walking its own    */
+/* segment sequentially: 64 rows progress per wave instruction, rows   */
+/* fold in row order (bit-exact vs the reference accumulators).        */
+/* ------------------------------------------------------------------ */
+
+__global__ __launch_bounds__(BLOCK) void k_fold3(const double* gval,
+        const uint32_t* gridx, const uint32_t* bucket_base,
+        const uint32_t* binoffs, const uint32_t* binlens, FoldChunk fc,
+        const int32_t* slot_of_widx, uint64_t* s_cnt, double* s_min,
+        double* s_max, double* s_sum, uint64_t* s_first) {
+    const int nk = fc.k_hi - fc.k_lo;
+    const int glocal_n = (fc.w_hi - fc.w_lo) * nk;
+    const int waves_per_bkt = (glocal_n + 63) / 64;
+    const int64_t gw = (blockIdx.x * (int64_t)BLOCK + threadIdx.x) >> 6;
+    const int bkt = (int)(gw / waves_per_bkt);
+    const int lane = threadIdx.x & 63;
+    const int g = (int)(gw % waves_per_bkt) * 64 + lane;
+    if (bkt >= NB) return;
+    const bool own = g < glocal_n;
+    const int my_widx = fc.w_lo + (own ? g / nk : 0);
+    const int my_kloc = fc.k_lo + (own ? g % nk : 0);
+
+    const uint32_t lo = bucket_base[bkt];
+    uint32_t off = 0, len = 0;
+    if (own) {
+        off = binoffs[(int64_t)bkt * GCAP + g];
+        len = binlens[(int64_t)bkt * GCAP + g];
+    }
     int64_t sidx = 0;
     uint64_t cnt = 0, fst = ~0ULL;
     double mn = 0.0, mx = 0.0, sm = 0.0;
     if (own) {
         int64_t slot = slot_of_widx[my_widx];
-        /* state slabs: [slot][5 fields][kcap] (window_op.cpp) */
         sidx = slot * (5 * fc.kcap) + (((int64_t)my_kloc << LOG_NB) | bkt);
         cnt = s_cnt[sidx];
         fst = s_first[sidx];
@@ -378,48 +486,28 @@ __global__ __launch_bounds__(BLOCK) void k_fold(const uint32_t* rmeta,
             mn = s_min[sidx];
             mx = s_max[sidx];
             sm = s_sum[sidx];
+        } else {
+            sm = 0.0;
         }
     }
-
-    for (uint32_t base = lo; base < hi; base += 64) {
-        const int nt = (int)min(64u, hi - base);
-        uint32_t m = 0, ri = 0;
-        double v = 0.0;
-        if (lane < nt) {
-            m = rmeta[base + lane];
-            ri = rridx[base + lane];
-            v = rvals[base + lane];
-        }
-        /* broadcast via readlane (uniform lane index -> scalar ops; __shfl
-         * would lower to ds_bpermute + lgkm waits, ~5x the issue cost) */
-        const uint64_t vb = (uint64_t)__double_as_longlong(v);
-        const uint32_t vlo = (uint32_t)vb, vhi = (uint32_t)(vb >> 32);
-        for (int s = 0; s < nt; s++) {
-            const uint32_t ms_ = (uint32_t)__builtin_amdgcn_readlane((int)m, s);
-            const uint32_t rs = (uint32_t)__builtin_amdgcn_readlane((int)ri, s);
-            const uint32_t lo32 = (uint32_t)__builtin_amdgcn_readlane((int)vlo, s);
-            const uint32_t hi32 = (uint32_t)__builtin_amdgcn_readlane((int)vhi, s);
-            const double vs = __longlong_as_double(
-                (long long)(((uint64_t)hi32 << 32) | lo32));
-            /* branchless: at most ONE lane matches each record, so exec-mask
-             * branches would serialize per record — selects instead.
-             * DF accumulator semantics: count non-null; min/max by strict
-             * compare (first non-null initializes); f64 sum in row order
-             * (see oracle.c pinning note). Selects follow the adds so no
-             * value is perturbed (e.g. no -0.0 + 0.0 normalization). */
-            const bool match = ((ms_ & 0xFFFFFu) == want) & own;
-            const bool upd = match & (bool)(ms_ >> META_VALID_SHIFT);
-            fst = (match && fst == ~0ULL) ? (((uint64_t)fc.batch_seq << 32) | rs)
-                                          : fst;
-            const bool fresh = cnt == 0;
-            mn = (upd && (fresh || vs < mn)) ? vs : mn;
-            mx = (upd && (fresh || vs > mx)) ? vs : mx;
-            const double sm2 = sm + vs;
-            sm = upd ? sm2 : sm;
-            cnt += upd ? 1 : 0;
-        }
+    if (own && len > 0 && fst == ~0ULL)
+        fst = ((uint64_t)fc.batch_seq << 32) | (gridx[lo + off] & 0x7FFFFFFFu);
+    uint32_t mlen = len;
+    for (int o = 32; o > 0; o >>= 1)
+        mlen = max(mlen, (uint32_t)__shfl_down((int)mlen, o));
+    mlen = (uint32_t)__builtin_amdgcn_readlane((int)mlen, 0);
+    for (uint32_t r = 0; r < mlen; r++) {
+        const bool act = own && r < len;
+        const uint32_t ri = act ? gridx[lo + off + r] : 0;
+        const double v = act ? gval[lo + off + r] : 0.0;
+        const bool upd = act && (ri >> 31);
+        const bool fresh = cnt == 0;
+        mn = (upd && (fresh || v < mn)) ? v : mn;
+        mx = (upd && (fresh || v > mx)) ? v : mx;
+        const double sm2 = sm + v;
+        sm = upd ? sm2 : sm;
+        cnt += upd ? 1 : 0;
     }
-
     if (own) {
         s_cnt[sidx] = cnt;
         s_first[sidx] = fst;
@@ -429,14 +517,20 @@ __global__ __launch_bounds__(BLOCK) void k_fold(const uint32_t* rmeta,
     }
 }
 
-void launch_fold(hipStream_t s, const uint32_t* d_meta, const uint32_t* d_ridx,
-                 const double* d_rvals, const uint32_t* d_bucket_base,
-                 const FoldChunk& fc, const int32_t* d_slot_of_widx,
-                 uint64_t* s_cnt, double* s_min, double* s_max, double* s_sum,
-                 uint64_t* s_first) {
-    hipLaunchKernelGGL(k_fold, dim3(NB / WAVES_PER_BLOCK), dim3(BLOCK), 0, s,
-                       d_meta, d_ridx, d_rvals, d_bucket_base, fc, d_slot_of_widx,
+void launch_fold3(hipStream_t s, const double* d_gval, const uint32_t* d_gridx,
+                  const uint32_t* d_bucket_base, const uint32_t* d_binoffs,
+                  const uint32_t* d_binlens, const FoldChunk& fc,
+                  const int32_t* d_slot_of_widx, uint64_t* s_cnt, double* s_min,
+                  double* s_max, double* s_sum, uint64_t* s_first) {
+    int nk = fc.k_hi - fc.k_lo;
+    int glocal_n = (fc.w_hi - fc.w_lo) * nk;
+    int waves_per_bkt = (glocal_n + 63) / 64;
+    int64_t waves = (int64_t)NB * waves_per_bkt;
+    int blocks = (int)((waves + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK);
+    hipLaunchKernelGGL(k_fold3, dim3(blocks), dim3(BLOCK), 0, s, d_gval, d_gridx,
+                       d_bucket_base, d_binoffs, d_binlens, fc, d_slot_of_widx,
                        s_cnt, s_min, s_max, s_sum, s_first);
 }
+
 
 } // namespace dz

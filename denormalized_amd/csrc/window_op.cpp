@@ -138,6 +138,10 @@ struct dz_window_op {
     uint32_t* d_meta = nullptr;
     uint32_t* d_ridx = nullptr;
     double* d_rvals = nullptr;
+    double* d_gval = nullptr;    /* group-segmented values (regroup output) */
+    uint32_t* d_gridx = nullptr; /* rowidx | valid<<31, segmented */
+    uint32_t* d_binoffs = nullptr;
+    uint32_t* d_binlens = nullptr;
     int64_t rec_cap = 0;
     uint64_t* d_scalars = nullptr;
     uint64_t* h_scalars = nullptr; /* pinned, 3 */
@@ -357,6 +361,8 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->s_base);
     hipFree(op->d_ghist); hipFree(op->d_gofs); hipFree(op->d_total); hipFree(op->d_base);
     hipFree(op->d_meta); hipFree(op->d_ridx); hipFree(op->d_rvals);
+    hipFree(op->d_gval); hipFree(op->d_gridx);
+    hipFree(op->d_binoffs); hipFree(op->d_binlens);
     hipFree(op->d_scalars); hipFree(op->d_slotmap);
     hipFree(op->d_ts); hipFree(op->d_kid); hipFree(op->d_vals); hipFree(op->d_valbm);
     if (op->h_scalars) hipHostFree(op->h_scalars);
@@ -595,10 +601,17 @@ static dz_status ensure_scratch(dz_window_op* op, int C, int64_t nrec) {
     }
     if (nrec > op->rec_cap) {
         hipFree(op->d_meta); hipFree(op->d_ridx); hipFree(op->d_rvals);
+        hipFree(op->d_gval); hipFree(op->d_gridx);
         CHK(op, hipMalloc(&op->d_meta, (size_t)nrec * 4));
         CHK(op, hipMalloc(&op->d_ridx, (size_t)nrec * 4));
         CHK(op, hipMalloc(&op->d_rvals, (size_t)nrec * 8));
+        CHK(op, hipMalloc(&op->d_gval, (size_t)nrec * 8));
+        CHK(op, hipMalloc(&op->d_gridx, (size_t)nrec * 4));
         op->rec_cap = nrec;
+    }
+    if (!op->d_binoffs) {
+        CHK(op, hipMalloc(&op->d_binoffs, (size_t)dz::NB * dz::FOLD_GCAP * 4));
+        CHK(op, hipMalloc(&op->d_binlens, (size_t)dz::NB * dz::FOLD_GCAP * 4));
     }
     return DZ_OK;
 }
@@ -607,6 +620,10 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
                            const int32_t* d_kid, const double* d_vals,
                            const uint8_t* d_valbm, bool keys_are_dense) {
     if (n <= 0) return DZ_OK; /* empty batch: reference emits empty (no-op) */
+    if (n >= (1LL << 31)) {
+        op->err = "batch exceeds 2^31 rows; push smaller batches";
+        return DZ_ERR;
+    }
     CHK(op, hipSetDevice(op->device));
 
     /* 1. batch watermark bounds + max key id (time.rs:31-57) */
@@ -712,9 +729,9 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
                            wp, op->d_gofs, op->d_meta, op->d_ridx, op->d_rvals);
     });
     int64_t klocs = op->kcap >> dz::LOG_NB;
-    for (int64_t k_lo = 0; k_lo < klocs; k_lo += 64) {
-        int32_t nk = (int32_t)std::min<int64_t>(64, klocs - k_lo);
-        int32_t wstep = 64 / nk;
+    for (int64_t k_lo = 0; k_lo < klocs; k_lo += dz::FOLD_GCAP) {
+        int32_t nk = (int32_t)std::min<int64_t>(dz::FOLD_GCAP, klocs - k_lo);
+        int32_t wstep = dz::FOLD_GCAP / nk;
         for (int64_t w_lo = 0; w_lo < nw; w_lo += wstep) {
             dz::FoldChunk fc;
             fc.w_lo = (int32_t)w_lo;
@@ -723,10 +740,16 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
             fc.k_hi = (int32_t)(k_lo + nk);
             fc.kcap = op->kcap;
             fc.batch_seq = op->batch_seq;
-            timed(op, "fold", (double)nrec_max * 16, [&] {
-                dz::launch_fold(op->stream, op->d_meta, op->d_ridx, op->d_rvals,
-                                op->d_base, fc, op->d_slotmap, op->s_cnt,
-                                op->s_min, op->s_max, op->s_sum, op->s_first);
+            timed(op, "regroup", (double)nrec_max * 32, [&] {
+                dz::launch_regroup(op->stream, op->d_meta, op->d_ridx,
+                                   op->d_rvals, op->d_base, fc, op->d_binoffs,
+                                   op->d_binlens, op->d_gval, op->d_gridx);
+            });
+            timed(op, "fold", (double)nrec_max * 12, [&] {
+                dz::launch_fold3(op->stream, op->d_gval, op->d_gridx, op->d_base,
+                                 op->d_binoffs, op->d_binlens, fc, op->d_slotmap,
+                                 op->s_cnt, op->s_min, op->s_max, op->s_sum,
+                                 op->s_first);
             });
         }
     }
