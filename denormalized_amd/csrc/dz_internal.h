@@ -10,17 +10,18 @@ namespace dz {
 /* Partition geometry. Rows are bucketed by (key_id & (NB-1)): with dense
  * dictionary ids this balances buckets to ±1 key. One wave folds one bucket,
  * so NB also sets fold parallelism (4096 waves = 16 waves/CU on 256 CUs). */
-constexpr int NB = 8192;
-constexpr int LOG_NB = 13;
+constexpr int NB = 512;
+constexpr int LOG_NB = 9;
 constexpr int BLOCK = 256;          /* 4 waves */
 constexpr int WAVES_PER_BLOCK = BLOCK / 64;
 constexpr int MAX_RANGES = 4096;    /* window frames touched by one batch */
+constexpr int ST_RECORDS = 2048;    /* scatter staging records per supertile */
 
-/* meta word packed per record: kloc (8b) | widx (12b) | valid (1b) */
-constexpr uint32_t META_KLOC_MASK = 0xFFu;
-constexpr int META_WIDX_SHIFT = 8;
+/* meta word packed per record: kloc (16b) | widx (12b) | valid (1b) */
+constexpr uint32_t META_KLOC_MASK = 0xFFFFu;
+constexpr int META_WIDX_SHIFT = 16;
 constexpr uint32_t META_WIDX_MASK = 0xFFFu;
-constexpr int META_VALID_SHIFT = 20;
+constexpr int META_VALID_SHIFT = 28;
 
 struct WinParams {
     int64_t s0;          /* first window start of this batch's range list */
@@ -59,9 +60,9 @@ void launch_scan(hipStream_t stream, const uint32_t* d_ghist, int C,
 
 void launch_scatter(hipStream_t stream, const int32_t* d_kid, const int64_t* d_ts,
                     const double* d_vals, const uint8_t* d_validity /*bitmap|null*/,
-                    int64_t n, int64_t chunk, int C, const WinParams& wp,
-                    const uint32_t* d_gofs, uint32_t* d_meta, uint32_t* d_ridx,
-                    double* d_rvals);
+                    int64_t n, int64_t chunk, int C, int32_t st_rows,
+                    const WinParams& wp, const uint32_t* d_gofs, uint32_t* d_meta,
+                    uint32_t* d_ridx, double* d_rvals);
 
 constexpr int FOLD_GCAP = 128; /* groups per bucket per fold chunk */
 

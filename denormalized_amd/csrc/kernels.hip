@@ -205,32 +205,23 @@ __global__ void k_scan_partial(const uint32_t* ghist, int C, int cs,
 }
 
 /* single block: bucket totals from psum + exclusive scan -> base[NB+1] */
-__global__ __launch_bounds__(1024) void k_scan_base(const uint32_t* psum,
-                                                    uint32_t* base) {
-    __shared__ uint32_t part[1024];
-    constexpr int PER = NB / 1024;
-    uint32_t loc[PER];
-    uint32_t s = 0;
-    for (int j = 0; j < PER; j++) {
-        int bkt = threadIdx.x * PER + j;
-        uint32_t t = 0;
-        for (int g = 0; g < SSPLIT; g++) t += psum[(int64_t)g * NB + bkt];
-        loc[j] = s;
-        s += t;
-    }
-    part[threadIdx.x] = s;
+__global__ __launch_bounds__(NB) void k_scan_base(const uint32_t* psum,
+                                                  uint32_t* base) {
+    __shared__ uint32_t part[NB];
+    const int bkt = threadIdx.x;
+    uint32_t t = 0;
+    for (int g = 0; g < SSPLIT; g++) t += psum[(int64_t)g * NB + bkt];
+    part[bkt] = t;
     __syncthreads();
-    /* Hillis-Steele inclusive scan over 1024 partials */
-    for (int o = 1; o < 1024; o <<= 1) {
-        uint32_t v = (threadIdx.x >= o) ? part[threadIdx.x - o] : 0;
+    /* Hillis-Steele inclusive scan over NB partials */
+    for (int o = 1; o < NB; o <<= 1) {
+        uint32_t v = (bkt >= o) ? part[bkt - o] : 0;
         __syncthreads();
-        part[threadIdx.x] += v;
+        part[bkt] += v;
         __syncthreads();
     }
-    uint32_t pre = (threadIdx.x > 0) ? part[threadIdx.x - 1] : 0;
-    for (int j = 0; j < PER; j++)
-        base[threadIdx.x * PER + j] = pre + loc[j];
-    if (threadIdx.x == 1023) base[NB] = part[1023];
+    base[bkt] = bkt ? part[bkt - 1] : 0;
+    if (bkt == NB - 1) base[NB] = part[NB - 1];
 }
 
 /* per-chunk running offsets within each segment:
@@ -256,7 +247,7 @@ void launch_scan(hipStream_t s, const uint32_t* d_ghist, int C,
     int cs = (C + SSPLIT - 1) / SSPLIT;
     hipLaunchKernelGGL(k_scan_partial, dim3(NB / 256, SSPLIT), dim3(256), 0, s,
                        d_ghist, C, cs, d_psum);
-    hipLaunchKernelGGL(k_scan_base, dim3(1), dim3(1024), 0, s, d_psum, d_base);
+    hipLaunchKernelGGL(k_scan_base, dim3(1), dim3(NB), 0, s, d_psum, d_base);
     hipLaunchKernelGGL(k_scan_offsets, dim3(NB / 256, SSPLIT), dim3(256), 0, s,
                        d_ghist, C, cs, d_psum, d_base, d_gofs);
 }
@@ -267,80 +258,140 @@ void launch_scan(hipStream_t s, const uint32_t* d_ghist, int C,
 
 __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
         const int64_t* ts, const double* vals, const uint8_t* validity,
-        int64_t n, int64_t chunk, WinParams wp, const uint32_t* gofs,
-        uint32_t* rmeta, uint32_t* rridx, double* rvals) {
-    __shared__ uint32_t cur[NB];
+        int64_t n, int64_t chunk, int32_t st_rows, WinParams wp,
+        const uint32_t* gofs, uint32_t* rmeta, uint32_t* rridx, double* rvals) {
+    /* LDS-staged stable partition: records of a supertile are ranked
+     * (row order preserved), placed bucket-major in LDS, then flushed so
+     * adjacent lanes write adjacent global addresses (~4-record runs at
+     * NB buckets instead of isolated 16 B scatters — the PMC-measured 6x
+     * write amplification of the direct form, profiles/hbm_traffic.json). */
+    __shared__ uint32_t cur[NB];   /* global cursors for this block's chunk */
+    __shared__ uint32_t cnt[NB];   /* per-supertile expanded record counts  */
+    __shared__ uint32_t offs[NB];  /* per-supertile exclusive bin prefix    */
+    __shared__ uint32_t stcur[NB]; /* running staging cursor per bin        */
+    __shared__ uint32_t scanbuf[BLOCK];
+    __shared__ uint32_t s_meta[ST_RECORDS];
+    __shared__ uint32_t s_ridx[ST_RECORDS];
+    __shared__ uint32_t s_dest[ST_RECORDS];
+    __shared__ double s_val[ST_RECORDS];
+    __shared__ uint32_t s_total;
+
     for (int t = threadIdx.x; t < NB; t += BLOCK)
         cur[t] = gofs[(int64_t)blockIdx.x * NB + t];
-    __syncthreads();
-
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
-    int64_t lo = blockIdx.x * chunk;
-    int64_t hi = i64min(n, lo + chunk);
+    const int64_t lo = blockIdx.x * chunk;
+    const int64_t hi = i64min(n, lo + chunk);
     constexpr uint32_t SENT = 0xFFFFFFFFu;
 
-    for (int64_t t0 = lo; t0 < hi; t0 += BLOCK) {
-        int64_t i = t0 + threadIdx.x;
-        uint32_t bkt = SENT;
-        int32_t jmin = 0, m = 0;
-        uint32_t kv = 0;
-        int64_t tsv = 0;
-        double v = 0.0;
-        uint32_t valid = 1;
-        if (i < hi) {
-            kv = (uint32_t)kid[i];
-            tsv = ts[i];
-            v = vals[i];
-            if (validity) valid = (validity[i >> 3] >> (i & 7)) & 1u;
-            row_windows(tsv, wp, &jmin, &m);
-            bkt = kv & (NB - 1);
+    for (int64_t st0 = lo; st0 < hi; st0 += st_rows) {
+        const int64_t st1 = i64min(hi, st0 + st_rows);
+        /* pass A: expanded counts per bucket */
+        for (int t = threadIdx.x; t < NB; t += BLOCK) cnt[t] = 0;
+        __syncthreads();
+        for (int64_t i = st0 + threadIdx.x; i < st1; i += BLOCK) {
+            int32_t jm, m;
+            row_windows(ts[i], wp, &jm, &m);
+            if (m > 0) atomicAdd(&cnt[kid[i] & (NB - 1)], (uint32_t)m);
         }
-        /* weighted intra-wave rank (row order), per-bucket wave totals,
-         * first lane per bucket. 64-wide loop keeps ranks deterministic. */
-        uint32_t r = 0, wtot = 0;
-        int fl = lane;
-        for (int j = 0; j < 64; j++) {
-            uint32_t bj = (uint32_t)__builtin_amdgcn_readlane((int)bkt, j);
-            uint32_t mj = (uint32_t)__builtin_amdgcn_readlane(m, j);
-            if (bj == bkt) {
-                if (j < lane) r += mj;
-                wtot += mj;
-                if (j < fl) fl = j;
+        __syncthreads();
+        /* block-level exclusive prefix over NB bins */
+        {
+            constexpr int PER = NB / BLOCK;
+            uint32_t loc[PER];
+            uint32_t s = 0;
+            for (int j = 0; j < PER; j++) {
+                loc[j] = s;
+                s += cnt[threadIdx.x * PER + j];
             }
-        }
-        /* wave-serialized cursor allocation keeps cross-wave row order */
-        uint32_t base = 0;
-        for (int w = 0; w < WAVES_PER_BLOCK; w++) {
-            if (wave == w) {
-                uint32_t pre = 0;
-                if (lane == fl && bkt != SENT) {
-                    pre = cur[bkt];
-                    cur[bkt] = pre + wtot;
-                }
-                pre = (uint32_t)__shfl((int)pre, fl);
-                base = pre + r;
-            }
+            scanbuf[threadIdx.x] = s;
             __syncthreads();
+            for (int o = 1; o < BLOCK; o <<= 1) {
+                uint32_t v = (threadIdx.x >= o) ? scanbuf[threadIdx.x - o] : 0;
+                __syncthreads();
+                scanbuf[threadIdx.x] += v;
+                __syncthreads();
+            }
+            uint32_t pre = threadIdx.x ? scanbuf[threadIdx.x - 1] : 0;
+            for (int j = 0; j < PER; j++) {
+                offs[threadIdx.x * PER + j] = pre + loc[j];
+                stcur[threadIdx.x * PER + j] = pre + loc[j];
+            }
+            if (threadIdx.x == BLOCK - 1) s_total = scanbuf[BLOCK - 1];
         }
-        uint32_t kloc = kv >> LOG_NB;
-        for (int jj = 0; jj < m; jj++) {
-            uint32_t d = base + jj;
-            rmeta[d] = kloc | ((uint32_t)(jmin + jj) << META_WIDX_SHIFT)
-                       | (valid << META_VALID_SHIFT);
-            rridx[d] = (uint32_t)i;
-            rvals[d] = v;
+        __syncthreads();
+        /* pass B: ranked placement into LDS staging (row order per bucket:
+         * weighted intra-wave ranks + wave-serialized staging cursors) */
+        for (int64_t t0 = st0; t0 < st1; t0 += BLOCK) {
+            const int64_t i = t0 + threadIdx.x;
+            uint32_t bkt = SENT;
+            int32_t jmin = 0, m = 0;
+            uint32_t kv = 0, valid = 1;
+            double v = 0.0;
+            if (i < st1) {
+                kv = (uint32_t)kid[i];
+                v = vals[i];
+                if (validity) valid = (validity[i >> 3] >> (i & 7)) & 1u;
+                row_windows(ts[i], wp, &jmin, &m);
+                bkt = kv & (NB - 1);
+            }
+            uint32_t r = 0, wtot = 0;
+            int fl = lane;
+            for (int j = 0; j < 64; j++) {
+                uint32_t bj = (uint32_t)__builtin_amdgcn_readlane((int)bkt, j);
+                uint32_t mj = (uint32_t)__builtin_amdgcn_readlane(m, j);
+                if (bj == bkt) {
+                    if (j < lane) r += mj;
+                    wtot += mj;
+                    if (j < fl) fl = j;
+                }
+            }
+            uint32_t base = 0;
+            for (int w = 0; w < WAVES_PER_BLOCK; w++) {
+                if (wave == w) {
+                    uint32_t pre = 0;
+                    if (lane == fl && bkt != SENT) {
+                        pre = stcur[bkt];
+                        stcur[bkt] = pre + wtot;
+                    }
+                    pre = (uint32_t)__shfl((int)pre, fl);
+                    base = pre + r;
+                }
+                __syncthreads();
+            }
+            const uint32_t kloc = kv >> LOG_NB;
+            for (int jj = 0; jj < m; jj++) {
+                const uint32_t p = base + jj;
+                s_meta[p] = kloc | ((uint32_t)(jmin + jj) << META_WIDX_SHIFT)
+                            | (valid << META_VALID_SHIFT);
+                s_ridx[p] = (uint32_t)i;
+                s_val[p] = v;
+                s_dest[p] = cur[bkt] + (p - offs[bkt]);
+            }
         }
+        __syncthreads();
+        /* flush: bucket-major staging => coalesced run writes */
+        const uint32_t tot = s_total;
+        for (uint32_t p = threadIdx.x; p < tot; p += BLOCK) {
+            const uint32_t d = s_dest[p];
+            rmeta[d] = s_meta[p];
+            rridx[d] = s_ridx[p];
+            rvals[d] = s_val[p];
+        }
+        __syncthreads();
+        for (int t = threadIdx.x; t < NB; t += BLOCK) cur[t] += cnt[t];
+        __syncthreads();
     }
 }
 
 void launch_scatter(hipStream_t s, const int32_t* d_kid, const int64_t* d_ts,
                     const double* d_vals, const uint8_t* d_validity, int64_t n,
-                    int64_t chunk, int C, const WinParams& wp,
+                    int64_t chunk, int C, int32_t st_rows, const WinParams& wp,
                     const uint32_t* d_gofs, uint32_t* d_meta, uint32_t* d_ridx,
                     double* d_rvals) {
     hipLaunchKernelGGL(k_scatter, dim3(C), dim3(BLOCK), 0, s, d_kid, d_ts, d_vals,
-                       d_validity, n, chunk, wp, d_gofs, d_meta, d_ridx, d_rvals);
+                       d_validity, n, chunk, st_rows, wp, d_gofs, d_meta, d_ridx,
+                       d_rvals);
 }
 
 /* ------------------------------------------------------------------ */

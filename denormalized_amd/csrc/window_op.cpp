@@ -260,8 +260,8 @@ static dz_status state_alloc(dz_window_op* op, int64_t kcap_new, int32_t nslots_
     /* allocate fresh arrays, memset to empty, copy any open-slot regions */
     kcap_new = std::max<int64_t>(kcap_new, dz::NB);
     kcap_new = (kcap_new + dz::NB - 1) / dz::NB * dz::NB;
-    if (kcap_new / dz::NB > 255) {
-        op->err = "key capacity exceeds 255*NB (~1.04M) — not supported yet";
+    if (kcap_new / dz::NB > 65535) {
+        op->err = "key capacity exceeds 65535*NB (~33M) — not supported yet";
         return DZ_ERR;
     }
     if (kcap_new == op->kcap && nslots_new <= op->nslots) return DZ_OK;
@@ -724,9 +724,11 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
         dz::launch_scan(op->stream, op->d_ghist, C, op->d_total, op->d_base,
                         op->d_gofs);
     });
-    timed(op, "scatter", (double)n * 20 + (double)nrec_max * 16, [&] {
+    int32_t st_rows = (int32_t)std::max<int64_t>(64, dz::ST_RECORDS / expand);
+    timed(op, "scatter", (double)n * 32 + (double)nrec_max * 16, [&] {
         dz::launch_scatter(op->stream, d_kid, d_ts, d_vals, d_valbm, n, chunk, C,
-                           wp, op->d_gofs, op->d_meta, op->d_ridx, op->d_rvals);
+                           st_rows, wp, op->d_gofs, op->d_meta, op->d_ridx,
+                           op->d_rvals);
     });
     int64_t klocs = op->kcap >> dz::LOG_NB;
     for (int64_t k_lo = 0; k_lo < klocs; k_lo += dz::FOLD_GCAP) {
