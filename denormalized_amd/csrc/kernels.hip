@@ -409,81 +409,136 @@ __global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
         const uint32_t* rridx, const double* rvals, const uint32_t* bucket_base,
         FoldChunk fc, uint32_t* binoffs, uint32_t* binlens, double* gval,
         uint32_t* gridx) {
-    __shared__ uint32_t lds[2 * GCAP * WAVES_PER_BLOCK];
-    const int bkt = (int)((blockIdx.x * (int64_t)BLOCK + threadIdx.x) >> 6);
+    /* ONE BLOCK per bucket, 4 waves cooperating:
+     *  pass 1: whole-bucket bin counts (LDS atomics) -> binoffs/binlens
+     *  per supertile: per-supertile counts + prefix, ranked placement into
+     *  LDS staging (bit-ballot same-bin masks, wave-serialized cursors keep
+     *  row order), then a bucket-major flush so writes coalesce. */
+    __shared__ uint32_t cnt[GCAP];    /* whole-bucket bin counts */
+    __shared__ uint32_t gcur[GCAP];   /* global (bucket-region) bin cursors */
+    __shared__ uint32_t stcnt[GCAP];  /* per-supertile bin counts */
+    __shared__ uint32_t stoffs[GCAP]; /* per-supertile bin prefix */
+    __shared__ uint32_t stcur[GCAP];
+    __shared__ uint32_t s_ridx[ST_RECORDS];
+    __shared__ uint32_t s_dest[ST_RECORDS];
+    __shared__ double s_val[ST_RECORDS];
+    __shared__ uint32_t s_total;
+
+    const int bkt = blockIdx.x;
     const int lane = threadIdx.x & 63;
-    const int wave = (threadIdx.x >> 6);
-    uint32_t* cnt = lds + 2 * GCAP * wave;
-    uint32_t* cur = cnt + GCAP;
-    if (bkt >= NB) return;
+    const int wave = threadIdx.x >> 6;
     const uint32_t lo = bucket_base[bkt];
     const uint32_t hi = bucket_base[bkt + 1];
     const int nk = fc.k_hi - fc.k_lo;
     const int glocal_n = (fc.w_hi - fc.w_lo) * nk;
-    for (int g = lane; g < GCAP; g += 64) cnt[g] = 0;
-    if (lo == hi) { /* still publish empty bins for the fold */
-        for (int g = lane; g < glocal_n; g += 64) {
-            binoffs[(int64_t)bkt * GCAP + g] = 0;
-            binlens[(int64_t)bkt * GCAP + g] = 0;
-        }
-        return;
-    }
-    /* pass 1: bin counts */
-    for (uint32_t base = lo + lane; base < hi; base += 64) {
-        const uint32_t ms_ = rmeta[base];
+    for (int g = threadIdx.x; g < GCAP; g += BLOCK) cnt[g] = 0;
+    __syncthreads();
+    /* pass 1: whole-bucket bin counts */
+    for (uint32_t i = lo + threadIdx.x; i < hi; i += BLOCK) {
+        const uint32_t ms_ = rmeta[i];
         const int widx = (int)((ms_ >> META_WIDX_SHIFT) & META_WIDX_MASK);
         const int kloc = (int)(ms_ & META_KLOC_MASK);
         if (widx >= fc.w_lo && widx < fc.w_hi && kloc >= fc.k_lo && kloc < fc.k_hi)
             atomicAdd(&cnt[(widx - fc.w_lo) * nk + (kloc - fc.k_lo)], 1u);
     }
-    /* exclusive prefix over GCAP bins (wave-serial by lane 0 is fine: tiny) */
-    if (lane == 0) {
+    __syncthreads();
+    if (threadIdx.x == 0) { /* tiny exclusive prefix over <=GCAP bins */
         uint32_t run = 0;
         for (int g = 0; g < GCAP; g++) {
             uint32_t t = cnt[g];
-            cur[g] = run;
+            gcur[g] = run;
             run += t;
         }
     }
-    /* publish segment layout for the fold */
-    for (int g = lane; g < glocal_n; g += 64) {
-        binoffs[(int64_t)bkt * GCAP + g] = cur[g];
+    __syncthreads();
+    for (int g = threadIdx.x; g < glocal_n; g += BLOCK) {
+        binoffs[(int64_t)bkt * GCAP + g] = gcur[g];
         binlens[(int64_t)bkt * GCAP + g] = cnt[g];
     }
-    /* pass 2: ranked placement (row order within each group) */
-    for (uint32_t base = lo; base < hi; base += 64) {
-        const int nt = (int)min(64u, hi - base);
-        uint32_t g = 0xFFu; /* sentinel: record outside this chunk */
-        uint32_t ri = 0;
-        double v = 0.0;
-        if (lane < nt) {
-            const uint32_t ms_ = rmeta[base + lane];
+    if (lo == hi) return;
+
+    for (uint32_t st0 = lo; st0 < hi; st0 += ST_RECORDS) {
+        const uint32_t st1 = min(hi, st0 + (uint32_t)ST_RECORDS);
+        /* per-supertile counts */
+        for (int g = threadIdx.x; g < GCAP; g += BLOCK) stcnt[g] = 0;
+        __syncthreads();
+        for (uint32_t i = st0 + threadIdx.x; i < st1; i += BLOCK) {
+            const uint32_t ms_ = rmeta[i];
             const int widx = (int)((ms_ >> META_WIDX_SHIFT) & META_WIDX_MASK);
             const int kloc = (int)(ms_ & META_KLOC_MASK);
             if (widx >= fc.w_lo && widx < fc.w_hi && kloc >= fc.k_lo &&
-                kloc < fc.k_hi) {
-                g = (uint32_t)((widx - fc.w_lo) * nk + (kloc - fc.k_lo));
-                ri = rridx[base + lane] | ((ms_ >> META_VALID_SHIFT) << 31);
-                v = rvals[base + lane];
+                kloc < fc.k_hi)
+                atomicAdd(&stcnt[(widx - fc.w_lo) * nk + (kloc - fc.k_lo)], 1u);
+        }
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            uint32_t run = 0;
+            for (int g = 0; g < GCAP; g++) {
+                uint32_t t = stcnt[g];
+                stoffs[g] = run;
+                stcur[g] = run;
+                run += t;
+            }
+            s_total = run;
+        }
+        __syncthreads();
+        /* ranked placement into staging */
+        for (uint32_t t0 = st0; t0 < st1; t0 += BLOCK) {
+            const uint32_t i = t0 + threadIdx.x;
+            uint32_t g = 0xFFu;
+            uint32_t ri = 0;
+            double v = 0.0;
+            if (i < st1) {
+                const uint32_t ms_ = rmeta[i];
+                const int widx = (int)((ms_ >> META_WIDX_SHIFT) & META_WIDX_MASK);
+                const int kloc = (int)(ms_ & META_KLOC_MASK);
+                if (widx >= fc.w_lo && widx < fc.w_hi && kloc >= fc.k_lo &&
+                    kloc < fc.k_hi) {
+                    g = (uint32_t)((widx - fc.w_lo) * nk + (kloc - fc.k_lo));
+                    ri = rridx[i] | ((ms_ >> META_VALID_SHIFT) << 31);
+                    v = rvals[i];
+                }
+            }
+            /* same-bin mask via bit-ballots over the 8 bin-id bits */
+            uint64_t same = ~0ULL;
+            for (int b = 0; b < 8; b++) {
+                uint64_t bb = __ballot((g >> b) & 1);
+                same &= ((g >> b) & 1) ? bb : ~bb;
+            }
+            const uint64_t below = (lane == 63) ? ~0ULL : ((1ULL << (lane + 1)) - 1);
+            const int rank = (int)__popcll(same & below) - 1;
+            const int leader = __ffsll((unsigned long long)same) - 1;
+            const uint32_t wtot = (uint32_t)__popcll(same);
+            uint32_t pos = 0;
+            for (int w = 0; w < WAVES_PER_BLOCK; w++) {
+                if (wave == w && g != 0xFFu) {
+                    uint32_t pre = 0;
+                    if (lane == leader) {
+                        pre = stcur[g];
+                        stcur[g] = pre + wtot;
+                    }
+                    pre = (uint32_t)__shfl((int)pre, leader);
+                    pos = pre + (uint32_t)rank;
+                }
+                __syncthreads();
+            }
+            if (g != 0xFFu) {
+                s_ridx[pos] = ri;
+                s_val[pos] = v;
+                s_dest[pos] = lo + gcur[g] + (pos - stoffs[g]);
             }
         }
-        /* same-group lane mask via bit-ballots over the 8 bin-id bits */
-        uint64_t same = ~0ULL;
-        for (int b = 0; b < 8; b++) {
-            uint64_t bb = __ballot((g >> b) & 1);
-            same &= ((g >> b) & 1) ? bb : ~bb;
+        __syncthreads();
+        /* flush (bin-major staging => coalesced runs) */
+        const uint32_t tot = s_total;
+        for (uint32_t p = threadIdx.x; p < tot; p += BLOCK) {
+            const uint32_t d = s_dest[p];
+            gval[d] = s_val[p];
+            gridx[d] = s_ridx[p];
         }
-        const uint64_t below = (lane == 63) ? ~0ULL : ((1ULL << (lane + 1)) - 1);
-        const int rank = (int)__popcll(same & below) - 1;
-        const int leader = __ffsll((unsigned long long)same) - 1;
-        uint32_t pos = 0;
-        if (g != 0xFFu) {
-            pos = cur[g] + (uint32_t)rank;
-            if (lane == leader) cur[g] += (uint32_t)__popcll(same);
-            gval[lo + pos] = v;
-            gridx[lo + pos] = ri;
-        }
-        __builtin_amdgcn_wave_barrier(); /* keep cur updates tile-ordered */
+        __syncthreads();
+        for (int g = threadIdx.x; g < GCAP; g += BLOCK) gcur[g] += stcnt[g];
+        __syncthreads();
     }
 }
 
@@ -491,9 +546,9 @@ void launch_regroup(hipStream_t s, const uint32_t* d_meta, const uint32_t* d_rid
                     const double* d_rvals, const uint32_t* d_bucket_base,
                     const FoldChunk& fc, uint32_t* d_binoffs, uint32_t* d_binlens,
                     double* d_gval, uint32_t* d_gridx) {
-    hipLaunchKernelGGL(k_regroup, dim3(NB / WAVES_PER_BLOCK), dim3(BLOCK), 0, s,
-                       d_meta, d_ridx, d_rvals, d_bucket_base, fc, d_binoffs,
-                       d_binlens, d_gval, d_gridx);
+    hipLaunchKernelGGL(k_regroup, dim3(NB), dim3(BLOCK), 0, s, d_meta, d_ridx,
+                       d_rvals, d_bucket_base, fc, d_binoffs, d_binlens, d_gval,
+                       d_gridx);
 }
 
 /* ------------------------------------------------------------------ */
