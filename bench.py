@@ -167,7 +167,9 @@ def main():
             dist.all_reduce(wm, op=dist.ReduceOp.MAX)
             op.advance_watermark(int(wm.item()))
         emitted = 0
-        for b in op.poll_all():
+        # non-blocking poll: emission is pipelined on the op's worker thread
+        # and overlaps the next step's kernels; finish() drains the tail.
+        for b in op.poll_all(drain=False):
             emitted += b["n_rows"]
         return emitted
 

@@ -93,6 +93,7 @@ def lib():
         L.dz_window_op_push_device.argtypes = [p, i64, p, p, p]
         L.dz_window_op_poll.argtypes = [p, ctypes.POINTER(ctypes.POINTER(DzOutBatch))]
         L.dz_window_op_finish.argtypes = [p]
+        L.dz_window_op_drain.argtypes = [p]
         L.dz_window_op_destroy.argtypes = [p]
         L.dz_last_error.restype = ctypes.c_char_p
         L.dz_last_error.argtypes = [p]
@@ -225,7 +226,14 @@ class WindowOp:
         res["window_end"] = np.ctypeslib.as_array(ob.window_end_ms, (n,)).copy() if n else np.zeros(0, np.int64)
         return res
 
-    def poll_all(self):
+    def drain(self):
+        """Wait until every triggered window's batch is pollable (emission is
+        pipelined on a worker thread)."""
+        self._check(self._L.dz_window_op_drain(self._h), "drain")
+
+    def poll_all(self, drain=True):
+        if drain:
+            self.drain()
         out = []
         while True:
             b = self.poll()

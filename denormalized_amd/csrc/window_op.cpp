@@ -16,6 +16,9 @@
 #include <map>
 #include <string>
 #include <unordered_map>
+#include <atomic>
+#include <condition_variable>
+#include <mutex>
 #include <thread>
 #include <vector>
 
@@ -105,9 +108,11 @@ struct dz_window_op {
      * is NOT assumed — emission sorts by first row; the dict only maps
      * key values <-> dense ids) */
     std::unordered_map<std::string, int32_t> dict_utf8;
-    std::vector<std::string> dict_strs;
+    std::deque<std::string> dict_strs;  /* deque: stable refs while the
+                                         * emission worker reads indices
+                                         * below its K snapshot */
     std::unordered_map<int64_t, int32_t> dict_i64;
-    std::vector<int64_t> dict_vals;
+    std::deque<int64_t> dict_vals;
     int64_t n_keys = 0; /* dense key count seen so far */
 
     /* persistent group state: one slab per slot, 5 contiguous fields of kcap
@@ -157,11 +162,28 @@ struct dz_window_op {
     char* h_stage = nullptr; /* pinned */
     size_t h_stage_cap = 0;
 
-    /* pinned emission staging: a small ring of whole-slot slabs so several
-     * window closes share one stream sync */
-    static constexpr int E_RING = 4;
-    uint64_t* e_ring[E_RING] = {nullptr, nullptr, nullptr, nullptr};
-    int64_t e_ring_kcap = 0;
+    /* async emission: pinned slab pool + worker thread. trigger_windows
+     * enqueues {D2H event, slab}; the worker waits the event, builds the
+     * output batch off the push critical path, and appends to outq. */
+    static constexpr int E_POOL = 8;
+    uint64_t* e_slabs[E_POOL] = {};
+    int64_t e_slab_kcap = 0;
+    struct EmitJob {
+        hipEvent_t ev;
+        int slab;
+        int64_t wstart, wend, n_keys, kcap;
+    };
+    std::deque<EmitJob> e_jobs;     /* guarded by e_mtx */
+    std::vector<int> e_free;        /* free slab indices, guarded by e_mtx */
+    std::vector<hipEvent_t> e_ev_pool; /* events for emission jobs (e_mtx) */
+    std::mutex e_mtx;
+    std::condition_variable e_cv;   /* job added / slab freed / drained */
+    std::thread e_worker;
+    bool e_stop = false;
+    int e_inflight = 0;
+    std::atomic<uint64_t> e_build_ns{0};
+    std::atomic<uint64_t> e_builds{0};
+    std::mutex out_mtx;             /* guards outq */
 
     /* filter pushdown */
     bool has_filter = false;
@@ -196,6 +218,8 @@ struct dz_window_op {
             return;                                                        \
         }                                                                  \
     } while (0)
+
+static void emit_worker_main(dz_window_op* op);
 
 static hipEvent_t get_event(dz_window_op* op) {
     if (!op->ev_pool.empty()) {
@@ -349,6 +373,7 @@ extern "C" dz_window_op* dz_window_op_create(const dz_window_desc* desc) {
         delete op;
         return nullptr;
     }
+    op->e_worker = std::thread(emit_worker_main, op);
     return op;
 }
 
@@ -356,8 +381,19 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     if (!op) return;
     hipSetDevice(op->device);
     hipStreamSynchronize(op->stream);
+    if (op->e_worker.joinable()) {
+        {
+            std::lock_guard<std::mutex> lk(op->e_mtx);
+            op->e_stop = true;
+        }
+        op->e_cv.notify_all();
+        op->e_worker.join();
+    }
     drain_events(op, true);
     for (auto e : op->ev_pool) hipEventDestroy(e);
+    for (auto e : op->e_ev_pool) hipEventDestroy(e);
+    for (auto& s : op->e_slabs)
+        if (s) hipHostFree(s);
     hipFree(op->s_base);
     hipFree(op->d_ghist); hipFree(op->d_gofs); hipFree(op->d_total); hipFree(op->d_base);
     hipFree(op->d_meta); hipFree(op->d_ridx); hipFree(op->d_rvals);
@@ -367,8 +403,6 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_ts); hipFree(op->d_kid); hipFree(op->d_vals); hipFree(op->d_valbm);
     if (op->h_scalars) hipHostFree(op->h_scalars);
     if (op->h_stage) hipHostFree(op->h_stage);
-    for (auto& r : op->e_ring)
-        if (r) hipHostFree(r);
     hipStreamDestroy(op->stream);
     delete op;
 }
@@ -411,11 +445,12 @@ static bool filter_pass(dz_window_op* op, int64_t row_cnt,
 }
 
 /* Build one emitted batch from a pinned copy of a slot slab
- * ([cnt][first][min][max][sum], each `kcap` 8-byte entries). */
+ * ([cnt][first][min][max][sum], each `kcap` 8-byte entries). Runs on the
+ * emission worker thread: touches only immutable config, the deque-backed
+ * dictionaries (indices < the job's n_keys snapshot) and the slab. */
 static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
-                           const uint64_t* slab, OutBuf* out) {
-    int64_t K = op->n_keys;
-    int64_t kcap = op->e_ring_kcap;
+                           int64_t K, int64_t kcap, const uint64_t* slab,
+                           OutBuf* out) {
     const uint64_t* f_cnt = slab;
     const uint64_t* f_first = slab + kcap;
     const double* f_min = (const double*)(slab + 2 * kcap);
@@ -529,6 +564,50 @@ static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
     *out = std::move(ob);
 }
 
+static void emit_drain(dz_window_op* op) {
+    std::unique_lock<std::mutex> lk(op->e_mtx);
+    op->e_cv.wait(lk, [&] { return op->e_inflight == 0; });
+}
+
+static void emit_worker_main(dz_window_op* op) {
+    hipSetDevice(op->device);
+    for (;;) {
+        dz_window_op::EmitJob job;
+        {
+            std::unique_lock<std::mutex> lk(op->e_mtx);
+            op->e_cv.wait(lk, [&] { return op->e_stop || !op->e_jobs.empty(); });
+            if (op->e_jobs.empty()) {
+                if (op->e_stop) return;
+                continue;
+            }
+            job = op->e_jobs.front();
+            op->e_jobs.pop_front();
+        }
+        hipEventSynchronize(job.ev);
+        auto t0 = std::chrono::steady_clock::now();
+        OutBuf ob;
+        build_emission(op, job.wstart, job.wend, job.n_keys, job.kcap,
+                       op->e_slabs[job.slab], &ob);
+        op->e_build_ns += (uint64_t)std::chrono::duration_cast<std::chrono::nanoseconds>(
+            std::chrono::steady_clock::now() - t0).count();
+        op->e_builds++;
+        {
+            std::lock_guard<std::mutex> lk(op->out_mtx);
+            op->outq.push_back(std::move(ob));
+        }
+        {
+            std::lock_guard<std::mutex> lk(op->e_mtx);
+            op->e_free.push_back(job.slab);
+            op->e_ev_pool.push_back(job.ev);
+            op->e_inflight--;
+        }
+        op->e_cv.notify_all();
+    }
+}
+
+/* trigger_windows (grouped_window_agg_stream.rs:220-253): closed windows are
+ * copied D2H asynchronously and built by the worker thread off the push
+ * critical path; dz_window_op_drain/finish wait for completion. */
 static dz_status trigger_windows(dz_window_op* op) {
     if (!op->has_wm) return DZ_OK;
     struct Closed { int64_t start, end; int32_t slot; };
@@ -542,44 +621,45 @@ static dz_status trigger_windows(dz_window_op* op) {
         }
     }
     if (closed.empty()) return DZ_OK;
-    /* (re)size the pinned ring */
-    if (op->e_ring_kcap != op->kcap) {
-        for (auto& r : op->e_ring)
-            if (r) { hipHostFree(r); r = nullptr; }
-        for (auto& r : op->e_ring)
-            CHK(op, hipHostMalloc((void**)&r, (size_t)op->kcap * 5 * 8));
-        op->e_ring_kcap = op->kcap;
+    if (op->e_slab_kcap != op->kcap) {
+        emit_drain(op);
+        for (auto& s : op->e_slabs) {
+            if (s) hipHostFree(s);
+            CHK(op, hipHostMalloc((void**)&s, (size_t)op->kcap * 5 * 8));
+        }
+        std::lock_guard<std::mutex> lk(op->e_mtx);
+        op->e_free.clear();
+        for (int i = 0; i < dz_window_op::E_POOL; i++) op->e_free.push_back(i);
+        op->e_slab_kcap = op->kcap;
     }
     size_t stride = (size_t)op->kcap * 5;
-    for (size_t g = 0; g < closed.size(); g += dz_window_op::E_RING) {
-        size_t gn = std::min<size_t>(dz_window_op::E_RING, closed.size() - g);
+    HostTimer ht(op, "h_emit_enqueue");
+    for (auto& c : closed) {
+        int slab;
+        hipEvent_t ev;
         {
-            HostTimer ht(op, "h_emit_d2h");
-            for (size_t i = 0; i < gn; i++)
-                CHK(op, hipMemcpyAsync(op->e_ring[i],
-                                       op->s_base + (size_t)closed[g + i].slot * stride,
-                                       stride * 8, hipMemcpyDeviceToHost, op->stream));
-            CHK(op, hipStreamSynchronize(op->stream));
-            drain_events(op, false);
+            std::unique_lock<std::mutex> lk(op->e_mtx);
+            op->e_cv.wait(lk, [&] { return !op->e_free.empty(); });
+            slab = op->e_free.back();
+            op->e_free.pop_back();
+            if (!op->e_ev_pool.empty()) {
+                ev = op->e_ev_pool.back();
+                op->e_ev_pool.pop_back();
+            } else {
+                hipEventCreate(&ev);
+            }
+            op->e_inflight++;
         }
+        CHK(op, hipMemcpyAsync(op->e_slabs[slab], op->s_base + (size_t)c.slot * stride,
+                               stride * 8, hipMemcpyDeviceToHost, op->stream));
+        CHK(op, hipEventRecord(ev, op->stream));
+        if (slot_reset(op, c.slot) != DZ_OK) return DZ_ERR;
+        op->free_slots.push_back(c.slot);
         {
-            /* builds are independent per window: run them on host threads */
-            HostTimer ht(op, "h_emit_build");
-            std::vector<OutBuf> obs(gn);
-            std::vector<std::thread> th;
-            for (size_t i = 1; i < gn; i++)
-                th.emplace_back(build_emission, op, closed[g + i].start,
-                                closed[g + i].end, op->e_ring[i], &obs[i]);
-            build_emission(op, closed[g].start, closed[g].end, op->e_ring[0],
-                           &obs[0]);
-            for (auto& t : th) t.join();
-            for (size_t i = 0; i < gn; i++)
-                op->outq.push_back(std::move(obs[i]));
+            std::lock_guard<std::mutex> lk(op->e_mtx);
+            op->e_jobs.push_back({ev, slab, c.start, c.end, op->n_keys, op->kcap});
         }
-        for (size_t i = 0; i < gn; i++) {
-            if (slot_reset(op, closed[g + i].slot) != DZ_OK) return DZ_ERR;
-            op->free_slots.push_back(closed[g + i].slot);
-        }
+        op->e_cv.notify_all();
     }
     return DZ_OK;
 }
@@ -883,9 +963,12 @@ extern "C" dz_status dz_window_op_push(dz_window_op* op, const dz_batch* batch) 
 extern "C" dz_status dz_window_op_poll(dz_window_op* op, const dz_out_batch** out) {
     if (!op || !out) return DZ_ERR;
     *out = nullptr;
-    if (op->outq.empty()) return DZ_OK;
-    op->current = std::move(op->outq.front());
-    op->outq.pop_front();
+    {
+        std::lock_guard<std::mutex> lk(op->out_mtx);
+        if (op->outq.empty()) return DZ_OK;
+        op->current = std::move(op->outq.front());
+        op->outq.pop_front();
+    }
     op->has_current = true;
     OutBuf& ob = op->current;
     ob.agg_ptrs.clear();
@@ -916,8 +999,15 @@ extern "C" dz_status dz_window_op_finish(dz_window_op* op) {
         op->has_wm = true;
     }
     if (trigger_windows(op) != DZ_OK) return DZ_ERR;
+    emit_drain(op);
     CHK(op, hipStreamSynchronize(op->stream));
     drain_events(op, true);
+    return DZ_OK;
+}
+
+extern "C" dz_status dz_window_op_drain(dz_window_op* op) {
+    if (!op) return DZ_ERR;
+    emit_drain(op);
     return DZ_OK;
 }
 
@@ -957,8 +1047,14 @@ extern "C" dz_status dz_window_op_kernel_stats(dz_window_op* op,
                                                dz_kernel_stat* out, int32_t cap,
                                                int32_t* n_out) {
     if (!op) return DZ_ERR;
+    emit_drain(op);
     CHK(op, hipStreamSynchronize(op->stream));
     drain_events(op, true);
+    {
+        KStatAcc& s = op->stats["h_emit_build"];
+        s.launches = op->e_builds.load();
+        s.ms = op->e_build_ns.load() / 1e6;
+    }
     int32_t n = 0;
     for (auto& kv : op->stats) {
         if (n >= cap) break;

@@ -156,6 +156,12 @@ dz_status dz_window_op_poll(dz_window_op* op, const dz_out_batch** out);
  * reference stream is unbounded and only closes on watermark advance). */
 dz_status dz_window_op_finish(dz_window_op* op);
 
+/* Emission is pipelined off the push path (a worker thread builds output
+ * batches while later pushes compute). Blocks until every window already
+ * triggered is available to poll — the synchronous-poll behaviour of the
+ * reference stream when a consumer needs it. finish() implies drain. */
+dz_status dz_window_op_drain(dz_window_op* op);
+
 void dz_window_op_destroy(dz_window_op* op);
 
 /* DataFusionError-by-value analog (crates/common/src/error/mod.rs:13-44). */
