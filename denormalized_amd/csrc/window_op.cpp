@@ -165,22 +165,27 @@ struct dz_window_op {
     /* async emission: pinned slab pool + worker thread. trigger_windows
      * enqueues {D2H event, slab}; the worker waits the event, builds the
      * output batch off the push critical path, and appends to outq. */
-    static constexpr int E_POOL = 8;
+    static constexpr int E_POOL = 16;
+    static constexpr int E_WORKERS = 4;
     uint64_t* e_slabs[E_POOL] = {};
     int64_t e_slab_kcap = 0;
     struct EmitJob {
         hipEvent_t ev;
         int slab;
         int64_t wstart, wend, n_keys, kcap;
+        uint64_t ticket;
     };
     std::deque<EmitJob> e_jobs;     /* guarded by e_mtx */
     std::vector<int> e_free;        /* free slab indices, guarded by e_mtx */
     std::vector<hipEvent_t> e_ev_pool; /* events for emission jobs (e_mtx) */
     std::mutex e_mtx;
     std::condition_variable e_cv;   /* job added / slab freed / drained */
-    std::thread e_worker;
+    std::vector<std::thread> e_workers;
     bool e_stop = false;
     int e_inflight = 0;
+    uint64_t e_ticket_next = 0;      /* next ticket to assign (e_mtx) */
+    uint64_t e_ticket_pop = 0;       /* next ticket poll may emit (out_mtx) */
+    std::map<uint64_t, OutBuf> e_done; /* completed out-of-order (out_mtx) */
     std::atomic<uint64_t> e_build_ns{0};
     std::atomic<uint64_t> e_builds{0};
     std::mutex out_mtx;             /* guards outq */
@@ -373,7 +378,8 @@ extern "C" dz_window_op* dz_window_op_create(const dz_window_desc* desc) {
         delete op;
         return nullptr;
     }
-    op->e_worker = std::thread(emit_worker_main, op);
+    for (int i = 0; i < dz_window_op::E_WORKERS; i++)
+        op->e_workers.emplace_back(emit_worker_main, op);
     return op;
 }
 
@@ -381,14 +387,13 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     if (!op) return;
     hipSetDevice(op->device);
     hipStreamSynchronize(op->stream);
-    if (op->e_worker.joinable()) {
-        {
-            std::lock_guard<std::mutex> lk(op->e_mtx);
-            op->e_stop = true;
-        }
-        op->e_cv.notify_all();
-        op->e_worker.join();
+    {
+        std::lock_guard<std::mutex> lk(op->e_mtx);
+        op->e_stop = true;
     }
+    op->e_cv.notify_all();
+    for (auto& w : op->e_workers)
+        if (w.joinable()) w.join();
     drain_events(op, true);
     for (auto e : op->ev_pool) hipEventDestroy(e);
     for (auto e : op->e_ev_pool) hipEventDestroy(e);
@@ -593,7 +598,13 @@ static void emit_worker_main(dz_window_op* op) {
         op->e_builds++;
         {
             std::lock_guard<std::mutex> lk(op->out_mtx);
-            op->outq.push_back(std::move(ob));
+            op->e_done.emplace(job.ticket, std::move(ob));
+            while (!op->e_done.empty() &&
+                   op->e_done.begin()->first == op->e_ticket_pop) {
+                op->outq.push_back(std::move(op->e_done.begin()->second));
+                op->e_done.erase(op->e_done.begin());
+                op->e_ticket_pop++;
+            }
         }
         {
             std::lock_guard<std::mutex> lk(op->e_mtx);
@@ -657,7 +668,8 @@ static dz_status trigger_windows(dz_window_op* op) {
         op->free_slots.push_back(c.slot);
         {
             std::lock_guard<std::mutex> lk(op->e_mtx);
-            op->e_jobs.push_back({ev, slab, c.start, c.end, op->n_keys, op->kcap});
+            op->e_jobs.push_back({ev, slab, c.start, c.end, op->n_keys, op->kcap,
+                                  op->e_ticket_next++});
         }
         op->e_cv.notify_all();
     }
