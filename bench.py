@@ -45,6 +45,8 @@ def parse_args():
     p.add_argument("--slide-ms", type=int, default=0)
     p.add_argument("--seed", type=int, default=42)
     p.add_argument("--no-filter", action="store_true")
+    p.add_argument("--dist-backend", default=None,
+                   help="torch.distributed backend override (default: nccl on GPU)")
     p.add_argument("--skip-cpu-baseline", action="store_true")
     p.add_argument("--cpu-sample-rows", type=int, default=8_000_000)
     return p.parse_args()
@@ -119,9 +121,9 @@ def main():
     if world > 1:
         import torch
         import torch.distributed as dist
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        backend = args.dist_backend or ("nccl" if torch.cuda.is_available() else "gloo")
         dist.init_process_group(backend=backend)
-        if torch.cuda.is_available():
+        if torch.cuda.is_available() and backend == "nccl":
             torch.cuda.set_device(local_rank)
     else:
         dist = None
@@ -130,6 +132,9 @@ def main():
     from denormalized_amd import _lib
 
     device = local_rank
+    if world > 1:
+        import torch
+        device = local_rank % max(1, torch.cuda.device_count())
     K, W = args.steps, args.warmup
     B = args.rows_per_step
     total_rows = (K + W) * B
@@ -162,7 +167,7 @@ def main():
         if dist is not None:
             import torch
             wm = torch.tensor([op.watermark], dtype=torch.int64)
-            if torch.cuda.is_available():
+            if torch.cuda.is_available() and dist.get_backend() == "nccl":
                 wm = wm.cuda()
             dist.all_reduce(wm, op=dist.ReduceOp.MAX)
             op.advance_watermark(int(wm.item()))
@@ -194,7 +199,7 @@ def main():
     if dist is not None:
         import torch
         e = torch.tensor([elapsed], dtype=torch.float64)
-        if torch.cuda.is_available():
+        if torch.cuda.is_available() and dist.get_backend() == "nccl":
             e = e.cuda()
         dist.all_reduce(e, op=dist.ReduceOp.MAX)
         elapsed = float(e.item())

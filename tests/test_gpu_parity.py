@@ -390,3 +390,53 @@ def test_randomized_stress_matrix():
             batches.append((ts.astype(np.int64), k, v))
         outs, exp = run_both(len_ms, slide, batches)
         assert_parity(outs, exp)
+
+
+def test_large_keyspace_multichunk_fold():
+    # 100k keys => klocs ~ 196 > FOLD_GCAP: exercises the k_lo chunking in
+    # push_core (multiple regroup+fold launches per batch)
+    rng = np.random.default_rng(77)
+    n = 400_000
+    ts = (1_000_000 + np.arange(n) // 400).astype(np.int64)
+    k = rng.integers(0, 100_000, n)
+    v = rng.uniform(0, 115, n)
+    outs, exp = run_both(1000, 0, [(ts, k, v)], n_keys_hint=100_000)
+    assert len(exp["key"]) > 50_000
+    assert_parity(outs, exp)
+
+
+def test_cfg3_shape_sliding_large_keys():
+    # cfg3 shape at reduced size: sliding 500/100, wide int keyspace
+    rng = np.random.default_rng(78)
+    n = 300_000
+    ts = (1_000_000 + np.arange(n) // 300).astype(np.int64)
+    k = rng.integers(0, 60_000, n)
+    v = rng.uniform(0, 115, n)
+    outs, exp = run_both(500, 100, [(ts, k, v)], n_keys_hint=60_000)
+    assert_parity(outs, exp)
+
+
+def test_key_skew_zipf_and_single_key():
+    rng = np.random.default_rng(79)
+    n = 200_000
+    ts = (1_000_000 + np.arange(n) // 200).astype(np.int64)
+    # heavy zipf skew: one bucket's segments much longer than others
+    k = np.minimum(rng.zipf(1.3, n) - 1, 499).astype(np.int64)
+    v = rng.uniform(0, 115, n)
+    outs, exp = run_both(1000, 0, [(ts, k, v)], n_keys_hint=500)
+    assert_parity(outs, exp)
+    # pathological single key: one group per window folds everything
+    k1 = np.zeros(n, np.int64)
+    outs, exp = run_both(1000, 0, [(ts, k1, v)], n_keys_hint=4)
+    assert_parity(outs, exp)
+
+
+def test_supertile_boundary_sizes():
+    # sizes straddling ST_RECORDS (2048) and chunk boundaries
+    rng = np.random.default_rng(80)
+    for n in (1, 63, 64, 65, 2047, 2048, 2049, 8193, 100_000):
+        ts = (1_000_000 + np.arange(n) // 50).astype(np.int64)
+        k = rng.integers(0, 97, n)
+        v = rng.uniform(0, 115, n)
+        outs, exp = run_both(1000, 0, [(ts, k, v)])
+        assert_parity(outs, exp)
