@@ -51,7 +51,7 @@ void launch_minmax(hipStream_t stream, const int64_t* d_ts, const int32_t* d_kid
 
 void launch_hist(hipStream_t stream, const int32_t* d_kid, const int64_t* d_ts,
                  int64_t n, int64_t chunk, int C, const WinParams& wp,
-                 uint32_t* d_ghist);
+                 uint32_t* d_ghist, uint64_t* d_scalars /*fused minmax or null*/);
 
 constexpr int SCAN_SSPLIT = 32; /* d_psum is SCAN_SSPLIT * NB u32 */
 void launch_scan(hipStream_t stream, const uint32_t* d_ghist, int C,

@@ -161,27 +161,68 @@ void launch_minmax(hipStream_t s, const int64_t* d_ts, const int32_t* d_kid,
 
 __global__ __launch_bounds__(BLOCK) void k_hist(const int32_t* kid,
         const int64_t* ts, int64_t n, int64_t chunk, WinParams wp,
-        uint32_t* ghist) {
+        uint32_t* ghist, uint64_t* scalars) {
+    /* bucket histogram; when `scalars` is non-null this launch ALSO reduces
+     * the batch min/max timestamp + max key id (the tumbling fast path fuses
+     * the watermark pass: one 12 B/row read instead of two). Tumbling rows
+     * have multiplicity 1, so ts is only read when reducing or sliding. */
     __shared__ uint32_t h[NB];
+    __shared__ uint64_t red[3][WAVES_PER_BLOCK];
     for (int t = threadIdx.x; t < NB; t += BLOCK) h[t] = 0;
     __syncthreads();
     int64_t lo = blockIdx.x * chunk;
     int64_t hi = i64min(n, lo + chunk);
+    uint64_t mn = ~0ULL, mx = 0, km = 0;
     for (int64_t i = lo + threadIdx.x; i < hi; i += BLOCK) {
-        int32_t jm, m;
-        row_windows(ts[i], wp, &jm, &m);
-        atomicAdd(&h[kid[i] & (NB - 1)], (uint32_t)m);
+        uint32_t k = (uint32_t)kid[i];
+        uint32_t m = 1;
+        if (wp.is_sliding) {
+            int32_t jm, mm;
+            row_windows(ts[i], wp, &jm, &mm);
+            m = (uint32_t)mm;
+        }
+        if (scalars) {
+            uint64_t t = map_i64(ts[i]);
+            mn = min(mn, t);
+            mx = max(mx, t);
+            km = max(km, (uint64_t)k);
+        }
+        if (m) atomicAdd(&h[k & (NB - 1)], m);
     }
     __syncthreads();
     for (int t = threadIdx.x; t < NB; t += BLOCK)
         ghist[(int64_t)blockIdx.x * NB + t] = h[t];
+    if (scalars) {
+        for (int o = 32; o > 0; o >>= 1) {
+            mn = min(mn, (uint64_t)__shfl_down((unsigned long long)mn, o));
+            mx = max(mx, (uint64_t)__shfl_down((unsigned long long)mx, o));
+            km = max(km, (uint64_t)__shfl_down((unsigned long long)km, o));
+        }
+        const int wave = threadIdx.x >> 6;
+        if ((threadIdx.x & 63) == 0) {
+            red[0][wave] = mn;
+            red[1][wave] = mx;
+            red[2][wave] = km;
+        }
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            for (int w = 1; w < WAVES_PER_BLOCK; w++) {
+                mn = min(mn, red[0][w]);
+                mx = max(mx, red[1][w]);
+                km = max(km, red[2][w]);
+            }
+            atomicMin((unsigned long long*)&scalars[0], (unsigned long long)mn);
+            atomicMax((unsigned long long*)&scalars[1], (unsigned long long)mx);
+            atomicMax((unsigned long long*)&scalars[2], (unsigned long long)km);
+        }
+    }
 }
 
 void launch_hist(hipStream_t s, const int32_t* d_kid, const int64_t* d_ts,
                  int64_t n, int64_t chunk, int C, const WinParams& wp,
-                 uint32_t* d_ghist) {
+                 uint32_t* d_ghist, uint64_t* d_scalars) {
     hipLaunchKernelGGL(k_hist, dim3(C), dim3(BLOCK), 0, s, d_kid, d_ts, n, chunk,
-                       wp, d_ghist);
+                       wp, d_ghist, d_scalars);
 }
 
 /* ------------------------------------------------------------------ */
@@ -290,9 +331,13 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
         for (int t = threadIdx.x; t < NB; t += BLOCK) cnt[t] = 0;
         __syncthreads();
         for (int64_t i = st0 + threadIdx.x; i < st1; i += BLOCK) {
-            int32_t jm, m;
-            row_windows(ts[i], wp, &jm, &m);
-            if (m > 0) atomicAdd(&cnt[kid[i] & (NB - 1)], (uint32_t)m);
+            uint32_t m = 1;
+            if (wp.is_sliding) {
+                int32_t jm, mm;
+                row_windows(ts[i], wp, &jm, &mm);
+                m = (uint32_t)mm;
+            }
+            if (m) atomicAdd(&cnt[kid[i] & (NB - 1)], m);
         }
         __syncthreads();
         /* block-level exclusive prefix over NB bins */

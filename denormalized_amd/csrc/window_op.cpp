@@ -103,6 +103,7 @@ struct dz_window_op {
     std::string err;
 
     hipStream_t stream = nullptr;
+    hipStream_t copy_stream = nullptr; /* emission D2H runs off the compute stream */
 
     /* dictionary (host side; first-seen insertion order == emitted id order
      * is NOT assumed — emission sorts by first row; the dict only maps
@@ -126,7 +127,8 @@ struct dz_window_op {
     double* s_min = nullptr;
     double* s_max = nullptr;
     double* s_sum = nullptr;
-    std::vector<int32_t> free_slots;
+    struct FreeSlot { int32_t slot; hipEvent_t ev; /* copy-done gate or null */ };
+    std::vector<FreeSlot> free_slots;
     struct OpenWin { int64_t end; int32_t slot; };
     std::map<int64_t, OpenWin> open; /* by window start (BTreeMap order) */
 
@@ -318,11 +320,13 @@ static dz_status state_alloc(dz_window_op* op, int64_t kcap_new, int32_t nslots_
                                    span * 8, hipMemcpyDeviceToDevice, op->stream));
     }
     CHK(op, hipStreamSynchronize(op->stream));
+    CHK(op, hipStreamSynchronize(op->copy_stream)); /* pending emission copies read s_base */
     hipFree(op->s_base);
     op->s_base = n_base;
     op->s_cnt = n_cnt; op->s_first = n_first; op->s_min = n_min;
     op->s_max = n_max; op->s_sum = n_sum;
-    for (int32_t s = op->nslots; s < nslots_new; s++) op->free_slots.push_back(s);
+    for (int32_t s = op->nslots; s < nslots_new; s++)
+        op->free_slots.push_back({s, nullptr});
     op->kcap = kcap_new;
     op->nslots = nslots_new;
     return DZ_OK;
@@ -364,7 +368,8 @@ extern "C" dz_window_op* dz_window_op_create(const dz_window_desc* desc) {
     op->device = desc->device;
     op->max_open = desc->max_open_windows > 0 ? desc->max_open_windows : 4096;
     if (hipSetDevice(op->device) != hipSuccess ||
-        hipStreamCreate(&op->stream) != hipSuccess) {
+        hipStreamCreate(&op->stream) != hipSuccess ||
+        hipStreamCreate(&op->copy_stream) != hipSuccess) {
         g_err = "hip device/stream init failed";
         delete op;
         return nullptr;
@@ -409,6 +414,9 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     if (op->h_scalars) hipHostFree(op->h_scalars);
     if (op->h_stage) hipHostFree(op->h_stage);
     hipStreamDestroy(op->stream);
+    hipStreamDestroy(op->copy_stream);
+    for (auto& fs : op->free_slots)
+        if (fs.ev) hipEventDestroy(fs.ev);
     delete op;
 }
 
@@ -645,27 +653,50 @@ static dz_status trigger_windows(dz_window_op* op) {
     }
     size_t stride = (size_t)op->kcap * 5;
     HostTimer ht(op, "h_emit_enqueue");
+    /* copies run on the copy stream AFTER the compute stream's folds */
+    hipEvent_t evA;
+    {
+        std::lock_guard<std::mutex> lk(op->e_mtx);
+        if (!op->e_ev_pool.empty()) {
+            evA = op->e_ev_pool.back();
+            op->e_ev_pool.pop_back();
+        } else {
+            hipEventCreate(&evA);
+        }
+    }
+    CHK(op, hipEventRecord(evA, op->stream));
+    CHK(op, hipStreamWaitEvent(op->copy_stream, evA, 0));
+    {
+        std::lock_guard<std::mutex> lk(op->e_mtx);
+        op->e_ev_pool.push_back(evA);
+    }
     for (auto& c : closed) {
         int slab;
-        hipEvent_t ev;
+        hipEvent_t ev, slot_ev;
         {
             std::unique_lock<std::mutex> lk(op->e_mtx);
             op->e_cv.wait(lk, [&] { return !op->e_free.empty(); });
             slab = op->e_free.back();
             op->e_free.pop_back();
-            if (!op->e_ev_pool.empty()) {
-                ev = op->e_ev_pool.back();
-                op->e_ev_pool.pop_back();
-            } else {
-                hipEventCreate(&ev);
-            }
+            auto take = [&]() -> hipEvent_t {
+                if (!op->e_ev_pool.empty()) {
+                    hipEvent_t e = op->e_ev_pool.back();
+                    op->e_ev_pool.pop_back();
+                    return e;
+                }
+                hipEvent_t e;
+                hipEventCreate(&e);
+                return e;
+            };
+            ev = take();
+            slot_ev = take();
             op->e_inflight++;
         }
         CHK(op, hipMemcpyAsync(op->e_slabs[slab], op->s_base + (size_t)c.slot * stride,
-                               stride * 8, hipMemcpyDeviceToHost, op->stream));
-        CHK(op, hipEventRecord(ev, op->stream));
-        if (slot_reset(op, c.slot) != DZ_OK) return DZ_ERR;
-        op->free_slots.push_back(c.slot);
+                               stride * 8, hipMemcpyDeviceToHost, op->copy_stream));
+        CHK(op, hipEventRecord(ev, op->copy_stream));
+        CHK(op, hipEventRecord(slot_ev, op->copy_stream));
+        op->free_slots.push_back({c.slot, slot_ev});
         {
             std::lock_guard<std::mutex> lk(op->e_mtx);
             op->e_jobs.push_back({ev, slab, c.start, c.end, op->n_keys, op->kcap,
@@ -718,13 +749,37 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
     }
     CHK(op, hipSetDevice(op->device));
 
-    /* 1. batch watermark bounds + max key id (time.rs:31-57) */
+    /* 1. batch watermark bounds + bucket histogram.
+     * Tumbling fast path: one fused 12 B/row pass computes the histogram AND
+     * min/max ts + max key id (the histogram needs no window params when
+     * every row has multiplicity 1). Sliding needs min/max first (the window
+     * grid anchors multiplicity), so it keeps a separate reduction pass. */
+    const bool sliding = op->slide_ms > 0;
+    int C = (int)std::min<int64_t>(512, std::max<int64_t>(1, (n + 8191) / 8192));
+    int64_t chunk = (n + C - 1) / C;
     CHK(op, hipMemsetAsync(op->d_scalars, 0xFF, 8, op->stream));
     CHK(op, hipMemsetAsync(op->d_scalars + 1, 0, 16, op->stream));
-    timed(op, "minmax", (double)n * 12, [&] {
-        dz::launch_minmax(op->stream, d_ts, keys_are_dense ? d_kid : nullptr, n,
-                          op->d_scalars);
-    });
+    dz::WinParams wp;
+    memset(&wp, 0, sizeof(wp));
+    wp.len_ms = op->len_ms;
+    wp.slide_ms = op->slide_ms;
+    wp.is_sliding = sliding;
+    if (!sliding) {
+        if (ensure_scratch(op, C, n) != DZ_OK) return DZ_ERR;
+        timed(op, "hist", (double)n * 12, [&] {
+            dz::launch_hist(op->stream, d_kid, d_ts, n, chunk, C, wp,
+                            op->d_ghist, op->d_scalars);
+        });
+        timed(op, "scan", (double)C * dz::NB * 12, [&] {
+            dz::launch_scan(op->stream, op->d_ghist, C, op->d_total, op->d_base,
+                            op->d_gofs);
+        });
+    } else {
+        timed(op, "minmax", (double)n * 12, [&] {
+            dz::launch_minmax(op->stream, d_ts, keys_are_dense ? d_kid : nullptr,
+                              n, op->d_scalars);
+        });
+    }
     CHK(op, hipMemcpyAsync(op->h_scalars, op->d_scalars, 24, hipMemcpyDeviceToHost,
                            op->stream));
     {
@@ -772,11 +827,16 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
                 if (state_alloc(op, op->kcap, op->nslots + grow) != DZ_OK)
                     return DZ_ERR;
             }
-            int32_t slot = op->free_slots.back();
+            dz_window_op::FreeSlot fs = op->free_slots.back();
             op->free_slots.pop_back();
-            if (slot_reset(op, slot) != DZ_OK) return DZ_ERR;
-            op->open[ws[r]] = {we[r], slot};
-            slotmap[r] = slot;
+            if (fs.ev) { /* emission copy of this slot may still be in flight */
+                CHK(op, hipStreamWaitEvent(op->stream, fs.ev, 0));
+                std::lock_guard<std::mutex> lk(op->e_mtx);
+                op->e_ev_pool.push_back(fs.ev);
+            }
+            if (slot_reset(op, fs.slot) != DZ_OK) return DZ_ERR;
+            op->open[ws[r]] = {we[r], fs.slot};
+            slotmap[r] = fs.slot;
         } else {
             slotmap[r] = it->second.slot;
         }
@@ -790,17 +850,11 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
         CHK(op, hipMemcpyAsync(op->d_slotmap, slotmap.data(), nw * 4,
                                hipMemcpyHostToDevice, op->stream));
 
-    dz::WinParams wp;
     wp.s0 = nw > 0 ? ws[0] : 0;
-    wp.len_ms = op->len_ms;
-    wp.slide_ms = op->slide_ms;
     wp.nw = (int32_t)nw;
-    wp.is_sliding = op->slide_ms > 0;
 
     /* 3. partition + fold */
-    int C = (int)std::min<int64_t>(512, std::max<int64_t>(1, (n + 8191) / 8192));
-    int64_t chunk = (n + C - 1) / C;
-    int64_t expand = wp.is_sliding
+    int64_t expand = sliding
         ? (op->len_ms + op->slide_ms - 1) / op->slide_ms + 1 : 1;
     int64_t nrec_max = n * expand;
     if (nrec_max >= (int64_t)UINT32_MAX) {
@@ -808,16 +862,18 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
         return DZ_ERR;
     }
     if (ensure_scratch(op, C, nrec_max) != DZ_OK) return DZ_ERR;
-
-    timed(op, "hist", (double)n * 12, [&] {
-        dz::launch_hist(op->stream, d_kid, d_ts, n, chunk, C, wp, op->d_ghist);
-    });
-    timed(op, "scan", (double)C * dz::NB * 12, [&] {
-        dz::launch_scan(op->stream, op->d_ghist, C, op->d_total, op->d_base,
-                        op->d_gofs);
-    });
+    if (sliding) {
+        timed(op, "hist", (double)n * 12, [&] {
+            dz::launch_hist(op->stream, d_kid, d_ts, n, chunk, C, wp,
+                            op->d_ghist, nullptr);
+        });
+        timed(op, "scan", (double)C * dz::NB * 12, [&] {
+            dz::launch_scan(op->stream, op->d_ghist, C, op->d_total, op->d_base,
+                            op->d_gofs);
+        });
+    }
     int32_t st_rows = (int32_t)std::max<int64_t>(64, dz::ST_RECORDS / expand);
-    timed(op, "scatter", (double)n * 32 + (double)nrec_max * 16, [&] {
+    timed(op, "scatter", (double)n * 24 + (double)nrec_max * 16, [&] {
         dz::launch_scatter(op->stream, d_kid, d_ts, d_vals, d_valbm, n, chunk, C,
                            st_rows, wp, op->d_gofs, op->d_meta, op->d_ridx,
                            op->d_rvals);
