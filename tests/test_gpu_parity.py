@@ -440,3 +440,62 @@ def test_supertile_boundary_sizes():
         v = rng.uniform(0, 115, n)
         outs, exp = run_both(1000, 0, [(ts, k, v)])
         assert_parity(outs, exp)
+
+
+def test_full_cfg2_prefix_parity():
+    """BASELINE cfg2 at full scale (100M rows, 10k keys, 1s tumbling,
+    filter off): the GPU processes all 100M device-resident rows; the oracle
+    processes only the first 4M rows. Windows are independent of later rows,
+    so every window fully contained in the prefix must match BIT-EXACTLY,
+    and whole-run invariants hold (counts sum to the row total)."""
+    import ctypes
+    from denormalized_amd import DeviceArray, WindowOp, _lib, generate, synchronize
+    n = 100_000_000
+    nkeys = 10_000
+    rpm = 1000  # 1k rows/ms => 1M rows per 1s window
+    d_ts = DeviceArray(0, n * 8)
+    d_kid = DeviceArray(0, n * 4)
+    d_vals = DeviceArray(0, n * 8)
+    generate(0, 42, 1_000_000, 0, n, nkeys, rpm, d_ts.ptr, None, d_kid.ptr,
+             d_vals.ptr)
+    synchronize(0)
+    op = WindowOp(length_ms=1000, key_kind=_lib.KEY_DENSE_INT64,
+                  n_keys_hint=nkeys)
+    step = 8_000_000
+    for off in range(0, n, step):
+        m = min(step, n - off)
+        op.push_device(m,
+                       ctypes.c_void_p(d_ts.ptr.value + off * 8),
+                       ctypes.c_void_p(d_kid.ptr.value + off * 4),
+                       ctypes.c_void_p(d_vals.ptr.value + off * 8))
+    op.finish()
+    outs = op.poll_all()
+    for a in (d_ts, d_kid, d_vals):
+        a.free()
+
+    # whole-run invariants
+    total_cnt = sum(int(b["count"].sum()) for b in outs)
+    assert total_cnt == n
+    n_windows = len({int(b["window_start"][0]) for b in outs if b["n_rows"]})
+    assert n_windows == n // (rpm * 1000)
+    assert all(b["min"].min() >= 0.0 for b in outs if b["n_rows"])
+    assert all(b["max"].max() < 115.0 for b in outs if b["n_rows"])
+
+    # prefix parity: oracle over the first 4M rows covers the first 3 windows
+    pre = 4_000_000
+    ts, kid, val = pyoracle.gen(42, 1_000_000, 0, pre, nkeys, rpm)
+    o = pyoracle.Oracle(1000, 0)
+    o.push(ts, kid, val)
+    o.finish()
+    exp = o.fetch()
+    o.close()
+    full = pre // (rpm * 1000) * 1000  # ms covered fully by the prefix
+    keep = exp["window_end"] <= 1_000_000 + full
+    gpu_pref = [b for b in outs
+                if b["n_rows"] and b["window_end"][0] <= 1_000_000 + full]
+    gk = np.concatenate([b["key"] for b in gpu_pref])
+    assert np.array_equal(gk, exp["key"][keep])
+    for f in ("count", "min", "max", "avg"):
+        gf = np.concatenate([b[f] for b in gpu_pref])
+        assert np.array_equal(gf, exp[f][keep]), f
+    op.close()
