@@ -64,7 +64,7 @@ void launch_scatter(hipStream_t stream, const int32_t* d_kid, const int64_t* d_t
                     const WinParams& wp, const uint32_t* d_gofs, uint32_t* d_meta,
                     uint32_t* d_ridx, double* d_rvals);
 
-constexpr int FOLD_GCAP = 128; /* groups per bucket per fold chunk */
+constexpr int FOLD_GCAP = 256; /* groups per bucket per fold chunk */
 
 void launch_regroup(hipStream_t stream, const uint32_t* d_meta,
                     const uint32_t* d_ridx, const double* d_rvals,

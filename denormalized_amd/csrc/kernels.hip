@@ -530,7 +530,7 @@ __global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
         /* ranked placement into staging */
         for (uint32_t t0 = st0; t0 < st1; t0 += BLOCK) {
             const uint32_t i = t0 + threadIdx.x;
-            uint32_t g = 0xFFu;
+            uint32_t g = 0x1FFu; /* sentinel above GCAP-1 */
             uint32_t ri = 0;
             double v = 0.0;
             if (i < st1) {
@@ -544,9 +544,9 @@ __global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
                     v = rvals[i];
                 }
             }
-            /* same-bin mask via bit-ballots over the 8 bin-id bits */
+            /* same-bin mask via bit-ballots over the 9 bin-id bits */
             uint64_t same = ~0ULL;
-            for (int b = 0; b < 8; b++) {
+            for (int b = 0; b < 9; b++) {
                 uint64_t bb = __ballot((g >> b) & 1);
                 same &= ((g >> b) & 1) ? bb : ~bb;
             }
@@ -556,7 +556,7 @@ __global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
             const uint32_t wtot = (uint32_t)__popcll(same);
             uint32_t pos = 0;
             for (int w = 0; w < WAVES_PER_BLOCK; w++) {
-                if (wave == w && g != 0xFFu) {
+                if (wave == w && g != 0x1FFu) {
                     uint32_t pre = 0;
                     if (lane == leader) {
                         pre = stcur[g];
@@ -567,7 +567,7 @@ __global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
                 }
                 __syncthreads();
             }
-            if (g != 0xFFu) {
+            if (g != 0x1FFu) {
                 s_ridx[pos] = ri;
                 s_val[pos] = v;
                 s_dest[pos] = lo + gcur[g] + (pos - stoffs[g]);
