@@ -219,8 +219,8 @@ def main():
         achieved = ALG_BYTES_PER_ROW * rows_per_launch / avg_s
         traffic = None
         tmap = read_traffic()
-        if tmap and dom in tmap:
-            traffic = tmap[dom]
+        if tmap:
+            traffic = tmap.get("per_kernel_bytes_per_launch", {}).get(dom)
         out = {
             "metric": "rows/sec through window()+filter() on synthetic sensor stream",
             "value": value,

@@ -40,8 +40,8 @@ class DataStream:
         """datastream.rs:178-196 (StreamingLogicalPlanBuilder::streaming_window,
         logical_plan/mod.rs:27-60). group_cols: [key column name]; aggs:
         [(op_name, value_col_name), ...]."""
-        if len(group_cols) != 1:
-            raise ValueError("hot-path shape: exactly one group column")
+        if len(group_cols) > 1:
+            raise ValueError("hot-path shape: at most one group column")
         if self._window is not None:
             raise ValueError("window() already applied")
         vcols = {c for _, c in aggs}
@@ -49,7 +49,7 @@ class DataStream:
             raise ValueError("hot-path shape: one aggregate input column")
         ds = self._copy()
         ds._window = {
-            "group_col": group_cols[0],
+            "group_col": group_cols[0] if group_cols else None,
             "aggs": list(aggs),
             "value_col": vcols.pop(),
             "length_ms": int(length_ms),
@@ -78,7 +78,8 @@ class DataStream:
         op = WindowOp(length_ms=w["length_ms"], slide_ms=w["slide_ms"],
                       aggs=[(name, 0) for name, _ in w["aggs"]],
                       key_kind=self._key_kind, device=self._ctx.device,
-                      n_keys_hint=n_keys_hint)
+                      n_keys_hint=n_keys_hint,
+                      no_group=w["group_col"] is None)
         for (col, cmp, lit) in self._filters:
             names = [name for name, _ in w["aggs"]]
             if col not in names:
@@ -93,8 +94,9 @@ class DataStream:
         w = self._window
         try:
             for b in self._batches:
-                op.push(b[self._ts_col], b[self._window["group_col"]],
-                        b[w["value_col"]], b.get("_validity"))
+                keys = b[w["group_col"]] if w["group_col"] is not None else None
+                op.push(b[self._ts_col], keys, b[w["value_col"]],
+                        b.get("_validity"))
                 for out in op.poll_all():
                     yield out
             op.finish()

@@ -137,8 +137,12 @@ class WindowOp:
     def __init__(self, length_ms, slide_ms=0, aggs=(("count", 0), ("min", 0),
                  ("max", 0), ("avg", 0)), key_kind=KEY_UTF8, n_keys_hint=1024,
                  device=0, max_open_windows=0, ts_col=0, group_col=1,
-                 value_col=2):
+                 value_col=2, no_group=False):
         self._L = lib()
+        self.no_group = no_group
+        if no_group:
+            group_col = -1
+            value_col = 1
         n = len(aggs)
         self._agg_arr = (DzAggDesc * n)()
         self.agg_names = []
@@ -168,9 +172,22 @@ class WindowOp:
         ts_ms = np.ascontiguousarray(ts_ms, np.int64)
         vals = np.ascontiguousarray(vals, np.float64)
         n = len(ts_ms)
+        keep = []
+        if self.no_group:
+            cols = (DzColumn * 2)()
+            cols[0] = DzColumn(n, None, None, _np_ptr(ts_ms).value if n else None)
+            bm = None
+            if val_valid_bitmap is not None:
+                bm = np.ascontiguousarray(val_valid_bitmap, np.uint8)
+                keep.append(bm)
+            cols[1] = DzColumn(n, _np_ptr(bm).value if bm is not None else None,
+                               None, _np_ptr(vals).value if n else None)
+            batch = DzBatch(n, 2, ctypes.cast(cols, ctypes.POINTER(DzColumn)))
+            self._check(self._L.dz_window_op_push(self._h, ctypes.byref(batch)),
+                        "push")
+            return
         cols = (DzColumn * 3)()
         cols[0] = DzColumn(n, None, None, _np_ptr(ts_ms).value if n else None)
-        keep = []
         if self.key_kind == KEY_UTF8:
             data = b"".join(k.encode() if isinstance(k, str) else bytes(k) for k in keys)
             offs = np.zeros(n + 1, np.int32)
@@ -207,7 +224,9 @@ class WindowOp:
         ob = outp.contents
         n = ob.n_rows
         res = {"n_rows": n}
-        if self.key_kind == KEY_UTF8:
+        if self.no_group:
+            pass  # global aggregate: no group column in the output schema
+        elif self.key_kind == KEY_UTF8:
             offs = np.ctypeslib.as_array(ob.key_offsets, (n + 1,)).copy() if n else np.zeros(1, np.int32)
             total = int(offs[-1]) if n else 0
             data = ctypes.string_at(ob.key_data, total) if total else b""

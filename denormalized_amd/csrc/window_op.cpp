@@ -96,6 +96,9 @@ struct dz_window_op {
     dz_window_type wtype;
     int64_t len_ms, slide_ms;
     int32_t ts_col, group_col;
+    bool no_group = false; /* global aggregate (no GROUP BY): Partial/Final
+                            * path of the reference, streaming_window.rs:640-1051;
+                            * one group (dense key 0) per window */
     dz_key_kind key_kind;
     std::vector<dz_agg_desc> aggs;
     int device = 0;
@@ -154,6 +157,8 @@ struct dz_window_op {
     uint64_t* h_scalars = nullptr; /* pinned, 3 */
     int32_t* d_slotmap = nullptr;
     int slotmap_cap = 0;
+    int32_t* d_zero_kid = nullptr; /* no_group: all rows in group 0 */
+    int64_t zero_cap = 0;
 
     /* input staging (host-batch path) */
     int64_t* d_ts = nullptr;
@@ -363,7 +368,8 @@ extern "C" dz_window_op* dz_window_op_create(const dz_window_desc* desc) {
     op->slide_ms = desc->window_type == DZ_WINDOW_SLIDING ? desc->slide_ms : 0;
     op->ts_col = desc->ts_col;
     op->group_col = desc->group_col;
-    op->key_kind = desc->key_kind;
+    op->no_group = desc->group_col < 0;
+    op->key_kind = op->no_group ? DZ_KEY_DENSE_INT64 : desc->key_kind;
     op->aggs.assign(desc->aggs, desc->aggs + desc->n_aggs);
     op->device = desc->device;
     op->max_open = desc->max_open_windows > 0 ? desc->max_open_windows : 4096;
@@ -409,7 +415,7 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_meta); hipFree(op->d_ridx); hipFree(op->d_rvals);
     hipFree(op->d_gval); hipFree(op->d_gridx);
     hipFree(op->d_binoffs); hipFree(op->d_binlens);
-    hipFree(op->d_scalars); hipFree(op->d_slotmap);
+    hipFree(op->d_scalars); hipFree(op->d_slotmap); hipFree(op->d_zero_kid);
     hipFree(op->d_ts); hipFree(op->d_kid); hipFree(op->d_vals); hipFree(op->d_valbm);
     if (op->h_scalars) hipHostFree(op->h_scalars);
     if (op->h_stage) hipHostFree(op->h_stage);
@@ -509,7 +515,10 @@ static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
     OutBuf ob;
     ob.agg_i64.resize(na);
     ob.agg_f64.resize(na);
-    if (op->key_kind == DZ_KEY_UTF8) {
+    if (op->no_group) {
+        /* global aggregate: output schema has no group column
+         * (create_schema with empty group exprs, streaming_window.rs:1096+) */
+    } else if (op->key_kind == DZ_KEY_UTF8) {
         ob.key_offsets.resize(n + 1);
         ob.key_offsets[0] = 0;
         size_t total = 0;
@@ -913,11 +922,26 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
     return trigger_windows(op);
 }
 
+static dz_status ensure_zero_kid(dz_window_op* op, int64_t n) {
+    if (n > op->zero_cap) {
+        hipFree(op->d_zero_kid);
+        CHK(op, hipMalloc(&op->d_zero_kid, (size_t)n * 4));
+        CHK(op, hipMemsetAsync(op->d_zero_kid, 0, (size_t)n * 4, op->stream));
+        op->zero_cap = n;
+    }
+    return DZ_OK;
+}
+
 extern "C" dz_status dz_window_op_push_device(dz_window_op* op, int64_t n_rows,
                                               const int64_t* d_ts_ms,
                                               const int32_t* d_key_ids,
                                               const double* d_vals) {
     if (!op) return DZ_ERR;
+    if (op->no_group || !d_key_ids) {
+        CHK(op, hipSetDevice(op->device));
+        if (ensure_zero_kid(op, n_rows) != DZ_OK) return DZ_ERR;
+        d_key_ids = op->d_zero_kid;
+    }
     return push_core(op, n_rows, d_ts_ms, d_key_ids, d_vals, nullptr, true);
 }
 
@@ -930,11 +954,12 @@ extern "C" dz_status dz_window_op_push(dz_window_op* op, const dz_batch* batch) 
     if (!batch) { op->err = "null batch"; return DZ_ERR; }
     int64_t n = batch->n_rows;
     if (n == 0) return DZ_OK;
-    int32_t need = std::max({op->ts_col, op->group_col});
+    int32_t need = std::max(op->ts_col, op->group_col);
     for (auto& a : op->aggs) need = std::max(need, a.input_col);
     if (batch->n_cols <= need) { op->err = "batch has too few columns"; return DZ_ERR; }
     const dz_column& tsc = batch->cols[op->ts_col];
-    const dz_column& kc = batch->cols[op->group_col];
+    static const dz_column k_none = {0, nullptr, nullptr, nullptr};
+    const dz_column& kc = op->no_group ? k_none : batch->cols[op->group_col];
     const dz_column& vc = batch->cols[op->aggs[0].input_col];
     for (auto& a : op->aggs)
         if (a.input_col != op->aggs[0].input_col) {
@@ -971,7 +996,10 @@ extern "C" dz_status dz_window_op_push(dz_window_op* op, const dz_batch* batch) 
 
     /* dictionary encode keys (first-seen dense ids; the per-frame insertion
      * order the reference emits is recovered at emission via first-row sort) */
-    if (op->key_kind == DZ_KEY_UTF8) {
+    if (op->no_group) {
+        memset(h_kid, 0, (size_t)n * 4);
+        op->n_keys = 1;
+    } else if (op->key_kind == DZ_KEY_UTF8) {
         const int32_t* offs = kc.offsets;
         const char* data = (const char*)kc.data;
         if (!offs || !data) { op->err = "utf8 key column needs offsets+data"; return DZ_ERR; }
