@@ -78,4 +78,7 @@ void launch_fold3(hipStream_t stream, const double* d_gval, const uint32_t* d_gr
                   const int32_t* d_slot_of_widx, uint64_t* s_cnt, double* s_min,
                   double* s_max, double* s_sum, uint64_t* s_first);
 
+void launch_reset_slots(hipStream_t stream, const int32_t* d_slots, int ns,
+                        int64_t kcap, uint64_t* s_cnt, uint64_t* s_first);
+
 } // namespace dz

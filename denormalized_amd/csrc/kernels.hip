@@ -301,16 +301,16 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
         const int64_t* ts, const double* vals, const uint8_t* validity,
         int64_t n, int64_t chunk, int32_t st_rows, WinParams wp,
         const uint32_t* gofs, uint32_t* rmeta, uint32_t* rridx, double* rvals) {
-    /* LDS-staged stable partition: records of a supertile are ranked
-     * (row order preserved), placed bucket-major in LDS, then flushed so
-     * adjacent lanes write adjacent global addresses (~4-record runs at
-     * NB buckets instead of isolated 16 B scatters — the PMC-measured 6x
-     * write amplification of the direct form, profiles/hbm_traffic.json). */
-    __shared__ uint32_t cur[NB];   /* global cursors for this block's chunk */
-    __shared__ uint32_t cnt[NB];   /* per-supertile expanded record counts  */
-    __shared__ uint32_t offs[NB];  /* per-supertile exclusive bin prefix    */
-    __shared__ uint32_t stcur[NB]; /* running staging cursor per bin        */
-    __shared__ uint32_t scanbuf[BLOCK];
+    /* LDS-staged stable partition. Each wave owns a CONTIGUOUS QUARTER of the
+     * supertile (wave order == row order), so staging cursors are per-wave
+     * private: no cross-wave serialization, ~4 block barriers per supertile.
+     * Records are placed bucket-major in LDS and flushed so adjacent lanes
+     * write adjacent global addresses (the direct form measured 6x write
+     * amplification — profiles/hbm_traffic.json). */
+    __shared__ uint32_t cur[NB];    /* global cursors for this block's chunk */
+    __shared__ uint32_t cnt4[WAVES_PER_BLOCK][NB]; /* per-wave-quarter counts */
+    __shared__ uint32_t offs[NB];   /* per-supertile exclusive bin prefix     */
+    __shared__ uint32_t wofs[WAVES_PER_BLOCK][NB]; /* per-wave staging cursor */
     __shared__ uint32_t s_meta[ST_RECORDS];
     __shared__ uint32_t s_ridx[ST_RECORDS];
     __shared__ uint32_t s_dest[ST_RECORDS];
@@ -327,28 +327,38 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
 
     for (int64_t st0 = lo; st0 < hi; st0 += st_rows) {
         const int64_t st1 = i64min(hi, st0 + st_rows);
-        /* pass A: expanded counts per bucket */
-        for (int t = threadIdx.x; t < NB; t += BLOCK) cnt[t] = 0;
+        /* wave-quarter bounds (contiguous: wave order == row order) */
+        const int64_t q = ((st1 - st0) + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+        const int64_t w0 = i64min(st1, st0 + wave * q);
+        const int64_t w1 = i64min(st1, w0 + q);
+        /* pass A: expanded counts per (wave, bucket) */
+        for (int t = threadIdx.x; t < NB; t += BLOCK)
+            for (int w = 0; w < WAVES_PER_BLOCK; w++) cnt4[w][t] = 0;
         __syncthreads();
-        for (int64_t i = st0 + threadIdx.x; i < st1; i += BLOCK) {
+        for (int64_t i = w0 + lane; i < w1; i += 64) {
             uint32_t m = 1;
             if (wp.is_sliding) {
                 int32_t jm, mm;
                 row_windows(ts[i], wp, &jm, &mm);
                 m = (uint32_t)mm;
             }
-            if (m) atomicAdd(&cnt[kid[i] & (NB - 1)], m);
+            if (m) atomicAdd(&cnt4[wave][(uint32_t)kid[i] & (NB - 1)], m);
         }
         __syncthreads();
-        /* block-level exclusive prefix over NB bins */
+        /* per-supertile prefix: offs (bin-major) + per-wave staging bases */
         {
             constexpr int PER = NB / BLOCK;
-            uint32_t loc[PER];
+            uint32_t tot[PER];
             uint32_t s = 0;
             for (int j = 0; j < PER; j++) {
-                loc[j] = s;
-                s += cnt[threadIdx.x * PER + j];
+                int b = threadIdx.x * PER + j;
+                uint32_t t = 0;
+                for (int w = 0; w < WAVES_PER_BLOCK; w++) t += cnt4[w][b];
+                tot[j] = s;
+                s += t;
             }
+            /* block scan over 256 partials */
+            __shared__ uint32_t scanbuf[BLOCK];
             scanbuf[threadIdx.x] = s;
             __syncthreads();
             for (int o = 1; o < BLOCK; o <<= 1) {
@@ -359,21 +369,25 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
             }
             uint32_t pre = threadIdx.x ? scanbuf[threadIdx.x - 1] : 0;
             for (int j = 0; j < PER; j++) {
-                offs[threadIdx.x * PER + j] = pre + loc[j];
-                stcur[threadIdx.x * PER + j] = pre + loc[j];
+                int b = threadIdx.x * PER + j;
+                uint32_t run = pre + tot[j];
+                offs[b] = run;
+                for (int w = 0; w < WAVES_PER_BLOCK; w++) {
+                    wofs[w][b] = run;
+                    run += cnt4[w][b];
+                }
             }
             if (threadIdx.x == BLOCK - 1) s_total = scanbuf[BLOCK - 1];
         }
         __syncthreads();
-        /* pass B: ranked placement into LDS staging (row order per bucket:
-         * weighted intra-wave ranks + wave-serialized staging cursors) */
-        for (int64_t t0 = st0; t0 < st1; t0 += BLOCK) {
-            const int64_t i = t0 + threadIdx.x;
+        /* pass B: ranked placement, PER-WAVE private cursors (no barriers) */
+        for (int64_t t0 = w0; t0 < w1; t0 += 64) {
+            const int64_t i = t0 + lane;
             uint32_t bkt = SENT;
             int32_t jmin = 0, m = 0;
             uint32_t kv = 0, valid = 1;
             double v = 0.0;
-            if (i < st1) {
+            if (i < w1) {
                 kv = (uint32_t)kid[i];
                 v = vals[i];
                 if (validity) valid = (validity[i >> 3] >> (i & 7)) & 1u;
@@ -382,27 +396,42 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
             }
             uint32_t r = 0, wtot = 0;
             int fl = lane;
-            for (int j = 0; j < 64; j++) {
-                uint32_t bj = (uint32_t)__builtin_amdgcn_readlane((int)bkt, j);
-                uint32_t mj = (uint32_t)__builtin_amdgcn_readlane(m, j);
-                if (bj == bkt) {
-                    if (j < lane) r += mj;
-                    wtot += mj;
-                    if (j < fl) fl = j;
+            if (!wp.is_sliding) {
+                /* m is 0/1: bit-ballot same-bucket mask (9+1 bits incl SENT) */
+                uint64_t same = ~0ULL;
+                for (int b = 0; b < 9; b++) {
+                    uint64_t bb = __ballot((bkt >> b) & 1);
+                    same &= ((bkt >> b) & 1) ? bb : ~bb;
+                }
+                {
+                    uint64_t bb = __ballot(bkt == SENT);
+                    same &= (bkt == SENT) ? bb : ~bb;
+                }
+                const uint64_t below =
+                    (lane == 63) ? ~0ULL : ((1ULL << (lane + 1)) - 1);
+                r = (uint32_t)__popcll(same & below) - 1;
+                wtot = (uint32_t)__popcll(same);
+                fl = __ffsll((unsigned long long)same) - 1;
+            } else {
+                for (int j = 0; j < 64; j++) {
+                    uint32_t bj = (uint32_t)__builtin_amdgcn_readlane((int)bkt, j);
+                    uint32_t mj = (uint32_t)__builtin_amdgcn_readlane(m, j);
+                    if (bj == bkt) {
+                        if (j < lane) r += mj;
+                        wtot += mj;
+                        if (j < fl) fl = j;
+                    }
                 }
             }
             uint32_t base = 0;
-            for (int w = 0; w < WAVES_PER_BLOCK; w++) {
-                if (wave == w) {
-                    uint32_t pre = 0;
-                    if (lane == fl && bkt != SENT) {
-                        pre = stcur[bkt];
-                        stcur[bkt] = pre + wtot;
-                    }
-                    pre = (uint32_t)__shfl((int)pre, fl);
-                    base = pre + r;
+            {
+                uint32_t pre = 0;
+                if (lane == fl && bkt != SENT) {
+                    pre = wofs[wave][bkt];
+                    wofs[wave][bkt] = pre + wtot;
                 }
-                __syncthreads();
+                pre = (uint32_t)__shfl((int)pre, fl);
+                base = pre + r;
             }
             const uint32_t kloc = kv >> LOG_NB;
             for (int jj = 0; jj < m; jj++) {
@@ -424,7 +453,11 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
             rvals[d] = s_val[p];
         }
         __syncthreads();
-        for (int t = threadIdx.x; t < NB; t += BLOCK) cur[t] += cnt[t];
+        for (int t = threadIdx.x; t < NB; t += BLOCK) {
+            uint32_t s = 0;
+            for (int w = 0; w < WAVES_PER_BLOCK; w++) s += cnt4[w][t];
+            cur[t] += s;
+        }
         __syncthreads();
     }
 }
@@ -461,9 +494,9 @@ __global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
      *  row order), then a bucket-major flush so writes coalesce. */
     __shared__ uint32_t cnt[GCAP];    /* whole-bucket bin counts */
     __shared__ uint32_t gcur[GCAP];   /* global (bucket-region) bin cursors */
-    __shared__ uint32_t stcnt[GCAP];  /* per-supertile bin counts */
+    __shared__ uint32_t stcnt4[WAVES_PER_BLOCK][GCAP]; /* per-wave-quarter */
     __shared__ uint32_t stoffs[GCAP]; /* per-supertile bin prefix */
-    __shared__ uint32_t stcur[GCAP];
+    __shared__ uint32_t wcur[WAVES_PER_BLOCK][GCAP];   /* per-wave cursors */
     __shared__ uint32_t s_ridx[ST_RECORDS];
     __shared__ uint32_t s_dest[ST_RECORDS];
     __shared__ double s_val[ST_RECORDS];
@@ -504,36 +537,47 @@ __global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
 
     for (uint32_t st0 = lo; st0 < hi; st0 += ST_RECORDS) {
         const uint32_t st1 = min(hi, st0 + (uint32_t)ST_RECORDS);
-        /* per-supertile counts */
-        for (int g = threadIdx.x; g < GCAP; g += BLOCK) stcnt[g] = 0;
+        /* wave-quarter bounds (contiguous: wave order == row order) */
+        const uint32_t q = ((st1 - st0) + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+        const uint32_t w0 = min(st1, st0 + (uint32_t)wave * q);
+        const uint32_t w1 = min(st1, w0 + q);
+        /* per-(wave, bin) counts for this supertile */
+        for (int g = threadIdx.x; g < GCAP; g += BLOCK)
+            for (int w = 0; w < WAVES_PER_BLOCK; w++) stcnt4[w][g] = 0;
         __syncthreads();
-        for (uint32_t i = st0 + threadIdx.x; i < st1; i += BLOCK) {
+        for (uint32_t i = w0 + lane; i < w1; i += 64) {
             const uint32_t ms_ = rmeta[i];
             const int widx = (int)((ms_ >> META_WIDX_SHIFT) & META_WIDX_MASK);
             const int kloc = (int)(ms_ & META_KLOC_MASK);
             if (widx >= fc.w_lo && widx < fc.w_hi && kloc >= fc.k_lo &&
                 kloc < fc.k_hi)
-                atomicAdd(&stcnt[(widx - fc.w_lo) * nk + (kloc - fc.k_lo)], 1u);
+                atomicAdd(&stcnt4[wave][(widx - fc.w_lo) * nk + (kloc - fc.k_lo)], 1u);
         }
         __syncthreads();
-        if (threadIdx.x == 0) {
+        if (threadIdx.x == 0) { /* bin prefix over <=GCAP bins */
             uint32_t run = 0;
             for (int g = 0; g < GCAP; g++) {
-                uint32_t t = stcnt[g];
                 stoffs[g] = run;
-                stcur[g] = run;
-                run += t;
+                for (int w = 0; w < WAVES_PER_BLOCK; w++) run += stcnt4[w][g];
             }
             s_total = run;
         }
         __syncthreads();
-        /* ranked placement into staging */
-        for (uint32_t t0 = st0; t0 < st1; t0 += BLOCK) {
-            const uint32_t i = t0 + threadIdx.x;
+        for (int g = threadIdx.x; g < GCAP; g += BLOCK) { /* per-wave bases */
+            uint32_t run = stoffs[g];
+            for (int w = 0; w < WAVES_PER_BLOCK; w++) {
+                wcur[w][g] = run;
+                run += stcnt4[w][g];
+            }
+        }
+        __syncthreads();
+        /* ranked placement into staging, per-wave private cursors */
+        for (uint32_t t0 = w0; t0 < w1; t0 += 64) {
+            const uint32_t i = t0 + lane;
             uint32_t g = 0x1FFu; /* sentinel above GCAP-1 */
             uint32_t ri = 0;
             double v = 0.0;
-            if (i < st1) {
+            if (i < w1) {
                 const uint32_t ms_ = rmeta[i];
                 const int widx = (int)((ms_ >> META_WIDX_SHIFT) & META_WIDX_MASK);
                 const int kloc = (int)(ms_ & META_KLOC_MASK);
@@ -555,17 +599,14 @@ __global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
             const int leader = __ffsll((unsigned long long)same) - 1;
             const uint32_t wtot = (uint32_t)__popcll(same);
             uint32_t pos = 0;
-            for (int w = 0; w < WAVES_PER_BLOCK; w++) {
-                if (wave == w && g != 0x1FFu) {
-                    uint32_t pre = 0;
-                    if (lane == leader) {
-                        pre = stcur[g];
-                        stcur[g] = pre + wtot;
-                    }
-                    pre = (uint32_t)__shfl((int)pre, leader);
-                    pos = pre + (uint32_t)rank;
+            {
+                uint32_t pre = 0;
+                if (lane == leader && g != 0x1FFu) {
+                    pre = wcur[wave][g];
+                    wcur[wave][g] = pre + wtot;
                 }
-                __syncthreads();
+                pre = (uint32_t)__shfl((int)pre, leader);
+                pos = pre + (uint32_t)rank;
             }
             if (g != 0x1FFu) {
                 s_ridx[pos] = ri;
@@ -582,7 +623,11 @@ __global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
             gridx[d] = s_ridx[p];
         }
         __syncthreads();
-        for (int g = threadIdx.x; g < GCAP; g += BLOCK) gcur[g] += stcnt[g];
+        for (int g = threadIdx.x; g < GCAP; g += BLOCK) {
+            uint32_t s = 0;
+            for (int w = 0; w < WAVES_PER_BLOCK; w++) s += stcnt4[w][g];
+            gcur[g] += s;
+        }
         __syncthreads();
     }
 }
@@ -683,5 +728,30 @@ void launch_fold3(hipStream_t s, const double* d_gval, const uint32_t* d_gridx,
                        s_cnt, s_min, s_max, s_sum, s_first);
 }
 
+
+/* batched reset of freshly (re)allocated window slots:
+ * cnt = 0, first = ~0 for `ns` slots in one launch */
+__global__ void k_reset_slots(const int32_t* slots, int ns, int64_t kcap,
+                              uint64_t* s_cnt, uint64_t* s_first) {
+    int64_t total = (int64_t)ns * kcap;
+    int64_t stride5 = 5 * kcap;
+    int64_t gstride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+         i += gstride) {
+        int64_t si = i / kcap, k = i % kcap;
+        int64_t base = slots[si] * stride5 + k;
+        s_cnt[base] = 0;
+        s_first[base] = ~0ULL;
+    }
+}
+
+void launch_reset_slots(hipStream_t s, const int32_t* d_slots, int ns,
+                        int64_t kcap, uint64_t* s_cnt, uint64_t* s_first) {
+    int64_t total = (int64_t)ns * kcap;
+    int blocks = (int)std::min<int64_t>((total + BLOCK - 1) / BLOCK, 2048);
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(k_reset_slots, dim3(blocks), dim3(BLOCK), 0, s, d_slots,
+                       ns, kcap, s_cnt, s_first);
+}
 
 } // namespace dz

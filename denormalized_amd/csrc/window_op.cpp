@@ -159,6 +159,8 @@ struct dz_window_op {
     int slotmap_cap = 0;
     int32_t* d_zero_kid = nullptr; /* no_group: all rows in group 0 */
     int64_t zero_cap = 0;
+    int32_t* d_resetlist = nullptr;
+    int resetlist_cap = 0;
 
     /* input staging (host-batch path) */
     int64_t* d_ts = nullptr;
@@ -337,12 +339,6 @@ static dz_status state_alloc(dz_window_op* op, int64_t kcap_new, int32_t nslots_
     return DZ_OK;
 }
 
-static dz_status slot_reset(dz_window_op* op, int32_t slot) {
-    size_t stride = (size_t)op->kcap * 5;
-    CHK(op, hipMemsetAsync(op->s_cnt + slot * stride, 0, op->kcap * 8, op->stream));
-    CHK(op, hipMemsetAsync(op->s_first + slot * stride, 0xFF, op->kcap * 8, op->stream));
-    return DZ_OK;
-}
 
 /* ------------------------------------------------------------------ */
 /* create / destroy                                                    */
@@ -416,6 +412,7 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_gval); hipFree(op->d_gridx);
     hipFree(op->d_binoffs); hipFree(op->d_binlens);
     hipFree(op->d_scalars); hipFree(op->d_slotmap); hipFree(op->d_zero_kid);
+    hipFree(op->d_resetlist);
     hipFree(op->d_ts); hipFree(op->d_kid); hipFree(op->d_vals); hipFree(op->d_valbm);
     if (op->h_scalars) hipHostFree(op->h_scalars);
     if (op->h_stage) hipHostFree(op->h_stage);
@@ -821,6 +818,7 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
         return DZ_ERR;
     }
     std::vector<int32_t> slotmap(nw);
+    std::vector<int32_t> reset_list;
     for (int64_t r = 0; r < nw; r++) {
         auto it = op->open.find(ws[r]);
         if (it == op->open.end()) {
@@ -843,7 +841,7 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
                 std::lock_guard<std::mutex> lk(op->e_mtx);
                 op->e_ev_pool.push_back(fs.ev);
             }
-            if (slot_reset(op, fs.slot) != DZ_OK) return DZ_ERR;
+            reset_list.push_back(fs.slot);
             op->open[ws[r]] = {we[r], fs.slot};
             slotmap[r] = fs.slot;
         } else {
@@ -854,6 +852,18 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
         hipFree(op->d_slotmap);
         CHK(op, hipMalloc(&op->d_slotmap, nw * 4));
         op->slotmap_cap = (int)nw;
+    }
+    if (!reset_list.empty()) { /* one batched reset launch for all new slots */
+        int ns = (int)reset_list.size();
+        if (ns > op->resetlist_cap) {
+            hipFree(op->d_resetlist);
+            CHK(op, hipMalloc(&op->d_resetlist, (size_t)ns * 4));
+            op->resetlist_cap = ns;
+        }
+        CHK(op, hipMemcpyAsync(op->d_resetlist, reset_list.data(), (size_t)ns * 4,
+                               hipMemcpyHostToDevice, op->stream));
+        dz::launch_reset_slots(op->stream, op->d_resetlist, ns, op->kcap,
+                               op->s_cnt, op->s_first);
     }
     if (nw > 0)
         CHK(op, hipMemcpyAsync(op->d_slotmap, slotmap.data(), nw * 4,
