@@ -1,0 +1,61 @@
+"""Property-based fuzz: the C oracle vs the independent pure-Python
+restatement on hypothesis-generated streams (window shapes, batch splits,
+late data, nulls). CPU-only."""
+import numpy as np
+from hypothesis import given, settings, strategies as st
+
+from oracle import pyoracle
+from tests.pyref import PyRef
+
+
+@st.composite
+def stream(draw):
+    len_ms = draw(st.sampled_from([250, 500, 1000, 1500, 2000, 5000]))
+    slide = draw(st.sampled_from([0, 100, 250, 500, 750, 1000, 3000]))
+    nkeys = draw(st.integers(1, 12))
+    nbatches = draw(st.integers(1, 5))
+    batches = []
+    t = draw(st.integers(10_000, 50_000))
+    for _ in range(nbatches):
+        n = draw(st.integers(1, 60))
+        deltas = draw(st.lists(st.integers(0, 400), min_size=n, max_size=n))
+        back = draw(st.integers(0, 3000))
+        start = max(0, t - back)
+        ts = np.array(start + np.cumsum(deltas), np.int64)
+        t = max(t, int(ts.max()))
+        keys = np.array(draw(st.lists(st.integers(0, nkeys - 1), min_size=n,
+                                      max_size=n)), np.int64)
+        vals = np.array(draw(st.lists(
+            st.floats(-1e6, 1e6, allow_nan=False, width=32),
+            min_size=n, max_size=n)), np.float64)
+        valid = np.array(draw(st.lists(st.booleans(), min_size=n, max_size=n)),
+                         np.uint8)
+        batches.append((ts, keys, vals, valid))
+    return len_ms, slide, batches
+
+
+@given(stream())
+@settings(max_examples=60, deadline=None)
+def test_oracle_matches_pyref(s):
+    len_ms, slide, batches = s
+    o = pyoracle.Oracle(len_ms, slide)
+    p = PyRef(len_ms, slide)
+    for ts, k, v, valid in batches:
+        o.push(ts, k, v, valid)
+        p.push(list(ts), list(k), list(v), list(valid))
+    o.finish()
+    p.finish()
+    got = o.fetch()
+    o.close()
+    exp = p.out
+    assert len(got["key"]) == len(exp)
+    for i, row in enumerate(exp):
+        assert int(got["key"][i]) == row[0]
+        assert int(got["count"][i]) == row[1]
+        assert int(got["valid"][i]) == row[6]
+        if row[6]:
+            assert float(got["min"][i]) == row[2]
+            assert float(got["max"][i]) == row[3]
+            assert float(got["avg"][i]) == row[4]
+        assert int(got["window_start"][i]) == row[7]
+        assert int(got["window_end"][i]) == row[8]
