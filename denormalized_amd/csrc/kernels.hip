@@ -308,9 +308,12 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
      * write adjacent global addresses (the direct form measured 6x write
      * amplification — profiles/hbm_traffic.json). */
     __shared__ uint32_t cur[NB];    /* global cursors for this block's chunk */
-    __shared__ uint32_t cnt4[WAVES_PER_BLOCK][NB]; /* per-wave-quarter counts */
+    __shared__ uint32_t cnt4[WAVES_PER_BLOCK][NB]; /* per-wave-quarter counts,
+                                     * converted IN PLACE to per-wave staging
+                                     * cursors by the prefix (saves 8 KiB LDS
+                                     * => 3 blocks/CU) */
     __shared__ uint32_t offs[NB];   /* per-supertile exclusive bin prefix     */
-    __shared__ uint32_t wofs[WAVES_PER_BLOCK][NB]; /* per-wave staging cursor */
+    auto wofs = cnt4;
     __shared__ uint32_t s_meta[ST_RECORDS];
     __shared__ uint32_t s_ridx[ST_RECORDS];
     __shared__ uint32_t s_dest[ST_RECORDS];
@@ -373,8 +376,9 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
                 uint32_t run = pre + tot[j];
                 offs[b] = run;
                 for (int w = 0; w < WAVES_PER_BLOCK; w++) {
-                    wofs[w][b] = run;
-                    run += cnt4[w][b];
+                    uint32_t c = cnt4[w][b];
+                    wofs[w][b] = run; /* overlays cnt4 */
+                    run += c;
                 }
             }
             if (threadIdx.x == BLOCK - 1) s_total = scanbuf[BLOCK - 1];
@@ -454,9 +458,9 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
         }
         __syncthreads();
         for (int t = threadIdx.x; t < NB; t += BLOCK) {
-            uint32_t s = 0;
-            for (int w = 0; w < WAVES_PER_BLOCK; w++) s += cnt4[w][t];
-            cur[t] += s;
+            /* bin total = offs delta (cnt4 was overlaid by the cursors) */
+            uint32_t nxt = (t + 1 < NB) ? offs[t + 1] : s_total;
+            cur[t] += nxt - offs[t];
         }
         __syncthreads();
     }
@@ -692,6 +696,7 @@ __global__ __launch_bounds__(BLOCK) void k_fold3(const double* gval,
     for (int o = 32; o > 0; o >>= 1)
         mlen = max(mlen, (uint32_t)__shfl_down((int)mlen, o));
     mlen = (uint32_t)__builtin_amdgcn_readlane((int)mlen, 0);
+#pragma unroll 4
     for (uint32_t r = 0; r < mlen; r++) {
         const bool act = own && r < len;
         const uint32_t ri = act ? gridx[lo + off + r] : 0;
