@@ -172,9 +172,11 @@ def main():
             dist.all_reduce(wm, op=dist.ReduceOp.MAX)
             op.advance_watermark(int(wm.item()))
         emitted = 0
-        # non-blocking poll: emission is pipelined on the op's worker thread
-        # and overlaps the next step's kernels; finish() drains the tail.
-        for b in op.poll_all(drain=False):
+        # non-blocking zero-copy poll: emission is pipelined on the op's
+        # worker pool and overlaps the next step's kernels; consuming the
+        # batch = reading the op-owned Arrow-style buffers (the C contract),
+        # exactly as the reference's downstream operator would.
+        for b in op.poll_all(drain=False, copy=False):
             emitted += b["n_rows"]
         return emitted
 
@@ -189,7 +191,7 @@ def main():
     for s in range(W, W + K):
         emitted += push_step(s)
     op.finish()
-    emitted += sum(b["n_rows"] for b in op.poll_all())
+    emitted += sum(b["n_rows"] for b in op.poll_all(copy=False))
     dz.synchronize(device)
     if dist is not None:
         dist.barrier()

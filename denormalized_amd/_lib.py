@@ -215,12 +215,21 @@ class WindowOp:
         self._check(self._L.dz_window_op_push_device(
             self._h, n, d_ts, d_kid32, d_vals), "push_device")
 
-    def poll(self):
-        """Returns a dict of numpy copies of one emitted batch, or None."""
+    def poll(self, copy=True):
+        """Returns a dict of numpy arrays for one emitted batch, or None.
+        copy=False returns zero-copy views of the op-owned buffers (valid
+        until the next poll on this handle — the C contract's ownership
+        rule); copy=True materialises."""
         outp = ctypes.POINTER(DzOutBatch)()
         self._check(self._L.dz_window_op_poll(self._h, ctypes.byref(outp)), "poll")
         if not outp:
             return None
+
+        def mk(p, n, ctype):
+            if not n:
+                return np.zeros(0, np.int64 if ctype == ctypes.c_int64 else np.float64)
+            a = np.ctypeslib.as_array(ctypes.cast(p, ctypes.POINTER(ctype)), (n,))
+            return a.copy() if copy else a
         ob = outp.contents
         n = ob.n_rows
         res = {"n_rows": n}
@@ -232,17 +241,20 @@ class WindowOp:
             data = ctypes.string_at(ob.key_data, total) if total else b""
             res["key"] = [data[offs[i]:offs[i + 1]].decode() for i in range(n)]
         else:
-            res["key"] = np.ctypeslib.as_array(ob.key_i64, (n,)).copy() if n else np.zeros(0, np.int64)
+            res["key"] = mk(ob.key_i64, n, ctypes.c_int64)
         for i, name in enumerate(self.agg_names):
             pt = ob.agg_cols[i]
             if AGG_BY_NAME.get(name, name) == AGG_COUNT or name == "count":
-                arr = np.ctypeslib.as_array(ctypes.cast(pt, ctypes.POINTER(ctypes.c_int64)), (n,)).copy() if n else np.zeros(0, np.int64)
+                res[name] = mk(pt, n, ctypes.c_int64)
             else:
-                arr = np.ctypeslib.as_array(ctypes.cast(pt, ctypes.POINTER(ctypes.c_double)), (n,)).copy() if n else np.zeros(0, np.float64)
-            res[name] = arr
-        res["valid"] = np.ctypeslib.as_array(ob.agg_valid, (n,)).copy() if n else np.zeros(0, np.uint8)
-        res["window_start"] = np.ctypeslib.as_array(ob.window_start_ms, (n,)).copy() if n else np.zeros(0, np.int64)
-        res["window_end"] = np.ctypeslib.as_array(ob.window_end_ms, (n,)).copy() if n else np.zeros(0, np.int64)
+                res[name] = mk(pt, n, ctypes.c_double)
+        if n:
+            v = np.ctypeslib.as_array(ob.agg_valid, (n,))
+            res["valid"] = v.copy() if copy else v
+        else:
+            res["valid"] = np.zeros(0, np.uint8)
+        res["window_start"] = mk(ob.window_start_ms, n, ctypes.c_int64)
+        res["window_end"] = mk(ob.window_end_ms, n, ctypes.c_int64)
         return res
 
     def drain(self):
@@ -250,12 +262,12 @@ class WindowOp:
         pipelined on a worker thread)."""
         self._check(self._L.dz_window_op_drain(self._h), "drain")
 
-    def poll_all(self, drain=True):
+    def poll_all(self, drain=True, copy=True):
         if drain:
             self.drain()
         out = []
         while True:
-            b = self.poll()
+            b = self.poll(copy=copy)
             if b is None:
                 return out
             out.append(b)

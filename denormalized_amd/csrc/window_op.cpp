@@ -174,7 +174,7 @@ struct dz_window_op {
     /* async emission: pinned slab pool + worker thread. trigger_windows
      * enqueues {D2H event, slab}; the worker waits the event, builds the
      * output batch off the push critical path, and appends to outq. */
-    static constexpr int E_POOL = 16;
+    static constexpr int E_POOL = 32;
     static constexpr int E_WORKERS = 4;
     uint64_t* e_slabs[E_POOL] = {};
     int64_t e_slab_kcap = 0;
