@@ -36,6 +36,8 @@ struct FoldChunk {
     int32_t k_lo, k_hi;  /* kloc range [k_lo, k_hi) */
     int64_t kcap;        /* state key capacity (multiple of NB) */
     uint32_t batch_seq;
+    int32_t bin_stride;  /* binoffs/binlens row stride (GCAP, or gtot in the
+                          * single-launch big-bin regime) */
 };
 
 /* Launch wrappers implemented in kernels.hip. All run on `stream`;
@@ -71,6 +73,12 @@ void launch_regroup(hipStream_t stream, const uint32_t* d_meta,
                     const uint32_t* d_bucket_base, const FoldChunk& fc,
                     uint32_t* d_binoffs, uint32_t* d_binlens, double* d_gval,
                     uint32_t* d_gridx);
+
+void launch_regroup_big(hipStream_t stream, const uint32_t* d_meta,
+                        const uint32_t* d_ridx, const double* d_rvals,
+                        const uint32_t* d_bucket_base, const FoldChunk& fc,
+                        int32_t gtot, uint32_t* d_binoffs, uint32_t* d_binlens,
+                        uint32_t* d_wqcnt, double* d_gval, uint32_t* d_gridx);
 
 void launch_fold3(hipStream_t stream, const double* d_gval, const uint32_t* d_gridx,
                   const uint32_t* d_bucket_base, const uint32_t* d_binoffs,

@@ -152,6 +152,9 @@ struct dz_window_op {
     uint32_t* d_gridx = nullptr; /* rowidx | valid<<31, segmented */
     uint32_t* d_binoffs = nullptr;
     uint32_t* d_binlens = nullptr;
+    int64_t bin_cap = 0;   /* bins per bucket the binoffs/lens arrays hold */
+    uint32_t* d_wqcnt = nullptr; /* big-bin regroup scratch [NB][4][gtot] */
+    int64_t wq_cap = 0;
     int64_t rec_cap = 0;
     uint64_t* d_scalars = nullptr;
     uint64_t* h_scalars = nullptr; /* pinned, 3 */
@@ -410,7 +413,7 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_ghist); hipFree(op->d_gofs); hipFree(op->d_total); hipFree(op->d_base);
     hipFree(op->d_meta); hipFree(op->d_ridx); hipFree(op->d_rvals);
     hipFree(op->d_gval); hipFree(op->d_gridx);
-    hipFree(op->d_binoffs); hipFree(op->d_binlens);
+    hipFree(op->d_binoffs); hipFree(op->d_binlens); hipFree(op->d_wqcnt);
     hipFree(op->d_scalars); hipFree(op->d_slotmap); hipFree(op->d_zero_kid);
     hipFree(op->d_resetlist);
     hipFree(op->d_ts); hipFree(op->d_kid); hipFree(op->d_vals); hipFree(op->d_valbm);
@@ -738,9 +741,16 @@ static dz_status ensure_scratch(dz_window_op* op, int C, int64_t nrec) {
         CHK(op, hipMalloc(&op->d_gridx, (size_t)nrec * 4));
         op->rec_cap = nrec;
     }
-    if (!op->d_binoffs) {
-        CHK(op, hipMalloc(&op->d_binoffs, (size_t)dz::NB * dz::FOLD_GCAP * 4));
-        CHK(op, hipMalloc(&op->d_binlens, (size_t)dz::NB * dz::FOLD_GCAP * 4));
+    return DZ_OK;
+}
+
+static dz_status ensure_bins(dz_window_op* op, int64_t bins_per_bucket) {
+    if (bins_per_bucket > op->bin_cap) {
+        hipFree(op->d_binoffs);
+        hipFree(op->d_binlens);
+        CHK(op, hipMalloc(&op->d_binoffs, (size_t)dz::NB * bins_per_bucket * 4));
+        CHK(op, hipMalloc(&op->d_binlens, (size_t)dz::NB * bins_per_bucket * 4));
+        op->bin_cap = bins_per_bucket;
     }
     return DZ_OK;
 }
@@ -898,28 +908,65 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
                            op->d_rvals);
     });
     int64_t klocs = op->kcap >> dz::LOG_NB;
-    for (int64_t k_lo = 0; k_lo < klocs; k_lo += dz::FOLD_GCAP) {
-        int32_t nk = (int32_t)std::min<int64_t>(dz::FOLD_GCAP, klocs - k_lo);
-        int32_t wstep = dz::FOLD_GCAP / nk;
-        for (int64_t w_lo = 0; w_lo < nw; w_lo += wstep) {
-            dz::FoldChunk fc;
-            fc.w_lo = (int32_t)w_lo;
-            fc.w_hi = (int32_t)std::min<int64_t>(nw, w_lo + wstep);
-            fc.k_lo = (int32_t)k_lo;
-            fc.k_hi = (int32_t)(k_lo + nk);
-            fc.kcap = op->kcap;
-            fc.batch_seq = op->batch_seq;
-            timed(op, "regroup", (double)nrec_max * 32, [&] {
-                dz::launch_regroup(op->stream, op->d_meta, op->d_ridx,
-                                   op->d_rvals, op->d_base, fc, op->d_binoffs,
-                                   op->d_binlens, op->d_gval, op->d_gridx);
-            });
-            timed(op, "fold", (double)nrec_max * 12, [&] {
-                dz::launch_fold3(op->stream, op->d_gval, op->d_gridx, op->d_base,
-                                 op->d_binoffs, op->d_binlens, fc, op->d_slotmap,
-                                 op->s_cnt, op->s_min, op->s_max, op->s_sum,
-                                 op->s_first);
-            });
+    int64_t gtot = klocs * nw;
+    constexpr int64_t BIG_GTOT_MAX = 65536;
+    if (gtot > dz::FOLD_GCAP && gtot <= BIG_GTOT_MAX) {
+        /* single-launch big-bin regime: bins in global scratch, no chunked
+         * re-reads (cfg3-class keyspaces) */
+        if (ensure_bins(op, gtot) != DZ_OK) return DZ_ERR;
+        if ((int64_t)4 * gtot > op->wq_cap) {
+            hipFree(op->d_wqcnt);
+            CHK(op, hipMalloc(&op->d_wqcnt, (size_t)dz::NB * 4 * gtot * 4));
+            op->wq_cap = 4 * gtot;
+        }
+        CHK(op, hipMemsetAsync(op->d_wqcnt, 0, (size_t)dz::NB * 4 * gtot * 4,
+                               op->stream));
+        dz::FoldChunk fc;
+        fc.w_lo = 0;
+        fc.w_hi = (int32_t)nw;
+        fc.k_lo = 0;
+        fc.k_hi = (int32_t)klocs;
+        fc.kcap = op->kcap;
+        fc.batch_seq = op->batch_seq;
+        fc.bin_stride = (int32_t)gtot;
+        timed(op, "regroup", (double)nrec_max * 32, [&] {
+            dz::launch_regroup_big(op->stream, op->d_meta, op->d_ridx,
+                                   op->d_rvals, op->d_base, fc, (int32_t)gtot,
+                                   op->d_binoffs, op->d_binlens, op->d_wqcnt,
+                                   op->d_gval, op->d_gridx);
+        });
+        timed(op, "fold", (double)nrec_max * 12, [&] {
+            dz::launch_fold3(op->stream, op->d_gval, op->d_gridx, op->d_base,
+                             op->d_binoffs, op->d_binlens, fc, op->d_slotmap,
+                             op->s_cnt, op->s_min, op->s_max, op->s_sum,
+                             op->s_first);
+        });
+    } else {
+        if (ensure_bins(op, dz::FOLD_GCAP) != DZ_OK) return DZ_ERR;
+        for (int64_t k_lo = 0; k_lo < klocs; k_lo += dz::FOLD_GCAP) {
+            int32_t nk = (int32_t)std::min<int64_t>(dz::FOLD_GCAP, klocs - k_lo);
+            int32_t wstep = dz::FOLD_GCAP / nk;
+            for (int64_t w_lo = 0; w_lo < nw; w_lo += wstep) {
+                dz::FoldChunk fc;
+                fc.w_lo = (int32_t)w_lo;
+                fc.w_hi = (int32_t)std::min<int64_t>(nw, w_lo + wstep);
+                fc.k_lo = (int32_t)k_lo;
+                fc.k_hi = (int32_t)(k_lo + nk);
+                fc.kcap = op->kcap;
+                fc.batch_seq = op->batch_seq;
+                fc.bin_stride = dz::FOLD_GCAP;
+                timed(op, "regroup", (double)nrec_max * 32, [&] {
+                    dz::launch_regroup(op->stream, op->d_meta, op->d_ridx,
+                                       op->d_rvals, op->d_base, fc, op->d_binoffs,
+                                       op->d_binlens, op->d_gval, op->d_gridx);
+                });
+                timed(op, "fold", (double)nrec_max * 12, [&] {
+                    dz::launch_fold3(op->stream, op->d_gval, op->d_gridx,
+                                     op->d_base, op->d_binoffs, op->d_binlens, fc,
+                                     op->d_slotmap, op->s_cnt, op->s_min,
+                                     op->s_max, op->s_sum, op->s_first);
+                });
+            }
         }
     }
     op->batch_seq++;
