@@ -86,6 +86,27 @@ void launch_fold3(hipStream_t stream, const double* d_gval, const uint32_t* d_gr
                   const int32_t* d_slot_of_widx, uint64_t* s_cnt, double* s_min,
                   double* s_max, double* s_sum, uint64_t* s_first);
 
+struct EmitFilter {
+    int32_t on;      /* 0 = no filter */
+    int32_t field;   /* 0 cnt, 1 min, 2 max, 3 sum, 4 avg */
+    int32_t cmp;     /* 0 < 1 <= 2 > 3 >= 4 == 5 != */
+    double lit;
+};
+
+/* device-side emission at window close: compact touched groups from the
+ * slot slab, stable radix sort by first-seen row, gather aggregate columns
+ * + filter flags. All launches on the given (copy) stream. */
+void launch_emission(hipStream_t stream, const uint64_t* slab_first,
+                     const uint64_t* slab_cnt, const double* slab_min,
+                     const double* slab_max, const double* slab_sum, int64_t K,
+                     uint64_t* ekeys, uint32_t* ekid, uint64_t* skeys,
+                     uint32_t* skid, uint32_t* counter, uint32_t* rhist,
+                     uint32_t* roffs, const EmitFilter& ef, uint32_t* okid,
+                     uint64_t* ocnt, double* omin, double* omax, double* osum,
+                     double* oavg, uint8_t* oflags);
+constexpr int EMIT_RCHUNK = 4096;
+constexpr int EMIT_RBINS = 2048;
+
 void launch_reset_slots(hipStream_t stream, const int32_t* d_slots, int ns,
                         int64_t kcap, uint64_t* s_cnt, uint64_t* s_first);
 

@@ -860,6 +860,240 @@ void launch_fold3(hipStream_t s, const double* d_gval, const uint32_t* d_gridx,
 }
 
 
+
+/* ------------------------------------------------------------------ */
+/* device-side emission: compact touched groups, stable radix sort by  */
+/* first-seen row (insertion order), gather aggregate columns + filter */
+/* flag. Runs on the op's copy stream at window close; the host worker */
+/* only formats the already-sorted columns.                            */
+/* ------------------------------------------------------------------ */
+
+constexpr int RDIG = 11;          /* radix digit bits */
+constexpr int RBINS = 1 << RDIG;  /* 2048 */
+constexpr int RCHUNK = 4096;      /* elements per radix block */
+constexpr int RPASSES = 6;        /* 6*11 = 66 >= 64 bits */
+
+__global__ void k_ecompact(const uint64_t* s_first, int64_t K, uint64_t* ekeys,
+                           uint32_t* ekid, uint32_t* counter) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < K;
+         k += stride) {
+        uint64_t f = s_first[k];
+        if (f != ~0ULL) {
+            uint32_t p = atomicAdd(counter, 1u);
+            ekeys[p] = f;
+            ekid[p] = (uint32_t)k;
+        }
+    }
+}
+
+__global__ __launch_bounds__(BLOCK) void k_rhist(const uint64_t* keys,
+        const uint32_t* counter, int shift, uint32_t* hist) {
+    __shared__ uint32_t h[RBINS];
+    const uint32_t nt = *counter;
+    for (int t = threadIdx.x; t < RBINS; t += BLOCK) h[t] = 0;
+    __syncthreads();
+    const uint32_t lo = blockIdx.x * RCHUNK;
+    const uint32_t hi = min(nt, lo + (uint32_t)RCHUNK);
+    for (uint32_t i = lo + threadIdx.x; i < hi; i += BLOCK)
+        atomicAdd(&h[(uint32_t)(keys[i] >> shift) & (RBINS - 1)], 1u);
+    __syncthreads();
+    for (int t = threadIdx.x; t < RBINS; t += BLOCK)
+        hist[(int64_t)blockIdx.x * RBINS + t] = h[t];
+}
+
+/* digit-major exclusive offsets: offs[b][d] = base[d] + sum_{b'<b} hist[b'][d] */
+__global__ __launch_bounds__(1024) void k_rscan(const uint32_t* hist, int nblk,
+                                                uint32_t* offs) {
+    __shared__ uint32_t part[1024];
+    constexpr int PER = RBINS / 1024; /* 2 */
+    uint32_t loc[PER];
+    uint32_t s = 0;
+    for (int j = 0; j < PER; j++) {
+        int d = threadIdx.x * PER + j;
+        uint32_t t = 0;
+        for (int b = 0; b < nblk; b++) t += hist[(int64_t)b * RBINS + d];
+        loc[j] = s;
+        s += t;
+    }
+    part[threadIdx.x] = s;
+    __syncthreads();
+    for (int o = 1; o < 1024; o <<= 1) {
+        uint32_t v = (threadIdx.x >= o) ? part[threadIdx.x - o] : 0;
+        __syncthreads();
+        part[threadIdx.x] += v;
+        __syncthreads();
+    }
+    uint32_t pre = threadIdx.x ? part[threadIdx.x - 1] : 0;
+    for (int j = 0; j < PER; j++) {
+        int d = threadIdx.x * PER + j;
+        uint32_t run = pre + loc[j];
+        for (int b = 0; b < nblk; b++) {
+            uint32_t t = hist[(int64_t)b * RBINS + d];
+            offs[(int64_t)b * RBINS + d] = run;
+            run += t;
+        }
+    }
+}
+
+__global__ __launch_bounds__(BLOCK) void k_rscatter(const uint64_t* keys,
+        const uint32_t* payload, const uint32_t* counter, int shift,
+        const uint32_t* offs, uint64_t* okeys, uint32_t* opayload) {
+    /* stable per-block scatter: wave-quarters (wave order == element order)
+     * with per-(wave,digit) counts -> private cursors */
+    __shared__ uint32_t cnt4[WAVES_PER_BLOCK][RBINS]; /* overlaid to cursors */
+    __shared__ uint32_t scanbuf[BLOCK];
+    const uint32_t nt = *counter;
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    const uint32_t lo = blockIdx.x * RCHUNK;
+    const uint32_t hi = min(nt, lo + (uint32_t)RCHUNK);
+    if (lo >= hi) return;
+    for (int t = threadIdx.x; t < RBINS; t += BLOCK)
+        for (int w = 0; w < WAVES_PER_BLOCK; w++) cnt4[w][t] = 0;
+    __syncthreads();
+    const uint32_t n = hi - lo;
+    const uint32_t q = (n + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+    const uint32_t w0 = lo + min(n, (uint32_t)wave * q);
+    const uint32_t w1 = lo + min(n, (uint32_t)(wave + 1) * q);
+    for (uint32_t i = w0 + lane; i < w1; i += 64)
+        atomicAdd(&cnt4[wave][(uint32_t)(keys[i] >> shift) & (RBINS - 1)], 1u);
+    __syncthreads();
+    {   /* per-digit block totals -> add global offs -> per-wave bases */
+        constexpr int PER = RBINS / BLOCK; /* 8 */
+        for (int j = 0; j < PER; j++) {
+            int d = threadIdx.x * PER + j;
+            uint32_t run = offs[(int64_t)blockIdx.x * RBINS + d];
+            for (int w = 0; w < WAVES_PER_BLOCK; w++) {
+                uint32_t c = cnt4[w][d];
+                cnt4[w][d] = run;
+                run += c;
+            }
+        }
+        (void)scanbuf;
+    }
+    __syncthreads();
+    for (uint32_t t0 = w0; t0 < w1; t0 += 64) {
+        const uint32_t i = t0 + lane;
+        const bool act = i < w1;
+        uint64_t key = act ? keys[i] : ~0ULL;
+        uint32_t pay = act ? payload[i] : 0;
+        uint32_t d = act ? ((uint32_t)(key >> shift) & (RBINS - 1)) : 0xFFFFu;
+        uint64_t same = ~0ULL;
+        for (int b = 0; b < RDIG; b++) {
+            uint64_t bb = __ballot((d >> b) & 1);
+            same &= ((d >> b) & 1) ? bb : ~bb;
+        }
+        {
+            uint64_t bb = __ballot(act);
+            same &= act ? bb : ~bb;
+        }
+        const uint64_t below = (lane == 63) ? ~0ULL : ((1ULL << (lane + 1)) - 1);
+        const int rank = (int)__popcll(same & below) - 1;
+        const int leader = __ffsll((unsigned long long)same) - 1;
+        const uint32_t wtot = (uint32_t)__popcll(same);
+        uint32_t pos = 0;
+        {
+            uint32_t pre = 0;
+            if (lane == leader && act) {
+                pre = cnt4[wave][d];
+                cnt4[wave][d] = pre + wtot;
+            }
+            pre = (uint32_t)__shfl((int)pre, leader);
+            pos = pre + (uint32_t)rank;
+        }
+        if (act) {
+            okeys[pos] = key;
+            opayload[pos] = pay;
+        }
+    }
+}
+
+/* gather the sorted groups' aggregates + filter flag into output columns */
+__global__ void k_egather(const uint32_t* skid, const uint32_t* counter,
+                          const uint64_t* s_cnt, const double* s_min,
+                          const double* s_max, const double* s_sum,
+                          EmitFilter ef, uint32_t* okid, uint64_t* ocnt,
+                          double* omin, double* omax, double* osum,
+                          double* oavg, uint8_t* oflags) {
+    const uint32_t nt = *counter;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nt;
+         i += stride) {
+        const uint32_t kid = skid[i];
+        const uint64_t c = s_cnt[kid];
+        const bool valid = c > 0;
+        const double mn = valid ? s_min[kid] : 0.0;
+        const double mx = valid ? s_max[kid] : 0.0;
+        const double sm = valid ? s_sum[kid] : 0.0;
+        const double av = valid ? sm / (double)c : 0.0;
+        bool pass = true;
+        if (ef.on) {
+            double v = 0.0;
+            bool fv = true;
+            switch (ef.field) {
+                case 0: v = (double)c; break;
+                case 1: v = mn; fv = valid; break;
+                case 2: v = mx; fv = valid; break;
+                case 3: v = sm; fv = valid; break;
+                default: v = av; fv = valid; break;
+            }
+            if (!fv) {
+                pass = false; /* NULL never passes a comparison filter */
+            } else {
+                switch (ef.cmp) {
+                    case 0: pass = v < ef.lit; break;
+                    case 1: pass = v <= ef.lit; break;
+                    case 2: pass = v > ef.lit; break;
+                    case 3: pass = v >= ef.lit; break;
+                    case 4: pass = v == ef.lit; break;
+                    default: pass = v != ef.lit; break;
+                }
+            }
+        }
+        okid[i] = kid;
+        ocnt[i] = c;
+        omin[i] = mn;
+        omax[i] = mx;
+        osum[i] = sm;
+        oavg[i] = av;
+        oflags[i] = (uint8_t)((valid ? 1 : 0) | (pass ? 2 : 0));
+    }
+}
+
+void launch_emission(hipStream_t s, const uint64_t* slab_first,
+                     const uint64_t* slab_cnt, const double* slab_min,
+                     const double* slab_max, const double* slab_sum, int64_t K,
+                     uint64_t* ekeys, uint32_t* ekid, uint64_t* skeys,
+                     uint32_t* skid, uint32_t* counter, uint32_t* rhist,
+                     uint32_t* roffs, const EmitFilter& ef, uint32_t* okid,
+                     uint64_t* ocnt, double* omin, double* omax, double* osum,
+                     double* oavg, uint8_t* oflags) {
+    int cblocks = (int)std::min<int64_t>((K + BLOCK - 1) / BLOCK, 2048);
+    hipLaunchKernelGGL(k_ecompact, dim3(cblocks), dim3(BLOCK), 0, s, slab_first,
+                       K, ekeys, ekid, counter);
+    int nblk = (int)((K + RCHUNK - 1) / RCHUNK);
+    uint64_t* ka = ekeys;
+    uint32_t* pa = ekid;
+    uint64_t* kb = skeys;
+    uint32_t* pb = skid;
+    for (int p = 0; p < RPASSES; p++) {
+        int shift = p * RDIG;
+        hipLaunchKernelGGL(k_rhist, dim3(nblk), dim3(BLOCK), 0, s, ka, counter,
+                           shift, rhist);
+        hipLaunchKernelGGL(k_rscan, dim3(1), dim3(1024), 0, s, rhist, nblk,
+                           roffs);
+        hipLaunchKernelGGL(k_rscatter, dim3(nblk), dim3(BLOCK), 0, s, ka, pa,
+                           counter, shift, roffs, kb, pb);
+        std::swap(ka, kb);
+        std::swap(pa, pb);
+    }
+    /* RPASSES is even => sorted data ends in (ekeys, ekid) */
+    hipLaunchKernelGGL(k_egather, dim3(cblocks), dim3(BLOCK), 0, s, ekid, counter,
+                       slab_cnt, slab_min, slab_max, slab_sum, ef, okid, ocnt,
+                       omin, omax, osum, oavg, oflags);
+}
+
 /* batched reset of freshly (re)allocated window slots:
  * cnt = 0, first = ~0 for `ns` slots in one launch */
 __global__ void k_reset_slots(const int32_t* slots, int ns, int64_t kcap,

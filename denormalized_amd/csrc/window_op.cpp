@@ -174,12 +174,26 @@ struct dz_window_op {
     char* h_stage = nullptr; /* pinned */
     size_t h_stage_cap = 0;
 
-    /* async emission: pinned slab pool + worker thread. trigger_windows
-     * enqueues {D2H event, slab}; the worker waits the event, builds the
-     * output batch off the push critical path, and appends to outq. */
-    static constexpr int E_POOL = 32;
+    /* async DEVICE-SIDE emission: at window close the copy stream runs
+     * compact -> stable radix sort by first-seen row -> gather+filter
+     * (kernels.hip launch_emission); a worker thread then copies exactly
+     * nt sorted rows D2H and formats them off the push critical path. */
+    static constexpr int E_POOL = 16;
     static constexpr int E_WORKERS = 4;
-    uint64_t* e_slabs[E_POOL] = {};
+    struct DevEmit { /* per-slot device scratch, carved from one alloc */
+        char* base = nullptr;
+        uint64_t* ekeys; uint32_t* ekid;
+        uint64_t* skeys; uint32_t* skid;
+        uint32_t* counter;
+        uint32_t* okid; uint64_t* ocnt;
+        double* omin; double* omax; double* osum; double* oavg;
+        uint8_t* oflags;
+    };
+    DevEmit e_dev[E_POOL];
+    uint64_t* e_slabs[E_POOL] = {}; /* pinned output staging per slot */
+    uint32_t* e_pcnt = nullptr;     /* pinned per-slot nt counters */
+    uint32_t* d_rhist = nullptr;    /* shared radix scratch (stream-serial) */
+    uint32_t* d_roffs = nullptr;
     int64_t e_slab_kcap = 0;
     struct EmitJob {
         hipEvent_t ev;
@@ -409,6 +423,11 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     for (auto e : op->e_ev_pool) hipEventDestroy(e);
     for (auto& s : op->e_slabs)
         if (s) hipHostFree(s);
+    for (auto& d : op->e_dev)
+        if (d.base) hipFree(d.base);
+    if (op->e_pcnt) hipHostFree(op->e_pcnt);
+    hipFree(op->d_rhist);
+    hipFree(op->d_roffs);
     hipFree(op->s_base);
     hipFree(op->d_ghist); hipFree(op->d_gofs); hipFree(op->d_total); hipFree(op->d_base);
     hipFree(op->d_meta); hipFree(op->d_ridx); hipFree(op->d_rvals);
@@ -435,86 +454,52 @@ extern "C" const char* dz_last_error(dz_window_op* op) {
 /* emission (trigger_windows, grouped_window_agg_stream.rs:220-253)    */
 /* ------------------------------------------------------------------ */
 
-static bool filter_pass(dz_window_op* op, int64_t row_cnt,
-                        double vmin, double vmax, double vsum, bool valid) {
-    if (!op->has_filter) return true;
-    double v;
-    dz_agg_op o = op->aggs[op->f_idx].op;
-    if (o == DZ_AGG_COUNT) {
-        v = (double)row_cnt;
-    } else {
-        if (!valid) return false; /* NULL never passes a comparison filter */
-        switch (o) {
-            case DZ_AGG_MIN: v = vmin; break;
-            case DZ_AGG_MAX: v = vmax; break;
-            case DZ_AGG_SUM: v = vsum; break;
-            case DZ_AGG_AVG: v = vsum / (double)row_cnt; break;
-            default: v = 0; break;
-        }
-    }
-    switch (op->f_cmp) {
-        case 0: return v < op->f_lit;
-        case 1: return v <= op->f_lit;
-        case 2: return v > op->f_lit;
-        case 3: return v >= op->f_lit;
-        case 4: return v == op->f_lit;
-        case 5: return v != op->f_lit;
-        default: return true;
-    }
+
+/* Pinned emission slab layout for kcap entries (45 bytes per entry):
+ * [kid u32 x kcap][flags u8 x kcap][cnt u64][min f64][max f64][sum f64][avg f64]
+ * (kcap is a multiple of NB=512, so every section stays 8-byte aligned). */
+struct SlabView {
+    const uint32_t* kid;
+    const uint8_t* flags;
+    const uint64_t* cnt;
+    const double* mn;
+    const double* mx;
+    const double* sm;
+    const double* av;
+};
+static SlabView slab_view(const uint64_t* slab, int64_t kcap) {
+    const char* p = (const char*)slab;
+    SlabView v;
+    v.kid = (const uint32_t*)p;
+    v.flags = (const uint8_t*)(p + kcap * 4);
+    v.cnt = (const uint64_t*)(p + kcap * 5);
+    v.mn = (const double*)(p + kcap * 13);
+    v.mx = (const double*)(p + kcap * 21);
+    v.sm = (const double*)(p + kcap * 29);
+    v.av = (const double*)(p + kcap * 37);
+    return v;
 }
+static constexpr int64_t SLAB_BYTES_PER_ENTRY = 45;
 
-/* Build one emitted batch from a pinned copy of a slot slab
- * ([cnt][first][min][max][sum], each `kcap` 8-byte entries). Runs on the
- * emission worker thread: touches only immutable config, the deque-backed
- * dictionaries (indices < the job's n_keys snapshot) and the slab. */
+/* Build one emitted batch from the device-sorted, device-filtered columns
+ * (insertion order already established by the GPU radix sort). Runs on the
+ * emission worker pool: touches only immutable config, the deque-backed
+ * dictionaries (indices < the job's n_keys snapshot) and the pinned slab. */
 static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
-                           int64_t K, int64_t kcap, const uint64_t* slab,
+                           uint32_t nt, int64_t kcap, const uint64_t* slab,
                            OutBuf* out) {
-    const uint64_t* f_cnt = slab;
-    const uint64_t* f_first = slab + kcap;
-    const double* f_min = (const double*)(slab + 2 * kcap);
-    const double* f_max = (const double*)(slab + 3 * kcap);
-    const double* f_sum = (const double*)(slab + 4 * kcap);
-
-    /* groups in first-seen (insertion) order: GroupValues emits insertion
-     * order; sort touched keys by first-row sequence (first values are
-     * distinct rows, so the order is exact). LSD radix (4 x 16-bit passes,
-     * passes over all-equal digits skipped) — ~4x faster than std::sort at
-     * the 10k-group scale this runs at per window close. */
-    std::vector<std::pair<uint64_t, int32_t>> touched, scratch;
-    touched.reserve(4096);
-    for (int64_t k = 0; k < K; k++)
-        if (f_first[k] != ~0ULL) touched.emplace_back(f_first[k], (int32_t)k);
-    if (touched.size() > 1) {
-        scratch.resize(touched.size());
-        uint32_t hist[2048];
-        for (int pass = 0; pass < 6; pass++) {
-            int sh = pass * 11;
-            memset(hist, 0, sizeof(hist));
-            for (auto& p : touched) hist[(p.first >> sh) & 0x7FF]++;
-            uint64_t d0 = (touched[0].first >> sh) & 0x7FF;
-            if (hist[d0] == touched.size()) continue; /* all-equal digit */
-            uint32_t run = 0;
-            for (int d = 0; d < 2048; d++) { uint32_t t = hist[d]; hist[d] = run; run += t; }
-            for (auto& p : touched) scratch[hist[(p.first >> sh) & 0x7FF]++] = p;
-            touched.swap(scratch);
-        }
-    }
-
-    /* filter pushdown (datastream.rs:94-105) — keep list */
-    std::vector<int32_t> rows;
-    rows.reserve(touched.size());
-    for (auto& p : touched) {
-        int32_t k = p.second;
-        int64_t cnt = (int64_t)f_cnt[k];
-        if (filter_pass(op, cnt, f_min[k], f_max[k], f_sum[k], cnt > 0))
-            rows.push_back(k);
-    }
-    size_t n = rows.size(), na = op->aggs.size();
-
+    SlabView v = slab_view(slab, kcap);
+    size_t na = op->aggs.size();
     OutBuf ob;
     ob.agg_i64.resize(na);
     ob.agg_f64.resize(na);
+    /* pass-filter compaction list (bit1 of flags; device applied the
+     * predicate — datastream.rs:94-105 semantics incl. NULL-never-passes) */
+    std::vector<uint32_t> rows;
+    rows.reserve(nt);
+    for (uint32_t i = 0; i < nt; i++)
+        if (v.flags[i] & 2) rows.push_back(i);
+    size_t n = rows.size();
     if (op->no_group) {
         /* global aggregate: output schema has no group column
          * (create_schema with empty group exprs, streaming_window.rs:1096+) */
@@ -522,64 +507,58 @@ static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
         ob.key_offsets.resize(n + 1);
         ob.key_offsets[0] = 0;
         size_t total = 0;
-        for (size_t i = 0; i < n; i++) total += op->dict_strs[rows[i]].size();
+        for (size_t i = 0; i < n; i++) total += op->dict_strs[v.kid[rows[i]]].size();
         ob.key_data.resize(total);
         size_t pos = 0;
         for (size_t i = 0; i < n; i++) {
-            const std::string& s = op->dict_strs[rows[i]];
+            const std::string& s = op->dict_strs[v.kid[rows[i]]];
             memcpy(ob.key_data.data() + pos, s.data(), s.size());
             pos += s.size();
             ob.key_offsets[i + 1] = (int32_t)pos;
         }
     } else if (op->key_kind == DZ_KEY_INT64) {
         ob.key_i64.resize(n);
-        for (size_t i = 0; i < n; i++) ob.key_i64[i] = op->dict_vals[rows[i]];
+        for (size_t i = 0; i < n; i++) ob.key_i64[i] = op->dict_vals[v.kid[rows[i]]];
     } else {
         ob.key_i64.resize(n);
-        for (size_t i = 0; i < n; i++) ob.key_i64[i] = rows[i];
+        for (size_t i = 0; i < n; i++) ob.key_i64[i] = v.kid[rows[i]];
     }
     for (size_t a = 0; a < na; a++) {
         switch (op->aggs[a].op) {
             case DZ_AGG_COUNT: {
                 auto& col = ob.agg_i64[a];
                 col.resize(n);
-                for (size_t i = 0; i < n; i++) col[i] = (int64_t)f_cnt[rows[i]];
+                for (size_t i = 0; i < n; i++) col[i] = (int64_t)v.cnt[rows[i]];
                 break;
             }
             case DZ_AGG_MIN: {
                 auto& col = ob.agg_f64[a];
                 col.resize(n);
-                for (size_t i = 0; i < n; i++)
-                    col[i] = f_cnt[rows[i]] ? f_min[rows[i]] : 0.0;
+                for (size_t i = 0; i < n; i++) col[i] = v.mn[rows[i]];
                 break;
             }
             case DZ_AGG_MAX: {
                 auto& col = ob.agg_f64[a];
                 col.resize(n);
-                for (size_t i = 0; i < n; i++)
-                    col[i] = f_cnt[rows[i]] ? f_max[rows[i]] : 0.0;
+                for (size_t i = 0; i < n; i++) col[i] = v.mx[rows[i]];
                 break;
             }
             case DZ_AGG_SUM: {
                 auto& col = ob.agg_f64[a];
                 col.resize(n);
-                for (size_t i = 0; i < n; i++)
-                    col[i] = f_cnt[rows[i]] ? f_sum[rows[i]] : 0.0;
+                for (size_t i = 0; i < n; i++) col[i] = v.sm[rows[i]];
                 break;
             }
             case DZ_AGG_AVG: {
                 auto& col = ob.agg_f64[a];
                 col.resize(n);
-                for (size_t i = 0; i < n; i++) {
-                    uint64_t c = f_cnt[rows[i]];
-                    col[i] = c ? f_sum[rows[i]] / (double)c : 0.0;
-                }
+                for (size_t i = 0; i < n; i++) col[i] = v.av[rows[i]];
                 break;
             }
         }
     }
     ob.agg_valid.resize(n);
-    for (size_t i = 0; i < n; i++) ob.agg_valid[i] = f_cnt[rows[i]] > 0 ? 1 : 0;
+    for (size_t i = 0; i < n; i++) ob.agg_valid[i] = v.flags[rows[i]] & 1;
     ob.wstart.assign(n, wstart);
     ob.wend.assign(n, wend);
     ob.view.n_rows = (int64_t)n;
@@ -605,10 +584,33 @@ static void emit_worker_main(dz_window_op* op) {
             job = op->e_jobs.front();
             op->e_jobs.pop_front();
         }
-        hipEventSynchronize(job.ev);
+        hipEventSynchronize(job.ev); /* device emission + counter D2H done */
+        const uint32_t nt = op->e_pcnt[job.slab];
         auto t0 = std::chrono::steady_clock::now();
+        if (nt > 0) {
+            /* second stage: copy exactly nt sorted rows into the pinned slab */
+            dz_window_op::DevEmit& d = op->e_dev[job.slab];
+            char* s = (char*)op->e_slabs[job.slab];
+            int64_t kc = job.kcap;
+            hipMemcpyAsync(s, d.okid, (size_t)nt * 4, hipMemcpyDeviceToHost,
+                           op->copy_stream);
+            hipMemcpyAsync(s + kc * 4, d.oflags, nt, hipMemcpyDeviceToHost,
+                           op->copy_stream);
+            hipMemcpyAsync(s + kc * 5, d.ocnt, (size_t)nt * 8,
+                           hipMemcpyDeviceToHost, op->copy_stream);
+            hipMemcpyAsync(s + kc * 13, d.omin, (size_t)nt * 8,
+                           hipMemcpyDeviceToHost, op->copy_stream);
+            hipMemcpyAsync(s + kc * 21, d.omax, (size_t)nt * 8,
+                           hipMemcpyDeviceToHost, op->copy_stream);
+            hipMemcpyAsync(s + kc * 29, d.osum, (size_t)nt * 8,
+                           hipMemcpyDeviceToHost, op->copy_stream);
+            hipMemcpyAsync(s + kc * 37, d.oavg, (size_t)nt * 8,
+                           hipMemcpyDeviceToHost, op->copy_stream);
+            hipEventRecord(job.ev, op->copy_stream);
+            hipEventSynchronize(job.ev);
+        }
         OutBuf ob;
-        build_emission(op, job.wstart, job.wend, job.n_keys, job.kcap,
+        build_emission(op, job.wstart, job.wend, nt, job.kcap,
                        op->e_slabs[job.slab], &ob);
         op->e_build_ns += (uint64_t)std::chrono::duration_cast<std::chrono::nanoseconds>(
             std::chrono::steady_clock::now() - t0).count();
@@ -651,9 +653,36 @@ static dz_status trigger_windows(dz_window_op* op) {
     if (closed.empty()) return DZ_OK;
     if (op->e_slab_kcap != op->kcap) {
         emit_drain(op);
-        for (auto& s : op->e_slabs) {
-            if (s) hipHostFree(s);
-            CHK(op, hipHostMalloc((void**)&s, (size_t)op->kcap * 5 * 8));
+        int64_t kc = op->kcap;
+        int64_t nblk = (kc + dz::EMIT_RCHUNK - 1) / dz::EMIT_RCHUNK;
+        hipFree(op->d_rhist);
+        hipFree(op->d_roffs);
+        CHK(op, hipMalloc(&op->d_rhist, (size_t)nblk * dz::EMIT_RBINS * 4));
+        CHK(op, hipMalloc(&op->d_roffs, (size_t)nblk * dz::EMIT_RBINS * 4));
+        if (!op->e_pcnt)
+            CHK(op, hipHostMalloc((void**)&op->e_pcnt, dz_window_op::E_POOL * 4));
+        for (int i = 0; i < dz_window_op::E_POOL; i++) {
+            auto& d = op->e_dev[i];
+            hipFree(d.base);
+            /* ekeys 8 + ekid 4 + skeys 8 + skid 4 + counter + out columns
+             * (kid 4 + cnt 8 + 4x f64 + flags 1) = ~69 B/entry */
+            CHK(op, hipMalloc(&d.base, (size_t)kc * 70 + 64));
+            char* p = d.base;
+            d.ekeys = (uint64_t*)p; p += kc * 8;
+            d.skeys = (uint64_t*)p; p += kc * 8;
+            d.ocnt = (uint64_t*)p; p += kc * 8;
+            d.omin = (double*)p; p += kc * 8;
+            d.omax = (double*)p; p += kc * 8;
+            d.osum = (double*)p; p += kc * 8;
+            d.oavg = (double*)p; p += kc * 8;
+            d.ekid = (uint32_t*)p; p += kc * 4;
+            d.skid = (uint32_t*)p; p += kc * 4;
+            d.okid = (uint32_t*)p; p += kc * 4;
+            d.oflags = (uint8_t*)p; p += kc;
+            d.counter = (uint32_t*)p;
+            if (op->e_slabs[i]) hipHostFree(op->e_slabs[i]);
+            CHK(op, hipHostMalloc((void**)&op->e_slabs[i],
+                                  (size_t)kc * SLAB_BYTES_PER_ENTRY + 16));
         }
         std::lock_guard<std::mutex> lk(op->e_mtx);
         op->e_free.clear();
@@ -662,7 +691,8 @@ static dz_status trigger_windows(dz_window_op* op) {
     }
     size_t stride = (size_t)op->kcap * 5;
     HostTimer ht(op, "h_emit_enqueue");
-    /* copies run on the copy stream AFTER the compute stream's folds */
+    /* emission kernels run on the copy stream AFTER the compute stream's
+     * folds for these slots */
     hipEvent_t evA;
     {
         std::lock_guard<std::mutex> lk(op->e_mtx);
@@ -678,6 +708,20 @@ static dz_status trigger_windows(dz_window_op* op) {
     {
         std::lock_guard<std::mutex> lk(op->e_mtx);
         op->e_ev_pool.push_back(evA);
+    }
+    dz::EmitFilter ef;
+    ef.on = op->has_filter ? 1 : 0;
+    ef.cmp = op->f_cmp;
+    ef.lit = op->f_lit;
+    ef.field = 4;
+    if (op->has_filter) {
+        switch (op->aggs[op->f_idx].op) {
+            case DZ_AGG_COUNT: ef.field = 0; break;
+            case DZ_AGG_MIN: ef.field = 1; break;
+            case DZ_AGG_MAX: ef.field = 2; break;
+            case DZ_AGG_SUM: ef.field = 3; break;
+            default: ef.field = 4; break;
+        }
     }
     for (auto& c : closed) {
         int slab;
@@ -701,8 +745,23 @@ static dz_status trigger_windows(dz_window_op* op) {
             slot_ev = take();
             op->e_inflight++;
         }
-        CHK(op, hipMemcpyAsync(op->e_slabs[slab], op->s_base + (size_t)c.slot * stride,
-                               stride * 8, hipMemcpyDeviceToHost, op->copy_stream));
+        dz_window_op::DevEmit& d = op->e_dev[slab];
+        const uint64_t* sl = op->s_base + (size_t)c.slot * stride;
+        CHK(op, hipMemsetAsync(d.counter, 0, 4, op->copy_stream));
+        if (op->n_keys > 0) {
+            dz::launch_emission(op->copy_stream,
+                                /*first*/ sl + op->kcap,
+                                /*cnt*/ sl,
+                                /*min*/ (const double*)(sl + 2 * op->kcap),
+                                /*max*/ (const double*)(sl + 3 * op->kcap),
+                                /*sum*/ (const double*)(sl + 4 * op->kcap),
+                                op->n_keys, d.ekeys, d.ekid, d.skeys, d.skid,
+                                d.counter, op->d_rhist, op->d_roffs, ef, d.okid,
+                                d.ocnt, d.omin, d.omax, d.osum, d.oavg,
+                                d.oflags);
+        }
+        CHK(op, hipMemcpyAsync(&op->e_pcnt[slab], d.counter, 4,
+                               hipMemcpyDeviceToHost, op->copy_stream));
         CHK(op, hipEventRecord(ev, op->copy_stream));
         CHK(op, hipEventRecord(slot_ev, op->copy_stream));
         op->free_slots.push_back({c.slot, slot_ev});
