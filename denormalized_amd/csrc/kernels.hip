@@ -1009,6 +1009,107 @@ __global__ __launch_bounds__(BLOCK) void k_rscatter(const uint64_t* keys,
     }
 }
 
+
+/* single-block LSD radix for small group counts (nt <= ~64k): all 6 passes
+ * in ONE launch (the multi-block chain is 18+ launches — too launch-heavy
+ * for the common 10k-group window close). Same wave-quarter stable
+ * placement as k_rscatter; global ping-pong between passes. */
+__global__ __launch_bounds__(BLOCK) void k_rsort_small(uint64_t* a_keys,
+        uint32_t* a_pay, uint64_t* b_keys, uint32_t* b_pay,
+        const uint32_t* counter) {
+    __shared__ uint32_t cnt4[WAVES_PER_BLOCK][RBINS];
+    __shared__ uint32_t scanbuf[BLOCK];
+    const uint32_t nt = *counter;
+    if (nt < 2) return;
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    uint64_t* ka = a_keys;
+    uint32_t* pa = a_pay;
+    uint64_t* kb = b_keys;
+    uint32_t* pb = b_pay;
+    for (int pass = 0; pass < RPASSES; pass++) {
+        const int shift = pass * RDIG;
+        for (int t = threadIdx.x; t < RBINS; t += BLOCK)
+            for (int w = 0; w < WAVES_PER_BLOCK; w++) cnt4[w][t] = 0;
+        __syncthreads();
+        const uint32_t q = (nt + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+        const uint32_t w0 = min(nt, (uint32_t)wave * q);
+        const uint32_t w1 = min(nt, w0 + q);
+        for (uint32_t i = w0 + lane; i < w1; i += 64)
+            atomicAdd(&cnt4[wave][(uint32_t)(ka[i] >> shift) & (RBINS - 1)], 1u);
+        __syncthreads();
+        { /* exclusive digit prefix + per-wave bases */
+            constexpr int PER = RBINS / BLOCK;
+            uint32_t loc[PER];
+            uint32_t s = 0;
+            for (int j = 0; j < PER; j++) {
+                int d = threadIdx.x * PER + j;
+                uint32_t t = 0;
+                for (int w = 0; w < WAVES_PER_BLOCK; w++) t += cnt4[w][d];
+                loc[j] = s;
+                s += t;
+            }
+            scanbuf[threadIdx.x] = s;
+            __syncthreads();
+            for (int o = 1; o < BLOCK; o <<= 1) {
+                uint32_t v = (threadIdx.x >= o) ? scanbuf[threadIdx.x - o] : 0;
+                __syncthreads();
+                scanbuf[threadIdx.x] += v;
+                __syncthreads();
+            }
+            uint32_t pre = threadIdx.x ? scanbuf[threadIdx.x - 1] : 0;
+            for (int j = 0; j < PER; j++) {
+                int d = threadIdx.x * PER + j;
+                uint32_t run = pre + loc[j];
+                for (int w = 0; w < WAVES_PER_BLOCK; w++) {
+                    uint32_t c = cnt4[w][d];
+                    cnt4[w][d] = run;
+                    run += c;
+                }
+            }
+        }
+        __syncthreads();
+        for (uint32_t t0 = w0; t0 < w1; t0 += 64) {
+            const uint32_t i = t0 + lane;
+            const bool act = i < w1;
+            uint64_t key = act ? ka[i] : ~0ULL;
+            uint32_t pay = act ? pa[i] : 0;
+            uint32_t d = act ? ((uint32_t)(key >> shift) & (RBINS - 1)) : 0xFFFFu;
+            uint64_t same = ~0ULL;
+            for (int b = 0; b < RDIG; b++) {
+                uint64_t bb = __ballot((d >> b) & 1);
+                same &= ((d >> b) & 1) ? bb : ~bb;
+            }
+            {
+                uint64_t bb = __ballot(act);
+                same &= act ? bb : ~bb;
+            }
+            const uint64_t below =
+                (lane == 63) ? ~0ULL : ((1ULL << (lane + 1)) - 1);
+            const int rank = (int)__popcll(same & below) - 1;
+            const int leader = __ffsll((unsigned long long)same) - 1;
+            const uint32_t wtot = (uint32_t)__popcll(same);
+            uint32_t pos = 0;
+            {
+                uint32_t pre = 0;
+                if (lane == leader && act) {
+                    pre = cnt4[wave][d];
+                    cnt4[wave][d] = pre + wtot;
+                }
+                pre = (uint32_t)__shfl((int)pre, leader);
+                pos = pre + (uint32_t)rank;
+            }
+            if (act) {
+                kb[pos] = key;
+                pb[pos] = pay;
+            }
+        }
+        __syncthreads();
+        uint64_t* tk = ka; ka = kb; kb = tk;
+        uint32_t* tp = pa; pa = pb; pb = tp;
+    }
+}
+
 /* gather the sorted groups' aggregates + filter flag into output columns */
 __global__ void k_egather(const uint32_t* skid, const uint32_t* counter,
                           const uint64_t* s_cnt, const double* s_min,
@@ -1072,21 +1173,27 @@ void launch_emission(hipStream_t s, const uint64_t* slab_first,
     int cblocks = (int)std::min<int64_t>((K + BLOCK - 1) / BLOCK, 2048);
     hipLaunchKernelGGL(k_ecompact, dim3(cblocks), dim3(BLOCK), 0, s, slab_first,
                        K, ekeys, ekid, counter);
-    int nblk = (int)((K + RCHUNK - 1) / RCHUNK);
-    uint64_t* ka = ekeys;
-    uint32_t* pa = ekid;
-    uint64_t* kb = skeys;
-    uint32_t* pb = skid;
-    for (int p = 0; p < RPASSES; p++) {
-        int shift = p * RDIG;
-        hipLaunchKernelGGL(k_rhist, dim3(nblk), dim3(BLOCK), 0, s, ka, counter,
-                           shift, rhist);
-        hipLaunchKernelGGL(k_rscan, dim3(1), dim3(1024), 0, s, rhist, nblk,
-                           roffs);
-        hipLaunchKernelGGL(k_rscatter, dim3(nblk), dim3(BLOCK), 0, s, ka, pa,
-                           counter, shift, roffs, kb, pb);
-        std::swap(ka, kb);
-        std::swap(pa, pb);
+    if (K <= 65536) {
+        /* single launch covers all 6 passes (ends back in ekeys/ekid) */
+        hipLaunchKernelGGL(k_rsort_small, dim3(1), dim3(BLOCK), 0, s, ekeys,
+                           ekid, skeys, skid, counter);
+    } else {
+        int nblk = (int)((K + RCHUNK - 1) / RCHUNK);
+        uint64_t* ka = ekeys;
+        uint32_t* pa = ekid;
+        uint64_t* kb = skeys;
+        uint32_t* pb = skid;
+        for (int p = 0; p < RPASSES; p++) {
+            int shift = p * RDIG;
+            hipLaunchKernelGGL(k_rhist, dim3(nblk), dim3(BLOCK), 0, s, ka,
+                               counter, shift, rhist);
+            hipLaunchKernelGGL(k_rscan, dim3(1), dim3(1024), 0, s, rhist, nblk,
+                               roffs);
+            hipLaunchKernelGGL(k_rscatter, dim3(nblk), dim3(BLOCK), 0, s, ka, pa,
+                               counter, shift, roffs, kb, pb);
+            std::swap(ka, kb);
+            std::swap(pa, pb);
+        }
     }
     /* RPASSES is even => sorted data ends in (ekeys, ekid) */
     hipLaunchKernelGGL(k_egather, dim3(cblocks), dim3(BLOCK), 0, s, ekid, counter,
