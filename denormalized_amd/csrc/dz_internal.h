@@ -93,17 +93,21 @@ struct EmitFilter {
     double lit;
 };
 
-/* device-side emission at window close: compact touched groups from the
- * slot slab, stable radix sort by first-seen row, gather aggregate columns
- * + filter flags. All launches on the given (copy) stream. */
-void launch_emission(hipStream_t stream, const uint64_t* slab_first,
-                     const uint64_t* slab_cnt, const double* slab_min,
-                     const double* slab_max, const double* slab_sum, int64_t K,
-                     uint64_t* ekeys, uint32_t* ekid, uint64_t* skeys,
-                     uint32_t* skid, uint32_t* counter, uint32_t* rhist,
-                     uint32_t* roffs, const EmitFilter& ef, uint32_t* okid,
-                     uint64_t* ocnt, double* omin, double* omax, double* osum,
-                     double* oavg, uint8_t* oflags);
+/* device-side emission at window close, two stages on the copy stream:
+ * slabread = compact touched groups + gather aggregate columns + filter
+ * flags (after which the window slot is reusable); sort = stable radix by
+ * first-seen row, leaving the sorted compact-index permutation in `skid`. */
+void launch_emission_slabread(hipStream_t stream, const uint64_t* slab_first,
+                              const uint64_t* slab_cnt, const double* slab_min,
+                              const double* slab_max, const double* slab_sum,
+                              int64_t K, uint64_t* ekeys, uint32_t* ekid,
+                              uint32_t* skid, uint32_t* counter,
+                              const EmitFilter& ef, uint64_t* ocnt, double* omin,
+                              double* omax, double* osum, double* oavg,
+                              uint8_t* oflags);
+void launch_emission_sort(hipStream_t stream, int64_t K, uint64_t* ekeys,
+                          uint64_t* skeys, uint32_t* skid, uint32_t* okid,
+                          uint32_t* counter, uint32_t* rhist, uint32_t* roffs);
 constexpr int EMIT_RCHUNK = 4096;
 constexpr int EMIT_RBINS = 2048;
 

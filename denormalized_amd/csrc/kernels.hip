@@ -874,7 +874,7 @@ constexpr int RCHUNK = 4096;      /* elements per radix block */
 constexpr int RPASSES = 6;        /* 6*11 = 66 >= 64 bits */
 
 __global__ void k_ecompact(const uint64_t* s_first, int64_t K, uint64_t* ekeys,
-                           uint32_t* ekid, uint32_t* counter) {
+                           uint32_t* ekid, uint32_t* eiota, uint32_t* counter) {
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < K;
          k += stride) {
@@ -883,6 +883,7 @@ __global__ void k_ecompact(const uint64_t* s_first, int64_t K, uint64_t* ekeys,
             uint32_t p = atomicAdd(counter, 1u);
             ekeys[p] = f;
             ekid[p] = (uint32_t)k;
+            eiota[p] = p; /* sort payload: compact position */
         }
     }
 }
@@ -1010,15 +1011,20 @@ __global__ __launch_bounds__(BLOCK) void k_rscatter(const uint64_t* keys,
 }
 
 
-/* single-block LSD radix for small group counts (nt <= ~64k): all 6 passes
- * in ONE launch (the multi-block chain is 18+ launches — too launch-heavy
- * for the common 10k-group window close). Same wave-quarter stable
- * placement as k_rscatter; global ping-pong between passes. */
-__global__ __launch_bounds__(BLOCK) void k_rsort_small(uint64_t* a_keys,
+/* single-block LSD radix for small group counts (nt <= ~64k): all passes in
+ * ONE launch; 1024 threads (16 wave-quarters) and all-equal digits skipped
+ * device-side (typical `first` keys use < 40 of the 64 bits). Stable
+ * wave-quarter placement; global ping-pong between active passes. The sort
+ * runs AFTER the slab readers, so it never gates window-slot reuse. */
+constexpr int SBLK = 1024;
+constexpr int SWAVES = SBLK / 64;
+
+__global__ __launch_bounds__(SBLK) void k_rsort_small(uint64_t* a_keys,
         uint32_t* a_pay, uint64_t* b_keys, uint32_t* b_pay,
         const uint32_t* counter) {
-    __shared__ uint32_t cnt4[WAVES_PER_BLOCK][RBINS];
-    __shared__ uint32_t scanbuf[BLOCK];
+    __shared__ uint32_t cnt4[SWAVES][RBINS];
+    __shared__ uint32_t scanbuf[SBLK];
+    __shared__ int allsame;
     const uint32_t nt = *counter;
     if (nt < 2) return;
     const int lane = threadIdx.x & 63;
@@ -1027,31 +1033,34 @@ __global__ __launch_bounds__(BLOCK) void k_rsort_small(uint64_t* a_keys,
     uint32_t* pa = a_pay;
     uint64_t* kb = b_keys;
     uint32_t* pb = b_pay;
+    const uint32_t q = (nt + SWAVES - 1) / SWAVES;
+    const uint32_t w0 = min(nt, (uint32_t)wave * q);
+    const uint32_t w1 = min(nt, w0 + q);
     for (int pass = 0; pass < RPASSES; pass++) {
         const int shift = pass * RDIG;
-        for (int t = threadIdx.x; t < RBINS; t += BLOCK)
-            for (int w = 0; w < WAVES_PER_BLOCK; w++) cnt4[w][t] = 0;
+        for (int t = threadIdx.x; t < RBINS; t += SBLK)
+            for (int w = 0; w < SWAVES; w++) cnt4[w][t] = 0;
+        if (threadIdx.x == 0) allsame = 0;
         __syncthreads();
-        const uint32_t q = (nt + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
-        const uint32_t w0 = min(nt, (uint32_t)wave * q);
-        const uint32_t w1 = min(nt, w0 + q);
         for (uint32_t i = w0 + lane; i < w1; i += 64)
             atomicAdd(&cnt4[wave][(uint32_t)(ka[i] >> shift) & (RBINS - 1)], 1u);
         __syncthreads();
-        { /* exclusive digit prefix + per-wave bases */
-            constexpr int PER = RBINS / BLOCK;
+        { /* exclusive digit prefix + per-wave bases; detect all-equal digit */
+            constexpr int PER = RBINS / SBLK; /* 2 */
             uint32_t loc[PER];
             uint32_t s = 0;
             for (int j = 0; j < PER; j++) {
                 int d = threadIdx.x * PER + j;
                 uint32_t t = 0;
-                for (int w = 0; w < WAVES_PER_BLOCK; w++) t += cnt4[w][d];
+                for (int w = 0; w < SWAVES; w++) t += cnt4[w][d];
+                if (t == nt) allsame = 1;
                 loc[j] = s;
                 s += t;
             }
             scanbuf[threadIdx.x] = s;
             __syncthreads();
-            for (int o = 1; o < BLOCK; o <<= 1) {
+            if (allsame) continue; /* this digit sorts nothing: skip pass */
+            for (int o = 1; o < SBLK; o <<= 1) {
                 uint32_t v = (threadIdx.x >= o) ? scanbuf[threadIdx.x - o] : 0;
                 __syncthreads();
                 scanbuf[threadIdx.x] += v;
@@ -1061,7 +1070,7 @@ __global__ __launch_bounds__(BLOCK) void k_rsort_small(uint64_t* a_keys,
             for (int j = 0; j < PER; j++) {
                 int d = threadIdx.x * PER + j;
                 uint32_t run = pre + loc[j];
-                for (int w = 0; w < WAVES_PER_BLOCK; w++) {
+                for (int w = 0; w < SWAVES; w++) {
                     uint32_t c = cnt4[w][d];
                     cnt4[w][d] = run;
                     run += c;
@@ -1108,20 +1117,28 @@ __global__ __launch_bounds__(BLOCK) void k_rsort_small(uint64_t* a_keys,
         uint64_t* tk = ka; ka = kb; kb = tk;
         uint32_t* tp = pa; pa = pb; pb = tp;
     }
+    /* variable number of swaps (skipped passes): ensure the result lands in
+     * (a_keys, a_pay) — copy back if it ended in b */
+    if (ka != a_keys) {
+        for (uint32_t i = threadIdx.x; i < nt; i += SBLK) {
+            a_keys[i] = ka[i];
+            a_pay[i] = pa[i];
+        }
+    }
 }
 
 /* gather the sorted groups' aggregates + filter flag into output columns */
-__global__ void k_egather(const uint32_t* skid, const uint32_t* counter,
+__global__ void k_egather(const uint32_t* ckid, const uint32_t* counter,
                           const uint64_t* s_cnt, const double* s_min,
                           const double* s_max, const double* s_sum,
-                          EmitFilter ef, uint32_t* okid, uint64_t* ocnt,
+                          EmitFilter ef, uint64_t* ocnt,
                           double* omin, double* omax, double* osum,
                           double* oavg, uint8_t* oflags) {
     const uint32_t nt = *counter;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nt;
          i += stride) {
-        const uint32_t kid = skid[i];
+        const uint32_t kid = ckid[i];
         const uint64_t c = s_cnt[kid];
         const bool valid = c > 0;
         const double mn = valid ? s_min[kid] : 0.0;
@@ -1152,7 +1169,6 @@ __global__ void k_egather(const uint32_t* skid, const uint32_t* counter,
                 }
             }
         }
-        okid[i] = kid;
         ocnt[i] = c;
         omin[i] = mn;
         omax[i] = mx;
@@ -1162,27 +1178,38 @@ __global__ void k_egather(const uint32_t* skid, const uint32_t* counter,
     }
 }
 
-void launch_emission(hipStream_t s, const uint64_t* slab_first,
-                     const uint64_t* slab_cnt, const double* slab_min,
-                     const double* slab_max, const double* slab_sum, int64_t K,
-                     uint64_t* ekeys, uint32_t* ekid, uint64_t* skeys,
-                     uint32_t* skid, uint32_t* counter, uint32_t* rhist,
-                     uint32_t* roffs, const EmitFilter& ef, uint32_t* okid,
-                     uint64_t* ocnt, double* omin, double* omax, double* osum,
-                     double* oavg, uint8_t* oflags) {
+void launch_emission_slabread(hipStream_t s, const uint64_t* slab_first,
+                              const uint64_t* slab_cnt, const double* slab_min,
+                              const double* slab_max, const double* slab_sum,
+                              int64_t K, uint64_t* ekeys, uint32_t* ekid,
+                              uint32_t* skid, uint32_t* counter,
+                              const EmitFilter& ef, uint64_t* ocnt, double* omin,
+                              double* omax, double* osum, double* oavg,
+                              uint8_t* oflags) {
     int cblocks = (int)std::min<int64_t>((K + BLOCK - 1) / BLOCK, 2048);
+    /* compact (reads slab.first) -> gather by compact order (reads the other
+     * slab fields). After these two launches the window slot is reusable. */
     hipLaunchKernelGGL(k_ecompact, dim3(cblocks), dim3(BLOCK), 0, s, slab_first,
-                       K, ekeys, ekid, counter);
+                       K, ekeys, ekid, skid, counter);
+    hipLaunchKernelGGL(k_egather, dim3(cblocks), dim3(BLOCK), 0, s, ekid, counter,
+                       slab_cnt, slab_min, slab_max, slab_sum, ef, ocnt, omin,
+                       omax, osum, oavg, oflags);
+}
+
+void launch_emission_sort(hipStream_t s, int64_t K, uint64_t* ekeys,
+                          uint64_t* skeys, uint32_t* skid, uint32_t* okid,
+                          uint32_t* counter, uint32_t* rhist, uint32_t* roffs) {
+    /* sort (first, compact-index) pairs: keys ekeys<->skeys, payload
+     * skid<->okid; both paths leave the sorted payload in skid */
     if (K <= 65536) {
-        /* single launch covers all 6 passes (ends back in ekeys/ekid) */
-        hipLaunchKernelGGL(k_rsort_small, dim3(1), dim3(BLOCK), 0, s, ekeys,
-                           ekid, skeys, skid, counter);
+        hipLaunchKernelGGL(k_rsort_small, dim3(1), dim3(SBLK), 0, s, ekeys, skid,
+                           skeys, okid, counter);
     } else {
         int nblk = (int)((K + RCHUNK - 1) / RCHUNK);
         uint64_t* ka = ekeys;
-        uint32_t* pa = ekid;
+        uint32_t* pa = skid;
         uint64_t* kb = skeys;
-        uint32_t* pb = skid;
+        uint32_t* pb = okid;
         for (int p = 0; p < RPASSES; p++) {
             int shift = p * RDIG;
             hipLaunchKernelGGL(k_rhist, dim3(nblk), dim3(BLOCK), 0, s, ka,
@@ -1194,11 +1221,8 @@ void launch_emission(hipStream_t s, const uint64_t* slab_first,
             std::swap(ka, kb);
             std::swap(pa, pb);
         }
+        /* RPASSES even => payload back in skid */
     }
-    /* RPASSES is even => sorted data ends in (ekeys, ekid) */
-    hipLaunchKernelGGL(k_egather, dim3(cblocks), dim3(BLOCK), 0, s, ekid, counter,
-                       slab_cnt, slab_min, slab_max, slab_sum, ef, okid, ocnt,
-                       omin, omax, osum, oavg, oflags);
 }
 
 /* batched reset of freshly (re)allocated window slots:

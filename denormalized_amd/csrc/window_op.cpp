@@ -455,11 +455,14 @@ extern "C" const char* dz_last_error(dz_window_op* op) {
 /* ------------------------------------------------------------------ */
 
 
-/* Pinned emission slab layout for kcap entries (45 bytes per entry):
- * [kid u32 x kcap][flags u8 x kcap][cnt u64][min f64][max f64][sum f64][avg f64]
- * (kcap is a multiple of NB=512, so every section stays 8-byte aligned). */
+/* Pinned emission slab layout for kcap entries (49 bytes per entry):
+ * [kid u32][sidx u32][flags u8][cnt u64][min f64][max f64][sum f64][avg f64]
+ * column sections; kid/cols are in COMPACT order, sidx is the sorted
+ * (insertion-order) permutation of compact indices. kcap is a multiple of
+ * NB=512, so every section stays 8-byte aligned. */
 struct SlabView {
     const uint32_t* kid;
+    const uint32_t* sidx;
     const uint8_t* flags;
     const uint64_t* cnt;
     const double* mn;
@@ -471,15 +474,16 @@ static SlabView slab_view(const uint64_t* slab, int64_t kcap) {
     const char* p = (const char*)slab;
     SlabView v;
     v.kid = (const uint32_t*)p;
-    v.flags = (const uint8_t*)(p + kcap * 4);
-    v.cnt = (const uint64_t*)(p + kcap * 5);
-    v.mn = (const double*)(p + kcap * 13);
-    v.mx = (const double*)(p + kcap * 21);
-    v.sm = (const double*)(p + kcap * 29);
-    v.av = (const double*)(p + kcap * 37);
+    v.sidx = (const uint32_t*)(p + kcap * 4);
+    v.flags = (const uint8_t*)(p + kcap * 8);
+    v.cnt = (const uint64_t*)(p + kcap * 9);
+    v.mn = (const double*)(p + kcap * 17);
+    v.mx = (const double*)(p + kcap * 25);
+    v.sm = (const double*)(p + kcap * 33);
+    v.av = (const double*)(p + kcap * 41);
     return v;
 }
-static constexpr int64_t SLAB_BYTES_PER_ENTRY = 45;
+static constexpr int64_t SLAB_BYTES_PER_ENTRY = 49;
 
 /* Build one emitted batch from the device-sorted, device-filtered columns
  * (insertion order already established by the GPU radix sort). Runs on the
@@ -493,12 +497,15 @@ static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
     OutBuf ob;
     ob.agg_i64.resize(na);
     ob.agg_f64.resize(na);
-    /* pass-filter compaction list (bit1 of flags; device applied the
-     * predicate — datastream.rs:94-105 semantics incl. NULL-never-passes) */
+    /* walk groups in insertion order (the device-sorted permutation) and
+     * keep filter passers (bit1 of flags; device applied the predicate —
+     * datastream.rs:94-105 semantics incl. NULL-never-passes) */
     std::vector<uint32_t> rows;
     rows.reserve(nt);
-    for (uint32_t i = 0; i < nt; i++)
-        if (v.flags[i] & 2) rows.push_back(i);
+    for (uint32_t i = 0; i < nt; i++) {
+        uint32_t j = v.sidx[i];
+        if (v.flags[j] & 2) rows.push_back(j);
+    }
     size_t n = rows.size();
     if (op->no_group) {
         /* global aggregate: output schema has no group column
@@ -592,19 +599,21 @@ static void emit_worker_main(dz_window_op* op) {
             dz_window_op::DevEmit& d = op->e_dev[job.slab];
             char* s = (char*)op->e_slabs[job.slab];
             int64_t kc = job.kcap;
-            hipMemcpyAsync(s, d.okid, (size_t)nt * 4, hipMemcpyDeviceToHost,
+            hipMemcpyAsync(s, d.ekid, (size_t)nt * 4, hipMemcpyDeviceToHost,
                            op->copy_stream);
-            hipMemcpyAsync(s + kc * 4, d.oflags, nt, hipMemcpyDeviceToHost,
+            hipMemcpyAsync(s + kc * 4, d.skid, (size_t)nt * 4,
+                           hipMemcpyDeviceToHost, op->copy_stream);
+            hipMemcpyAsync(s + kc * 8, d.oflags, nt, hipMemcpyDeviceToHost,
                            op->copy_stream);
-            hipMemcpyAsync(s + kc * 5, d.ocnt, (size_t)nt * 8,
+            hipMemcpyAsync(s + kc * 9, d.ocnt, (size_t)nt * 8,
                            hipMemcpyDeviceToHost, op->copy_stream);
-            hipMemcpyAsync(s + kc * 13, d.omin, (size_t)nt * 8,
+            hipMemcpyAsync(s + kc * 17, d.omin, (size_t)nt * 8,
                            hipMemcpyDeviceToHost, op->copy_stream);
-            hipMemcpyAsync(s + kc * 21, d.omax, (size_t)nt * 8,
+            hipMemcpyAsync(s + kc * 25, d.omax, (size_t)nt * 8,
                            hipMemcpyDeviceToHost, op->copy_stream);
-            hipMemcpyAsync(s + kc * 29, d.osum, (size_t)nt * 8,
+            hipMemcpyAsync(s + kc * 33, d.osum, (size_t)nt * 8,
                            hipMemcpyDeviceToHost, op->copy_stream);
-            hipMemcpyAsync(s + kc * 37, d.oavg, (size_t)nt * 8,
+            hipMemcpyAsync(s + kc * 41, d.oavg, (size_t)nt * 8,
                            hipMemcpyDeviceToHost, op->copy_stream);
             hipEventRecord(job.ev, op->copy_stream);
             hipEventSynchronize(job.ev);
@@ -749,22 +758,27 @@ static dz_status trigger_windows(dz_window_op* op) {
         const uint64_t* sl = op->s_base + (size_t)c.slot * stride;
         CHK(op, hipMemsetAsync(d.counter, 0, 4, op->copy_stream));
         if (op->n_keys > 0) {
-            dz::launch_emission(op->copy_stream,
-                                /*first*/ sl + op->kcap,
-                                /*cnt*/ sl,
-                                /*min*/ (const double*)(sl + 2 * op->kcap),
-                                /*max*/ (const double*)(sl + 3 * op->kcap),
-                                /*sum*/ (const double*)(sl + 4 * op->kcap),
-                                op->n_keys, d.ekeys, d.ekid, d.skeys, d.skid,
-                                d.counter, op->d_rhist, op->d_roffs, ef, d.okid,
-                                d.ocnt, d.omin, d.omax, d.osum, d.oavg,
-                                d.oflags);
+            dz::launch_emission_slabread(op->copy_stream,
+                                         /*first*/ sl + op->kcap,
+                                         /*cnt*/ sl,
+                                         /*min*/ (const double*)(sl + 2 * op->kcap),
+                                         /*max*/ (const double*)(sl + 3 * op->kcap),
+                                         /*sum*/ (const double*)(sl + 4 * op->kcap),
+                                         op->n_keys, d.ekeys, d.ekid, d.skid,
+                                         d.counter, ef, d.ocnt, d.omin, d.omax,
+                                         d.osum, d.oavg, d.oflags);
         }
+        /* the slot is reusable as soon as the slab readers above are done —
+         * the sort below never gates the compute stream */
+        CHK(op, hipEventRecord(slot_ev, op->copy_stream));
+        op->free_slots.push_back({c.slot, slot_ev});
+        if (op->n_keys > 0)
+            dz::launch_emission_sort(op->copy_stream, op->n_keys, d.ekeys,
+                                     d.skeys, d.skid, d.okid, d.counter,
+                                     op->d_rhist, op->d_roffs);
         CHK(op, hipMemcpyAsync(&op->e_pcnt[slab], d.counter, 4,
                                hipMemcpyDeviceToHost, op->copy_stream));
         CHK(op, hipEventRecord(ev, op->copy_stream));
-        CHK(op, hipEventRecord(slot_ev, op->copy_stream));
-        op->free_slots.push_back({c.slot, slot_ev});
         {
             std::lock_guard<std::mutex> lk(op->e_mtx);
             op->e_jobs.push_back({ev, slab, c.start, c.end, op->n_keys, op->kcap,
