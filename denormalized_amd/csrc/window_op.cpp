@@ -599,33 +599,9 @@ static void emit_worker_main(dz_window_op* op) {
             job = op->e_jobs.front();
             op->e_jobs.pop_front();
         }
-        event_spin(job.ev); /* device emission + counter D2H done */
+        event_spin(job.ev); /* device emission + column D2H complete */
         const uint32_t nt = op->e_pcnt[job.slab];
         auto t0 = std::chrono::steady_clock::now();
-        if (nt > 0) {
-            /* second stage: copy exactly nt sorted rows into the pinned slab */
-            dz_window_op::DevEmit& d = op->e_dev[job.slab];
-            char* s = (char*)op->e_slabs[job.slab];
-            int64_t kc = job.kcap;
-            hipMemcpyAsync(s, d.ekid, (size_t)nt * 4, hipMemcpyDeviceToHost,
-                           op->copy_stream);
-            hipMemcpyAsync(s + kc * 4, d.skid, (size_t)nt * 4,
-                           hipMemcpyDeviceToHost, op->copy_stream);
-            hipMemcpyAsync(s + kc * 8, d.oflags, nt, hipMemcpyDeviceToHost,
-                           op->copy_stream);
-            hipMemcpyAsync(s + kc * 9, d.ocnt, (size_t)nt * 8,
-                           hipMemcpyDeviceToHost, op->copy_stream);
-            hipMemcpyAsync(s + kc * 17, d.omin, (size_t)nt * 8,
-                           hipMemcpyDeviceToHost, op->copy_stream);
-            hipMemcpyAsync(s + kc * 25, d.omax, (size_t)nt * 8,
-                           hipMemcpyDeviceToHost, op->copy_stream);
-            hipMemcpyAsync(s + kc * 33, d.osum, (size_t)nt * 8,
-                           hipMemcpyDeviceToHost, op->copy_stream);
-            hipMemcpyAsync(s + kc * 41, d.oavg, (size_t)nt * 8,
-                           hipMemcpyDeviceToHost, op->copy_stream);
-            hipEventRecord(job.ev, op->copy_stream);
-            event_spin(job.ev);
-        }
         OutBuf ob;
         build_emission(op, job.wstart, job.wend, nt, job.kcap,
                        op->e_slabs[job.slab], &ob);
@@ -780,10 +756,32 @@ static dz_status trigger_windows(dz_window_op* op) {
          * the sort below never gates the compute stream */
         CHK(op, hipEventRecord(slot_ev, op->copy_stream));
         op->free_slots.push_back({c.slot, slot_ev});
-        if (op->n_keys > 0)
+        if (op->n_keys > 0) {
             dz::launch_emission_sort(op->copy_stream, op->n_keys, d.ekeys,
                                      d.skeys, d.skid, d.okid, d.counter,
                                      op->d_rhist, op->d_roffs);
+            /* full-capacity column D2H (nt is not host-known yet; the slab
+             * sections beyond nt are simply unread by the builder) */
+            char* s = (char*)op->e_slabs[slab];
+            size_t kc = (size_t)op->kcap;
+            size_t K = (size_t)op->n_keys;
+            CHK(op, hipMemcpyAsync(s, d.ekid, K * 4, hipMemcpyDeviceToHost,
+                                   op->copy_stream));
+            CHK(op, hipMemcpyAsync(s + kc * 4, d.skid, K * 4,
+                                   hipMemcpyDeviceToHost, op->copy_stream));
+            CHK(op, hipMemcpyAsync(s + kc * 8, d.oflags, K,
+                                   hipMemcpyDeviceToHost, op->copy_stream));
+            CHK(op, hipMemcpyAsync(s + kc * 9, d.ocnt, K * 8,
+                                   hipMemcpyDeviceToHost, op->copy_stream));
+            CHK(op, hipMemcpyAsync(s + kc * 17, d.omin, K * 8,
+                                   hipMemcpyDeviceToHost, op->copy_stream));
+            CHK(op, hipMemcpyAsync(s + kc * 25, d.omax, K * 8,
+                                   hipMemcpyDeviceToHost, op->copy_stream));
+            CHK(op, hipMemcpyAsync(s + kc * 33, d.osum, K * 8,
+                                   hipMemcpyDeviceToHost, op->copy_stream));
+            CHK(op, hipMemcpyAsync(s + kc * 41, d.oavg, K * 8,
+                                   hipMemcpyDeviceToHost, op->copy_stream));
+        }
         CHK(op, hipMemcpyAsync(&op->e_pcnt[slab], d.counter, 4,
                                hipMemcpyDeviceToHost, op->copy_stream));
         CHK(op, hipEventRecord(ev, op->copy_stream));
