@@ -716,6 +716,13 @@ static dz_status trigger_windows(dz_window_op* op) {
             default: ef.field = 4; break;
         }
     }
+    /* phase 1: slab readers only (compact+gather) for EVERY closed window,
+     * so all slot-release events precede the sorts/copies on the copy
+     * stream — window slots come back to the compute stream immediately
+     * after its folds, not behind emission work. */
+    struct Pending { int slab; hipEvent_t ev; };
+    std::vector<Pending> pend;
+    pend.reserve(closed.size());
     for (auto& c : closed) {
         int slab;
         hipEvent_t ev, slot_ev;
@@ -752,16 +759,20 @@ static dz_status trigger_windows(dz_window_op* op) {
                                          d.counter, ef, d.ocnt, d.omin, d.omax,
                                          d.osum, d.oavg, d.oflags);
         }
-        /* the slot is reusable as soon as the slab readers above are done —
-         * the sort below never gates the compute stream */
         CHK(op, hipEventRecord(slot_ev, op->copy_stream));
         op->free_slots.push_back({c.slot, slot_ev});
+        pend.push_back({slab, ev});
+    }
+    /* phase 2: sorts + column D2H + job hand-off */
+    for (size_t ci = 0; ci < closed.size(); ci++) {
+        auto& c = closed[ci];
+        int slab = pend[ci].slab;
+        hipEvent_t ev = pend[ci].ev;
+        dz_window_op::DevEmit& d = op->e_dev[slab];
         if (op->n_keys > 0) {
             dz::launch_emission_sort(op->copy_stream, op->n_keys, d.ekeys,
                                      d.skeys, d.skid, d.okid, d.counter,
                                      op->d_rhist, op->d_roffs);
-            /* full-capacity column D2H (nt is not host-known yet; the slab
-             * sections beyond nt are simply unread by the builder) */
             char* s = (char*)op->e_slabs[slab];
             size_t kc = (size_t)op->kcap;
             size_t K = (size_t)op->n_keys;
