@@ -716,14 +716,20 @@ static dz_status trigger_windows(dz_window_op* op) {
             default: ef.field = 4; break;
         }
     }
-    /* phase 1: slab readers only (compact+gather) for EVERY closed window,
-     * so all slot-release events precede the sorts/copies on the copy
-     * stream — window slots come back to the compute stream immediately
-     * after its folds, not behind emission work. */
+    /* Two phases per GROUP of closes: slab readers (compact+gather) for the
+     * whole group first, so the slot-release events precede the group's
+     * sorts/copies on the copy stream — window slots come back to the
+     * compute stream right after its folds, not behind emission work.
+     * Groups are bounded by half the slab pool: phase 1 must never wait for
+     * slabs whose release depends on this group's phase 2 (deadlock). */
     struct Pending { int slab; hipEvent_t ev; };
+    constexpr size_t EGROUP = dz_window_op::E_POOL / 2;
+    for (size_t g0 = 0; g0 < closed.size(); g0 += EGROUP) {
+    const size_t g1 = std::min(closed.size(), g0 + EGROUP);
     std::vector<Pending> pend;
-    pend.reserve(closed.size());
-    for (auto& c : closed) {
+    pend.reserve(g1 - g0);
+    for (size_t ci = g0; ci < g1; ci++) {
+        auto& c = closed[ci];
         int slab;
         hipEvent_t ev, slot_ev;
         {
@@ -764,10 +770,10 @@ static dz_status trigger_windows(dz_window_op* op) {
         pend.push_back({slab, ev});
     }
     /* phase 2: sorts + column D2H + job hand-off */
-    for (size_t ci = 0; ci < closed.size(); ci++) {
+    for (size_t ci = g0; ci < g1; ci++) {
         auto& c = closed[ci];
-        int slab = pend[ci].slab;
-        hipEvent_t ev = pend[ci].ev;
+        int slab = pend[ci - g0].slab;
+        hipEvent_t ev = pend[ci - g0].ev;
         dz_window_op::DevEmit& d = op->e_dev[slab];
         if (op->n_keys > 0) {
             dz::launch_emission_sort(op->copy_stream, op->n_keys, d.ekeys,
@@ -803,6 +809,7 @@ static dz_status trigger_windows(dz_window_op* op) {
         }
         op->e_cv.notify_all();
     }
+    } /* group loop */
     return DZ_OK;
 }
 
