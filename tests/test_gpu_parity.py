@@ -583,3 +583,29 @@ def test_global_partial_final_merge_two_shards():
         assert row["max"] == exp["max"][i]
         # sum merges in rank order: tolerance vs single-stream row order
         assert abs(row["sum"] - exp["sum"][i]) <= 1e-9 * abs(exp["sum"][i])
+
+
+def test_device_emission_filter_large_keyspace():
+    # >64k keys takes the DEVICE emission path (compact+sort+gather+filter
+    # on GPU); verify filter semantics match the oracle there too
+    rng = np.random.default_rng(93)
+    n = 300_000
+    ts = (1_000_000 + np.arange(n) // 300).astype(np.int64)
+    k = rng.integers(0, 80_000, n)
+    v = rng.uniform(0, 115, n)
+    op = make_op(1000, n_keys_hint=80_000)
+    op.set_filter("max", ">", 90.0)
+    o = pyoracle.Oracle(1000, 0)
+    op.push(ts, k, v)
+    o.push(ts, k, v)
+    op.finish()
+    o.finish()
+    outs = op.poll_all()
+    exp = o.fetch()
+    keep = exp["max"] > 90.0
+    assert np.array_equal(np.asarray(cat(outs, "key")), exp["key"][keep])
+    assert np.array_equal(cat(outs, "count"), exp["count"][keep])
+    assert np.array_equal(cat(outs, "min"), exp["min"][keep])
+    assert np.array_equal(cat(outs, "avg"), exp["avg"][keep])
+    op.close()
+    o.close()
