@@ -400,6 +400,7 @@ extern "C" dz_window_op* dz_window_op_create(const dz_window_desc* desc) {
     if (state_alloc(op, hint, 4) != DZ_OK) {
         g_err = op->err;
         hipStreamDestroy(op->stream);
+        hipStreamDestroy(op->copy_stream);
         delete op;
         return nullptr;
     }
