@@ -365,7 +365,7 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
             scanbuf[threadIdx.x] = s;
             __syncthreads();
             for (int o = 1; o < BLOCK; o <<= 1) {
-                uint32_t v = (threadIdx.x >= o) ? scanbuf[threadIdx.x - o] : 0;
+                uint32_t v = (threadIdx.x >= (unsigned)o) ? scanbuf[threadIdx.x - o] : 0;
                 __syncthreads();
                 scanbuf[threadIdx.x] += v;
                 __syncthreads();
@@ -697,7 +697,7 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_big(const uint32_t* rmeta,
     scanbuf[threadIdx.x] = local;
     __syncthreads();
     for (int o = 1; o < BLOCK; o <<= 1) {
-        uint32_t v = (threadIdx.x >= o) ? scanbuf[threadIdx.x - o] : 0;
+        uint32_t v = (threadIdx.x >= (unsigned)o) ? scanbuf[threadIdx.x - o] : 0;
         __syncthreads();
         scanbuf[threadIdx.x] += v;
         __syncthreads();
@@ -920,7 +920,7 @@ __global__ __launch_bounds__(1024) void k_rscan(const uint32_t* hist, int nblk,
     part[threadIdx.x] = s;
     __syncthreads();
     for (int o = 1; o < 1024; o <<= 1) {
-        uint32_t v = (threadIdx.x >= o) ? part[threadIdx.x - o] : 0;
+        uint32_t v = (threadIdx.x >= (unsigned)o) ? part[threadIdx.x - o] : 0;
         __syncthreads();
         part[threadIdx.x] += v;
         __syncthreads();
@@ -1061,7 +1061,7 @@ __global__ __launch_bounds__(SBLK) void k_rsort_small(uint64_t* a_keys,
             __syncthreads();
             if (allsame) continue; /* this digit sorts nothing: skip pass */
             for (int o = 1; o < SBLK; o <<= 1) {
-                uint32_t v = (threadIdx.x >= o) ? scanbuf[threadIdx.x - o] : 0;
+                uint32_t v = (threadIdx.x >= (unsigned)o) ? scanbuf[threadIdx.x - o] : 0;
                 __syncthreads();
                 scanbuf[threadIdx.x] += v;
                 __syncthreads();

@@ -242,15 +242,6 @@ struct dz_window_op {
         }                                                                  \
     } while (0)
 
-#define CHKV(op, call)                                                     \
-    do {                                                                   \
-        hipError_t e_ = (call);                                            \
-        if (e_ != hipSuccess) {                                            \
-            (op)->err = std::string(#call) + ": " + hipGetErrorString(e_); \
-            return;                                                        \
-        }                                                                  \
-    } while (0)
-
 static void emit_worker_main(dz_window_op* op);
 
 static hipEvent_t get_event(dz_window_op* op) {
