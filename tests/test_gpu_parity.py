@@ -609,3 +609,35 @@ def test_device_emission_filter_large_keyspace():
     assert np.array_equal(cat(outs, "avg"), exp["avg"][keep])
     op.close()
     o.close()
+
+
+def test_two_concurrent_ops_one_device():
+    # two independent pipelines on one GPU (each op owns streams + worker
+    # threads): no cross-talk, both bit-exact
+    rng = np.random.default_rng(94)
+    n = 60_000
+    ts = (1_000_000 + np.arange(n) // 60).astype(np.int64)
+    k = rng.integers(0, 333, n)
+    v = rng.uniform(0, 115, n)
+    opA = make_op(1000, n_keys_hint=400)
+    opB = make_op(500, 250, n_keys_hint=400)
+    oA = pyoracle.Oracle(1000, 0)
+    oB = pyoracle.Oracle(500, 250)
+    step = 10_000
+    for lo in range(0, n, step):
+        sl = slice(lo, lo + step)
+        opA.push(ts[sl], k[sl], v[sl])
+        opB.push(ts[sl], k[sl], v[sl])
+        oA.push(ts[sl], k[sl], v[sl])
+        oB.push(ts[sl], k[sl], v[sl])
+    outsA, outsB = [], []
+    opA.finish()
+    opB.finish()
+    outsA = opA.poll_all()
+    outsB = opB.poll_all()
+    oA.finish()
+    oB.finish()
+    assert_parity(outsA, oA.fetch())
+    assert_parity(outsB, oB.fetch())
+    for x in (opA, opB, oA, oB):
+        x.close()
