@@ -78,7 +78,7 @@ void launch_regroup_big(hipStream_t stream, const uint32_t* d_meta,
                         const uint32_t* d_ridx, const double* d_rvals,
                         const uint32_t* d_bucket_base, const FoldChunk& fc,
                         int32_t gtot, uint32_t* d_binoffs, uint32_t* d_binlens,
-                        uint32_t* d_wqcnt, double* d_gval, uint32_t* d_gridx);
+                        double* d_gval, uint32_t* d_gridx);
 
 void launch_fold3(hipStream_t stream, const double* d_gval, const uint32_t* d_gridx,
                   const uint32_t* d_bucket_base, const uint32_t* d_binoffs,

@@ -659,42 +659,42 @@ void launch_regroup(hipStream_t s, const uint32_t* d_meta, const uint32_t* d_rid
 __global__ __launch_bounds__(BLOCK) void k_regroup_big(const uint32_t* rmeta,
         const uint32_t* rridx, const double* rvals, const uint32_t* bucket_base,
         FoldChunk fc, int32_t gtot, uint32_t* binoffs, uint32_t* binlens,
-        uint32_t* wq_cnt /* [NB][WAVES][gtot] scratch */, double* gval,
-        uint32_t* gridx) {
+        double* gval, uint32_t* gridx) {
+    /* big-bin regime (GCAP < gtot <= ~37k, e.g. cfg3's 1M keys): bin
+     * counters/cursors live in DYNAMIC LDS (gtot*4 bytes), counted with LDS
+     * atomics and claimed through wave-serialized phases over interleaved
+     * 64-record tiles (global tile order == row order). The earlier global-
+     * scratch form was HBM-atomic-bound (~6.5 ms per 40M records). */
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    uint32_t* cnt = (uint32_t*)smem; /* becomes the cursor array in place */
     __shared__ uint32_t scanbuf[BLOCK];
     const int bkt = blockIdx.x;
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
     const uint32_t lo = bucket_base[bkt];
     const uint32_t hi = bucket_base[bkt + 1];
-    const int nk = fc.k_hi - fc.k_lo; /* == klocs here (single launch) */
-    uint32_t* mycnt = wq_cnt + ((int64_t)bkt * WAVES_PER_BLOCK) * gtot;
+    const int nk = fc.k_hi - fc.k_lo; /* == klocs (single launch) */
     uint32_t* offs = binoffs + (int64_t)bkt * gtot;
     uint32_t* lens = binlens + (int64_t)bkt * gtot;
-    /* wq_cnt is zeroed by the host (one hipMemsetAsync) before launch */
-    /* wave-quarter bounds over the WHOLE bucket (wave order == row order) */
-    const uint32_t nrec = hi - lo;
-    const uint32_t q = (nrec + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
-    const uint32_t w0 = lo + min(nrec, (uint32_t)wave * q);
-    const uint32_t w1 = lo + min(nrec, (uint32_t)(wave + 1) * q);
-    /* pass 1: per-(wave, bin) counts */
-    for (uint32_t i = w0 + lane; i < w1; i += 64) {
+    for (int g = threadIdx.x; g < gtot; g += BLOCK) cnt[g] = 0;
+    __syncthreads();
+    /* pass 1: bin counts (LDS atomics) */
+    for (uint32_t i = lo + threadIdx.x; i < hi; i += BLOCK) {
         const uint32_t ms_ = rmeta[i];
         const int widx = (int)((ms_ >> META_WIDX_SHIFT) & META_WIDX_MASK);
         const int kloc = (int)(ms_ & META_KLOC_MASK);
-        atomicAdd(&mycnt[(int64_t)wave * gtot + widx * nk + kloc], 1u);
+        atomicAdd(&cnt[widx * nk + kloc], 1u);
     }
     __syncthreads();
-    /* exclusive prefix over gtot bins (bin-major, then per-wave bases).
-     * two sweeps: per-thread local runs + block scan of 256 partials. */
+    /* exclusive prefix in place (publishing binoffs/binlens as we go) */
     const int per = (gtot + BLOCK - 1) / BLOCK;
     const int b0 = threadIdx.x * per;
     const int b1 = min(gtot, b0 + per);
-    uint32_t local = 0;
-    for (int b = b0; b < b1; b++)
-        for (int w = 0; w < WAVES_PER_BLOCK; w++)
-            local += mycnt[(int64_t)w * gtot + b];
-    scanbuf[threadIdx.x] = local;
+    {
+        uint32_t s = 0;
+        for (int b = b0; b < b1; b++) s += cnt[b];
+        scanbuf[threadIdx.x] = s;
+    }
     __syncthreads();
     for (int o = 1; o < BLOCK; o <<= 1) {
         uint32_t v = (threadIdx.x >= (unsigned)o) ? scanbuf[threadIdx.x - o] : 0;
@@ -702,26 +702,27 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_big(const uint32_t* rmeta,
         scanbuf[threadIdx.x] += v;
         __syncthreads();
     }
-    uint32_t run = threadIdx.x ? scanbuf[threadIdx.x - 1] : 0;
-    for (int b = b0; b < b1; b++) {
-        offs[b] = run;
-        uint32_t t = 0;
-        for (int w = 0; w < WAVES_PER_BLOCK; w++) {
-            uint32_t c = mycnt[(int64_t)w * gtot + b];
-            mycnt[(int64_t)w * gtot + b] = run + t; /* in-place: wave base */
-            t += c;
+    {
+        uint32_t run = threadIdx.x ? scanbuf[threadIdx.x - 1] : 0;
+        for (int b = b0; b < b1; b++) {
+            uint32_t t = cnt[b];
+            offs[b] = run;
+            lens[b] = t;
+            cnt[b] = run; /* cursor */
+            run += t;
         }
-        lens[b] = t;
-        run += t;
     }
     __syncthreads();
-    /* pass 2: placement with per-wave private cursors (plain RMW) */
-    for (uint32_t t0 = w0; t0 < w1; t0 += 64) {
+    /* pass 2: placement. Interleaved 64-record tiles (wave w owns tiles
+     * 4t+w) with claim phases serialized in wave order per tile group, so
+     * cursor claims follow global row order. */
+    for (uint32_t grp = lo; grp < hi; grp += BLOCK) {
+        const uint32_t t0 = grp + (uint32_t)wave * 64;
         const uint32_t i = t0 + lane;
+        const bool act = t0 < hi && i < hi;
         uint32_t g = 0;
         uint32_t ri = 0;
         double v = 0.0;
-        bool act = i < w1;
         if (act) {
             const uint32_t ms_ = rmeta[i];
             const int widx = (int)((ms_ >> META_WIDX_SHIFT) & META_WIDX_MASK);
@@ -730,7 +731,7 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_big(const uint32_t* rmeta,
             ri = rridx[i] | ((ms_ >> META_VALID_SHIFT) << 31);
             v = rvals[i];
         }
-        /* ranks via readlane loop (bins up to 64K: ballots too wide) */
+        /* intra-wave ranks via readlane loop (bins too wide for ballots) */
         uint32_t r = 0, wtot = 0;
         int fl = lane;
         const uint32_t gk = act ? g : 0xFFFFFFFFu;
@@ -743,21 +744,22 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_big(const uint32_t* rmeta,
             }
         }
         uint32_t pos = 0;
-        {
-            uint32_t pre = 0;
-            if (lane == fl && act) {
-                uint32_t* cp = &mycnt[(int64_t)wave * gtot + g];
-                pre = *cp;
-                *cp = pre + wtot;
+        for (int w = 0; w < WAVES_PER_BLOCK; w++) {
+            if (wave == w && act) {
+                uint32_t pre = 0;
+                if (lane == fl) {
+                    pre = cnt[g];
+                    cnt[g] = pre + wtot;
+                }
+                pre = (uint32_t)__shfl((int)pre, fl);
+                pos = pre + r;
             }
-            pre = (uint32_t)__shfl((int)pre, fl);
-            pos = pre + r;
+            __syncthreads();
         }
         if (act) {
             gval[lo + pos] = v;
             gridx[lo + pos] = ri;
         }
-        __builtin_amdgcn_wave_barrier();
     }
 }
 
@@ -765,10 +767,11 @@ void launch_regroup_big(hipStream_t s, const uint32_t* d_meta,
                         const uint32_t* d_ridx, const double* d_rvals,
                         const uint32_t* d_bucket_base, const FoldChunk& fc,
                         int32_t gtot, uint32_t* d_binoffs, uint32_t* d_binlens,
-                        uint32_t* d_wqcnt, double* d_gval, uint32_t* d_gridx) {
-    hipLaunchKernelGGL(k_regroup_big, dim3(NB), dim3(BLOCK), 0, s, d_meta,
+                        double* d_gval, uint32_t* d_gridx) {
+    size_t lds = (size_t)gtot * 4;
+    hipLaunchKernelGGL(k_regroup_big, dim3(NB), dim3(BLOCK), lds, s, d_meta,
                        d_ridx, d_rvals, d_bucket_base, fc, gtot, d_binoffs,
-                       d_binlens, d_wqcnt, d_gval, d_gridx);
+                       d_binlens, d_gval, d_gridx);
 }
 
 /* ------------------------------------------------------------------ */

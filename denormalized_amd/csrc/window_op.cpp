@@ -153,8 +153,6 @@ struct dz_window_op {
     uint32_t* d_binoffs = nullptr;
     uint32_t* d_binlens = nullptr;
     int64_t bin_cap = 0;   /* bins per bucket the binoffs/lens arrays hold */
-    uint32_t* d_wqcnt = nullptr; /* big-bin regroup scratch [NB][4][gtot] */
-    int64_t wq_cap = 0;
     int64_t rec_cap = 0;
     uint64_t* d_scalars = nullptr;
     uint64_t* h_scalars = nullptr; /* pinned, 3 */
@@ -425,7 +423,7 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_ghist); hipFree(op->d_gofs); hipFree(op->d_total); hipFree(op->d_base);
     hipFree(op->d_meta); hipFree(op->d_ridx); hipFree(op->d_rvals);
     hipFree(op->d_gval); hipFree(op->d_gridx);
-    hipFree(op->d_binoffs); hipFree(op->d_binlens); hipFree(op->d_wqcnt);
+    hipFree(op->d_binoffs); hipFree(op->d_binlens);
     hipFree(op->d_scalars); hipFree(op->d_slotmap); hipFree(op->d_zero_kid);
     hipFree(op->d_resetlist);
     hipFree(op->d_ts); hipFree(op->d_kid); hipFree(op->d_vals); hipFree(op->d_valbm);
@@ -1162,18 +1160,11 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
     });
     int64_t klocs = op->kcap >> dz::LOG_NB;
     int64_t gtot = klocs * nw;
-    constexpr int64_t BIG_GTOT_MAX = 65536;
+    constexpr int64_t BIG_GTOT_MAX = 37888; /* bins fit dynamic LDS (x4 B) */
     if (gtot > dz::FOLD_GCAP && gtot <= BIG_GTOT_MAX) {
         /* single-launch big-bin regime: bins in global scratch, no chunked
          * re-reads (cfg3-class keyspaces) */
         if (ensure_bins(op, gtot) != DZ_OK) return DZ_ERR;
-        if ((int64_t)4 * gtot > op->wq_cap) {
-            hipFree(op->d_wqcnt);
-            CHK(op, hipMalloc(&op->d_wqcnt, (size_t)dz::NB * 4 * gtot * 4));
-            op->wq_cap = 4 * gtot;
-        }
-        CHK(op, hipMemsetAsync(op->d_wqcnt, 0, (size_t)dz::NB * 4 * gtot * 4,
-                               op->stream));
         dz::FoldChunk fc;
         fc.w_lo = 0;
         fc.w_hi = (int32_t)nw;
@@ -1185,7 +1176,7 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
         timed(op, "regroup", (double)nrec_max * 32, [&] {
             dz::launch_regroup_big(op->stream, op->d_meta, op->d_ridx,
                                    op->d_rvals, op->d_base, fc, (int32_t)gtot,
-                                   op->d_binoffs, op->d_binlens, op->d_wqcnt,
+                                   op->d_binoffs, op->d_binlens,
                                    op->d_gval, op->d_gridx);
         });
         timed(op, "fold", (double)nrec_max * 12, [&] {
