@@ -101,13 +101,14 @@ void launch_emission_slabread(hipStream_t stream, const uint64_t* slab_first,
                               const uint64_t* slab_cnt, const double* slab_min,
                               const double* slab_max, const double* slab_sum,
                               int64_t K, uint64_t* ekeys, uint32_t* ekid,
-                              uint32_t* skid, uint32_t* counter,
+                              uint64_t* fkeys, uint32_t* fkid, uint32_t* fiota,
+                              uint32_t* counter, uint32_t* counter2,
                               const EmitFilter& ef, uint64_t* ocnt, double* omin,
                               double* omax, double* osum, double* oavg,
                               uint8_t* oflags);
-void launch_emission_sort(hipStream_t stream, int64_t K, uint64_t* ekeys,
-                          uint64_t* skeys, uint32_t* skid, uint32_t* okid,
-                          uint32_t* counter, uint32_t* rhist, uint32_t* roffs);
+void launch_emission_sort(hipStream_t stream, int64_t K, uint64_t* fkeys,
+                          uint64_t* skeys, uint32_t* fiota, uint32_t* okid,
+                          uint32_t* counter2, uint32_t* rhist, uint32_t* roffs);
 constexpr int EMIT_RCHUNK = 4096;
 constexpr int EMIT_RBINS = 2048;
 

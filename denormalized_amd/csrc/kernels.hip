@@ -1130,34 +1130,35 @@ __global__ __launch_bounds__(SBLK) void k_rsort_small(uint64_t* a_keys,
     }
 }
 
-/* gather the sorted groups' aggregates + filter flag into output columns */
-__global__ void k_egather(const uint32_t* ckid, const uint32_t* counter,
-                          const uint64_t* s_cnt, const double* s_min,
-                          const double* s_max, const double* s_sum,
-                          EmitFilter ef, uint64_t* ocnt,
-                          double* omin, double* omax, double* osum,
-                          double* oavg, uint8_t* oflags) {
+
+/* filter-compact: evaluate the pushed-down predicate on each touched group
+ * (straight from the slot slab) and pack the passers:
+ * {first, kid, iota} triples for the sort + gather stages. */
+__global__ void k_efilter(const uint64_t* ekeys, const uint32_t* ekid,
+                          const uint32_t* counter, const uint64_t* s_cnt,
+                          const double* s_min, const double* s_max,
+                          const double* s_sum, EmitFilter ef, uint64_t* fkeys,
+                          uint32_t* fkid, uint32_t* fiota, uint32_t* counter2) {
     const uint32_t nt = *counter;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nt;
          i += stride) {
-        const uint32_t kid = ckid[i];
-        const uint64_t c = s_cnt[kid];
-        const bool valid = c > 0;
-        const double mn = valid ? s_min[kid] : 0.0;
-        const double mx = valid ? s_max[kid] : 0.0;
-        const double sm = valid ? s_sum[kid] : 0.0;
-        const double av = valid ? sm / (double)c : 0.0;
+        const uint32_t kid = ekid[i];
         bool pass = true;
         if (ef.on) {
+            const uint64_t c = s_cnt[kid];
+            const bool valid = c > 0;
             double v = 0.0;
             bool fv = true;
             switch (ef.field) {
                 case 0: v = (double)c; break;
-                case 1: v = mn; fv = valid; break;
-                case 2: v = mx; fv = valid; break;
-                case 3: v = sm; fv = valid; break;
-                default: v = av; fv = valid; break;
+                case 1: v = valid ? s_min[kid] : 0.0; fv = valid; break;
+                case 2: v = valid ? s_max[kid] : 0.0; fv = valid; break;
+                case 3: v = valid ? s_sum[kid] : 0.0; fv = valid; break;
+                default:
+                    v = valid ? s_sum[kid] / (double)c : 0.0;
+                    fv = valid;
+                    break;
             }
             if (!fv) {
                 pass = false; /* NULL never passes a comparison filter */
@@ -1172,12 +1173,34 @@ __global__ void k_egather(const uint32_t* ckid, const uint32_t* counter,
                 }
             }
         }
+        if (pass) {
+            uint32_t p = atomicAdd(counter2, 1u);
+            fkeys[p] = ekeys[i];
+            fkid[p] = kid;
+            fiota[p] = p;
+        }
+    }
+}
+
+/* gather the passing groups' aggregate columns (by packed order) */
+__global__ void k_egather(const uint32_t* fkid, const uint32_t* counter2,
+                          const uint64_t* s_cnt, const double* s_min,
+                          const double* s_max, const double* s_sum,
+                          uint64_t* ocnt, double* omin, double* omax,
+                          double* osum, double* oavg, uint8_t* oflags) {
+    const uint32_t nt = *counter2;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nt;
+         i += stride) {
+        const uint32_t kid = fkid[i];
+        const uint64_t c = s_cnt[kid];
+        const bool valid = c > 0;
         ocnt[i] = c;
-        omin[i] = mn;
-        omax[i] = mx;
-        osum[i] = sm;
-        oavg[i] = av;
-        oflags[i] = (uint8_t)((valid ? 1 : 0) | (pass ? 2 : 0));
+        omin[i] = valid ? s_min[kid] : 0.0;
+        omax[i] = valid ? s_max[kid] : 0.0;
+        osum[i] = valid ? s_sum[kid] : 0.0;
+        oavg[i] = valid ? s_sum[kid] / (double)c : 0.0;
+        oflags[i] = (uint8_t)((valid ? 1 : 0) | 2);
     }
 }
 
@@ -1185,23 +1208,32 @@ void launch_emission_slabread(hipStream_t s, const uint64_t* slab_first,
                               const uint64_t* slab_cnt, const double* slab_min,
                               const double* slab_max, const double* slab_sum,
                               int64_t K, uint64_t* ekeys, uint32_t* ekid,
-                              uint32_t* skid, uint32_t* counter,
+                              uint64_t* fkeys, uint32_t* fkid, uint32_t* fiota,
+                              uint32_t* counter, uint32_t* counter2,
                               const EmitFilter& ef, uint64_t* ocnt, double* omin,
                               double* omax, double* osum, double* oavg,
                               uint8_t* oflags) {
     int cblocks = (int)std::min<int64_t>((K + BLOCK - 1) / BLOCK, 2048);
-    /* compact (reads slab.first) -> gather by compact order (reads the other
-     * slab fields). After these two launches the window slot is reusable. */
+    /* compact touched groups, filter-compact the passers, gather their
+     * columns — all reads of the window slot's slab happen HERE, so the
+     * slot is reusable right after these launches. */
     hipLaunchKernelGGL(k_ecompact, dim3(cblocks), dim3(BLOCK), 0, s, slab_first,
-                       K, ekeys, ekid, skid, counter);
-    hipLaunchKernelGGL(k_egather, dim3(cblocks), dim3(BLOCK), 0, s, ekid, counter,
-                       slab_cnt, slab_min, slab_max, slab_sum, ef, ocnt, omin,
-                       omax, osum, oavg, oflags);
+                       K, ekeys, ekid, fiota /*scratch, rewritten below*/,
+                       counter);
+    hipLaunchKernelGGL(k_efilter, dim3(cblocks), dim3(BLOCK), 0, s, ekeys, ekid,
+                       counter, slab_cnt, slab_min, slab_max, slab_sum, ef,
+                       fkeys, fkid, fiota, counter2);
+    hipLaunchKernelGGL(k_egather, dim3(cblocks), dim3(BLOCK), 0, s, fkid,
+                       counter2, slab_cnt, slab_min, slab_max, slab_sum, ocnt,
+                       omin, omax, osum, oavg, oflags);
 }
 
-void launch_emission_sort(hipStream_t s, int64_t K, uint64_t* ekeys,
-                          uint64_t* skeys, uint32_t* skid, uint32_t* okid,
-                          uint32_t* counter, uint32_t* rhist, uint32_t* roffs) {
+void launch_emission_sort(hipStream_t s, int64_t K, uint64_t* fkeys,
+                          uint64_t* skeys, uint32_t* fiota, uint32_t* okid,
+                          uint32_t* counter2, uint32_t* rhist, uint32_t* roffs) {
+    uint64_t* ekeys = fkeys;
+    uint32_t* skid = fiota;
+    uint32_t* counter = counter2;
     /* sort (first, compact-index) pairs: keys ekeys<->skeys, payload
      * skid<->okid; both paths leave the sorted payload in skid */
     if (K <= 65536) {

@@ -180,10 +180,12 @@ struct dz_window_op {
     static constexpr int E_WORKERS = 4;
     struct DevEmit { /* per-slot device scratch, carved from one alloc */
         char* base = nullptr;
-        uint64_t* ekeys; uint32_t* ekid;
-        uint64_t* skeys; uint32_t* skid;
-        uint32_t* counter;
-        uint32_t* okid; uint64_t* ocnt;
+        uint64_t* ekeys; uint32_t* ekid;      /* touched groups (compact) */
+        uint64_t* fkeys; uint32_t* fkid;      /* filter passers (packed)  */
+        uint32_t* fiota;                      /* sort payload / sidx out  */
+        uint64_t* skeys; uint32_t* okid;      /* sort ping-pong buffers   */
+        uint32_t* counter;                    /* [0]=touched, [1]=passers */
+        uint64_t* ocnt;
         double* omin; double* omax; double* osum; double* oavg;
         uint8_t* oflags;
     };
@@ -738,12 +740,38 @@ static void emit_worker_main(dz_window_op* op) {
             job = op->e_jobs.front();
             op->e_jobs.pop_front();
         }
-        event_spin(job.ev); /* emission D2H complete */
+        event_spin(job.ev); /* emission kernels (+ host-path D2H) complete */
         auto t0 = std::chrono::steady_clock::now();
         OutBuf ob;
         if (job.device) {
-            build_emission(op, job.wstart, job.wend, op->e_pcnt[job.slab],
-                           job.kcap, op->e_slabs[job.slab], &ob);
+            /* copy exactly nt2 packed rows (the filter already ran on
+             * device, so this is the final output volume, not the keyspace) */
+            const uint32_t nt = op->e_pcnt[job.slab];
+            if (nt > 0) {
+                dz_window_op::DevEmit& d = op->e_dev[job.slab];
+                char* s = (char*)op->e_slabs[job.slab];
+                int64_t kc = job.kcap;
+                hipMemcpyAsync(s, d.fkid, (size_t)nt * 4, hipMemcpyDeviceToHost,
+                               op->copy_stream);
+                hipMemcpyAsync(s + kc * 4, d.fiota, (size_t)nt * 4,
+                               hipMemcpyDeviceToHost, op->copy_stream);
+                hipMemcpyAsync(s + kc * 8, d.oflags, nt, hipMemcpyDeviceToHost,
+                               op->copy_stream);
+                hipMemcpyAsync(s + kc * 9, d.ocnt, (size_t)nt * 8,
+                               hipMemcpyDeviceToHost, op->copy_stream);
+                hipMemcpyAsync(s + kc * 17, d.omin, (size_t)nt * 8,
+                               hipMemcpyDeviceToHost, op->copy_stream);
+                hipMemcpyAsync(s + kc * 25, d.omax, (size_t)nt * 8,
+                               hipMemcpyDeviceToHost, op->copy_stream);
+                hipMemcpyAsync(s + kc * 33, d.osum, (size_t)nt * 8,
+                               hipMemcpyDeviceToHost, op->copy_stream);
+                hipMemcpyAsync(s + kc * 41, d.oavg, (size_t)nt * 8,
+                               hipMemcpyDeviceToHost, op->copy_stream);
+                hipEventRecord(job.ev, op->copy_stream);
+                event_spin(job.ev);
+            }
+            build_emission(op, job.wstart, job.wend, nt, job.kcap,
+                           op->e_slabs[job.slab], &ob);
         } else {
             build_emission_host(op, job.wstart, job.wend, job.n_keys, job.kcap,
                                 op->e_slabs[job.slab], &ob);
@@ -800,11 +828,12 @@ static dz_status trigger_windows(dz_window_op* op) {
         for (int i = 0; i < dz_window_op::E_POOL; i++) {
             auto& d = op->e_dev[i];
             hipFree(d.base);
-            /* ekeys 8 + ekid 4 + skeys 8 + skid 4 + counter + out columns
-             * (kid 4 + cnt 8 + 4x f64 + flags 1) = ~69 B/entry */
-            CHK(op, hipMalloc(&d.base, (size_t)kc * 70 + 64));
+            /* 3x u64 keys + 5x u64/f64 columns + 4x u32 + flags + counters
+             * = ~82 B/entry */
+            CHK(op, hipMalloc(&d.base, (size_t)kc * 84 + 64));
             char* p = d.base;
             d.ekeys = (uint64_t*)p; p += kc * 8;
+            d.fkeys = (uint64_t*)p; p += kc * 8;
             d.skeys = (uint64_t*)p; p += kc * 8;
             d.ocnt = (uint64_t*)p; p += kc * 8;
             d.omin = (double*)p; p += kc * 8;
@@ -812,7 +841,8 @@ static dz_status trigger_windows(dz_window_op* op) {
             d.osum = (double*)p; p += kc * 8;
             d.oavg = (double*)p; p += kc * 8;
             d.ekid = (uint32_t*)p; p += kc * 4;
-            d.skid = (uint32_t*)p; p += kc * 4;
+            d.fkid = (uint32_t*)p; p += kc * 4;
+            d.fiota = (uint32_t*)p; p += kc * 4;
             d.okid = (uint32_t*)p; p += kc * 4;
             d.oflags = (uint8_t*)p; p += kc;
             d.counter = (uint32_t*)p;
@@ -904,7 +934,7 @@ static dz_status trigger_windows(dz_window_op* op) {
             CHK(op, hipMemcpyAsync(op->e_slabs[slab], sl, stride * 8,
                                    hipMemcpyDeviceToHost, op->copy_stream));
         } else {
-        CHK(op, hipMemsetAsync(d.counter, 0, 4, op->copy_stream));
+        CHK(op, hipMemsetAsync(d.counter, 0, 8, op->copy_stream));
         if (op->n_keys > 0) {
             dz::launch_emission_slabread(op->copy_stream,
                                          /*first*/ sl + op->kcap,
@@ -912,9 +942,10 @@ static dz_status trigger_windows(dz_window_op* op) {
                                          /*min*/ (const double*)(sl + 2 * op->kcap),
                                          /*max*/ (const double*)(sl + 3 * op->kcap),
                                          /*sum*/ (const double*)(sl + 4 * op->kcap),
-                                         op->n_keys, d.ekeys, d.ekid, d.skid,
-                                         d.counter, ef, d.ocnt, d.omin, d.omax,
-                                         d.osum, d.oavg, d.oflags);
+                                         op->n_keys, d.ekeys, d.ekid, d.fkeys,
+                                         d.fkid, d.fiota, d.counter,
+                                         d.counter + 1, ef, d.ocnt, d.omin,
+                                         d.omax, d.osum, d.oavg, d.oflags);
         }
         }
         CHK(op, hipEventRecord(slot_ev, op->copy_stream));
@@ -929,31 +960,12 @@ static dz_status trigger_windows(dz_window_op* op) {
         dz_window_op::DevEmit& d = op->e_dev[slab];
         const bool dev_path = op->n_keys > 65536;
         if (dev_path && op->n_keys > 0) {
-            dz::launch_emission_sort(op->copy_stream, op->n_keys, d.ekeys,
-                                     d.skeys, d.skid, d.okid, d.counter,
+            dz::launch_emission_sort(op->copy_stream, op->n_keys, d.fkeys,
+                                     d.skeys, d.fiota, d.okid, d.counter + 1,
                                      op->d_rhist, op->d_roffs);
-            char* s = (char*)op->e_slabs[slab];
-            size_t kc = (size_t)op->kcap;
-            size_t K = (size_t)op->n_keys;
-            CHK(op, hipMemcpyAsync(s, d.ekid, K * 4, hipMemcpyDeviceToHost,
-                                   op->copy_stream));
-            CHK(op, hipMemcpyAsync(s + kc * 4, d.skid, K * 4,
-                                   hipMemcpyDeviceToHost, op->copy_stream));
-            CHK(op, hipMemcpyAsync(s + kc * 8, d.oflags, K,
-                                   hipMemcpyDeviceToHost, op->copy_stream));
-            CHK(op, hipMemcpyAsync(s + kc * 9, d.ocnt, K * 8,
-                                   hipMemcpyDeviceToHost, op->copy_stream));
-            CHK(op, hipMemcpyAsync(s + kc * 17, d.omin, K * 8,
-                                   hipMemcpyDeviceToHost, op->copy_stream));
-            CHK(op, hipMemcpyAsync(s + kc * 25, d.omax, K * 8,
-                                   hipMemcpyDeviceToHost, op->copy_stream));
-            CHK(op, hipMemcpyAsync(s + kc * 33, d.osum, K * 8,
-                                   hipMemcpyDeviceToHost, op->copy_stream));
-            CHK(op, hipMemcpyAsync(s + kc * 41, d.oavg, K * 8,
-                                   hipMemcpyDeviceToHost, op->copy_stream));
         }
         if (dev_path)
-            CHK(op, hipMemcpyAsync(&op->e_pcnt[slab], d.counter, 4,
+            CHK(op, hipMemcpyAsync(&op->e_pcnt[slab], d.counter + 1, 4,
                                    hipMemcpyDeviceToHost, op->copy_stream));
         CHK(op, hipEventRecord(ev, op->copy_stream));
         {
