@@ -237,10 +237,14 @@ def main():
             "dtype": "f64",
             "data": "synthetic",
             "config": {
-                "workload": f"cfg2: {K * B / 1e6:.0f}M rows/GPU, {args.keys} keys, "
-                            f"{args.window_ms}ms tumbling"
-                            + (f"/{args.slide_ms}ms slide" if args.slide_ms else "")
-                            + ", count/min/max/avg + filter(max>113)",
+                "workload": ("cfg2: " if (args.keys == 10_000 and not args.slide_ms
+                                           and args.window_ms == 1000) else "")
+                            + f"{K * B / 1e6:.0f}M rows/GPU, {args.keys} keys, "
+                            + f"{args.window_ms}ms "
+                            + (f"sliding/{args.slide_ms}ms hop" if args.slide_ms
+                               else "tumbling")
+                            + ", count/min/max/avg"
+                            + ("" if args.no_filter else " + filter(max>113)"),
                 "rows_per_step": B,
                 "keys_per_gpu": args.keys,
                 "rows_per_ms": args.rows_per_ms,

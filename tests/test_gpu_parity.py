@@ -641,3 +641,20 @@ def test_two_concurrent_ops_one_device():
     assert_parity(outsB, oB.fetch())
     for x in (opA, opB, oA, oB):
         x.close()
+
+
+def test_fully_shuffled_timestamps():
+    # worst-case disorder: every batch's timestamps fully shuffled => frames
+    # are created/re-created and re-emitted continuously
+    rng = np.random.default_rng(95)
+    batches = []
+    for b in range(3):
+        n = 8000
+        ts = rng.integers(1_000_000, 1_020_000, n).astype(np.int64)
+        k = rng.integers(0, 50, n)
+        v = rng.uniform(0, 115, n)
+        batches.append((ts, k, v))
+    outs, exp = run_both(1000, 0, batches)
+    assert_parity(outs, exp)
+    outs, exp = run_both(2000, 500, batches)
+    assert_parity(outs, exp)
