@@ -658,3 +658,73 @@ def test_fully_shuffled_timestamps():
     assert_parity(outs, exp)
     outs, exp = run_both(2000, 500, batches)
     assert_parity(outs, exp)
+
+
+def test_state_growth_with_open_windows():
+    # key capacity grows while windows are OPEN: state_alloc must migrate
+    # open-slot state to the bigger slabs without losing accumulators
+    rng = np.random.default_rng(96)
+    op = make_op(10_000, n_keys_hint=8)  # 10s window stays open across pushes
+    o = pyoracle.Oracle(10_000, 0)
+    t0 = 1_000_000
+    for step, nkeys in enumerate([10, 200, 5000, 60_000]):
+        n = 50_000
+        ts = (t0 + step * 1000 + np.arange(n) // 100).astype(np.int64)
+        k = rng.integers(0, nkeys, n)
+        v = rng.uniform(0, 115, n)
+        op.push(ts, k, v)
+        o.push(ts, k, v)
+        assert op.open_windows == o.open_frames == 1
+    op.finish()
+    o.finish()
+    assert_parity(op.poll_all(), o.fetch())
+    op.close()
+    o.close()
+
+
+def test_emission_path_switch_midstream():
+    # the dictionary crosses the 64k-key device-emission threshold between
+    # window closes: host-built and device-built batches interleave
+    rng = np.random.default_rng(97)
+    op = make_op(1000, n_keys_hint=64)
+    o = pyoracle.Oracle(1000, 0)
+    t0 = 1_000_000
+    outs = []
+    for step, nkeys in enumerate([1000, 30_000, 90_000, 120_000]):
+        n = 150_000
+        ts = (t0 + step * 1500 + np.arange(n) // 150).astype(np.int64)
+        k = rng.integers(0, nkeys, n)
+        v = rng.uniform(0, 115, n)
+        op.push(ts, k, v)
+        o.push(ts, k, v)
+        outs += op.poll_all()
+    op.finish()
+    o.finish()
+    outs += op.poll_all()
+    assert_parity(outs, o.fetch())
+    op.close()
+    o.close()
+
+
+def test_soak_many_small_pushes():
+    # slab/event/slot recycling over many pushes (threaded emission soak)
+    rng = np.random.default_rng(98)
+    op = make_op(500, n_keys_hint=128)
+    o = pyoracle.Oracle(500, 0)
+    t = 1_000_000
+    outs = []
+    for _ in range(200):
+        n = int(rng.integers(100, 3000))
+        ts = (t + np.cumsum(rng.integers(0, 3, n))).astype(np.int64)
+        t = int(ts.max())
+        k = rng.integers(0, 128, n)
+        v = rng.uniform(0, 115, n)
+        op.push(ts, k, v)
+        o.push(ts, k, v)
+        outs += op.poll_all()
+    op.finish()
+    o.finish()
+    outs += op.poll_all()
+    assert_parity(outs, o.fetch())
+    op.close()
+    o.close()
