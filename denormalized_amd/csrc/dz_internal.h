@@ -36,7 +36,8 @@ struct FoldChunk {
     int32_t k_lo, k_hi;  /* kloc range [k_lo, k_hi) */
     int64_t kcap;        /* state key capacity (multiple of NB) */
     uint32_t batch_seq;
-    int32_t bin_stride;  /* binoffs/binlens row stride (GCAP, or gtot in the
+    int32_t bin_stride;  /* binoffs/binlens row stride per bucket */
+    int32_t tl_nw;       /* two-level mode: windows per batch (0 = off) */  /* binoffs/binlens row stride (GCAP, or gtot in the
                           * single-launch big-bin regime) */
 };
 
@@ -74,11 +75,18 @@ void launch_regroup(hipStream_t stream, const uint32_t* d_meta,
                     uint32_t* d_binoffs, uint32_t* d_binlens, double* d_gval,
                     uint32_t* d_gridx);
 
-void launch_regroup_big(hipStream_t stream, const uint32_t* d_meta,
-                        const uint32_t* d_ridx, const double* d_rvals,
-                        const uint32_t* d_bucket_base, const FoldChunk& fc,
-                        int32_t gtot, uint32_t* d_binoffs, uint32_t* d_binlens,
-                        double* d_gval, uint32_t* d_gridx);
+void launch_regroup_l1(hipStream_t stream, const uint32_t* d_meta,
+                       const uint32_t* d_ridx, const double* d_rvals,
+                       const uint32_t* d_bucket_base, const FoldChunk& fc,
+                       uint32_t* d_b1offs, uint32_t* d_b1lens, uint32_t* d_meta2,
+                       uint32_t* d_ridx2, double* d_rvals2);
+
+void launch_regroup_l2(hipStream_t stream, const uint32_t* d_meta2,
+                       const uint32_t* d_ridx2, const double* d_rvals2,
+                       const uint32_t* d_bucket_base, const FoldChunk& fc,
+                       int nb1, const uint32_t* d_b1offs, const uint32_t* d_b1lens,
+                       uint32_t* d_binoffs, uint32_t* d_binlens, double* d_gval,
+                       uint32_t* d_gridx);
 
 void launch_fold3(hipStream_t stream, const double* d_gval, const uint32_t* d_gridx,
                   const uint32_t* d_bucket_base, const uint32_t* d_binoffs,

@@ -487,20 +487,31 @@ void launch_scatter(hipStream_t s, const int32_t* d_kid, const int64_t* d_ts,
 
 constexpr int GCAP = FOLD_GCAP; /* bins (groups) per bucket per chunk */
 
-__global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
+/* One templated stable-split kernel, three modes:
+ *  DIRECT (gtot <= GCAP): bins = (widx,kloc) groups, output 12 B records
+ *  L1 (two-level, first pass): bins = (kloc>>8, widx) <= 256, output FULL
+ *     16 B records (meta travels to L2)
+ *  L2 (two-level, second pass): one block per (bucket, bin1) segment of the
+ *     L1 output; bins = kloc & 255, output 12 B records + final group
+ *     segment offsets for the fold
+ * Structure (shared): whole-segment bin counts (LDS atomics), prefix, then
+ * per-supertile ranked placement (wave-quarters: wave order == row order,
+ * private cursors, bit-ballot same-bin masks) into LDS staging flushed
+ * bin-major so writes coalesce. */
+enum { RG_DIRECT = 0, RG_L1 = 1, RG_L2 = 2 };
+
+template <int MODE>
+__global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
         const uint32_t* rridx, const double* rvals, const uint32_t* bucket_base,
-        FoldChunk fc, uint32_t* binoffs, uint32_t* binlens, double* gval,
-        uint32_t* gridx) {
-    /* ONE BLOCK per bucket, 4 waves cooperating:
-     *  pass 1: whole-bucket bin counts (LDS atomics) -> binoffs/binlens
-     *  per supertile: per-supertile counts + prefix, ranked placement into
-     *  LDS staging (bit-ballot same-bin masks, wave-serialized cursors keep
-     *  row order), then a bucket-major flush so writes coalesce. */
-    __shared__ uint32_t cnt[GCAP];    /* whole-bucket bin counts */
-    __shared__ uint32_t gcur[GCAP];   /* global (bucket-region) bin cursors */
+        FoldChunk fc, const uint32_t* b1offs, const uint32_t* b1lens,
+        uint32_t* binoffs, uint32_t* binlens, uint32_t* ometa, uint32_t* oridx,
+        double* oval) {
+    __shared__ uint32_t cnt[GCAP];    /* whole-segment bin counts */
+    __shared__ uint32_t gcur[GCAP];   /* segment-region bin cursors */
     __shared__ uint32_t stcnt4[WAVES_PER_BLOCK][GCAP]; /* per-wave-quarter */
     __shared__ uint32_t stoffs[GCAP]; /* per-supertile bin prefix */
     __shared__ uint32_t wcur[WAVES_PER_BLOCK][GCAP];   /* per-wave cursors */
+    __shared__ uint32_t s_meta[ST_RECORDS]; /* RG_L1 only */
     __shared__ uint32_t s_ridx[ST_RECORDS];
     __shared__ uint32_t s_dest[ST_RECORDS];
     __shared__ double s_val[ST_RECORDS];
@@ -509,19 +520,45 @@ __global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
     const int bkt = blockIdx.x;
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
-    const uint32_t lo = bucket_base[bkt];
-    const uint32_t hi = bucket_base[bkt + 1];
+    const uint32_t bklo = bucket_base[bkt];
+    uint32_t lo = bklo, hi = bucket_base[bkt + 1];
     const int nk = fc.k_hi - fc.k_lo;
-    const int glocal_n = (fc.w_hi - fc.w_lo) * nk;
+    int nbins, obase;
+    if (MODE == RG_DIRECT) {
+        nbins = (fc.w_hi - fc.w_lo) * nk;
+        obase = bkt * fc.bin_stride;
+    } else if (MODE == RG_L1) {
+        nbins = ((nk + 255) >> 8) * fc.tl_nw;
+        obase = bkt * 256;
+    } else { /* RG_L2 */
+        const int bin1 = blockIdx.y;
+        nbins = 256;
+        lo = bklo + b1offs[bkt * 256 + bin1];
+        hi = lo + b1lens[bkt * 256 + bin1];
+        obase = bkt * fc.bin_stride + bin1 * 256;
+    }
     for (int g = threadIdx.x; g < GCAP; g += BLOCK) cnt[g] = 0;
     __syncthreads();
-    /* pass 1: whole-bucket bin counts */
-    for (uint32_t i = lo + threadIdx.x; i < hi; i += BLOCK) {
-        const uint32_t ms_ = rmeta[i];
+
+    auto bin_of = [&](uint32_t ms_) -> uint32_t {
         const int widx = (int)((ms_ >> META_WIDX_SHIFT) & META_WIDX_MASK);
         const int kloc = (int)(ms_ & META_KLOC_MASK);
-        if (widx >= fc.w_lo && widx < fc.w_hi && kloc >= fc.k_lo && kloc < fc.k_hi)
-            atomicAdd(&cnt[(widx - fc.w_lo) * nk + (kloc - fc.k_lo)], 1u);
+        if (MODE == RG_DIRECT) {
+            if (widx < fc.w_lo || widx >= fc.w_hi || kloc < fc.k_lo ||
+                kloc >= fc.k_hi)
+                return 0x1FFu; /* outside this chunk */
+            return (uint32_t)((widx - fc.w_lo) * nk + (kloc - fc.k_lo));
+        } else if (MODE == RG_L1) {
+            return (uint32_t)((kloc >> 8) * fc.tl_nw + widx);
+        } else {
+            return (uint32_t)(kloc & 255);
+        }
+    };
+
+    /* pass 1: whole-segment bin counts */
+    for (uint32_t i = lo + threadIdx.x; i < hi; i += BLOCK) {
+        const uint32_t g = bin_of(rmeta[i]);
+        if (g != 0x1FFu) atomicAdd(&cnt[g], 1u);
     }
     __syncthreads();
     if (threadIdx.x == 0) { /* tiny exclusive prefix over <=GCAP bins */
@@ -533,9 +570,12 @@ __global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
         }
     }
     __syncthreads();
-    for (int g = threadIdx.x; g < glocal_n; g += BLOCK) {
-        binoffs[(int64_t)bkt * GCAP + g] = gcur[g];
-        binlens[(int64_t)bkt * GCAP + g] = cnt[g];
+    /* publish segment layout for the next stage / fold (offsets are
+     * relative to the BUCKET region start) */
+    const uint32_t relbase = lo - bklo;
+    for (int g = threadIdx.x; g < nbins; g += BLOCK) {
+        binoffs[obase + g] = relbase + gcur[g];
+        binlens[obase + g] = cnt[g];
     }
     if (lo == hi) return;
 
@@ -545,17 +585,12 @@ __global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
         const uint32_t q = ((st1 - st0) + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
         const uint32_t w0 = min(st1, st0 + (uint32_t)wave * q);
         const uint32_t w1 = min(st1, w0 + q);
-        /* per-(wave, bin) counts for this supertile */
         for (int g = threadIdx.x; g < GCAP; g += BLOCK)
             for (int w = 0; w < WAVES_PER_BLOCK; w++) stcnt4[w][g] = 0;
         __syncthreads();
         for (uint32_t i = w0 + lane; i < w1; i += 64) {
-            const uint32_t ms_ = rmeta[i];
-            const int widx = (int)((ms_ >> META_WIDX_SHIFT) & META_WIDX_MASK);
-            const int kloc = (int)(ms_ & META_KLOC_MASK);
-            if (widx >= fc.w_lo && widx < fc.w_hi && kloc >= fc.k_lo &&
-                kloc < fc.k_hi)
-                atomicAdd(&stcnt4[wave][(widx - fc.w_lo) * nk + (kloc - fc.k_lo)], 1u);
+            const uint32_t g = bin_of(rmeta[i]);
+            if (g != 0x1FFu) atomicAdd(&stcnt4[wave][g], 1u);
         }
         __syncthreads();
         if (threadIdx.x == 0) { /* bin prefix over <=GCAP bins */
@@ -579,16 +614,17 @@ __global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
         for (uint32_t t0 = w0; t0 < w1; t0 += 64) {
             const uint32_t i = t0 + lane;
             uint32_t g = 0x1FFu; /* sentinel above GCAP-1 */
-            uint32_t ri = 0;
+            uint32_t ms = 0, ri = 0;
             double v = 0.0;
             if (i < w1) {
-                const uint32_t ms_ = rmeta[i];
-                const int widx = (int)((ms_ >> META_WIDX_SHIFT) & META_WIDX_MASK);
-                const int kloc = (int)(ms_ & META_KLOC_MASK);
-                if (widx >= fc.w_lo && widx < fc.w_hi && kloc >= fc.k_lo &&
-                    kloc < fc.k_hi) {
-                    g = (uint32_t)((widx - fc.w_lo) * nk + (kloc - fc.k_lo));
-                    ri = rridx[i] | ((ms_ >> META_VALID_SHIFT) << 31);
+                ms = rmeta[i];
+                g = bin_of(ms);
+                if (g != 0x1FFu) {
+                    /* fold records (DIRECT, L2) carry validity in bit 31;
+                     * L1 passes the raw rowidx through (meta travels) */
+                    ri = (MODE == RG_L1)
+                             ? rridx[i]
+                             : (rridx[i] | ((ms >> META_VALID_SHIFT) << 31));
                     v = rvals[i];
                 }
             }
@@ -615,6 +651,7 @@ __global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
             if (g != 0x1FFu) {
                 s_ridx[pos] = ri;
                 s_val[pos] = v;
+                if (MODE == RG_L1) s_meta[pos] = ms;
                 s_dest[pos] = lo + gcur[g] + (pos - stoffs[g]);
             }
         }
@@ -623,8 +660,9 @@ __global__ __launch_bounds__(BLOCK) void k_regroup(const uint32_t* rmeta,
         const uint32_t tot = s_total;
         for (uint32_t p = threadIdx.x; p < tot; p += BLOCK) {
             const uint32_t d = s_dest[p];
-            gval[d] = s_val[p];
-            gridx[d] = s_ridx[p];
+            oval[d] = s_val[p];
+            oridx[d] = s_ridx[p];
+            if (MODE == RG_L1) ometa[d] = s_meta[p];
         }
         __syncthreads();
         for (int g = threadIdx.x; g < GCAP; g += BLOCK) {
@@ -640,138 +678,30 @@ void launch_regroup(hipStream_t s, const uint32_t* d_meta, const uint32_t* d_rid
                     const double* d_rvals, const uint32_t* d_bucket_base,
                     const FoldChunk& fc, uint32_t* d_binoffs, uint32_t* d_binlens,
                     double* d_gval, uint32_t* d_gridx) {
-    hipLaunchKernelGGL(k_regroup, dim3(NB), dim3(BLOCK), 0, s, d_meta, d_ridx,
-                       d_rvals, d_bucket_base, fc, d_binoffs, d_binlens, d_gval,
-                       d_gridx);
+    hipLaunchKernelGGL(k_regroup_t<RG_DIRECT>, dim3(NB), dim3(BLOCK), 0, s,
+                       d_meta, d_ridx, d_rvals, d_bucket_base, fc, nullptr,
+                       nullptr, d_binoffs, d_binlens, nullptr, d_gridx, d_gval);
 }
 
-
-/* ------------------------------------------------------------------ */
-/* regroup_big — the G_local > GCAP regime (large keyspaces, e.g. cfg3's
- * 1M keys): bins live in GLOBAL scratch (block-private => L2-resident)
- * instead of LDS, the whole (window x kloc) bin space is split in ONE
- * launch (the chunked LDS form would re-read the bucket per 256-bin
- * chunk). Wave-quarters span the whole bucket region, so cursors are
- * per-wave private plain RMW. Output writes are scattered (runs < 1 at
- * this bin density) — still far cheaper than chunked re-reads.        */
-/* ------------------------------------------------------------------ */
-
-__global__ __launch_bounds__(BLOCK) void k_regroup_big(const uint32_t* rmeta,
-        const uint32_t* rridx, const double* rvals, const uint32_t* bucket_base,
-        FoldChunk fc, int32_t gtot, uint32_t* binoffs, uint32_t* binlens,
-        double* gval, uint32_t* gridx) {
-    /* big-bin regime (GCAP < gtot <= ~37k, e.g. cfg3's 1M keys): bin
-     * counters/cursors live in DYNAMIC LDS (gtot*4 bytes), counted with LDS
-     * atomics and claimed through wave-serialized phases over interleaved
-     * 64-record tiles (global tile order == row order). The earlier global-
-     * scratch form was HBM-atomic-bound (~6.5 ms per 40M records). */
-    extern __shared__ __attribute__((aligned(16))) char smem[];
-    uint32_t* cnt = (uint32_t*)smem; /* becomes the cursor array in place */
-    __shared__ uint32_t scanbuf[BLOCK];
-    const int bkt = blockIdx.x;
-    const int lane = threadIdx.x & 63;
-    const int wave = threadIdx.x >> 6;
-    const uint32_t lo = bucket_base[bkt];
-    const uint32_t hi = bucket_base[bkt + 1];
-    const int nk = fc.k_hi - fc.k_lo; /* == klocs (single launch) */
-    uint32_t* offs = binoffs + (int64_t)bkt * gtot;
-    uint32_t* lens = binlens + (int64_t)bkt * gtot;
-    for (int g = threadIdx.x; g < gtot; g += BLOCK) cnt[g] = 0;
-    __syncthreads();
-    /* pass 1: bin counts (LDS atomics) */
-    for (uint32_t i = lo + threadIdx.x; i < hi; i += BLOCK) {
-        const uint32_t ms_ = rmeta[i];
-        const int widx = (int)((ms_ >> META_WIDX_SHIFT) & META_WIDX_MASK);
-        const int kloc = (int)(ms_ & META_KLOC_MASK);
-        atomicAdd(&cnt[widx * nk + kloc], 1u);
-    }
-    __syncthreads();
-    /* exclusive prefix in place (publishing binoffs/binlens as we go) */
-    const int per = (gtot + BLOCK - 1) / BLOCK;
-    const int b0 = threadIdx.x * per;
-    const int b1 = min(gtot, b0 + per);
-    {
-        uint32_t s = 0;
-        for (int b = b0; b < b1; b++) s += cnt[b];
-        scanbuf[threadIdx.x] = s;
-    }
-    __syncthreads();
-    for (int o = 1; o < BLOCK; o <<= 1) {
-        uint32_t v = (threadIdx.x >= (unsigned)o) ? scanbuf[threadIdx.x - o] : 0;
-        __syncthreads();
-        scanbuf[threadIdx.x] += v;
-        __syncthreads();
-    }
-    {
-        uint32_t run = threadIdx.x ? scanbuf[threadIdx.x - 1] : 0;
-        for (int b = b0; b < b1; b++) {
-            uint32_t t = cnt[b];
-            offs[b] = run;
-            lens[b] = t;
-            cnt[b] = run; /* cursor */
-            run += t;
-        }
-    }
-    __syncthreads();
-    /* pass 2: placement. Interleaved 64-record tiles (wave w owns tiles
-     * 4t+w) with claim phases serialized in wave order per tile group, so
-     * cursor claims follow global row order. */
-    for (uint32_t grp = lo; grp < hi; grp += BLOCK) {
-        const uint32_t t0 = grp + (uint32_t)wave * 64;
-        const uint32_t i = t0 + lane;
-        const bool act = t0 < hi && i < hi;
-        uint32_t g = 0;
-        uint32_t ri = 0;
-        double v = 0.0;
-        if (act) {
-            const uint32_t ms_ = rmeta[i];
-            const int widx = (int)((ms_ >> META_WIDX_SHIFT) & META_WIDX_MASK);
-            const int kloc = (int)(ms_ & META_KLOC_MASK);
-            g = (uint32_t)(widx * nk + kloc);
-            ri = rridx[i] | ((ms_ >> META_VALID_SHIFT) << 31);
-            v = rvals[i];
-        }
-        /* intra-wave ranks via readlane loop (bins too wide for ballots) */
-        uint32_t r = 0, wtot = 0;
-        int fl = lane;
-        const uint32_t gk = act ? g : 0xFFFFFFFFu;
-        for (int j = 0; j < 64; j++) {
-            uint32_t gj = (uint32_t)__builtin_amdgcn_readlane((int)gk, j);
-            if (gj == gk) {
-                if (j < lane) r++;
-                wtot++;
-                if (j < fl) fl = j;
-            }
-        }
-        uint32_t pos = 0;
-        for (int w = 0; w < WAVES_PER_BLOCK; w++) {
-            if (wave == w && act) {
-                uint32_t pre = 0;
-                if (lane == fl) {
-                    pre = cnt[g];
-                    cnt[g] = pre + wtot;
-                }
-                pre = (uint32_t)__shfl((int)pre, fl);
-                pos = pre + r;
-            }
-            __syncthreads();
-        }
-        if (act) {
-            gval[lo + pos] = v;
-            gridx[lo + pos] = ri;
-        }
-    }
+void launch_regroup_l1(hipStream_t s, const uint32_t* d_meta,
+                       const uint32_t* d_ridx, const double* d_rvals,
+                       const uint32_t* d_bucket_base, const FoldChunk& fc,
+                       uint32_t* d_b1offs, uint32_t* d_b1lens, uint32_t* d_meta2,
+                       uint32_t* d_ridx2, double* d_rvals2) {
+    hipLaunchKernelGGL(k_regroup_t<RG_L1>, dim3(NB), dim3(BLOCK), 0, s, d_meta,
+                       d_ridx, d_rvals, d_bucket_base, fc, nullptr, nullptr,
+                       d_b1offs, d_b1lens, d_meta2, d_ridx2, d_rvals2);
 }
 
-void launch_regroup_big(hipStream_t s, const uint32_t* d_meta,
-                        const uint32_t* d_ridx, const double* d_rvals,
-                        const uint32_t* d_bucket_base, const FoldChunk& fc,
-                        int32_t gtot, uint32_t* d_binoffs, uint32_t* d_binlens,
-                        double* d_gval, uint32_t* d_gridx) {
-    size_t lds = (size_t)gtot * 4;
-    hipLaunchKernelGGL(k_regroup_big, dim3(NB), dim3(BLOCK), lds, s, d_meta,
-                       d_ridx, d_rvals, d_bucket_base, fc, gtot, d_binoffs,
-                       d_binlens, d_gval, d_gridx);
+void launch_regroup_l2(hipStream_t s, const uint32_t* d_meta2,
+                       const uint32_t* d_ridx2, const double* d_rvals2,
+                       const uint32_t* d_bucket_base, const FoldChunk& fc,
+                       int nb1, const uint32_t* d_b1offs, const uint32_t* d_b1lens,
+                       uint32_t* d_binoffs, uint32_t* d_binlens, double* d_gval,
+                       uint32_t* d_gridx) {
+    hipLaunchKernelGGL(k_regroup_t<RG_L2>, dim3(NB, nb1), dim3(BLOCK), 0, s,
+                       d_meta2, d_ridx2, d_rvals2, d_bucket_base, fc, d_b1offs,
+                       d_b1lens, d_binoffs, d_binlens, nullptr, d_gridx, d_gval);
 }
 
 /* ------------------------------------------------------------------ */
@@ -786,16 +716,24 @@ __global__ __launch_bounds__(BLOCK) void k_fold3(const double* gval,
         const int32_t* slot_of_widx, uint64_t* s_cnt, double* s_min,
         double* s_max, double* s_sum, uint64_t* s_first) {
     const int nk = fc.k_hi - fc.k_lo;
-    const int glocal_n = (fc.w_hi - fc.w_lo) * nk;
+    const int glocal_n = fc.tl_nw ? fc.bin_stride : (fc.w_hi - fc.w_lo) * nk;
     const int waves_per_bkt = (glocal_n + 63) / 64;
     const int64_t gw = (blockIdx.x * (int64_t)BLOCK + threadIdx.x) >> 6;
     const int bkt = (int)(gw / waves_per_bkt);
     const int lane = threadIdx.x & 63;
     const int g = (int)(gw % waves_per_bkt) * 64 + lane;
     if (bkt >= NB) return;
-    const bool own = g < glocal_n;
-    const int my_widx = fc.w_lo + (own ? g / nk : 0);
-    const int my_kloc = fc.k_lo + (own ? g % nk : 0);
+    bool own = g < glocal_n;
+    int my_widx, my_kloc;
+    if (fc.tl_nw) {
+        /* two-level numbering: g = ((kloc>>8)*nw + widx)*256 + (kloc&255) */
+        my_widx = (g >> 8) % fc.tl_nw;
+        my_kloc = ((g >> 8) / fc.tl_nw) * 256 + (g & 255);
+        own = own && my_kloc < nk && my_widx < fc.w_hi;
+    } else {
+        my_widx = fc.w_lo + (own ? g / nk : 0);
+        my_kloc = fc.k_lo + (own ? g % nk : 0);
+    }
 
     const uint32_t lo = bucket_base[bkt];
     uint32_t off = 0, len = 0;

@@ -153,6 +153,12 @@ struct dz_window_op {
     uint32_t* d_binoffs = nullptr;
     uint32_t* d_binlens = nullptr;
     int64_t bin_cap = 0;   /* bins per bucket the binoffs/lens arrays hold */
+    uint32_t* d_b1offs = nullptr; /* two-level L1 segment layout [NB][256] */
+    uint32_t* d_b1lens = nullptr;
+    uint32_t* d_meta2 = nullptr;  /* two-level intermediate records */
+    uint32_t* d_ridx2 = nullptr;
+    double* d_rvals2 = nullptr;
+    int64_t l2_cap = 0;
     int64_t rec_cap = 0;
     uint64_t* d_scalars = nullptr;
     uint64_t* h_scalars = nullptr; /* pinned, 3 */
@@ -426,6 +432,8 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_meta); hipFree(op->d_ridx); hipFree(op->d_rvals);
     hipFree(op->d_gval); hipFree(op->d_gridx);
     hipFree(op->d_binoffs); hipFree(op->d_binlens);
+    hipFree(op->d_b1offs); hipFree(op->d_b1lens);
+    hipFree(op->d_meta2); hipFree(op->d_ridx2); hipFree(op->d_rvals2);
     hipFree(op->d_scalars); hipFree(op->d_slotmap); hipFree(op->d_zero_kid);
     hipFree(op->d_resetlist);
     hipFree(op->d_ts); hipFree(op->d_kid); hipFree(op->d_vals); hipFree(op->d_valbm);
@@ -1172,11 +1180,25 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
     });
     int64_t klocs = op->kcap >> dz::LOG_NB;
     int64_t gtot = klocs * nw;
-    constexpr int64_t BIG_GTOT_MAX = 37888; /* bins fit dynamic LDS (x4 B) */
-    if (gtot > dz::FOLD_GCAP && gtot <= BIG_GTOT_MAX) {
-        /* single-launch big-bin regime: bins in global scratch, no chunked
-         * re-reads (cfg3-class keyspaces) */
-        if (ensure_bins(op, gtot) != DZ_OK) return DZ_ERR;
+    int64_t khigh = (klocs + 255) >> 8;
+    int64_t nb1 = khigh * nw;
+    if (gtot > dz::FOLD_GCAP && nb1 <= 256) {
+        /* two-level regime (large keyspaces, e.g. cfg3's 1M keys): split by
+         * (kloc>>8, widx) then by kloc&255 — both <=256-bin LDS passes at
+         * full occupancy, no chunked re-reads */
+        int64_t stride = nb1 * 256;
+        if (ensure_bins(op, stride) != DZ_OK) return DZ_ERR;
+        if (!op->d_b1offs) {
+            CHK(op, hipMalloc(&op->d_b1offs, (size_t)dz::NB * 256 * 4));
+            CHK(op, hipMalloc(&op->d_b1lens, (size_t)dz::NB * 256 * 4));
+        }
+        if (nrec_max > op->l2_cap) {
+            hipFree(op->d_meta2); hipFree(op->d_ridx2); hipFree(op->d_rvals2);
+            CHK(op, hipMalloc(&op->d_meta2, (size_t)nrec_max * 4));
+            CHK(op, hipMalloc(&op->d_ridx2, (size_t)nrec_max * 4));
+            CHK(op, hipMalloc(&op->d_rvals2, (size_t)nrec_max * 8));
+            op->l2_cap = nrec_max;
+        }
         dz::FoldChunk fc;
         fc.w_lo = 0;
         fc.w_hi = (int32_t)nw;
@@ -1184,12 +1206,19 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
         fc.k_hi = (int32_t)klocs;
         fc.kcap = op->kcap;
         fc.batch_seq = op->batch_seq;
-        fc.bin_stride = (int32_t)gtot;
-        timed(op, "regroup", (double)nrec_max * 32, [&] {
-            dz::launch_regroup_big(op->stream, op->d_meta, op->d_ridx,
-                                   op->d_rvals, op->d_base, fc, (int32_t)gtot,
-                                   op->d_binoffs, op->d_binlens,
-                                   op->d_gval, op->d_gridx);
+        fc.bin_stride = (int32_t)stride;
+        fc.tl_nw = (int32_t)nw;
+        timed(op, "regroup", (double)nrec_max * 36, [&] {
+            dz::launch_regroup_l1(op->stream, op->d_meta, op->d_ridx,
+                                  op->d_rvals, op->d_base, fc, op->d_b1offs,
+                                  op->d_b1lens, op->d_meta2, op->d_ridx2,
+                                  op->d_rvals2);
+        });
+        timed(op, "regroup", (double)nrec_max * 36, [&] {
+            dz::launch_regroup_l2(op->stream, op->d_meta2, op->d_ridx2,
+                                  op->d_rvals2, op->d_base, fc, (int)nb1,
+                                  op->d_b1offs, op->d_b1lens, op->d_binoffs,
+                                  op->d_binlens, op->d_gval, op->d_gridx);
         });
         timed(op, "fold", (double)nrec_max * 12, [&] {
             dz::launch_fold3(op->stream, op->d_gval, op->d_gridx, op->d_base,
@@ -1211,6 +1240,7 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
                 fc.kcap = op->kcap;
                 fc.batch_seq = op->batch_seq;
                 fc.bin_stride = dz::FOLD_GCAP;
+                fc.tl_nw = 0;
                 timed(op, "regroup", (double)nrec_max * 32, [&] {
                     dz::launch_regroup(op->stream, op->d_meta, op->d_ridx,
                                        op->d_rvals, op->d_base, fc, op->d_binoffs,
