@@ -791,7 +791,8 @@ void launch_fold3(hipStream_t s, const double* d_gval, const uint32_t* d_gridx,
                   const int32_t* d_slot_of_widx, uint64_t* s_cnt, double* s_min,
                   double* s_max, double* s_sum, uint64_t* s_first) {
     int nk = fc.k_hi - fc.k_lo;
-    int glocal_n = (fc.w_hi - fc.w_lo) * nk;
+    /* must mirror the kernel's numbering: two-level mode spans bin_stride */
+    int glocal_n = fc.tl_nw ? fc.bin_stride : (fc.w_hi - fc.w_lo) * nk;
     int waves_per_bkt = (glocal_n + 63) / 64;
     int64_t waves = (int64_t)NB * waves_per_bkt;
     int blocks = (int)((waves + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK);
