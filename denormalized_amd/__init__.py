@@ -108,6 +108,12 @@ class DataStream:
     def collect(self, n_keys_hint=1024):
         return list(self.run(n_keys_hint))
 
+    def sink_python(self, callback, n_keys_hint=1024):
+        """Feed each emitted batch to `callback` (the PyDataStream.sink_python
+        analog, py-denormalized/src/datastream.rs:232-270)."""
+        for out in self.run(n_keys_hint):
+            callback(out)
+
     def print_stream(self):
         for out in self.run():
             n = out["n_rows"]
