@@ -815,18 +815,41 @@ constexpr int RBINS = 1 << RDIG;  /* 2048 */
 constexpr int RCHUNK = 4096;      /* elements per radix block */
 constexpr int RPASSES = 6;        /* 6*11 = 66 >= 64 bits */
 
-__global__ void k_ecompact(const uint64_t* s_first, int64_t K, uint64_t* ekeys,
-                           uint32_t* ekid, uint32_t* eiota, uint32_t* counter) {
+__global__ __launch_bounds__(BLOCK) void k_ecompact(const uint64_t* s_first,
+        int64_t K, uint64_t* ekeys, uint32_t* ekid, uint32_t* eiota,
+        uint32_t* counter) {
+    /* one counter atomic per BLOCK (a returning per-thread atomic on one
+     * word serialises); block-local order is irrelevant pre-sort */
+    __shared__ uint32_t base;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t k = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; k < K;
-         k += stride) {
-        uint64_t f = s_first[k];
-        if (f != ~0ULL) {
-            uint32_t p = atomicAdd(counter, 1u);
+    for (int64_t k0 = (int64_t)blockIdx.x * blockDim.x; k0 < K;
+         k0 += stride) {
+        int64_t k = k0 + threadIdx.x;
+        uint64_t f = (k < K) ? s_first[k] : ~0ULL;
+        const bool hit = f != ~0ULL;
+        /* block count + intra-block rank via wave ballots + LDS */
+        __shared__ uint32_t wsum[WAVES_PER_BLOCK];
+        const int lane = threadIdx.x & 63;
+        const int wave = threadIdx.x >> 6;
+        const uint64_t m = __ballot(hit);
+        const uint64_t below = (lane == 63) ? ~0ULL : ((1ULL << (lane + 1)) - 1);
+        const uint32_t wrank = (uint32_t)__popcll(m & below) - (hit ? 1 : 0);
+        if (lane == 0) wsum[wave] = (uint32_t)__popcll(m);
+        __syncthreads();
+        uint32_t wbase = 0, tot = 0;
+        for (int w = 0; w < WAVES_PER_BLOCK; w++) {
+            if (w < wave) wbase += wsum[w];
+            tot += wsum[w];
+        }
+        if (threadIdx.x == 0) base = tot ? atomicAdd(counter, tot) : 0u;
+        __syncthreads();
+        if (hit) {
+            uint32_t p = base + wbase + wrank;
             ekeys[p] = f;
             ekid[p] = (uint32_t)k;
-            eiota[p] = p; /* sort payload: compact position */
+            eiota[p] = p;
         }
+        __syncthreads();
     }
 }
 
@@ -845,17 +868,32 @@ __global__ __launch_bounds__(BLOCK) void k_rhist(const uint64_t* keys,
         hist[(int64_t)blockIdx.x * RBINS + t] = h[t];
 }
 
-/* digit-major exclusive offsets: offs[b][d] = base[d] + sum_{b'<b} hist[b'][d] */
-__global__ __launch_bounds__(1024) void k_rscan(const uint32_t* hist, int nblk,
-                                                uint32_t* offs) {
+/* digit-major exclusive offsets: offs[b][d] = base[d] + sum_{b'<b} hist[b'][d]
+ * Parallel 3-stage form: the single-block version was 44% of cfg3's GPU time
+ * (profiles/r01_kernel_stats_cfg3.csv). */
+constexpr int RSEG = 16;
+
+__global__ void k_rscan_a(const uint32_t* hist, int nblk, int bs,
+                          uint32_t* psum) {
+    int d = blockIdx.x * blockDim.x + threadIdx.x;
+    int seg = blockIdx.y;
+    if (d >= RBINS) return;
+    int b0 = seg * bs, b1 = min(nblk, b0 + bs);
+    uint32_t s = 0;
+    for (int b = b0; b < b1; b++) s += hist[(int64_t)b * RBINS + d];
+    psum[(int64_t)seg * RBINS + d] = s;
+}
+
+__global__ __launch_bounds__(1024) void k_rscan_b(const uint32_t* psum,
+                                                  uint32_t* dbase) {
     __shared__ uint32_t part[1024];
-    constexpr int PER = RBINS / 1024; /* 2 */
+    constexpr int PER = RBINS / 1024;
     uint32_t loc[PER];
     uint32_t s = 0;
     for (int j = 0; j < PER; j++) {
         int d = threadIdx.x * PER + j;
         uint32_t t = 0;
-        for (int b = 0; b < nblk; b++) t += hist[(int64_t)b * RBINS + d];
+        for (int g = 0; g < RSEG; g++) t += psum[(int64_t)g * RBINS + d];
         loc[j] = s;
         s += t;
     }
@@ -868,14 +906,22 @@ __global__ __launch_bounds__(1024) void k_rscan(const uint32_t* hist, int nblk,
         __syncthreads();
     }
     uint32_t pre = threadIdx.x ? part[threadIdx.x - 1] : 0;
-    for (int j = 0; j < PER; j++) {
-        int d = threadIdx.x * PER + j;
-        uint32_t run = pre + loc[j];
-        for (int b = 0; b < nblk; b++) {
-            uint32_t t = hist[(int64_t)b * RBINS + d];
-            offs[(int64_t)b * RBINS + d] = run;
-            run += t;
-        }
+    for (int j = 0; j < PER; j++) dbase[threadIdx.x * PER + j] = pre + loc[j];
+}
+
+__global__ void k_rscan_c(const uint32_t* hist, const uint32_t* psum,
+                          const uint32_t* dbase, int nblk, int bs,
+                          uint32_t* offs) {
+    int d = blockIdx.x * blockDim.x + threadIdx.x;
+    int seg = blockIdx.y;
+    if (d >= RBINS) return;
+    uint32_t run = dbase[d];
+    for (int g = 0; g < seg; g++) run += psum[(int64_t)g * RBINS + d];
+    int b0 = seg * bs, b1 = min(nblk, b0 + bs);
+    for (int b = b0; b < b1; b++) {
+        uint32_t t = hist[(int64_t)b * RBINS + d];
+        offs[(int64_t)b * RBINS + d] = run;
+        run += t;
     }
 }
 
@@ -1073,18 +1119,23 @@ __global__ __launch_bounds__(SBLK) void k_rsort_small(uint64_t* a_keys,
 /* filter-compact: evaluate the pushed-down predicate on each touched group
  * (straight from the slot slab) and pack the passers:
  * {first, kid, iota} triples for the sort + gather stages. */
-__global__ void k_efilter(const uint64_t* ekeys, const uint32_t* ekid,
-                          const uint32_t* counter, const uint64_t* s_cnt,
-                          const double* s_min, const double* s_max,
-                          const double* s_sum, EmitFilter ef, uint64_t* fkeys,
-                          uint32_t* fkid, uint32_t* fiota, uint32_t* counter2) {
+__global__ __launch_bounds__(BLOCK) void k_efilter(const uint64_t* ekeys,
+        const uint32_t* ekid, const uint32_t* counter, const uint64_t* s_cnt,
+        const double* s_min, const double* s_max, const double* s_sum,
+        EmitFilter ef, uint64_t* fkeys, uint32_t* fkid, uint32_t* fiota,
+        uint32_t* counter2) {
+    __shared__ uint32_t base;
+    __shared__ uint32_t wsum[WAVES_PER_BLOCK];
     const uint32_t nt = *counter;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nt;
-         i += stride) {
-        const uint32_t kid = ekid[i];
-        bool pass = true;
-        if (ef.on) {
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    for (int64_t i0 = (int64_t)blockIdx.x * blockDim.x; i0 < nt; i0 += stride) {
+        int64_t i = i0 + threadIdx.x;
+        const bool act = i < nt;
+        const uint32_t kid = act ? ekid[i] : 0;
+        bool pass = act;
+        if (act && ef.on) {
             const uint64_t c = s_cnt[kid];
             const bool valid = c > 0;
             double v = 0.0;
@@ -1112,12 +1163,25 @@ __global__ void k_efilter(const uint64_t* ekeys, const uint32_t* ekid,
                 }
             }
         }
+        const uint64_t m = __ballot(pass);
+        const uint64_t below = (lane == 63) ? ~0ULL : ((1ULL << (lane + 1)) - 1);
+        const uint32_t wrank = (uint32_t)__popcll(m & below) - (pass ? 1 : 0);
+        if (lane == 0) wsum[wave] = (uint32_t)__popcll(m);
+        __syncthreads();
+        uint32_t wbase = 0, tot = 0;
+        for (int w = 0; w < WAVES_PER_BLOCK; w++) {
+            if (w < wave) wbase += wsum[w];
+            tot += wsum[w];
+        }
+        if (threadIdx.x == 0) base = tot ? atomicAdd(counter2, tot) : 0u;
+        __syncthreads();
         if (pass) {
-            uint32_t p = atomicAdd(counter2, 1u);
+            uint32_t p = base + wbase + wrank;
             fkeys[p] = ekeys[i];
             fkid[p] = kid;
             fiota[p] = p;
         }
+        __syncthreads();
     }
 }
 
@@ -1184,12 +1248,18 @@ void launch_emission_sort(hipStream_t s, int64_t K, uint64_t* fkeys,
         uint32_t* pa = skid;
         uint64_t* kb = skeys;
         uint32_t* pb = okid;
+        int bs = (nblk + RSEG - 1) / RSEG;
+        uint32_t* psum = rhist + (int64_t)nblk * RBINS;  /* scratch tail */
+        uint32_t* dbase = psum + (int64_t)RSEG * RBINS;
         for (int p = 0; p < RPASSES; p++) {
             int shift = p * RDIG;
             hipLaunchKernelGGL(k_rhist, dim3(nblk), dim3(BLOCK), 0, s, ka,
                                counter, shift, rhist);
-            hipLaunchKernelGGL(k_rscan, dim3(1), dim3(1024), 0, s, rhist, nblk,
-                               roffs);
+            hipLaunchKernelGGL(k_rscan_a, dim3(RBINS / 256, RSEG), dim3(256), 0,
+                               s, rhist, nblk, bs, psum);
+            hipLaunchKernelGGL(k_rscan_b, dim3(1), dim3(1024), 0, s, psum, dbase);
+            hipLaunchKernelGGL(k_rscan_c, dim3(RBINS / 256, RSEG), dim3(256), 0,
+                               s, rhist, psum, dbase, nblk, bs, roffs);
             hipLaunchKernelGGL(k_rscatter, dim3(nblk), dim3(BLOCK), 0, s, ka, pa,
                                counter, shift, roffs, kb, pb);
             std::swap(ka, kb);

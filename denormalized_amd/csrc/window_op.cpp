@@ -829,7 +829,9 @@ static dz_status trigger_windows(dz_window_op* op) {
         int64_t nblk = (kc + dz::EMIT_RCHUNK - 1) / dz::EMIT_RCHUNK;
         hipFree(op->d_rhist);
         hipFree(op->d_roffs);
-        CHK(op, hipMalloc(&op->d_rhist, (size_t)nblk * dz::EMIT_RBINS * 4));
+        /* + tail: RSEG partial rows + digit bases for the parallel scan */
+        CHK(op, hipMalloc(&op->d_rhist,
+                          (size_t)(nblk + 17) * dz::EMIT_RBINS * 4));
         CHK(op, hipMalloc(&op->d_roffs, (size_t)nblk * dz::EMIT_RBINS * 4));
         if (!op->e_pcnt)
             CHK(op, hipHostMalloc((void**)&op->e_pcnt, dz_window_op::E_POOL * 4));
