@@ -249,6 +249,7 @@ struct dz_window_op {
     } while (0)
 
 static void emit_worker_main(dz_window_op* op);
+static dz_status ensure_emission(dz_window_op* op);
 
 static hipEvent_t get_event(dz_window_op* op) {
     if (!op->ev_pool.empty()) {
@@ -400,6 +401,11 @@ extern "C" dz_window_op* dz_window_op_create(const dz_window_desc* desc) {
         hipStreamDestroy(op->copy_stream);
         delete op;
         return nullptr;
+    }
+    if (ensure_emission(op) != DZ_OK) {
+        g_err = op->err;
+        /* continue: trigger retries; create-time prealloc is a fast-path */
+        op->err.clear();
     }
     for (int i = 0; i < dz_window_op::E_WORKERS; i++)
         op->e_workers.emplace_back(emit_worker_main, op);
@@ -807,23 +813,9 @@ static void emit_worker_main(dz_window_op* op) {
     }
 }
 
-/* trigger_windows (grouped_window_agg_stream.rs:220-253): closed windows are
- * copied D2H asynchronously and built by the worker thread off the push
- * critical path; dz_window_op_drain/finish wait for completion. */
-static dz_status trigger_windows(dz_window_op* op) {
-    if (!op->has_wm) return DZ_OK;
-    struct Closed { int64_t start, end; int32_t slot; };
-    std::vector<Closed> closed;
-    for (auto it = op->open.begin(); it != op->open.end();) {
-        if (op->watermark >= it->second.end) {
-            closed.push_back({it->first, it->second.end, it->second.slot});
-            it = op->open.erase(it);
-        } else {
-            ++it;
-        }
-    }
-    if (closed.empty()) return DZ_OK;
-    if (op->e_slab_kcap != op->kcap) {
+static dz_status ensure_emission(dz_window_op* op) {
+    if (op->e_slab_kcap == op->kcap) return DZ_OK;
+    {
         emit_drain(op);
         int64_t kc = op->kcap;
         int64_t nblk = (kc + dz::EMIT_RCHUNK - 1) / dz::EMIT_RCHUNK;
@@ -865,6 +857,26 @@ static dz_status trigger_windows(dz_window_op* op) {
         for (int i = 0; i < dz_window_op::E_POOL; i++) op->e_free.push_back(i);
         op->e_slab_kcap = op->kcap;
     }
+    return DZ_OK;
+}
+
+/* trigger_windows (grouped_window_agg_stream.rs:220-253): closed windows are
+ * copied D2H asynchronously and built by the worker thread off the push
+ * critical path; dz_window_op_drain/finish wait for completion. */
+static dz_status trigger_windows(dz_window_op* op) {
+    if (!op->has_wm) return DZ_OK;
+    struct Closed { int64_t start, end; int32_t slot; };
+    std::vector<Closed> closed;
+    for (auto it = op->open.begin(); it != op->open.end();) {
+        if (op->watermark >= it->second.end) {
+            closed.push_back({it->first, it->second.end, it->second.slot});
+            it = op->open.erase(it);
+        } else {
+            ++it;
+        }
+    }
+    if (closed.empty()) return DZ_OK;
+    if (ensure_emission(op) != DZ_OK) return DZ_ERR;
     size_t stride = (size_t)op->kcap * 5;
     HostTimer ht(op, "h_emit_enqueue");
     /* emission kernels run on the copy stream AFTER the compute stream's
