@@ -999,121 +999,6 @@ __global__ __launch_bounds__(BLOCK) void k_rscatter(const uint64_t* keys,
 }
 
 
-/* single-block LSD radix for small group counts (nt <= ~64k): all passes in
- * ONE launch; 1024 threads (16 wave-quarters) and all-equal digits skipped
- * device-side (typical `first` keys use < 40 of the 64 bits). Stable
- * wave-quarter placement; global ping-pong between active passes. The sort
- * runs AFTER the slab readers, so it never gates window-slot reuse. */
-constexpr int SBLK = 1024;
-constexpr int SWAVES = SBLK / 64;
-
-__global__ __launch_bounds__(SBLK) void k_rsort_small(uint64_t* a_keys,
-        uint32_t* a_pay, uint64_t* b_keys, uint32_t* b_pay,
-        const uint32_t* counter) {
-    __shared__ uint32_t cnt4[SWAVES][RBINS];
-    __shared__ uint32_t scanbuf[SBLK];
-    __shared__ int allsame;
-    const uint32_t nt = *counter;
-    if (nt < 2) return;
-    const int lane = threadIdx.x & 63;
-    const int wave = threadIdx.x >> 6;
-    uint64_t* ka = a_keys;
-    uint32_t* pa = a_pay;
-    uint64_t* kb = b_keys;
-    uint32_t* pb = b_pay;
-    const uint32_t q = (nt + SWAVES - 1) / SWAVES;
-    const uint32_t w0 = min(nt, (uint32_t)wave * q);
-    const uint32_t w1 = min(nt, w0 + q);
-    for (int pass = 0; pass < RPASSES; pass++) {
-        const int shift = pass * RDIG;
-        for (int t = threadIdx.x; t < RBINS; t += SBLK)
-            for (int w = 0; w < SWAVES; w++) cnt4[w][t] = 0;
-        if (threadIdx.x == 0) allsame = 0;
-        __syncthreads();
-        for (uint32_t i = w0 + lane; i < w1; i += 64)
-            atomicAdd(&cnt4[wave][(uint32_t)(ka[i] >> shift) & (RBINS - 1)], 1u);
-        __syncthreads();
-        { /* exclusive digit prefix + per-wave bases; detect all-equal digit */
-            constexpr int PER = RBINS / SBLK; /* 2 */
-            uint32_t loc[PER];
-            uint32_t s = 0;
-            for (int j = 0; j < PER; j++) {
-                int d = threadIdx.x * PER + j;
-                uint32_t t = 0;
-                for (int w = 0; w < SWAVES; w++) t += cnt4[w][d];
-                if (t == nt) allsame = 1;
-                loc[j] = s;
-                s += t;
-            }
-            scanbuf[threadIdx.x] = s;
-            __syncthreads();
-            if (allsame) continue; /* this digit sorts nothing: skip pass */
-            for (int o = 1; o < SBLK; o <<= 1) {
-                uint32_t v = (threadIdx.x >= (unsigned)o) ? scanbuf[threadIdx.x - o] : 0;
-                __syncthreads();
-                scanbuf[threadIdx.x] += v;
-                __syncthreads();
-            }
-            uint32_t pre = threadIdx.x ? scanbuf[threadIdx.x - 1] : 0;
-            for (int j = 0; j < PER; j++) {
-                int d = threadIdx.x * PER + j;
-                uint32_t run = pre + loc[j];
-                for (int w = 0; w < SWAVES; w++) {
-                    uint32_t c = cnt4[w][d];
-                    cnt4[w][d] = run;
-                    run += c;
-                }
-            }
-        }
-        __syncthreads();
-        for (uint32_t t0 = w0; t0 < w1; t0 += 64) {
-            const uint32_t i = t0 + lane;
-            const bool act = i < w1;
-            uint64_t key = act ? ka[i] : ~0ULL;
-            uint32_t pay = act ? pa[i] : 0;
-            uint32_t d = act ? ((uint32_t)(key >> shift) & (RBINS - 1)) : 0xFFFFu;
-            uint64_t same = ~0ULL;
-            for (int b = 0; b < RDIG; b++) {
-                uint64_t bb = __ballot((d >> b) & 1);
-                same &= ((d >> b) & 1) ? bb : ~bb;
-            }
-            {
-                uint64_t bb = __ballot(act);
-                same &= act ? bb : ~bb;
-            }
-            const uint64_t below =
-                (lane == 63) ? ~0ULL : ((1ULL << (lane + 1)) - 1);
-            const int rank = (int)__popcll(same & below) - 1;
-            const int leader = __ffsll((unsigned long long)same) - 1;
-            const uint32_t wtot = (uint32_t)__popcll(same);
-            uint32_t pos = 0;
-            {
-                uint32_t pre = 0;
-                if (lane == leader && act) {
-                    pre = cnt4[wave][d];
-                    cnt4[wave][d] = pre + wtot;
-                }
-                pre = (uint32_t)__shfl((int)pre, leader);
-                pos = pre + (uint32_t)rank;
-            }
-            if (act) {
-                kb[pos] = key;
-                pb[pos] = pay;
-            }
-        }
-        __syncthreads();
-        uint64_t* tk = ka; ka = kb; kb = tk;
-        uint32_t* tp = pa; pa = pb; pb = tp;
-    }
-    /* variable number of swaps (skipped passes): ensure the result lands in
-     * (a_keys, a_pay) — copy back if it ended in b */
-    if (ka != a_keys) {
-        for (uint32_t i = threadIdx.x; i < nt; i += SBLK) {
-            a_keys[i] = ka[i];
-            a_pay[i] = pa[i];
-        }
-    }
-}
 
 
 /* filter-compact: evaluate the pushed-down predicate on each touched group
@@ -1238,11 +1123,10 @@ void launch_emission_sort(hipStream_t s, int64_t K, uint64_t* fkeys,
     uint32_t* skid = fiota;
     uint32_t* counter = counter2;
     /* sort (first, compact-index) pairs: keys ekeys<->skeys, payload
-     * skid<->okid; both paths leave the sorted payload in skid */
-    if (K <= 65536) {
-        hipLaunchKernelGGL(k_rsort_small, dim3(1), dim3(SBLK), 0, s, ekeys, skid,
-                           skeys, okid, counter);
-    } else {
+     * skid<->okid; RPASSES is even, so the sorted payload lands in skid.
+     * (The device path only runs above the 64k-key host-emission cutoff,
+     * so the multi-block form is always the right one.) */
+    {
         int nblk = (int)((K + RCHUNK - 1) / RCHUNK);
         uint64_t* ka = ekeys;
         uint32_t* pa = skid;
@@ -1265,7 +1149,6 @@ void launch_emission_sort(hipStream_t s, int64_t K, uint64_t* fkeys,
             std::swap(ka, kb);
             std::swap(pa, pb);
         }
-        /* RPASSES even => payload back in skid */
     }
 }
 
