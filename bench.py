@@ -36,7 +36,7 @@ ALG_BYTES_PER_ROW = 20.0  # key id 4B + reading 8B + ts 8B
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=12)
+    p.add_argument("--steps", type=int, default=24)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--rows-per-step", type=int, default=8_000_000)
     p.add_argument("--keys", type=int, default=10_000)
