@@ -373,7 +373,7 @@ def test_push_device_dense_parity_1M():
 
 def test_randomized_stress_matrix():
     rng = np.random.default_rng(1234)
-    for case in range(6):
+    for case in range(10):
         len_ms = int(rng.choice([500, 1000, 1500, 2000]))
         slide = int(rng.choice([0, 0, 250, 500]))
         if slide > len_ms:
@@ -728,3 +728,66 @@ def test_soak_many_small_pushes():
     assert_parity(outs, o.fetch())
     op.close()
     o.close()
+
+
+def test_shared_watermark_two_partitions():
+    # two partition streams share a watermark (the reference's Arc<Mutex>
+    # across partitions, streaming_window.rs:210): the merged watermark is
+    # the max of batch minimums seen by EITHER partition, injected via
+    # dz_window_op_advance_watermark. Partition B lags; A's watermark must
+    # still close B's windows at the right time.
+    opA = make_op(1000, n_keys_hint=16)
+    opB = make_op(1000, n_keys_hint=16)
+    oA = pyoracle.Oracle(1000, 0)
+    oB = pyoracle.Oracle(1000, 0)
+    outsA, outsB = [], []
+    wm = None
+    for step in range(5):
+        tA = 1_000_000 + step * 700
+        tB = 1_000_000 + step * 650  # lags behind A
+        tsA = (tA + np.arange(500) % 600).astype(np.int64)
+        tsB = (tB + np.arange(500) % 500).astype(np.int64)
+        kA = (np.arange(500) % 7).astype(np.int64)
+        vA = np.linspace(0, 100, 500)
+        opA.push(tsA, kA, vA)
+        opB.push(tsB, kA, vA)
+        oA.push(tsA, kA, vA)
+        oB.push(tsB, kA, vA)
+        wm = max(opA.watermark, opB.watermark)
+        for op_ in (opA, opB):
+            op_.advance_watermark(wm)
+        # mirror the shared watermark on the oracles via a sentinel push? the
+        # oracle lacks an inject API — emulate by asserting GPU watermark
+        # equals the analytical max-of-mins and comparing at finish instead
+        assert opA.watermark == opB.watermark == wm
+        outsA += opA.poll_all()
+        outsB += opB.poll_all()
+    opA.finish()
+    opB.finish()
+    outsA += opA.poll_all()
+    outsB += opB.poll_all()
+    oA.finish()
+    oB.finish()
+    # content parity per partition (timing of emission differs from the
+    # isolated oracles, but window contents must match exactly)
+    expA, expB = oA.fetch(), oB.fetch()
+
+    def by_window(outs):
+        m = {}
+        for b in outs:
+            for i in range(b["n_rows"]):
+                m.setdefault(int(b["window_start"][i]), []).append(
+                    (int(b["key"][i]), int(b["count"][i]), float(b["avg"][i])))
+        return m
+
+    def by_window_exp(e):
+        m = {}
+        for i in range(len(e["key"])):
+            m.setdefault(int(e["window_start"][i]), []).append(
+                (int(e["key"][i]), int(e["count"][i]), float(e["avg"][i])))
+        return m
+
+    assert by_window(outsA) == by_window_exp(expA)
+    assert by_window(outsB) == by_window_exp(expB)
+    for x in (opA, opB, oA, oB):
+        x.close()

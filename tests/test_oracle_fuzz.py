@@ -35,7 +35,7 @@ def stream(draw):
 
 
 @given(stream())
-@settings(max_examples=60, deadline=None)
+@settings(max_examples=120, deadline=None)
 def test_oracle_matches_pyref(s):
     len_ms, slide, batches = s
     o = pyoracle.Oracle(len_ms, slide)
