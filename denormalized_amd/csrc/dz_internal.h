@@ -61,34 +61,33 @@ void launch_scan(hipStream_t stream, const uint32_t* d_ghist, int C,
                  uint32_t* d_psum, uint32_t* d_bucket_base /*NB+1*/,
                  uint32_t* d_gofs);
 
+/* record payloads are 16 B uint4s: {val lo, val hi, rowidx, unused}; the
+ * rowidx word carries the validity bit (bit 31) in FOLD-stage records */
 void launch_scatter(hipStream_t stream, const int32_t* d_kid, const int64_t* d_ts,
                     const double* d_vals, const uint8_t* d_validity /*bitmap|null*/,
                     int64_t n, int64_t chunk, int C, int32_t st_rows,
                     const WinParams& wp, const uint32_t* d_gofs, uint32_t* d_meta,
-                    uint32_t* d_ridx, double* d_rvals);
+                    uint4* d_grec);
 
 constexpr int FOLD_GCAP = 256; /* groups per bucket per fold chunk */
 
 void launch_regroup(hipStream_t stream, const uint32_t* d_meta,
-                    const uint32_t* d_ridx, const double* d_rvals,
-                    const uint32_t* d_bucket_base, const FoldChunk& fc,
-                    uint32_t* d_binoffs, uint32_t* d_binlens, double* d_gval,
-                    uint32_t* d_gridx);
+                    const uint4* d_grec, const uint32_t* d_bucket_base,
+                    const FoldChunk& fc, uint32_t* d_binoffs, uint32_t* d_binlens,
+                    uint4* d_gfrec);
 
 void launch_regroup_l1(hipStream_t stream, const uint32_t* d_meta,
-                       const uint32_t* d_ridx, const double* d_rvals,
-                       const uint32_t* d_bucket_base, const FoldChunk& fc,
-                       uint32_t* d_b1offs, uint32_t* d_b1lens, uint32_t* d_meta2,
-                       uint32_t* d_ridx2, double* d_rvals2);
+                       const uint4* d_grec, const uint32_t* d_bucket_base,
+                       const FoldChunk& fc, uint32_t* d_b1offs,
+                       uint32_t* d_b1lens, uint32_t* d_meta2, uint4* d_grec2);
 
 void launch_regroup_l2(hipStream_t stream, const uint32_t* d_meta2,
-                       const uint32_t* d_ridx2, const double* d_rvals2,
-                       const uint32_t* d_bucket_base, const FoldChunk& fc,
-                       int nb1, const uint32_t* d_b1offs, const uint32_t* d_b1lens,
-                       uint32_t* d_binoffs, uint32_t* d_binlens, double* d_gval,
-                       uint32_t* d_gridx);
+                       const uint4* d_grec2, const uint32_t* d_bucket_base,
+                       const FoldChunk& fc, int nb1, const uint32_t* d_b1offs,
+                       const uint32_t* d_b1lens, uint32_t* d_binoffs,
+                       uint32_t* d_binlens, uint4* d_gfrec);
 
-void launch_fold3(hipStream_t stream, const double* d_gval, const uint32_t* d_gridx,
+void launch_fold3(hipStream_t stream, const uint4* d_gfrec,
                   const uint32_t* d_bucket_base, const uint32_t* d_binoffs,
                   const uint32_t* d_binlens, const FoldChunk& fc,
                   const int32_t* d_slot_of_widx, uint64_t* s_cnt, double* s_min,

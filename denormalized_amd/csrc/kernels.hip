@@ -300,7 +300,7 @@ void launch_scan(hipStream_t s, const uint32_t* d_ghist, int C,
 __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
         const int64_t* ts, const double* vals, const uint8_t* validity,
         int64_t n, int64_t chunk, int32_t st_rows, WinParams wp,
-        const uint32_t* gofs, uint32_t* rmeta, uint32_t* rridx, double* rvals) {
+        const uint32_t* gofs, uint32_t* rmeta, uint4* grec) {
     /* LDS-staged stable partition. Each wave owns a CONTIGUOUS QUARTER of the
      * supertile (wave order == row order), so staging cursors are per-wave
      * private: no cross-wave serialization, ~4 block barriers per supertile.
@@ -315,9 +315,8 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
     __shared__ uint32_t offs[NB];   /* per-supertile exclusive bin prefix     */
     auto wofs = cnt4;
     __shared__ uint32_t s_meta[ST_RECORDS];
-    __shared__ uint32_t s_ridx[ST_RECORDS];
     __shared__ uint32_t s_dest[ST_RECORDS];
-    __shared__ double s_val[ST_RECORDS];
+    __shared__ uint4 s_rec[ST_RECORDS];
     __shared__ uint32_t s_total;
 
     for (int t = threadIdx.x; t < NB; t += BLOCK)
@@ -438,12 +437,14 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
                 base = pre + r;
             }
             const uint32_t kloc = kv >> LOG_NB;
+            const uint64_t vb = (uint64_t)__double_as_longlong(v);
+            const uint4 rec = make_uint4((uint32_t)vb, (uint32_t)(vb >> 32),
+                                         (uint32_t)i, 0u);
             for (int jj = 0; jj < m; jj++) {
                 const uint32_t p = base + jj;
                 s_meta[p] = kloc | ((uint32_t)(jmin + jj) << META_WIDX_SHIFT)
                             | (valid << META_VALID_SHIFT);
-                s_ridx[p] = (uint32_t)i;
-                s_val[p] = v;
+                s_rec[p] = rec;
                 s_dest[p] = cur[bkt] + (p - offs[bkt]);
             }
         }
@@ -453,8 +454,7 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
         for (uint32_t p = threadIdx.x; p < tot; p += BLOCK) {
             const uint32_t d = s_dest[p];
             rmeta[d] = s_meta[p];
-            rridx[d] = s_ridx[p];
-            rvals[d] = s_val[p];
+            grec[d] = s_rec[p];
         }
         __syncthreads();
         for (int t = threadIdx.x; t < NB; t += BLOCK) {
@@ -469,11 +469,9 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
 void launch_scatter(hipStream_t s, const int32_t* d_kid, const int64_t* d_ts,
                     const double* d_vals, const uint8_t* d_validity, int64_t n,
                     int64_t chunk, int C, int32_t st_rows, const WinParams& wp,
-                    const uint32_t* d_gofs, uint32_t* d_meta, uint32_t* d_ridx,
-                    double* d_rvals) {
+                    const uint32_t* d_gofs, uint32_t* d_meta, uint4* d_grec) {
     hipLaunchKernelGGL(k_scatter, dim3(C), dim3(BLOCK), 0, s, d_kid, d_ts, d_vals,
-                       d_validity, n, chunk, st_rows, wp, d_gofs, d_meta, d_ridx,
-                       d_rvals);
+                       d_validity, n, chunk, st_rows, wp, d_gofs, d_meta, d_grec);
 }
 
 /* ------------------------------------------------------------------ */
@@ -502,19 +500,17 @@ enum { RG_DIRECT = 0, RG_L1 = 1, RG_L2 = 2 };
 
 template <int MODE>
 __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
-        const uint32_t* rridx, const double* rvals, const uint32_t* bucket_base,
-        FoldChunk fc, const uint32_t* b1offs, const uint32_t* b1lens,
-        uint32_t* binoffs, uint32_t* binlens, uint32_t* ometa, uint32_t* oridx,
-        double* oval) {
+        const uint4* rrec, const uint32_t* bucket_base, FoldChunk fc,
+        const uint32_t* b1offs, const uint32_t* b1lens, uint32_t* binoffs,
+        uint32_t* binlens, uint32_t* ometa, uint4* orec) {
     __shared__ uint32_t cnt[GCAP];    /* whole-segment bin counts */
     __shared__ uint32_t gcur[GCAP];   /* segment-region bin cursors */
     __shared__ uint32_t stcnt4[WAVES_PER_BLOCK][GCAP]; /* per-wave-quarter */
     __shared__ uint32_t stoffs[GCAP]; /* per-supertile bin prefix */
     __shared__ uint32_t wcur[WAVES_PER_BLOCK][GCAP];   /* per-wave cursors */
     __shared__ uint32_t s_meta[ST_RECORDS]; /* RG_L1 only */
-    __shared__ uint32_t s_ridx[ST_RECORDS];
     __shared__ uint32_t s_dest[ST_RECORDS];
-    __shared__ double s_val[ST_RECORDS];
+    __shared__ uint4 s_rec[ST_RECORDS];
     __shared__ uint32_t s_total;
 
     const int bkt = blockIdx.x;
@@ -614,18 +610,17 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
         for (uint32_t t0 = w0; t0 < w1; t0 += 64) {
             const uint32_t i = t0 + lane;
             uint32_t g = 0x1FFu; /* sentinel above GCAP-1 */
-            uint32_t ms = 0, ri = 0;
-            double v = 0.0;
+            uint32_t ms = 0;
+            uint4 rec = make_uint4(0u, 0u, 0u, 0u);
             if (i < w1) {
                 ms = rmeta[i];
                 g = bin_of(ms);
                 if (g != 0x1FFu) {
+                    rec = rrec[i];
                     /* fold records (DIRECT, L2) carry validity in bit 31;
                      * L1 passes the raw rowidx through (meta travels) */
-                    ri = (MODE == RG_L1)
-                             ? rridx[i]
-                             : (rridx[i] | ((ms >> META_VALID_SHIFT) << 31));
-                    v = rvals[i];
+                    if (MODE != RG_L1)
+                        rec.z |= (ms >> META_VALID_SHIFT) << 31;
                 }
             }
             /* same-bin mask via bit-ballots over the 9 bin-id bits */
@@ -649,8 +644,7 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
                 pos = pre + (uint32_t)rank;
             }
             if (g != 0x1FFu) {
-                s_ridx[pos] = ri;
-                s_val[pos] = v;
+                s_rec[pos] = rec;
                 if (MODE == RG_L1) s_meta[pos] = ms;
                 s_dest[pos] = lo + gcur[g] + (pos - stoffs[g]);
             }
@@ -660,8 +654,7 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
         const uint32_t tot = s_total;
         for (uint32_t p = threadIdx.x; p < tot; p += BLOCK) {
             const uint32_t d = s_dest[p];
-            oval[d] = s_val[p];
-            oridx[d] = s_ridx[p];
+            orec[d] = s_rec[p];
             if (MODE == RG_L1) ometa[d] = s_meta[p];
         }
         __syncthreads();
@@ -674,34 +667,30 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
     }
 }
 
-void launch_regroup(hipStream_t s, const uint32_t* d_meta, const uint32_t* d_ridx,
-                    const double* d_rvals, const uint32_t* d_bucket_base,
-                    const FoldChunk& fc, uint32_t* d_binoffs, uint32_t* d_binlens,
-                    double* d_gval, uint32_t* d_gridx) {
+void launch_regroup(hipStream_t s, const uint32_t* d_meta, const uint4* d_grec,
+                    const uint32_t* d_bucket_base, const FoldChunk& fc,
+                    uint32_t* d_binoffs, uint32_t* d_binlens, uint4* d_gfrec) {
     hipLaunchKernelGGL(k_regroup_t<RG_DIRECT>, dim3(NB), dim3(BLOCK), 0, s,
-                       d_meta, d_ridx, d_rvals, d_bucket_base, fc, nullptr,
-                       nullptr, d_binoffs, d_binlens, nullptr, d_gridx, d_gval);
+                       d_meta, d_grec, d_bucket_base, fc, nullptr, nullptr,
+                       d_binoffs, d_binlens, nullptr, d_gfrec);
 }
 
-void launch_regroup_l1(hipStream_t s, const uint32_t* d_meta,
-                       const uint32_t* d_ridx, const double* d_rvals,
+void launch_regroup_l1(hipStream_t s, const uint32_t* d_meta, const uint4* d_grec,
                        const uint32_t* d_bucket_base, const FoldChunk& fc,
                        uint32_t* d_b1offs, uint32_t* d_b1lens, uint32_t* d_meta2,
-                       uint32_t* d_ridx2, double* d_rvals2) {
+                       uint4* d_grec2) {
     hipLaunchKernelGGL(k_regroup_t<RG_L1>, dim3(NB), dim3(BLOCK), 0, s, d_meta,
-                       d_ridx, d_rvals, d_bucket_base, fc, nullptr, nullptr,
-                       d_b1offs, d_b1lens, d_meta2, d_ridx2, d_rvals2);
+                       d_grec, d_bucket_base, fc, nullptr, nullptr, d_b1offs,
+                       d_b1lens, d_meta2, d_grec2);
 }
 
-void launch_regroup_l2(hipStream_t s, const uint32_t* d_meta2,
-                       const uint32_t* d_ridx2, const double* d_rvals2,
+void launch_regroup_l2(hipStream_t s, const uint32_t* d_meta2, const uint4* d_grec2,
                        const uint32_t* d_bucket_base, const FoldChunk& fc,
                        int nb1, const uint32_t* d_b1offs, const uint32_t* d_b1lens,
-                       uint32_t* d_binoffs, uint32_t* d_binlens, double* d_gval,
-                       uint32_t* d_gridx) {
+                       uint32_t* d_binoffs, uint32_t* d_binlens, uint4* d_gfrec) {
     hipLaunchKernelGGL(k_regroup_t<RG_L2>, dim3(NB, nb1), dim3(BLOCK), 0, s,
-                       d_meta2, d_ridx2, d_rvals2, d_bucket_base, fc, d_b1offs,
-                       d_b1lens, d_binoffs, d_binlens, nullptr, d_gridx, d_gval);
+                       d_meta2, d_grec2, d_bucket_base, fc, d_b1offs, d_b1lens,
+                       d_binoffs, d_binlens, nullptr, d_gfrec);
 }
 
 /* ------------------------------------------------------------------ */
@@ -710,11 +699,11 @@ void launch_regroup_l2(hipStream_t s, const uint32_t* d_meta2,
 /* fold in row order (bit-exact vs the reference accumulators).        */
 /* ------------------------------------------------------------------ */
 
-__global__ __launch_bounds__(BLOCK) void k_fold3(const double* gval,
-        const uint32_t* gridx, const uint32_t* bucket_base,
-        const uint32_t* binoffs, const uint32_t* binlens, FoldChunk fc,
-        const int32_t* slot_of_widx, uint64_t* s_cnt, double* s_min,
-        double* s_max, double* s_sum, uint64_t* s_first) {
+__global__ __launch_bounds__(BLOCK) void k_fold3(const uint4* gfrec,
+        const uint32_t* bucket_base, const uint32_t* binoffs,
+        const uint32_t* binlens, FoldChunk fc, const int32_t* slot_of_widx,
+        uint64_t* s_cnt, double* s_min, double* s_max, double* s_sum,
+        uint64_t* s_first) {
     const int nk = fc.k_hi - fc.k_lo;
     const int glocal_n = fc.tl_nw ? fc.bin_stride : (fc.w_hi - fc.w_lo) * nk;
     const int waves_per_bkt = (glocal_n + 63) / 64;
@@ -758,7 +747,7 @@ __global__ __launch_bounds__(BLOCK) void k_fold3(const double* gval,
         }
     }
     if (own && len > 0 && fst == ~0ULL)
-        fst = ((uint64_t)fc.batch_seq << 32) | (gridx[lo + off] & 0x7FFFFFFFu);
+        fst = ((uint64_t)fc.batch_seq << 32) | (gfrec[lo + off].z & 0x7FFFFFFFu);
     uint32_t mlen = len;
     for (int o = 32; o > 0; o >>= 1)
         mlen = max(mlen, (uint32_t)__shfl_down((int)mlen, o));
@@ -766,9 +755,10 @@ __global__ __launch_bounds__(BLOCK) void k_fold3(const double* gval,
 #pragma unroll 4
     for (uint32_t r = 0; r < mlen; r++) {
         const bool act = own && r < len;
-        const uint32_t ri = act ? gridx[lo + off + r] : 0;
-        const double v = act ? gval[lo + off + r] : 0.0;
-        const bool upd = act && (ri >> 31);
+        const uint4 rec = act ? gfrec[lo + off + r] : make_uint4(0u, 0u, 0u, 0u);
+        const double v = __longlong_as_double(
+            (long long)(((uint64_t)rec.y << 32) | rec.x));
+        const bool upd = act && (rec.z >> 31);
         const bool fresh = cnt == 0;
         mn = (upd && (fresh || v < mn)) ? v : mn;
         mx = (upd && (fresh || v > mx)) ? v : mx;
@@ -785,7 +775,7 @@ __global__ __launch_bounds__(BLOCK) void k_fold3(const double* gval,
     }
 }
 
-void launch_fold3(hipStream_t s, const double* d_gval, const uint32_t* d_gridx,
+void launch_fold3(hipStream_t s, const uint4* d_gfrec,
                   const uint32_t* d_bucket_base, const uint32_t* d_binoffs,
                   const uint32_t* d_binlens, const FoldChunk& fc,
                   const int32_t* d_slot_of_widx, uint64_t* s_cnt, double* s_min,
@@ -796,7 +786,7 @@ void launch_fold3(hipStream_t s, const double* d_gval, const uint32_t* d_gridx,
     int waves_per_bkt = (glocal_n + 63) / 64;
     int64_t waves = (int64_t)NB * waves_per_bkt;
     int blocks = (int)((waves + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK);
-    hipLaunchKernelGGL(k_fold3, dim3(blocks), dim3(BLOCK), 0, s, d_gval, d_gridx,
+    hipLaunchKernelGGL(k_fold3, dim3(blocks), dim3(BLOCK), 0, s, d_gfrec,
                        d_bucket_base, d_binoffs, d_binlens, fc, d_slot_of_widx,
                        s_cnt, s_min, s_max, s_sum, s_first);
 }

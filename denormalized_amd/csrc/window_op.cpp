@@ -146,18 +146,15 @@ struct dz_window_op {
     uint32_t* d_base = nullptr;
     int C_cap = 0;
     uint32_t* d_meta = nullptr;
-    uint32_t* d_ridx = nullptr;
-    double* d_rvals = nullptr;
-    double* d_gval = nullptr;    /* group-segmented values (regroup output) */
-    uint32_t* d_gridx = nullptr; /* rowidx | valid<<31, segmented */
+    uint4* d_grec = nullptr;  /* 16 B {val, rowidx} records (scatter output) */
+    uint4* d_gfrec = nullptr; /* group-segmented fold records */
     uint32_t* d_binoffs = nullptr;
     uint32_t* d_binlens = nullptr;
     int64_t bin_cap = 0;   /* bins per bucket the binoffs/lens arrays hold */
     uint32_t* d_b1offs = nullptr; /* two-level L1 segment layout [NB][256] */
     uint32_t* d_b1lens = nullptr;
     uint32_t* d_meta2 = nullptr;  /* two-level intermediate records */
-    uint32_t* d_ridx2 = nullptr;
-    double* d_rvals2 = nullptr;
+    uint4* d_grec2 = nullptr;
     int64_t l2_cap = 0;
     int64_t rec_cap = 0;
     uint64_t* d_scalars = nullptr;
@@ -439,7 +436,7 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_gval); hipFree(op->d_gridx);
     hipFree(op->d_binoffs); hipFree(op->d_binlens);
     hipFree(op->d_b1offs); hipFree(op->d_b1lens);
-    hipFree(op->d_meta2); hipFree(op->d_ridx2); hipFree(op->d_rvals2);
+    hipFree(op->d_meta2); hipFree(op->d_grec2);
     hipFree(op->d_scalars); hipFree(op->d_slotmap); hipFree(op->d_zero_kid);
     hipFree(op->d_resetlist);
     hipFree(op->d_ts); hipFree(op->d_kid); hipFree(op->d_vals); hipFree(op->d_valbm);
@@ -1017,13 +1014,10 @@ static dz_status ensure_scratch(dz_window_op* op, int C, int64_t nrec) {
         op->C_cap = C;
     }
     if (nrec > op->rec_cap) {
-        hipFree(op->d_meta); hipFree(op->d_ridx); hipFree(op->d_rvals);
-        hipFree(op->d_gval); hipFree(op->d_gridx);
+        hipFree(op->d_meta); hipFree(op->d_grec); hipFree(op->d_gfrec);
         CHK(op, hipMalloc(&op->d_meta, (size_t)nrec * 4));
-        CHK(op, hipMalloc(&op->d_ridx, (size_t)nrec * 4));
-        CHK(op, hipMalloc(&op->d_rvals, (size_t)nrec * 8));
-        CHK(op, hipMalloc(&op->d_gval, (size_t)nrec * 8));
-        CHK(op, hipMalloc(&op->d_gridx, (size_t)nrec * 4));
+        CHK(op, hipMalloc(&op->d_grec, (size_t)nrec * 16));
+        CHK(op, hipMalloc(&op->d_gfrec, (size_t)nrec * 16));
         op->rec_cap = nrec;
     }
     return DZ_OK;
@@ -1207,7 +1201,7 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
             CHK(op, hipMalloc(&op->d_b1lens, (size_t)dz::NB * 256 * 4));
         }
         if (nrec_max > op->l2_cap) {
-            hipFree(op->d_meta2); hipFree(op->d_ridx2); hipFree(op->d_rvals2);
+            hipFree(op->d_meta2); hipFree(op->d_grec2);
             CHK(op, hipMalloc(&op->d_meta2, (size_t)nrec_max * 4));
             CHK(op, hipMalloc(&op->d_ridx2, (size_t)nrec_max * 4));
             CHK(op, hipMalloc(&op->d_rvals2, (size_t)nrec_max * 8));
@@ -1222,23 +1216,21 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
         fc.batch_seq = op->batch_seq;
         fc.bin_stride = (int32_t)stride;
         fc.tl_nw = (int32_t)nw;
-        timed(op, "regroup", (double)nrec_max * 36, [&] {
-            dz::launch_regroup_l1(op->stream, op->d_meta, op->d_ridx,
-                                  op->d_rvals, op->d_base, fc, op->d_b1offs,
-                                  op->d_b1lens, op->d_meta2, op->d_ridx2,
-                                  op->d_rvals2);
+        timed(op, "regroup", (double)nrec_max * 40, [&] {
+            dz::launch_regroup_l1(op->stream, op->d_meta, op->d_grec, op->d_base,
+                                  fc, op->d_b1offs, op->d_b1lens, op->d_meta2,
+                                  op->d_grec2);
         });
-        timed(op, "regroup", (double)nrec_max * 36, [&] {
-            dz::launch_regroup_l2(op->stream, op->d_meta2, op->d_ridx2,
-                                  op->d_rvals2, op->d_base, fc, (int)nb1,
-                                  op->d_b1offs, op->d_b1lens, op->d_binoffs,
-                                  op->d_binlens, op->d_gval, op->d_gridx);
+        timed(op, "regroup", (double)nrec_max * 40, [&] {
+            dz::launch_regroup_l2(op->stream, op->d_meta2, op->d_grec2,
+                                  op->d_base, fc, (int)nb1, op->d_b1offs,
+                                  op->d_b1lens, op->d_binoffs, op->d_binlens,
+                                  op->d_gfrec);
         });
-        timed(op, "fold", (double)nrec_max * 12, [&] {
-            dz::launch_fold3(op->stream, op->d_gval, op->d_gridx, op->d_base,
-                             op->d_binoffs, op->d_binlens, fc, op->d_slotmap,
-                             op->s_cnt, op->s_min, op->s_max, op->s_sum,
-                             op->s_first);
+        timed(op, "fold", (double)nrec_max * 16, [&] {
+            dz::launch_fold3(op->stream, op->d_gfrec, op->d_base, op->d_binoffs,
+                             op->d_binlens, fc, op->d_slotmap, op->s_cnt,
+                             op->s_min, op->s_max, op->s_sum, op->s_first);
         });
     } else {
         if (ensure_bins(op, dz::FOLD_GCAP) != DZ_OK) return DZ_ERR;
@@ -1255,14 +1247,14 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
                 fc.batch_seq = op->batch_seq;
                 fc.bin_stride = dz::FOLD_GCAP;
                 fc.tl_nw = 0;
-                timed(op, "regroup", (double)nrec_max * 32, [&] {
-                    dz::launch_regroup(op->stream, op->d_meta, op->d_ridx,
-                                       op->d_rvals, op->d_base, fc, op->d_binoffs,
-                                       op->d_binlens, op->d_gval, op->d_gridx);
+                timed(op, "regroup", (double)nrec_max * 40, [&] {
+                    dz::launch_regroup(op->stream, op->d_meta, op->d_grec,
+                                       op->d_base, fc, op->d_binoffs,
+                                       op->d_binlens, op->d_gfrec);
                 });
-                timed(op, "fold", (double)nrec_max * 12, [&] {
-                    dz::launch_fold3(op->stream, op->d_gval, op->d_gridx,
-                                     op->d_base, op->d_binoffs, op->d_binlens, fc,
+                timed(op, "fold", (double)nrec_max * 16, [&] {
+                    dz::launch_fold3(op->stream, op->d_gfrec, op->d_base,
+                                     op->d_binoffs, op->d_binlens, fc,
                                      op->d_slotmap, op->s_cnt, op->s_min,
                                      op->s_max, op->s_sum, op->s_first);
                 });
