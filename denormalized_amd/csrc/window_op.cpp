@@ -432,8 +432,7 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_roffs);
     hipFree(op->s_base);
     hipFree(op->d_ghist); hipFree(op->d_gofs); hipFree(op->d_total); hipFree(op->d_base);
-    hipFree(op->d_meta); hipFree(op->d_ridx); hipFree(op->d_rvals);
-    hipFree(op->d_gval); hipFree(op->d_gridx);
+    hipFree(op->d_meta); hipFree(op->d_grec); hipFree(op->d_gfrec);
     hipFree(op->d_binoffs); hipFree(op->d_binlens);
     hipFree(op->d_b1offs); hipFree(op->d_b1lens);
     hipFree(op->d_meta2); hipFree(op->d_grec2);
@@ -1181,10 +1180,9 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
         });
     }
     int32_t st_rows = (int32_t)std::max<int64_t>(64, dz::ST_RECORDS / expand);
-    timed(op, "scatter", (double)n * 24 + (double)nrec_max * 16, [&] {
+    timed(op, "scatter", (double)n * 24 + (double)nrec_max * 20, [&] {
         dz::launch_scatter(op->stream, d_kid, d_ts, d_vals, d_valbm, n, chunk, C,
-                           st_rows, wp, op->d_gofs, op->d_meta, op->d_ridx,
-                           op->d_rvals);
+                           st_rows, wp, op->d_gofs, op->d_meta, op->d_grec);
     });
     int64_t klocs = op->kcap >> dz::LOG_NB;
     int64_t gtot = klocs * nw;
@@ -1203,8 +1201,7 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
         if (nrec_max > op->l2_cap) {
             hipFree(op->d_meta2); hipFree(op->d_grec2);
             CHK(op, hipMalloc(&op->d_meta2, (size_t)nrec_max * 4));
-            CHK(op, hipMalloc(&op->d_ridx2, (size_t)nrec_max * 4));
-            CHK(op, hipMalloc(&op->d_rvals2, (size_t)nrec_max * 8));
+            CHK(op, hipMalloc(&op->d_grec2, (size_t)nrec_max * 16));
             op->l2_cap = nrec_max;
         }
         dz::FoldChunk fc;
