@@ -254,3 +254,31 @@ def test_golden_fixtures():
         got = rows(out)
         exp = [tuple(r) for r in case["expected"]]
         assert [tuple(g) for g in got] == exp, case["name"]
+
+
+def test_bench_generator_distribution_crosscheck():
+    # oracle vs independent restatement on the BENCH's generator distribution
+    # (1M rows, cfg2-shaped), batched like the bench pushes
+    ts, kid, val = pyoracle.gen(42, 1_000_000, 0, 1_000_000, 10_000, 1000)
+    o = pyoracle.Oracle(1000, 0)
+    p = PyRef(1000, 0)
+    step = 250_000
+    for lo in range(0, 1_000_000, step):
+        sl = slice(lo, lo + step)
+        o.push(ts[sl], kid[sl], val[sl])
+        p.push(ts[sl].tolist(), kid[sl].tolist(), val[sl].tolist())
+    o.finish()
+    p.finish()
+    got = o.fetch()
+    assert len(got["key"]) == len(p.out) > 9000
+    for i in (0, 1, len(p.out) // 2, len(p.out) - 1):
+        row = p.out[i]
+        assert int(got["key"][i]) == row[0]
+        assert int(got["count"][i]) == row[1]
+        assert float(got["avg"][i]) == row[4]
+    # full-array equality
+    import numpy as np
+    assert np.array_equal(got["key"], np.array([r[0] for r in p.out]))
+    assert np.array_equal(got["count"], np.array([r[1] for r in p.out]))
+    assert np.array_equal(got["sum"], np.array([r[5] for r in p.out]))
+    o.close()
