@@ -139,8 +139,10 @@ struct dz_window_op {
     bool has_wm = false;
     uint32_t batch_seq = 0;
 
-    /* scratch */
-    uint32_t* d_ghist = nullptr;
+    /* scratch (d_ghist/d_scalars double-buffered: the deferred device-push
+     * pipeline runs batch N's reduction on i_stream while batch N-1's scan
+     * still reads the other buffer on the compute stream) */
+    uint32_t* d_ghist[2] = {};
     uint32_t* d_gofs = nullptr;
     uint32_t* d_total = nullptr;
     uint32_t* d_base = nullptr;
@@ -157,8 +159,8 @@ struct dz_window_op {
     uint4* d_grec2 = nullptr;
     int64_t l2_cap = 0;
     int64_t rec_cap = 0;
-    uint64_t* d_scalars = nullptr;
-    uint64_t* h_scalars = nullptr; /* pinned, 3 */
+    uint64_t* d_scalars[2] = {};
+    uint64_t* h_scalars = nullptr; /* pinned, 2 x 3 (per pipeline buffer) */
     int32_t* d_slotmap = nullptr;
     int slotmap_cap = 0;
     int32_t* d_zero_kid = nullptr; /* no_group: all rows in group 0 */
@@ -174,6 +176,33 @@ struct dz_window_op {
     int64_t in_cap = 0;
     char* h_stage = nullptr; /* pinned */
     size_t h_stage_cap = 0;
+
+    /* deferred device-push pipeline: push_device enqueues input staging plus
+     * the min/max(/hist) reduction on i_stream and returns; the window
+     * decisions and partition/fold launches run at the NEXT call into the op
+     * (or at a poll once the reduction is host-visible), so the host's
+     * decision wait overlaps the previous step's device execution instead of
+     * serializing behind it. */
+    hipStream_t i_stream = nullptr;
+    struct Pend {
+        bool active = false, sliding = false, dense = false, deferred = false;
+        int buf = 0;
+        int C = 0;
+        int64_t n = 0, chunk = 0;
+        const int64_t* ts = nullptr;
+        const int32_t* kid = nullptr;
+        const double* vals = nullptr;
+        const uint8_t* valbm = nullptr;
+    } pend;
+    int next_buf = 0;
+    int64_t* d_sts[2] = {};   /* op-owned device input ring (device-push) */
+    int32_t* d_skid[2] = {};
+    double* d_svals[2] = {};
+    int64_t s_in_cap[2] = {0, 0};
+    hipEvent_t ev_ready[2] = {};    /* scalars D2H host-visible */
+    hipEvent_t ev_staged[2] = {};   /* caller inputs copied into the ring */
+    hipEvent_t ev_consumed[2] = {}; /* scatter that read buffer b completed */
+    bool consumed_valid[2] = {false, false};
 
     /* async DEVICE-SIDE emission: at window close the copy stream runs
      * compact -> stable radix sort by first-seen row -> gather+filter
@@ -247,6 +276,7 @@ struct dz_window_op {
 
 static void emit_worker_main(dz_window_op* op);
 static dz_status ensure_emission(dz_window_op* op);
+static dz_status process_pending(dz_window_op* op);
 
 static hipEvent_t get_event(dz_window_op* op) {
     if (!op->ev_pool.empty()) {
@@ -287,12 +317,18 @@ struct HostTimer {
 };
 
 template <typename F>
-static void timed(dz_window_op* op, const char* name, double bytes, F&& fn) {
+static void timed_on(dz_window_op* op, hipStream_t s, const char* name,
+                     double bytes, F&& fn) {
     hipEvent_t a = get_event(op), b = get_event(op);
-    hipEventRecord(a, op->stream);
+    hipEventRecord(a, s);
     fn();
-    hipEventRecord(b, op->stream);
+    hipEventRecord(b, s);
     op->pending.push_back({name, a, b, bytes});
+}
+
+template <typename F>
+static void timed(dz_window_op* op, const char* name, double bytes, F&& fn) {
+    timed_on(op, op->stream, name, bytes, std::forward<F>(fn));
 }
 
 HostTimer::~HostTimer() {
@@ -389,8 +425,15 @@ extern "C" dz_window_op* dz_window_op_create(const dz_window_desc* desc) {
         delete op;
         return nullptr;
     }
-    hipMalloc(&op->d_scalars, 3 * 8);
-    hipHostMalloc((void**)&op->h_scalars, 3 * 8);
+    hipMalloc(&op->d_scalars[0], 3 * 8);
+    hipMalloc(&op->d_scalars[1], 3 * 8);
+    hipHostMalloc((void**)&op->h_scalars, 2 * 3 * 8);
+    hipStreamCreateWithFlags(&op->i_stream, hipStreamNonBlocking);
+    for (int i = 0; i < 2; i++) {
+        hipEventCreateWithFlags(&op->ev_ready[i], hipEventDisableTiming);
+        hipEventCreateWithFlags(&op->ev_staged[i], hipEventDisableTiming);
+        hipEventCreateWithFlags(&op->ev_consumed[i], hipEventDisableTiming);
+    }
     int64_t hint = std::max<int64_t>(desc->n_keys_hint, 1);
     if (state_alloc(op, hint, 4) != DZ_OK) {
         g_err = op->err;
@@ -412,7 +455,9 @@ extern "C" dz_window_op* dz_window_op_create(const dz_window_desc* desc) {
 extern "C" void dz_window_op_destroy(dz_window_op* op) {
     if (!op) return;
     hipSetDevice(op->device);
+    process_pending(op); /* flush any deferred device push (best effort) */
     hipStreamSynchronize(op->stream);
+    if (op->i_stream) hipStreamSynchronize(op->i_stream);
     {
         std::lock_guard<std::mutex> lk(op->e_mtx);
         op->e_stop = true;
@@ -431,18 +476,27 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_rhist);
     hipFree(op->d_roffs);
     hipFree(op->s_base);
-    hipFree(op->d_ghist); hipFree(op->d_gofs); hipFree(op->d_total); hipFree(op->d_base);
+    hipFree(op->d_ghist[0]); hipFree(op->d_ghist[1]);
+    hipFree(op->d_gofs); hipFree(op->d_total); hipFree(op->d_base);
     hipFree(op->d_meta); hipFree(op->d_grec); hipFree(op->d_gfrec);
     hipFree(op->d_binoffs); hipFree(op->d_binlens);
     hipFree(op->d_b1offs); hipFree(op->d_b1lens);
     hipFree(op->d_meta2); hipFree(op->d_grec2);
-    hipFree(op->d_scalars); hipFree(op->d_slotmap); hipFree(op->d_zero_kid);
+    hipFree(op->d_scalars[0]); hipFree(op->d_scalars[1]);
+    hipFree(op->d_slotmap); hipFree(op->d_zero_kid);
     hipFree(op->d_resetlist);
     hipFree(op->d_ts); hipFree(op->d_kid); hipFree(op->d_vals); hipFree(op->d_valbm);
+    for (int i = 0; i < 2; i++) {
+        hipFree(op->d_sts[i]); hipFree(op->d_skid[i]); hipFree(op->d_svals[i]);
+        if (op->ev_ready[i]) hipEventDestroy(op->ev_ready[i]);
+        if (op->ev_staged[i]) hipEventDestroy(op->ev_staged[i]);
+        if (op->ev_consumed[i]) hipEventDestroy(op->ev_consumed[i]);
+    }
     if (op->h_scalars) hipHostFree(op->h_scalars);
     if (op->h_stage) hipHostFree(op->h_stage);
     hipStreamDestroy(op->stream);
     hipStreamDestroy(op->copy_stream);
+    if (op->i_stream) hipStreamDestroy(op->i_stream);
     for (auto& fs : op->free_slots)
         if (fs.ev) hipEventDestroy(fs.ev);
     delete op;
@@ -1001,10 +1055,20 @@ static dz_status trigger_windows(dz_window_op* op) {
 /* the push core (device-resident inputs)                              */
 /* ------------------------------------------------------------------ */
 
+/* Growth paths free device buffers that in-flight kernels of the previous
+ * (pipelined) step may still read: quiesce both streams first. hipFree's
+ * implicit device sync would cover it, but growth is rare — be explicit. */
+static void quiesce(dz_window_op* op) {
+    hipStreamSynchronize(op->stream);
+    if (op->i_stream) hipStreamSynchronize(op->i_stream);
+}
+
 static dz_status ensure_scratch(dz_window_op* op, int C, int64_t nrec) {
     if (C > op->C_cap) {
-        hipFree(op->d_ghist); hipFree(op->d_gofs);
-        CHK(op, hipMalloc(&op->d_ghist, (size_t)C * dz::NB * 4));
+        quiesce(op);
+        hipFree(op->d_ghist[0]); hipFree(op->d_ghist[1]); hipFree(op->d_gofs);
+        CHK(op, hipMalloc(&op->d_ghist[0], (size_t)C * dz::NB * 4));
+        CHK(op, hipMalloc(&op->d_ghist[1], (size_t)C * dz::NB * 4));
         CHK(op, hipMalloc(&op->d_gofs, (size_t)C * dz::NB * 4));
         if (!op->d_total) {
             CHK(op, hipMalloc(&op->d_total, (size_t)dz::SCAN_SSPLIT * dz::NB * 4));
@@ -1013,6 +1077,7 @@ static dz_status ensure_scratch(dz_window_op* op, int C, int64_t nrec) {
         op->C_cap = C;
     }
     if (nrec > op->rec_cap) {
+        quiesce(op);
         hipFree(op->d_meta); hipFree(op->d_grec); hipFree(op->d_gfrec);
         CHK(op, hipMalloc(&op->d_meta, (size_t)nrec * 4));
         CHK(op, hipMalloc(&op->d_grec, (size_t)nrec * 16));
@@ -1033,26 +1098,63 @@ static dz_status ensure_bins(dz_window_op* op, int64_t bins_per_bucket) {
     return DZ_OK;
 }
 
-static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
-                           const int32_t* d_kid, const double* d_vals,
-                           const uint8_t* d_valbm, bool keys_are_dense) {
-    if (n <= 0) return DZ_OK; /* empty batch: reference emits empty (no-op) */
+/* Section 1 of a push: stage inputs and run the reduction whose scalars
+ * (batch min/max ts, max key id) the host must read before it can lay out
+ * window frames. deferred=true (device-push pipeline) runs it on i_stream
+ * against an op-owned staging copy of the inputs, records ev_ready, and
+ * leaves the rest for process_pending() at the next call into the op;
+ * deferred=false (host-batch path) runs in-stream on the compute stream.
+ * Tumbling fast path: one fused 12 B/row pass computes the histogram AND
+ * min/max ts + max key id (the histogram needs no window params when every
+ * row has multiplicity 1). Sliding needs min/max first (the window grid
+ * anchors multiplicity), so it keeps a separate reduction pass. */
+static dz_status stage_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
+                            const int32_t* d_kid, const double* d_vals,
+                            const uint8_t* d_valbm, bool keys_are_dense,
+                            bool deferred) {
     if (n >= (1LL << 31)) {
         op->err = "batch exceeds 2^31 rows; push smaller batches";
         return DZ_ERR;
     }
-    CHK(op, hipSetDevice(op->device));
-
-    /* 1. batch watermark bounds + bucket histogram.
-     * Tumbling fast path: one fused 12 B/row pass computes the histogram AND
-     * min/max ts + max key id (the histogram needs no window params when
-     * every row has multiplicity 1). Sliding needs min/max first (the window
-     * grid anchors multiplicity), so it keeps a separate reduction pass. */
     const bool sliding = op->slide_ms > 0;
     int C = (int)std::min<int64_t>(512, std::max<int64_t>(1, (n + 8191) / 8192));
     int64_t chunk = (n + C - 1) / C;
-    CHK(op, hipMemsetAsync(op->d_scalars, 0xFF, 8, op->stream));
-    CHK(op, hipMemsetAsync(op->d_scalars + 1, 0, 16, op->stream));
+    int b = op->next_buf;
+    op->next_buf ^= 1;
+    hipStream_t s = deferred ? op->i_stream : op->stream;
+    if (deferred) {
+        HostTimer ht(op, "h_stage");
+        if (n > op->s_in_cap[b]) {
+            quiesce(op);
+            hipFree(op->d_sts[b]); hipFree(op->d_skid[b]); hipFree(op->d_svals[b]);
+            CHK(op, hipMalloc(&op->d_sts[b], (size_t)n * 8));
+            CHK(op, hipMalloc(&op->d_skid[b], (size_t)n * 4));
+            CHK(op, hipMalloc(&op->d_svals[b], (size_t)n * 8));
+            op->s_in_cap[b] = n;
+            op->consumed_valid[b] = false;
+        }
+        /* buffer b was last read by the scatter two pushes ago */
+        if (op->consumed_valid[b])
+            CHK(op, hipStreamWaitEvent(s, op->ev_consumed[b], 0));
+        CHK(op, hipMemcpyAsync(op->d_sts[b], d_ts, (size_t)n * 8,
+                               hipMemcpyDeviceToDevice, s));
+        CHK(op, hipMemcpyAsync(op->d_skid[b], d_kid, (size_t)n * 4,
+                               hipMemcpyDeviceToDevice, s));
+        CHK(op, hipMemcpyAsync(op->d_svals[b], d_vals, (size_t)n * 8,
+                               hipMemcpyDeviceToDevice, s));
+        CHK(op, hipEventRecord(op->ev_staged[b], s));
+        d_ts = op->d_sts[b];
+        d_kid = op->d_skid[b];
+        d_vals = op->d_svals[b];
+    } else if (op->consumed_valid[b]) {
+        /* host-batch path interleaved with device pushes: the compute stream
+         * must not overwrite d_ghist[b]/d_scalars[b] before the i_stream
+         * reduction that last wrote them was consumed — in-stream ordering
+         * covers compute-stream reuse, this wait covers the cross-stream case */
+        CHK(op, hipStreamWaitEvent(s, op->ev_consumed[b], 0));
+    }
+    CHK(op, hipMemsetAsync(op->d_scalars[b], 0xFF, 8, s));
+    CHK(op, hipMemsetAsync(op->d_scalars[b] + 1, 0, 16, s));
     dz::WinParams wp;
     memset(&wp, 0, sizeof(wp));
     wp.len_ms = op->len_ms;
@@ -1060,36 +1162,84 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
     wp.is_sliding = sliding;
     if (!sliding) {
         if (ensure_scratch(op, C, n) != DZ_OK) return DZ_ERR;
-        timed(op, "hist", (double)n * 12, [&] {
-            dz::launch_hist(op->stream, d_kid, d_ts, n, chunk, C, wp,
-                            op->d_ghist, op->d_scalars);
-        });
-        timed(op, "scan", (double)C * dz::NB * 12, [&] {
-            dz::launch_scan(op->stream, op->d_ghist, C, op->d_total, op->d_base,
-                            op->d_gofs);
+        timed_on(op, s, "hist", (double)n * 12, [&] {
+            dz::launch_hist(s, d_kid, d_ts, n, chunk, C, wp,
+                            op->d_ghist[b], op->d_scalars[b]);
         });
     } else {
-        timed(op, "minmax", (double)n * 12, [&] {
-            dz::launch_minmax(op->stream, d_ts, keys_are_dense ? d_kid : nullptr,
-                              n, op->d_scalars);
+        timed_on(op, s, "minmax", (double)n * 12, [&] {
+            dz::launch_minmax(s, d_ts, keys_are_dense ? d_kid : nullptr,
+                              n, op->d_scalars[b]);
         });
     }
-    CHK(op, hipMemcpyAsync(op->h_scalars, op->d_scalars, 24, hipMemcpyDeviceToHost,
-                           op->stream));
+    CHK(op, hipMemcpyAsync(op->h_scalars + 3 * b, op->d_scalars[b], 24,
+                           hipMemcpyDeviceToHost, s));
+    CHK(op, hipEventRecord(op->ev_ready[b], s));
+    op->pend.active = true;
+    op->pend.sliding = sliding;
+    op->pend.dense = keys_are_dense;
+    op->pend.deferred = deferred;
+    op->pend.buf = b;
+    op->pend.C = C;
+    op->pend.n = n;
+    op->pend.chunk = chunk;
+    op->pend.ts = d_ts;
+    op->pend.kid = d_kid;
+    op->pend.vals = d_vals;
+    op->pend.valbm = d_valbm;
+    if (deferred) {
+        /* the caller may release its input buffers once we return: wait for
+         * the ring copies (typically tens of µs — the ring slot is idle) */
+        HostTimer ht(op, "h_stage");
+        event_spin(op->ev_staged[b]);
+    }
+    return DZ_OK;
+}
+
+/* Sections 2-4 of a push: read the staged scalars, lay out window frames,
+ * partition + fold, advance the watermark and trigger closes. */
+static dz_status process_pending(dz_window_op* op) {
+    if (!op->pend.active) return DZ_OK;
+    CHK(op, hipSetDevice(op->device));
+    dz_window_op::Pend P = op->pend;
+    op->pend.active = false;
+    const bool sliding = P.sliding;
+    const bool keys_are_dense = P.dense;
+    const int64_t n = P.n, chunk = P.chunk;
+    const int C = P.C;
+    const int b = P.buf;
+    const int64_t* d_ts = P.ts;
+    const int32_t* d_kid = P.kid;
+    const double* d_vals = P.vals;
+    const uint8_t* d_valbm = P.valbm;
     {
         HostTimer ht(op, "h_minmax_sync");
-        CHK(op, hipStreamSynchronize(op->stream));
+        event_spin(op->ev_ready[b]);
     }
+    /* compute-stream work below consumes i_stream-written buffers */
+    CHK(op, hipStreamWaitEvent(op->stream, op->ev_ready[b], 0));
     drain_events(op, false);
-    int64_t mn = (int64_t)(op->h_scalars[0] ^ 0x8000000000000000ULL);
-    int64_t mx = (int64_t)(op->h_scalars[1] ^ 0x8000000000000000ULL);
+    dz::WinParams wp;
+    memset(&wp, 0, sizeof(wp));
+    wp.len_ms = op->len_ms;
+    wp.slide_ms = op->slide_ms;
+    wp.is_sliding = sliding;
+    const uint64_t* hs = op->h_scalars + 3 * b;
+    int64_t mn = (int64_t)(hs[0] ^ 0x8000000000000000ULL);
+    int64_t mx = (int64_t)(hs[1] ^ 0x8000000000000000ULL);
     if (mn < 0) {
         op->err = "negative timestamps not supported (reference SystemTime panics too)";
         return DZ_ERR;
     }
     if (keys_are_dense) {
-        int64_t kmax = (int64_t)op->h_scalars[2];
+        int64_t kmax = (int64_t)hs[2];
         op->n_keys = std::max(op->n_keys, kmax + 1);
+    }
+    if (!sliding) {
+        timed(op, "scan", (double)C * dz::NB * 12, [&] {
+            dz::launch_scan(op->stream, op->d_ghist[b], C, op->d_total,
+                            op->d_base, op->d_gofs);
+        });
     }
     if (op->n_keys > op->kcap) {
         if (state_alloc(op, std::max(op->n_keys, op->kcap * 2), op->nslots) != DZ_OK)
@@ -1172,11 +1322,11 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
     if (sliding) {
         timed(op, "hist", (double)n * 12, [&] {
             dz::launch_hist(op->stream, d_kid, d_ts, n, chunk, C, wp,
-                            op->d_ghist, nullptr);
+                            op->d_ghist[b], nullptr);
         });
         timed(op, "scan", (double)C * dz::NB * 12, [&] {
-            dz::launch_scan(op->stream, op->d_ghist, C, op->d_total, op->d_base,
-                            op->d_gofs);
+            dz::launch_scan(op->stream, op->d_ghist[b], C, op->d_total,
+                            op->d_base, op->d_gofs);
         });
     }
     int32_t st_rows = (int32_t)std::max<int64_t>(64, dz::ST_RECORDS / expand);
@@ -1184,6 +1334,10 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
         dz::launch_scatter(op->stream, d_kid, d_ts, d_vals, d_valbm, n, chunk, C,
                            st_rows, wp, op->d_gofs, op->d_meta, op->d_grec);
     });
+    /* scatter is the last reader of the staged inputs and of d_ghist[b]:
+     * once it completes, pipeline buffer b may be restaged */
+    CHK(op, hipEventRecord(op->ev_consumed[b], op->stream));
+    op->consumed_valid[b] = true;
     int64_t klocs = op->kcap >> dz::LOG_NB;
     int64_t gtot = klocs * nw;
     int64_t khigh = (klocs + 255) >> 8;
@@ -1268,11 +1422,26 @@ static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
     return trigger_windows(op);
 }
 
+/* Synchronous push: flush any pending device-push, stage this batch on the
+ * compute stream, process it before returning (host-batch semantics). */
+static dz_status push_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
+                           const int32_t* d_kid, const double* d_vals,
+                           const uint8_t* d_valbm, bool keys_are_dense) {
+    if (process_pending(op) != DZ_OK) return DZ_ERR;
+    if (n <= 0) return DZ_OK; /* empty batch: reference emits empty (no-op) */
+    if (stage_core(op, n, d_ts, d_kid, d_vals, d_valbm, keys_are_dense,
+                   /*deferred=*/false) != DZ_OK)
+        return DZ_ERR;
+    return process_pending(op);
+}
+
 static dz_status ensure_zero_kid(dz_window_op* op, int64_t n) {
     if (n > op->zero_cap) {
         hipFree(op->d_zero_kid);
         CHK(op, hipMalloc(&op->d_zero_kid, (size_t)n * 4));
         CHK(op, hipMemsetAsync(op->d_zero_kid, 0, (size_t)n * 4, op->stream));
+        /* i_stream reductions read d_zero_kid: make the fill visible there */
+        CHK(op, hipStreamSynchronize(op->stream));
         op->zero_cap = n;
     }
     return DZ_OK;
@@ -1283,12 +1452,18 @@ extern "C" dz_status dz_window_op_push_device(dz_window_op* op, int64_t n_rows,
                                               const int32_t* d_key_ids,
                                               const double* d_vals) {
     if (!op) return DZ_ERR;
+    CHK(op, hipSetDevice(op->device));
     if (op->no_group || !d_key_ids) {
-        CHK(op, hipSetDevice(op->device));
         if (ensure_zero_kid(op, n_rows) != DZ_OK) return DZ_ERR;
         d_key_ids = op->d_zero_kid;
     }
-    return push_core(op, n_rows, d_ts_ms, d_key_ids, d_vals, nullptr, true);
+    /* deferred pipeline: process the previous device push (its reduction has
+     * long overlapped our caller's time between pushes), then stage this one
+     * and return — its heavy phase runs at the next call into the op. */
+    if (process_pending(op) != DZ_OK) return DZ_ERR;
+    if (n_rows <= 0) return DZ_OK;
+    return stage_core(op, n_rows, d_ts_ms, d_key_ids, d_vals, nullptr,
+                      /*keys_are_dense=*/true, /*deferred=*/true);
 }
 
 /* ------------------------------------------------------------------ */
@@ -1405,6 +1580,15 @@ extern "C" dz_status dz_window_op_push(dz_window_op* op, const dz_batch* batch) 
 extern "C" dz_status dz_window_op_poll(dz_window_op* op, const dz_out_batch** out) {
     if (!op || !out) return DZ_ERR;
     *out = nullptr;
+    if (op->pend.active) {
+        /* opportunistic, non-blocking: finish a deferred device push whose
+         * reduction is already host-visible (keeps a poll loop live without
+         * ever stalling the pipelined caller) */
+        hipSetDevice(op->device);
+        if (hipEventQuery(op->ev_ready[op->pend.buf]) == hipSuccess &&
+            process_pending(op) != DZ_OK)
+            return DZ_ERR;
+    }
     {
         std::lock_guard<std::mutex> lk(op->out_mtx);
         if (op->outq.empty()) return DZ_OK;
@@ -1434,6 +1618,7 @@ extern "C" dz_status dz_window_op_poll(dz_window_op* op, const dz_out_batch** ou
 extern "C" dz_status dz_window_op_finish(dz_window_op* op) {
     if (!op) return DZ_ERR;
     CHK(op, hipSetDevice(op->device));
+    if (process_pending(op) != DZ_OK) return DZ_ERR;
     int64_t mx = op->has_wm ? op->watermark : INT64_MIN;
     for (auto& kv : op->open) mx = std::max(mx, kv.second.end);
     if (mx != INT64_MIN) {
@@ -1449,6 +1634,10 @@ extern "C" dz_status dz_window_op_finish(dz_window_op* op) {
 
 extern "C" dz_status dz_window_op_drain(dz_window_op* op) {
     if (!op) return DZ_ERR;
+    if (op->pend.active) {
+        CHK(op, hipSetDevice(op->device));
+        if (process_pending(op) != DZ_OK) return DZ_ERR;
+    }
     emit_drain(op);
     return DZ_OK;
 }
@@ -1456,6 +1645,9 @@ extern "C" dz_status dz_window_op_drain(dz_window_op* op) {
 extern "C" dz_status dz_window_op_advance_watermark(dz_window_op* op, int64_t wm) {
     if (!op) return DZ_ERR;
     CHK(op, hipSetDevice(op->device));
+    /* the external watermark may close windows the deferred batch still has
+     * rows for: fold it in first */
+    if (process_pending(op) != DZ_OK) return DZ_ERR;
     if (!op->has_wm || op->watermark <= wm) {
         op->watermark = wm;
         op->has_wm = true;
@@ -1468,12 +1660,21 @@ extern "C" int64_t dz_window_op_watermark(dz_window_op* op) {
 }
 
 extern "C" int64_t dz_window_op_open_windows(dz_window_op* op) {
-    return op ? (int64_t)op->open.size() : 0;
+    if (!op) return 0;
+    if (op->pend.active) {
+        hipSetDevice(op->device);
+        process_pending(op);
+    }
+    return (int64_t)op->open.size();
 }
 
 extern "C" dz_status dz_window_op_set_filter(dz_window_op* op, int32_t agg_idx,
                                              int32_t cmp, double literal) {
     if (!op) return DZ_ERR;
+    if (op->pend.active) { /* keep filter-applies-at-trigger-time semantics */
+        CHK(op, hipSetDevice(op->device));
+        if (process_pending(op) != DZ_OK) return DZ_ERR;
+    }
     if (agg_idx < 0 || agg_idx >= (int32_t)op->aggs.size()) {
         op->err = "filter agg index out of range";
         return DZ_ERR;
@@ -1489,6 +1690,8 @@ extern "C" dz_status dz_window_op_kernel_stats(dz_window_op* op,
                                                dz_kernel_stat* out, int32_t cap,
                                                int32_t* n_out) {
     if (!op) return DZ_ERR;
+    CHK(op, hipSetDevice(op->device));
+    if (process_pending(op) != DZ_OK) return DZ_ERR;
     emit_drain(op);
     CHK(op, hipStreamSynchronize(op->stream));
     drain_events(op, true);

@@ -138,10 +138,16 @@ dz_window_op* dz_window_op_create(const dz_window_desc* desc);
  * shared watermark, triggers closed windows. */
 dz_status dz_window_op_push(dz_window_op* op, const dz_batch* batch);
 
-/* Device-resident push: same semantics as dz_window_op_push but the three
+/* Device-resident push: same results as dz_window_op_push but the three
  * buffers already live in the op's device HBM (keys as dense int32 ids).
  * This is the bench's timed entry (inputs resident per measurement contract);
- * the host-buffer path above is the reference-shaped boundary. */
+ * the host-buffer path above is the reference-shaped boundary.
+ * Pipelined: the call stages the inputs (they may be freed once it returns)
+ * and enqueues the batch's reduction, then returns; routing/aggregation and
+ * any window closes complete at the NEXT call into the op (push/poll/
+ * advance_watermark/finish/destroy all flush — poll opportunistically, the
+ * rest unconditionally). Emitted batches and the local watermark may
+ * therefore lag one device-pushed batch; results are identical. */
 dz_status dz_window_op_push_device(dz_window_op* op, int64_t n_rows,
                                    const int64_t* d_ts_ms,
                                    const int32_t* d_key_ids,
