@@ -208,7 +208,12 @@ struct dz_window_op {
      * compact -> stable radix sort by first-seen row -> gather+filter
      * (kernels.hip launch_emission); a worker thread then copies exactly
      * nt sorted rows D2H and formats them off the push critical path. */
-    static constexpr int E_POOL = 16;
+    /* 32 slabs: a slab is held from enqueue until the worker's host build
+     * completes (~fold latency + build); with the deferred push pipeline the
+     * push thread no longer idles in a stream sync each step, so trigger
+     * bursts (≈8 closes/step at cfg2) need 2-3 steps of slack or the enqueue
+     * path sleeps on e_cv waiting for a free slab. */
+    static constexpr int E_POOL = 32;
     static constexpr int E_WORKERS = 4;
     struct DevEmit { /* per-slot device scratch, carved from one alloc */
         char* base = nullptr;
