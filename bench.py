@@ -47,6 +47,10 @@ def parse_args():
     p.add_argument("--no-filter", action="store_true")
     p.add_argument("--dist-backend", default=None,
                    help="torch.distributed backend override (default: nccl on GPU)")
+    p.add_argument("--staged-push", action="store_true",
+                   help="use the staging-copy push instead of the zero-copy "
+                        "borrowed push (also the fallback for A/B runs "
+                        "against engine builds without the borrowed entry)")
     p.add_argument("--skip-cpu-baseline", action="store_true")
     p.add_argument("--cpu-sample-rows", type=int, default=8_000_000)
     return p.parse_args()
@@ -160,10 +164,14 @@ def main():
 
     def push_step(step):
         off = step * B
+        # borrowed (zero-copy) push: the pre-generated stream stays resident
+        # and untouched for the whole run, exactly the lifetime the borrowed
+        # contract asks for
         op.push_device(B,
                        ctypes.c_void_p(d_ts.ptr.value + off * 8),
                        ctypes.c_void_p(d_kid.ptr.value + off * 4),
-                       ctypes.c_void_p(d_vals.ptr.value + off * 8))
+                       ctypes.c_void_p(d_vals.ptr.value + off * 8),
+                       borrowed=not args.staged_push)
         if dist is not None:
             import torch
             wm = torch.tensor([op.watermark], dtype=torch.int64)

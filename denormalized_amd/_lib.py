@@ -9,7 +9,8 @@ import os
 import numpy as np
 
 _DIR = os.path.dirname(os.path.abspath(__file__))
-_SO = os.path.join(_DIR, "_dzengine.so")
+# DZ_ENGINE_SO: A/B hook — point at an alternatively-built engine .so
+_SO = os.environ.get("DZ_ENGINE_SO") or os.path.join(_DIR, "_dzengine.so")
 
 DZ_OK = 0
 
@@ -91,6 +92,10 @@ def lib():
         L.dz_window_op_create.argtypes = [ctypes.POINTER(DzWindowDesc)]
         L.dz_window_op_push.argtypes = [p, ctypes.POINTER(DzBatch)]
         L.dz_window_op_push_device.argtypes = [p, i64, p, p, p]
+        try:
+            L.dz_window_op_push_device_borrowed.argtypes = [p, i64, p, p, p]
+        except AttributeError:  # older engine build (A/B hook)
+            pass
         L.dz_window_op_poll.argtypes = [p, ctypes.POINTER(ctypes.POINTER(DzOutBatch))]
         L.dz_window_op_finish.argtypes = [p]
         L.dz_window_op_drain.argtypes = [p]
@@ -211,9 +216,12 @@ class WindowOp:
         batch = DzBatch(n, 3, ctypes.cast(cols, ctypes.POINTER(DzColumn)))
         self._check(self._L.dz_window_op_push(self._h, ctypes.byref(batch)), "push")
 
-    def push_device(self, n, d_ts, d_kid32, d_vals):
-        self._check(self._L.dz_window_op_push_device(
-            self._h, n, d_ts, d_kid32, d_vals), "push_device")
+    def push_device(self, n, d_ts, d_kid32, d_vals, borrowed=False):
+        """borrowed=True: zero-copy — the op reads these buffers until the
+        NEXT call on the op; they must stay valid and unmodified until then."""
+        fn = (self._L.dz_window_op_push_device_borrowed if borrowed
+              else self._L.dz_window_op_push_device)
+        self._check(fn(self._h, n, d_ts, d_kid32, d_vals), "push_device")
 
     def poll(self, copy=True):
         """Returns a dict of numpy arrays for one emitted batch, or None.

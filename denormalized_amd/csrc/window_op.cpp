@@ -1116,7 +1116,7 @@ static dz_status ensure_bins(dz_window_op* op, int64_t bins_per_bucket) {
 static dz_status stage_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
                             const int32_t* d_kid, const double* d_vals,
                             const uint8_t* d_valbm, bool keys_are_dense,
-                            bool deferred) {
+                            bool deferred, bool borrow = false) {
     if (n >= (1LL << 31)) {
         op->err = "batch exceeds 2^31 rows; push smaller batches";
         return DZ_ERR;
@@ -1127,7 +1127,7 @@ static dz_status stage_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
     int b = op->next_buf;
     op->next_buf ^= 1;
     hipStream_t s = deferred ? op->i_stream : op->stream;
-    if (deferred) {
+    if (deferred && !borrow) {
         HostTimer ht(op, "h_stage");
         if (n > op->s_in_cap[b]) {
             quiesce(op);
@@ -1136,7 +1136,6 @@ static dz_status stage_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
             CHK(op, hipMalloc(&op->d_skid[b], (size_t)n * 4));
             CHK(op, hipMalloc(&op->d_svals[b], (size_t)n * 8));
             op->s_in_cap[b] = n;
-            op->consumed_valid[b] = false;
         }
         /* buffer b was last read by the scatter two pushes ago */
         if (op->consumed_valid[b])
@@ -1151,6 +1150,11 @@ static dz_status stage_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
         d_ts = op->d_sts[b];
         d_kid = op->d_skid[b];
         d_vals = op->d_svals[b];
+    } else if (deferred && op->consumed_valid[b]) {
+        /* borrowed push: caller keeps the inputs valid; the i_stream
+         * reduction still writes d_ghist[b]/d_scalars[b], whose previous
+         * contents the compute stream reads until that batch's scatter */
+        CHK(op, hipStreamWaitEvent(s, op->ev_consumed[b], 0));
     } else if (op->consumed_valid[b]) {
         /* host-batch path interleaved with device pushes: the compute stream
          * must not overwrite d_ghist[b]/d_scalars[b] before the i_stream
@@ -1192,7 +1196,7 @@ static dz_status stage_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
     op->pend.kid = d_kid;
     op->pend.vals = d_vals;
     op->pend.valbm = d_valbm;
-    if (deferred) {
+    if (deferred && !borrow) {
         /* the caller may release its input buffers once we return: wait for
          * the ring copies (typically tens of µs — the ring slot is idle) */
         HostTimer ht(op, "h_stage");
@@ -1469,6 +1473,28 @@ extern "C" dz_status dz_window_op_push_device(dz_window_op* op, int64_t n_rows,
     if (n_rows <= 0) return DZ_OK;
     return stage_core(op, n_rows, d_ts_ms, d_key_ids, d_vals, nullptr,
                       /*keys_are_dense=*/true, /*deferred=*/true);
+}
+
+/* Zero-copy variant: the op reads the caller's buffers directly, so they
+ * must stay valid AND unmodified until the NEXT call into the op (the
+ * deferred phase reads them then). For callers that own a resident stream
+ * buffer — the bench, or an FFI caller holding the batch across its poll
+ * loop iteration — this skips the staging copy (the D2D blit competes with
+ * the compute kernels for CUs). */
+extern "C" dz_status dz_window_op_push_device_borrowed(
+        dz_window_op* op, int64_t n_rows, const int64_t* d_ts_ms,
+        const int32_t* d_key_ids, const double* d_vals) {
+    if (!op) return DZ_ERR;
+    CHK(op, hipSetDevice(op->device));
+    if (op->no_group || !d_key_ids) {
+        if (ensure_zero_kid(op, n_rows) != DZ_OK) return DZ_ERR;
+        d_key_ids = op->d_zero_kid;
+    }
+    if (process_pending(op) != DZ_OK) return DZ_ERR;
+    if (n_rows <= 0) return DZ_OK;
+    return stage_core(op, n_rows, d_ts_ms, d_key_ids, d_vals, nullptr,
+                      /*keys_are_dense=*/true, /*deferred=*/true,
+                      /*borrow=*/true);
 }
 
 /* ------------------------------------------------------------------ */

@@ -153,6 +153,18 @@ dz_status dz_window_op_push_device(dz_window_op* op, int64_t n_rows,
                                    const int32_t* d_key_ids,
                                    const double* d_vals);
 
+/* Zero-copy variant of dz_window_op_push_device: the op reads the caller's
+ * device buffers directly instead of staging a copy, so they must remain
+ * valid AND unmodified until the NEXT call into the op (when the deferred
+ * phase consumes them). Use when the caller owns a resident stream buffer
+ * (the bench; an FFI caller holding the RecordBatch across its poll-loop
+ * iteration); use the staging variant above when buffer lifetime past the
+ * call cannot be guaranteed. */
+dz_status dz_window_op_push_device_borrowed(dz_window_op* op, int64_t n_rows,
+                                            const int64_t* d_ts_ms,
+                                            const int32_t* d_key_ids,
+                                            const double* d_vals);
+
 /* Retrieve emitted closed windows (the stream's output RecordBatch,
  * trigger_windows :220-253). *out = NULL when nothing is pending.
  * The returned batch stays valid until the next poll/destroy. */

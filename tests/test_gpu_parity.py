@@ -791,3 +791,53 @@ def test_shared_watermark_two_partitions():
     assert by_window(outsB) == by_window_exp(expB)
     for x in (opA, opB, oA, oB):
         x.close()
+
+
+@pytest.mark.gpu
+def test_push_device_borrowed_parity():
+    # zero-copy push: op reads caller buffers until the next call; results
+    # must match the staged push and the oracle bit-exactly
+    from denormalized_amd import DeviceArray, WindowOp, _lib, generate, synchronize
+    n, nkeys = 500_000, 700
+    d_ts = DeviceArray(0, n * 8)
+    d_kid = DeviceArray(0, n * 4)
+    d_vals = DeviceArray(0, n * 8)
+    generate(0, 123, 2_000_000, 0, n, nkeys, 500, d_ts.ptr, None, d_kid.ptr,
+             d_vals.ptr)
+    synchronize(0)
+    outs = {}
+    for borrowed in (False, True):
+        op = WindowOp(length_ms=1000, key_kind=_lib.KEY_DENSE_INT64,
+                      n_keys_hint=nkeys)
+        import ctypes
+        step = n // 4
+        for off in range(0, n, step):
+            m = min(step, n - off)
+            op.push_device(m,
+                           ctypes.c_void_p(d_ts.ptr.value + off * 8),
+                           ctypes.c_void_p(d_kid.ptr.value + off * 4),
+                           ctypes.c_void_p(d_vals.ptr.value + off * 8),
+                           borrowed=borrowed)
+        op.finish()
+        bs = op.poll_all()
+        outs[borrowed] = {
+            f: np.concatenate([b[f] for b in bs if b["n_rows"]])
+            for f in ("key", "count", "min", "max", "avg", "window_start")
+        }
+        op.close()
+    for f, a in outs[False].items():
+        assert np.array_equal(a, outs[True][f]), f
+
+    ts, kid, val = pyoracle.gen(123, 2_000_000, 0, n, nkeys, 500)
+    o = pyoracle.Oracle(1000, 0)
+    step = n // 4
+    for off in range(0, n, step):
+        sl = slice(off, off + step)
+        o.push(ts[sl], kid[sl], val[sl])
+    o.finish()
+    exp = o.fetch()
+    o.close()
+    for f in ("key", "count", "min", "max", "avg"):
+        assert np.array_equal(outs[True][f], exp[f]), f
+    for a in (d_ts, d_kid, d_vals):
+        a.free()
