@@ -983,6 +983,7 @@ static dz_status trigger_windows(dz_window_op* op) {
         int slab;
         hipEvent_t ev, slot_ev;
         {
+            HostTimer htw(op, "h_emit_slabwait");
             std::unique_lock<std::mutex> lk(op->e_mtx);
             op->e_cv.wait(lk, [&] { return !op->e_free.empty(); });
             slab = op->e_free.back();
@@ -1050,8 +1051,8 @@ static dz_status trigger_windows(dz_window_op* op) {
             op->e_jobs.push_back({ev, slab, c.start, c.end, op->n_keys, op->kcap,
                                   op->e_ticket_next++, dev_path});
         }
-        op->e_cv.notify_all();
     }
+    op->e_cv.notify_all(); /* one wakeup per group, not per close */
     } /* group loop */
     return DZ_OK;
 }
