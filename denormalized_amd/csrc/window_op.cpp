@@ -18,6 +18,7 @@
 #include <unordered_map>
 #include <atomic>
 #include <condition_variable>
+#include <memory>
 #include <mutex>
 #include <thread>
 #include <vector>
@@ -238,7 +239,16 @@ struct dz_window_op {
         int64_t wstart, wend, n_keys, kcap;
         uint64_t ticket;
         bool device; /* device-sorted columns vs raw host slab */
+        /* host-path closes of one trigger group share ONE event (their slab
+         * copies enqueue back-to-back); the last sibling to finish returns
+         * it to the pool */
+        std::shared_ptr<std::atomic<int>> grp_left;
     };
+    /* slot-release frontier for group-batched host emission: re-recorded on
+     * the copy stream after each group's slab reads; a window slot freed
+     * with a null per-slot event gates its reset on this instead */
+    hipEvent_t e_frontier = nullptr;
+    bool e_frontier_rec = false;
     std::deque<EmitJob> e_jobs;     /* guarded by e_mtx */
     std::vector<int> e_free;        /* free slab indices, guarded by e_mtx */
     std::vector<hipEvent_t> e_ev_pool; /* events for emission jobs (e_mtx) */
@@ -504,6 +514,7 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     if (op->i_stream) hipStreamDestroy(op->i_stream);
     for (auto& fs : op->free_slots)
         if (fs.ev) hipEventDestroy(fs.ev);
+    if (op->e_frontier) hipEventDestroy(op->e_frontier);
     delete op;
 }
 
@@ -861,7 +872,10 @@ static void emit_worker_main(dz_window_op* op) {
         {
             std::lock_guard<std::mutex> lk(op->e_mtx);
             op->e_free.push_back(job.slab);
-            op->e_ev_pool.push_back(job.ev);
+            /* a group-shared event goes back to the pool only once every
+             * sibling is past its spin (i.e. fully built) */
+            if (!job.grp_left || job.grp_left->fetch_sub(1) == 1)
+                op->e_ev_pool.push_back(job.ev);
             op->e_inflight--;
         }
         op->e_cv.notify_all();
@@ -974,8 +988,69 @@ static dz_status trigger_windows(dz_window_op* op) {
      * slabs whose release depends on this group's phase 2 (deadlock). */
     struct Pending { int slab; hipEvent_t ev; };
     constexpr size_t EGROUP = dz_window_op::E_POOL / 2;
+    const bool dev_path_g = op->n_keys > 65536;
     for (size_t g0 = 0; g0 < closed.size(); g0 += EGROUP) {
     const size_t g1 = std::min(closed.size(), g0 + EGROUP);
+    if (!dev_path_g) {
+        /* host-built closes (small keyspaces): batch the whole group behind
+         * ONE shared event — per close only the slab D2H enqueue remains on
+         * the push thread (the per-close event/job API overhead measured
+         * ~55 µs x ~8 closes/step at cfg2) */
+        hipEvent_t gev;
+        {
+            std::lock_guard<std::mutex> lk(op->e_mtx);
+            if (!op->e_ev_pool.empty()) {
+                gev = op->e_ev_pool.back();
+                op->e_ev_pool.pop_back();
+            } else {
+                hipEventCreate(&gev);
+            }
+        }
+        std::vector<dz_window_op::EmitJob> jobs;
+        jobs.reserve(g1 - g0);
+        for (size_t ci = g0; ci < g1; ci++) {
+            auto& c = closed[ci];
+            int slab;
+            {
+                HostTimer htw(op, "h_emit_slabwait");
+                std::unique_lock<std::mutex> lk(op->e_mtx);
+                op->e_cv.wait(lk, [&] { return !op->e_free.empty(); });
+                slab = op->e_free.back();
+                op->e_free.pop_back();
+            }
+            CHK(op, hipMemcpyAsync(op->e_slabs[slab],
+                                   op->s_base + (size_t)c.slot * stride,
+                                   stride * 8, hipMemcpyDeviceToHost,
+                                   op->copy_stream));
+            dz_window_op::EmitJob j;
+            j.ev = gev;
+            j.slab = slab;
+            j.wstart = c.start;
+            j.wend = c.end;
+            j.n_keys = op->n_keys;
+            j.kcap = op->kcap;
+            j.device = false;
+            jobs.push_back(std::move(j));
+            op->free_slots.push_back({c.slot, nullptr});
+        }
+        CHK(op, hipEventRecord(gev, op->copy_stream));
+        if (!op->e_frontier)
+            hipEventCreateWithFlags(&op->e_frontier, hipEventDisableTiming);
+        CHK(op, hipEventRecord(op->e_frontier, op->copy_stream));
+        op->e_frontier_rec = true;
+        auto left = std::make_shared<std::atomic<int>>((int)jobs.size());
+        {
+            std::lock_guard<std::mutex> lk(op->e_mtx);
+            for (auto& j : jobs) {
+                j.ticket = op->e_ticket_next++;
+                j.grp_left = left;
+                op->e_jobs.push_back(std::move(j));
+                op->e_inflight++;
+            }
+        }
+        op->e_cv.notify_all();
+        continue;
+    }
     std::vector<Pending> pend;
     pend.reserve(g1 - g0);
     for (size_t ci = g0; ci < g1; ci++) {
@@ -1288,6 +1363,10 @@ static dz_status process_pending(dz_window_op* op) {
                 CHK(op, hipStreamWaitEvent(op->stream, fs.ev, 0));
                 std::lock_guard<std::mutex> lk(op->e_mtx);
                 op->e_ev_pool.push_back(fs.ev);
+            } else if (op->e_frontier_rec) {
+                /* group-batched host emission gates slot reuse on the shared
+                 * read frontier (re-records only push it later — safe) */
+                CHK(op, hipStreamWaitEvent(op->stream, op->e_frontier, 0));
             }
             reset_list.push_back(fs.slot);
             op->open[ws[r]] = {we[r], fs.slot};
