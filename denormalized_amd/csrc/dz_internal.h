@@ -122,4 +122,12 @@ constexpr int EMIT_RBINS = 2048;
 void launch_reset_slots(hipStream_t stream, const int32_t* d_slots, int ns,
                         int64_t kcap, uint64_t* s_cnt, uint64_t* s_first);
 
+/* host-path emission: pack up to 16 closing window slots' state slabs into
+ * one contiguous staging area (one launch + ONE big D2H replaces a per-close
+ * copy on the push thread; slot ids travel by value in the kernel args) */
+struct EGatherSlots { int32_t s[16]; };
+void launch_egather_slabs(hipStream_t stream, const uint64_t* s_base,
+                          int64_t stride_u64, EGatherSlots slots, int gcount,
+                          uint64_t* out);
+
 } // namespace dz

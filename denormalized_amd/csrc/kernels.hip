@@ -1168,3 +1168,29 @@ void launch_reset_slots(hipStream_t s, const int32_t* d_slots, int ns,
 }
 
 } // namespace dz
+
+/* ------------------------------------------------------------------ */
+/* host-path emission: pack closing slots' slabs into one staging run  */
+/* (grid.y = slot index; no per-element division)                      */
+/* ------------------------------------------------------------------ */
+
+__global__ void k_egather_slabs(const uint64_t* __restrict__ s_base,
+                                int64_t stride_u64, dz::EGatherSlots slots,
+                                uint64_t* __restrict__ out) {
+    const int g = blockIdx.y;
+    const uint64_t* src = s_base + (int64_t)slots.s[g] * stride_u64;
+    uint64_t* dst = out + (int64_t)g * stride_u64;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < stride_u64; i += (int64_t)gridDim.x * blockDim.x)
+        dst[i] = src[i];
+}
+
+void dz::launch_egather_slabs(hipStream_t s, const uint64_t* s_base,
+                              int64_t stride_u64, dz::EGatherSlots slots,
+                              int gcount, uint64_t* out) {
+    int bx = (int)std::min<int64_t>((stride_u64 + dz::BLOCK - 1) / dz::BLOCK,
+                                    512);
+    dim3 grid(bx, gcount);
+    hipLaunchKernelGGL(k_egather_slabs, grid, dim3(dz::BLOCK), 0, s, s_base,
+                       stride_u64, slots, out);
+}

@@ -240,10 +240,20 @@ struct dz_window_op {
         uint64_t ticket;
         bool device; /* device-sorted columns vs raw host slab */
         /* host-path closes of one trigger group share ONE event (their slab
-         * copies enqueue back-to-back); the last sibling to finish returns
-         * it to the pool */
+         * reads ride one gather + one D2H); the last sibling to finish
+         * returns the event and the group buffer to their pools */
         std::shared_ptr<std::atomic<int>> grp_left;
+        int gbuf = -1;  /* pinned group buffer index, or -1 (own slab) */
+        int goff = 0;   /* this close's slab offset within the group buffer */
     };
+    /* pinned group buffers for batched host-path emission (gather kernel
+     * packs up to EGROUP slots contiguously on device, one D2H lands here) */
+    static constexpr int E_GBUFS = 4;
+    uint64_t* e_gbufs[E_GBUFS] = {};
+    std::vector<int> e_gfree;            /* guarded by e_mtx */
+    uint64_t* d_egather = nullptr;       /* shared device staging (copy-stream
+                                          * ordered, so one buffer suffices) */
+    int64_t e_gbuf_kcap = 0;
     /* slot-release frontier for group-batched host emission: re-recorded on
      * the copy stream after each group's slab reads; a window slot freed
      * with a null per-slot event gates its reset on this instead */
@@ -485,6 +495,9 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     for (auto e : op->e_ev_pool) hipEventDestroy(e);
     for (auto& s : op->e_slabs)
         if (s) hipHostFree(s);
+    for (auto& g : op->e_gbufs)
+        if (g) hipHostFree(g);
+    hipFree(op->d_egather);
     for (auto& d : op->e_dev)
         if (d.base) hipFree(d.base);
     if (op->e_pcnt) hipHostFree(op->e_pcnt);
@@ -853,8 +866,11 @@ static void emit_worker_main(dz_window_op* op) {
             build_emission(op, job.wstart, job.wend, nt, job.kcap,
                            op->e_slabs[job.slab], &ob);
         } else {
+            const uint64_t* slab = job.gbuf >= 0
+                ? op->e_gbufs[job.gbuf] + (size_t)job.goff * job.kcap * 5
+                : op->e_slabs[job.slab];
             build_emission_host(op, job.wstart, job.wend, job.n_keys, job.kcap,
-                                op->e_slabs[job.slab], &ob);
+                                slab, &ob);
         }
         op->e_build_ns += (uint64_t)std::chrono::duration_cast<std::chrono::nanoseconds>(
             std::chrono::steady_clock::now() - t0).count();
@@ -871,11 +887,13 @@ static void emit_worker_main(dz_window_op* op) {
         }
         {
             std::lock_guard<std::mutex> lk(op->e_mtx);
-            op->e_free.push_back(job.slab);
-            /* a group-shared event goes back to the pool only once every
+            if (job.slab >= 0) op->e_free.push_back(job.slab);
+            /* group-shared resources go back to their pools only once every
              * sibling is past its spin (i.e. fully built) */
-            if (!job.grp_left || job.grp_left->fetch_sub(1) == 1)
+            if (!job.grp_left || job.grp_left->fetch_sub(1) == 1) {
                 op->e_ev_pool.push_back(job.ev);
+                if (job.gbuf >= 0) op->e_gfree.push_back(job.gbuf);
+            }
             op->e_inflight--;
         }
         op->e_cv.notify_all();
@@ -921,9 +939,29 @@ static dz_status ensure_emission(dz_window_op* op) {
             CHK(op, hipHostMalloc((void**)&op->e_slabs[i],
                                   (size_t)kc * SLAB_BYTES_PER_ENTRY + 16));
         }
+        /* group buffers for batched host-path emission (small keyspaces
+         * only: a group buffer is EGROUP whole slabs) */
+        for (int i = 0; i < dz_window_op::E_GBUFS; i++)
+            if (op->e_gbufs[i]) {
+                hipHostFree(op->e_gbufs[i]);
+                op->e_gbufs[i] = nullptr;
+            }
+        hipFree(op->d_egather);
+        op->d_egather = nullptr;
+        bool have_gbufs = kc <= 65536;
+        if (have_gbufs) {
+            size_t gbytes = (size_t)(dz_window_op::E_POOL / 2) * kc * 5 * 8;
+            CHK(op, hipMalloc(&op->d_egather, gbytes));
+            for (int i = 0; i < dz_window_op::E_GBUFS; i++)
+                CHK(op, hipHostMalloc((void**)&op->e_gbufs[i], gbytes));
+        }
         std::lock_guard<std::mutex> lk(op->e_mtx);
         op->e_free.clear();
         for (int i = 0; i < dz_window_op::E_POOL; i++) op->e_free.push_back(i);
+        op->e_gfree.clear();
+        if (have_gbufs)
+            for (int i = 0; i < dz_window_op::E_GBUFS; i++)
+                op->e_gfree.push_back(i);
         op->e_slab_kcap = op->kcap;
     }
     return DZ_OK;
@@ -991,14 +1029,20 @@ static dz_status trigger_windows(dz_window_op* op) {
     const bool dev_path_g = op->n_keys > 65536;
     for (size_t g0 = 0; g0 < closed.size(); g0 += EGROUP) {
     const size_t g1 = std::min(closed.size(), g0 + EGROUP);
-    if (!dev_path_g) {
-        /* host-built closes (small keyspaces): batch the whole group behind
-         * ONE shared event — per close only the slab D2H enqueue remains on
-         * the push thread (the per-close event/job API overhead measured
-         * ~55 µs x ~8 closes/step at cfg2) */
+    if (!dev_path_g && op->e_gbufs[0]) {
+        /* host-built closes (small keyspaces): pack the whole group's slabs
+         * with one gather launch + ONE D2H into a pinned group buffer behind
+         * ONE shared event — the push thread pays a fixed ~4 hip calls per
+         * group instead of ~5 per close (measured ~55 µs x ~8 closes/step
+         * at cfg2) */
         hipEvent_t gev;
+        int gbuf;
         {
-            std::lock_guard<std::mutex> lk(op->e_mtx);
+            HostTimer htw(op, "h_emit_slabwait");
+            std::unique_lock<std::mutex> lk(op->e_mtx);
+            op->e_cv.wait(lk, [&] { return !op->e_gfree.empty(); });
+            gbuf = op->e_gfree.back();
+            op->e_gfree.pop_back();
             if (!op->e_ev_pool.empty()) {
                 gev = op->e_ev_pool.back();
                 op->e_ev_pool.pop_back();
@@ -1006,44 +1050,39 @@ static dz_status trigger_windows(dz_window_op* op) {
                 hipEventCreate(&gev);
             }
         }
-        std::vector<dz_window_op::EmitJob> jobs;
-        jobs.reserve(g1 - g0);
+        dz::EGatherSlots gs;
+        const int gcount = (int)(g1 - g0);
         for (size_t ci = g0; ci < g1; ci++) {
-            auto& c = closed[ci];
-            int slab;
-            {
-                HostTimer htw(op, "h_emit_slabwait");
-                std::unique_lock<std::mutex> lk(op->e_mtx);
-                op->e_cv.wait(lk, [&] { return !op->e_free.empty(); });
-                slab = op->e_free.back();
-                op->e_free.pop_back();
-            }
-            CHK(op, hipMemcpyAsync(op->e_slabs[slab],
-                                   op->s_base + (size_t)c.slot * stride,
-                                   stride * 8, hipMemcpyDeviceToHost,
-                                   op->copy_stream));
-            dz_window_op::EmitJob j;
-            j.ev = gev;
-            j.slab = slab;
-            j.wstart = c.start;
-            j.wend = c.end;
-            j.n_keys = op->n_keys;
-            j.kcap = op->kcap;
-            j.device = false;
-            jobs.push_back(std::move(j));
-            op->free_slots.push_back({c.slot, nullptr});
+            gs.s[ci - g0] = closed[ci].slot;
+            op->free_slots.push_back({closed[ci].slot, nullptr});
         }
-        CHK(op, hipEventRecord(gev, op->copy_stream));
+        dz::launch_egather_slabs(op->copy_stream, op->s_base, stride, gs,
+                                 gcount, op->d_egather);
+        /* slots are readable again once the gather (not the D2H) is done */
         if (!op->e_frontier)
             hipEventCreateWithFlags(&op->e_frontier, hipEventDisableTiming);
         CHK(op, hipEventRecord(op->e_frontier, op->copy_stream));
         op->e_frontier_rec = true;
-        auto left = std::make_shared<std::atomic<int>>((int)jobs.size());
+        CHK(op, hipMemcpyAsync(op->e_gbufs[gbuf], op->d_egather,
+                               (size_t)gcount * stride * 8,
+                               hipMemcpyDeviceToHost, op->copy_stream));
+        CHK(op, hipEventRecord(gev, op->copy_stream));
+        auto left = std::make_shared<std::atomic<int>>(gcount);
         {
             std::lock_guard<std::mutex> lk(op->e_mtx);
-            for (auto& j : jobs) {
-                j.ticket = op->e_ticket_next++;
+            for (size_t ci = g0; ci < g1; ci++) {
+                dz_window_op::EmitJob j;
+                j.ev = gev;
+                j.slab = -1;
+                j.wstart = closed[ci].start;
+                j.wend = closed[ci].end;
+                j.n_keys = op->n_keys;
+                j.kcap = op->kcap;
+                j.device = false;
                 j.grp_left = left;
+                j.gbuf = gbuf;
+                j.goff = (int)(ci - g0);
+                j.ticket = op->e_ticket_next++;
                 op->e_jobs.push_back(std::move(j));
                 op->e_inflight++;
             }
