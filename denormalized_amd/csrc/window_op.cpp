@@ -998,8 +998,11 @@ static dz_status trigger_windows(dz_window_op* op) {
             hipEventCreate(&evA);
         }
     }
-    CHK(op, hipEventRecord(evA, op->stream));
-    CHK(op, hipStreamWaitEvent(op->copy_stream, evA, 0));
+    {
+        HostTimer htf(op, "h_trig_fence");
+        CHK(op, hipEventRecord(evA, op->stream));
+        CHK(op, hipStreamWaitEvent(op->copy_stream, evA, 0));
+    }
     {
         std::lock_guard<std::mutex> lk(op->e_mtx);
         op->e_ev_pool.push_back(evA);
@@ -1056,6 +1059,8 @@ static dz_status trigger_windows(dz_window_op* op) {
             gs.s[ci - g0] = closed[ci].slot;
             op->free_slots.push_back({closed[ci].slot, nullptr});
         }
+        {
+        HostTimer htg(op, "h_trig_gather");
         dz::launch_egather_slabs(op->copy_stream, op->s_base, stride, gs,
                                  gcount, op->d_egather);
         /* slots are readable again once the gather (not the D2H) is done */
@@ -1067,8 +1072,10 @@ static dz_status trigger_windows(dz_window_op* op) {
                                (size_t)gcount * stride * 8,
                                hipMemcpyDeviceToHost, op->copy_stream));
         CHK(op, hipEventRecord(gev, op->copy_stream));
+        }
         auto left = std::make_shared<std::atomic<int>>(gcount);
         {
+            HostTimer htj(op, "h_trig_jobs");
             std::lock_guard<std::mutex> lk(op->e_mtx);
             for (size_t ci = g0; ci < g1; ci++) {
                 dz_window_op::EmitJob j;
