@@ -131,7 +131,12 @@ struct dz_window_op {
     double* s_min = nullptr;
     double* s_max = nullptr;
     double* s_sum = nullptr;
-    struct FreeSlot { int32_t slot; hipEvent_t ev; /* copy-done gate or null */ };
+    struct FreeSlot {
+        int32_t slot;
+        hipEvent_t ev;     /* copy-done gate or null */
+        bool ev_owned = true; /* false: a rotating frontier event — wait on
+                               * reuse but never return it to the pool */
+    };
     std::vector<FreeSlot> free_slots;
     struct OpenWin { int64_t end; int32_t slot; };
     std::map<int64_t, OpenWin> open; /* by window start (BTreeMap order) */
@@ -254,11 +259,15 @@ struct dz_window_op {
     uint64_t* d_egather = nullptr;       /* shared device staging (copy-stream
                                           * ordered, so one buffer suffices) */
     int64_t e_gbuf_kcap = 0;
-    /* slot-release frontier for group-batched host emission: re-recorded on
-     * the copy stream after each group's slab reads; a window slot freed
-     * with a null per-slot event gates its reset on this instead */
-    hipEvent_t e_frontier = nullptr;
-    bool e_frontier_rec = false;
+    /* slot-release frontier events for group-batched host emission: each
+     * group records the next event of this rotation after its gather; the
+     * freed slots' reset waits reference it WITHOUT taking ownership (a
+     * later re-record only pushes the observed instant later — safe).
+     * Rotating avoids re-recording a still-pending event, which the runtime
+     * serves with an internal wait. */
+    static constexpr int E_FRONTIERS = 4;
+    hipEvent_t e_frontier[E_FRONTIERS] = {};
+    int e_frontier_idx = 0;
     std::deque<EmitJob> e_jobs;     /* guarded by e_mtx */
     std::vector<int> e_free;        /* free slab indices, guarded by e_mtx */
     std::vector<hipEvent_t> e_ev_pool; /* events for emission jobs (e_mtx) */
@@ -527,7 +536,8 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     if (op->i_stream) hipStreamDestroy(op->i_stream);
     for (auto& fs : op->free_slots)
         if (fs.ev) hipEventDestroy(fs.ev);
-    if (op->e_frontier) hipEventDestroy(op->e_frontier);
+    for (auto e : op->e_frontier)
+        if (e) hipEventDestroy(e);
     delete op;
 }
 
@@ -1055,23 +1065,30 @@ static dz_status trigger_windows(dz_window_op* op) {
         }
         dz::EGatherSlots gs;
         const int gcount = (int)(g1 - g0);
+        /* slots are readable again once the gather (not the D2H) is done:
+         * gate their reuse on the next rotating frontier event */
+        hipEvent_t fr = op->e_frontier[op->e_frontier_idx];
+        if (!fr) {
+            hipEventCreateWithFlags(&fr, hipEventDisableTiming);
+            op->e_frontier[op->e_frontier_idx] = fr;
+        }
+        op->e_frontier_idx = (op->e_frontier_idx + 1) % dz_window_op::E_FRONTIERS;
         for (size_t ci = g0; ci < g1; ci++) {
             gs.s[ci - g0] = closed[ci].slot;
-            op->free_slots.push_back({closed[ci].slot, nullptr});
+            op->free_slots.push_back({closed[ci].slot, fr, false});
         }
         {
-        HostTimer htg(op, "h_trig_gather");
-        dz::launch_egather_slabs(op->copy_stream, op->s_base, stride, gs,
-                                 gcount, op->d_egather);
-        /* slots are readable again once the gather (not the D2H) is done */
-        if (!op->e_frontier)
-            hipEventCreateWithFlags(&op->e_frontier, hipEventDisableTiming);
-        CHK(op, hipEventRecord(op->e_frontier, op->copy_stream));
-        op->e_frontier_rec = true;
-        CHK(op, hipMemcpyAsync(op->e_gbufs[gbuf], op->d_egather,
-                               (size_t)gcount * stride * 8,
-                               hipMemcpyDeviceToHost, op->copy_stream));
-        CHK(op, hipEventRecord(gev, op->copy_stream));
+            HostTimer htg(op, "h_trig_gather");
+            dz::launch_egather_slabs(op->copy_stream, op->s_base, stride, gs,
+                                     gcount, op->d_egather);
+            CHK(op, hipEventRecord(fr, op->copy_stream));
+        }
+        {
+            HostTimer htc(op, "h_trig_copy");
+            CHK(op, hipMemcpyAsync(op->e_gbufs[gbuf], op->d_egather,
+                                   (size_t)gcount * stride * 8,
+                                   hipMemcpyDeviceToHost, op->copy_stream));
+            CHK(op, hipEventRecord(gev, op->copy_stream));
         }
         auto left = std::make_shared<std::atomic<int>>(gcount);
         {
@@ -1405,14 +1422,12 @@ static dz_status process_pending(dz_window_op* op) {
             }
             dz_window_op::FreeSlot fs = op->free_slots.back();
             op->free_slots.pop_back();
-            if (fs.ev) { /* emission copy of this slot may still be in flight */
+            if (fs.ev) { /* emission read of this slot may still be in flight */
                 CHK(op, hipStreamWaitEvent(op->stream, fs.ev, 0));
-                std::lock_guard<std::mutex> lk(op->e_mtx);
-                op->e_ev_pool.push_back(fs.ev);
-            } else if (op->e_frontier_rec) {
-                /* group-batched host emission gates slot reuse on the shared
-                 * read frontier (re-records only push it later — safe) */
-                CHK(op, hipStreamWaitEvent(op->stream, op->e_frontier, 0));
+                if (fs.ev_owned) {
+                    std::lock_guard<std::mutex> lk(op->e_mtx);
+                    op->e_ev_pool.push_back(fs.ev);
+                }
             }
             reset_list.push_back(fs.slot);
             op->open[ws[r]] = {we[r], fs.slot};
