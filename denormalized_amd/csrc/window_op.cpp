@@ -254,12 +254,15 @@ struct dz_window_op {
     /* pinned group buffers for batched host-path emission (gather kernel
      * packs up to EGROUP slots contiguously on device, one D2H lands here) */
     static constexpr int E_GBUFS = 4;
-    uint64_t* e_gbufs[E_GBUFS] = {};
+    uint64_t* e_gbufs[E_GBUFS] = {};     /* pinned, host view */
+    uint64_t* e_gbufs_dev[E_GBUFS] = {}; /* device view of the same memory:
+                                          * the gather kernel writes output
+                                          * straight over the host link — a
+                                          * hipMemcpyAsync D2H here performs
+                                          * the transfer at ENQUEUE time on
+                                          * the calling thread (measured),
+                                          * a kernel store does not */
     std::vector<int> e_gfree;            /* guarded by e_mtx */
-    uint64_t* d_egather = nullptr;       /* device staging, one region per
-                                          * group buffer (workers D2H their
-                                          * own slice after the gather) */
-    size_t e_gregion_u64 = 0;            /* u64s per gbuf region */
     int64_t e_gbuf_kcap = 0;
     /* slot-release frontier events for group-batched host emission: each
      * group records the next event of this rotation after its gather; the
@@ -508,7 +511,6 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
         if (s) hipHostFree(s);
     for (auto& g : op->e_gbufs)
         if (g) hipHostFree(g);
-    hipFree(op->d_egather);
     for (auto& d : op->e_dev)
         if (d.base) hipFree(d.base);
     if (op->e_pcnt) hipHostFree(op->e_pcnt);
@@ -878,38 +880,11 @@ static void emit_worker_main(dz_window_op* op) {
             build_emission(op, job.wstart, job.wend, nt, job.kcap,
                            op->e_slabs[job.slab], &ob);
         } else {
-            const uint64_t* slab;
-            if (job.gbuf >= 0) {
-                /* job.ev = gather done; pull this close's slab slice down
-                 * here, off the push thread (the D2H transfer runs at
-                 * enqueue time) */
-                size_t off = (size_t)job.goff * job.kcap * 5;
-                uint64_t* dst = op->e_gbufs[job.gbuf] + off;
-                hipEvent_t me;
-                {
-                    std::lock_guard<std::mutex> lk(op->e_mtx);
-                    if (!op->e_ev_pool.empty()) {
-                        me = op->e_ev_pool.back();
-                        op->e_ev_pool.pop_back();
-                    } else {
-                        hipEventCreate(&me);
-                    }
-                }
-                hipMemcpyAsync(dst,
-                               op->d_egather +
-                                   (size_t)job.gbuf * op->e_gregion_u64 + off,
-                               (size_t)job.kcap * 5 * 8,
-                               hipMemcpyDeviceToHost, op->copy_stream);
-                hipEventRecord(me, op->copy_stream);
-                event_spin(me);
-                {
-                    std::lock_guard<std::mutex> lk(op->e_mtx);
-                    op->e_ev_pool.push_back(me);
-                }
-                slab = dst;
-            } else {
-                slab = op->e_slabs[job.slab];
-            }
+            /* gbuf jobs: job.ev = gather done, and the gather wrote this
+             * close's slab directly into the pinned group buffer */
+            const uint64_t* slab = job.gbuf >= 0
+                ? op->e_gbufs[job.gbuf] + (size_t)job.goff * job.kcap * 5
+                : op->e_slabs[job.slab];
             build_emission_host(op, job.wstart, job.wend, job.n_keys, job.kcap,
                                 slab, &ob);
         }
@@ -987,16 +962,15 @@ static dz_status ensure_emission(dz_window_op* op) {
                 hipHostFree(op->e_gbufs[i]);
                 op->e_gbufs[i] = nullptr;
             }
-        hipFree(op->d_egather);
-        op->d_egather = nullptr;
         bool have_gbufs = kc <= 65536;
         if (have_gbufs) {
             size_t gbytes = (size_t)(dz_window_op::E_POOL / 2) * kc * 5 * 8;
-            op->e_gregion_u64 = gbytes / 8;
-            CHK(op, hipMalloc(&op->d_egather,
-                              gbytes * dz_window_op::E_GBUFS));
-            for (int i = 0; i < dz_window_op::E_GBUFS; i++)
-                CHK(op, hipHostMalloc((void**)&op->e_gbufs[i], gbytes));
+            for (int i = 0; i < dz_window_op::E_GBUFS; i++) {
+                CHK(op, hipHostMalloc((void**)&op->e_gbufs[i], gbytes,
+                                      hipHostMallocMapped));
+                CHK(op, hipHostGetDevicePointer((void**)&op->e_gbufs_dev[i],
+                                                op->e_gbufs[i], 0));
+            }
         }
         std::lock_guard<std::mutex> lk(op->e_mtx);
         op->e_free.clear();
@@ -1111,14 +1085,13 @@ static dz_status trigger_windows(dz_window_op* op) {
             op->free_slots.push_back({closed[ci].slot, fr, false});
         }
         {
-            /* the push thread only launches the gather; the D2H of each
-             * slab slice moves to the worker that builds it (hipMemcpyAsync
-             * D2H was measured performing the transfer at enqueue time,
-             * ~0.36 ms/step on the push thread) */
+            /* ONE launch reads the closing slots and writes their slabs
+             * straight into the pinned group buffer over the host link —
+             * fully asynchronous for the push thread (hipMemcpyAsync D2H
+             * was measured performing the transfer at ENQUEUE time) */
             HostTimer htg(op, "h_trig_gather");
             dz::launch_egather_slabs(op->copy_stream, op->s_base, stride, gs,
-                                     gcount,
-                                     op->d_egather + (size_t)gbuf * op->e_gregion_u64);
+                                     gcount, op->e_gbufs_dev[gbuf]);
             CHK(op, hipEventRecord(fr, op->copy_stream));
             CHK(op, hipEventRecord(gev, op->copy_stream));
         }
