@@ -149,9 +149,9 @@ struct dz_window_op {
      * pipeline runs batch N's reduction on i_stream while batch N-1's scan
      * still reads the other buffer on the compute stream) */
     uint32_t* d_ghist[2] = {};
-    uint32_t* d_gofs = nullptr;
-    uint32_t* d_total = nullptr;
-    uint32_t* d_base = nullptr;
+    uint32_t* d_gofs[2] = {};
+    uint32_t* d_total[2] = {};
+    uint32_t* d_base[2] = {};
     int C_cap = 0;
     uint32_t* d_meta = nullptr;
     uint4* d_grec = nullptr;  /* 16 B {val, rowidx} records (scatter output) */
@@ -467,7 +467,15 @@ extern "C" dz_window_op* dz_window_op_create(const dz_window_desc* desc) {
     hipMalloc(&op->d_scalars[0], 3 * 8);
     hipMalloc(&op->d_scalars[1], 3 * 8);
     hipHostMalloc((void**)&op->h_scalars, 2 * 3 * 8);
-    hipStreamCreateWithFlags(&op->i_stream, hipStreamNonBlocking);
+    {
+        /* high-priority ingest stream: the next batch's reduction should
+         * fill gaps ahead of the current batch's queued partition/fold, not
+         * drain after it — its result gates the host's next decisions */
+        int least = 0, greatest = 0;
+        hipDeviceGetStreamPriorityRange(&least, &greatest);
+        hipStreamCreateWithPriority(&op->i_stream, hipStreamNonBlocking,
+                                    greatest);
+    }
     for (int i = 0; i < 2; i++) {
         hipEventCreateWithFlags(&op->ev_ready[i], hipEventDisableTiming);
         hipEventCreateWithFlags(&op->ev_staged[i], hipEventDisableTiming);
@@ -518,7 +526,9 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_roffs);
     hipFree(op->s_base);
     hipFree(op->d_ghist[0]); hipFree(op->d_ghist[1]);
-    hipFree(op->d_gofs); hipFree(op->d_total); hipFree(op->d_base);
+    hipFree(op->d_gofs[0]); hipFree(op->d_gofs[1]);
+    hipFree(op->d_total[0]); hipFree(op->d_total[1]);
+    hipFree(op->d_base[0]); hipFree(op->d_base[1]);
     hipFree(op->d_meta); hipFree(op->d_grec); hipFree(op->d_gfrec);
     hipFree(op->d_binoffs); hipFree(op->d_binlens);
     hipFree(op->d_b1offs); hipFree(op->d_b1lens);
@@ -1215,13 +1225,15 @@ static void quiesce(dz_window_op* op) {
 static dz_status ensure_scratch(dz_window_op* op, int C, int64_t nrec) {
     if (C > op->C_cap) {
         quiesce(op);
-        hipFree(op->d_ghist[0]); hipFree(op->d_ghist[1]); hipFree(op->d_gofs);
-        CHK(op, hipMalloc(&op->d_ghist[0], (size_t)C * dz::NB * 4));
-        CHK(op, hipMalloc(&op->d_ghist[1], (size_t)C * dz::NB * 4));
-        CHK(op, hipMalloc(&op->d_gofs, (size_t)C * dz::NB * 4));
-        if (!op->d_total) {
-            CHK(op, hipMalloc(&op->d_total, (size_t)dz::SCAN_SSPLIT * dz::NB * 4));
-            CHK(op, hipMalloc(&op->d_base, (dz::NB + 1) * 4));
+        for (int i = 0; i < 2; i++) {
+            hipFree(op->d_ghist[i]); hipFree(op->d_gofs[i]);
+            CHK(op, hipMalloc(&op->d_ghist[i], (size_t)C * dz::NB * 4));
+            CHK(op, hipMalloc(&op->d_gofs[i], (size_t)C * dz::NB * 4));
+            if (!op->d_total[i]) {
+                CHK(op, hipMalloc(&op->d_total[i],
+                                  (size_t)dz::SCAN_SSPLIT * dz::NB * 4));
+                CHK(op, hipMalloc(&op->d_base[i], (dz::NB + 1) * 4));
+            }
         }
         op->C_cap = C;
     }
@@ -1319,6 +1331,13 @@ static dz_status stage_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
             dz::launch_hist(s, d_kid, d_ts, n, chunk, C, wp,
                             op->d_ghist[b], op->d_scalars[b]);
         });
+        /* the scan depends only on the histogram (not on the host's window
+         * decisions): run it here so process_pending's critical path starts
+         * at the scatter */
+        timed_on(op, s, "scan", (double)C * dz::NB * 12, [&] {
+            dz::launch_scan(s, op->d_ghist[b], C, op->d_total[b],
+                            op->d_base[b], op->d_gofs[b]);
+        });
     } else {
         timed_on(op, s, "minmax", (double)n * 12, [&] {
             dz::launch_minmax(s, d_ts, keys_are_dense ? d_kid : nullptr,
@@ -1388,12 +1407,7 @@ static dz_status process_pending(dz_window_op* op) {
         int64_t kmax = (int64_t)hs[2];
         op->n_keys = std::max(op->n_keys, kmax + 1);
     }
-    if (!sliding) {
-        timed(op, "scan", (double)C * dz::NB * 12, [&] {
-            dz::launch_scan(op->stream, op->d_ghist[b], C, op->d_total,
-                            op->d_base, op->d_gofs);
-        });
-    }
+
     if (op->n_keys > op->kcap) {
         if (state_alloc(op, std::max(op->n_keys, op->kcap * 2), op->nslots) != DZ_OK)
             return DZ_ERR;
@@ -1480,19 +1494,15 @@ static dz_status process_pending(dz_window_op* op) {
                             op->d_ghist[b], nullptr);
         });
         timed(op, "scan", (double)C * dz::NB * 12, [&] {
-            dz::launch_scan(op->stream, op->d_ghist[b], C, op->d_total,
-                            op->d_base, op->d_gofs);
+            dz::launch_scan(op->stream, op->d_ghist[b], C, op->d_total[b],
+                            op->d_base[b], op->d_gofs[b]);
         });
     }
     int32_t st_rows = (int32_t)std::max<int64_t>(64, dz::ST_RECORDS / expand);
     timed(op, "scatter", (double)n * 24 + (double)nrec_max * 20, [&] {
         dz::launch_scatter(op->stream, d_kid, d_ts, d_vals, d_valbm, n, chunk, C,
-                           st_rows, wp, op->d_gofs, op->d_meta, op->d_grec);
+                           st_rows, wp, op->d_gofs[b], op->d_meta, op->d_grec);
     });
-    /* scatter is the last reader of the staged inputs and of d_ghist[b]:
-     * once it completes, pipeline buffer b may be restaged */
-    CHK(op, hipEventRecord(op->ev_consumed[b], op->stream));
-    op->consumed_valid[b] = true;
     int64_t klocs = op->kcap >> dz::LOG_NB;
     int64_t gtot = klocs * nw;
     int64_t khigh = (klocs + 255) >> 8;
@@ -1523,18 +1533,20 @@ static dz_status process_pending(dz_window_op* op) {
         fc.bin_stride = (int32_t)stride;
         fc.tl_nw = (int32_t)nw;
         timed(op, "regroup", (double)nrec_max * 40, [&] {
-            dz::launch_regroup_l1(op->stream, op->d_meta, op->d_grec, op->d_base,
+            dz::launch_regroup_l1(op->stream, op->d_meta, op->d_grec,
+                                  op->d_base[b],
                                   fc, op->d_b1offs, op->d_b1lens, op->d_meta2,
                                   op->d_grec2);
         });
         timed(op, "regroup", (double)nrec_max * 40, [&] {
             dz::launch_regroup_l2(op->stream, op->d_meta2, op->d_grec2,
-                                  op->d_base, fc, (int)nb1, op->d_b1offs,
+                                  op->d_base[b], fc, (int)nb1, op->d_b1offs,
                                   op->d_b1lens, op->d_binoffs, op->d_binlens,
                                   op->d_gfrec);
         });
         timed(op, "fold", (double)nrec_max * 16, [&] {
-            dz::launch_fold3(op->stream, op->d_gfrec, op->d_base, op->d_binoffs,
+            dz::launch_fold3(op->stream, op->d_gfrec, op->d_base[b],
+                             op->d_binoffs,
                              op->d_binlens, fc, op->d_slotmap, op->s_cnt,
                              op->s_min, op->s_max, op->s_sum, op->s_first);
         });
@@ -1555,11 +1567,11 @@ static dz_status process_pending(dz_window_op* op) {
                 fc.tl_nw = 0;
                 timed(op, "regroup", (double)nrec_max * 40, [&] {
                     dz::launch_regroup(op->stream, op->d_meta, op->d_grec,
-                                       op->d_base, fc, op->d_binoffs,
+                                       op->d_base[b], fc, op->d_binoffs,
                                        op->d_binlens, op->d_gfrec);
                 });
                 timed(op, "fold", (double)nrec_max * 16, [&] {
-                    dz::launch_fold3(op->stream, op->d_gfrec, op->d_base,
+                    dz::launch_fold3(op->stream, op->d_gfrec, op->d_base[b],
                                      op->d_binoffs, op->d_binlens, fc,
                                      op->d_slotmap, op->s_cnt, op->s_min,
                                      op->s_max, op->s_sum, op->s_first);
@@ -1567,6 +1579,11 @@ static dz_status process_pending(dz_window_op* op) {
             }
         }
     }
+    /* the fold is the last reader of the per-buffer pipeline state (the
+     * scatter reads the staged inputs + d_gofs[b]; regroup/fold read
+     * d_base[b]): once it completes, buffer b may be restaged */
+    CHK(op, hipEventRecord(op->ev_consumed[b], op->stream));
+    op->consumed_valid[b] = true;
     op->batch_seq++;
 
     /* 4. watermark (running max of batch minimums, :255-266) + trigger */
