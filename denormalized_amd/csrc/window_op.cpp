@@ -1370,11 +1370,17 @@ static dz_status stage_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
 
 /* Sections 2-4 of a push: read the staged scalars, lay out window frames,
  * partition + fold, advance the watermark and trigger closes. */
+static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P);
+
 static dz_status process_pending(dz_window_op* op) {
     if (!op->pend.active) return DZ_OK;
     CHK(op, hipSetDevice(op->device));
     dz_window_op::Pend P = op->pend;
     op->pend.active = false;
+    return process_batch(op, P);
+}
+
+static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
     const bool sliding = P.sliding;
     const bool keys_are_dense = P.dense;
     const int64_t n = P.n, chunk = P.chunk;
@@ -1629,13 +1635,25 @@ extern "C" dz_status dz_window_op_push_device(dz_window_op* op, int64_t n_rows,
         if (ensure_zero_kid(op, n_rows) != DZ_OK) return DZ_ERR;
         d_key_ids = op->d_zero_kid;
     }
-    /* deferred pipeline: process the previous device push (its reduction has
-     * long overlapped our caller's time between pushes), then stage this one
-     * and return — its heavy phase runs at the next call into the op. */
-    if (process_pending(op) != DZ_OK) return DZ_ERR;
-    if (n_rows <= 0) return DZ_OK;
-    return stage_core(op, n_rows, d_ts_ms, d_key_ids, d_vals, nullptr,
-                      /*keys_are_dense=*/true, /*deferred=*/true);
+    /* deferred pipeline: STAGE this batch first — its reduction starts on a
+     * near-idle device (the previous step's kernels have drained during the
+     * caller's gap) and runs ahead of the heavy kernels launched just below
+     * — then process the previous push, whose reduction is long visible. */
+    if (n_rows <= 0) return process_pending(op);
+    int C = (int)std::min<int64_t>(512,
+                                   std::max<int64_t>(1, (n_rows + 8191) / 8192));
+    if (op->pend.active &&
+        (C > op->C_cap || (op->slide_ms == 0 && n_rows > op->rec_cap))) {
+        /* scratch growth quiesces the device: drain the pipeline first */
+        if (process_pending(op) != DZ_OK) return DZ_ERR;
+    }
+    dz_window_op::Pend prev = op->pend;
+    op->pend.active = false;
+    if (stage_core(op, n_rows, d_ts_ms, d_key_ids, d_vals, nullptr,
+                   /*keys_are_dense=*/true, /*deferred=*/true) != DZ_OK)
+        return DZ_ERR;
+    if (prev.active) return process_batch(op, prev);
+    return DZ_OK;
 }
 
 /* Zero-copy variant: the op reads the caller's buffers directly, so they
@@ -1653,11 +1671,22 @@ extern "C" dz_status dz_window_op_push_device_borrowed(
         if (ensure_zero_kid(op, n_rows) != DZ_OK) return DZ_ERR;
         d_key_ids = op->d_zero_kid;
     }
-    if (process_pending(op) != DZ_OK) return DZ_ERR;
-    if (n_rows <= 0) return DZ_OK;
-    return stage_core(op, n_rows, d_ts_ms, d_key_ids, d_vals, nullptr,
-                      /*keys_are_dense=*/true, /*deferred=*/true,
-                      /*borrow=*/true);
+    /* stage first, then process the previous push — see push_device */
+    if (n_rows <= 0) return process_pending(op);
+    int C = (int)std::min<int64_t>(512,
+                                   std::max<int64_t>(1, (n_rows + 8191) / 8192));
+    if (op->pend.active &&
+        (C > op->C_cap || (op->slide_ms == 0 && n_rows > op->rec_cap))) {
+        if (process_pending(op) != DZ_OK) return DZ_ERR;
+    }
+    dz_window_op::Pend prev = op->pend;
+    op->pend.active = false;
+    if (stage_core(op, n_rows, d_ts_ms, d_key_ids, d_vals, nullptr,
+                   /*keys_are_dense=*/true, /*deferred=*/true,
+                   /*borrow=*/true) != DZ_OK)
+        return DZ_ERR;
+    if (prev.active) return process_batch(op, prev);
+    return DZ_OK;
 }
 
 /* ------------------------------------------------------------------ */
