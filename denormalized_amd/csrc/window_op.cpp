@@ -167,12 +167,14 @@ struct dz_window_op {
     int64_t rec_cap = 0;
     uint64_t* d_scalars[2] = {};
     uint64_t* h_scalars = nullptr; /* pinned, 2 x 3 (per pipeline buffer) */
-    int32_t* d_slotmap = nullptr;
-    int slotmap_cap = 0;
+    int32_t* d_slotmap = nullptr;   /* MAX_RANGES, allocated at create */
+    int32_t* h_slotmap[2] = {};     /* pinned upload staging, per pipeline
+                                     * buffer (host reuse must not outrun
+                                     * the async H2D) */
     int32_t* d_zero_kid = nullptr; /* no_group: all rows in group 0 */
     int64_t zero_cap = 0;
-    int32_t* d_resetlist = nullptr;
-    int resetlist_cap = 0;
+    int32_t* d_resetlist = nullptr; /* MAX_RANGES, allocated at create */
+    int32_t* h_resetlist[2] = {};
 
     /* input staging (host-batch path) */
     int64_t* d_ts = nullptr;
@@ -467,6 +469,12 @@ extern "C" dz_window_op* dz_window_op_create(const dz_window_desc* desc) {
     hipMalloc(&op->d_scalars[0], 3 * 8);
     hipMalloc(&op->d_scalars[1], 3 * 8);
     hipHostMalloc((void**)&op->h_scalars, 2 * 3 * 8);
+    hipMalloc(&op->d_slotmap, (size_t)dz::MAX_RANGES * 4);
+    hipMalloc(&op->d_resetlist, (size_t)dz::MAX_RANGES * 4);
+    for (int i = 0; i < 2; i++) {
+        hipHostMalloc((void**)&op->h_slotmap[i], (size_t)dz::MAX_RANGES * 4);
+        hipHostMalloc((void**)&op->h_resetlist[i], (size_t)dz::MAX_RANGES * 4);
+    }
     {
         /* high-priority ingest stream: the next batch's reduction should
          * fill gaps ahead of the current batch's queued partition/fold, not
@@ -544,6 +552,10 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
         if (op->ev_consumed[i]) hipEventDestroy(op->ev_consumed[i]);
     }
     if (op->h_scalars) hipHostFree(op->h_scalars);
+    for (int i = 0; i < 2; i++) {
+        if (op->h_slotmap[i]) hipHostFree(op->h_slotmap[i]);
+        if (op->h_resetlist[i]) hipHostFree(op->h_resetlist[i]);
+    }
     if (op->h_stage) hipHostFree(op->h_stage);
     hipStreamDestroy(op->stream);
     hipStreamDestroy(op->copy_stream);
@@ -1461,26 +1473,22 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
             slotmap[r] = it->second.slot;
         }
     }
-    if ((int)nw > op->slotmap_cap) {
-        hipFree(op->d_slotmap);
-        CHK(op, hipMalloc(&op->d_slotmap, nw * 4));
-        op->slotmap_cap = (int)nw;
-    }
-    if (!reset_list.empty()) { /* one batched reset launch for all new slots */
+    if (!reset_list.empty()) { /* one batched reset launch for all new slots
+                                 * (pinned staging: a pageable-source async
+                                 * copy stages synchronously at enqueue) */
         int ns = (int)reset_list.size();
-        if (ns > op->resetlist_cap) {
-            hipFree(op->d_resetlist);
-            CHK(op, hipMalloc(&op->d_resetlist, (size_t)ns * 4));
-            op->resetlist_cap = ns;
-        }
-        CHK(op, hipMemcpyAsync(op->d_resetlist, reset_list.data(), (size_t)ns * 4,
-                               hipMemcpyHostToDevice, op->stream));
+        memcpy(op->h_resetlist[b], reset_list.data(), (size_t)ns * 4);
+        CHK(op, hipMemcpyAsync(op->d_resetlist, op->h_resetlist[b],
+                               (size_t)ns * 4, hipMemcpyHostToDevice,
+                               op->stream));
         dz::launch_reset_slots(op->stream, op->d_resetlist, ns, op->kcap,
                                op->s_cnt, op->s_first);
     }
-    if (nw > 0)
-        CHK(op, hipMemcpyAsync(op->d_slotmap, slotmap.data(), nw * 4,
+    if (nw > 0) {
+        memcpy(op->h_slotmap[b], slotmap.data(), (size_t)nw * 4);
+        CHK(op, hipMemcpyAsync(op->d_slotmap, op->h_slotmap[b], nw * 4,
                                hipMemcpyHostToDevice, op->stream));
+    }
 
     wp.s0 = nw > 0 ? ws[0] : 0;
     wp.nw = (int32_t)nw;
