@@ -1813,12 +1813,25 @@ extern "C" dz_status dz_window_op_poll(dz_window_op* op, const dz_out_batch** ou
     *out = nullptr;
     if (op->pend.active) {
         /* opportunistic, non-blocking: finish a deferred device push whose
-         * reduction is already host-visible (keeps a poll loop live without
-         * ever stalling the pipelined caller) */
-        hipSetDevice(op->device);
-        if (hipEventQuery(op->ev_ready[op->pend.buf]) == hipSuccess &&
-            process_pending(op) != DZ_OK)
-            return DZ_ERR;
+         * reduction is already host-visible — but only when the caller would
+         * otherwise see nothing at all (keeps a poll-only loop live without
+         * perturbing the push/process phase of a pipelined caller, whose
+         * next push will process it anyway) */
+        bool starved;
+        {
+            std::lock_guard<std::mutex> lk(op->out_mtx);
+            starved = op->outq.empty();
+        }
+        if (starved) {
+            std::lock_guard<std::mutex> lk(op->e_mtx);
+            starved = op->e_inflight == 0 && op->e_jobs.empty();
+        }
+        if (starved) {
+            hipSetDevice(op->device);
+            if (hipEventQuery(op->ev_ready[op->pend.buf]) == hipSuccess &&
+                process_pending(op) != DZ_OK)
+                return DZ_ERR;
+        }
     }
     {
         std::lock_guard<std::mutex> lk(op->out_mtx);
