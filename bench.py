@@ -52,6 +52,8 @@ def parse_args():
                         "borrowed push (also the fallback for A/B runs "
                         "against engine builds without the borrowed entry)")
     p.add_argument("--skip-cpu-baseline", action="store_true")
+    p.add_argument("--debug-steps", action="store_true",
+                   help="print per-step wall times (push / poll split) to stderr")
     p.add_argument("--cpu-sample-rows", type=int, default=8_000_000)
     return p.parse_args()
 
@@ -196,8 +198,26 @@ def main():
         dist.barrier()
 
     t0 = time.perf_counter()
-    for s in range(W, W + K):
-        emitted += push_step(s)
+    if args.debug_steps:
+        marks = []
+        for s in range(W, W + K):
+            a = time.perf_counter()
+            off = s * B
+            op.push_device(B,
+                           ctypes.c_void_p(d_ts.ptr.value + off * 8),
+                           ctypes.c_void_p(d_kid.ptr.value + off * 4),
+                           ctypes.c_void_p(d_vals.ptr.value + off * 8),
+                           borrowed=not args.staged_push)
+            m = time.perf_counter()
+            for bt in op.poll_all(drain=False, copy=False):
+                emitted += bt["n_rows"]
+            marks.append((m - a, time.perf_counter() - m))
+        for i, (pu, po) in enumerate(marks):
+            print(f"step {i} push {pu*1e3:7.3f} ms poll {po*1e3:7.3f} ms",
+                  file=sys.stderr)
+    else:
+        for s in range(W, W + K):
+            emitted += push_step(s)
     op.finish()
     emitted += sum(b["n_rows"] for b in op.poll_all(copy=False))
     dz.synchronize(device)
