@@ -36,8 +36,8 @@ ALG_BYTES_PER_ROW = 20.0  # key id 4B + reading 8B + ts 8B
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=24)
-    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--steps", type=int, default=96)
+    p.add_argument("--warmup", type=int, default=6)
     p.add_argument("--rows-per-step", type=int, default=8_000_000)
     p.add_argument("--keys", type=int, default=10_000)
     p.add_argument("--rows-per-ms", type=int, default=1000)
@@ -212,13 +212,16 @@ def main():
             for bt in op.poll_all(drain=False, copy=False):
                 emitted += bt["n_rows"]
             marks.append((m - a, time.perf_counter() - m))
-        for i, (pu, po) in enumerate(marks):
-            print(f"step {i} push {pu*1e3:7.3f} ms poll {po*1e3:7.3f} ms",
-                  file=sys.stderr)
+        tf = time.perf_counter()
     else:
         for s in range(W, W + K):
             emitted += push_step(s)
     op.finish()
+    if args.debug_steps:
+        print(f"finish {(time.perf_counter()-tf)*1e3:.3f} ms", file=sys.stderr)
+        for i, (pu, po) in enumerate(marks):
+            print(f"step {i} push {pu*1e3:7.3f} ms poll {po*1e3:7.3f} ms",
+                  file=sys.stderr)
     emitted += sum(b["n_rows"] for b in op.poll_all(copy=False))
     dz.synchronize(device)
     if dist is not None:
