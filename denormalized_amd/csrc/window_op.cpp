@@ -107,7 +107,13 @@ struct dz_window_op {
     std::string err;
 
     hipStream_t stream = nullptr;
-    hipStream_t copy_stream = nullptr; /* emission D2H runs off the compute stream */
+    /* emission runs off the compute stream; the per-close device chains
+     * (slabread -> sort -> gather) of different closes are independent, so
+     * they round-robin over E_CSTREAMS streams (each with its own radix
+     * scratch) instead of serializing on one */
+    static constexpr int E_CSTREAMS = 4;
+    hipStream_t copy_stream = nullptr;   /* == c_streams[0] */
+    hipStream_t c_streams[E_CSTREAMS] = {};
 
     /* dictionary (host side; first-seen insertion order == emitted id order
      * is NOT assumed — emission sorts by first row; the dict only maps
@@ -237,8 +243,8 @@ struct dz_window_op {
     DevEmit e_dev[E_POOL];
     uint64_t* e_slabs[E_POOL] = {}; /* pinned output staging per slot */
     uint32_t* e_pcnt = nullptr;     /* pinned per-slot nt counters */
-    uint32_t* d_rhist = nullptr;    /* shared radix scratch (stream-serial) */
-    uint32_t* d_roffs = nullptr;
+    uint32_t* d_rhist[E_CSTREAMS] = {}; /* radix scratch, one per c_stream */
+    uint32_t* d_roffs[E_CSTREAMS] = {};
     int64_t e_slab_kcap = 0;
     struct EmitJob {
         hipEvent_t ev;
@@ -246,6 +252,7 @@ struct dz_window_op {
         int64_t wstart, wend, n_keys, kcap;
         uint64_t ticket;
         bool device; /* device-sorted columns vs raw host slab */
+        int cs = 0;  /* c_streams index this close's device chain runs on */
         /* host-path closes of one trigger group share ONE event (their slab
          * reads ride one gather + one D2H); the last sibling to finish
          * returns the event and the group buffer to their pools */
@@ -417,7 +424,9 @@ static dz_status state_alloc(dz_window_op* op, int64_t kcap_new, int32_t nslots_
                                    span * 8, hipMemcpyDeviceToDevice, op->stream));
     }
     CHK(op, hipStreamSynchronize(op->stream));
-    CHK(op, hipStreamSynchronize(op->copy_stream)); /* pending emission copies read s_base */
+    for (int s = 0; s < dz_window_op::E_CSTREAMS; s++) /* emission reads s_base */
+        if (op->c_streams[s])
+            CHK(op, hipStreamSynchronize(op->c_streams[s]));
     hipFree(op->s_base);
     op->s_base = n_base;
     op->s_cnt = n_cnt; op->s_first = n_first; op->s_min = n_min;
@@ -459,12 +468,20 @@ extern "C" dz_window_op* dz_window_op_create(const dz_window_desc* desc) {
     op->aggs.assign(desc->aggs, desc->aggs + desc->n_aggs);
     op->device = desc->device;
     op->max_open = desc->max_open_windows > 0 ? desc->max_open_windows : 4096;
+    bool cs_ok = true;
     if (hipSetDevice(op->device) != hipSuccess ||
         hipStreamCreate(&op->stream) != hipSuccess ||
         hipStreamCreate(&op->copy_stream) != hipSuccess) {
         g_err = "hip device/stream init failed";
         delete op;
         return nullptr;
+    }
+    op->c_streams[0] = op->copy_stream;
+    for (int i = 1; i < dz_window_op::E_CSTREAMS; i++)
+        cs_ok = cs_ok && hipStreamCreate(&op->c_streams[i]) == hipSuccess;
+    if (!cs_ok) { /* degrade to one emission stream */
+        for (int i = 1; i < dz_window_op::E_CSTREAMS; i++)
+            op->c_streams[i] = op->copy_stream;
     }
     hipMalloc(&op->d_scalars[0], 3 * 8);
     hipMalloc(&op->d_scalars[1], 3 * 8);
@@ -530,8 +547,10 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     for (auto& d : op->e_dev)
         if (d.base) hipFree(d.base);
     if (op->e_pcnt) hipHostFree(op->e_pcnt);
-    hipFree(op->d_rhist);
-    hipFree(op->d_roffs);
+    for (int s = 0; s < dz_window_op::E_CSTREAMS; s++) {
+        hipFree(op->d_rhist[s]);
+        hipFree(op->d_roffs[s]);
+    }
     hipFree(op->s_base);
     hipFree(op->d_ghist[0]); hipFree(op->d_ghist[1]);
     hipFree(op->d_gofs[0]); hipFree(op->d_gofs[1]);
@@ -559,6 +578,9 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     if (op->h_stage) hipHostFree(op->h_stage);
     hipStreamDestroy(op->stream);
     hipStreamDestroy(op->copy_stream);
+    for (int s = 1; s < dz_window_op::E_CSTREAMS; s++)
+        if (op->c_streams[s] && op->c_streams[s] != op->copy_stream)
+            hipStreamDestroy(op->c_streams[s]);
     if (op->i_stream) hipStreamDestroy(op->i_stream);
     for (auto& fs : op->free_slots)
         if (fs.ev) hipEventDestroy(fs.ev);
@@ -880,23 +902,24 @@ static void emit_worker_main(dz_window_op* op) {
                 dz_window_op::DevEmit& d = op->e_dev[job.slab];
                 char* s = (char*)op->e_slabs[job.slab];
                 int64_t kc = job.kcap;
+                hipStream_t wcs = op->c_streams[job.cs];
                 hipMemcpyAsync(s, d.fkid, (size_t)nt * 4, hipMemcpyDeviceToHost,
-                               op->copy_stream);
+                               wcs);
                 hipMemcpyAsync(s + kc * 4, d.fiota, (size_t)nt * 4,
-                               hipMemcpyDeviceToHost, op->copy_stream);
+                               hipMemcpyDeviceToHost, wcs);
                 hipMemcpyAsync(s + kc * 8, d.oflags, nt, hipMemcpyDeviceToHost,
-                               op->copy_stream);
+                               wcs);
                 hipMemcpyAsync(s + kc * 9, d.ocnt, (size_t)nt * 8,
-                               hipMemcpyDeviceToHost, op->copy_stream);
+                               hipMemcpyDeviceToHost, wcs);
                 hipMemcpyAsync(s + kc * 17, d.omin, (size_t)nt * 8,
-                               hipMemcpyDeviceToHost, op->copy_stream);
+                               hipMemcpyDeviceToHost, wcs);
                 hipMemcpyAsync(s + kc * 25, d.omax, (size_t)nt * 8,
-                               hipMemcpyDeviceToHost, op->copy_stream);
+                               hipMemcpyDeviceToHost, wcs);
                 hipMemcpyAsync(s + kc * 33, d.osum, (size_t)nt * 8,
-                               hipMemcpyDeviceToHost, op->copy_stream);
+                               hipMemcpyDeviceToHost, wcs);
                 hipMemcpyAsync(s + kc * 41, d.oavg, (size_t)nt * 8,
-                               hipMemcpyDeviceToHost, op->copy_stream);
-                hipEventRecord(job.ev, op->copy_stream);
+                               hipMemcpyDeviceToHost, wcs);
+                hipEventRecord(job.ev, wcs);
                 event_spin(job.ev);
             }
             build_emission(op, job.wstart, job.wend, nt, job.kcap,
@@ -944,12 +967,15 @@ static dz_status ensure_emission(dz_window_op* op) {
         emit_drain(op);
         int64_t kc = op->kcap;
         int64_t nblk = (kc + dz::EMIT_RCHUNK - 1) / dz::EMIT_RCHUNK;
-        hipFree(op->d_rhist);
-        hipFree(op->d_roffs);
-        /* + tail: RSEG partial rows + digit bases for the parallel scan */
-        CHK(op, hipMalloc(&op->d_rhist,
-                          (size_t)(nblk + 17) * dz::EMIT_RBINS * 4));
-        CHK(op, hipMalloc(&op->d_roffs, (size_t)nblk * dz::EMIT_RBINS * 4));
+        for (int s = 0; s < dz_window_op::E_CSTREAMS; s++) {
+            hipFree(op->d_rhist[s]);
+            hipFree(op->d_roffs[s]);
+            /* + tail: RSEG partial rows + digit bases for the parallel scan */
+            CHK(op, hipMalloc(&op->d_rhist[s],
+                              (size_t)(nblk + 17) * dz::EMIT_RBINS * 4));
+            CHK(op, hipMalloc(&op->d_roffs[s],
+                              (size_t)nblk * dz::EMIT_RBINS * 4));
+        }
         if (!op->e_pcnt)
             CHK(op, hipHostMalloc((void**)&op->e_pcnt, dz_window_op::E_POOL * 4));
         for (int i = 0; i < dz_window_op::E_POOL; i++) {
@@ -1040,7 +1066,8 @@ static dz_status trigger_windows(dz_window_op* op) {
     {
         HostTimer htf(op, "h_trig_fence");
         CHK(op, hipEventRecord(evA, op->stream));
-        CHK(op, hipStreamWaitEvent(op->copy_stream, evA, 0));
+        for (int s = 0; s < dz_window_op::E_CSTREAMS; s++)
+            CHK(op, hipStreamWaitEvent(op->c_streams[s], evA, 0));
     }
     {
         std::lock_guard<std::mutex> lk(op->e_mtx);
@@ -1170,16 +1197,17 @@ static dz_status trigger_windows(dz_window_op* op) {
         dz_window_op::DevEmit& d = op->e_dev[slab];
         const uint64_t* sl = op->s_base + (size_t)c.slot * stride;
         const bool dev_path = op->n_keys > 65536;
+        hipStream_t cs = op->c_streams[(ci - g0) % dz_window_op::E_CSTREAMS];
         if (!dev_path) {
             /* small keyspaces: ONE pinned copy of the raw slab; the worker
              * sorts/filters on host (15 stream ops per close would cost more
              * than the whole host build at this scale) */
             CHK(op, hipMemcpyAsync(op->e_slabs[slab], sl, stride * 8,
-                                   hipMemcpyDeviceToHost, op->copy_stream));
+                                   hipMemcpyDeviceToHost, cs));
         } else {
-        CHK(op, hipMemsetAsync(d.counter, 0, 8, op->copy_stream));
+        CHK(op, hipMemsetAsync(d.counter, 0, 8, cs));
         if (op->n_keys > 0) {
-            dz::launch_emission_slabread(op->copy_stream,
+            dz::launch_emission_slabread(cs,
                                          /*first*/ sl + op->kcap,
                                          /*cnt*/ sl,
                                          /*min*/ (const double*)(sl + 2 * op->kcap),
@@ -1191,7 +1219,7 @@ static dz_status trigger_windows(dz_window_op* op) {
                                          d.omax, d.osum, d.oavg, d.oflags);
         }
         }
-        CHK(op, hipEventRecord(slot_ev, op->copy_stream));
+        CHK(op, hipEventRecord(slot_ev, cs));
         op->free_slots.push_back({c.slot, slot_ev});
         pend.push_back({slab, ev});
     }
@@ -1202,19 +1230,30 @@ static dz_status trigger_windows(dz_window_op* op) {
         hipEvent_t ev = pend[ci - g0].ev;
         dz_window_op::DevEmit& d = op->e_dev[slab];
         const bool dev_path = op->n_keys > 65536;
+        int csi = (int)((ci - g0) % dz_window_op::E_CSTREAMS);
+        hipStream_t cs = op->c_streams[csi];
         if (dev_path && op->n_keys > 0) {
-            dz::launch_emission_sort(op->copy_stream, op->n_keys, d.fkeys,
+            dz::launch_emission_sort(cs, op->n_keys, d.fkeys,
                                      d.skeys, d.fiota, d.okid, d.counter + 1,
-                                     op->d_rhist, op->d_roffs);
+                                     op->d_rhist[csi], op->d_roffs[csi]);
         }
         if (dev_path)
             CHK(op, hipMemcpyAsync(&op->e_pcnt[slab], d.counter + 1, 4,
-                                   hipMemcpyDeviceToHost, op->copy_stream));
-        CHK(op, hipEventRecord(ev, op->copy_stream));
+                                   hipMemcpyDeviceToHost, cs));
+        CHK(op, hipEventRecord(ev, cs));
         {
             std::lock_guard<std::mutex> lk(op->e_mtx);
-            op->e_jobs.push_back({ev, slab, c.start, c.end, op->n_keys, op->kcap,
-                                  op->e_ticket_next++, dev_path});
+            dz_window_op::EmitJob j;
+            j.ev = ev;
+            j.slab = slab;
+            j.wstart = c.start;
+            j.wend = c.end;
+            j.n_keys = op->n_keys;
+            j.kcap = op->kcap;
+            j.ticket = op->e_ticket_next++;
+            j.device = dev_path;
+            j.cs = csi;
+            op->e_jobs.push_back(std::move(j));
         }
     }
     op->e_cv.notify_all(); /* one wakeup per group, not per close */
