@@ -228,10 +228,7 @@ struct dz_window_op {
      * bursts (≈8 closes/step at cfg2) need 2-3 steps of slack or the enqueue
      * path sleeps on e_cv waiting for a free slab. */
     static constexpr int E_POOL = 32;
-    static constexpr int E_WORKERS = 8; /* cfg3-scale builds are ~2 ms each
-                                         * (nt up to ~1M rows); 4 workers
-                                         * saturated and back-pressured the
-                                         * slab pool */
+    static constexpr int E_WORKERS = 4;
     struct DevEmit { /* per-slot device scratch, carved from one alloc */
         char* base = nullptr;
         uint64_t* ekeys; uint32_t* ekid;      /* touched groups (compact) */
