@@ -841,3 +841,22 @@ def test_push_device_borrowed_parity():
         assert np.array_equal(outs[True][f], exp[f]), f
     for a in (d_ts, d_kid, d_vals):
         a.free()
+
+
+@pytest.mark.gpu
+def test_device_emission_many_concurrent_closes():
+    # many device-path closes in flight at once: their per-close device
+    # chains round-robin over the emission copy streams; results must stay
+    # bit-exact and ticket-ordered regardless of chain interleaving
+    rng = np.random.default_rng(94)
+    n = 2_000_000
+    # 25 windows of 1s, ~80k rows each, 80k keys => device emission path,
+    # closes arrive in bursts of ~12 per push
+    ts = (1_000_000 + np.arange(n) // 80).astype(np.int64)
+    k = rng.integers(0, 80_000, n)
+    v = rng.uniform(0, 115, n)
+    outs, exp = run_both(1000, 0, [(ts[:n // 2], k[:n // 2], v[:n // 2]),
+                                   (ts[n // 2:], k[n // 2:], v[n // 2:])],
+                         n_keys_hint=80_000)
+    assert len({int(b["window_start"][0]) for b in outs if b["n_rows"]}) >= 20
+    assert_parity(outs, exp)
