@@ -887,6 +887,14 @@ static void event_spin(hipEvent_t ev) {
         std::this_thread::yield();
 }
 
+/* worker-side variant: a tight hipEventQuery loop from several worker
+ * threads contends the runtime lock against the push thread's own queries
+ * and launches; the workers can afford 20 µs of extra latency */
+static void event_spin_relaxed(hipEvent_t ev) {
+    while (hipEventQuery(ev) != hipSuccess)
+        std::this_thread::sleep_for(std::chrono::microseconds(20));
+}
+
 static void emit_worker_main(dz_window_op* op) {
     hipSetDevice(op->device);
     for (;;) {
@@ -901,7 +909,7 @@ static void emit_worker_main(dz_window_op* op) {
             job = op->e_jobs.front();
             op->e_jobs.pop_front();
         }
-        event_spin(job.ev); /* emission kernels (+ host-path D2H) complete */
+        event_spin_relaxed(job.ev); /* emission chain complete */
         auto t0 = std::chrono::steady_clock::now();
         OutBuf ob;
         if (job.device) {
@@ -930,7 +938,7 @@ static void emit_worker_main(dz_window_op* op) {
                 hipMemcpyAsync(s + kc * 41, d.oavg, (size_t)nt * 8,
                                hipMemcpyDeviceToHost, wcs);
                 hipEventRecord(job.ev, wcs);
-                event_spin(job.ev);
+                event_spin_relaxed(job.ev);
             }
             build_emission(op, job.wstart, job.wend, nt, job.kcap,
                            op->e_slabs[job.slab], &ob);
