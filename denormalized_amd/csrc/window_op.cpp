@@ -1461,7 +1461,24 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
     const uint8_t* d_valbm = P.valbm;
     {
         HostTimer ht(op, "h_minmax_sync");
-        event_spin(op->ev_ready[b]);
+        static const bool dbg = getenv("DZ_SPIN_DEBUG") != nullptr;
+        if (dbg) {
+            auto t0 = std::chrono::steady_clock::now();
+            long it = 0;
+            auto tq0 = t0;
+            while (hipEventQuery(op->ev_ready[b]) != hipSuccess) {
+                std::this_thread::yield();
+                it++;
+            }
+            auto t1 = std::chrono::steady_clock::now();
+            double first_q = std::chrono::duration<double, std::micro>(
+                                 std::chrono::steady_clock::now() - tq0).count();
+            (void)first_q;
+            fprintf(stderr, "[spin] iters=%ld us=%.1f\n", it,
+                    std::chrono::duration<double, std::micro>(t1 - t0).count());
+        } else {
+            event_spin(op->ev_ready[b]);
+        }
     }
     /* compute-stream work below consumes i_stream-written buffers */
     CHK(op, hipStreamWaitEvent(op->stream, op->ev_ready[b], 0));
