@@ -1465,15 +1465,11 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
         if (dbg) {
             auto t0 = std::chrono::steady_clock::now();
             long it = 0;
-            auto tq0 = t0;
             while (hipEventQuery(op->ev_ready[b]) != hipSuccess) {
                 std::this_thread::yield();
                 it++;
             }
             auto t1 = std::chrono::steady_clock::now();
-            double first_q = std::chrono::duration<double, std::micro>(
-                                 std::chrono::steady_clock::now() - tq0).count();
-            (void)first_q;
             fprintf(stderr, "[spin] iters=%ld us=%.1f\n", it,
                     std::chrono::duration<double, std::micro>(t1 - t0).count());
         } else {
