@@ -860,3 +860,51 @@ def test_device_emission_many_concurrent_closes():
                          n_keys_hint=80_000)
     assert len({int(b["window_start"][0]) for b in outs if b["n_rows"]}) >= 20
     assert_parity(outs, exp)
+
+
+@pytest.mark.gpu
+def test_soak_mixed_push_kinds():
+    # interleave host pushes (synchronous), staged device pushes (deferred)
+    # and borrowed device pushes (zero-copy deferred) on ONE op, with polls
+    # and external watermark advances between them: the pipeline must flush
+    # and order correctly across every transition
+    from denormalized_amd import DeviceArray, WindowOp, _lib
+    rng = np.random.default_rng(101)
+    nkeys = 300
+    op = WindowOp(length_ms=1000, key_kind=_lib.KEY_DENSE_INT64,
+                  n_keys_hint=nkeys)
+    o = pyoracle.Oracle(1000, 0)
+    t = 2_000_000
+    outs = []
+    bufs = []
+    for step in range(60):
+        n = int(rng.integers(500, 20_000))
+        ts = (t + np.cumsum(rng.integers(0, 2, n))).astype(np.int64)
+        t = int(ts.max())
+        k = rng.integers(0, nkeys, n)
+        v = rng.uniform(0, 115, n)
+        kind = step % 3
+        if kind == 0:
+            op.push(ts, k, v)
+        else:
+            d_ts = DeviceArray(0, n * 8)
+            d_k = DeviceArray(0, n * 4)
+            d_v = DeviceArray(0, n * 8)
+            d_ts.from_host(ts)
+            d_k.from_host(k.astype(np.int32))
+            d_v.from_host(v)
+            op.push_device(n, d_ts.ptr, d_k.ptr, d_v.ptr, borrowed=(kind == 2))
+            bufs.append((d_ts, d_k, d_v))  # keep alive (borrowed contract)
+        o.push(ts, k, v)
+        if step % 7 == 3:
+            op.advance_watermark(int(op.watermark))
+        outs += op.poll_all(drain=False)
+    op.finish()
+    o.finish()
+    outs += op.poll_all()
+    assert_parity(outs, o.fetch())
+    op.close()
+    for bs in bufs:
+        for a in bs:
+            a.free()
+    o.close()
