@@ -37,8 +37,7 @@ struct FoldChunk {
     int64_t kcap;        /* state key capacity (multiple of NB) */
     uint32_t batch_seq;
     int32_t bin_stride;  /* binoffs/binlens row stride per bucket */
-    int32_t tl_nw;       /* two-level mode: windows per batch (0 = off) */  /* binoffs/binlens row stride (GCAP, or gtot in the
-                          * single-launch big-bin regime) */
+    int32_t tl_nw;       /* two-level mode: windows per batch (0 = off) */
 };
 
 /* Launch wrappers implemented in kernels.hip. All run on `stream`;
@@ -129,5 +128,6 @@ struct EGatherSlots { int32_t s[16]; };
 void launch_egather_slabs(hipStream_t stream, const uint64_t* s_base,
                           int64_t stride_u64, EGatherSlots slots, int gcount,
                           uint64_t* out);
+void launch_arm_scalars(hipStream_t stream, uint64_t* scalars);
 
 } // namespace dz

@@ -1201,3 +1201,16 @@ void dz::launch_egather_slabs(hipStream_t s, const uint64_t* s_base,
     hipLaunchKernelGGL(k_egather_slabs, grid, dim3(dz::BLOCK), 0, s, s_base,
                        stride_u64, slots, out);
 }
+
+/* arm the per-push scalar block (min=+inf pattern, max/kid=0) in ONE tiny
+ * launch: the generic fill kernel costs ~20 us per call on a busy device
+ * and two of them sat on the ingest reduction's critical path */
+__global__ void k_arm_scalars(uint64_t* __restrict__ s) {
+    s[0] = ~0ULL;
+    s[1] = 0ULL;
+    s[2] = 0ULL;
+}
+
+void dz::launch_arm_scalars(hipStream_t st, uint64_t* s) {
+    hipLaunchKernelGGL(k_arm_scalars, dim3(1), dim3(1), 0, st, s);
+}

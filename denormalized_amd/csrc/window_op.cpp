@@ -1387,8 +1387,7 @@ static dz_status stage_core(dz_window_op* op, int64_t n, const int64_t* d_ts,
          * covers compute-stream reuse, this wait covers the cross-stream case */
         CHK(op, hipStreamWaitEvent(s, op->ev_consumed[b], 0));
     }
-    CHK(op, hipMemsetAsync(op->d_scalars[b], 0xFF, 8, s));
-    CHK(op, hipMemsetAsync(op->d_scalars[b] + 1, 0, 16, s));
+    dz::launch_arm_scalars(s, op->d_scalars[b]);
     dz::WinParams wp;
     memset(&wp, 0, sizeof(wp));
     wp.len_ms = op->len_ms;
