@@ -129,5 +129,6 @@ void launch_egather_slabs(hipStream_t stream, const uint64_t* s_base,
                           int64_t stride_u64, EGatherSlots slots, int gcount,
                           uint64_t* out);
 void launch_arm_scalars(hipStream_t stream, uint64_t* scalars);
+void launch_zero_counters(hipStream_t stream, uint32_t* two_u32);
 
 } // namespace dz

@@ -1214,3 +1214,12 @@ __global__ void k_arm_scalars(uint64_t* __restrict__ s) {
 void dz::launch_arm_scalars(hipStream_t st, uint64_t* s) {
     hipLaunchKernelGGL(k_arm_scalars, dim3(1), dim3(1), 0, st, s);
 }
+
+__global__ void k_zero2(uint32_t* __restrict__ p) {
+    p[0] = 0;
+    p[1] = 0;
+}
+
+void dz::launch_zero_counters(hipStream_t st, uint32_t* p) {
+    hipLaunchKernelGGL(k_zero2, dim3(1), dim3(1), 0, st, p);
+}

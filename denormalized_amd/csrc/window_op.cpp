@@ -1223,7 +1223,7 @@ static dz_status trigger_windows(dz_window_op* op) {
             CHK(op, hipMemcpyAsync(op->e_slabs[slab], sl, stride * 8,
                                    hipMemcpyDeviceToHost, cs));
         } else {
-        CHK(op, hipMemsetAsync(d.counter, 0, 8, cs));
+        dz::launch_zero_counters(cs, d.counter);
         if (op->n_keys > 0) {
             dz::launch_emission_slabread(cs,
                                          /*first*/ sl + op->kcap,
