@@ -166,6 +166,20 @@ def main():
 
     def push_step(step):
         off = step * B
+        if dist is not None:
+            # exchange the (one-batch-lagged) local watermarks BEFORE the
+            # push: advance_watermark flushes the deferred previous batch at
+            # exactly the point the push would have processed it anyway, so
+            # the multi-rank step keeps the single-rank pipelining; the
+            # one-batch lag is the reference's own async watermark
+            # propagation semantics (late rows re-emit their frame)
+            import torch
+            wm = torch.tensor([op.watermark], dtype=torch.int64)
+            if torch.cuda.is_available() and dist.get_backend() == "nccl":
+                wm = wm.cuda()
+            dist.all_reduce(wm, op=dist.ReduceOp.MAX)
+            if int(wm.item()) != -(2**63):  # unset on the very first step
+                op.advance_watermark(int(wm.item()))
         # borrowed (zero-copy) push: the pre-generated stream stays resident
         # and untouched for the whole run, exactly the lifetime the borrowed
         # contract asks for
@@ -174,13 +188,6 @@ def main():
                        ctypes.c_void_p(d_kid.ptr.value + off * 4),
                        ctypes.c_void_p(d_vals.ptr.value + off * 8),
                        borrowed=not args.staged_push)
-        if dist is not None:
-            import torch
-            wm = torch.tensor([op.watermark], dtype=torch.int64)
-            if torch.cuda.is_available() and dist.get_backend() == "nccl":
-                wm = wm.cuda()
-            dist.all_reduce(wm, op=dist.ReduceOp.MAX)
-            op.advance_watermark(int(wm.item()))
         emitted = 0
         # non-blocking zero-copy poll: emission is pipelined on the op's
         # worker pool and overlaps the next step's kernels; consuming the
