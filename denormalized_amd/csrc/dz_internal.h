@@ -130,5 +130,11 @@ void launch_egather_slabs(hipStream_t stream, const uint64_t* s_base,
                           uint64_t* out);
 void launch_arm_scalars(hipStream_t stream, uint64_t* scalars);
 void launch_zero_counters(hipStream_t stream, uint32_t* two_u32);
+void launch_emission_permute(hipStream_t stream, int64_t K,
+                             const uint32_t* counter2, const uint32_t* sidx,
+                             const uint32_t* fkid, const uint64_t* ocnt,
+                             const double* omin, const double* omax,
+                             const double* osum, const double* oavg,
+                             const uint8_t* oflags, char* out);
 
 } // namespace dz

@@ -1223,3 +1223,57 @@ __global__ void k_zero2(uint32_t* __restrict__ p) {
 void dz::launch_zero_counters(hipStream_t st, uint32_t* p) {
     hipLaunchKernelGGL(k_zero2, dim3(1), dim3(1), 0, st, p);
 }
+
+/* apply the sorted permutation to every output column ON DEVICE, packing
+ * the final-order columns into one contiguous block: the worker then pulls
+ * ONE span and builds with sequential copies (the host-side gather through
+ * sidx over ~1M-row closes was the cfg3 build bottleneck). Layout (by nt,
+ * device-computed from counter2): [key i64][cnt u64][min][max][sum][avg]
+ * [kid u32][flags u8] = 53 B/row. */
+__global__ void k_epermute(const uint32_t* __restrict__ counter2,
+                           const uint32_t* __restrict__ sidx,
+                           const uint32_t* __restrict__ fkid,
+                           const uint64_t* __restrict__ ocnt,
+                           const double* __restrict__ omin,
+                           const double* __restrict__ omax,
+                           const double* __restrict__ osum,
+                           const double* __restrict__ oavg,
+                           const uint8_t* __restrict__ oflags,
+                           char* __restrict__ out) {
+    const uint32_t nt = *counter2;
+    int64_t* pkey = (int64_t*)out;
+    uint64_t* pcnt = (uint64_t*)(out + (size_t)nt * 8);
+    double* pmin = (double*)(out + (size_t)nt * 16);
+    double* pmax = (double*)(out + (size_t)nt * 24);
+    double* psum = (double*)(out + (size_t)nt * 32);
+    double* pavg = (double*)(out + (size_t)nt * 40);
+    uint32_t* pkid = (uint32_t*)(out + (size_t)nt * 48);
+    uint8_t* pfl = (uint8_t*)(out + (size_t)nt * 52);
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nt;
+         i += stride) {
+        const uint32_t j = sidx[i];
+        const uint32_t k = fkid[j];
+        pkey[i] = (int64_t)k;
+        pcnt[i] = ocnt[j];
+        pmin[i] = omin[j];
+        pmax[i] = omax[j];
+        psum[i] = osum[j];
+        pavg[i] = oavg[j];
+        pkid[i] = k;
+        pfl[i] = oflags[j];
+    }
+}
+
+void dz::launch_emission_permute(hipStream_t st, int64_t K,
+                                 const uint32_t* counter2, const uint32_t* sidx,
+                                 const uint32_t* fkid, const uint64_t* ocnt,
+                                 const double* omin, const double* omax,
+                                 const double* osum, const double* oavg,
+                                 const uint8_t* oflags, char* out) {
+    int blocks = (int)std::min<int64_t>((K + dz::BLOCK - 1) / dz::BLOCK, 2048);
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(k_epermute, dim3(blocks), dim3(dz::BLOCK), 0, st,
+                       counter2, sidx, fkid, ocnt, omin, omax, osum, oavg,
+                       oflags, out);
+}

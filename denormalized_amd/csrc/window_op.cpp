@@ -240,6 +240,7 @@ struct dz_window_op {
         uint64_t* ocnt;
         double* omin; double* omax; double* osum; double* oavg;
         uint8_t* oflags;
+        char* pout;  /* permuted, packed final-order columns (53 B/entry) */
     };
     DevEmit e_dev[E_POOL];
     uint64_t* e_slabs[E_POOL] = {}; /* pinned output staging per slot */
@@ -762,53 +763,30 @@ static void build_emission_host(dz_window_op* op, int64_t wstart, int64_t wend,
  * column sections; kid/cols are in COMPACT order, sidx is the sorted
  * (insertion-order) permutation of compact indices. kcap is a multiple of
  * NB=512, so every section stays 8-byte aligned. */
-struct SlabView {
-    const uint32_t* kid;
-    const uint32_t* sidx;
-    const uint8_t* flags;
-    const uint64_t* cnt;
-    const double* mn;
-    const double* mx;
-    const double* sm;
-    const double* av;
-};
-static SlabView slab_view(const uint64_t* slab, int64_t kcap) {
-    const char* p = (const char*)slab;
-    SlabView v;
-    v.kid = (const uint32_t*)p;
-    v.sidx = (const uint32_t*)(p + kcap * 4);
-    v.flags = (const uint8_t*)(p + kcap * 8);
-    v.cnt = (const uint64_t*)(p + kcap * 9);
-    v.mn = (const double*)(p + kcap * 17);
-    v.mx = (const double*)(p + kcap * 25);
-    v.sm = (const double*)(p + kcap * 33);
-    v.av = (const double*)(p + kcap * 41);
-    return v;
-}
-static constexpr int64_t SLAB_BYTES_PER_ENTRY = 49;
+static constexpr int64_t SLAB_BYTES_PER_ENTRY = 56;
 
 /* Build one emitted batch from the device-sorted, device-filtered columns
  * (insertion order already established by the GPU radix sort). Runs on the
  * emission worker pool: touches only immutable config, the deque-backed
  * dictionaries (indices < the job's n_keys snapshot) and the pinned slab. */
 static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
-                           uint32_t nt, int64_t kcap, const uint64_t* slab,
-                           OutBuf* out) {
-    SlabView v = slab_view(slab, kcap);
+                           uint32_t nt, const uint64_t* slab, OutBuf* out) {
+    /* packed final-order columns from k_epermute (53 B/row): sequential
+     * reads only — the per-row permutation gather already ran on device */
+    const char* p = (const char*)slab;
+    const int64_t* pkey = (const int64_t*)p;
+    const uint64_t* pcnt = (const uint64_t*)(p + (size_t)nt * 8);
+    const double* pmin = (const double*)(p + (size_t)nt * 16);
+    const double* pmax = (const double*)(p + (size_t)nt * 24);
+    const double* psum = (const double*)(p + (size_t)nt * 32);
+    const double* pavg = (const double*)(p + (size_t)nt * 40);
+    const uint32_t* pkid = (const uint32_t*)(p + (size_t)nt * 48);
+    const uint8_t* pfl = (const uint8_t*)(p + (size_t)nt * 52);
+    size_t n = nt;
     size_t na = op->aggs.size();
     OutBuf ob;
     ob.agg_i64.resize(na);
     ob.agg_f64.resize(na);
-    /* walk groups in insertion order (the device-sorted permutation) and
-     * keep filter passers (bit1 of flags; device applied the predicate —
-     * datastream.rs:94-105 semantics incl. NULL-never-passes) */
-    std::vector<uint32_t> rows;
-    rows.reserve(nt);
-    for (uint32_t i = 0; i < nt; i++) {
-        uint32_t j = v.sidx[i];
-        if (v.flags[j] & 2) rows.push_back(j);
-    }
-    size_t n = rows.size();
     if (op->no_group) {
         /* global aggregate: output schema has no group column
          * (create_schema with empty group exprs, streaming_window.rs:1096+) */
@@ -816,58 +794,35 @@ static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
         ob.key_offsets.resize(n + 1);
         ob.key_offsets[0] = 0;
         size_t total = 0;
-        for (size_t i = 0; i < n; i++) total += op->dict_strs[v.kid[rows[i]]].size();
+        for (size_t i = 0; i < n; i++) total += op->dict_strs[pkid[i]].size();
         ob.key_data.resize(total);
         size_t pos = 0;
         for (size_t i = 0; i < n; i++) {
-            const std::string& s = op->dict_strs[v.kid[rows[i]]];
+            const std::string& s = op->dict_strs[pkid[i]];
             memcpy(ob.key_data.data() + pos, s.data(), s.size());
             pos += s.size();
             ob.key_offsets[i + 1] = (int32_t)pos;
         }
     } else if (op->key_kind == DZ_KEY_INT64) {
         ob.key_i64.resize(n);
-        for (size_t i = 0; i < n; i++) ob.key_i64[i] = op->dict_vals[v.kid[rows[i]]];
+        for (size_t i = 0; i < n; i++) ob.key_i64[i] = op->dict_vals[pkid[i]];
     } else {
-        ob.key_i64.resize(n);
-        for (size_t i = 0; i < n; i++) ob.key_i64[i] = v.kid[rows[i]];
+        ob.key_i64.assign(pkey, pkey + n);
     }
     for (size_t a = 0; a < na; a++) {
         switch (op->aggs[a].op) {
-            case DZ_AGG_COUNT: {
-                auto& col = ob.agg_i64[a];
-                col.resize(n);
-                for (size_t i = 0; i < n; i++) col[i] = (int64_t)v.cnt[rows[i]];
+            case DZ_AGG_COUNT:
+                ob.agg_i64[a].assign((const int64_t*)pcnt,
+                                     (const int64_t*)pcnt + n);
                 break;
-            }
-            case DZ_AGG_MIN: {
-                auto& col = ob.agg_f64[a];
-                col.resize(n);
-                for (size_t i = 0; i < n; i++) col[i] = v.mn[rows[i]];
-                break;
-            }
-            case DZ_AGG_MAX: {
-                auto& col = ob.agg_f64[a];
-                col.resize(n);
-                for (size_t i = 0; i < n; i++) col[i] = v.mx[rows[i]];
-                break;
-            }
-            case DZ_AGG_SUM: {
-                auto& col = ob.agg_f64[a];
-                col.resize(n);
-                for (size_t i = 0; i < n; i++) col[i] = v.sm[rows[i]];
-                break;
-            }
-            case DZ_AGG_AVG: {
-                auto& col = ob.agg_f64[a];
-                col.resize(n);
-                for (size_t i = 0; i < n; i++) col[i] = v.av[rows[i]];
-                break;
-            }
+            case DZ_AGG_MIN: ob.agg_f64[a].assign(pmin, pmin + n); break;
+            case DZ_AGG_MAX: ob.agg_f64[a].assign(pmax, pmax + n); break;
+            case DZ_AGG_SUM: ob.agg_f64[a].assign(psum, psum + n); break;
+            case DZ_AGG_AVG: ob.agg_f64[a].assign(pavg, pavg + n); break;
         }
     }
     ob.agg_valid.resize(n);
-    for (size_t i = 0; i < n; i++) ob.agg_valid[i] = v.flags[rows[i]] & 1;
+    for (size_t i = 0; i < n; i++) ob.agg_valid[i] = pfl[i] & 1;
     ob.wstart.assign(n, wstart);
     ob.wend.assign(n, wend);
     ob.view.n_rows = (int64_t)n;
@@ -918,29 +873,15 @@ static void emit_worker_main(dz_window_op* op) {
             const uint32_t nt = op->e_pcnt[job.slab];
             if (nt > 0) {
                 dz_window_op::DevEmit& d = op->e_dev[job.slab];
-                char* s = (char*)op->e_slabs[job.slab];
-                int64_t kc = job.kcap;
+                /* the device already applied the sorted permutation and
+                 * packed every final-order column: ONE contiguous pull */
                 hipStream_t wcs = op->c_streams[job.cs];
-                hipMemcpyAsync(s, d.fkid, (size_t)nt * 4, hipMemcpyDeviceToHost,
-                               wcs);
-                hipMemcpyAsync(s + kc * 4, d.fiota, (size_t)nt * 4,
-                               hipMemcpyDeviceToHost, wcs);
-                hipMemcpyAsync(s + kc * 8, d.oflags, nt, hipMemcpyDeviceToHost,
-                               wcs);
-                hipMemcpyAsync(s + kc * 9, d.ocnt, (size_t)nt * 8,
-                               hipMemcpyDeviceToHost, wcs);
-                hipMemcpyAsync(s + kc * 17, d.omin, (size_t)nt * 8,
-                               hipMemcpyDeviceToHost, wcs);
-                hipMemcpyAsync(s + kc * 25, d.omax, (size_t)nt * 8,
-                               hipMemcpyDeviceToHost, wcs);
-                hipMemcpyAsync(s + kc * 33, d.osum, (size_t)nt * 8,
-                               hipMemcpyDeviceToHost, wcs);
-                hipMemcpyAsync(s + kc * 41, d.oavg, (size_t)nt * 8,
+                hipMemcpyAsync(op->e_slabs[job.slab], d.pout, (size_t)nt * 53,
                                hipMemcpyDeviceToHost, wcs);
                 hipEventRecord(job.ev, wcs);
                 event_spin_relaxed(job.ev);
             }
-            build_emission(op, job.wstart, job.wend, nt, job.kcap,
+            build_emission(op, job.wstart, job.wend, nt,
                            op->e_slabs[job.slab], &ob);
         } else {
             /* gbuf jobs: job.ev = gather done, and the gather wrote this
@@ -1000,8 +941,8 @@ static dz_status ensure_emission(dz_window_op* op) {
             auto& d = op->e_dev[i];
             hipFree(d.base);
             /* 3x u64 keys + 5x u64/f64 columns + 4x u32 + flags + counters
-             * = ~82 B/entry */
-            CHK(op, hipMalloc(&d.base, (size_t)kc * 84 + 64));
+             * + the 53 B/entry packed permuted-output block */
+            CHK(op, hipMalloc(&d.base, (size_t)kc * 140 + 128));
             char* p = d.base;
             d.ekeys = (uint64_t*)p; p += kc * 8;
             d.fkeys = (uint64_t*)p; p += kc * 8;
@@ -1016,7 +957,8 @@ static dz_status ensure_emission(dz_window_op* op) {
             d.fiota = (uint32_t*)p; p += kc * 4;
             d.okid = (uint32_t*)p; p += kc * 4;
             d.oflags = (uint8_t*)p; p += kc;
-            d.counter = (uint32_t*)p;
+            d.counter = (uint32_t*)p; p += 64;
+            d.pout = p;
             if (op->e_slabs[i]) hipHostFree(op->e_slabs[i]);
             CHK(op, hipHostMalloc((void**)&op->e_slabs[i],
                                   (size_t)kc * SLAB_BYTES_PER_ENTRY + 16));
@@ -1254,6 +1196,12 @@ static dz_status trigger_windows(dz_window_op* op) {
             dz::launch_emission_sort(cs, op->n_keys, d.fkeys,
                                      d.skeys, d.fiota, d.okid, d.counter + 1,
                                      op->d_rhist[csi], op->d_roffs[csi]);
+            /* pack the final-order columns on device: the worker pulls one
+             * contiguous span and builds with sequential copies */
+            dz::launch_emission_permute(cs, op->n_keys, d.counter + 1,
+                                        d.fiota, d.fkid, d.ocnt, d.omin,
+                                        d.omax, d.osum, d.oavg, d.oflags,
+                                        d.pout);
         }
         if (dev_path)
             CHK(op, hipMemcpyAsync(&op->e_pcnt[slab], d.counter + 1, 4,
