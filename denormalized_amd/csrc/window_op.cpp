@@ -308,6 +308,10 @@ struct dz_window_op {
     std::deque<OutBuf> outq;
     OutBuf current;
     bool has_current = false;
+    /* consumed OutBuf husks recycled to the build workers: vector capacity
+     * is retained, so ~1M-row closes stop paying allocation + page faults
+     * on every build (the reason 8 workers ran SLOWER than 4) */
+    std::vector<OutBuf> ob_pool;  /* guarded by out_mtx */
 
     /* timing */
     std::vector<hipEvent_t> ev_pool;
@@ -687,7 +691,7 @@ static void build_emission_host(dz_window_op* op, int64_t wstart, int64_t wend,
     }
     size_t n = rows.size(), na = op->aggs.size();
 
-    OutBuf ob;
+    OutBuf& ob = *out;
     ob.agg_i64.resize(na);
     ob.agg_f64.resize(na);
     if (op->key_kind == DZ_KEY_UTF8) {
@@ -755,7 +759,6 @@ static void build_emission_host(dz_window_op* op, int64_t wstart, int64_t wend,
     ob.wstart.assign(n, wstart);
     ob.wend.assign(n, wend);
     ob.view.n_rows = (int64_t)n;
-    *out = std::move(ob);
 }
 
 /* Pinned emission slab layout for kcap entries (49 bytes per entry):
@@ -784,7 +787,7 @@ static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
     const uint8_t* pfl = (const uint8_t*)(p + (size_t)nt * 52);
     size_t n = nt;
     size_t na = op->aggs.size();
-    OutBuf ob;
+    OutBuf& ob = *out;
     ob.agg_i64.resize(na);
     ob.agg_f64.resize(na);
     if (op->no_group) {
@@ -826,7 +829,22 @@ static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
     ob.wstart.assign(n, wstart);
     ob.wend.assign(n, wend);
     ob.view.n_rows = (int64_t)n;
-    *out = std::move(ob);
+}
+
+static OutBuf take_outbuf(dz_window_op* op) {
+    std::lock_guard<std::mutex> lk(op->out_mtx);
+    if (op->ob_pool.empty()) return OutBuf();
+    OutBuf ob = std::move(op->ob_pool.back());
+    op->ob_pool.pop_back();
+    ob.key_i64.clear();
+    ob.key_offsets.clear();
+    ob.key_data.clear();
+    for (auto& c : ob.agg_i64) c.clear();
+    for (auto& c : ob.agg_f64) c.clear();
+    ob.agg_valid.clear();
+    ob.wstart.clear();
+    ob.wend.clear();
+    return ob;
 }
 
 static void emit_drain(dz_window_op* op) {
@@ -866,7 +884,7 @@ static void emit_worker_main(dz_window_op* op) {
         }
         event_spin_relaxed(job.ev); /* emission chain complete */
         auto t0 = std::chrono::steady_clock::now();
-        OutBuf ob;
+        OutBuf ob = take_outbuf(op);
         if (job.device) {
             /* copy exactly nt2 packed rows (the filter already ran on
              * device, so this is the final output volume, not the keyspace) */
@@ -1853,6 +1871,8 @@ extern "C" dz_status dz_window_op_poll(dz_window_op* op, const dz_out_batch** ou
     {
         std::lock_guard<std::mutex> lk(op->out_mtx);
         if (op->outq.empty()) return DZ_OK;
+        if (op->has_current && op->ob_pool.size() < 16)
+            op->ob_pool.push_back(std::move(op->current));
         op->current = std::move(op->outq.front());
         op->outq.pop_front();
     }
