@@ -1261,7 +1261,8 @@ __global__ void k_epermute(const uint32_t* __restrict__ counter2,
         psum[i] = osum[j];
         pavg[i] = oavg[j];
         pkid[i] = k;
-        pfl[i] = oflags[j];
+        pfl[i] = oflags[j] & 1; /* packed validity is already 0/1 so the
+                                 * host build can take it verbatim */
     }
 }
 

@@ -824,8 +824,7 @@ static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
             case DZ_AGG_AVG: ob.agg_f64[a].assign(pavg, pavg + n); break;
         }
     }
-    ob.agg_valid.resize(n);
-    for (size_t i = 0; i < n; i++) ob.agg_valid[i] = pfl[i] & 1;
+    ob.agg_valid.assign(pfl, pfl + n);
     ob.wstart.assign(n, wstart);
     ob.wend.assign(n, wend);
     ob.view.n_rows = (int64_t)n;
