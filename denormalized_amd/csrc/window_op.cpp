@@ -887,16 +887,25 @@ static void emit_worker_main(dz_window_op* op) {
         if (job.device) {
             /* copy exactly nt2 packed rows (the filter already ran on
              * device, so this is the final output volume, not the keyspace) */
-            const uint32_t nt = op->e_pcnt[job.slab];
+            uint32_t nt = op->e_pcnt[job.slab];
             if (nt > 0) {
                 dz_window_op::DevEmit& d = op->e_dev[job.slab];
                 /* the device already applied the sorted permutation and
                  * packed every final-order column: ONE contiguous pull */
                 hipStream_t wcs = op->c_streams[job.cs];
-                hipMemcpyAsync(op->e_slabs[job.slab], d.pout, (size_t)nt * 53,
-                               hipMemcpyDeviceToHost, wcs);
-                hipEventRecord(job.ev, wcs);
-                event_spin_relaxed(job.ev);
+                hipError_t e = hipMemcpyAsync(op->e_slabs[job.slab], d.pout,
+                                              (size_t)nt * 53,
+                                              hipMemcpyDeviceToHost, wcs);
+                if (e == hipSuccess) e = hipEventRecord(job.ev, wcs);
+                if (e != hipSuccess) {
+                    /* surface on the op; emit an empty batch rather than
+                     * formatting garbage */
+                    op->err = std::string("emission copy failed: ") +
+                              hipGetErrorString(e);
+                    nt = 0;
+                } else {
+                    event_spin_relaxed(job.ev);
+                }
             }
             build_emission(op, job.wstart, job.wend, nt,
                            op->e_slabs[job.slab], &ob);
