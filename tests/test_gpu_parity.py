@@ -908,3 +908,32 @@ def test_soak_mixed_push_kinds():
         for a in bs:
             a.free()
     o.close()
+
+
+@pytest.mark.gpu
+def test_negative_timestamps_rejected():
+    # the reference's SystemTime arithmetic panics on pre-epoch timestamps;
+    # we surface a loud error instead. Host pushes are synchronous (error at
+    # the push); device pushes are pipelined (error by the next call into
+    # the op — finish at the latest, per the header contract).
+    op = make_op(1000)
+    with pytest.raises(RuntimeError, match="negative"):
+        op.push(np.array([-5], np.int64), np.array([1], np.int64),
+                np.array([1.0]))
+    op.close()
+
+    from denormalized_amd import DeviceArray, WindowOp, _lib
+    op = WindowOp(length_ms=1000, key_kind=_lib.KEY_DENSE_INT64,
+                  n_keys_hint=4)
+    d_ts = DeviceArray(0, 8)
+    d_k = DeviceArray(0, 4)
+    d_v = DeviceArray(0, 8)
+    d_ts.from_host(np.array([-7], np.int64))
+    d_k.from_host(np.array([0], np.int32))
+    d_v.from_host(np.array([2.0]))
+    op.push_device(1, d_ts.ptr, d_k.ptr, d_v.ptr, borrowed=True)
+    with pytest.raises(RuntimeError, match="negative"):
+        op.finish()
+    op.close()
+    for a in (d_ts, d_k, d_v):
+        a.free()
