@@ -81,6 +81,14 @@ struct KStatAcc {
 };
 
 struct OutBuf {
+    /* zero-copy device-path emission: the batch's aggregate columns (and
+     * the dense key column) are served as views into the pinned packed
+     * span; the OutBuf holds the slab until the consumer moves to the next
+     * batch (poll pop releases it). hold_slab == -1 => fully materialized. */
+    const char* packed = nullptr;
+    uint32_t packed_nt = 0;
+    int hold_slab = -1;
+    uint32_t slab_gen = 0;  /* pool generation the held slab belongs to */
     std::vector<int64_t> key_i64;
     std::vector<int32_t> key_offsets;
     std::vector<char> key_data;
@@ -275,6 +283,11 @@ struct dz_window_op {
                                           * a kernel store does not */
     std::vector<int> e_gfree;            /* guarded by e_mtx */
     int64_t e_gbuf_kcap = 0;
+    /* zero-copy OutBufs hold pinned slabs across a pool regrow (keyspace
+     * growth): old allocations go to a graveyard until destroy, and a
+     * generation counter stops a stale hold from re-entering the new pool */
+    uint32_t e_slab_gen = 0;             /* guarded by e_mtx */
+    std::vector<uint64_t*> e_graveyard;
     /* slot-release frontier events for group-batched host emission: each
      * group records the next event of this rotation after its gather; the
      * freed slots' reset waits reference it WITHOUT taking ownership (a
@@ -557,6 +570,8 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     for (auto e : op->e_ev_pool) hipEventDestroy(e);
     for (auto& s : op->e_slabs)
         if (s) hipHostFree(s);
+    for (auto* g : op->e_graveyard)
+        hipHostFree(g);
     for (auto& g : op->e_gbufs)
         if (g) hipHostFree(g);
     for (auto& d : op->e_dev)
@@ -773,9 +788,13 @@ static constexpr int64_t SLAB_BYTES_PER_ENTRY = 56;
  * emission worker pool: touches only immutable config, the deque-backed
  * dictionaries (indices < the job's n_keys snapshot) and the pinned slab. */
 static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
-                           uint32_t nt, const uint64_t* slab, OutBuf* out) {
+                           uint32_t nt, const uint64_t* slab, OutBuf* out,
+                           int slab_idx = -1, bool zero_copy = false) {
     /* packed final-order columns from k_epermute (53 B/row): sequential
-     * reads only — the per-row permutation gather already ran on device */
+     * reads only — the per-row permutation gather already ran on device.
+     * zero_copy: serve the aggregate columns (and dense keys + validity)
+     * as VIEWS into the pinned span; the OutBuf keeps the slab until the
+     * consumer's next poll. */
     const char* p = (const char*)slab;
     const int64_t* pkey = (const int64_t*)p;
     const uint64_t* pcnt = (const uint64_t*)(p + (size_t)nt * 8);
@@ -790,6 +809,11 @@ static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
     OutBuf& ob = *out;
     ob.agg_i64.resize(na);
     ob.agg_f64.resize(na);
+    if (zero_copy && n > 0) {
+        ob.packed = p;
+        ob.packed_nt = nt;
+        ob.hold_slab = slab_idx;
+    }
     if (op->no_group) {
         /* global aggregate: output schema has no group column
          * (create_schema with empty group exprs, streaming_window.rs:1096+) */
@@ -809,9 +833,10 @@ static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
     } else if (op->key_kind == DZ_KEY_INT64) {
         ob.key_i64.resize(n);
         for (size_t i = 0; i < n; i++) ob.key_i64[i] = op->dict_vals[pkid[i]];
-    } else {
+    } else if (!ob.packed) {
         ob.key_i64.assign(pkey, pkey + n);
     }
+    if (!ob.packed)
     for (size_t a = 0; a < na; a++) {
         switch (op->aggs[a].op) {
             case DZ_AGG_COUNT:
@@ -824,7 +849,7 @@ static void build_emission(dz_window_op* op, int64_t wstart, int64_t wend,
             case DZ_AGG_AVG: ob.agg_f64[a].assign(pavg, pavg + n); break;
         }
     }
-    ob.agg_valid.assign(pfl, pfl + n);
+    if (!ob.packed) ob.agg_valid.assign(pfl, pfl + n);
     ob.wstart.assign(n, wstart);
     ob.wend.assign(n, wend);
     ob.view.n_rows = (int64_t)n;
@@ -835,6 +860,9 @@ static OutBuf take_outbuf(dz_window_op* op) {
     if (op->ob_pool.empty()) return OutBuf();
     OutBuf ob = std::move(op->ob_pool.back());
     op->ob_pool.pop_back();
+    ob.packed = nullptr;
+    ob.packed_nt = 0;
+    ob.hold_slab = -1;
     ob.key_i64.clear();
     ob.key_offsets.clear();
     ob.key_data.clear();
@@ -907,8 +935,20 @@ static void emit_worker_main(dz_window_op* op) {
                     event_spin_relaxed(job.ev);
                 }
             }
+            bool zc = false;
+            uint32_t gen = 0;
+            if (nt >= 65536) {
+                /* big outputs skip the materializing copies; LIVENESS: only
+                 * while the slab pool stays deep — if consumers lag, fall
+                 * back to copy+release so trigger_windows can always make
+                 * progress */
+                std::lock_guard<std::mutex> lk(op->e_mtx);
+                zc = op->e_free.size() >= 8;
+                gen = op->e_slab_gen;
+            }
             build_emission(op, job.wstart, job.wend, nt,
-                           op->e_slabs[job.slab], &ob);
+                           op->e_slabs[job.slab], &ob, job.slab, zc);
+            ob.slab_gen = gen;
         } else {
             /* gbuf jobs: job.ev = gather done, and the gather wrote this
              * close's slab directly into the pinned group buffer */
@@ -921,6 +961,7 @@ static void emit_worker_main(dz_window_op* op) {
         op->e_build_ns += (uint64_t)std::chrono::duration_cast<std::chrono::nanoseconds>(
             std::chrono::steady_clock::now() - t0).count();
         op->e_builds++;
+        const int ob_holds_slab = ob.hold_slab;
         {
             std::lock_guard<std::mutex> lk(op->out_mtx);
             op->e_done.emplace(job.ticket, std::move(ob));
@@ -933,7 +974,8 @@ static void emit_worker_main(dz_window_op* op) {
         }
         {
             std::lock_guard<std::mutex> lk(op->e_mtx);
-            if (job.slab >= 0) op->e_free.push_back(job.slab);
+            if (job.slab >= 0 && ob_holds_slab != job.slab)
+                op->e_free.push_back(job.slab);
             /* group-shared resources go back to their pools only once every
              * sibling is past its spin (i.e. fully built) */
             if (!job.grp_left || job.grp_left->fetch_sub(1) == 1) {
@@ -985,7 +1027,7 @@ static dz_status ensure_emission(dz_window_op* op) {
             d.oflags = (uint8_t*)p; p += kc;
             d.counter = (uint32_t*)p; p += 64;
             d.pout = p;
-            if (op->e_slabs[i]) hipHostFree(op->e_slabs[i]);
+            if (op->e_slabs[i]) op->e_graveyard.push_back(op->e_slabs[i]);
             CHK(op, hipHostMalloc((void**)&op->e_slabs[i],
                                   (size_t)kc * SLAB_BYTES_PER_ENTRY + 16));
         }
@@ -1007,6 +1049,7 @@ static dz_status ensure_emission(dz_window_op* op) {
             }
         }
         std::lock_guard<std::mutex> lk(op->e_mtx);
+        op->e_slab_gen++;
         op->e_free.clear();
         for (int i = 0; i < dz_window_op::E_POOL; i++) op->e_free.push_back(i);
         op->e_gfree.clear();
@@ -1876,28 +1919,75 @@ extern "C" dz_status dz_window_op_poll(dz_window_op* op, const dz_out_batch** ou
                 return DZ_ERR;
         }
     }
+    int release = -1;
+    uint32_t release_gen = 0;
     {
         std::lock_guard<std::mutex> lk(op->out_mtx);
         if (op->outq.empty()) return DZ_OK;
-        if (op->has_current && op->ob_pool.size() < 16)
-            op->ob_pool.push_back(std::move(op->current));
+        if (op->has_current) {
+            /* the previous batch's validity ends here: release its slab
+             * (zero-copy batches) and recycle the buffer husk */
+            release = op->current.hold_slab;
+            release_gen = op->current.slab_gen;
+            op->current.hold_slab = -1;
+            op->current.packed = nullptr;
+            if (op->ob_pool.size() < 16)
+                op->ob_pool.push_back(std::move(op->current));
+        }
         op->current = std::move(op->outq.front());
         op->outq.pop_front();
+    }
+    if (release >= 0) {
+        {
+            std::lock_guard<std::mutex> lk(op->e_mtx);
+            /* a hold that survived a pool regrow references graveyarded
+             * memory: its slot number must not re-enter the new pool */
+            if (release_gen == op->e_slab_gen)
+                op->e_free.push_back(release);
+        }
+        op->e_cv.notify_all();
     }
     op->has_current = true;
     OutBuf& ob = op->current;
     ob.agg_ptrs.clear();
-    for (size_t a = 0; a < op->aggs.size(); a++) {
-        if (op->aggs[a].op == DZ_AGG_COUNT)
-            ob.agg_ptrs.push_back((const void*)ob.agg_i64[a].data());
-        else
-            ob.agg_ptrs.push_back((const void*)ob.agg_f64[a].data());
+    if (ob.packed) {
+        /* zero-copy batch: aggregate columns (+ dense keys + validity) are
+         * views into the pinned packed span the OutBuf holds */
+        const char* p = ob.packed;
+        const size_t ntp = ob.packed_nt;
+        for (size_t a = 0; a < op->aggs.size(); a++) {
+            const void* ptr = nullptr;
+            switch (op->aggs[a].op) {
+                case DZ_AGG_COUNT: ptr = p + ntp * 8; break;
+                case DZ_AGG_MIN: ptr = p + ntp * 16; break;
+                case DZ_AGG_MAX: ptr = p + ntp * 24; break;
+                case DZ_AGG_SUM: ptr = p + ntp * 32; break;
+                case DZ_AGG_AVG: ptr = p + ntp * 40; break;
+            }
+            ob.agg_ptrs.push_back(ptr);
+        }
+        bool dense_key = !op->no_group && op->key_kind == DZ_KEY_DENSE_INT64;
+        ob.view.key_i64 = dense_key ? (const int64_t*)p
+                                    : (ob.key_i64.empty() ? nullptr
+                                                          : ob.key_i64.data());
+        ob.view.key_offsets =
+            ob.key_offsets.empty() ? nullptr : ob.key_offsets.data();
+        ob.view.key_data = ob.key_data.empty() ? nullptr : ob.key_data.data();
+        ob.view.agg_valid = (const uint8_t*)(p + ntp * 52);
+    } else {
+        for (size_t a = 0; a < op->aggs.size(); a++) {
+            if (op->aggs[a].op == DZ_AGG_COUNT)
+                ob.agg_ptrs.push_back((const void*)ob.agg_i64[a].data());
+            else
+                ob.agg_ptrs.push_back((const void*)ob.agg_f64[a].data());
+        }
+        ob.view.key_i64 = ob.key_i64.empty() ? nullptr : ob.key_i64.data();
+        ob.view.key_offsets =
+            ob.key_offsets.empty() ? nullptr : ob.key_offsets.data();
+        ob.view.key_data = ob.key_data.empty() ? nullptr : ob.key_data.data();
+        ob.view.agg_valid = ob.agg_valid.data();
     }
-    ob.view.key_i64 = ob.key_i64.empty() ? nullptr : ob.key_i64.data();
-    ob.view.key_offsets = ob.key_offsets.empty() ? nullptr : ob.key_offsets.data();
-    ob.view.key_data = ob.key_data.empty() ? nullptr : ob.key_data.data();
     ob.view.agg_cols = ob.agg_ptrs.data();
-    ob.view.agg_valid = ob.agg_valid.data();
     ob.view.window_start_ms = ob.wstart.data();
     ob.view.window_end_ms = ob.wend.data();
     *out = &ob.view;
