@@ -310,6 +310,8 @@ struct dz_window_op {
     std::map<uint64_t, OutBuf> e_done; /* completed out-of-order (out_mtx) */
     std::atomic<uint64_t> e_build_ns{0};
     std::atomic<uint64_t> e_builds{0};
+    std::atomic<uint64_t> e_zc_builds{0};   /* zero-copy vs materializing */
+    std::atomic<uint64_t> e_copy_builds{0}; /* device-path builds */
     std::mutex out_mtx;             /* guards outq */
 
     /* filter pushdown */
@@ -937,7 +939,7 @@ static void emit_worker_main(dz_window_op* op) {
             }
             bool zc = false;
             uint32_t gen = 0;
-            if (nt >= 65536) {
+            if (nt >= 16384) {
                 /* big outputs skip the materializing copies; LIVENESS: only
                  * while the slab pool stays deep — if consumers lag, fall
                  * back to copy+release so trigger_windows can always make
@@ -949,6 +951,7 @@ static void emit_worker_main(dz_window_op* op) {
             build_emission(op, job.wstart, job.wend, nt,
                            op->e_slabs[job.slab], &ob, job.slab, zc);
             ob.slab_gen = gen;
+            (zc ? op->e_zc_builds : op->e_copy_builds)++;
         } else {
             /* gbuf jobs: job.ev = gather done, and the gather wrote this
              * close's slab directly into the pinned group buffer */
@@ -2078,6 +2081,8 @@ extern "C" dz_status dz_window_op_kernel_stats(dz_window_op* op,
         KStatAcc& s = op->stats["h_emit_build"];
         s.launches = op->e_builds.load();
         s.ms = op->e_build_ns.load() / 1e6;
+        op->stats["h_emit_zc"].launches = op->e_zc_builds.load();
+        op->stats["h_emit_copybuild"].launches = op->e_copy_builds.load();
     }
     int32_t n = 0;
     for (auto& kv : op->stats) {
