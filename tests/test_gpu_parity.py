@@ -937,3 +937,40 @@ def test_negative_timestamps_rejected():
     op.close()
     for a in (d_ts, d_k, d_v):
         a.free()
+
+
+@pytest.mark.gpu
+def test_randomized_config_matrix():
+    # 20 seeded random configurations (window length/hop, key kind, keyspace,
+    # batch count/size, null density) — every one must be bit-exact. This is
+    # the engine-vs-oracle analog of the CPU-side oracle-vs-restatement fuzz.
+    rng = np.random.default_rng(2026)
+    from denormalized_amd import _lib
+    for case in range(20):
+        len_ms = int(rng.choice([250, 500, 1000, 1500, 3000]))
+        slide_ms = int(rng.choice([0, 0, 100, 250, len_ms]))
+        if slide_ms > len_ms * 3:
+            slide_ms = 0
+        nkeys = int(rng.choice([1, 7, 100, 3000, 70_000]))
+        nb = int(rng.integers(1, 5))
+        rows = int(rng.integers(200, 40_000))
+        rate = max(1, rows // max(1, int(rng.integers(50, 4000))))
+        with_nulls = bool(rng.random() < 0.3)
+        t = 1_000_000
+        batches, valids = [], []
+        for _ in range(nb):
+            ts = (t + np.cumsum(rng.integers(0, max(1, 1000 // rate), rows))
+                  ).astype(np.int64)
+            t = int(ts.max())
+            k = rng.integers(0, nkeys, rows)
+            v = rng.uniform(-10, 125, rows)
+            batches.append((ts, k, v))
+            valids.append((rng.random(rows) > 0.2) if with_nulls else None)
+        outs, exp = run_both(len_ms, slide_ms, batches,
+                             n_keys_hint=min(nkeys, 256), valids=valids)
+        try:
+            assert_parity(outs, exp)
+        except AssertionError as e:
+            raise AssertionError(
+                f"case {case}: len={len_ms} slide={slide_ms} keys={nkeys} "
+                f"rows={rows}x{nb} nulls={with_nulls}: {e}") from e
