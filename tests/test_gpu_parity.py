@@ -958,9 +958,14 @@ def test_randomized_config_matrix():
         with_nulls = bool(rng.random() < 0.3)
         t = 1_000_000
         batches, valids = [], []
+        hop = slide_ms if slide_ms else len_ms
         for _ in range(nb):
             ts = (t + np.cumsum(rng.integers(0, max(1, 1000 // rate), rows))
                   ).astype(np.int64)
+            span = int(ts.max()) - t
+            cap = hop * 2000  # stay inside the 4096-windows-per-batch envelope
+            if span > cap:
+                ts = (t + (ts - t) * cap // span).astype(np.int64)
             t = int(ts.max())
             k = rng.integers(0, nkeys, rows)
             v = rng.uniform(-10, 125, rows)
@@ -974,3 +979,15 @@ def test_randomized_config_matrix():
             raise AssertionError(
                 f"case {case}: len={len_ms} slide={slide_ms} keys={nkeys} "
                 f"rows={rows}x{nb} nulls={with_nulls}: {e}") from e
+
+
+@pytest.mark.gpu
+def test_batch_window_span_cap_is_loud():
+    # one batch may span at most MAX_RANGES (4096) windows — a documented
+    # envelope (streaming batches never legitimately span hours); exceeding
+    # it must error loudly, not truncate
+    op = make_op(1000)
+    ts = np.array([0, 5_000_000_000], np.int64)  # ~5M windows apart
+    with pytest.raises(RuntimeError, match="windows"):
+        op.push(ts, np.array([1, 1], np.int64), np.array([1.0, 2.0]))
+    op.close()
