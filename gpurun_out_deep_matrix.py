@@ -26,8 +26,6 @@ for case in range(100):
         v = rng.uniform(-50, 200, rows)
         batches.append((ts, k, v))
         valids.append((rng.random(rows) > 0.15) if with_nulls else None)
-    print(f"case {case}: len={len_ms} slide={slide_ms} keys={nkeys} "
-          f"rows={rows}x{nb} nulls={with_nulls}", flush=True)
     try:
         outs, exp = run_both(len_ms, slide_ms, batches,
                              n_keys_hint=min(nkeys, 128), valids=valids)
