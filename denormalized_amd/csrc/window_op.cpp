@@ -1585,6 +1585,13 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
     /* 3. partition + fold */
     int64_t expand = sliding
         ? (op->len_ms + op->slide_ms - 1) / op->slide_ms + 1 : 1;
+    if (expand > dz::ST_RECORDS) {
+        /* one row would overflow the scatter's LDS staging on its own */
+        op->err = "window length / hop ratio exceeds " +
+                  std::to_string(dz::ST_RECORDS) + " (one row would land in >" +
+                  std::to_string(dz::ST_RECORDS) + " windows)";
+        return DZ_ERR;
+    }
     int64_t nrec_max = n * expand;
     if (nrec_max >= (int64_t)UINT32_MAX) {
         op->err = "batch expands past 2^32 records; push smaller batches";
@@ -1601,7 +1608,10 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
                             op->d_base[b], op->d_gofs[b]);
         });
     }
-    int32_t st_rows = (int32_t)std::max<int64_t>(64, dz::ST_RECORDS / expand);
+    /* supertile row budget: st_rows * expand staged records must fit the
+     * ST_RECORDS LDS staging (a 64-row floor overflowed it for window/hop
+     * ratios above 32 — found by the randomized deep matrix) */
+    int32_t st_rows = (int32_t)std::max<int64_t>(1, dz::ST_RECORDS / expand);
     timed(op, "scatter", (double)n * 24 + (double)nrec_max * 20, [&] {
         dz::launch_scatter(op->stream, d_kid, d_ts, d_vals, d_valbm, n, chunk, C,
                            st_rows, wp, op->d_gofs[b], op->d_meta, op->d_grec);

@@ -991,3 +991,31 @@ def test_batch_window_span_cap_is_loud():
     with pytest.raises(RuntimeError, match="windows"):
         op.push(ts, np.array([1, 1], np.int64), np.array([1.0, 2.0]))
     op.close()
+
+
+@pytest.mark.gpu
+def test_extreme_window_hop_ratio():
+    # len/hop = 50 => every row lands in ~50 windows (51 staged records per
+    # row): the scatter's supertile row budget must shrink so staging never
+    # overflows LDS (the 64-row floor crashed with a GPU memory fault —
+    # found by scripts/deep_matrix.py case 22)
+    rng = np.random.default_rng(22)
+    n = 39_066
+    t = 1_000_000
+    ts = (t + np.cumsum(rng.integers(0, 12, n))).astype(np.int64)
+    k = rng.integers(0, 7, n)
+    v = rng.uniform(-50, 200, n)
+    outs, exp = run_both(5000, 100, [(ts, k, v)], n_keys_hint=7)
+    assert len(exp["key"]) > 0
+    assert_parity(outs, exp)
+
+
+@pytest.mark.gpu
+def test_window_hop_ratio_envelope_is_loud():
+    # ratios beyond ST_RECORDS (2048) would overflow staging from a single
+    # row: rejected loudly (a 2048x-overlapping window is not a real config)
+    op = make_op(5_000_000, slide_ms=1)
+    with pytest.raises(RuntimeError, match="ratio exceeds"):
+        op.push(np.array([1_000_000], np.int64), np.array([1], np.int64),
+                np.array([1.0]))
+    op.close()
