@@ -1014,7 +1014,7 @@ def test_extreme_window_hop_ratio():
 def test_window_hop_ratio_envelope_is_loud():
     # ratios beyond ST_RECORDS (2048) would overflow staging from a single
     # row: rejected loudly (a 2048x-overlapping window is not a real config)
-    op = make_op(5_000_000, slide_ms=1)
+    op = make_op(3_000_000, slide_ms=1000)  # expand ~3001 (window list ~3000, under MAX_RANGES)
     with pytest.raises(RuntimeError, match="ratio exceeds"):
         op.push(np.array([1_000_000], np.int64), np.array([1], np.int64),
                 np.array([1.0]))
