@@ -3,7 +3,8 @@ sys.path.insert(0, "/root/repo")
 import numpy as np
 from tests.test_gpu_parity import run_both, assert_parity
 
-rng = np.random.default_rng(777)
+import os
+rng = np.random.default_rng(int(os.environ.get("DM_SEED", "777")))
 fails = 0
 for case in range(100):
     len_ms = int(rng.choice([250, 333, 500, 1000, 1500, 2500, 5000]))
