@@ -7,13 +7,14 @@ metric is quoted on: 10k keys, 1s tumbling window, count/min/max/avg(reading)
 + filter(max > 113), f64 readings, synthetic sensor stream (seeded; spec in
 DESIGN.md §Generator). A step = one push of --rows-per-step rows through the
 operator (device-resident inputs) including triggered window emission; the
-default run covers ~100M rows ≈ cfg2.
+default run covers 768M rows (≥ cfg2's 100M; longer runs amortize warmup so
+the steady state dominates).
 
 Contract: `python bench.py --gpus N --steps K --warmup W`. For N>1 the driver
 launches one rank per GPU via torch.distributed.run (RCCL); rows shard by key
-across ranks (the reference's RepartitionExec::Hash analog, SURVEY §8e), the
-only collectives are the shared-watermark all-reduce (MAX) and final count
-gather. scaling=weak: per-rank rows fixed as N grows.
+across ranks (the reference's RepartitionExec::Hash analog, SURVEY §8e); the
+only collective is the shared-watermark all-reduce (MAX), exchanged before
+each push. scaling=weak: per-rank rows fixed as N grows.
 
 Rank 0 prints ONE JSON line. cpu_baseline: the CPU oracle (kind "port") timed
 on this box's host cores on a bounded sample of the same workload.
