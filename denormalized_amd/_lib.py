@@ -303,9 +303,12 @@ class WindowOp:
         return self._L.dz_window_op_open_windows(self._h)
 
     def kernel_stats(self):
-        arr = (DzKernelStat * 16)()
+        # 32 slots: the stat map currently holds up to 16 entries (6 device
+        # kernels + 10 host timers) — headroom so a new timer can never
+        # silently push the roofline's dominant kernel out of the window
+        arr = (DzKernelStat * 32)()
         nout = ctypes.c_int32()
-        self._check(self._L.dz_window_op_kernel_stats(self._h, arr, 16,
+        self._check(self._L.dz_window_op_kernel_stats(self._h, arr, 32,
                                                       ctypes.byref(nout)), "stats")
         return {arr[i].name.decode(): {
             "launches": arr[i].launches, "total_ms": arr[i].total_ms,
