@@ -11,10 +11,14 @@ default run covers 768M rows (≥ cfg2's 100M; longer runs amortize warmup so
 the steady state dominates).
 
 Contract: `python bench.py --gpus N --steps K --warmup W`. For N>1 the driver
-launches one rank per GPU via torch.distributed.run (RCCL); rows shard by key
-across ranks (the reference's RepartitionExec::Hash analog, SURVEY §8e); the
-only collective is the shared-watermark all-reduce (MAX), exchanged before
-each push. scaling=weak: per-rank rows fixed as N grows.
+launches one rank per GPU via torch.distributed.run (RCCL). N-GPU mode
+measures WEAK scaling over N independent key partitions: each rank generates
+its own rank-distinct key shard (rank-local dense ids over a rank-seeded
+stream), the in-engine analog of the rows a RepartitionExec::Hash(group_by)
+exchange would deliver to that rank — the hash exchange itself
+is not in the timed path (each GPU owns its keys for the whole run). The only
+collective is the shared-watermark all-reduce (MAX), exchanged before each
+push. scaling=weak: per-rank rows fixed as N grows.
 
 Rank 0 prints ONE JSON line. cpu_baseline: the CPU oracle (kind "port") timed
 on this box's host cores on a bounded sample of the same workload.
@@ -192,9 +196,10 @@ def main():
         emitted = 0
         # non-blocking zero-copy poll: emission is pipelined on the op's
         # worker pool and overlaps the next step's kernels; consuming the
-        # batch = reading the op-owned Arrow-style buffers (the C contract),
-        # exactly as the reference's downstream operator would.
-        for b in op.poll_all(drain=False, copy=False):
+        # batch = reading the op-owned Arrow-style buffers (the C contract:
+        # each batch is read before the next poll invalidates it), exactly
+        # as the reference's downstream operator would.
+        for b in op.poll_iter(copy=False):
             emitted += b["n_rows"]
         return emitted
 
@@ -220,7 +225,7 @@ def main():
                            ctypes.c_void_p(d_vals.ptr.value + off * 8),
                            borrowed=not args.staged_push)
             m = time.perf_counter()
-            for bt in op.poll_all(drain=False, copy=False):
+            for bt in op.poll_iter(copy=False):
                 emitted += bt["n_rows"]
             marks.append((m - a, time.perf_counter() - m))
         tf = time.perf_counter()
@@ -233,7 +238,8 @@ def main():
         for i, (pu, po) in enumerate(marks):
             print(f"step {i} push {pu*1e3:7.3f} ms poll {po*1e3:7.3f} ms",
                   file=sys.stderr)
-    emitted += sum(b["n_rows"] for b in op.poll_all(copy=False))
+    op.drain()
+    emitted += sum(b["n_rows"] for b in op.poll_iter(copy=False))
     dz.synchronize(device)
     if dist is not None:
         dist.barrier()
@@ -254,18 +260,49 @@ def main():
     if rank == 0:
         rows_timed = K * B * world  # whole-job rows through the timed region
         value = rows_timed / elapsed
-        # roofline for the dominant DEVICE kernel (host-phase timers excluded)
+        # roofline for the dominant DEVICE kernel (host-phase timers excluded):
+        # achieved = the kernel's PMC-calibrated HBM bytes per launch (from
+        # the committed rocprofv3 --pmc run, profiles/hbm_traffic.json) over
+        # its HIP-event average launch time. Falls back to the algorithmic
+        # estimate only when no PMC calibration exists for the kernel.
         dev = {k: v for k, v in stats.items() if not k.startswith("h_")}
         dom = max(dev, key=lambda k: dev[k]["total_ms"])
         d = stats[dom]
         launches_timed = d["launches"]
         avg_s = (d["total_ms"] / 1000.0) / max(1, d["launches"])
         rows_per_launch = (K + W) * B / max(1, d["launches"])
-        achieved = ALG_BYTES_PER_ROW * rows_per_launch / avg_s
         traffic = None
         tmap = read_traffic()
+        per_k = tmap.get("per_kernel_bytes_per_launch", {}) if tmap else {}
         if tmap:
-            traffic = tmap.get("per_kernel_bytes_per_launch", {}).get(dom)
+            traffic = per_k.get(dom)
+        if traffic:
+            achieved = traffic / avg_s
+            achieved_basis = "pmc"
+        else:
+            achieved = ALG_BYTES_PER_ROW * rows_per_launch / avg_s
+            achieved_basis = "algorithmic"
+        # whole-path view: PMC bytes moved per step (sum over the pipeline's
+        # kernels, launches from this run) and algorithmic bytes per step,
+        # each over the measured wall step time — the gap between the two
+        # fracs IS the pipeline's traffic amplification.
+        step_s = elapsed / K
+        pmc_bytes_per_step = None
+        if per_k:
+            tot = 0.0
+            for kname, kstat in dev.items():
+                bpl = per_k.get(kname)
+                if bpl and kname != "gen":  # gen runs pre-timed-region
+                    tot += bpl * kstat["launches"] / (K + W)
+            pmc_bytes_per_step = tot if tot > 0 else None
+        path = {
+            "alg_Bps": ALG_BYTES_PER_ROW * B / step_s,
+            "alg_frac": ALG_BYTES_PER_ROW * B / step_s / HBM_PEAK,
+        }
+        if pmc_bytes_per_step:
+            path["pmc_Bps"] = pmc_bytes_per_step / step_s
+            path["pmc_frac"] = pmc_bytes_per_step / step_s / HBM_PEAK
+            path["amplification"] = pmc_bytes_per_step / (ALG_BYTES_PER_ROW * B)
         out = {
             "metric": "rows/sec through window()+filter() on synthetic sensor stream",
             "value": value,
@@ -280,9 +317,11 @@ def main():
             "dtype": "f64",
             "data": "synthetic",
             "config": {
-                "workload": ("cfg2: " if (args.keys == 10_000 and not args.slide_ms
-                                           and args.window_ms == 1000) else "")
-                            + f"{K * B / 1e6:.0f}M rows/GPU, {args.keys} keys, "
+                "workload": ("cfg2 (dense-int key variant): "
+                             if (args.keys == 10_000 and not args.slide_ms
+                                 and args.window_ms == 1000) else "")
+                            + f"{K * B / 1e6:.0f}M rows/GPU, "
+                            + f"{args.keys} dense-int keys, "
                             + f"{args.window_ms}ms "
                             + (f"sliding/{args.slide_ms}ms hop" if args.slide_ms
                                else "tumbling")
@@ -298,11 +337,13 @@ def main():
                 "bound": "hbm",
                 "kernel": dom,
                 "achieved": achieved,
+                "achieved_basis": achieved_basis,
                 "peak": HBM_PEAK,
                 "unit": "B/s",
                 "frac": achieved / HBM_PEAK,
                 "traffic": traffic,
                 "launches": launches_timed,
+                "path": path,
                 "kernel_ms_total": {k: round(v["total_ms"], 3) for k, v in stats.items()},
             },
             "cpu_baseline": None,

@@ -62,7 +62,10 @@ class DzOutBatch(ctypes.Structure):
         ("n_rows", ctypes.c_int64),
         ("key_i64", ctypes.POINTER(ctypes.c_int64)),
         ("key_offsets", ctypes.POINTER(ctypes.c_int32)),
-        ("key_data", ctypes.c_char_p),
+        # c_void_p, NOT c_char_p: key bytes are arbitrary (may contain NUL);
+        # a c_char_p attribute read returns a NUL-truncated copy, and
+        # string_at on that copy could read out of bounds.
+        ("key_data", ctypes.c_void_p),
         ("agg_cols", ctypes.POINTER(ctypes.c_void_p)),
         ("agg_valid", ctypes.POINTER(ctypes.c_uint8)),
         ("window_start_ms", ctypes.POINTER(ctypes.c_int64)),
@@ -271,14 +274,32 @@ class WindowOp:
         self._check(self._L.dz_window_op_drain(self._h), "drain")
 
     def poll_all(self, drain=True, copy=True):
+        """Collect every pollable batch as owned arrays. The C ownership
+        contract invalidates a batch's buffers at the next poll on the
+        handle, so a LIST of live views is impossible: poll_all always
+        materialises (the copy parameter is kept for signature
+        compatibility but no longer returns aliased views). Zero-copy
+        consumption goes through poll(copy=False) or poll_iter(), reading
+        each batch before advancing."""
+        del copy
         if drain:
             self.drain()
         out = []
         while True:
-            b = self.poll(copy=copy)
+            b = self.poll(copy=True)
             if b is None:
                 return out
             out.append(b)
+
+    def poll_iter(self, copy=False):
+        """Generator form of poll_all: yields one batch at a time; with
+        copy=False the yielded views are valid only until the next
+        iteration (exactly the single-poll contract)."""
+        while True:
+            b = self.poll(copy=copy)
+            if b is None:
+                return
+            yield b
 
     def finish(self):
         self._check(self._L.dz_window_op_finish(self._h), "finish")

@@ -73,6 +73,42 @@ struct PendingEvent {
     double bytes;
 };
 
+/* Append-only dictionary store shared between the push thread (appends) and
+ * the emission workers (read indices below their job's n_keys snapshot).
+ * A deque here was an unsynchronized race: push_back mutates the deque's
+ * internal block map, which a concurrent operator[] walks. This store never
+ * mutates published state: chunks are fixed arrays whose pointers publish
+ * once (release store), elements are written before the job hand-off (the
+ * e_mtx pair gives the happens-before), and appends to a chunk touch only
+ * indices no worker's snapshot covers. */
+template <typename T>
+struct ChunkedDict {
+    static constexpr int CHUNK_LOG = 12; /* 4096 entries per chunk */
+    static constexpr size_t CHUNK = 1u << CHUNK_LOG;
+    static constexpr size_t MAX_CHUNKS = 8192; /* 33M entries = the key cap */
+    std::atomic<T*> chunks[MAX_CHUNKS] = {};
+    size_t n = 0; /* push thread only; workers use their job snapshot */
+    void push_back(T v) {
+        size_t c = n >> CHUNK_LOG;
+        T* p = chunks[c].load(std::memory_order_relaxed);
+        if (!p) {
+            p = new T[CHUNK];
+            chunks[c].store(p, std::memory_order_release);
+        }
+        p[n & (CHUNK - 1)] = std::move(v);
+        n++;
+    }
+    const T& operator[](size_t i) const {
+        return chunks[i >> CHUNK_LOG].load(std::memory_order_acquire)
+                     [i & (CHUNK - 1)];
+    }
+    size_t size() const { return n; }
+    ~ChunkedDict() {
+        for (size_t c = 0; c <= (n >> CHUNK_LOG) && c < MAX_CHUNKS; c++)
+            delete[] chunks[c].load(std::memory_order_relaxed);
+    }
+};
+
 struct KStatAcc {
     uint64_t launches = 0;
     uint64_t timed = 0;   /* launches that carried HIP timing events */
@@ -126,13 +162,13 @@ struct dz_window_op {
 
     /* dictionary (host side; first-seen insertion order == emitted id order
      * is NOT assumed — emission sorts by first row; the dict only maps
-     * key values <-> dense ids) */
+     * key values <-> dense ids). ChunkedDict: append-only, safe for the
+     * emission workers to read below their job's n_keys snapshot while the
+     * push thread appends. */
     std::unordered_map<std::string, int32_t> dict_utf8;
-    std::deque<std::string> dict_strs;  /* deque: stable refs while the
-                                         * emission worker reads indices
-                                         * below its K snapshot */
+    ChunkedDict<std::string> dict_strs;
     std::unordered_map<int64_t, int32_t> dict_i64;
-    std::deque<int64_t> dict_vals;
+    ChunkedDict<int64_t> dict_vals;
     int64_t n_keys = 0; /* dense key count seen so far */
 
     /* persistent group state: one slab per slot, 5 contiguous fields of kcap
