@@ -70,27 +70,27 @@ void launch_scatter(hipStream_t stream, const int32_t* d_kid, const int64_t* d_t
 
 constexpr int FOLD_GCAP = 256; /* groups per bucket per fold chunk */
 
-void launch_regroup(hipStream_t stream, const uint32_t* d_meta,
-                    const uint4* d_grec, const uint32_t* d_bucket_base,
-                    const FoldChunk& fc, uint32_t* d_binoffs, uint32_t* d_binlens,
-                    uint4* d_gfrec);
+/* fused stable-split + fold: per-(window,key) bins staged in LDS fold
+ * straight into register accumulators seeded from the window-slot slabs —
+ * no reordered records in HBM */
+void launch_regroup_fold(hipStream_t stream, const uint32_t* d_meta,
+                         const uint4* d_grec, const uint32_t* d_bucket_base,
+                         const FoldChunk& fc, const int32_t* d_slot_of_widx,
+                         uint64_t* s_cnt, double* s_min, double* s_max,
+                         double* s_sum, uint64_t* s_first);
 
 void launch_regroup_l1(hipStream_t stream, const uint32_t* d_meta,
                        const uint4* d_grec, const uint32_t* d_bucket_base,
                        const FoldChunk& fc, uint32_t* d_b1offs,
                        uint32_t* d_b1lens, uint32_t* d_meta2, uint4* d_grec2);
 
-void launch_regroup_l2(hipStream_t stream, const uint32_t* d_meta2,
-                       const uint4* d_grec2, const uint32_t* d_bucket_base,
-                       const FoldChunk& fc, int nb1, const uint32_t* d_b1offs,
-                       const uint32_t* d_b1lens, uint32_t* d_binoffs,
-                       uint32_t* d_binlens, uint4* d_gfrec);
-
-void launch_fold3(hipStream_t stream, const uint4* d_gfrec,
-                  const uint32_t* d_bucket_base, const uint32_t* d_binoffs,
-                  const uint32_t* d_binlens, const FoldChunk& fc,
-                  const int32_t* d_slot_of_widx, uint64_t* s_cnt, double* s_min,
-                  double* s_max, double* s_sum, uint64_t* s_first);
+void launch_regroup_l2_fold(hipStream_t stream, const uint32_t* d_meta2,
+                            const uint4* d_grec2, const uint32_t* d_bucket_base,
+                            const FoldChunk& fc, int nb1,
+                            const uint32_t* d_b1offs, const uint32_t* d_b1lens,
+                            const int32_t* d_slot_of_widx, uint64_t* s_cnt,
+                            double* s_min, double* s_max, double* s_sum,
+                            uint64_t* s_first);
 
 struct EmitFilter {
     int32_t on;      /* 0 = no filter */

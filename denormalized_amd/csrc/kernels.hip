@@ -13,10 +13,15 @@
  *   k_scatter  stable partition: records (meta,rowidx,value) land in bucket
  *              regions IN GLOBAL ROW ORDER (weighted intra-wave ranks via
  *              wave-64 shuffles + wave-serialized LDS cursors)
- *   k_fold     one wave per bucket; each lane owns one (window,key) group and
- *              folds its rows IN ROW ORDER — count/min/max and the f64 sum
- *              are bit-identical to the reference's sequential accumulator
- *              updates (update_batch call order, grouped_window_agg_stream.rs:533)
+ *   k_regroup_t<RG_*_FOLD>  one block per bucket (or (bucket,bin1) segment):
+ *              stable per-supertile split of the bucket's records into
+ *              per-(window,key) bins IN LDS, then each bin's staged segment
+ *              folds directly into a per-thread register accumulator IN ROW
+ *              ORDER — count/min/max and the f64 sum are bit-identical to
+ *              the reference's sequential accumulator updates (update_batch
+ *              call order, grouped_window_agg_stream.rs:533). The reordered
+ *              records never touch HBM (the unfused regroup+fold pair they
+ *              replace moved ~32 B/row of pure materialization traffic).
  *
  * No atomics on the data path (only LDS histogram counts); every kernel is
  * deterministic. Roofline: HBM bandwidth (no contraction => no MFMA).
@@ -492,31 +497,42 @@ void launch_scatter(hipStream_t s, const int32_t* d_kid, const int64_t* d_ts,
 
 constexpr int GCAP = FOLD_GCAP; /* bins (groups) per bucket per chunk */
 
-/* One templated stable-split kernel, three modes:
- *  DIRECT (gtot <= GCAP): bins = (widx,kloc) groups, output 12 B records
+/* One templated stable-split kernel, four modes:
+ *  DIRECT_FOLD (gtot <= GCAP per chunk): bins = (widx,kloc) groups; each
+ *     supertile's LDS-staged bins fold DIRECTLY into per-bin register
+ *     accumulators (one thread owns one bin), seeded from / written back to
+ *     the persistent window-slot slab. No reordered records ever touch HBM:
+ *     the unfused form measured 324 MB regroup write + 159 MB fold read per
+ *     8M-row cfg2 launch (profiles/hbm_traffic.json) — pure materialization
+ *     round-trip, all of it gone here. Row-order folding is preserved:
+ *     supertiles advance in row order and staged records within a bin are
+ *     ranked in row order, so each bin's accumulator sees its rows in
+ *     exactly the reference's update_batch order (bit-exact f64 sum).
  *  L1 (two-level, first pass): bins = (kloc>>8, widx) <= 256, output FULL
  *     16 B records (meta travels to L2)
- *  L2 (two-level, second pass): one block per (bucket, bin1) segment of the
- *     L1 output; bins = kloc & 255, output 12 B records + final group
- *     segment offsets for the fold
- * Structure (shared): whole-segment bin counts (LDS atomics), prefix, then
- * per-supertile ranked placement (wave-quarters: wave order == row order,
- * private cursors, bit-ballot same-bin masks) into LDS staging flushed
- * bin-major so writes coalesce. */
-enum { RG_DIRECT = 0, RG_L1 = 1, RG_L2 = 2 };
+ *  L2_FOLD (two-level, second pass): one block per (bucket, bin1) segment
+ *     of the L1 output; bins = kloc & 255, folded like DIRECT_FOLD
+ * Structure (shared): per-supertile ranked placement (wave-quarters: wave
+ * order == row order, private cursors, bit-ballot same-bin masks) into LDS
+ * staging; L1 flushes bin-major so writes coalesce, fold modes consume the
+ * staging in place. */
+enum { RG_L1 = 1, RG_DIRECT_FOLD = 3, RG_L2_FOLD = 4 };
 
 template <int MODE>
 __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
         const uint4* rrec, const uint32_t* bucket_base, FoldChunk fc,
         const uint32_t* b1offs, const uint32_t* b1lens, uint32_t* binoffs,
-        uint32_t* binlens, uint32_t* ometa, uint4* orec) {
-    __shared__ uint32_t cnt[GCAP];    /* whole-segment bin counts */
-    __shared__ uint32_t gcur[GCAP];   /* segment-region bin cursors */
+        uint32_t* binlens, uint32_t* ometa, uint4* orec,
+        const int32_t* slot_of_widx, uint64_t* s_cnt, double* s_min,
+        double* s_max, double* s_sum, uint64_t* s_first) {
+    constexpr bool FOLD = (MODE == RG_DIRECT_FOLD || MODE == RG_L2_FOLD);
+    __shared__ uint32_t cnt[GCAP];    /* whole-segment bin counts (L1) */
+    __shared__ uint32_t gcur[GCAP];   /* segment-region bin cursors (L1) */
     __shared__ uint32_t stcnt4[WAVES_PER_BLOCK][GCAP]; /* per-wave-quarter */
     __shared__ uint32_t stoffs[GCAP]; /* per-supertile bin prefix */
     __shared__ uint32_t wcur[WAVES_PER_BLOCK][GCAP];   /* per-wave cursors */
     __shared__ uint32_t s_meta[ST_RECORDS]; /* RG_L1 only */
-    __shared__ uint32_t s_dest[ST_RECORDS];
+    __shared__ uint32_t s_dest[ST_RECORDS]; /* RG_L1 only */
     __shared__ uint4 s_rec[ST_RECORDS];
     __shared__ uint32_t s_total;
 
@@ -526,27 +542,23 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
     const uint32_t bklo = bucket_base[bkt];
     uint32_t lo = bklo, hi = bucket_base[bkt + 1];
     const int nk = fc.k_hi - fc.k_lo;
-    int nbins, obase;
-    if (MODE == RG_DIRECT) {
+    int nbins, obase = 0;
+    if (MODE == RG_DIRECT_FOLD) {
         nbins = (fc.w_hi - fc.w_lo) * nk;
-        obase = bkt * fc.bin_stride;
     } else if (MODE == RG_L1) {
         nbins = ((nk + 255) >> 8) * fc.tl_nw;
         obase = bkt * 256;
-    } else { /* RG_L2 */
+    } else { /* RG_L2_FOLD */
         const int bin1 = blockIdx.y;
         nbins = 256;
         lo = bklo + b1offs[bkt * 256 + bin1];
         hi = lo + b1lens[bkt * 256 + bin1];
-        obase = bkt * fc.bin_stride + bin1 * 256;
     }
-    for (int g = threadIdx.x; g < GCAP; g += BLOCK) cnt[g] = 0;
-    __syncthreads();
 
     auto bin_of = [&](uint32_t ms_) -> uint32_t {
         const int widx = (int)((ms_ >> META_WIDX_SHIFT) & META_WIDX_MASK);
         const int kloc = (int)(ms_ & META_KLOC_MASK);
-        if (MODE == RG_DIRECT) {
+        if (MODE == RG_DIRECT_FOLD) {
             if (widx < fc.w_lo || widx >= fc.w_hi || kloc < fc.k_lo ||
                 kloc >= fc.k_hi)
                 return 0x1FFu; /* outside this chunk */
@@ -558,30 +570,64 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
         }
     };
 
-    /* pass 1: whole-segment bin counts */
-    for (uint32_t i = lo + threadIdx.x; i < hi; i += BLOCK) {
-        const uint32_t g = bin_of(rmeta[i]);
-        if (g != 0x1FFu) atomicAdd(&cnt[g], 1u);
-    }
-    __syncthreads();
-    if (threadIdx.x == 0) { /* tiny exclusive prefix over <=GCAP bins */
-        uint32_t run = 0;
-        for (int g = 0; g < GCAP; g++) {
-            uint32_t t = cnt[g];
-            gcur[g] = run;
-            run += t;
+    /* fold modes: thread t owns bin t — accumulator lives in registers,
+     * seeded from the persistent slab exactly like the standalone fold */
+    int64_t f_sidx = 0;
+    bool f_own = false;
+    uint64_t f_cnt = 0, f_fst = ~0ULL;
+    double f_mn = 0.0, f_mx = 0.0, f_sm = 0.0;
+    if (FOLD) {
+        const int g = threadIdx.x;
+        int my_widx = 0, my_kloc = 0;
+        if (MODE == RG_DIRECT_FOLD) {
+            f_own = g < nbins;
+            my_widx = fc.w_lo + (f_own ? g / nk : 0);
+            my_kloc = fc.k_lo + (f_own ? g % nk : 0);
+        } else { /* RG_L2_FOLD: bin1 = (kloc>>8)*nw + widx; bin = kloc&255 */
+            const int bin1 = blockIdx.y;
+            my_widx = bin1 % fc.tl_nw;
+            my_kloc = (bin1 / fc.tl_nw) * 256 + g;
+            f_own = my_kloc < nk && my_widx < fc.w_hi;
+        }
+        if (f_own) {
+            int64_t slot = slot_of_widx[my_widx];
+            f_sidx = slot * (5 * fc.kcap) + (((int64_t)my_kloc << LOG_NB) | bkt);
+            f_cnt = s_cnt[f_sidx];
+            f_fst = s_first[f_sidx];
+            if (f_cnt > 0) {
+                f_mn = s_min[f_sidx];
+                f_mx = s_max[f_sidx];
+                f_sm = s_sum[f_sidx];
+            }
         }
     }
-    __syncthreads();
-    /* publish segment layout for the next stage / fold (offsets are
-     * relative to the BUCKET region start) */
-    const uint32_t relbase = lo - bklo;
-    for (int g = threadIdx.x; g < nbins; g += BLOCK) {
-        binoffs[obase + g] = relbase + gcur[g];
-        binlens[obase + g] = cnt[g];
-    }
-    if (lo == hi) return;
 
+    if (MODE == RG_L1) {
+        for (int g = threadIdx.x; g < GCAP; g += BLOCK) cnt[g] = 0;
+        __syncthreads();
+        /* pass 1: whole-segment bin counts */
+        for (uint32_t i = lo + threadIdx.x; i < hi; i += BLOCK) {
+            const uint32_t g = bin_of(rmeta[i]);
+            if (g != 0x1FFu) atomicAdd(&cnt[g], 1u);
+        }
+        __syncthreads();
+        if (threadIdx.x == 0) { /* tiny exclusive prefix over <=GCAP bins */
+            uint32_t run = 0;
+            for (int g = 0; g < GCAP; g++) {
+                uint32_t t = cnt[g];
+                gcur[g] = run;
+                run += t;
+            }
+        }
+        __syncthreads();
+        /* publish segment layout for the next stage (offsets are
+         * relative to the BUCKET region start) */
+        const uint32_t relbase = lo - bklo;
+        for (int g = threadIdx.x; g < nbins; g += BLOCK) {
+            binoffs[obase + g] = relbase + gcur[g];
+            binlens[obase + g] = cnt[g];
+        }
+    }
     for (uint32_t st0 = lo; st0 < hi; st0 += ST_RECORDS) {
         const uint32_t st1 = min(hi, st0 + (uint32_t)ST_RECORDS);
         /* wave-quarter bounds (contiguous: wave order == row order) */
@@ -624,8 +670,8 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
                 g = bin_of(ms);
                 if (g != 0x1FFu) {
                     rec = rrec[i];
-                    /* fold records (DIRECT, L2) carry validity in bit 31;
-                     * L1 passes the raw rowidx through (meta travels) */
+                    /* fold records carry validity in bit 31; L1 passes the
+                     * raw rowidx through (meta travels) */
                     if (MODE != RG_L1)
                         rec.z |= (ms >> META_VALID_SHIFT) << 31;
                 }
@@ -652,34 +698,73 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
             }
             if (g != 0x1FFu) {
                 s_rec[pos] = rec;
-                if (MODE == RG_L1) s_meta[pos] = ms;
-                s_dest[pos] = lo + gcur[g] + (pos - stoffs[g]);
+                if (MODE == RG_L1) {
+                    s_meta[pos] = ms;
+                    s_dest[pos] = lo + gcur[g] + (pos - stoffs[g]);
+                }
             }
         }
         __syncthreads();
-        /* flush (bin-major staging => coalesced runs) */
-        const uint32_t tot = s_total;
-        for (uint32_t p = threadIdx.x; p < tot; p += BLOCK) {
-            const uint32_t d = s_dest[p];
-            orec[d] = s_rec[p];
-            if (MODE == RG_L1) ometa[d] = s_meta[p];
+        if (FOLD) {
+            /* consume the staging in place: thread t folds bin t's staged
+             * segment (row-ordered) into its register accumulator — the
+             * reference's sequential update_batch order, no HBM round-trip */
+            const int g = threadIdx.x;
+            const uint32_t seg0 = stoffs[g];
+            const uint32_t seg1 = (g + 1 < GCAP) ? stoffs[g + 1] : s_total;
+            if (f_own && seg1 > seg0) {
+                if (f_fst == ~0ULL)
+                    f_fst = ((uint64_t)fc.batch_seq << 32) |
+                            (s_rec[seg0].z & 0x7FFFFFFFu);
+                for (uint32_t r = seg0; r < seg1; r++) {
+                    const uint4 rec = s_rec[r];
+                    const double v = __longlong_as_double(
+                        (long long)(((uint64_t)rec.y << 32) | rec.x));
+                    if (rec.z >> 31) {
+                        const bool fresh = f_cnt == 0;
+                        f_mn = (fresh || v < f_mn) ? v : f_mn;
+                        f_mx = (fresh || v > f_mx) ? v : f_mx;
+                        f_sm += v;
+                        f_cnt++;
+                    }
+                }
+            }
+            __syncthreads();
+        } else {
+            /* L1 flush (bin-major staging => coalesced runs) */
+            const uint32_t tot = s_total;
+            for (uint32_t p = threadIdx.x; p < tot; p += BLOCK) {
+                const uint32_t d = s_dest[p];
+                orec[d] = s_rec[p];
+                ometa[d] = s_meta[p];
+            }
+            __syncthreads();
+            for (int g = threadIdx.x; g < GCAP; g += BLOCK) {
+                uint32_t s = 0;
+                for (int w = 0; w < WAVES_PER_BLOCK; w++) s += stcnt4[w][g];
+                gcur[g] += s;
+            }
+            __syncthreads();
         }
-        __syncthreads();
-        for (int g = threadIdx.x; g < GCAP; g += BLOCK) {
-            uint32_t s = 0;
-            for (int w = 0; w < WAVES_PER_BLOCK; w++) s += stcnt4[w][g];
-            gcur[g] += s;
-        }
-        __syncthreads();
+    }
+    if (FOLD && f_own) {
+        s_cnt[f_sidx] = f_cnt;
+        s_first[f_sidx] = f_fst;
+        s_min[f_sidx] = f_mn;
+        s_max[f_sidx] = f_mx;
+        s_sum[f_sidx] = f_sm;
     }
 }
 
-void launch_regroup(hipStream_t s, const uint32_t* d_meta, const uint4* d_grec,
-                    const uint32_t* d_bucket_base, const FoldChunk& fc,
-                    uint32_t* d_binoffs, uint32_t* d_binlens, uint4* d_gfrec) {
-    hipLaunchKernelGGL(k_regroup_t<RG_DIRECT>, dim3(NB), dim3(BLOCK), 0, s,
+void launch_regroup_fold(hipStream_t s, const uint32_t* d_meta,
+                         const uint4* d_grec, const uint32_t* d_bucket_base,
+                         const FoldChunk& fc, const int32_t* d_slot_of_widx,
+                         uint64_t* s_cnt, double* s_min, double* s_max,
+                         double* s_sum, uint64_t* s_first) {
+    hipLaunchKernelGGL(k_regroup_t<RG_DIRECT_FOLD>, dim3(NB), dim3(BLOCK), 0, s,
                        d_meta, d_grec, d_bucket_base, fc, nullptr, nullptr,
-                       d_binoffs, d_binlens, nullptr, d_gfrec);
+                       nullptr, nullptr, nullptr, nullptr, d_slot_of_widx,
+                       s_cnt, s_min, s_max, s_sum, s_first);
 }
 
 void launch_regroup_l1(hipStream_t s, const uint32_t* d_meta, const uint4* d_grec,
@@ -688,117 +773,22 @@ void launch_regroup_l1(hipStream_t s, const uint32_t* d_meta, const uint4* d_gre
                        uint4* d_grec2) {
     hipLaunchKernelGGL(k_regroup_t<RG_L1>, dim3(NB), dim3(BLOCK), 0, s, d_meta,
                        d_grec, d_bucket_base, fc, nullptr, nullptr, d_b1offs,
-                       d_b1lens, d_meta2, d_grec2);
+                       d_b1lens, d_meta2, d_grec2, nullptr, nullptr, nullptr,
+                       nullptr, nullptr, nullptr);
 }
 
-void launch_regroup_l2(hipStream_t s, const uint32_t* d_meta2, const uint4* d_grec2,
-                       const uint32_t* d_bucket_base, const FoldChunk& fc,
-                       int nb1, const uint32_t* d_b1offs, const uint32_t* d_b1lens,
-                       uint32_t* d_binoffs, uint32_t* d_binlens, uint4* d_gfrec) {
-    hipLaunchKernelGGL(k_regroup_t<RG_L2>, dim3(NB, nb1), dim3(BLOCK), 0, s,
+void launch_regroup_l2_fold(hipStream_t s, const uint32_t* d_meta2,
+                            const uint4* d_grec2, const uint32_t* d_bucket_base,
+                            const FoldChunk& fc, int nb1,
+                            const uint32_t* d_b1offs, const uint32_t* d_b1lens,
+                            const int32_t* d_slot_of_widx, uint64_t* s_cnt,
+                            double* s_min, double* s_max, double* s_sum,
+                            uint64_t* s_first) {
+    hipLaunchKernelGGL(k_regroup_t<RG_L2_FOLD>, dim3(NB, nb1), dim3(BLOCK), 0, s,
                        d_meta2, d_grec2, d_bucket_base, fc, d_b1offs, d_b1lens,
-                       d_binoffs, d_binlens, nullptr, d_gfrec);
-}
-
-/* ------------------------------------------------------------------ */
-/* segmented fold — one LANE per (window,key) group walking its own    */
-/* segment sequentially: 64 rows progress per wave instruction, rows   */
-/* fold in row order (bit-exact vs the reference accumulators).        */
-/* ------------------------------------------------------------------ */
-
-__global__ __launch_bounds__(BLOCK) void k_fold3(const uint4* gfrec,
-        const uint32_t* bucket_base, const uint32_t* binoffs,
-        const uint32_t* binlens, FoldChunk fc, const int32_t* slot_of_widx,
-        uint64_t* s_cnt, double* s_min, double* s_max, double* s_sum,
-        uint64_t* s_first) {
-    const int nk = fc.k_hi - fc.k_lo;
-    const int glocal_n = fc.tl_nw ? fc.bin_stride : (fc.w_hi - fc.w_lo) * nk;
-    const int waves_per_bkt = (glocal_n + 63) / 64;
-    const int64_t gw = (blockIdx.x * (int64_t)BLOCK + threadIdx.x) >> 6;
-    const int bkt = (int)(gw / waves_per_bkt);
-    const int lane = threadIdx.x & 63;
-    const int g = (int)(gw % waves_per_bkt) * 64 + lane;
-    if (bkt >= NB) return;
-    bool own = g < glocal_n;
-    int my_widx, my_kloc;
-    if (fc.tl_nw) {
-        /* two-level numbering: g = ((kloc>>8)*nw + widx)*256 + (kloc&255) */
-        my_widx = (g >> 8) % fc.tl_nw;
-        my_kloc = ((g >> 8) / fc.tl_nw) * 256 + (g & 255);
-        own = own && my_kloc < nk && my_widx < fc.w_hi;
-    } else {
-        my_widx = fc.w_lo + (own ? g / nk : 0);
-        my_kloc = fc.k_lo + (own ? g % nk : 0);
-    }
-
-    const uint32_t lo = bucket_base[bkt];
-    uint32_t off = 0, len = 0;
-    if (own) {
-        off = binoffs[(int64_t)bkt * fc.bin_stride + g];
-        len = binlens[(int64_t)bkt * fc.bin_stride + g];
-    }
-    int64_t sidx = 0;
-    uint64_t cnt = 0, fst = ~0ULL;
-    double mn = 0.0, mx = 0.0, sm = 0.0;
-    if (own) {
-        int64_t slot = slot_of_widx[my_widx];
-        sidx = slot * (5 * fc.kcap) + (((int64_t)my_kloc << LOG_NB) | bkt);
-        cnt = s_cnt[sidx];
-        fst = s_first[sidx];
-        if (cnt > 0) {
-            mn = s_min[sidx];
-            mx = s_max[sidx];
-            sm = s_sum[sidx];
-        } else {
-            sm = 0.0;
-        }
-    }
-    if (own && len > 0 && fst == ~0ULL)
-        fst = ((uint64_t)fc.batch_seq << 32) | (gfrec[lo + off].z & 0x7FFFFFFFu);
-    uint32_t mlen = len;
-    for (int o = 32; o > 0; o >>= 1)
-        mlen = max(mlen, (uint32_t)__shfl_down((int)mlen, o));
-    mlen = (uint32_t)__builtin_amdgcn_readlane((int)mlen, 0);
-#pragma unroll 4
-    for (uint32_t r = 0; r < mlen; r++) {
-        const bool act = own && r < len;
-        const uint4 rec = act ? gfrec[lo + off + r] : make_uint4(0u, 0u, 0u, 0u);
-        const double v = __longlong_as_double(
-            (long long)(((uint64_t)rec.y << 32) | rec.x));
-        const bool upd = act && (rec.z >> 31);
-        const bool fresh = cnt == 0;
-        mn = (upd && (fresh || v < mn)) ? v : mn;
-        mx = (upd && (fresh || v > mx)) ? v : mx;
-        const double sm2 = sm + v;
-        sm = upd ? sm2 : sm;
-        cnt += upd ? 1 : 0;
-    }
-    if (own) {
-        s_cnt[sidx] = cnt;
-        s_first[sidx] = fst;
-        s_min[sidx] = mn;
-        s_max[sidx] = mx;
-        s_sum[sidx] = sm;
-    }
-}
-
-void launch_fold3(hipStream_t s, const uint4* d_gfrec,
-                  const uint32_t* d_bucket_base, const uint32_t* d_binoffs,
-                  const uint32_t* d_binlens, const FoldChunk& fc,
-                  const int32_t* d_slot_of_widx, uint64_t* s_cnt, double* s_min,
-                  double* s_max, double* s_sum, uint64_t* s_first) {
-    int nk = fc.k_hi - fc.k_lo;
-    /* must mirror the kernel's numbering: two-level mode spans bin_stride */
-    int glocal_n = fc.tl_nw ? fc.bin_stride : (fc.w_hi - fc.w_lo) * nk;
-    int waves_per_bkt = (glocal_n + 63) / 64;
-    int64_t waves = (int64_t)NB * waves_per_bkt;
-    int blocks = (int)((waves + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK);
-    hipLaunchKernelGGL(k_fold3, dim3(blocks), dim3(BLOCK), 0, s, d_gfrec,
-                       d_bucket_base, d_binoffs, d_binlens, fc, d_slot_of_widx,
+                       nullptr, nullptr, nullptr, nullptr, d_slot_of_widx,
                        s_cnt, s_min, s_max, s_sum, s_first);
 }
-
-
 
 /* ------------------------------------------------------------------ */
 /* device-side emission: compact touched groups, stable radix sort by  */

@@ -206,10 +206,6 @@ struct dz_window_op {
     int C_cap = 0;
     uint32_t* d_meta = nullptr;
     uint4* d_grec = nullptr;  /* 16 B {val, rowidx} records (scatter output) */
-    uint4* d_gfrec = nullptr; /* group-segmented fold records */
-    uint32_t* d_binoffs = nullptr;
-    uint32_t* d_binlens = nullptr;
-    int64_t bin_cap = 0;   /* bins per bucket the binoffs/lens arrays hold */
     uint32_t* d_b1offs = nullptr; /* two-level L1 segment layout [NB][256] */
     uint32_t* d_b1lens = nullptr;
     uint32_t* d_meta2 = nullptr;  /* two-level intermediate records */
@@ -624,8 +620,7 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_gofs[0]); hipFree(op->d_gofs[1]);
     hipFree(op->d_total[0]); hipFree(op->d_total[1]);
     hipFree(op->d_base[0]); hipFree(op->d_base[1]);
-    hipFree(op->d_meta); hipFree(op->d_grec); hipFree(op->d_gfrec);
-    hipFree(op->d_binoffs); hipFree(op->d_binlens);
+    hipFree(op->d_meta); hipFree(op->d_grec);
     hipFree(op->d_b1offs); hipFree(op->d_b1lens);
     hipFree(op->d_meta2); hipFree(op->d_grec2);
     hipFree(op->d_scalars[0]); hipFree(op->d_scalars[1]);
@@ -1364,22 +1359,10 @@ static dz_status ensure_scratch(dz_window_op* op, int C, int64_t nrec) {
     }
     if (nrec > op->rec_cap) {
         quiesce(op);
-        hipFree(op->d_meta); hipFree(op->d_grec); hipFree(op->d_gfrec);
+        hipFree(op->d_meta); hipFree(op->d_grec);
         CHK(op, hipMalloc(&op->d_meta, (size_t)nrec * 4));
         CHK(op, hipMalloc(&op->d_grec, (size_t)nrec * 16));
-        CHK(op, hipMalloc(&op->d_gfrec, (size_t)nrec * 16));
         op->rec_cap = nrec;
-    }
-    return DZ_OK;
-}
-
-static dz_status ensure_bins(dz_window_op* op, int64_t bins_per_bucket) {
-    if (bins_per_bucket > op->bin_cap) {
-        hipFree(op->d_binoffs);
-        hipFree(op->d_binlens);
-        CHK(op, hipMalloc(&op->d_binoffs, (size_t)dz::NB * bins_per_bucket * 4));
-        CHK(op, hipMalloc(&op->d_binlens, (size_t)dz::NB * bins_per_bucket * 4));
-        op->bin_cap = bins_per_bucket;
     }
     return DZ_OK;
 }
@@ -1658,10 +1641,9 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
     int64_t nb1 = khigh * nw;
     if (gtot > dz::FOLD_GCAP && nb1 <= 256) {
         /* two-level regime (large keyspaces, e.g. cfg3's 1M keys): split by
-         * (kloc>>8, widx) then by kloc&255 — both <=256-bin LDS passes at
-         * full occupancy, no chunked re-reads */
-        int64_t stride = nb1 * 256;
-        if (ensure_bins(op, stride) != DZ_OK) return DZ_ERR;
+         * (kloc>>8, widx), then the second pass splits each bin1 by kloc&255
+         * AND folds the staged bins in place (fused — the split-out records
+         * and the fold's re-read of them never touch HBM) */
         if (!op->d_b1offs) {
             CHK(op, hipMalloc(&op->d_b1offs, (size_t)dz::NB * 256 * 4));
             CHK(op, hipMalloc(&op->d_b1lens, (size_t)dz::NB * 256 * 4));
@@ -1679,7 +1661,7 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
         fc.k_hi = (int32_t)klocs;
         fc.kcap = op->kcap;
         fc.batch_seq = op->batch_seq;
-        fc.bin_stride = (int32_t)stride;
+        fc.bin_stride = (int32_t)(nb1 * 256);
         fc.tl_nw = (int32_t)nw;
         timed(op, "regroup", (double)nrec_max * 40, [&] {
             dz::launch_regroup_l1(op->stream, op->d_meta, op->d_grec,
@@ -1687,20 +1669,14 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
                                   fc, op->d_b1offs, op->d_b1lens, op->d_meta2,
                                   op->d_grec2);
         });
-        timed(op, "regroup", (double)nrec_max * 40, [&] {
-            dz::launch_regroup_l2(op->stream, op->d_meta2, op->d_grec2,
-                                  op->d_base[b], fc, (int)nb1, op->d_b1offs,
-                                  op->d_b1lens, op->d_binoffs, op->d_binlens,
-                                  op->d_gfrec);
-        });
-        timed(op, "fold", (double)nrec_max * 16, [&] {
-            dz::launch_fold3(op->stream, op->d_gfrec, op->d_base[b],
-                             op->d_binoffs,
-                             op->d_binlens, fc, op->d_slotmap, op->s_cnt,
-                             op->s_min, op->s_max, op->s_sum, op->s_first);
+        timed(op, "regfold", (double)nrec_max * 24, [&] {
+            dz::launch_regroup_l2_fold(op->stream, op->d_meta2, op->d_grec2,
+                                       op->d_base[b], fc, (int)nb1,
+                                       op->d_b1offs, op->d_b1lens,
+                                       op->d_slotmap, op->s_cnt, op->s_min,
+                                       op->s_max, op->s_sum, op->s_first);
         });
     } else {
-        if (ensure_bins(op, dz::FOLD_GCAP) != DZ_OK) return DZ_ERR;
         for (int64_t k_lo = 0; k_lo < klocs; k_lo += dz::FOLD_GCAP) {
             int32_t nk = (int32_t)std::min<int64_t>(dz::FOLD_GCAP, klocs - k_lo);
             int32_t wstep = dz::FOLD_GCAP / nk;
@@ -1714,16 +1690,11 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
                 fc.batch_seq = op->batch_seq;
                 fc.bin_stride = dz::FOLD_GCAP;
                 fc.tl_nw = 0;
-                timed(op, "regroup", (double)nrec_max * 40, [&] {
-                    dz::launch_regroup(op->stream, op->d_meta, op->d_grec,
-                                       op->d_base[b], fc, op->d_binoffs,
-                                       op->d_binlens, op->d_gfrec);
-                });
-                timed(op, "fold", (double)nrec_max * 16, [&] {
-                    dz::launch_fold3(op->stream, op->d_gfrec, op->d_base[b],
-                                     op->d_binoffs, op->d_binlens, fc,
-                                     op->d_slotmap, op->s_cnt, op->s_min,
-                                     op->s_max, op->s_sum, op->s_first);
+                timed(op, "regfold", (double)nrec_max * 24, [&] {
+                    dz::launch_regroup_fold(op->stream, op->d_meta, op->d_grec,
+                                            op->d_base[b], fc, op->d_slotmap,
+                                            op->s_cnt, op->s_min, op->s_max,
+                                            op->s_sum, op->s_first);
                 });
             }
         }
