@@ -60,32 +60,33 @@ void launch_scan(hipStream_t stream, const uint32_t* d_ghist, int C,
                  uint32_t* d_psum, uint32_t* d_bucket_base /*NB+1*/,
                  uint32_t* d_gofs);
 
-/* record payloads are 16 B uint4s: {val lo, val hi, rowidx, unused}; the
- * rowidx word carries the validity bit (bit 31) in FOLD-stage records */
+/* record payloads are 16 B uint4s: {val lo, val hi, rowidx, meta}; the
+ * rowidx word carries the validity bit (bit 31) in FOLD-stage records and
+ * .w carries the meta word (kloc|widx|valid) — there is no separate meta
+ * array */
 void launch_scatter(hipStream_t stream, const int32_t* d_kid, const int64_t* d_ts,
                     const double* d_vals, const uint8_t* d_validity /*bitmap|null*/,
                     int64_t n, int64_t chunk, int C, int32_t st_rows,
-                    const WinParams& wp, const uint32_t* d_gofs, uint32_t* d_meta,
-                    uint4* d_grec);
+                    const WinParams& wp, const uint32_t* d_gofs, uint4* d_grec);
 
 constexpr int FOLD_GCAP = 256; /* groups per bucket per fold chunk */
 
 /* fused stable-split + fold: per-(window,key) bins staged in LDS fold
  * straight into register accumulators seeded from the window-slot slabs —
  * no reordered records in HBM */
-void launch_regroup_fold(hipStream_t stream, const uint32_t* d_meta,
-                         const uint4* d_grec, const uint32_t* d_bucket_base,
+void launch_regroup_fold(hipStream_t stream, const uint4* d_grec,
+                         const uint32_t* d_bucket_base,
                          const FoldChunk& fc, const int32_t* d_slot_of_widx,
                          uint64_t* s_cnt, double* s_min, double* s_max,
                          double* s_sum, uint64_t* s_first);
 
-void launch_regroup_l1(hipStream_t stream, const uint32_t* d_meta,
-                       const uint4* d_grec, const uint32_t* d_bucket_base,
+void launch_regroup_l1(hipStream_t stream, const uint4* d_grec,
+                       const uint32_t* d_bucket_base,
                        const FoldChunk& fc, uint32_t* d_b1offs,
-                       uint32_t* d_b1lens, uint32_t* d_meta2, uint4* d_grec2);
+                       uint32_t* d_b1lens, uint4* d_grec2);
 
-void launch_regroup_l2_fold(hipStream_t stream, const uint32_t* d_meta2,
-                            const uint4* d_grec2, const uint32_t* d_bucket_base,
+void launch_regroup_l2_fold(hipStream_t stream, const uint4* d_grec2,
+                            const uint32_t* d_bucket_base,
                             const FoldChunk& fc, int nb1,
                             const uint32_t* d_b1offs, const uint32_t* d_b1lens,
                             const int32_t* d_slot_of_widx, uint64_t* s_cnt,

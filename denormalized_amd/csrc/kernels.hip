@@ -305,13 +305,18 @@ void launch_scan(hipStream_t s, const uint32_t* d_ghist, int C,
 __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
         const int64_t* ts, const double* vals, const uint8_t* validity,
         int64_t n, int64_t chunk, int32_t st_rows, WinParams wp,
-        const uint32_t* gofs, uint32_t* rmeta, uint4* grec) {
+        const uint32_t* gofs, uint4* grec) {
     /* LDS-staged stable partition. Each wave owns a CONTIGUOUS QUARTER of the
      * supertile (wave order == row order), so staging cursors are per-wave
      * private: no cross-wave serialization, ~4 block barriers per supertile.
      * Records are placed bucket-major in LDS and flushed so adjacent lanes
      * write adjacent global addresses (the direct form measured 6x write
-     * amplification — profiles/hbm_traffic.json). */
+     * amplification — profiles/hbm_traffic.json). The flush additionally
+     * CARRIES the <=3 records that would leave a bucket's cursor
+     * mid-64B-sector over to the next supertile's run (resid/r_res), so a
+     * (chunk,bucket) pays at most one partial head + one partial tail
+     * sector instead of one partial sector per ~64 B run (the unaligned
+     * form measured 1.95x write amplification). */
     __shared__ uint32_t cur[NB];    /* global cursors for this block's chunk */
     __shared__ uint32_t cnt4[WAVES_PER_BLOCK][NB]; /* per-wave-quarter counts,
                                      * converted IN PLACE to per-wave staging
@@ -321,14 +326,16 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
     auto wofs = cnt4;
     /* staged records carry their meta in .w (the payload's spare lane) and
      * their destination is re-derived at flush by a 9-step binary search
-     * over offs[] — dropping the separate s_meta/s_dest arrays takes the
-     * block from 62.5 to 46 KiB LDS: 2 -> 3 blocks/CU on a latency-bound
-     * kernel */
+     * over offs[] — no per-record dest/meta staging arrays */
     __shared__ uint4 s_rec[ST_RECORDS];
+    __shared__ uint4 resid[NB][3];  /* sector-carry records per bucket */
+    __shared__ uint32_t r_res[NB];
     __shared__ uint32_t s_total;
 
-    for (int t = threadIdx.x; t < NB; t += BLOCK)
+    for (int t = threadIdx.x; t < NB; t += BLOCK) {
         cur[t] = gofs[(int64_t)blockIdx.x * NB + t];
+        r_res[t] = 0;
+    }
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
     const int64_t lo = blockIdx.x * chunk;
@@ -457,22 +464,50 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
         __syncthreads();
         /* flush: bucket-major staging => coalesced run writes; the record's
          * bin (and so its destination) falls out of a binary search over
-         * the bin prefix */
+         * the bin prefix. Pending order per bucket = carried residuals
+         * (older rows) first, then this supertile's staged run; only the
+         * sector-aligned prefix of the pending sequence is written unless
+         * this is the chunk's final supertile. */
+        const bool final_st = (st1 == hi);
         const uint32_t tot = s_total;
         for (uint32_t p = threadIdx.x; p < tot; p += BLOCK) {
             uint32_t b = 0;
             for (int stp = NB >> 1; stp; stp >>= 1)
                 if (b + stp < NB && offs[b + stp] <= p) b += stp;
-            const uint32_t d = cur[b] + (p - offs[b]);
-            const uint4 q = s_rec[p];
-            rmeta[d] = q.w;
-            grec[d] = q;
+            const uint32_t nb_ = ((b + 1 < NB) ? offs[b + 1] : tot) - offs[b];
+            const uint32_t rb = r_res[b];
+            const uint32_t cb = cur[b];
+            uint32_t wb = rb + nb_; /* final: flush all pending */
+            if (!final_st) {
+                const uint32_t Eb = (cb + wb) & ~3u;
+                wb = (Eb > cb) ? (Eb - cb) : 0u;
+            }
+            const uint32_t jp = rb + (p - offs[b]); /* pending index */
+            if (jp < wb) grec[cb + jp] = s_rec[p];
         }
         __syncthreads();
+        /* residual head writes + carry update (thread-owned per bucket) */
         for (int t = threadIdx.x; t < NB; t += BLOCK) {
-            /* bin total = offs delta (cnt4 was overlaid by the cursors) */
-            uint32_t nxt = (t + 1 < NB) ? offs[t + 1] : s_total;
-            cur[t] += nxt - offs[t];
+            const uint32_t nb_ = ((t + 1 < NB) ? offs[t + 1] : tot) - offs[t];
+            const uint32_t rb = r_res[t];
+            const uint32_t Tb = rb + nb_;
+            const uint32_t cb = cur[t];
+            uint32_t wb = Tb;
+            if (!final_st) {
+                const uint32_t Eb = (cb + Tb) & ~3u;
+                wb = (Eb > cb) ? (Eb - cb) : 0u;
+            }
+            for (uint32_t j = 0; j < rb && j < wb; j++)
+                grec[cb + j] = resid[t][j];
+            const uint32_t nr = Tb - wb; /* <= 3 */
+            uint4 tmp[3];
+            for (uint32_t j = 0; j < nr; j++) {
+                const uint32_t i = wb + j;
+                tmp[j] = (i < rb) ? resid[t][i] : s_rec[offs[t] + (i - rb)];
+            }
+            for (uint32_t j = 0; j < nr; j++) resid[t][j] = tmp[j];
+            r_res[t] = nr;
+            cur[t] = cb + wb;
         }
         __syncthreads();
     }
@@ -481,9 +516,9 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
 void launch_scatter(hipStream_t s, const int32_t* d_kid, const int64_t* d_ts,
                     const double* d_vals, const uint8_t* d_validity, int64_t n,
                     int64_t chunk, int C, int32_t st_rows, const WinParams& wp,
-                    const uint32_t* d_gofs, uint32_t* d_meta, uint4* d_grec) {
+                    const uint32_t* d_gofs, uint4* d_grec) {
     hipLaunchKernelGGL(k_scatter, dim3(C), dim3(BLOCK), 0, s, d_kid, d_ts, d_vals,
-                       d_validity, n, chunk, st_rows, wp, d_gofs, d_meta, d_grec);
+                       d_validity, n, chunk, st_rows, wp, d_gofs, d_grec);
 }
 
 /* ------------------------------------------------------------------ */
@@ -519,10 +554,10 @@ constexpr int GCAP = FOLD_GCAP; /* bins (groups) per bucket per chunk */
 enum { RG_L1 = 1, RG_DIRECT_FOLD = 3, RG_L2_FOLD = 4 };
 
 template <int MODE>
-__global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
-        const uint4* rrec, const uint32_t* bucket_base, FoldChunk fc,
+__global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint4* rrec,
+        const uint32_t* bucket_base, FoldChunk fc,
         const uint32_t* b1offs, const uint32_t* b1lens, uint32_t* binoffs,
-        uint32_t* binlens, uint32_t* ometa, uint4* orec,
+        uint32_t* binlens, uint4* orec,
         const int32_t* slot_of_widx, uint64_t* s_cnt, double* s_min,
         double* s_max, double* s_sum, uint64_t* s_first) {
     constexpr bool FOLD = (MODE == RG_DIRECT_FOLD || MODE == RG_L2_FOLD);
@@ -531,9 +566,8 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
     __shared__ uint32_t stcnt4[WAVES_PER_BLOCK][GCAP]; /* per-wave-quarter */
     __shared__ uint32_t stoffs[GCAP]; /* per-supertile bin prefix */
     __shared__ uint32_t wcur[WAVES_PER_BLOCK][GCAP];   /* per-wave cursors */
-    __shared__ uint32_t s_meta[ST_RECORDS]; /* RG_L1 only */
     __shared__ uint32_t s_dest[ST_RECORDS]; /* RG_L1 only */
-    __shared__ uint4 s_rec[ST_RECORDS];
+    __shared__ uint4 s_rec[ST_RECORDS];     /* meta travels in .w */
     __shared__ uint32_t s_total;
 
     const int bkt = blockIdx.x;
@@ -605,9 +639,11 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
     if (MODE == RG_L1) {
         for (int g = threadIdx.x; g < GCAP; g += BLOCK) cnt[g] = 0;
         __syncthreads();
-        /* pass 1: whole-segment bin counts */
+        /* pass 1: whole-segment bin counts (meta rides in each record's
+         * .w lane; the supertile passes below re-read the same lines out
+         * of L2/MALL) */
         for (uint32_t i = lo + threadIdx.x; i < hi; i += BLOCK) {
-            const uint32_t g = bin_of(rmeta[i]);
+            const uint32_t g = bin_of(rrec[i].w);
             if (g != 0x1FFu) atomicAdd(&cnt[g], 1u);
         }
         __syncthreads();
@@ -638,7 +674,7 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
             for (int w = 0; w < WAVES_PER_BLOCK; w++) stcnt4[w][g] = 0;
         __syncthreads();
         for (uint32_t i = w0 + lane; i < w1; i += 64) {
-            const uint32_t g = bin_of(rmeta[i]);
+            const uint32_t g = bin_of(rrec[i].w);
             if (g != 0x1FFu) atomicAdd(&stcnt4[wave][g], 1u);
         }
         __syncthreads();
@@ -663,18 +699,14 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
         for (uint32_t t0 = w0; t0 < w1; t0 += 64) {
             const uint32_t i = t0 + lane;
             uint32_t g = 0x1FFu; /* sentinel above GCAP-1 */
-            uint32_t ms = 0;
             uint4 rec = make_uint4(0u, 0u, 0u, 0u);
             if (i < w1) {
-                ms = rmeta[i];
-                g = bin_of(ms);
-                if (g != 0x1FFu) {
-                    rec = rrec[i];
-                    /* fold records carry validity in bit 31; L1 passes the
-                     * raw rowidx through (meta travels) */
-                    if (MODE != RG_L1)
-                        rec.z |= (ms >> META_VALID_SHIFT) << 31;
-                }
+                rec = rrec[i];
+                g = bin_of(rec.w);
+                /* fold records carry validity in bit 31; L1 passes the
+                 * raw rowidx through (meta stays in .w) */
+                if (MODE != RG_L1 && g != 0x1FFu)
+                    rec.z |= (rec.w >> META_VALID_SHIFT) << 31;
             }
             /* same-bin mask via bit-ballots over the 9 bin-id bits */
             uint64_t same = ~0ULL;
@@ -698,10 +730,8 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
             }
             if (g != 0x1FFu) {
                 s_rec[pos] = rec;
-                if (MODE == RG_L1) {
-                    s_meta[pos] = ms;
+                if (MODE == RG_L1)
                     s_dest[pos] = lo + gcur[g] + (pos - stoffs[g]);
-                }
             }
         }
         __syncthreads();
@@ -733,11 +763,8 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
         } else {
             /* L1 flush (bin-major staging => coalesced runs) */
             const uint32_t tot = s_total;
-            for (uint32_t p = threadIdx.x; p < tot; p += BLOCK) {
-                const uint32_t d = s_dest[p];
-                orec[d] = s_rec[p];
-                ometa[d] = s_meta[p];
-            }
+            for (uint32_t p = threadIdx.x; p < tot; p += BLOCK)
+                orec[s_dest[p]] = s_rec[p];
             __syncthreads();
             for (int g = threadIdx.x; g < GCAP; g += BLOCK) {
                 uint32_t s = 0;
@@ -756,37 +783,36 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint32_t* rmeta,
     }
 }
 
-void launch_regroup_fold(hipStream_t s, const uint32_t* d_meta,
-                         const uint4* d_grec, const uint32_t* d_bucket_base,
+void launch_regroup_fold(hipStream_t s, const uint4* d_grec,
+                         const uint32_t* d_bucket_base,
                          const FoldChunk& fc, const int32_t* d_slot_of_widx,
                          uint64_t* s_cnt, double* s_min, double* s_max,
                          double* s_sum, uint64_t* s_first) {
     hipLaunchKernelGGL(k_regroup_t<RG_DIRECT_FOLD>, dim3(NB), dim3(BLOCK), 0, s,
-                       d_meta, d_grec, d_bucket_base, fc, nullptr, nullptr,
-                       nullptr, nullptr, nullptr, nullptr, d_slot_of_widx,
+                       d_grec, d_bucket_base, fc, nullptr, nullptr,
+                       nullptr, nullptr, nullptr, d_slot_of_widx,
                        s_cnt, s_min, s_max, s_sum, s_first);
 }
 
-void launch_regroup_l1(hipStream_t s, const uint32_t* d_meta, const uint4* d_grec,
+void launch_regroup_l1(hipStream_t s, const uint4* d_grec,
                        const uint32_t* d_bucket_base, const FoldChunk& fc,
-                       uint32_t* d_b1offs, uint32_t* d_b1lens, uint32_t* d_meta2,
-                       uint4* d_grec2) {
-    hipLaunchKernelGGL(k_regroup_t<RG_L1>, dim3(NB), dim3(BLOCK), 0, s, d_meta,
+                       uint32_t* d_b1offs, uint32_t* d_b1lens, uint4* d_grec2) {
+    hipLaunchKernelGGL(k_regroup_t<RG_L1>, dim3(NB), dim3(BLOCK), 0, s,
                        d_grec, d_bucket_base, fc, nullptr, nullptr, d_b1offs,
-                       d_b1lens, d_meta2, d_grec2, nullptr, nullptr, nullptr,
+                       d_b1lens, d_grec2, nullptr, nullptr, nullptr,
                        nullptr, nullptr, nullptr);
 }
 
-void launch_regroup_l2_fold(hipStream_t s, const uint32_t* d_meta2,
-                            const uint4* d_grec2, const uint32_t* d_bucket_base,
+void launch_regroup_l2_fold(hipStream_t s, const uint4* d_grec2,
+                            const uint32_t* d_bucket_base,
                             const FoldChunk& fc, int nb1,
                             const uint32_t* d_b1offs, const uint32_t* d_b1lens,
                             const int32_t* d_slot_of_widx, uint64_t* s_cnt,
                             double* s_min, double* s_max, double* s_sum,
                             uint64_t* s_first) {
     hipLaunchKernelGGL(k_regroup_t<RG_L2_FOLD>, dim3(NB, nb1), dim3(BLOCK), 0, s,
-                       d_meta2, d_grec2, d_bucket_base, fc, d_b1offs, d_b1lens,
-                       nullptr, nullptr, nullptr, nullptr, d_slot_of_widx,
+                       d_grec2, d_bucket_base, fc, d_b1offs, d_b1lens,
+                       nullptr, nullptr, nullptr, d_slot_of_widx,
                        s_cnt, s_min, s_max, s_sum, s_first);
 }
 

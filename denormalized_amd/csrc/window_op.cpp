@@ -204,12 +204,10 @@ struct dz_window_op {
     uint32_t* d_total[2] = {};
     uint32_t* d_base[2] = {};
     int C_cap = 0;
-    uint32_t* d_meta = nullptr;
-    uint4* d_grec = nullptr;  /* 16 B {val, rowidx} records (scatter output) */
+    uint4* d_grec = nullptr;  /* 16 B {val, rowidx, meta} records (scatter out) */
     uint32_t* d_b1offs = nullptr; /* two-level L1 segment layout [NB][256] */
     uint32_t* d_b1lens = nullptr;
-    uint32_t* d_meta2 = nullptr;  /* two-level intermediate records */
-    uint4* d_grec2 = nullptr;
+    uint4* d_grec2 = nullptr;     /* two-level intermediate records */
     int64_t l2_cap = 0;
     int64_t rec_cap = 0;
     uint64_t* d_scalars[2] = {};
@@ -620,9 +618,9 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_gofs[0]); hipFree(op->d_gofs[1]);
     hipFree(op->d_total[0]); hipFree(op->d_total[1]);
     hipFree(op->d_base[0]); hipFree(op->d_base[1]);
-    hipFree(op->d_meta); hipFree(op->d_grec);
+    hipFree(op->d_grec);
     hipFree(op->d_b1offs); hipFree(op->d_b1lens);
-    hipFree(op->d_meta2); hipFree(op->d_grec2);
+    hipFree(op->d_grec2);
     hipFree(op->d_scalars[0]); hipFree(op->d_scalars[1]);
     hipFree(op->d_slotmap); hipFree(op->d_zero_kid);
     hipFree(op->d_resetlist);
@@ -1359,8 +1357,7 @@ static dz_status ensure_scratch(dz_window_op* op, int C, int64_t nrec) {
     }
     if (nrec > op->rec_cap) {
         quiesce(op);
-        hipFree(op->d_meta); hipFree(op->d_grec);
-        CHK(op, hipMalloc(&op->d_meta, (size_t)nrec * 4));
+        hipFree(op->d_grec);
         CHK(op, hipMalloc(&op->d_grec, (size_t)nrec * 16));
         op->rec_cap = nrec;
     }
@@ -1633,7 +1630,7 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
     int32_t st_rows = (int32_t)std::max<int64_t>(1, dz::ST_RECORDS / expand);
     timed(op, "scatter", (double)n * 24 + (double)nrec_max * 20, [&] {
         dz::launch_scatter(op->stream, d_kid, d_ts, d_vals, d_valbm, n, chunk, C,
-                           st_rows, wp, op->d_gofs[b], op->d_meta, op->d_grec);
+                           st_rows, wp, op->d_gofs[b], op->d_grec);
     });
     int64_t klocs = op->kcap >> dz::LOG_NB;
     int64_t gtot = klocs * nw;
@@ -1649,8 +1646,7 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
             CHK(op, hipMalloc(&op->d_b1lens, (size_t)dz::NB * 256 * 4));
         }
         if (nrec_max > op->l2_cap) {
-            hipFree(op->d_meta2); hipFree(op->d_grec2);
-            CHK(op, hipMalloc(&op->d_meta2, (size_t)nrec_max * 4));
+            hipFree(op->d_grec2);
             CHK(op, hipMalloc(&op->d_grec2, (size_t)nrec_max * 16));
             op->l2_cap = nrec_max;
         }
@@ -1664,13 +1660,12 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
         fc.bin_stride = (int32_t)(nb1 * 256);
         fc.tl_nw = (int32_t)nw;
         timed(op, "regroup", (double)nrec_max * 40, [&] {
-            dz::launch_regroup_l1(op->stream, op->d_meta, op->d_grec,
-                                  op->d_base[b],
-                                  fc, op->d_b1offs, op->d_b1lens, op->d_meta2,
+            dz::launch_regroup_l1(op->stream, op->d_grec, op->d_base[b],
+                                  fc, op->d_b1offs, op->d_b1lens,
                                   op->d_grec2);
         });
         timed(op, "regfold", (double)nrec_max * 24, [&] {
-            dz::launch_regroup_l2_fold(op->stream, op->d_meta2, op->d_grec2,
+            dz::launch_regroup_l2_fold(op->stream, op->d_grec2,
                                        op->d_base[b], fc, (int)nb1,
                                        op->d_b1offs, op->d_b1lens,
                                        op->d_slotmap, op->s_cnt, op->s_min,
@@ -1691,7 +1686,7 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
                 fc.bin_stride = dz::FOLD_GCAP;
                 fc.tl_nw = 0;
                 timed(op, "regfold", (double)nrec_max * 24, [&] {
-                    dz::launch_regroup_fold(op->stream, op->d_meta, op->d_grec,
+                    dz::launch_regroup_fold(op->stream, op->d_grec,
                                             op->d_base[b], fc, op->d_slotmap,
                                             op->s_cnt, op->s_min, op->s_max,
                                             op->s_sum, op->s_first);

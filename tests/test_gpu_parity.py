@@ -356,7 +356,7 @@ def test_push_device_dense_parity_1M():
     op.finish()
     outs = op.poll_all()
     stats = op.kernel_stats()
-    assert stats["fold"]["launches"] >= 2 and stats["scatter"]["total_ms"] > 0
+    assert stats["regfold"]["launches"] >= 2 and stats["scatter"]["total_ms"] > 0
 
     ts, kid, val = pyoracle.gen(99, 5_000_000, 0, n, nkeys, 1000)
     o = pyoracle.Oracle(1000, 0)
