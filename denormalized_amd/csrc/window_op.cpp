@@ -912,18 +912,36 @@ static void emit_drain(dz_window_op* op) {
 
 /* hipEventSynchronize pays ~0.5-1 ms blocked-wakeup latency per call; the
  * emission workers poll instead (hipEventQuery + yield): the events complete
- * in tens of microseconds once the copy stream reaches them. */
+ * in tens of microseconds once the copy stream reaches them.
+ * TERMINAL errors (a device fault poisons every query with e.g.
+ * hipErrorIllegalAddress) must break the loop — spinning on them forever
+ * turns one bad kernel into a process hang; the next CHK'd hip call after
+ * the spin reports the sticky error. */
 static void event_spin(hipEvent_t ev) {
-    while (hipEventQuery(ev) != hipSuccess)
+    hipError_t e;
+    while ((e = hipEventQuery(ev)) != hipSuccess) {
+        if (e != hipErrorNotReady) {
+            fprintf(stderr, "dz: event_spin aborted: %s\n",
+                    hipGetErrorString(e));
+            return;
+        }
         std::this_thread::yield();
+    }
 }
 
 /* worker-side variant: a tight hipEventQuery loop from several worker
  * threads contends the runtime lock against the push thread's own queries
  * and launches; the workers can afford 20 µs of extra latency */
 static void event_spin_relaxed(hipEvent_t ev) {
-    while (hipEventQuery(ev) != hipSuccess)
+    hipError_t e;
+    while ((e = hipEventQuery(ev)) != hipSuccess) {
+        if (e != hipErrorNotReady) {
+            fprintf(stderr, "dz: event_spin aborted: %s\n",
+                    hipGetErrorString(e));
+            return;
+        }
         std::this_thread::sleep_for(std::chrono::microseconds(20));
+    }
 }
 
 static void emit_worker_main(dz_window_op* op) {
