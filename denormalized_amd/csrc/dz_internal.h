@@ -67,7 +67,8 @@ void launch_scan(hipStream_t stream, const uint32_t* d_ghist, int C,
 void launch_scatter(hipStream_t stream, const int32_t* d_kid, const int64_t* d_ts,
                     const double* d_vals, const uint8_t* d_validity /*bitmap|null*/,
                     int64_t n, int64_t chunk, int C, int32_t st_rows,
-                    const WinParams& wp, const uint32_t* d_gofs, uint4* d_grec);
+                    const WinParams& wp, const uint32_t* d_gofs, uint4* d_grec,
+                    uint32_t rec_limit, uint32_t* d_dbg /*bounds-guard cells*/);
 
 constexpr int FOLD_GCAP = 256; /* groups per bucket per fold chunk */
 
@@ -78,7 +79,8 @@ void launch_regroup_fold(hipStream_t stream, const uint4* d_grec,
                          const uint32_t* d_bucket_base,
                          const FoldChunk& fc, const int32_t* d_slot_of_widx,
                          uint64_t* s_cnt, double* s_min, double* s_max,
-                         double* s_sum, uint64_t* s_first);
+                         double* s_sum, uint64_t* s_first, int64_t slab_cells,
+                         uint32_t* d_dbg);
 
 void launch_regroup_l1(hipStream_t stream, const uint4* d_grec,
                        const uint32_t* d_bucket_base,
@@ -91,7 +93,8 @@ void launch_regroup_l2_fold(hipStream_t stream, const uint4* d_grec2,
                             const uint32_t* d_b1offs, const uint32_t* d_b1lens,
                             const int32_t* d_slot_of_widx, uint64_t* s_cnt,
                             double* s_min, double* s_max, double* s_sum,
-                            uint64_t* s_first);
+                            uint64_t* s_first, int64_t slab_cells,
+                            uint32_t* d_dbg);
 
 struct EmitFilter {
     int32_t on;      /* 0 = no filter */

@@ -305,7 +305,8 @@ void launch_scan(hipStream_t s, const uint32_t* d_ghist, int C,
 __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
         const int64_t* ts, const double* vals, const uint8_t* validity,
         int64_t n, int64_t chunk, int32_t st_rows, WinParams wp,
-        const uint32_t* gofs, uint4* grec) {
+        const uint32_t* gofs, uint4* grec, uint32_t rec_limit,
+        uint32_t* dbg) {
     /* LDS-staged stable partition. Each wave owns a CONTIGUOUS QUARTER of the
      * supertile (wave order == row order), so staging cursors are per-wave
      * private: no cross-wave serialization, ~4 block barriers per supertile.
@@ -483,7 +484,11 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
                 wb = (Eb > cb) ? (Eb - cb) : 0u;
             }
             const uint32_t jp = rb + (p - offs[b]); /* pending index */
-            if (jp < wb) grec[cb + jp] = s_rec[p];
+            if (jp < wb) {
+                const uint32_t d = cb + jp;
+                if (d < rec_limit) grec[d] = s_rec[p];
+                else dbg[0] = 1; /* bounds guard: flag, never corrupt */
+            }
         }
         __syncthreads();
         /* residual head writes + carry update (thread-owned per bucket) */
@@ -497,8 +502,10 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
                 const uint32_t Eb = (cb + Tb) & ~3u;
                 wb = (Eb > cb) ? (Eb - cb) : 0u;
             }
-            for (uint32_t j = 0; j < rb && j < wb; j++)
-                grec[cb + j] = resid[t][j];
+            for (uint32_t j = 0; j < rb && j < wb; j++) {
+                if (cb + j < rec_limit) grec[cb + j] = resid[t][j];
+                else dbg[0] = 2;
+            }
             const uint32_t nr = Tb - wb; /* <= 3 */
             uint4 tmp[3];
             for (uint32_t j = 0; j < nr; j++) {
@@ -516,9 +523,11 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
 void launch_scatter(hipStream_t s, const int32_t* d_kid, const int64_t* d_ts,
                     const double* d_vals, const uint8_t* d_validity, int64_t n,
                     int64_t chunk, int C, int32_t st_rows, const WinParams& wp,
-                    const uint32_t* d_gofs, uint4* d_grec) {
+                    const uint32_t* d_gofs, uint4* d_grec, uint32_t rec_limit,
+                    uint32_t* d_dbg) {
     hipLaunchKernelGGL(k_scatter, dim3(C), dim3(BLOCK), 0, s, d_kid, d_ts, d_vals,
-                       d_validity, n, chunk, st_rows, wp, d_gofs, d_grec);
+                       d_validity, n, chunk, st_rows, wp, d_gofs, d_grec,
+                       rec_limit, d_dbg);
 }
 
 /* ------------------------------------------------------------------ */
@@ -559,7 +568,8 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint4* rrec,
         const uint32_t* b1offs, const uint32_t* b1lens, uint32_t* binoffs,
         uint32_t* binlens, uint4* orec,
         const int32_t* slot_of_widx, uint64_t* s_cnt, double* s_min,
-        double* s_max, double* s_sum, uint64_t* s_first) {
+        double* s_max, double* s_sum, uint64_t* s_first, int64_t slab_cells,
+        uint32_t* dbg) {
     constexpr bool FOLD = (MODE == RG_DIRECT_FOLD || MODE == RG_L2_FOLD);
     __shared__ uint32_t cnt[GCAP];    /* whole-segment bin counts (L1) */
     __shared__ uint32_t gcur[GCAP];   /* segment-region bin cursors (L1) */
@@ -626,6 +636,12 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint4* rrec,
         if (f_own) {
             int64_t slot = slot_of_widx[my_widx];
             f_sidx = slot * (5 * fc.kcap) + (((int64_t)my_kloc << LOG_NB) | bkt);
+            if (slot < 0 || f_sidx < 0 || f_sidx + 4 * fc.kcap >= slab_cells) {
+                dbg[1] = 1; /* bounds guard: disown, never corrupt */
+                f_own = false;
+            }
+        }
+        if (f_own) {
             f_cnt = s_cnt[f_sidx];
             f_fst = s_first[f_sidx];
             if (f_cnt > 0) {
@@ -787,11 +803,12 @@ void launch_regroup_fold(hipStream_t s, const uint4* d_grec,
                          const uint32_t* d_bucket_base,
                          const FoldChunk& fc, const int32_t* d_slot_of_widx,
                          uint64_t* s_cnt, double* s_min, double* s_max,
-                         double* s_sum, uint64_t* s_first) {
+                         double* s_sum, uint64_t* s_first, int64_t slab_cells,
+                         uint32_t* d_dbg) {
     hipLaunchKernelGGL(k_regroup_t<RG_DIRECT_FOLD>, dim3(NB), dim3(BLOCK), 0, s,
                        d_grec, d_bucket_base, fc, nullptr, nullptr,
                        nullptr, nullptr, nullptr, d_slot_of_widx,
-                       s_cnt, s_min, s_max, s_sum, s_first);
+                       s_cnt, s_min, s_max, s_sum, s_first, slab_cells, d_dbg);
 }
 
 void launch_regroup_l1(hipStream_t s, const uint4* d_grec,
@@ -800,7 +817,7 @@ void launch_regroup_l1(hipStream_t s, const uint4* d_grec,
     hipLaunchKernelGGL(k_regroup_t<RG_L1>, dim3(NB), dim3(BLOCK), 0, s,
                        d_grec, d_bucket_base, fc, nullptr, nullptr, d_b1offs,
                        d_b1lens, d_grec2, nullptr, nullptr, nullptr,
-                       nullptr, nullptr, nullptr);
+                       nullptr, nullptr, nullptr, 0, nullptr);
 }
 
 void launch_regroup_l2_fold(hipStream_t s, const uint4* d_grec2,
@@ -809,11 +826,12 @@ void launch_regroup_l2_fold(hipStream_t s, const uint4* d_grec2,
                             const uint32_t* d_b1offs, const uint32_t* d_b1lens,
                             const int32_t* d_slot_of_widx, uint64_t* s_cnt,
                             double* s_min, double* s_max, double* s_sum,
-                            uint64_t* s_first) {
+                            uint64_t* s_first, int64_t slab_cells,
+                            uint32_t* d_dbg) {
     hipLaunchKernelGGL(k_regroup_t<RG_L2_FOLD>, dim3(NB, nb1), dim3(BLOCK), 0, s,
                        d_grec2, d_bucket_base, fc, d_b1offs, d_b1lens,
                        nullptr, nullptr, nullptr, d_slot_of_widx,
-                       s_cnt, s_min, s_max, s_sum, s_first);
+                       s_cnt, s_min, s_max, s_sum, s_first, slab_cells, d_dbg);
 }
 
 /* ------------------------------------------------------------------ */

@@ -220,6 +220,8 @@ struct dz_window_op {
     int64_t zero_cap = 0;
     int32_t* d_resetlist = nullptr; /* MAX_RANGES, allocated at create */
     int32_t* h_resetlist[2] = {};
+    uint32_t* d_dbg = nullptr; /* kernel bounds-guard cells (4 u32, zeroed) */
+    bool dbg_checked_err = false;
 
     /* input staging (host-batch path) */
     int64_t* d_ts = nullptr;
@@ -548,6 +550,8 @@ extern "C" dz_window_op* dz_window_op_create(const dz_window_desc* desc) {
     hipHostMalloc((void**)&op->h_scalars, 2 * 3 * 8);
     hipMalloc(&op->d_slotmap, (size_t)dz::MAX_RANGES * 4);
     hipMalloc(&op->d_resetlist, (size_t)dz::MAX_RANGES * 4);
+    hipMalloc(&op->d_dbg, 16);
+    hipMemset(op->d_dbg, 0, 16);
     for (int i = 0; i < 2; i++) {
         hipHostMalloc((void**)&op->h_slotmap[i], (size_t)dz::MAX_RANGES * 4);
         hipHostMalloc((void**)&op->h_resetlist[i], (size_t)dz::MAX_RANGES * 4);
@@ -623,7 +627,7 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_grec2);
     hipFree(op->d_scalars[0]); hipFree(op->d_scalars[1]);
     hipFree(op->d_slotmap); hipFree(op->d_zero_kid);
-    hipFree(op->d_resetlist);
+    hipFree(op->d_resetlist); hipFree(op->d_dbg);
     hipFree(op->d_ts); hipFree(op->d_kid); hipFree(op->d_vals); hipFree(op->d_valbm);
     for (int i = 0; i < 2; i++) {
         hipFree(op->d_sts[i]); hipFree(op->d_skid[i]); hipFree(op->d_svals[i]);
@@ -1648,7 +1652,8 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
     int32_t st_rows = (int32_t)std::max<int64_t>(1, dz::ST_RECORDS / expand);
     timed(op, "scatter", (double)n * 24 + (double)nrec_max * 20, [&] {
         dz::launch_scatter(op->stream, d_kid, d_ts, d_vals, d_valbm, n, chunk, C,
-                           st_rows, wp, op->d_gofs[b], op->d_grec);
+                           st_rows, wp, op->d_gofs[b], op->d_grec,
+                           (uint32_t)op->rec_cap, op->d_dbg);
     });
     int64_t klocs = op->kcap >> dz::LOG_NB;
     int64_t gtot = klocs * nw;
@@ -1687,7 +1692,9 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
                                        op->d_base[b], fc, (int)nb1,
                                        op->d_b1offs, op->d_b1lens,
                                        op->d_slotmap, op->s_cnt, op->s_min,
-                                       op->s_max, op->s_sum, op->s_first);
+                                       op->s_max, op->s_sum, op->s_first,
+                                       (int64_t)op->nslots * 5 * op->kcap,
+                                       op->d_dbg);
         });
     } else {
         for (int64_t k_lo = 0; k_lo < klocs; k_lo += dz::FOLD_GCAP) {
@@ -1707,7 +1714,9 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
                     dz::launch_regroup_fold(op->stream, op->d_grec,
                                             op->d_base[b], fc, op->d_slotmap,
                                             op->s_cnt, op->s_min, op->s_max,
-                                            op->s_sum, op->s_first);
+                                            op->s_sum, op->s_first,
+                                            (int64_t)op->nslots * 5 * op->kcap,
+                                            op->d_dbg);
                 });
             }
         }
@@ -2027,6 +2036,24 @@ extern "C" dz_status dz_window_op_poll(dz_window_op* op, const dz_out_batch** ou
     return DZ_OK;
 }
 
+/* read the kernels' bounds-guard cells; a nonzero cell means a kernel
+ * detected an out-of-range write it refused to perform — surface it as a
+ * hard error (results may be incomplete, never corrupted) */
+static dz_status check_dbg(dz_window_op* op) {
+    uint32_t cells[4] = {0, 0, 0, 0};
+    if (op->d_dbg &&
+        hipMemcpy(cells, op->d_dbg, 16, hipMemcpyDeviceToHost) == hipSuccess) {
+        for (int i = 0; i < 4; i++)
+            if (cells[i]) {
+                op->err = "internal: kernel bounds guard tripped (cell " +
+                          std::to_string(i) + " = " + std::to_string(cells[i]) +
+                          ") — a kernel skipped an out-of-range write";
+                return DZ_ERR;
+            }
+    }
+    return DZ_OK;
+}
+
 extern "C" dz_status dz_window_op_finish(dz_window_op* op) {
     if (!op) return DZ_ERR;
     CHK(op, hipSetDevice(op->device));
@@ -2041,7 +2068,7 @@ extern "C" dz_status dz_window_op_finish(dz_window_op* op) {
     emit_drain(op);
     CHK(op, hipStreamSynchronize(op->stream));
     drain_events(op, true);
-    return DZ_OK;
+    return check_dbg(op);
 }
 
 extern "C" dz_status dz_window_op_drain(dz_window_op* op) {
@@ -2107,6 +2134,7 @@ extern "C" dz_status dz_window_op_kernel_stats(dz_window_op* op,
     emit_drain(op);
     CHK(op, hipStreamSynchronize(op->stream));
     drain_events(op, true);
+    if (check_dbg(op) != DZ_OK) return DZ_ERR;
     {
         KStatAcc& s = op->stats["h_emit_build"];
         s.launches = op->e_builds.load();
