@@ -10,9 +10,9 @@ REPO=${GRAFT_REPO_ROOT:-/root/repo}
 BENCH="python $REPO/bench.py --steps 12 --warmup 3 --skip-cpu-baseline"
 OUT=$REPO/gpurun_out
 rm -rf /tmp/p1 /tmp/p2 /tmp/p3
-rocprofv3 --kernel-trace --stats -d /tmp/p1 -o ks -- $BENCH >/dev/null 2>&1 || true
-rocprofv3 --pmc FETCH_SIZE -d /tmp/p2 -o f -- $BENCH >/dev/null 2>&1 || true
-rocprofv3 --pmc WRITE_SIZE -d /tmp/p3 -o w -- $BENCH >/dev/null 2>&1 || true
+rocprofv3 --kernel-trace --stats --output-format csv -d /tmp/p1 -o ks -- $BENCH || true
+rocprofv3 --pmc FETCH_SIZE --output-format csv -d /tmp/p2 -o f -- $BENCH || true
+rocprofv3 --pmc WRITE_SIZE --output-format csv -d /tmp/p3 -o w -- $BENCH || true
 cp /tmp/p1/*kernel_stats.csv $OUT/r02_kernel_stats.csv 2>/dev/null || true
 python3 - <<'EOF'
 import csv, glob, json, os
