@@ -312,12 +312,13 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
      * private: no cross-wave serialization, ~4 block barriers per supertile.
      * Records are placed bucket-major in LDS and flushed so adjacent lanes
      * write adjacent global addresses (the direct form measured 6x write
-     * amplification — profiles/hbm_traffic.json). The flush additionally
-     * CARRIES the <=3 records that would leave a bucket's cursor
-     * mid-64B-sector over to the next supertile's run (resid/r_res), so a
-     * (chunk,bucket) pays at most one partial head + one partial tail
-     * sector instead of one partial sector per ~64 B run (the unaligned
-     * form measured 1.95x write amplification). */
+     * amplification — profiles/hbm_traffic.json). A sector-aligned carrying
+     * variant (hold back <=3 records per bucket to end runs on 64 B
+     * boundaries) was measured in round 2 and REMOVED: whole-pipeline HBM
+     * traffic is ~1 TB/s against the ~6.3 TB/s achievable — the pipeline is
+     * latency-bound, and the carrying's extra flush work cost more kernel
+     * time than the write-sector savings returned (profiles/r02_pmc.json:
+     * the isolated residual-head stores also re-opened sectors). */
     __shared__ uint32_t cur[NB];    /* global cursors for this block's chunk */
     __shared__ uint32_t cnt4[WAVES_PER_BLOCK][NB]; /* per-wave-quarter counts,
                                      * converted IN PLACE to per-wave staging
@@ -329,14 +330,10 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
      * their destination is re-derived at flush by a 9-step binary search
      * over offs[] — no per-record dest/meta staging arrays */
     __shared__ uint4 s_rec[ST_RECORDS];
-    __shared__ uint4 resid[NB][3];  /* sector-carry records per bucket */
-    __shared__ uint32_t r_res[NB];
     __shared__ uint32_t s_total;
 
-    for (int t = threadIdx.x; t < NB; t += BLOCK) {
+    for (int t = threadIdx.x; t < NB; t += BLOCK)
         cur[t] = gofs[(int64_t)blockIdx.x * NB + t];
-        r_res[t] = 0;
-    }
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
     const int64_t lo = blockIdx.x * chunk;
@@ -465,56 +462,21 @@ __global__ __launch_bounds__(BLOCK) void k_scatter(const int32_t* kid,
         __syncthreads();
         /* flush: bucket-major staging => coalesced run writes; the record's
          * bin (and so its destination) falls out of a binary search over
-         * the bin prefix. Pending order per bucket = carried residuals
-         * (older rows) first, then this supertile's staged run; only the
-         * sector-aligned prefix of the pending sequence is written unless
-         * this is the chunk's final supertile. */
-        const bool final_st = (st1 == hi);
+         * the bin prefix */
         const uint32_t tot = s_total;
         for (uint32_t p = threadIdx.x; p < tot; p += BLOCK) {
             uint32_t b = 0;
             for (int stp = NB >> 1; stp; stp >>= 1)
                 if (b + stp < NB && offs[b + stp] <= p) b += stp;
-            const uint32_t nb_ = ((b + 1 < NB) ? offs[b + 1] : tot) - offs[b];
-            const uint32_t rb = r_res[b];
-            const uint32_t cb = cur[b];
-            uint32_t wb = rb + nb_; /* final: flush all pending */
-            if (!final_st) {
-                const uint32_t Eb = (cb + wb) & ~3u;
-                wb = (Eb > cb) ? (Eb - cb) : 0u;
-            }
-            const uint32_t jp = rb + (p - offs[b]); /* pending index */
-            if (jp < wb) {
-                const uint32_t d = cb + jp;
-                if (d < rec_limit) grec[d] = s_rec[p];
-                else dbg[0] = 1; /* bounds guard: flag, never corrupt */
-            }
+            const uint32_t d = cur[b] + (p - offs[b]);
+            if (d < rec_limit) grec[d] = s_rec[p];
+            else dbg[0] = 1; /* bounds guard: flag, never corrupt */
         }
         __syncthreads();
-        /* residual head writes + carry update (thread-owned per bucket) */
         for (int t = threadIdx.x; t < NB; t += BLOCK) {
-            const uint32_t nb_ = ((t + 1 < NB) ? offs[t + 1] : tot) - offs[t];
-            const uint32_t rb = r_res[t];
-            const uint32_t Tb = rb + nb_;
-            const uint32_t cb = cur[t];
-            uint32_t wb = Tb;
-            if (!final_st) {
-                const uint32_t Eb = (cb + Tb) & ~3u;
-                wb = (Eb > cb) ? (Eb - cb) : 0u;
-            }
-            for (uint32_t j = 0; j < rb && j < wb; j++) {
-                if (cb + j < rec_limit) grec[cb + j] = resid[t][j];
-                else dbg[0] = 2;
-            }
-            const uint32_t nr = Tb - wb; /* <= 3 */
-            uint4 tmp[3];
-            for (uint32_t j = 0; j < nr; j++) {
-                const uint32_t i = wb + j;
-                tmp[j] = (i < rb) ? resid[t][i] : s_rec[offs[t] + (i - rb)];
-            }
-            for (uint32_t j = 0; j < nr; j++) resid[t][j] = tmp[j];
-            r_res[t] = nr;
-            cur[t] = cb + wb;
+            /* bin total = offs delta (cnt4 was overlaid by the cursors) */
+            uint32_t nxt = (t + 1 < NB) ? offs[t + 1] : s_total;
+            cur[t] += nxt - offs[t];
         }
         __syncthreads();
     }
@@ -578,7 +540,9 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint4* rrec,
     __shared__ uint32_t wcur[WAVES_PER_BLOCK][GCAP];   /* per-wave cursors */
     __shared__ uint32_t s_dest[ST_RECORDS]; /* RG_L1 only */
     __shared__ uint4 s_rec[ST_RECORDS];     /* meta travels in .w */
+    __shared__ uint32_t scanbuf[BLOCK];
     __shared__ uint32_t s_total;
+    static_assert(GCAP == BLOCK, "parallel bin prefix maps one thread per bin");
 
     const int bkt = blockIdx.x;
     const int lane = threadIdx.x & 63;
@@ -694,20 +658,26 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint4* rrec,
             if (g != 0x1FFu) atomicAdd(&stcnt4[wave][g], 1u);
         }
         __syncthreads();
-        if (threadIdx.x == 0) { /* bin prefix over <=GCAP bins */
-            uint32_t run = 0;
-            for (int g = 0; g < GCAP; g++) {
-                stoffs[g] = run;
-                for (int w = 0; w < WAVES_PER_BLOCK; w++) run += stcnt4[w][g];
+        {   /* parallel bin prefix (GCAP == BLOCK: one thread per bin) — the
+             * serial thread-0 form cost ~1k lone LDS reads per supertile */
+            uint32_t tg = 0;
+            for (int w = 0; w < WAVES_PER_BLOCK; w++)
+                tg += stcnt4[w][threadIdx.x];
+            scanbuf[threadIdx.x] = tg;
+            __syncthreads();
+            for (int o = 1; o < BLOCK; o <<= 1) {
+                uint32_t v =
+                    (threadIdx.x >= (unsigned)o) ? scanbuf[threadIdx.x - o] : 0;
+                __syncthreads();
+                scanbuf[threadIdx.x] += v;
+                __syncthreads();
             }
-            s_total = run;
-        }
-        __syncthreads();
-        for (int g = threadIdx.x; g < GCAP; g += BLOCK) { /* per-wave bases */
-            uint32_t run = stoffs[g];
-            for (int w = 0; w < WAVES_PER_BLOCK; w++) {
-                wcur[w][g] = run;
-                run += stcnt4[w][g];
+            uint32_t run = threadIdx.x ? scanbuf[threadIdx.x - 1] : 0;
+            stoffs[threadIdx.x] = run;
+            if (threadIdx.x == BLOCK - 1) s_total = scanbuf[threadIdx.x];
+            for (int w = 0; w < WAVES_PER_BLOCK; w++) { /* per-wave bases */
+                wcur[w][threadIdx.x] = run;
+                run += stcnt4[w][threadIdx.x];
             }
         }
         __syncthreads();
