@@ -49,6 +49,10 @@ def parse_args():
     p.add_argument("--window-ms", type=int, default=1000)
     p.add_argument("--slide-ms", type=int, default=0)
     p.add_argument("--seed", type=int, default=42)
+    p.add_argument("--key-kind", choices=["dense", "utf8"], default="dense",
+                   help="utf8 = raw 'sensor_{k}' string keys pushed to the op "
+                        "and interned ON DEVICE inside the timed region "
+                        "(cfg2 as written); dense = pre-densified int ids")
     p.add_argument("--no-filter", action="store_true")
     p.add_argument("--dist-backend", default=None,
                    help="torch.distributed backend override (default: nccl on GPU)")
@@ -153,24 +157,71 @@ def main():
     # pre-generate the whole stream into HBM (inputs resident when the timed
     # region starts). Each rank owns a disjoint key shard (weak scaling):
     # same time distribution, rank-distinct keys (ids are rank-local).
+    utf8 = args.key_kind == "utf8"
+    gseed = args.seed + 1000003 * rank  # rank-distinct draws per shard
     d_ts = dz.DeviceArray(device, total_rows * 8)
-    d_kid = dz.DeviceArray(device, total_rows * 4)
+    d_kid = None if utf8 else dz.DeviceArray(device, total_rows * 4)
     d_vals = dz.DeviceArray(device, total_rows * 8)
-    # rank-distinct seed => independent key/value draws per shard
-    dz.generate(device, args.seed + 1000003 * rank, 1_000_000, 0, total_rows,
-                args.keys, args.rows_per_ms, d_ts.ptr, None, d_kid.ptr,
-                d_vals.ptr)
+    dz.generate(device, gseed, 1_000_000, 0, total_rows, args.keys,
+                args.rows_per_ms, d_ts.ptr, None,
+                d_kid.ptr if d_kid else None, d_vals.ptr)
+    d_offs = d_data = None
+    step_data_base = []
+    if utf8:
+        # utf8 key column, per-step Arrow slices: lens on device, host cumsum
+        # into per-step int32 offsets, then the device fill pass. All
+        # pre-generation — the timed region starts with everything resident.
+        d_lens = dz.DeviceArray(device, B * 4)
+        d_offs = dz.DeviceArray(device, (K + W) * (B + 1) * 4)
+        host_offs = []
+        total_bytes = 0
+        for s in range(K + W):
+            dz.generate_utf8(device, gseed, s * B, B, args.keys,
+                             d_lens=d_lens.ptr)
+            lens = d_lens.to_host(np.int32, B)
+            o = np.zeros(B + 1, np.int32)
+            np.cumsum(lens, out=o[1:])
+            host_offs.append(o)
+            step_data_base.append(total_bytes)
+            total_bytes += int(o[-1])
+        d_data = dz.DeviceArray(device, total_bytes)
+        L = _lib.lib()
+        for s in range(K + W):
+            optr = ctypes.c_void_p(d_offs.ptr.value + s * (B + 1) * 4)
+            L.dz_memcpy_h2d(optr, host_offs[s].ctypes.data_as(ctypes.c_void_p),
+                            (B + 1) * 4)
+            dz.generate_utf8(device, gseed, s * B, B, args.keys,
+                             d_offsets=optr,
+                             d_key_data=ctypes.c_void_p(
+                                 d_data.ptr.value + step_data_base[s]))
+        d_lens.free()
+        del host_offs
     dz.synchronize(device)
 
     op = dz.WindowOp(length_ms=args.window_ms, slide_ms=args.slide_ms,
                      aggs=[("count", 0), ("min", 0), ("max", 0), ("avg", 0)],
-                     key_kind=_lib.KEY_DENSE_INT64, n_keys_hint=args.keys,
-                     device=device)
+                     key_kind=_lib.KEY_UTF8 if utf8 else _lib.KEY_DENSE_INT64,
+                     n_keys_hint=args.keys, device=device)
     if not args.no_filter:
         op.set_filter("max", ">", 113.0)
 
-    def push_step(step):
+    def do_push(step):
         off = step * B
+        if utf8:
+            op.push_device_utf8(
+                B,
+                ctypes.c_void_p(d_ts.ptr.value + off * 8),
+                ctypes.c_void_p(d_offs.ptr.value + step * (B + 1) * 4),
+                ctypes.c_void_p(d_data.ptr.value + step_data_base[step]),
+                ctypes.c_void_p(d_vals.ptr.value + off * 8))
+        else:
+            op.push_device(B,
+                           ctypes.c_void_p(d_ts.ptr.value + off * 8),
+                           ctypes.c_void_p(d_kid.ptr.value + off * 4),
+                           ctypes.c_void_p(d_vals.ptr.value + off * 8),
+                           borrowed=not args.staged_push)
+
+    def push_step(step):
         if dist is not None:
             # exchange the (one-batch-lagged) local watermarks BEFORE the
             # push: advance_watermark flushes the deferred previous batch at
@@ -188,11 +239,7 @@ def main():
         # borrowed (zero-copy) push: the pre-generated stream stays resident
         # and untouched for the whole run, exactly the lifetime the borrowed
         # contract asks for
-        op.push_device(B,
-                       ctypes.c_void_p(d_ts.ptr.value + off * 8),
-                       ctypes.c_void_p(d_kid.ptr.value + off * 4),
-                       ctypes.c_void_p(d_vals.ptr.value + off * 8),
-                       borrowed=not args.staged_push)
+        do_push(step)
         emitted = 0
         # non-blocking zero-copy poll: emission is pipelined on the op's
         # worker pool and overlaps the next step's kernels; consuming the
@@ -218,12 +265,7 @@ def main():
         marks = []
         for s in range(W, W + K):
             a = time.perf_counter()
-            off = s * B
-            op.push_device(B,
-                           ctypes.c_void_p(d_ts.ptr.value + off * 8),
-                           ctypes.c_void_p(d_kid.ptr.value + off * 4),
-                           ctypes.c_void_p(d_vals.ptr.value + off * 8),
-                           borrowed=not args.staged_push)
+            do_push(s)
             m = time.perf_counter()
             for bt in op.poll_iter(copy=False):
                 emitted += bt["n_rows"]
@@ -317,11 +359,15 @@ def main():
             "dtype": "f64",
             "data": "synthetic",
             "config": {
-                "workload": ("cfg2 (dense-int key variant): "
+                "workload": (("cfg2-utf8 (as written, device-interned "
+                              "string keys): " if utf8 else
+                              "cfg2 (dense-int key variant): ")
                              if (args.keys == 10_000 and not args.slide_ms
                                  and args.window_ms == 1000) else "")
                             + f"{K * B / 1e6:.0f}M rows/GPU, "
-                            + f"{args.keys} dense-int keys, "
+                            + (f"{args.keys} utf8 'sensor_{{k}}' keys "
+                               "(per-row device intern in the timed region), "
+                               if utf8 else f"{args.keys} dense-int keys, ")
                             + f"{args.window_ms}ms "
                             + (f"sliding/{args.slide_ms}ms hop" if args.slide_ms
                                else "tumbling")
@@ -350,10 +396,17 @@ def main():
         }
         if not args.skip_cpu_baseline and world == 1:
             out["cpu_baseline"] = cpu_baseline(args)
+            if utf8:
+                out["cpu_baseline"]["sample"] += (
+                    "; NB the CPU leg aggregates pre-densified int keys — "
+                    "per-row string interning is EXCLUDED from the CPU "
+                    "baseline but INCLUDED in the GPU value (conservative "
+                    "for the GPU/CPU ratio)")
         print(json.dumps(out))
     op.close()
-    for a in (d_ts, d_kid, d_vals):
-        a.free()
+    for a in (d_ts, d_kid, d_vals, d_offs, d_data):
+        if a is not None:
+            a.free()
     if dist is not None:
         dist.destroy_process_group()
 

@@ -19,7 +19,8 @@ All per-row compute runs in hand-written CDNA4 HIP kernels behind the C ABI
 (include/denormalized_amd.h); there is no CPU fallback.
 """
 from . import _lib
-from ._lib import WindowOp, DeviceArray, generate, synchronize  # noqa: F401
+from ._lib import (WindowOp, DeviceArray, generate, generate_utf8,  # noqa: F401
+                   synchronize)
 
 __version__ = "0.1"
 

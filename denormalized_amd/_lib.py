@@ -99,6 +99,9 @@ def lib():
             L.dz_window_op_push_device_borrowed.argtypes = [p, i64, p, p, p]
         except AttributeError:  # older engine build (A/B hook)
             pass
+        L.dz_window_op_push_device_utf8.argtypes = [p, i64, p, p, p, p]
+        L.dz_generate_utf8.argtypes = [ctypes.c_int32, ctypes.c_uint64, i64,
+                                       i64, i64, p, p, p]
         L.dz_window_op_poll.argtypes = [p, ctypes.POINTER(ctypes.POINTER(DzOutBatch))]
         L.dz_window_op_finish.argtypes = [p]
         L.dz_window_op_drain.argtypes = [p]
@@ -225,6 +228,14 @@ class WindowOp:
         fn = (self._L.dz_window_op_push_device_borrowed if borrowed
               else self._L.dz_window_op_push_device)
         self._check(fn(self._h, n, d_ts, d_kid32, d_vals), "push_device")
+
+    def push_device_utf8(self, n, d_ts, d_key_offsets, d_key_data, d_vals):
+        """Raw utf8 keys interned ON DEVICE (GroupValues::intern analog).
+        Borrowed semantics: all four buffers stay valid until the next call
+        on the op. Requires key_kind KEY_UTF8."""
+        self._check(self._L.dz_window_op_push_device_utf8(
+            self._h, n, d_ts, d_key_offsets, d_key_data, d_vals),
+            "push_device_utf8")
 
     def poll(self, copy=True):
         """Returns a dict of numpy arrays for one emitted batch, or None.
@@ -389,6 +400,16 @@ def generate(device, seed, t0_ms, start_row, n_rows, n_keys, rows_per_ms,
     if L.dz_generate(device, seed, t0_ms, start_row, n_rows, n_keys, rows_per_ms,
                      d_ts, d_keys, d_kid32, d_vals) != DZ_OK:
         raise RuntimeError(f"dz_generate failed: {_err(L)}")
+
+
+def generate_utf8(device, seed, start_row, n_rows, n_keys, d_lens=None,
+                  d_offsets=None, d_key_data=None):
+    """Synthetic "sensor_{k}" key column: lens pass (host cumsums into
+    offsets) and/or fill pass."""
+    L = lib()
+    if L.dz_generate_utf8(device, seed, start_row, n_rows, n_keys, d_lens,
+                          d_offsets, d_key_data) != DZ_OK:
+        raise RuntimeError(f"dz_generate_utf8 failed: {_err(L)}")
 
 
 def synchronize(device=0):

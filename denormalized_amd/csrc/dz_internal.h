@@ -46,6 +46,18 @@ void launch_gen(hipStream_t stream, uint64_t seed, int64_t t0, int64_t start_row
                 int64_t n, int64_t nkeys, int64_t rows_per_ms,
                 int64_t* d_ts, int64_t* d_keys, int32_t* d_kid, double* d_vals);
 
+/* device utf8 intern (GroupValues::intern analog at device rate) */
+void launch_intern(hipStream_t stream, const int32_t* d_offs, const char* d_data,
+                   int64_t n, uint64_t* tab_fp, uint32_t* tab_id,
+                   uint32_t p_mask, uint32_t* id_off, uint32_t* id_len,
+                   char* pool, uint32_t* ctrs, uint32_t id_cap,
+                   uint32_t pool_cap, int32_t* out_kid, uint32_t* d_dbg);
+
+/* synthetic utf8 key generator ("sensor_{k}"): lens pass and/or fill pass */
+void launch_gen_utf8(hipStream_t stream, uint64_t seed, int64_t start_row,
+                     int64_t n, int64_t nkeys, int32_t* d_lens,
+                     const int32_t* d_offs, char* d_data);
+
 /* d_scalars: [0]=min ts (order-mapped u64), [1]=max ts, [2]=max kid (u64).
  * Host must pre-set [0]=UINT64_MAX, [1]=0, [2]=0. */
 void launch_minmax(hipStream_t stream, const int64_t* d_ts, const int32_t* d_kid,

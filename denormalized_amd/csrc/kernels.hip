@@ -107,6 +107,176 @@ void launch_gen(hipStream_t s, uint64_t seed, int64_t t0, int64_t start_row,
 }
 
 /* ------------------------------------------------------------------ */
+/* device utf8 intern — GroupValues::intern for string keys            */
+/* (grouped_window_agg_stream.rs:512) AT DEVICE RATE: open-address     */
+/* claim on a 64-bit FNV fingerprint, byte-verified against a device   */
+/* string pool, dense ids from a global counter. Id VALUES are         */
+/* schedule-dependent and never surface: bucketing only needs density, */
+/* per-group row order (and with it bit-exactness) is preserved by the */
+/* partition pipeline, and emission orders groups by first-seen row    */
+/* and emits the original bytes from the pool.                         */
+/* ------------------------------------------------------------------ */
+
+__device__ __forceinline__ uint64_t fnv1a64(const char* p, int32_t len) {
+    uint64_t h = 1469598103934665603ULL;
+    for (int32_t i = 0; i < len; i++) {
+        h ^= (uint8_t)p[i];
+        h *= 1099511628211ULL;
+    }
+    return h;
+}
+
+__global__ __launch_bounds__(BLOCK) void k_intern(const int32_t* offs,
+        const char* data, int64_t n, uint64_t* tab_fp, uint32_t* tab_id,
+        uint32_t p_mask, uint32_t* id_off, uint32_t* id_len, char* pool,
+        uint32_t* ctrs /* [0]=next id, [1]=pool cursor */, uint32_t id_cap,
+        uint32_t pool_cap, int32_t* out_kid, uint32_t* dbg) {
+    const int lane = threadIdx.x & 63;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x - lane + lane;
+         i - lane < n; i += stride) {
+        const bool act = i < n;
+        int32_t o0 = 0, len = 0;
+        uint64_t fp = 0;
+        if (act) {
+            o0 = offs[i];
+            len = offs[i + 1] - o0;
+            fp = fnv1a64(data + o0, len);
+            if (!fp) fp = 1; /* 0 marks an empty table slot */
+        }
+        /* wave dedupe by full 64-bit fingerprint: only LEADERS probe, so
+         * a claim's publisher can never share a wave with its waiter
+         * (intra-wave spin-wait on a divergent branch would deadlock) */
+        uint64_t same = ~0ULL;
+        for (int b = 0; b < 64; b++) {
+            uint64_t bb = __ballot((fp >> b) & 1);
+            same &= ((fp >> b) & 1) ? bb : ~bb;
+        }
+        {
+            uint64_t bb = __ballot(act);
+            same &= act ? bb : ~bb;
+        }
+        const int leader = __ffsll((unsigned long long)same) - 1;
+        int32_t id = -1;
+        if (act && lane == leader) {
+            uint32_t slot = (uint32_t)fp & p_mask;
+            for (long spin_guard = 0;; slot = (slot + 1) & p_mask) {
+                uint64_t got = (uint64_t)atomicCAS(
+                    (unsigned long long*)&tab_fp[slot], 0ULL,
+                    (unsigned long long)fp);
+                if (got == 0) { /* claimed: allocate id + pool bytes */
+                    uint32_t nid = atomicAdd(&ctrs[0], 1u);
+                    uint32_t po = atomicAdd(&ctrs[1], (uint32_t)len);
+                    if (nid >= id_cap || po + (uint32_t)len > pool_cap) {
+                        dbg[3] = nid >= id_cap ? 1 : 2; /* capacity guard */
+                        id = 0;
+                        break;
+                    }
+                    for (int32_t j = 0; j < len; j++)
+                        pool[po + j] = data[o0 + j];
+                    id_off[nid] = po;
+                    id_len[nid] = (uint32_t)len;
+                    __threadfence();
+                    atomicExch(&tab_id[slot], nid); /* publish */
+                    id = (int32_t)nid;
+                    break;
+                }
+                if (got == fp) { /* candidate: wait for publish, verify */
+                    uint32_t cand;
+                    while ((cand = atomicAdd(&tab_id[slot], 0u)) == ~0u) {
+                        if (++spin_guard > (64LL << 20)) {
+                            dbg[3] = 3; /* fp64-collision deadlock guard */
+                            break;
+                        }
+                    }
+                    if (cand == ~0u) { id = 0; break; }
+                    bool eq = id_len[cand] == (uint32_t)len;
+                    const uint32_t co = id_off[cand];
+                    for (int32_t j = 0; eq && j < len; j++)
+                        eq = pool[co + j] == data[o0 + j];
+                    if (eq) {
+                        id = (int32_t)cand;
+                        break;
+                    }
+                    /* full-fp collision between different keys: probe on */
+                }
+                if (++spin_guard > (64LL << 20)) {
+                    dbg[3] = 4; /* table full (probed every slot) */
+                    id = 0;
+                    break;
+                }
+            }
+        }
+        id = __shfl(id, leader >= 0 ? leader : 0);
+        if (act) out_kid[i] = id;
+    }
+}
+
+void launch_intern(hipStream_t s, const int32_t* d_offs, const char* d_data,
+                   int64_t n, uint64_t* tab_fp, uint32_t* tab_id,
+                   uint32_t p_mask, uint32_t* id_off, uint32_t* id_len,
+                   char* pool, uint32_t* ctrs, uint32_t id_cap,
+                   uint32_t pool_cap, int32_t* out_kid, uint32_t* dbg) {
+    int blocks = (int)std::min<int64_t>((n + BLOCK - 1) / BLOCK, 2048);
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(k_intern, dim3(blocks), dim3(BLOCK), 0, s, d_offs,
+                       d_data, n, tab_fp, tab_id, p_mask, id_off, id_len, pool,
+                       ctrs, id_cap, pool_cap, out_kid, dbg);
+}
+
+/* synthetic utf8 key generator: "sensor_{k}" with k from the SAME splitmix
+ * draw as the dense generator (spec in DESIGN.md §Generator) — lengths
+ * first (host cumsums them into offsets), then the bytes */
+__device__ __forceinline__ int32_t dec_digits(uint64_t k) {
+    int32_t d = 1;
+    while (k >= 10) { k /= 10; d++; }
+    return d;
+}
+
+__global__ void k_gen_keylens(uint64_t seed, int64_t start_row, int64_t n,
+                              int64_t nkeys, int32_t* lens) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        uint64_t r = splitmix64(seed ^ (0x9e3779b97f4a7c15ULL *
+                                        (uint64_t)(start_row + i + 1)));
+        lens[i] = 7 + dec_digits(r % (uint64_t)nkeys);
+    }
+}
+
+__global__ void k_gen_keyfill(uint64_t seed, int64_t start_row, int64_t n,
+                              int64_t nkeys, const int32_t* offs, char* data) {
+    const char* pfx = "sensor_";
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        uint64_t r = splitmix64(seed ^ (0x9e3779b97f4a7c15ULL *
+                                        (uint64_t)(start_row + i + 1)));
+        uint64_t k = r % (uint64_t)nkeys;
+        char* p = data + offs[i];
+        for (int j = 0; j < 7; j++) p[j] = pfx[j];
+        int32_t len = offs[i + 1] - offs[i];
+        for (int32_t j = len - 1; j >= 7; j--) {
+            p[j] = (char)('0' + (k % 10));
+            k /= 10;
+        }
+    }
+}
+
+void launch_gen_utf8(hipStream_t s, uint64_t seed, int64_t start_row, int64_t n,
+                     int64_t nkeys, int32_t* d_lens, const int32_t* d_offs,
+                     char* d_data) {
+    int blocks = (int)std::min<int64_t>((n + BLOCK - 1) / BLOCK, 2048);
+    if (blocks < 1) blocks = 1;
+    if (d_lens)
+        hipLaunchKernelGGL(k_gen_keylens, dim3(blocks), dim3(BLOCK), 0, s, seed,
+                           start_row, n, nkeys, d_lens);
+    if (d_data)
+        hipLaunchKernelGGL(k_gen_keyfill, dim3(blocks), dim3(BLOCK), 0, s, seed,
+                           start_row, n, nkeys, d_offs, d_data);
+}
+
+/* ------------------------------------------------------------------ */
 /* batch min/max ts + max kid                                          */
 /* ------------------------------------------------------------------ */
 

@@ -223,6 +223,20 @@ struct dz_window_op {
     uint32_t* d_dbg = nullptr; /* kernel bounds-guard cells (4 u32, zeroed) */
     bool dbg_checked_err = false;
 
+    /* device utf8 intern (GroupValues::intern at device rate). Fixed-capacity
+     * from n_keys_hint at create; capacity overflow flags d_dbg[3]. */
+    uint64_t* d_itab_fp = nullptr;  /* open-address fingerprint table */
+    uint32_t* d_itab_id = nullptr;
+    uint32_t i_pmask = 0;
+    uint32_t* d_ioff = nullptr;     /* per-id {pool offset, byte length} */
+    uint32_t* d_ilen = nullptr;
+    char* d_ipool = nullptr;        /* first-seen key bytes */
+    uint32_t* d_ictrs = nullptr;    /* [0]=next id, [1]=pool cursor */
+    uint32_t i_idcap = 0, i_poolcap = 0;
+    int32_t* d_ikid[2] = {};        /* interned dense ids, per pipeline buf */
+    int64_t i_kid_cap[2] = {0, 0};
+    int64_t mirror_keys = 0;        /* ids mirrored into dict_strs so far */
+
     /* input staging (host-batch path) */
     int64_t* d_ts = nullptr;
     int32_t* d_kid = nullptr;
@@ -378,6 +392,7 @@ struct dz_window_op {
 static void emit_worker_main(dz_window_op* op);
 static dz_status ensure_emission(dz_window_op* op);
 static dz_status process_pending(dz_window_op* op);
+static dz_status intern_sync_mirror(dz_window_op* op);
 
 static hipEvent_t get_event(dz_window_op* op) {
     if (!op->ev_pool.empty()) {
@@ -628,6 +643,10 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_scalars[0]); hipFree(op->d_scalars[1]);
     hipFree(op->d_slotmap); hipFree(op->d_zero_kid);
     hipFree(op->d_resetlist); hipFree(op->d_dbg);
+    hipFree(op->d_itab_fp); hipFree(op->d_itab_id);
+    hipFree(op->d_ioff); hipFree(op->d_ilen);
+    hipFree(op->d_ipool); hipFree(op->d_ictrs);
+    hipFree(op->d_ikid[0]); hipFree(op->d_ikid[1]);
     hipFree(op->d_ts); hipFree(op->d_kid); hipFree(op->d_vals); hipFree(op->d_valbm);
     for (int i = 0; i < 2; i++) {
         hipFree(op->d_sts[i]); hipFree(op->d_skid[i]); hipFree(op->d_svals[i]);
@@ -1733,6 +1752,8 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
         op->watermark = mn;
         op->has_wm = true;
     }
+    if (op->d_itab_fp && intern_sync_mirror(op) != DZ_OK)
+        return DZ_ERR; /* newly interned key bytes -> emission dictionary */
     return trigger_windows(op);
 }
 
@@ -1826,6 +1847,132 @@ extern "C" dz_status dz_window_op_push_device_borrowed(
 }
 
 /* ------------------------------------------------------------------ */
+/* device utf8 push: GroupValues::intern (grouped_window_agg_stream.rs:512)
+ * at device rate — open-address fingerprint table + device string pool;
+ * the rest of the pipeline sees dense int32 ids like every other path.  */
+/* ------------------------------------------------------------------ */
+
+static dz_status ensure_intern(dz_window_op* op, int64_t n) {
+    if (!op->d_itab_fp) {
+        uint64_t want = 4 * (uint64_t)std::max<int64_t>(op->kcap, 1);
+        uint32_t P = 1u << 16;
+        while (P < want && P < (1u << 26)) P <<= 1;
+        op->i_pmask = P - 1;
+        op->i_idcap = P / 2;
+        op->i_poolcap = (uint32_t)std::min<uint64_t>((uint64_t)op->i_idcap * 64,
+                                                     1u << 31);
+        CHK(op, hipMalloc(&op->d_itab_fp, (size_t)P * 8));
+        CHK(op, hipMalloc(&op->d_itab_id, (size_t)P * 4));
+        CHK(op, hipMalloc(&op->d_ioff, (size_t)op->i_idcap * 4));
+        CHK(op, hipMalloc(&op->d_ilen, (size_t)op->i_idcap * 4));
+        CHK(op, hipMalloc(&op->d_ipool, op->i_poolcap));
+        CHK(op, hipMalloc(&op->d_ictrs, 8));
+        CHK(op, hipMemsetAsync(op->d_itab_fp, 0, (size_t)P * 8, op->stream));
+        CHK(op, hipMemsetAsync(op->d_itab_id, 0xFF, (size_t)P * 4, op->stream));
+        CHK(op, hipMemsetAsync(op->d_ictrs, 0, 8, op->stream));
+        /* the intern runs on i_stream: make the init visible there */
+        CHK(op, hipStreamSynchronize(op->stream));
+    }
+    int b = op->next_buf;
+    if (n > op->i_kid_cap[b]) {
+        quiesce(op);
+        hipFree(op->d_ikid[b]);
+        CHK(op, hipMalloc(&op->d_ikid[b], (size_t)n * 4));
+        op->i_kid_cap[b] = n;
+    }
+    return DZ_OK;
+}
+
+/* pull newly interned key bytes into the host dictionary mirror so the
+ * emission workers can format them (append-only ChunkedDict: safe alongside
+ * in-flight reads below their snapshots). Runs on the push thread AFTER the
+ * batch's scalars are host-visible, i.e. the intern that allocated these
+ * ids has completed. */
+static dz_status intern_sync_mirror(dz_window_op* op) {
+    if (op->mirror_keys >= op->n_keys) return DZ_OK;
+    int64_t lo = op->mirror_keys, hi = op->n_keys;
+    std::vector<uint32_t> offs(hi - lo), lens(hi - lo);
+    CHK(op, hipMemcpy(offs.data(), op->d_ioff + lo, (size_t)(hi - lo) * 4,
+                      hipMemcpyDeviceToHost));
+    CHK(op, hipMemcpy(lens.data(), op->d_ilen + lo, (size_t)(hi - lo) * 4,
+                      hipMemcpyDeviceToHost));
+    uint32_t pmax = 0;
+    for (size_t i = 0; i < offs.size(); i++)
+        pmax = std::max(pmax, offs[i] + lens[i]);
+    std::vector<char> pool(pmax);
+    if (pmax)
+        CHK(op, hipMemcpy(pool.data(), op->d_ipool, pmax,
+                          hipMemcpyDeviceToHost));
+    for (size_t i = 0; i < offs.size(); i++)
+        op->dict_strs.push_back(std::string(pool.data() + offs[i], lens[i]));
+    op->mirror_keys = hi;
+    return DZ_OK;
+}
+
+extern "C" dz_status dz_window_op_push_device_utf8(dz_window_op* op,
+        int64_t n_rows, const int64_t* d_ts_ms, const int32_t* d_key_offsets,
+        const char* d_key_data, const double* d_vals) {
+    if (!op) return DZ_ERR;
+    if (op->key_kind != DZ_KEY_UTF8 || op->no_group) {
+        op->err = "push_device_utf8 requires key_kind DZ_KEY_UTF8";
+        return DZ_ERR;
+    }
+    if (!op->dict_utf8.empty()) {
+        op->err = "cannot mix host utf8 pushes and device utf8 pushes on one "
+                  "operator (two dictionaries would assign conflicting ids)";
+        return DZ_ERR;
+    }
+    CHK(op, hipSetDevice(op->device));
+    if (n_rows <= 0) return process_pending(op);
+    int C = (int)std::min<int64_t>(512,
+                                   std::max<int64_t>(1, (n_rows + 8191) / 8192));
+    if (op->pend.active &&
+        (C > op->C_cap || (op->slide_ms == 0 && n_rows > op->rec_cap))) {
+        if (process_pending(op) != DZ_OK) return DZ_ERR;
+    }
+    if (ensure_intern(op, n_rows) != DZ_OK) return DZ_ERR;
+    dz_window_op::Pend prev = op->pend;
+    op->pend.active = false;
+    /* intern on the ingest stream ahead of the reduction; the output ring
+     * buffer is gated like the input staging (the scatter two pushes back
+     * was its last reader) */
+    int b = op->next_buf;
+    if (op->consumed_valid[b])
+        CHK(op, hipStreamWaitEvent(op->i_stream, op->ev_consumed[b], 0));
+    timed_on(op, op->i_stream, "intern", (double)n_rows * 18, [&] {
+        dz::launch_intern(op->i_stream, d_key_offsets, d_key_data, n_rows,
+                          op->d_itab_fp, op->d_itab_id, op->i_pmask,
+                          op->d_ioff, op->d_ilen, op->d_ipool, op->d_ictrs,
+                          op->i_idcap, op->i_poolcap, op->d_ikid[b],
+                          op->d_dbg);
+    });
+    if (stage_core(op, n_rows, d_ts_ms, op->d_ikid[b], d_vals, nullptr,
+                   /*keys_are_dense=*/true, /*deferred=*/true,
+                   /*borrow=*/true) != DZ_OK)
+        return DZ_ERR;
+    if (prev.active) return process_batch(op, prev);
+    return DZ_OK;
+}
+
+extern "C" dz_status dz_generate_utf8(int32_t device, uint64_t seed,
+                                      int64_t start_row, int64_t n_rows,
+                                      int64_t n_keys, int32_t* d_lens,
+                                      const int32_t* d_offsets,
+                                      char* d_key_data) {
+    if (hipSetDevice(device) != hipSuccess) {
+        g_err = "hipSetDevice failed";
+        return DZ_ERR;
+    }
+    dz::launch_gen_utf8(nullptr, seed, start_row, n_rows, n_keys, d_lens,
+                        d_offsets, d_key_data);
+    if (hipGetLastError() != hipSuccess) {
+        g_err = "dz_generate_utf8 launch failed";
+        return DZ_ERR;
+    }
+    return DZ_OK;
+}
+
+/* ------------------------------------------------------------------ */
 /* host-batch push: dictionary-encode + stage + H2D                    */
 /* ------------------------------------------------------------------ */
 
@@ -1883,6 +2030,12 @@ extern "C" dz_status dz_window_op_push(dz_window_op* op, const dz_batch* batch) 
         const int32_t* offs = kc.offsets;
         const char* data = (const char*)kc.data;
         if (!offs || !data) { op->err = "utf8 key column needs offsets+data"; return DZ_ERR; }
+        if (op->d_itab_fp) {
+            op->err = "cannot mix host utf8 pushes and device utf8 pushes on "
+                      "one operator (two dictionaries would assign "
+                      "conflicting ids)";
+            return DZ_ERR;
+        }
         for (int64_t i = 0; i < n; i++) {
             std::string s(data + offs[i], data + offs[i + 1]);
             auto it = op->dict_utf8.find(s);

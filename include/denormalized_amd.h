@@ -165,6 +165,20 @@ dz_status dz_window_op_push_device_borrowed(dz_window_op* op, int64_t n_rows,
                                             const int32_t* d_key_ids,
                                             const double* d_vals);
 
+/* Device-resident push with RAW UTF8 KEYS: the per-row string interning the
+ * reference pays inside GroupValues::intern (grouped_window_agg_stream.rs:512)
+ * runs ON DEVICE (open-address fingerprint table + device string pool), so
+ * key materialization is inside the measured path. Borrowed semantics: the
+ * offsets/data/ts/vals buffers stay valid and unmodified until the NEXT call
+ * into the op. Requires key_kind DZ_KEY_UTF8. Key capacity is sized from
+ * n_keys_hint at create (table 4x hint, pool 64 B/key average) and growth
+ * past it fails loudly — size the hint for the stream's cardinality. */
+dz_status dz_window_op_push_device_utf8(dz_window_op* op, int64_t n_rows,
+                                        const int64_t* d_ts_ms,
+                                        const int32_t* d_key_offsets,
+                                        const char* d_key_data,
+                                        const double* d_vals);
+
 /* Retrieve emitted closed windows (the stream's output RecordBatch,
  * trigger_windows :220-253). *out = NULL when nothing is pending.
  * The returned batch stays valid until the next poll/destroy. */
@@ -212,6 +226,15 @@ dz_status dz_generate(int32_t device, uint64_t seed, int64_t t0_ms,
                       int64_t start_row, int64_t n_rows, int64_t n_keys,
                       int64_t rows_per_ms, int64_t* d_ts_ms,
                       int64_t* d_keys, int32_t* d_key_ids, double* d_vals);
+
+/* Synthetic utf8 key column generator ("sensor_{k}", k from the same seeded
+ * draw as dz_generate): pass 1 writes per-row byte lengths to d_lens (host
+ * cumsums them into offsets — int32 offsets bound one batch to 2 GiB of key
+ * bytes); pass 2 fills d_data at d_offsets. Either pointer pair may be NULL
+ * to run one pass. */
+dz_status dz_generate_utf8(int32_t device, uint64_t seed, int64_t start_row,
+                           int64_t n_rows, int64_t n_keys, int32_t* d_lens,
+                           const int32_t* d_offsets, char* d_key_data);
 
 /* Device memory helpers for the bench path (thin wrappers so no HIP types
  * cross the boundary). */

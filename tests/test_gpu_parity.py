@@ -1019,3 +1019,98 @@ def test_window_hop_ratio_envelope_is_loud():
         op.push(np.array([1_000_000], np.int64), np.array([1], np.int64),
                 np.array([1.0]))
     op.close()
+
+
+# ------------------------------------------------- device utf8 intern path
+
+def _utf8_cols(kid, nkeys):
+    names = [f"sensor_{i}".encode() for i in range(nkeys)]
+    data = b"".join(names[k] for k in kid)
+    offs = np.zeros(len(kid) + 1, np.int32)
+    np.cumsum([len(names[k]) for k in kid], out=offs[1:])
+    return offs, np.frombuffer(data, np.uint8)
+
+
+def test_device_utf8_generator_matches_python():
+    # k_gen_keylens/k_gen_keyfill produce "sensor_{k}" for the same splitmix
+    # draw as the dense generator
+    from denormalized_amd import DeviceArray, generate_utf8, synchronize
+    n, nkeys, seed = 4096, 137, 77
+    d_lens = DeviceArray(0, n * 4)
+    generate_utf8(0, seed, 500, n, nkeys, d_lens=d_lens.ptr)
+    synchronize(0)
+    lens = d_lens.to_host(np.int32, n)
+    offs = np.zeros(n + 1, np.int32)
+    np.cumsum(lens, out=offs[1:])
+    d_offs = DeviceArray(0, (n + 1) * 4)
+    d_offs.from_host(offs)
+    d_data = DeviceArray(0, int(offs[-1]))
+    generate_utf8(0, seed, 500, n, nkeys, d_offsets=d_offs.ptr,
+                  d_key_data=d_data.ptr)
+    synchronize(0)
+    data = d_data.to_host(np.uint8, int(offs[-1])).tobytes()
+    _, kid, _ = pyoracle.gen(seed, 0, 500, n, nkeys, 10)
+    for i in range(n):
+        assert data[offs[i]:offs[i + 1]] == f"sensor_{kid[i]}".encode()
+    for a in (d_lens, d_offs, d_data):
+        a.free()
+
+
+def test_device_utf8_intern_parity():
+    # raw utf8 keys pushed to the device intern vs the oracle on the same
+    # stream; key growth across batches exercises the emission-dictionary
+    # mirror. Ids are schedule-dependent; outputs must still match exactly
+    # (per-group row order + first-seen emission order are id-independent).
+    from denormalized_amd import DeviceArray, _lib
+    nkeys = 700
+    names = [f"sensor_{i}" for i in range(nkeys)]
+    op = make_op(1000, key_kind=_lib.KEY_UTF8, n_keys_hint=nkeys)
+    o = pyoracle.Oracle(1000, 0)
+    outs = []
+    keep = []
+    rng = np.random.default_rng(5150)
+    for b in range(3):
+        n = 60_000
+        ts = (1_000_000 + np.arange(b * n, (b + 1) * n) // 40).astype(np.int64)
+        # batch 0 uses a third of the keyspace, later batches all of it
+        kid = rng.integers(0, nkeys // 3 if b == 0 else nkeys, n)
+        v = rng.uniform(0, 115, n)
+        offs, data = _utf8_cols(kid, nkeys)
+        d_ts = DeviceArray(0, n * 8); d_ts.from_host(ts)
+        d_of = DeviceArray(0, offs.nbytes); d_of.from_host(offs)
+        d_da = DeviceArray(0, max(1, data.nbytes)); d_da.from_host(data)
+        d_v = DeviceArray(0, n * 8); d_v.from_host(v)
+        keep.append((d_ts, d_of, d_da, d_v))  # borrowed until next call
+        op.push_device_utf8(n, d_ts.ptr, d_of.ptr, d_da.ptr, d_v.ptr)
+        outs += op.poll_all()
+        o.push(ts, kid, v)
+    op.finish()
+    outs += op.poll_all()
+    o.finish()
+    exp = o.fetch()
+    assert len(exp["key"]) > 0
+    assert_parity(outs, exp, utf8_keys=names)
+    op.close()
+    o.close()
+    for bufs in keep:
+        for a in bufs:
+            a.free()
+
+
+def test_utf8_mixing_guard():
+    from denormalized_amd import DeviceArray, _lib
+    op = make_op(1000, key_kind=_lib.KEY_UTF8, n_keys_hint=8)
+    ts = np.arange(1_000_000, 1_000_100, dtype=np.int64)
+    v = np.linspace(0, 1, 100)
+    op.push(ts, [f"k{i % 4}" for i in range(100)], v)  # host dictionary path
+    kid = np.arange(100) % 4
+    offs, data = _utf8_cols(kid, 8)
+    d_ts = DeviceArray(0, ts.nbytes); d_ts.from_host(ts)
+    d_of = DeviceArray(0, offs.nbytes); d_of.from_host(offs)
+    d_da = DeviceArray(0, data.nbytes); d_da.from_host(data)
+    d_v = DeviceArray(0, v.nbytes); d_v.from_host(v)
+    with pytest.raises(RuntimeError, match="cannot mix"):
+        op.push_device_utf8(100, d_ts.ptr, d_of.ptr, d_da.ptr, d_v.ptr)
+    op.close()
+    for a in (d_ts, d_of, d_da, d_v):
+        a.free()
