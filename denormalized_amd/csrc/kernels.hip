@@ -126,102 +126,131 @@ __device__ __forceinline__ uint64_t fnv1a64(const char* p, int32_t len) {
     return h;
 }
 
-__global__ __launch_bounds__(BLOCK) void k_intern(const int32_t* offs,
-        const char* data, int64_t n, uint64_t* tab_fp, uint32_t* tab_id,
-        uint32_t p_mask, uint32_t* id_off, uint32_t* id_len, char* pool,
-        uint32_t* ctrs /* [0]=next id, [1]=pool cursor */, uint32_t id_cap,
-        uint32_t pool_cap, int32_t* out_kid, uint32_t* dbg) {
-    const int lane = threadIdx.x & 63;
+__global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
+        const char* data, int64_t n, uint64_t* tab_fp, uint32_t* tab_row,
+        uint32_t p_mask, uint32_t* dbg) {
+    /* phase 1: every row probes; exactly one row CASes each new
+     * fingerprint in, recording itself as the claiming row. No lane ever
+     * waits on another (a publish-wait design can cycle across waves). */
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x - lane + lane;
-         i - lane < n; i += stride) {
-        const bool act = i < n;
-        int32_t o0 = 0, len = 0;
-        uint64_t fp = 0;
-        if (act) {
-            o0 = offs[i];
-            len = offs[i + 1] - o0;
-            fp = fnv1a64(data + o0, len);
-            if (!fp) fp = 1; /* 0 marks an empty table slot */
-        }
-        /* wave dedupe by full 64-bit fingerprint: only LEADERS probe, so
-         * a claim's publisher can never share a wave with its waiter
-         * (intra-wave spin-wait on a divergent branch would deadlock) */
-        uint64_t same = ~0ULL;
-        for (int b = 0; b < 64; b++) {
-            uint64_t bb = __ballot((fp >> b) & 1);
-            same &= ((fp >> b) & 1) ? bb : ~bb;
-        }
-        {
-            uint64_t bb = __ballot(act);
-            same &= act ? bb : ~bb;
-        }
-        const int leader = __ffsll((unsigned long long)same) - 1;
-        int32_t id = -1;
-        if (act && lane == leader) {
-            uint32_t slot = (uint32_t)fp & p_mask;
-            for (long spin_guard = 0;; slot = (slot + 1) & p_mask) {
-                uint64_t got = (uint64_t)atomicCAS(
-                    (unsigned long long*)&tab_fp[slot], 0ULL,
-                    (unsigned long long)fp);
-                if (got == 0) { /* claimed: allocate id + pool bytes */
-                    uint32_t nid = atomicAdd(&ctrs[0], 1u);
-                    uint32_t po = atomicAdd(&ctrs[1], (uint32_t)len);
-                    if (nid >= id_cap || po + (uint32_t)len > pool_cap) {
-                        dbg[3] = nid >= id_cap ? 1 : 2; /* capacity guard */
-                        id = 0;
-                        break;
-                    }
-                    for (int32_t j = 0; j < len; j++)
-                        pool[po + j] = data[o0 + j];
-                    id_off[nid] = po;
-                    id_len[nid] = (uint32_t)len;
-                    __threadfence();
-                    atomicExch(&tab_id[slot], nid); /* publish */
-                    id = (int32_t)nid;
-                    break;
-                }
-                if (got == fp) { /* candidate: wait for publish, verify */
-                    uint32_t cand;
-                    while ((cand = atomicAdd(&tab_id[slot], 0u)) == ~0u) {
-                        if (++spin_guard > (64LL << 20)) {
-                            dbg[3] = 3; /* fp64-collision deadlock guard */
-                            break;
-                        }
-                    }
-                    if (cand == ~0u) { id = 0; break; }
-                    bool eq = id_len[cand] == (uint32_t)len;
-                    const uint32_t co = id_off[cand];
-                    for (int32_t j = 0; eq && j < len; j++)
-                        eq = pool[co + j] == data[o0 + j];
-                    if (eq) {
-                        id = (int32_t)cand;
-                        break;
-                    }
-                    /* full-fp collision between different keys: probe on */
-                }
-                if (++spin_guard > (64LL << 20)) {
-                    dbg[3] = 4; /* table full (probed every slot) */
-                    id = 0;
-                    break;
-                }
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        const int32_t o0 = offs[i];
+        const int32_t len = offs[i + 1] - o0;
+        uint64_t fp = fnv1a64(data + o0, len);
+        if (!fp) fp = 1; /* 0 marks an empty slot */
+        uint32_t slot = (uint32_t)fp & p_mask;
+        for (uint32_t probes = 0;; slot = (slot + 1) & p_mask) {
+            uint64_t got = (uint64_t)atomicCAS(
+                (unsigned long long*)&tab_fp[slot], 0ULL,
+                (unsigned long long)fp);
+            if (got == 0) {
+                tab_row[slot] = (uint32_t)i; /* claimed: I define the bytes */
+                break;
+            }
+            if (got == fp) break; /* same key (fp64 exact; bytes verified in
+                                   * the lookup phase) */
+            if (++probes > p_mask) {
+                dbg[3] = 4; /* table full */
+                break;
             }
         }
-        id = __shfl(id, leader >= 0 ? leader : 0);
-        if (act) out_kid[i] = id;
+    }
+}
+
+__global__ __launch_bounds__(BLOCK) void k_intern_assign(uint64_t* tab_fp,
+        uint32_t* tab_row, uint32_t* tab_id, uint32_t p_count,
+        const int32_t* offs, const char* data, uint32_t* id_off,
+        uint32_t* id_len, char* pool, uint32_t* ctrs, uint32_t id_cap,
+        uint32_t pool_cap, uint32_t* dbg) {
+    /* phase 2 (after claim completes, stream-ordered): one thread per slot;
+     * slots claimed this batch (fp set, id still unassigned) get a dense id
+     * and their first-seen bytes copied into the persistent pool. */
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t sl = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         sl < p_count; sl += stride) {
+        if (tab_fp[sl] == 0 || tab_id[sl] != ~0u) continue;
+        const uint32_t r = tab_row[sl];
+        const int32_t o0 = offs[r];
+        const int32_t len = offs[r + 1] - o0;
+        const uint32_t nid = atomicAdd(&ctrs[0], 1u);
+        const uint32_t po = atomicAdd(&ctrs[1], (uint32_t)len);
+        if (nid >= id_cap || po + (uint32_t)len > pool_cap) {
+            dbg[3] = nid >= id_cap ? 1 : 2; /* capacity guard */
+            continue;
+        }
+        for (int32_t j = 0; j < len; j++) pool[po + j] = data[o0 + j];
+        id_off[nid] = po;
+        id_len[nid] = (uint32_t)len;
+        tab_id[sl] = nid;
+    }
+}
+
+__global__ __launch_bounds__(BLOCK) void k_intern_lookup(const int32_t* offs,
+        const char* data, int64_t n, const uint64_t* tab_fp,
+        const uint32_t* tab_id, uint32_t p_mask, const uint32_t* id_off,
+        const uint32_t* id_len, const char* pool, int32_t* out_kid,
+        uint32_t* dbg) {
+    /* phase 3 (after assign): probe again, byte-verify candidates against
+     * the pool (catches distinct keys sharing a truncated... full-64-bit
+     * fingerprint), emit the dense id per row. */
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        const int32_t o0 = offs[i];
+        const int32_t len = offs[i + 1] - o0;
+        uint64_t fp = fnv1a64(data + o0, len);
+        if (!fp) fp = 1;
+        uint32_t slot = (uint32_t)fp & p_mask;
+        int32_t id = 0;
+        for (uint32_t probes = 0;; slot = (slot + 1) & p_mask) {
+            const uint64_t got = tab_fp[slot];
+            if (got == fp) {
+                const uint32_t cand = tab_id[slot];
+                if (cand == ~0u) { dbg[3] = 5; break; } /* capacity overflow
+                                                         * left it unassigned */
+                bool eq = id_len[cand] == (uint32_t)len;
+                const uint32_t co = id_off[cand];
+                for (int32_t j = 0; eq && j < len; j++)
+                    eq = pool[co + j] == data[o0 + j];
+                if (eq) {
+                    id = (int32_t)cand;
+                    break;
+                }
+                /* full-fp collision between different keys: probe on (the
+                 * colliding key then never claimed its own slot — flag) */
+                dbg[3] = 6;
+            } else if (got == 0) {
+                dbg[3] = 7; /* should be impossible: claim inserted us */
+                break;
+            }
+            if (++probes > p_mask) {
+                dbg[3] = 4;
+                break;
+            }
+        }
+        out_kid[i] = id;
     }
 }
 
 void launch_intern(hipStream_t s, const int32_t* d_offs, const char* d_data,
                    int64_t n, uint64_t* tab_fp, uint32_t* tab_id,
-                   uint32_t p_mask, uint32_t* id_off, uint32_t* id_len,
-                   char* pool, uint32_t* ctrs, uint32_t id_cap,
-                   uint32_t pool_cap, int32_t* out_kid, uint32_t* dbg) {
+                   uint32_t* tab_row, uint32_t p_mask, uint32_t* id_off,
+                   uint32_t* id_len, char* pool, uint32_t* ctrs,
+                   uint32_t id_cap, uint32_t pool_cap, int32_t* out_kid,
+                   uint32_t* dbg) {
     int blocks = (int)std::min<int64_t>((n + BLOCK - 1) / BLOCK, 2048);
     if (blocks < 1) blocks = 1;
-    hipLaunchKernelGGL(k_intern, dim3(blocks), dim3(BLOCK), 0, s, d_offs,
-                       d_data, n, tab_fp, tab_id, p_mask, id_off, id_len, pool,
-                       ctrs, id_cap, pool_cap, out_kid, dbg);
+    const uint32_t P = p_mask + 1;
+    int sblocks = (int)std::min<uint32_t>((P + BLOCK - 1) / BLOCK, 2048);
+    hipLaunchKernelGGL(k_intern_claim, dim3(blocks), dim3(BLOCK), 0, s, d_offs,
+                       d_data, n, tab_fp, tab_row, p_mask, dbg);
+    hipLaunchKernelGGL(k_intern_assign, dim3(sblocks), dim3(BLOCK), 0, s,
+                       tab_fp, tab_row, tab_id, P, d_offs, d_data, id_off,
+                       id_len, pool, ctrs, id_cap, pool_cap, dbg);
+    hipLaunchKernelGGL(k_intern_lookup, dim3(blocks), dim3(BLOCK), 0, s,
+                       d_offs, d_data, n, tab_fp, tab_id, p_mask, id_off,
+                       id_len, pool, out_kid, dbg);
 }
 
 /* synthetic utf8 key generator: "sensor_{k}" with k from the SAME splitmix

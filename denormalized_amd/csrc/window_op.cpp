@@ -227,6 +227,7 @@ struct dz_window_op {
      * from n_keys_hint at create; capacity overflow flags d_dbg[3]. */
     uint64_t* d_itab_fp = nullptr;  /* open-address fingerprint table */
     uint32_t* d_itab_id = nullptr;
+    uint32_t* d_itab_row = nullptr; /* claiming row, valid within one claim */
     uint32_t i_pmask = 0;
     uint32_t* d_ioff = nullptr;     /* per-id {pool offset, byte length} */
     uint32_t* d_ilen = nullptr;
@@ -643,7 +644,7 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_scalars[0]); hipFree(op->d_scalars[1]);
     hipFree(op->d_slotmap); hipFree(op->d_zero_kid);
     hipFree(op->d_resetlist); hipFree(op->d_dbg);
-    hipFree(op->d_itab_fp); hipFree(op->d_itab_id);
+    hipFree(op->d_itab_fp); hipFree(op->d_itab_id); hipFree(op->d_itab_row);
     hipFree(op->d_ioff); hipFree(op->d_ilen);
     hipFree(op->d_ipool); hipFree(op->d_ictrs);
     hipFree(op->d_ikid[0]); hipFree(op->d_ikid[1]);
@@ -1863,6 +1864,7 @@ static dz_status ensure_intern(dz_window_op* op, int64_t n) {
                                                      1u << 31);
         CHK(op, hipMalloc(&op->d_itab_fp, (size_t)P * 8));
         CHK(op, hipMalloc(&op->d_itab_id, (size_t)P * 4));
+        CHK(op, hipMalloc(&op->d_itab_row, (size_t)P * 4));
         CHK(op, hipMalloc(&op->d_ioff, (size_t)op->i_idcap * 4));
         CHK(op, hipMalloc(&op->d_ilen, (size_t)op->i_idcap * 4));
         CHK(op, hipMalloc(&op->d_ipool, op->i_poolcap));
@@ -1941,10 +1943,10 @@ extern "C" dz_status dz_window_op_push_device_utf8(dz_window_op* op,
         CHK(op, hipStreamWaitEvent(op->i_stream, op->ev_consumed[b], 0));
     timed_on(op, op->i_stream, "intern", (double)n_rows * 18, [&] {
         dz::launch_intern(op->i_stream, d_key_offsets, d_key_data, n_rows,
-                          op->d_itab_fp, op->d_itab_id, op->i_pmask,
-                          op->d_ioff, op->d_ilen, op->d_ipool, op->d_ictrs,
-                          op->i_idcap, op->i_poolcap, op->d_ikid[b],
-                          op->d_dbg);
+                          op->d_itab_fp, op->d_itab_id, op->d_itab_row,
+                          op->i_pmask, op->d_ioff, op->d_ilen, op->d_ipool,
+                          op->d_ictrs, op->i_idcap, op->i_poolcap,
+                          op->d_ikid[b], op->d_dbg);
     });
     if (stage_core(op, n_rows, d_ts_ms, op->d_ikid[b], d_vals, nullptr,
                    /*keys_are_dense=*/true, /*deferred=*/true,
