@@ -258,10 +258,23 @@ class WindowOp:
         if self.no_group:
             pass  # global aggregate: no group column in the output schema
         elif self.key_kind == KEY_UTF8:
-            offs = np.ctypeslib.as_array(ob.key_offsets, (n + 1,)).copy() if n else np.zeros(1, np.int32)
-            total = int(offs[-1]) if n else 0
-            data = ctypes.string_at(ob.key_data, total) if total else b""
-            res["key"] = [data[offs[i]:offs[i + 1]].decode() for i in range(n)]
+            if copy:
+                offs = (np.ctypeslib.as_array(ob.key_offsets, (n + 1,)).copy()
+                        if n else np.zeros(1, np.int32))
+                total = int(offs[-1]) if n else 0
+                data = ctypes.string_at(ob.key_data, total) if total else b""
+                res["key"] = [data[offs[i]:offs[i + 1]].decode()
+                              for i in range(n)]
+            else:
+                # zero-copy: raw Arrow views (decoding every key into Python
+                # strings costs more than the whole GPU step)
+                offs = (np.ctypeslib.as_array(ob.key_offsets, (n + 1,))
+                        if n else np.zeros(1, np.int32))
+                total = int(offs[-1]) if n else 0
+                res["key_offsets"] = offs
+                res["key_data"] = (np.ctypeslib.as_array(
+                    ctypes.cast(ob.key_data, ctypes.POINTER(ctypes.c_uint8)),
+                    (total,)) if total else np.zeros(0, np.uint8))
         else:
             res["key"] = mk(ob.key_i64, n, ctypes.c_int64)
         for i, name in enumerate(self.agg_names):

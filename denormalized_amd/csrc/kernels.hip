@@ -128,7 +128,8 @@ __device__ __forceinline__ uint64_t fnv1a64(const char* p, int32_t len) {
 
 __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
         const char* data, int64_t n, uint64_t* tab_fp, uint32_t* tab_row,
-        uint32_t p_mask, uint32_t* dbg) {
+        uint32_t p_mask, int32_t* out_kid /* slot scratch for the lookup */,
+        uint32_t* dbg) {
     /* phase 1: every row probes; exactly one row CASes each new
      * fingerprint in, recording itself as the claiming row. No lane ever
      * waits on another (a publish-wait design can cycle across waves). */
@@ -141,12 +142,17 @@ __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
         if (!fp) fp = 1; /* 0 marks an empty slot */
         uint32_t slot = (uint32_t)fp & p_mask;
         for (uint32_t probes = 0;; slot = (slot + 1) & p_mask) {
-            uint64_t got = (uint64_t)atomicCAS(
-                (unsigned long long*)&tab_fp[slot], 0ULL,
-                (unsigned long long)fp);
+            uint64_t got = tab_fp[slot]; /* read first: the steady state
+                                          * (key already present) must not
+                                          * pay an atomic per row */
             if (got == 0) {
-                tab_row[slot] = (uint32_t)i; /* claimed: I define the bytes */
-                break;
+                got = (uint64_t)atomicCAS(
+                    (unsigned long long*)&tab_fp[slot], 0ULL,
+                    (unsigned long long)fp);
+                if (got == 0) {
+                    tab_row[slot] = (uint32_t)i; /* claimed: I define bytes */
+                    break;
+                }
             }
             if (got == fp) break; /* same key (fp64 exact; bytes verified in
                                    * the lookup phase) */
@@ -155,6 +161,7 @@ __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
                 break;
             }
         }
+        out_kid[i] = (int32_t)slot; /* hand the slot to the lookup phase */
     }
 }
 
@@ -187,47 +194,30 @@ __global__ __launch_bounds__(BLOCK) void k_intern_assign(uint64_t* tab_fp,
 }
 
 __global__ __launch_bounds__(BLOCK) void k_intern_lookup(const int32_t* offs,
-        const char* data, int64_t n, const uint64_t* tab_fp,
-        const uint32_t* tab_id, uint32_t p_mask, const uint32_t* id_off,
-        const uint32_t* id_len, const char* pool, int32_t* out_kid,
-        uint32_t* dbg) {
-    /* phase 3 (after assign): probe again, byte-verify candidates against
-     * the pool (catches distinct keys sharing a truncated... full-64-bit
-     * fingerprint), emit the dense id per row. */
+        const char* data, int64_t n, const uint32_t* tab_id,
+        const uint32_t* id_off, const uint32_t* id_len, const char* pool,
+        int32_t* out_kid, uint32_t* dbg) {
+    /* phase 3 (after assign): resolve each row's slot (stashed by the claim
+     * phase) to its dense id, byte-verifying against the pool — distinct
+     * keys sharing a full 64-bit fingerprint cannot be interned and are
+     * FLAGGED (results then fail loudly via the guard cells). */
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += stride) {
-        const int32_t o0 = offs[i];
-        const int32_t len = offs[i + 1] - o0;
-        uint64_t fp = fnv1a64(data + o0, len);
-        if (!fp) fp = 1;
-        uint32_t slot = (uint32_t)fp & p_mask;
+        const uint32_t slot = (uint32_t)out_kid[i];
+        const uint32_t cand = tab_id[slot];
         int32_t id = 0;
-        for (uint32_t probes = 0;; slot = (slot + 1) & p_mask) {
-            const uint64_t got = tab_fp[slot];
-            if (got == fp) {
-                const uint32_t cand = tab_id[slot];
-                if (cand == ~0u) { dbg[3] = 5; break; } /* capacity overflow
-                                                         * left it unassigned */
-                bool eq = id_len[cand] == (uint32_t)len;
-                const uint32_t co = id_off[cand];
-                for (int32_t j = 0; eq && j < len; j++)
-                    eq = pool[co + j] == data[o0 + j];
-                if (eq) {
-                    id = (int32_t)cand;
-                    break;
-                }
-                /* full-fp collision between different keys: probe on (the
-                 * colliding key then never claimed its own slot — flag) */
-                dbg[3] = 6;
-            } else if (got == 0) {
-                dbg[3] = 7; /* should be impossible: claim inserted us */
-                break;
-            }
-            if (++probes > p_mask) {
-                dbg[3] = 4;
-                break;
-            }
+        if (cand == ~0u) {
+            dbg[3] = 5; /* capacity overflow left the slot unassigned */
+        } else {
+            const int32_t o0 = offs[i];
+            const int32_t len = offs[i + 1] - o0;
+            bool eq = id_len[cand] == (uint32_t)len;
+            const uint32_t co = id_off[cand];
+            for (int32_t j = 0; eq && j < len; j++)
+                eq = pool[co + j] == data[o0 + j];
+            if (eq) id = (int32_t)cand;
+            else dbg[3] = 6; /* fp64 collision between distinct keys */
         }
         out_kid[i] = id;
     }
@@ -244,12 +234,12 @@ void launch_intern(hipStream_t s, const int32_t* d_offs, const char* d_data,
     const uint32_t P = p_mask + 1;
     int sblocks = (int)std::min<uint32_t>((P + BLOCK - 1) / BLOCK, 2048);
     hipLaunchKernelGGL(k_intern_claim, dim3(blocks), dim3(BLOCK), 0, s, d_offs,
-                       d_data, n, tab_fp, tab_row, p_mask, dbg);
+                       d_data, n, tab_fp, tab_row, p_mask, out_kid, dbg);
     hipLaunchKernelGGL(k_intern_assign, dim3(sblocks), dim3(BLOCK), 0, s,
                        tab_fp, tab_row, tab_id, P, d_offs, d_data, id_off,
                        id_len, pool, ctrs, id_cap, pool_cap, dbg);
     hipLaunchKernelGGL(k_intern_lookup, dim3(blocks), dim3(BLOCK), 0, s,
-                       d_offs, d_data, n, tab_fp, tab_id, p_mask, id_off,
+                       d_offs, d_data, n, tab_id, id_off,
                        id_len, pool, out_kid, dbg);
 }
 
