@@ -2,13 +2,16 @@
 """bench.py — measures BASELINE.json's metric: rows/sec through
 window()+filter() on the synthetic sensor stream.
 
-Workload (config.workload): BASELINE cfg2 — the largest single-GPU config the
-metric is quoted on: 10k keys, 1s tumbling window, count/min/max/avg(reading)
+Workload (config.workload): BASELINE cfg2 AS WRITTEN — the largest
+single-GPU config the metric is quoted on: 10k utf8 "sensor_{k}" keys
+(per-row device interning inside the timed region — the GroupValues::intern
+cost the reference pays), 1s tumbling window, count/min/max/avg(reading)
 + filter(max > 113), f64 readings, synthetic sensor stream (seeded; spec in
 DESIGN.md §Generator). A step = one push of --rows-per-step rows through the
 operator (device-resident inputs) including triggered window emission; the
 default run covers 768M rows (≥ cfg2's 100M; longer runs amortize warmup so
-the steady state dominates).
+the steady state dominates). --key-kind dense measures the pre-densified
+int-key variant (round-1's headline).
 
 Contract: `python bench.py --gpus N --steps K --warmup W`. For N>1 the driver
 launches one rank per GPU via torch.distributed.run (RCCL). N-GPU mode
@@ -49,10 +52,12 @@ def parse_args():
     p.add_argument("--window-ms", type=int, default=1000)
     p.add_argument("--slide-ms", type=int, default=0)
     p.add_argument("--seed", type=int, default=42)
-    p.add_argument("--key-kind", choices=["dense", "utf8"], default="dense",
-                   help="utf8 = raw 'sensor_{k}' string keys pushed to the op "
-                        "and interned ON DEVICE inside the timed region "
-                        "(cfg2 as written); dense = pre-densified int ids")
+    p.add_argument("--key-kind", choices=["dense", "utf8"], default="utf8",
+                   help="utf8 (default) = raw 'sensor_{k}' string keys pushed "
+                        "to the op and interned ON DEVICE inside the timed "
+                        "region — BASELINE cfg2 as written; dense = "
+                        "pre-densified int ids (the round-1 variant, kept "
+                        "for comparison)")
     p.add_argument("--no-filter", action="store_true")
     p.add_argument("--dist-backend", default=None,
                    help="torch.distributed backend override (default: nccl on GPU)")
@@ -190,6 +195,7 @@ def main():
             optr = ctypes.c_void_p(d_offs.ptr.value + s * (B + 1) * 4)
             L.dz_memcpy_h2d(optr, host_offs[s].ctypes.data_as(ctypes.c_void_p),
                             (B + 1) * 4)
+            host_offs[s] = None  # cap host memory at one step's offsets
             dz.generate_utf8(device, gseed, s * B, B, args.keys,
                              d_offsets=optr,
                              d_key_data=ctypes.c_void_p(
