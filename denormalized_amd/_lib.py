@@ -126,6 +126,18 @@ def lib():
         L.dz_device_synchronize.argtypes = [ctypes.c_int32]
         L.dz_memcpy_d2h.argtypes = [p, p, ctypes.c_size_t]
         L.dz_memcpy_h2d.argtypes = [p, p, ctypes.c_size_t]
+        L.dz_join_op_create.restype = p
+        L.dz_join_op_create.argtypes = [ctypes.c_int32, i64]
+        L.dz_join_op_destroy.argtypes = [p]
+        L.dz_join_last_error.restype = ctypes.c_char_p
+        L.dz_join_last_error.argtypes = [p]
+        L.dz_join_op_push_build.argtypes = [p, i64, p, p]
+        L.dz_join_op_push_probe.argtypes = [p, i64, p, p, p]
+        L.dz_join_op_matches.argtypes = [p, ctypes.POINTER(i64),
+                                         ctypes.POINTER(p), ctypes.POINTER(p),
+                                         ctypes.POINTER(p)]
+        L.dz_join_op_unmatched.restype = i64
+        L.dz_join_op_unmatched.argtypes = [p]
         L.dz_debug_windows_for_range.restype = i64
         L.dz_debug_windows_for_range.argtypes = [i64, i64, i64, i64, p, p, i64]
         L.dz_version.restype = ctypes.c_char_p
@@ -363,6 +375,62 @@ class WindowOp:
     def close(self):
         if getattr(self, "_h", None):
             self._L.dz_window_op_destroy(self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+class JoinOp:
+    """Stream inner equi-join on trip_id (BASELINE cfg5): build side
+    (trip_id -> driver_id), probe side (ts, trip_id, value). Matches are
+    device-resident columns shaped for WindowOp.push_device(borrowed=True)
+    — see include/denormalized_amd.h for the emission discipline."""
+
+    def __init__(self, device=0, n_trips_hint=1 << 20):
+        self._L = lib()
+        self._h = self._L.dz_join_op_create(device, n_trips_hint)
+        if not self._h:
+            m = self._L.dz_join_last_error(None)
+            raise RuntimeError(f"dz_join_op_create failed: "
+                               f"{m.decode() if m else 'unknown'}")
+
+    def _check(self, st, what):
+        if st != DZ_OK:
+            m = self._L.dz_join_last_error(self._h)
+            raise RuntimeError(f"{what} failed: "
+                               f"{m.decode() if m else 'unknown'}")
+
+    def push_build(self, n, d_trips, d_drivers):
+        self._check(self._L.dz_join_op_push_build(self._h, n, d_trips,
+                                                  d_drivers), "push_build")
+
+    def push_probe(self, n, d_ts, d_trips, d_vals):
+        self._check(self._L.dz_join_op_push_probe(self._h, n, d_ts, d_trips,
+                                                  d_vals), "push_probe")
+
+    def matches(self):
+        """(n, d_ts, d_kid32, d_vals) of the LAST push — device pointers,
+        valid until the second-next push on this op."""
+        n = ctypes.c_int64()
+        ts = ctypes.c_void_p()
+        kid = ctypes.c_void_p()
+        val = ctypes.c_void_p()
+        self._check(self._L.dz_join_op_matches(
+            self._h, ctypes.byref(n), ctypes.byref(ts), ctypes.byref(kid),
+            ctypes.byref(val)), "matches")
+        return n.value, ts, kid, val
+
+    @property
+    def unmatched(self):
+        return self._L.dz_join_op_unmatched(self._h)
+
+    def close(self):
+        if getattr(self, "_h", None):
+            self._L.dz_join_op_destroy(self._h)
             self._h = None
 
     def __del__(self):

@@ -146,6 +146,22 @@ void launch_egather_slabs(hipStream_t stream, const uint64_t* s_base,
                           int64_t stride_u64, EGatherSlots slots, int gcount,
                           uint64_t* out);
 void launch_arm_scalars(hipStream_t stream, uint64_t* scalars);
+
+void launch_fill_i64(hipStream_t stream, int64_t* d_p, int64_t n, int64_t v);
+
+/* stream join (cfg5): open-address trip->driver build table + stable
+ * order-preserving probe/emit (kernels.hip §stream join) */
+void launch_join_build(hipStream_t stream, const int64_t* d_trips,
+                       const int64_t* d_drivers, int64_t n, int64_t* tab_trip,
+                       int64_t* tab_drv, uint64_t p_mask, uint32_t* d_dbg);
+void launch_join_probe(hipStream_t stream, const int64_t* d_ts,
+                       const int64_t* d_trips, const double* d_vals, int64_t n,
+                       int64_t chunk, int C, const int64_t* tab_trip,
+                       const int64_t* tab_drv, uint64_t p_mask,
+                       int32_t* d_drvtmp, uint32_t* d_jcnt, uint32_t* d_mbase,
+                       uint32_t* d_ubase, uint32_t* d_tot, int64_t* o_ts,
+                       int32_t* o_kid, double* o_val, int64_t ubuf_base,
+                       int64_t* u_ts, int64_t* u_trip, double* u_val);
 void launch_zero_counters(hipStream_t stream, uint32_t* two_u32);
 void launch_emission_permute(hipStream_t stream, int64_t K,
                              const uint32_t* counter2, const uint32_t* sidx,

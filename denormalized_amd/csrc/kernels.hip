@@ -1502,3 +1502,183 @@ void dz::launch_emission_permute(hipStream_t st, int64_t K,
                        counter2, sidx, fkid, ocnt, omin, omax, osum, oavg,
                        oflags, out);
 }
+
+/* ------------------------------------------------------------------ */
+/* stream join (BASELINE cfg5): inner equi-join on trip_id, build side */
+/* (trip -> driver) in an open-address device table; probe batches     */
+/* emit matched rows IN ROW ORDER (stable per-chunk compaction) and    */
+/* buffer unmatched rows IN ROW ORDER for re-probe on build growth —   */
+/* the emission discipline oracle.c::orc_join_* restates.              */
+/* ------------------------------------------------------------------ */
+
+namespace dz {
+
+constexpr int64_t JEMPTY = INT64_MIN;
+
+__device__ __forceinline__ uint64_t jhash(int64_t trip) {
+    return (uint64_t)trip * 0x9e3779b97f4a7c15ULL;
+}
+
+__global__ void k_join_build(const int64_t* trips, const int64_t* drivers,
+                             int64_t n, int64_t* tab_trip, int64_t* tab_drv,
+                             uint64_t p_mask, uint32_t* dbg) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        const int64_t trip = trips[i];
+        uint64_t slot = jhash(trip) & p_mask;
+        for (uint64_t probes = 0;; slot = (slot + 1) & p_mask) {
+            int64_t got = tab_trip[slot];
+            if (got == JEMPTY) {
+                got = (int64_t)atomicCAS((unsigned long long*)&tab_trip[slot],
+                                         (unsigned long long)JEMPTY,
+                                         (unsigned long long)trip);
+                if (got == JEMPTY) got = trip; /* claimed */
+            }
+            if (got == trip) {
+                /* later duplicates overwrite (dimension update); the winner
+                 * among duplicates WITHIN one batch is unspecified — same
+                 * contract as the oracle documents */
+                tab_drv[slot] = drivers[i];
+                break;
+            }
+            if (++probes > p_mask) {
+                dbg[3] = 8; /* join table full */
+                break;
+            }
+        }
+    }
+}
+
+/* probe pass 1: resolve each row's driver (or -1) into drv_tmp and count
+ * matches per chunk (one block per chunk) */
+__global__ __launch_bounds__(64) void k_join_mark(const int64_t* trips,
+        int64_t n, int64_t chunk, const int64_t* tab_trip,
+        const int64_t* tab_drv, uint64_t p_mask, int32_t* drv_tmp,
+        uint32_t* jcnt) {
+    const int64_t lo = blockIdx.x * chunk;
+    const int64_t hi = i64min(n, lo + chunk);
+    uint32_t m = 0;
+    for (int64_t i = lo + threadIdx.x; i < hi; i += 64) {
+        const int64_t trip = trips[i];
+        uint64_t slot = jhash(trip) & p_mask;
+        int32_t drv = -1;
+        for (;;) {
+            const int64_t got = tab_trip[slot];
+            if (got == trip) {
+                drv = (int32_t)tab_drv[slot];
+                break;
+            }
+            if (got == JEMPTY) break;
+            slot = (slot + 1) & p_mask;
+        }
+        drv_tmp[i] = drv;
+        m += drv >= 0 ? 1 : 0;
+    }
+    for (int o = 32; o > 0; o >>= 1) m += (uint32_t)__shfl_down((int)m, o);
+    if (threadIdx.x == 0) jcnt[blockIdx.x] = m;
+}
+
+/* exclusive scan of per-chunk matched counts + complement (single block):
+ * mbase/ubase per chunk; totals in tot[0] (matched) / tot[1] (unmatched) */
+__global__ __launch_bounds__(1024) void k_join_scan(const uint32_t* jcnt,
+        int C, int64_t n, int64_t chunk, uint32_t* mbase, uint32_t* ubase,
+        uint32_t* tot) {
+    __shared__ uint32_t runm, runu;
+    if (threadIdx.x == 0) {
+        uint32_t rm = 0, ru = 0;
+        for (int c = 0; c < C; c++) {
+            const uint32_t rows =
+                (uint32_t)(i64min(n, (int64_t)(c + 1) * chunk) -
+                           i64min(n, (int64_t)c * chunk));
+            mbase[c] = rm;
+            ubase[c] = ru;
+            rm += jcnt[c];
+            ru += rows - jcnt[c];
+        }
+        runm = rm;
+        runu = ru;
+        tot[0] = rm;
+        tot[1] = ru;
+    }
+}
+
+/* probe pass 2: stable per-chunk split — matched rows (ts, driver-as-kid,
+ * val) to the output block, unmatched (ts, trip, val) appended to the
+ * unmatched buffer; one 64-thread block per chunk, wave ballots keep row
+ * order */
+__global__ __launch_bounds__(64) void k_join_emit(const int64_t* ts,
+        const int64_t* trips, const double* vals, const int32_t* drv_tmp,
+        int64_t n, int64_t chunk, const uint32_t* mbase, const uint32_t* ubase,
+        int64_t* o_ts, int32_t* o_kid, double* o_val, int64_t ubuf_base,
+        int64_t* u_ts, int64_t* u_trip, double* u_val) {
+    const int64_t lo = blockIdx.x * chunk;
+    const int64_t hi = i64min(n, lo + chunk);
+    uint32_t mcur = mbase[blockIdx.x];
+    uint32_t ucur = ubase[blockIdx.x];
+    const int lane = threadIdx.x;
+    for (int64_t i0 = lo; i0 < hi; i0 += 64) {
+        const int64_t i = i0 + lane;
+        const bool act = i < hi;
+        const int32_t drv = act ? drv_tmp[i] : -1;
+        const bool hit = act && drv >= 0;
+        const uint64_t mm = __ballot(hit);
+        const uint64_t um = __ballot(act && !hit);
+        const uint64_t below = (lane == 63) ? ~0ULL : ((1ULL << (lane + 1)) - 1);
+        if (hit) {
+            const uint32_t p = mcur + (uint32_t)__popcll(mm & below) - 1;
+            o_ts[p] = ts[i];
+            o_kid[p] = drv;
+            o_val[p] = vals[i];
+        } else if (act) {
+            const int64_t p = ubuf_base + ucur +
+                              (uint32_t)__popcll(um & below) - 1;
+            u_ts[p] = ts[i];
+            u_trip[p] = trips[i];
+            u_val[p] = vals[i];
+        }
+        mcur += (uint32_t)__popcll(mm);
+        ucur += (uint32_t)__popcll(um);
+    }
+}
+
+__global__ void k_fill_i64(int64_t* p, int64_t n, int64_t v) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        p[i] = v;
+}
+
+void launch_fill_i64(hipStream_t s, int64_t* d_p, int64_t n, int64_t v) {
+    int blocks = (int)std::min<int64_t>((n + BLOCK - 1) / BLOCK, 2048);
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(k_fill_i64, dim3(blocks), dim3(BLOCK), 0, s, d_p, n, v);
+}
+
+void launch_join_build(hipStream_t s, const int64_t* d_trips,
+                       const int64_t* d_drivers, int64_t n, int64_t* tab_trip,
+                       int64_t* tab_drv, uint64_t p_mask, uint32_t* d_dbg) {
+    int blocks = (int)std::min<int64_t>((n + BLOCK - 1) / BLOCK, 2048);
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(k_join_build, dim3(blocks), dim3(BLOCK), 0, s, d_trips,
+                       d_drivers, n, tab_trip, tab_drv, p_mask, d_dbg);
+}
+
+void launch_join_probe(hipStream_t s, const int64_t* d_ts,
+                       const int64_t* d_trips, const double* d_vals, int64_t n,
+                       int64_t chunk, int C, const int64_t* tab_trip,
+                       const int64_t* tab_drv, uint64_t p_mask,
+                       int32_t* d_drvtmp, uint32_t* d_jcnt, uint32_t* d_mbase,
+                       uint32_t* d_ubase, uint32_t* d_tot, int64_t* o_ts,
+                       int32_t* o_kid, double* o_val, int64_t ubuf_base,
+                       int64_t* u_ts, int64_t* u_trip, double* u_val) {
+    hipLaunchKernelGGL(k_join_mark, dim3(C), dim3(64), 0, s, d_trips, n, chunk,
+                       tab_trip, tab_drv, p_mask, d_drvtmp, d_jcnt);
+    hipLaunchKernelGGL(k_join_scan, dim3(1), dim3(1024), 0, s, d_jcnt, C, n,
+                       chunk, d_mbase, d_ubase, d_tot);
+    hipLaunchKernelGGL(k_join_emit, dim3(C), dim3(64), 0, s, d_ts, d_trips,
+                       d_vals, d_drvtmp, n, chunk, d_mbase, d_ubase, o_ts,
+                       o_kid, o_val, ubuf_base, u_ts, u_trip, u_val);
+}
+
+} // namespace dz

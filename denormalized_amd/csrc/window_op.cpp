@@ -2358,3 +2358,251 @@ extern "C" dz_status dz_memcpy_d2h(void* dst, const void* src, size_t bytes) {
 extern "C" dz_status dz_memcpy_h2d(void* dst, const void* src, size_t bytes) {
     return hipMemcpy(dst, src, bytes, hipMemcpyHostToDevice) == hipSuccess ? DZ_OK : DZ_ERR;
 }
+
+/* ------------------------------------------------------------------ */
+/* stream join operator (BASELINE cfg5): inner equi-join on trip_id    */
+/* feeding the windowed group-by. The reference lowers .join() to      */
+/* DataFusion's inner hash join (datastream.rs:126-175); the streaming */
+/* emission discipline here matches oracle.c::orc_join_* exactly:      */
+/* matched probe rows emit IN ROW ORDER, unmatched rows buffer in row  */
+/* order and re-emit (original order) when their build row arrives.    */
+/* Outputs are DEVICE-RESIDENT (ts, dense driver kid, value) columns   */
+/* sized for a zero-copy borrowed push into a dz_window_op — the       */
+/* join+window pipeline stays on HBM end to end.                       */
+/* ------------------------------------------------------------------ */
+
+struct dz_join_op {
+    int device = 0;
+    std::string err;
+    hipStream_t stream = nullptr;
+    int64_t* tab_trip = nullptr; /* open-address (trip -> driver), JEMPTY=empty */
+    int64_t* tab_drv = nullptr;
+    uint64_t p_mask = 0;
+    /* unmatched ping-pong buffers (order-preserving) */
+    int64_t* u_ts[2] = {};
+    int64_t* u_trip[2] = {};
+    double* u_val[2] = {};
+    int64_t u_cap = 0, u_n = 0;
+    int u_cur = 0;
+    /* matched output, double-buffered: buffer b stays valid until the
+     * SECOND-next push on the join op (a window op's borrowed push reads
+     * it until the window's next call) */
+    int64_t* o_ts[2] = {};
+    int32_t* o_kid[2] = {};
+    double* o_val[2] = {};
+    int64_t o_cap = 0;
+    int o_cur = 0;
+    int64_t o_n = 0;
+    /* probe scratch */
+    int32_t* d_drvtmp = nullptr;
+    int64_t drv_cap = 0;
+    uint32_t* d_jcnt = nullptr;  /* [512] counts + mbase + ubase + tot */
+    uint32_t* h_tot = nullptr;   /* pinned {matched, unmatched} */
+    uint32_t* d_dbg = nullptr;
+};
+
+extern "C" void dz_join_op_destroy(dz_join_op* op);
+
+#define JCHK(op, call)                                                     \
+    do {                                                                   \
+        hipError_t e_ = (call);                                            \
+        if (e_ != hipSuccess) {                                            \
+            (op)->err = std::string(#call) + ": " + hipGetErrorString(e_); \
+            return DZ_ERR;                                                 \
+        }                                                                  \
+    } while (0)
+
+extern "C" dz_join_op* dz_join_op_create(int32_t device, int64_t n_trips_hint) {
+    g_err.clear();
+    int ndev = 0;
+    if (hipGetDeviceCount(&ndev) != hipSuccess || ndev <= device) {
+        g_err = "no HIP device available (this operator has no CPU fallback)";
+        return nullptr;
+    }
+    auto* op = new dz_join_op();
+    op->device = device;
+    uint64_t want = 4 * (uint64_t)std::max<int64_t>(n_trips_hint, 1);
+    uint64_t P = 1u << 16;
+    while (P < want && P < (1ull << 28)) P <<= 1;
+    op->p_mask = P - 1;
+    if (hipSetDevice(device) != hipSuccess ||
+        hipStreamCreate(&op->stream) != hipSuccess ||
+        hipMalloc(&op->tab_trip, P * 8) != hipSuccess ||
+        hipMalloc(&op->tab_drv, P * 8) != hipSuccess ||
+        hipMalloc(&op->d_jcnt, (3 * 512 + 2) * 4) != hipSuccess ||
+        hipMalloc(&op->d_dbg, 16) != hipSuccess ||
+        hipHostMalloc((void**)&op->h_tot, 8) != hipSuccess) {
+        g_err = "join op allocation failed";
+        delete op;
+        return nullptr;
+    }
+    hipMemset(op->d_dbg, 0, 16);
+    dz::launch_fill_i64(op->stream, op->tab_trip, (int64_t)P, INT64_MIN);
+    if (hipStreamSynchronize(op->stream) != hipSuccess) {
+        g_err = "join table init failed";
+        dz_join_op_destroy(op);
+        return nullptr;
+    }
+    return op;
+}
+
+extern "C" void dz_join_op_destroy(dz_join_op* op) {
+    if (!op) return;
+    hipSetDevice(op->device);
+    if (op->stream) hipStreamSynchronize(op->stream);
+    hipFree(op->tab_trip); hipFree(op->tab_drv);
+    for (int i = 0; i < 2; i++) {
+        hipFree(op->u_ts[i]); hipFree(op->u_trip[i]); hipFree(op->u_val[i]);
+        hipFree(op->o_ts[i]); hipFree(op->o_kid[i]); hipFree(op->o_val[i]);
+    }
+    hipFree(op->d_drvtmp); hipFree(op->d_jcnt); hipFree(op->d_dbg);
+    if (op->h_tot) hipHostFree(op->h_tot);
+    if (op->stream) hipStreamDestroy(op->stream);
+    delete op;
+}
+
+extern "C" const char* dz_join_last_error(dz_join_op* op) {
+    if (!op) return g_err.empty() ? nullptr : g_err.c_str();
+    return op->err.empty() ? nullptr : op->err.c_str();
+}
+
+static dz_status join_ensure_probe(dz_join_op* op, int64_t n) {
+    if (n > op->drv_cap) {
+        JCHK(op, hipStreamSynchronize(op->stream));
+        hipFree(op->d_drvtmp);
+        JCHK(op, hipMalloc(&op->d_drvtmp, (size_t)n * 4));
+        op->drv_cap = n;
+    }
+    /* output must hold this probe's matches OR a full buffer re-probe */
+    int64_t need_o = std::max(n, op->u_n);
+    if (need_o > op->o_cap) {
+        JCHK(op, hipStreamSynchronize(op->stream));
+        for (int i = 0; i < 2; i++) {
+            hipFree(op->o_ts[i]); hipFree(op->o_kid[i]); hipFree(op->o_val[i]);
+            JCHK(op, hipMalloc(&op->o_ts[i], (size_t)need_o * 8));
+            JCHK(op, hipMalloc(&op->o_kid[i], (size_t)need_o * 4));
+            JCHK(op, hipMalloc(&op->o_val[i], (size_t)need_o * 8));
+        }
+        op->o_cap = need_o;
+    }
+    int64_t need_u = op->u_n + n;
+    if (need_u > op->u_cap) {
+        JCHK(op, hipStreamSynchronize(op->stream));
+        int64_t cap = std::max<int64_t>(need_u, op->u_cap * 2);
+        for (int i = 0; i < 2; i++) {
+            int64_t* nts; int64_t* ntr; double* nv;
+            JCHK(op, hipMalloc(&nts, (size_t)cap * 8));
+            JCHK(op, hipMalloc(&ntr, (size_t)cap * 8));
+            JCHK(op, hipMalloc(&nv, (size_t)cap * 8));
+            if (op->u_n > 0 && i == op->u_cur) {
+                JCHK(op, hipMemcpy(nts, op->u_ts[i], (size_t)op->u_n * 8,
+                                   hipMemcpyDeviceToDevice));
+                JCHK(op, hipMemcpy(ntr, op->u_trip[i], (size_t)op->u_n * 8,
+                                   hipMemcpyDeviceToDevice));
+                JCHK(op, hipMemcpy(nv, op->u_val[i], (size_t)op->u_n * 8,
+                                   hipMemcpyDeviceToDevice));
+            }
+            hipFree(op->u_ts[i]); hipFree(op->u_trip[i]); hipFree(op->u_val[i]);
+            op->u_ts[i] = nts; op->u_trip[i] = ntr; op->u_val[i] = nv;
+        }
+        op->u_cap = cap;
+    }
+    return DZ_OK;
+}
+
+static dz_status join_check_dbg(dz_join_op* op) {
+    uint32_t cells[4] = {0, 0, 0, 0};
+    if (hipMemcpy(cells, op->d_dbg, 16, hipMemcpyDeviceToHost) == hipSuccess &&
+        cells[3] == 8) {
+        op->err = "join build table full — raise n_trips_hint";
+        return DZ_ERR;
+    }
+    return DZ_OK;
+}
+
+extern "C" dz_status dz_join_op_push_build(dz_join_op* op, int64_t n,
+                                           const int64_t* d_trip_ids,
+                                           const int64_t* d_driver_ids) {
+    if (!op) return DZ_ERR;
+    JCHK(op, hipSetDevice(op->device));
+    op->o_n = 0;
+    if (n > 0)
+        dz::launch_join_build(op->stream, d_trip_ids, d_driver_ids, n,
+                              op->tab_trip, op->tab_drv, op->p_mask, op->d_dbg);
+    if (op->u_n > 0) {
+        /* re-probe the unmatched buffer: newly matched rows emit in their
+         * original buffered order; the rest compact into the other buffer */
+        if (join_ensure_probe(op, 0) != DZ_OK) return DZ_ERR;
+        const int src = op->u_cur, dst = src ^ 1;
+        int C = (int)std::min<int64_t>(512,
+                                       std::max<int64_t>(1, (op->u_n + 8191) / 8192));
+        int64_t chunk = (op->u_n + C - 1) / C;
+        const int nxt = op->o_cur ^ 1;
+        dz::launch_join_probe(op->stream, op->u_ts[src], op->u_trip[src],
+                              op->u_val[src], op->u_n, chunk, C, op->tab_trip,
+                              op->tab_drv, op->p_mask, op->d_drvtmp,
+                              op->d_jcnt, op->d_jcnt + 512, op->d_jcnt + 1024,
+                              op->d_jcnt + 1536, op->o_ts[nxt], op->o_kid[nxt],
+                              op->o_val[nxt], 0, op->u_ts[dst],
+                              op->u_trip[dst], op->u_val[dst]);
+        JCHK(op, hipMemcpyAsync(op->h_tot, op->d_jcnt + 1536, 8,
+                                hipMemcpyDeviceToHost, op->stream));
+        JCHK(op, hipStreamSynchronize(op->stream));
+        op->o_cur = nxt;
+        op->o_n = op->h_tot[0];
+        op->u_n = op->h_tot[1];
+        op->u_cur = dst;
+    } else {
+        JCHK(op, hipStreamSynchronize(op->stream));
+    }
+    return join_check_dbg(op);
+}
+
+extern "C" dz_status dz_join_op_push_probe(dz_join_op* op, int64_t n,
+                                           const int64_t* d_ts_ms,
+                                           const int64_t* d_trip_ids,
+                                           const double* d_vals) {
+    if (!op) return DZ_ERR;
+    JCHK(op, hipSetDevice(op->device));
+    op->o_n = 0;
+    if (n <= 0) return DZ_OK;
+    if (n > op->drv_cap || op->u_n + n > op->u_cap ||
+        std::max(n, op->u_n) > op->o_cap)
+        if (join_ensure_probe(op, n) != DZ_OK) return DZ_ERR;
+    int C = (int)std::min<int64_t>(512, std::max<int64_t>(1, (n + 8191) / 8192));
+    int64_t chunk = (n + C - 1) / C;
+    const int nxt = op->o_cur ^ 1;
+    const int ub = op->u_cur;
+    dz::launch_join_probe(op->stream, d_ts_ms, d_trip_ids, d_vals, n, chunk, C,
+                          op->tab_trip, op->tab_drv, op->p_mask, op->d_drvtmp,
+                          op->d_jcnt, op->d_jcnt + 512, op->d_jcnt + 1024,
+                          op->d_jcnt + 1536, op->o_ts[nxt], op->o_kid[nxt],
+                          op->o_val[nxt], op->u_n, op->u_ts[ub],
+                          op->u_trip[ub], op->u_val[ub]);
+    JCHK(op, hipMemcpyAsync(op->h_tot, op->d_jcnt + 1536, 8,
+                            hipMemcpyDeviceToHost, op->stream));
+    JCHK(op, hipStreamSynchronize(op->stream));
+    op->o_cur = nxt;
+    op->o_n = op->h_tot[0];
+    op->u_n += op->h_tot[1];
+    return join_check_dbg(op);
+}
+
+/* matched output of the LAST push (device-resident). Valid until the
+ * SECOND-next push on this join op (double-buffered), which covers a
+ * dz_window_op borrowed push consuming it across one pipeline step. */
+extern "C" dz_status dz_join_op_matches(dz_join_op* op, int64_t* n_out,
+                                        const int64_t** d_ts,
+                                        const int32_t** d_kid,
+                                        const double** d_vals) {
+    if (!op || !n_out) return DZ_ERR;
+    *n_out = op->o_n;
+    if (d_ts) *d_ts = op->o_ts[op->o_cur];
+    if (d_kid) *d_kid = op->o_kid[op->o_cur];
+    if (d_vals) *d_vals = op->o_val[op->o_cur];
+    return DZ_OK;
+}
+
+extern "C" int64_t dz_join_op_unmatched(dz_join_op* op) {
+    return op ? op->u_n : 0;
+}

@@ -218,6 +218,49 @@ dz_status dz_window_op_set_filter(dz_window_op* op, int32_t agg_idx,
                                   int32_t cmp, double literal);
 
 /* ------------------------------------------------------------------ */
+/* Stream join (BASELINE cfg5): inner equi-join on trip_id feeding the
+ * windowed group-by. The reference lowers DataStream::join to DataFusion's
+ * inner hash join (crates/core/src/datastream.rs:126-175; its only join
+ * example composes two streams, examples/examples/stream_join.rs). The
+ * streaming discipline (restated in oracle/oracle.c orc_join_*): build side
+ * = (trip_id -> driver_id) dimension table (later duplicates overwrite;
+ * the winner among duplicates WITHIN one build batch is unspecified); each
+ * probe batch emits its matching rows IN ROW ORDER; unmatched rows buffer
+ * in row order and re-emit (original order) when their build row arrives;
+ * rows never matched are dropped (inner join). Outputs are device-resident
+ * (ts, driver id as a dense int32 key, value) columns shaped for a
+ * zero-copy dz_window_op_push_device_borrowed — driver ids must fit int32.
+ * Build-table capacity is sized from n_trips_hint (4x, pow2) and overflow
+ * fails loudly. */
+
+typedef struct dz_join_op dz_join_op;
+
+dz_join_op* dz_join_op_create(int32_t device, int64_t n_trips_hint);
+void dz_join_op_destroy(dz_join_op* op);
+const char* dz_join_last_error(dz_join_op* op);
+
+/* Device-resident build push (trip_id -> driver_id rows). Also re-probes
+ * the unmatched buffer: newly matched rows become this call's matches. */
+dz_status dz_join_op_push_build(dz_join_op* op, int64_t n_rows,
+                                const int64_t* d_trip_ids,
+                                const int64_t* d_driver_ids);
+
+/* Device-resident probe push (event rows). Matching rows become this
+ * call's matches; the rest join the unmatched buffer. */
+dz_status dz_join_op_push_probe(dz_join_op* op, int64_t n_rows,
+                                const int64_t* d_ts_ms,
+                                const int64_t* d_trip_ids,
+                                const double* d_vals);
+
+/* The LAST push's matched rows (device pointers, valid until the
+ * second-next push on this op — double-buffered so a window op's borrowed
+ * push can consume them across one pipeline step). */
+dz_status dz_join_op_matches(dz_join_op* op, int64_t* n_out,
+                             const int64_t** d_ts_ms, const int32_t** d_kids,
+                             const double** d_vals);
+int64_t dz_join_op_unmatched(dz_join_op* op);
+
+/* ------------------------------------------------------------------ */
 /* Synthetic sensor stream generator, on device (bench/test input; spec in
  * DESIGN.md §Generator, bit-identical to oracle orc_gen; mirrors
  * examples/examples/emit_measurements.rs:30-67). Any output pointer may be

@@ -428,3 +428,141 @@ void orc_gen(uint64_t seed, int64_t t0_ms, int64_t start_row, int64_t nrows,
         if (val) val[i] = (double)(splitmix64(r) >> 11) * (1.0 / 9007199254740992.0) * 115.0;
     }
 }
+
+/* ------------------------------------------------------------------ */
+/* Stream join restatement (BASELINE cfg5: rideshare-style two-topic   */
+/* inner equi-join on trip_id feeding the windowed group-by).          */
+/*                                                                     */
+/* Reference anchors: DataStream::join/join_on lowers to DataFusion's  */
+/* standard inner hash join (crates/core/src/datastream.rs:126-175;    */
+/* the arithmetic lives in the unvendored DataFusion fork — parity     */
+/* pinning as for the aggregates, see the header note). The STREAMING  */
+/* emission discipline is defined HERE (the reference's only join      */
+/* example joins two closed-window outputs, examples/stream_join.rs):  */
+/*  - build side (trips): (trip_id -> driver_id), at most one row per  */
+/*    trip_id (later duplicates overwrite — dimension-table update);   */
+/*  - probe side (events): each pushed batch emits its matching rows   */
+/*    IN ROW ORDER; rows with no match yet are buffered IN ROW ORDER;  */
+/*  - each build push re-probes the buffer and emits newly matched     */
+/*    rows in their original buffered order (late-data semantics of    */
+/*    the downstream window op apply);                                 */
+/*  - inner join: rows never matched are dropped.                      */
+/* The GPU join implements the SAME discipline, so join->window        */
+/* pipelines compare bit-exactly.                                      */
+/* ------------------------------------------------------------------ */
+
+typedef struct {
+    int64_t cap, mask, n;   /* open-address trip table, power-of-two */
+    int64_t* trips;         /* INT64_MIN = empty */
+    int64_t* drivers;
+    /* unmatched probe rows, in arrival order */
+    int64_t un_cap, un_n;
+    int64_t* u_ts; int64_t* u_trip; double* u_val;
+    /* matched output, cleared by fetch */
+    int64_t o_cap, o_n;
+    int64_t* o_ts; int64_t* o_driver; double* o_val;
+} OrcJoin;
+
+static void oj_out(OrcJoin* j, int64_t ts, int64_t drv, double v) {
+    if (j->o_n == j->o_cap) {
+        j->o_cap = j->o_cap ? j->o_cap * 2 : 1024;
+        j->o_ts = (int64_t*)realloc(j->o_ts, j->o_cap * 8);
+        j->o_driver = (int64_t*)realloc(j->o_driver, j->o_cap * 8);
+        j->o_val = (double*)realloc(j->o_val, j->o_cap * 8);
+    }
+    j->o_ts[j->o_n] = ts; j->o_driver[j->o_n] = drv; j->o_val[j->o_n] = v;
+    j->o_n++;
+}
+
+static int64_t oj_find(OrcJoin* j, int64_t trip) {
+    uint64_t h = (uint64_t)trip * 0x9e3779b97f4a7c15ULL;
+    int64_t s = (int64_t)(h & (uint64_t)j->mask);
+    while (j->trips[s] != INT64_MIN) {
+        if (j->trips[s] == trip) return s;
+        s = (s + 1) & j->mask;
+    }
+    return ~s; /* not found; ~slot = insertion point */
+}
+
+OrcJoin* orc_join_create(void) {
+    OrcJoin* j = (OrcJoin*)calloc(1, sizeof(OrcJoin));
+    j->cap = 1024; j->mask = j->cap - 1;
+    j->trips = (int64_t*)malloc(j->cap * 8);
+    j->drivers = (int64_t*)malloc(j->cap * 8);
+    for (int64_t i = 0; i < j->cap; i++) j->trips[i] = INT64_MIN;
+    return j;
+}
+
+void orc_join_destroy(OrcJoin* j) {
+    if (!j) return;
+    free(j->trips); free(j->drivers);
+    free(j->u_ts); free(j->u_trip); free(j->u_val);
+    free(j->o_ts); free(j->o_driver); free(j->o_val);
+    free(j);
+}
+
+void orc_join_push_build(OrcJoin* j, int64_t n, const int64_t* trips,
+                         const int64_t* drivers) {
+    for (int64_t i = 0; i < n; i++) {
+        if (j->n * 2 >= j->cap) { /* grow + rehash */
+            int64_t oc = j->cap;
+            int64_t* ot = j->trips; int64_t* od = j->drivers;
+            j->cap *= 2; j->mask = j->cap - 1; j->n = 0;
+            j->trips = (int64_t*)malloc(j->cap * 8);
+            j->drivers = (int64_t*)malloc(j->cap * 8);
+            for (int64_t s = 0; s < j->cap; s++) j->trips[s] = INT64_MIN;
+            for (int64_t s = 0; s < oc; s++)
+                if (ot[s] != INT64_MIN) {
+                    int64_t ns = oj_find(j, ot[s]);
+                    ns = ~ns;
+                    j->trips[ns] = ot[s]; j->drivers[ns] = od[s]; j->n++;
+                }
+            free(ot); free(od);
+        }
+        int64_t s = oj_find(j, trips[i]);
+        if (s < 0) { s = ~s; j->trips[s] = trips[i]; j->n++; }
+        j->drivers[s] = drivers[i]; /* later duplicate overwrites */
+    }
+    /* re-probe the unmatched buffer in original order */
+    int64_t w = 0;
+    for (int64_t i = 0; i < j->un_n; i++) {
+        int64_t s = oj_find(j, j->u_trip[i]);
+        if (s >= 0) {
+            oj_out(j, j->u_ts[i], j->drivers[s], j->u_val[i]);
+        } else {
+            j->u_ts[w] = j->u_ts[i]; j->u_trip[w] = j->u_trip[i];
+            j->u_val[w] = j->u_val[i]; w++;
+        }
+    }
+    j->un_n = w;
+}
+
+void orc_join_push_probe(OrcJoin* j, int64_t n, const int64_t* ts,
+                         const int64_t* trips, const double* vals) {
+    for (int64_t i = 0; i < n; i++) {
+        int64_t s = oj_find(j, trips[i]);
+        if (s >= 0) {
+            oj_out(j, ts[i], j->drivers[s], vals[i]);
+        } else {
+            if (j->un_n == j->un_cap) {
+                j->un_cap = j->un_cap ? j->un_cap * 2 : 1024;
+                j->u_ts = (int64_t*)realloc(j->u_ts, j->un_cap * 8);
+                j->u_trip = (int64_t*)realloc(j->u_trip, j->un_cap * 8);
+                j->u_val = (double*)realloc(j->u_val, j->un_cap * 8);
+            }
+            j->u_ts[j->un_n] = ts[i]; j->u_trip[j->un_n] = trips[i];
+            j->u_val[j->un_n] = vals[i]; j->un_n++;
+        }
+    }
+}
+
+int64_t orc_join_out_rows(OrcJoin* j) { return j->o_n; }
+int64_t orc_join_unmatched(OrcJoin* j) { return j->un_n; }
+
+void orc_join_out_fetch(OrcJoin* j, int64_t* ts, int64_t* drivers,
+                        double* vals) {
+    memcpy(ts, j->o_ts, j->o_n * 8);
+    memcpy(drivers, j->o_driver, j->o_n * 8);
+    memcpy(vals, j->o_val, j->o_n * 8);
+    j->o_n = 0;
+}

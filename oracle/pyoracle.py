@@ -136,3 +136,70 @@ class Oracle:
             self.close()
         except Exception:
             pass
+
+
+class JoinOracle:
+    """Inner equi-join restatement (oracle.c orc_join_*): build side
+    trip_id -> driver_id, probe rows emit in row order, unmatched rows
+    buffer and re-emit in original order when their build row arrives."""
+
+    def __init__(self):
+        self._L = lib()
+        L = self._L
+        import ctypes
+        p = ctypes.c_void_p
+        i64 = ctypes.c_int64
+        if not getattr(L, "_join_boot", False):
+            L.orc_join_create.restype = p
+            L.orc_join_destroy.argtypes = [p]
+            L.orc_join_push_build.argtypes = [p, i64, p, p]
+            L.orc_join_push_probe.argtypes = [p, i64, p, p, p]
+            L.orc_join_out_rows.restype = i64
+            L.orc_join_out_rows.argtypes = [p]
+            L.orc_join_unmatched.restype = i64
+            L.orc_join_unmatched.argtypes = [p]
+            L.orc_join_out_fetch.argtypes = [p, p, p, p]
+            L._join_boot = True
+        self._h = L.orc_join_create()
+
+    def push_build(self, trips, drivers):
+        import numpy as np
+        trips = np.ascontiguousarray(trips, np.int64)
+        drivers = np.ascontiguousarray(drivers, np.int64)
+        self._L.orc_join_push_build(self._h, len(trips), _ptr(trips),
+                                    _ptr(drivers))
+
+    def push_probe(self, ts, trips, vals):
+        import numpy as np
+        ts = np.ascontiguousarray(ts, np.int64)
+        trips = np.ascontiguousarray(trips, np.int64)
+        vals = np.ascontiguousarray(vals, np.float64)
+        self._L.orc_join_push_probe(self._h, len(ts), _ptr(ts), _ptr(trips),
+                                    _ptr(vals))
+
+    @property
+    def unmatched(self):
+        return self._L.orc_join_unmatched(self._h)
+
+    def fetch(self):
+        import numpy as np
+        n = self._L.orc_join_out_rows(self._h)
+        ts = np.empty(n, np.int64)
+        drv = np.empty(n, np.int64)
+        val = np.empty(n, np.float64)
+        if n:
+            self._L.orc_join_out_fetch(self._h, _ptr(ts), _ptr(drv), _ptr(val))
+        else:
+            self._L.orc_join_out_fetch(self._h, _ptr(ts), _ptr(drv), _ptr(val))
+        return ts, drv, val
+
+    def close(self):
+        if getattr(self, "_h", None):
+            self._L.orc_join_destroy(self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
