@@ -2467,11 +2467,13 @@ extern "C" const char* dz_join_last_error(dz_join_op* op) {
 }
 
 static dz_status join_ensure_probe(dz_join_op* op, int64_t n) {
-    if (n > op->drv_cap) {
+    /* the buffer re-probe on a build push marks u_n rows, not n */
+    int64_t need_drv = std::max(n, op->u_n);
+    if (need_drv > op->drv_cap) {
         JCHK(op, hipStreamSynchronize(op->stream));
         hipFree(op->d_drvtmp);
-        JCHK(op, hipMalloc(&op->d_drvtmp, (size_t)n * 4));
-        op->drv_cap = n;
+        JCHK(op, hipMalloc(&op->d_drvtmp, (size_t)need_drv * 4));
+        op->drv_cap = need_drv;
     }
     /* output must hold this probe's matches OR a full buffer re-probe */
     int64_t need_o = std::max(n, op->u_n);
@@ -2566,9 +2568,7 @@ extern "C" dz_status dz_join_op_push_probe(dz_join_op* op, int64_t n,
     JCHK(op, hipSetDevice(op->device));
     op->o_n = 0;
     if (n <= 0) return DZ_OK;
-    if (n > op->drv_cap || op->u_n + n > op->u_cap ||
-        std::max(n, op->u_n) > op->o_cap)
-        if (join_ensure_probe(op, n) != DZ_OK) return DZ_ERR;
+    if (join_ensure_probe(op, n) != DZ_OK) return DZ_ERR;
     int C = (int)std::min<int64_t>(512, std::max<int64_t>(1, (n + 8191) / 8192));
     int64_t chunk = (n + C - 1) / C;
     const int nxt = op->o_cur ^ 1;
