@@ -59,6 +59,13 @@ def parse_args():
                         "pre-densified int ids (the round-1 variant, kept "
                         "for comparison)")
     p.add_argument("--no-filter", action="store_true")
+    p.add_argument("--cfg5", action="store_true",
+                   help="BASELINE cfg5 pipeline: inner join on trip_id "
+                        "(1M-trip build side) feeding the 1s tumbling window "
+                        "group-by driver; value = probe rows/s through "
+                        "join()+window(). Per-GPU shard; rows all match "
+                        "(build side resident before the timed region).")
+    p.add_argument("--trips", type=int, default=1_000_000)
     p.add_argument("--dist-backend", default=None,
                    help="torch.distributed backend override (default: nccl on GPU)")
     p.add_argument("--staged-push", action="store_true",
@@ -162,14 +169,34 @@ def main():
     # pre-generate the whole stream into HBM (inputs resident when the timed
     # region starts). Each rank owns a disjoint key shard (weak scaling):
     # same time distribution, rank-distinct keys (ids are rank-local).
-    utf8 = args.key_kind == "utf8"
+    utf8 = args.key_kind == "utf8" and not args.cfg5
     gseed = args.seed + 1000003 * rank  # rank-distinct draws per shard
     d_ts = dz.DeviceArray(device, total_rows * 8)
-    d_kid = None if utf8 else dz.DeviceArray(device, total_rows * 4)
+    d_kid = None if (utf8 or args.cfg5) else dz.DeviceArray(device,
+                                                            total_rows * 4)
+    d_trips = join = None
+    if args.cfg5:
+        # probe stream: trip ids (int64) from the same seeded draw; build
+        # side (trip -> driver) fully resident before the timed region
+        d_trips = dz.DeviceArray(device, total_rows * 8)
     d_vals = dz.DeviceArray(device, total_rows * 8)
-    dz.generate(device, gseed, 1_000_000, 0, total_rows, args.keys,
-                args.rows_per_ms, d_ts.ptr, None,
+    dz.generate(device, gseed, 1_000_000, 0, total_rows,
+                args.trips if args.cfg5 else args.keys,
+                args.rows_per_ms, d_ts.ptr,
+                d_trips.ptr if args.cfg5 else None,
                 d_kid.ptr if d_kid else None, d_vals.ptr)
+    if args.cfg5:
+        rng = np.random.default_rng(gseed)
+        trips_h = np.arange(args.trips, dtype=np.int64)
+        drivers_h = rng.integers(0, args.keys, args.trips).astype(np.int64)
+        d_btr = dz.DeviceArray(device, trips_h.nbytes)
+        d_btr.from_host(trips_h)
+        d_bdr = dz.DeviceArray(device, drivers_h.nbytes)
+        d_bdr.from_host(drivers_h)
+        join = dz.JoinOp(device=device, n_trips_hint=args.trips)
+        join.push_build(args.trips, d_btr.ptr, d_bdr.ptr)
+        d_btr.free()
+        d_bdr.free()
     d_offs = d_data = None
     step_data_base = []
     if utf8:
@@ -213,6 +240,19 @@ def main():
 
     def do_push(step):
         off = step * B
+        if args.cfg5:
+            # join -> window, all on device: probe the build table, feed the
+            # matched (ts, driver kid, value) columns to the window op
+            # zero-copy (the join double-buffers its output, covering the
+            # window's one-step borrowed lifetime)
+            join.push_probe(B,
+                            ctypes.c_void_p(d_ts.ptr.value + off * 8),
+                            ctypes.c_void_p(d_trips.ptr.value + off * 8),
+                            ctypes.c_void_p(d_vals.ptr.value + off * 8))
+            nm, pts, pkid, pval = join.matches()
+            if nm:
+                op.push_device(nm, pts, pkid, pval, borrowed=True)
+            return
         if utf8:
             op.push_device_utf8(
                 B,
@@ -352,7 +392,10 @@ def main():
             path["pmc_frac"] = pmc_bytes_per_step / step_s / HBM_PEAK
             path["amplification"] = pmc_bytes_per_step / (ALG_BYTES_PER_ROW * B)
         out = {
-            "metric": "rows/sec through window()+filter() on synthetic sensor stream",
+            "metric": ("probe rows/sec through join()+window() on synthetic "
+                       "rideshare stream" if args.cfg5 else
+                       "rows/sec through window()+filter() on synthetic "
+                       "sensor stream"),
             "value": value,
             "unit": "rows/s",
             "n_gpus": world,
@@ -365,11 +408,14 @@ def main():
             "dtype": "f64",
             "data": "synthetic",
             "config": {
-                "workload": (("cfg2-utf8 (as written, device-interned "
-                              "string keys): " if utf8 else
-                              "cfg2 (dense-int key variant): ")
-                             if (args.keys == 10_000 and not args.slide_ms
-                                 and args.window_ms == 1000) else "")
+                "workload": ("cfg5 (per-GPU shard): join(trip_id, "
+                             f"{args.trips} trips)->window group-by driver, "
+                             if args.cfg5 else
+                             (("cfg2-utf8 (as written, device-interned "
+                               "string keys): " if utf8 else
+                               "cfg2 (dense-int key variant): ")
+                              if (args.keys == 10_000 and not args.slide_ms
+                                  and args.window_ms == 1000) else ""))
                             + f"{K * B / 1e6:.0f}M rows/GPU, "
                             + (f"{args.keys} utf8 'sensor_{{k}}' keys "
                                "(per-row device intern in the timed region), "
@@ -410,7 +456,9 @@ def main():
                     "for the GPU/CPU ratio)")
         print(json.dumps(out))
     op.close()
-    for a in (d_ts, d_kid, d_vals, d_offs, d_data):
+    if join is not None:
+        join.close()
+    for a in (d_ts, d_kid, d_vals, d_offs, d_data, d_trips):
         if a is not None:
             a.free()
     if dist is not None:
