@@ -19,8 +19,8 @@ All per-row compute runs in hand-written CDNA4 HIP kernels behind the C ABI
 (include/denormalized_amd.h); there is no CPU fallback.
 """
 from . import _lib
-from ._lib import (WindowOp, JoinOp, DeviceArray, generate,  # noqa: F401
-                   generate_utf8, synchronize)
+from ._lib import (WindowOp, JoinOp, JsonDecoder, DeviceArray,  # noqa: F401
+                   generate, generate_utf8, generate_json, synchronize)
 
 __version__ = "0.1"
 

@@ -138,6 +138,20 @@ def lib():
                                          ctypes.POINTER(p)]
         L.dz_join_op_unmatched.restype = i64
         L.dz_join_op_unmatched.argtypes = [p]
+        L.dz_json_decoder_create.restype = p
+        L.dz_json_decoder_create.argtypes = [ctypes.c_int32, ctypes.c_char_p,
+                                             ctypes.c_char_p, ctypes.c_char_p]
+        L.dz_json_decoder_destroy.argtypes = [p]
+        L.dz_json_decoder_last_error.restype = ctypes.c_char_p
+        L.dz_json_decoder_last_error.argtypes = [p]
+        L.dz_json_decode.argtypes = [p, p, i64]
+        L.dz_json_decoder_batch.argtypes = [p, ctypes.POINTER(i64),
+                                            ctypes.POINTER(p),
+                                            ctypes.POINTER(p),
+                                            ctypes.POINTER(p),
+                                            ctypes.POINTER(p)]
+        L.dz_generate_json.argtypes = [ctypes.c_int32, ctypes.c_uint64, i64,
+                                       i64, i64, i64, i64, p, p, p]
         L.dz_debug_windows_for_range.restype = i64
         L.dz_debug_windows_for_range.argtypes = [i64, i64, i64, i64, p, p, i64]
         L.dz_version.restype = ctypes.c_char_p
@@ -384,6 +398,58 @@ class WindowOp:
             pass
 
 
+class JsonDecoder:
+    """Device JSON decode (from_topic's decode stage): newline-delimited
+    records -> (ts, utf8 key column, value) device columns shaped for
+    WindowOp.push_device_utf8. See include/denormalized_amd.h for the
+    documented subset."""
+
+    def __init__(self, device=0, ts_field="occurred_at_ms",
+                 key_field="sensor_name", val_field="reading"):
+        self._L = lib()
+        self._h = self._L.dz_json_decoder_create(
+            device, ts_field.encode(), key_field.encode(), val_field.encode())
+        if not self._h:
+            m = self._L.dz_json_decoder_last_error(None)
+            raise RuntimeError(f"dz_json_decoder_create failed: "
+                               f"{m.decode() if m else 'unknown'}")
+
+    def _check(self, st, what):
+        if st != DZ_OK:
+            m = self._L.dz_json_decoder_last_error(self._h)
+            raise RuntimeError(f"{what} failed: "
+                               f"{m.decode() if m else 'unknown'}")
+
+    def decode(self, d_bytes, n_bytes):
+        self._check(self._L.dz_json_decode(self._h, d_bytes, n_bytes),
+                    "json decode")
+        return self.batch()
+
+    def batch(self):
+        """(n, d_ts, d_key_offsets, d_key_data, d_vals) of the LAST decode
+        — device pointers, valid until the second-next decode."""
+        n = ctypes.c_int64()
+        ts = ctypes.c_void_p()
+        ko = ctypes.c_void_p()
+        kd = ctypes.c_void_p()
+        v = ctypes.c_void_p()
+        self._check(self._L.dz_json_decoder_batch(
+            self._h, ctypes.byref(n), ctypes.byref(ts), ctypes.byref(ko),
+            ctypes.byref(kd), ctypes.byref(v)), "batch")
+        return n.value, ts, ko, kd, v
+
+    def close(self):
+        if getattr(self, "_h", None):
+            self._L.dz_json_decoder_destroy(self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
 class JoinOp:
     """Stream inner equi-join on trip_id (BASELINE cfg5): build side
     (trip_id -> driver_id), probe side (ts, trip_id, value). Matches are
@@ -481,6 +547,14 @@ def generate(device, seed, t0_ms, start_row, n_rows, n_keys, rows_per_ms,
     if L.dz_generate(device, seed, t0_ms, start_row, n_rows, n_keys, rows_per_ms,
                      d_ts, d_keys, d_kid32, d_vals) != DZ_OK:
         raise RuntimeError(f"dz_generate failed: {_err(L)}")
+
+
+def generate_json(device, seed, t0_ms, start_row, n_rows, n_keys,
+                  rows_per_ms, d_lens=None, d_offsets=None, d_data=None):
+    L = lib()
+    if L.dz_generate_json(device, seed, t0_ms, start_row, n_rows, n_keys,
+                          rows_per_ms, d_lens, d_offsets, d_data) != DZ_OK:
+        raise RuntimeError(f"dz_generate_json failed: {_err(L)}")
 
 
 def generate_utf8(device, seed, start_row, n_rows, n_keys, d_lens=None,

@@ -149,6 +149,29 @@ void launch_arm_scalars(hipStream_t stream, uint64_t* scalars);
 
 void launch_fill_i64(hipStream_t stream, int64_t* d_p, int64_t n, int64_t v);
 
+/* JSON ingest (kernels.hip §JSON ingest): newline count/scan, then
+ * emit+parse+key compaction with the record count known host-side */
+struct JsonFields {
+    char ts_name[32];
+    char key_name[32];
+    char val_name[32];
+    int32_t ts_len, key_len, val_len;
+};
+void launch_json_count(hipStream_t stream, const char* d_buf, int64_t nbytes,
+                       int C, int64_t chunk, int32_t tail_record,
+                       uint32_t* d_cnt, uint32_t* d_base, uint32_t* d_tot);
+void launch_json_parse(hipStream_t stream, const char* d_buf, int64_t nbytes,
+                       int C, int64_t chunk, const JsonFields& jf,
+                       const uint32_t* d_base, const uint32_t* d_tot,
+                       int64_t* d_recoff, int64_t nrec, int64_t* o_ts,
+                       int64_t* o_kbeg, int32_t* o_klen, double* o_val,
+                       uint32_t* d_blocksum, uint32_t* d_ktot, int32_t* o_koff,
+                       char* o_kdata, uint32_t* d_dbg);
+void launch_gen_json(hipStream_t stream, uint64_t seed, int64_t t0,
+                     int64_t start_row, int64_t n, int64_t nkeys,
+                     int64_t rows_per_ms, int32_t* d_lens,
+                     const int64_t* d_offs, char* d_data);
+
 /* stream join (cfg5): open-address trip->driver build table + stable
  * order-preserving probe/emit (kernels.hip §stream join) */
 void launch_join_build(hipStream_t stream, const int64_t* d_trips,

@@ -1682,3 +1682,463 @@ void launch_join_probe(hipStream_t s, const int64_t* d_ts,
 }
 
 } // namespace dz
+
+/* ------------------------------------------------------------------ */
+/* JSON ingest: the reference decodes Kafka payload      */
+/* bytes to RecordBatches on the host via serde_json                   */
+/* (crates/core/src/formats/decoders/json.rs:23-46 over the stream     */
+/* read path kafka_stream_read.rs:165-296). Here the decode runs ON    */
+/* DEVICE: newline-split, schema-directed field parse (int64 ts, utf8  */
+/* key, f64 reading; other fields of any JSON shape are skipped), and  */
+/* a compact Arrow key column — shaped to feed the utf8-intern push.   */
+/* Number parsing uses the EXACT Clinger fast path (<=15 significant   */
+/* digits and 10^|e|<=22: both factors exactly representable => the    */
+/* correctly-rounded double, identical to strtod); longer literals     */
+/* flag the debug cell and the decode fails loudly (documented         */
+/* subset — the synthetic sensor streams print 6-decimal fixed).       */
+/* ------------------------------------------------------------------ */
+
+namespace dz {
+
+__global__ __launch_bounds__(64) void k_json_nl_count(const char* b, int64_t n,
+        int64_t chunk, uint32_t* cnt) {
+    const int64_t lo = blockIdx.x * chunk;
+    const int64_t hi = i64min(n, lo + chunk);
+    uint32_t m = 0;
+    for (int64_t i = lo + threadIdx.x; i < hi; i += 64)
+        m += b[i] == '\n' ? 1 : 0;
+    for (int o = 32; o > 0; o >>= 1) m += (uint32_t)__shfl_down((int)m, o);
+    if (threadIdx.x == 0) cnt[blockIdx.x] = m;
+}
+
+/* exclusive scan over the C chunk counts + total records:
+ * rec_total = newlines + (tail bytes after the last newline ? 1 : 0).
+ * The tail test needs the LAST byte: done host-side (flag arg). */
+__global__ void k_json_nl_scan(const uint32_t* cnt, int C, uint32_t* base,
+                               int32_t tail_record, uint32_t* tot) {
+    if (threadIdx.x == 0 && blockIdx.x == 0) {
+        uint32_t run = 1; /* record 0 starts at byte 0 */
+        for (int c = 0; c < C; c++) {
+            base[c] = run;
+            run += cnt[c];
+        }
+        /* run now counts 1 + newlines = candidate starts; the start right
+         * after a trailing newline is not a record */
+        tot[0] = tail_record ? run : run - 1;
+    }
+}
+
+/* stable start-offset emission: record i starts at 0 or right after the
+ * i-1'th newline */
+__global__ __launch_bounds__(64) void k_json_nl_emit(const char* b, int64_t n,
+        int64_t chunk, const uint32_t* base, const uint32_t* tot,
+        int64_t* rec_off) {
+    if (blockIdx.x == 0 && threadIdx.x == 0) rec_off[0] = 0;
+    const int64_t lo = blockIdx.x * chunk;
+    const int64_t hi = i64min(n, lo + chunk);
+    uint32_t cur = base[blockIdx.x];
+    const uint32_t nt = tot[0];
+    for (int64_t i0 = lo; i0 < hi; i0 += 64) {
+        const int64_t i = i0 + threadIdx.x;
+        const bool nl = i < hi && b[i] == '\n';
+        const uint64_t m = __ballot(nl);
+        const uint64_t below =
+            (threadIdx.x == 63) ? ~0ULL : ((1ULL << (threadIdx.x + 1)) - 1);
+        if (nl) {
+            const uint32_t p = cur + (uint32_t)__popcll(m & below) - 1;
+            if (p < nt) rec_off[p] = i + 1;
+        }
+        cur += (uint32_t)__popcll(m);
+    }
+}
+
+/* schema-directed per-record parse (JsonFields in dz_internal.h) */
+__device__ __forceinline__ bool jf_eq(const char* a, int32_t alen,
+                                      const char* b, int32_t blen) {
+    if (alen != blen) return false;
+    for (int32_t i = 0; i < alen; i++)
+        if (a[i] != b[i]) return false;
+    return true;
+}
+
+__device__ __forceinline__ const char* j_ws(const char* p, const char* e) {
+    while (p < e && (*p == ' ' || *p == '\t' || *p == '\r')) p++;
+    return p;
+}
+
+__global__ __launch_bounds__(BLOCK) void k_json_parse(const char* buf,
+        const int64_t* rec_off, const uint32_t* tot, int64_t nbytes,
+        JsonFields jf, int64_t* o_ts, int64_t* o_kbeg, int32_t* o_klen,
+        double* o_val, uint32_t* dbg) {
+    const double p10[23] = {1e0, 1e1, 1e2, 1e3, 1e4, 1e5, 1e6, 1e7, 1e8, 1e9,
+                            1e10, 1e11, 1e12, 1e13, 1e14, 1e15, 1e16, 1e17,
+                            1e18, 1e19, 1e20, 1e21, 1e22};
+    const uint32_t nt = tot[0];
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; r < nt;
+         r += stride) {
+        const char* p = buf + rec_off[r];
+        const char* e = (r + 1 < nt) ? buf + rec_off[r + 1] - 1 : buf + nbytes;
+        while (e > p && (*(e - 1) == '\n' || *(e - 1) == '\r')) e--;
+        int64_t ts = 0;
+        int64_t kbeg = -1;
+        int32_t klen = 0;
+        double val = 0.0;
+        bool have_ts = false, have_key = false, have_val = false, bad = false;
+        p = j_ws(p, e);
+        if (p >= e || *p != '{') bad = true;
+        if (!bad) p++;
+        while (!bad) {
+            p = j_ws(p, e);
+            if (p < e && *p == '}') break;
+            if (p < e && *p == ',') { p++; continue; }
+            if (p >= e || *p != '"') { bad = true; break; }
+            const char* name = ++p;
+            while (p < e && *p != '"') {
+                if (*p == '\\') { bad = true; break; } /* escaped field names
+                                                        * unsupported */
+                p++;
+            }
+            if (bad || p >= e) { bad = true; break; }
+            const int32_t nlen = (int32_t)(p - name);
+            p++;
+            p = j_ws(p, e);
+            if (p >= e || *p != ':') { bad = true; break; }
+            p = j_ws(p + 1, e);
+            if (p >= e) { bad = true; break; }
+            const bool is_ts = jf_eq(name, nlen, jf.ts_name, jf.ts_len);
+            const bool is_key = jf_eq(name, nlen, jf.key_name, jf.key_len);
+            const bool is_val = jf_eq(name, nlen, jf.val_name, jf.val_len);
+            if (*p == '"') { /* string value */
+                const char* s = ++p;
+                while (p < e && *p != '"') {
+                    if (*p == '\\') { bad = true; break; } /* escapes: out of
+                                                            * the subset */
+                    p++;
+                }
+                if (bad || p >= e) { bad = true; break; }
+                if (is_key) {
+                    kbeg = (int64_t)(s - buf);
+                    klen = (int32_t)(p - s);
+                    have_key = true;
+                } else if (is_ts || is_val) {
+                    bad = true; /* wrong type for a schema field */
+                    break;
+                }
+                p++;
+            } else if (*p == '{' || *p == '[') { /* skip nested structure */
+                if (is_ts || is_key || is_val) { bad = true; break; }
+                int depth = 0;
+                bool instr = false;
+                while (p < e) {
+                    const char c = *p;
+                    if (instr) {
+                        if (c == '\\' && p + 1 < e) p++;
+                        else if (c == '"') instr = false;
+                    } else if (c == '"') {
+                        instr = true;
+                    } else if (c == '{' || c == '[') {
+                        depth++;
+                    } else if (c == '}' || c == ']') {
+                        depth--;
+                        if (depth == 0) { p++; break; }
+                    }
+                    p++;
+                }
+                if (depth != 0) { bad = true; break; }
+            } else { /* number / literal */
+                bool neg = false;
+                if (*p == '-') { neg = true; p++; }
+                if (p < e && (*p == 't' || *p == 'f' || *p == 'n')) {
+                    /* true/false/null: skip letters */
+                    if (is_ts || is_key || is_val) { bad = true; break; }
+                    while (p < e && *p >= 'a' && *p <= 'z') p++;
+                    continue;
+                }
+                uint64_t mant = 0;
+                int32_t ndig = 0, frac = 0, exp10 = 0;
+                while (p < e && *p >= '0' && *p <= '9') {
+                    if (ndig < 19) mant = mant * 10 + (uint64_t)(*p - '0');
+                    ndig++;
+                    p++;
+                }
+                if (p < e && *p == '.') {
+                    p++;
+                    while (p < e && *p >= '0' && *p <= '9') {
+                        if (ndig < 19) {
+                            mant = mant * 10 + (uint64_t)(*p - '0');
+                            ndig++;
+                            frac++;
+                        } else {
+                            ndig++;
+                        }
+                        p++;
+                    }
+                }
+                if (p < e && (*p == 'e' || *p == 'E')) {
+                    p++;
+                    bool eneg = false;
+                    if (p < e && (*p == '+' || *p == '-')) {
+                        eneg = *p == '-';
+                        p++;
+                    }
+                    int32_t ev = 0;
+                    while (p < e && *p >= '0' && *p <= '9') {
+                        ev = ev * 10 + (*p - '0');
+                        p++;
+                    }
+                    exp10 = eneg ? -ev : ev;
+                }
+                if (is_ts) {
+                    if (frac || exp10) { bad = true; break; }
+                    ts = neg ? -(int64_t)mant : (int64_t)mant;
+                    have_ts = true;
+                } else if (is_val) {
+                    const int32_t e10 = exp10 - frac;
+                    if (ndig > 15 || e10 > 22 || e10 < -22) {
+                        dbg[2] = 2; /* outside the exact fast path */
+                        bad = true;
+                        break;
+                    }
+                    double d = (double)mant;
+                    d = e10 >= 0 ? d * p10[e10] : d / p10[-e10];
+                    val = neg ? -d : d;
+                    have_val = true;
+                }
+            }
+        }
+        if (bad || !have_ts || !have_key || !have_val) {
+            dbg[2] = bad ? 1 : 3; /* malformed / missing schema field */
+            ts = 0; kbeg = rec_off[r]; klen = 0; val = 0.0;
+        }
+        o_ts[r] = ts;
+        o_kbeg[r] = kbeg;
+        o_klen[r] = klen;
+        o_val[r] = val;
+    }
+}
+
+/* generic exclusive scan (u32) for the key-length -> offsets pass:
+ * per-block totals, single-block scan of totals, add-back */
+constexpr int JS_ELEMS = 4096; /* per block */
+
+__global__ __launch_bounds__(BLOCK) void k_scanu32_a(const int32_t* in,
+        int64_t n, uint32_t* blocksum) {
+    const int64_t lo = (int64_t)blockIdx.x * JS_ELEMS;
+    const int64_t hi = i64min(n, lo + JS_ELEMS);
+    uint32_t s = 0;
+    for (int64_t i = lo + threadIdx.x; i < hi; i += BLOCK)
+        s += (uint32_t)in[i];
+    __shared__ uint32_t red[WAVES_PER_BLOCK];
+    for (int o = 32; o > 0; o >>= 1) s += (uint32_t)__shfl_down((int)s, o);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = s;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        uint32_t t = 0;
+        for (int w = 0; w < WAVES_PER_BLOCK; w++) t += red[w];
+        blocksum[blockIdx.x] = t;
+    }
+}
+
+__global__ __launch_bounds__(1024) void k_scanu32_b(uint32_t* blocksum,
+        int nb, uint32_t* total) {
+    /* single block: exclusive scan over nb block sums (loops if nb > 1024) */
+    __shared__ uint32_t carry;
+    if (threadIdx.x == 0) carry = 0;
+    __syncthreads();
+    __shared__ uint32_t buf[1024];
+    for (int b0 = 0; b0 < nb; b0 += 1024) {
+        const int i = b0 + threadIdx.x;
+        uint32_t v = (i < nb) ? blocksum[i] : 0;
+        buf[threadIdx.x] = v;
+        __syncthreads();
+        for (int o = 1; o < 1024; o <<= 1) {
+            uint32_t t = (threadIdx.x >= (unsigned)o) ? buf[threadIdx.x - o] : 0;
+            __syncthreads();
+            buf[threadIdx.x] += t;
+            __syncthreads();
+        }
+        if (i < nb)
+            blocksum[i] = carry + (threadIdx.x ? buf[threadIdx.x - 1] : 0);
+        __syncthreads();
+        if (threadIdx.x == 0) carry += buf[1023];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0 && total) total[0] = carry;
+}
+
+__global__ __launch_bounds__(BLOCK) void k_scanu32_c(const int32_t* in,
+        int64_t n, const uint32_t* blocksum, int32_t* out /* n+1 */) {
+    /* per-block local exclusive scan + base add-back */
+    __shared__ uint32_t thr[BLOCK];
+    const int64_t lo = (int64_t)blockIdx.x * JS_ELEMS;
+    const int64_t hi = i64min(n, lo + JS_ELEMS);
+    constexpr int PER = JS_ELEMS / BLOCK;
+    uint32_t loc[PER];
+    uint32_t s = 0;
+    for (int j = 0; j < PER; j++) {
+        const int64_t i = lo + (int64_t)threadIdx.x * PER + j;
+        loc[j] = s;
+        if (i < hi) s += (uint32_t)in[i];
+    }
+    thr[threadIdx.x] = s;
+    __syncthreads();
+    for (int o = 1; o < BLOCK; o <<= 1) {
+        uint32_t t = (threadIdx.x >= (unsigned)o) ? thr[threadIdx.x - o] : 0;
+        __syncthreads();
+        thr[threadIdx.x] += t;
+        __syncthreads();
+    }
+    const uint32_t base = blocksum[blockIdx.x] +
+                          (threadIdx.x ? thr[threadIdx.x - 1] : 0);
+    for (int j = 0; j < PER; j++) {
+        const int64_t i = lo + (int64_t)threadIdx.x * PER + j;
+        if (i < hi) out[i] = (int32_t)(base + loc[j]);
+    }
+    if (blockIdx.x == 0 && threadIdx.x == 0) out[n] = 0; /* patched below */
+}
+
+__global__ void k_scanu32_tail(int64_t n, const uint32_t* total, int32_t* out) {
+    out[n] = (int32_t)total[0];
+}
+
+/* gather the key bytes into the compact Arrow data buffer */
+__global__ __launch_bounds__(BLOCK) void k_json_copykeys(const char* buf,
+        const int64_t* kbeg, const int32_t* klen, const int32_t* koff,
+        const uint32_t* tot, char* kdata) {
+    const uint32_t nt = tot[0];
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; r < nt;
+         r += stride) {
+        const char* s = buf + kbeg[r];
+        char* d = kdata + koff[r];
+        const int32_t l = klen[r];
+        for (int32_t j = 0; j < l; j++) d[j] = s[j];
+    }
+}
+
+void launch_json_count(hipStream_t s, const char* d_buf, int64_t nbytes,
+                       int C, int64_t chunk, int32_t tail_record,
+                       uint32_t* d_cnt, uint32_t* d_base, uint32_t* d_tot) {
+    hipLaunchKernelGGL(k_json_nl_count, dim3(C), dim3(64), 0, s, d_buf, nbytes,
+                       chunk, d_cnt);
+    hipLaunchKernelGGL(k_json_nl_scan, dim3(1), dim3(64), 0, s, d_cnt, C,
+                       d_base, tail_record, d_tot);
+}
+
+void launch_json_parse(hipStream_t s, const char* d_buf, int64_t nbytes,
+                       int C, int64_t chunk, const JsonFields& jf,
+                       const uint32_t* d_base, const uint32_t* d_tot,
+                       int64_t* d_recoff, int64_t nrec, int64_t* o_ts,
+                       int64_t* o_kbeg, int32_t* o_klen, double* o_val,
+                       uint32_t* d_blocksum, uint32_t* d_ktot, int32_t* o_koff,
+                       char* o_kdata, uint32_t* d_dbg) {
+    hipLaunchKernelGGL(k_json_nl_emit, dim3(C), dim3(64), 0, s, d_buf, nbytes,
+                       chunk, d_base, d_tot, d_recoff);
+    int pblocks = (int)std::min<int64_t>((nrec + BLOCK - 1) / BLOCK, 2048);
+    if (pblocks < 1) pblocks = 1;
+    hipLaunchKernelGGL(k_json_parse, dim3(pblocks), dim3(BLOCK), 0, s, d_buf,
+                       d_recoff, d_tot, nbytes, jf, o_ts, o_kbeg, o_klen,
+                       o_val, d_dbg);
+    const int nb = (int)((nrec + JS_ELEMS - 1) / JS_ELEMS);
+    hipLaunchKernelGGL(k_scanu32_a, dim3(nb), dim3(BLOCK), 0, s, o_klen,
+                       nrec, d_blocksum);
+    hipLaunchKernelGGL(k_scanu32_b, dim3(1), dim3(1024), 0, s, d_blocksum, nb,
+                       d_ktot);
+    hipLaunchKernelGGL(k_scanu32_c, dim3(nb), dim3(BLOCK), 0, s, o_klen,
+                       nrec, d_blocksum, o_koff);
+    hipLaunchKernelGGL(k_scanu32_tail, dim3(1), dim3(1), 0, s, nrec,
+                       d_ktot, o_koff);
+    hipLaunchKernelGGL(k_json_copykeys, dim3(pblocks), dim3(BLOCK), 0, s,
+                       d_buf, o_kbeg, o_klen, o_koff, d_tot, o_kdata);
+}
+
+} // namespace dz
+
+/* synthetic on-wire JSON generator: one newline-delimited record per row of
+ * the same seeded sensor stream (emit_measurements.rs shape), reading
+ * printed as fixed 6-decimal (inside the decoder's exact parse subset):
+ * {"occurred_at_ms":T,"sensor_name":"sensor_K","reading":III.FFFFFF}\n */
+namespace dz {
+
+__device__ __forceinline__ void jgen_row(uint64_t seed, int64_t t0,
+        int64_t rows_per_ms, int64_t nkeys, int64_t gi, int64_t* ts,
+        uint64_t* k, uint64_t* r6) {
+    uint64_t r = splitmix64(seed ^ (0x9e3779b97f4a7c15ULL * (uint64_t)(gi + 1)));
+    *k = r % (uint64_t)nkeys;
+    *ts = t0 + gi / rows_per_ms;
+    const double v =
+        (double)(splitmix64(r) >> 11) * (1.0 / 9007199254740992.0) * 115.0;
+    *r6 = (uint64_t)(v * 1e6 + 0.5);
+}
+
+__global__ void k_gen_json_lens(uint64_t seed, int64_t t0, int64_t start_row,
+                                int64_t n, int64_t nkeys, int64_t rows_per_ms,
+                                int32_t* lens) {
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        int64_t ts;
+        uint64_t k, r6;
+        jgen_row(seed, t0, rows_per_ms, nkeys, start_row + i, &ts, &k, &r6);
+        lens[i] = 18 + dec_digits((uint64_t)ts) + 23 + dec_digits(k) + 12 +
+                  dec_digits(r6 / 1000000) + 1 + 6 + 2;
+    }
+}
+
+__device__ __forceinline__ char* jgen_u64(char* p, uint64_t v) {
+    const int32_t d = dec_digits(v);
+    for (int32_t j = d - 1; j >= 0; j--) {
+        p[j] = (char)('0' + (v % 10));
+        v /= 10;
+    }
+    return p + d;
+}
+
+__global__ void k_gen_json_fill(uint64_t seed, int64_t t0, int64_t start_row,
+                                int64_t n, int64_t nkeys, int64_t rows_per_ms,
+                                const int64_t* offs, char* data) {
+    const char* A = "{\"occurred_at_ms\":";
+    const char* B = ",\"sensor_name\":\"sensor_";
+    const char* Cs = "\",\"reading\":";
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        int64_t ts;
+        uint64_t k, r6;
+        jgen_row(seed, t0, rows_per_ms, nkeys, start_row + i, &ts, &k, &r6);
+        char* p = data + offs[i];
+        for (int j = 0; j < 18; j++) *p++ = A[j];
+        p = jgen_u64(p, (uint64_t)ts);
+        for (int j = 0; j < 23; j++) *p++ = B[j];
+        p = jgen_u64(p, k);
+        for (int j = 0; j < 12; j++) *p++ = Cs[j];
+        p = jgen_u64(p, r6 / 1000000);
+        *p++ = '.';
+        uint64_t f = r6 % 1000000;
+        for (int32_t j = 5; j >= 0; j--) {
+            p[j] = (char)('0' + (f % 10));
+            f /= 10;
+        }
+        p += 6;
+        *p++ = '}';
+        *p++ = '\n';
+    }
+}
+
+void launch_gen_json(hipStream_t s, uint64_t seed, int64_t t0,
+                     int64_t start_row, int64_t n, int64_t nkeys,
+                     int64_t rows_per_ms, int32_t* d_lens,
+                     const int64_t* d_offs, char* d_data) {
+    int blocks = (int)std::min<int64_t>((n + BLOCK - 1) / BLOCK, 2048);
+    if (blocks < 1) blocks = 1;
+    if (d_lens)
+        hipLaunchKernelGGL(k_gen_json_lens, dim3(blocks), dim3(BLOCK), 0, s,
+                           seed, t0, start_row, n, nkeys, rows_per_ms, d_lens);
+    if (d_data)
+        hipLaunchKernelGGL(k_gen_json_fill, dim3(blocks), dim3(BLOCK), 0, s,
+                           seed, t0, start_row, n, nkeys, rows_per_ms, d_offs,
+                           d_data);
+}
+
+} // namespace dz

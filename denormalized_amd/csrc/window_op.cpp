@@ -2606,3 +2606,205 @@ extern "C" dz_status dz_join_op_matches(dz_join_op* op, int64_t* n_out,
 extern "C" int64_t dz_join_op_unmatched(dz_join_op* op) {
     return op ? op->u_n : 0;
 }
+
+/* ------------------------------------------------------------------ */
+/* JSON ingest decoder: device-side replacement for the  */
+/* reference's serde_json decode of Kafka payload bytes                */
+/* (formats/decoders/json.rs:23-46 via kafka_stream_read.rs:165-296).  */
+/* Newline-delimited records -> (ts int64, utf8 key column, f64 value) */
+/* device columns shaped for dz_window_op_push_device_utf8. Documented */
+/* subset (flagged loudly otherwise): no escape sequences inside the   */
+/* schema fields' strings/names, numeric literals within the exact     */
+/* Clinger fast path (<=15 significant digits, |decimal exp| <= 22).   */
+/* ------------------------------------------------------------------ */
+
+struct dz_json_decoder {
+    int device = 0;
+    std::string err;
+    hipStream_t stream = nullptr;
+    dz::JsonFields jf;
+    uint32_t* d_cnt = nullptr;   /* [512] + base [512] + tot [1] + ktot [1] */
+    uint32_t* h_tot = nullptr;   /* pinned */
+    uint32_t* d_dbg = nullptr;
+    uint32_t* d_blocksum = nullptr;
+    int64_t bs_cap = 0;
+    /* per-decode outputs, double-buffered (valid until the second-next
+     * decode: a utf8 borrowed push consumes them across one step) */
+    int64_t* d_recoff[2] = {};
+    int64_t* o_ts[2] = {};
+    int64_t* o_kbeg[2] = {};
+    int32_t* o_klen[2] = {};
+    double* o_val[2] = {};
+    int32_t* o_koff[2] = {};
+    char* o_kdata[2] = {};
+    int64_t rec_cap = 0, kdata_cap = 0;
+    int cur = 0;
+    int64_t n_rec = 0;
+};
+
+extern "C" dz_json_decoder* dz_json_decoder_create(int32_t device,
+                                                   const char* ts_field,
+                                                   const char* key_field,
+                                                   const char* val_field) {
+    g_err.clear();
+    int ndev = 0;
+    if (hipGetDeviceCount(&ndev) != hipSuccess || ndev <= device) {
+        g_err = "no HIP device available (this operator has no CPU fallback)";
+        return nullptr;
+    }
+    if (!ts_field || !key_field || !val_field ||
+        strlen(ts_field) > 31 || strlen(key_field) > 31 ||
+        strlen(val_field) > 31) {
+        g_err = "field names must be 1..31 bytes";
+        return nullptr;
+    }
+    auto* d = new dz_json_decoder();
+    d->device = device;
+    snprintf(d->jf.ts_name, 32, "%s", ts_field);
+    snprintf(d->jf.key_name, 32, "%s", key_field);
+    snprintf(d->jf.val_name, 32, "%s", val_field);
+    d->jf.ts_len = (int32_t)strlen(ts_field);
+    d->jf.key_len = (int32_t)strlen(key_field);
+    d->jf.val_len = (int32_t)strlen(val_field);
+    if (hipSetDevice(device) != hipSuccess ||
+        hipStreamCreate(&d->stream) != hipSuccess ||
+        hipMalloc(&d->d_cnt, (512 + 512 + 2) * 4) != hipSuccess ||
+        hipMalloc(&d->d_dbg, 16) != hipSuccess ||
+        hipHostMalloc((void**)&d->h_tot, 8) != hipSuccess) {
+        g_err = "json decoder allocation failed";
+        delete d;
+        return nullptr;
+    }
+    hipMemset(d->d_dbg, 0, 16);
+    return d;
+}
+
+extern "C" void dz_json_decoder_destroy(dz_json_decoder* d) {
+    if (!d) return;
+    hipSetDevice(d->device);
+    if (d->stream) hipStreamSynchronize(d->stream);
+    hipFree(d->d_cnt); hipFree(d->d_dbg); hipFree(d->d_blocksum);
+    for (int i = 0; i < 2; i++) {
+        hipFree(d->d_recoff[i]); hipFree(d->o_ts[i]); hipFree(d->o_kbeg[i]);
+        hipFree(d->o_klen[i]); hipFree(d->o_val[i]); hipFree(d->o_koff[i]);
+        hipFree(d->o_kdata[i]);
+    }
+    if (d->h_tot) hipHostFree(d->h_tot);
+    if (d->stream) hipStreamDestroy(d->stream);
+    delete d;
+}
+
+extern "C" const char* dz_json_decoder_last_error(dz_json_decoder* d) {
+    if (!d) return g_err.empty() ? nullptr : g_err.c_str();
+    return d->err.empty() ? nullptr : d->err.c_str();
+}
+
+#define DCHK(op, call)                                                     \
+    do {                                                                   \
+        hipError_t e_ = (call);                                            \
+        if (e_ != hipSuccess) {                                            \
+            (op)->err = std::string(#call) + ": " + hipGetErrorString(e_); \
+            return DZ_ERR;                                                 \
+        }                                                                  \
+    } while (0)
+
+extern "C" dz_status dz_json_decode(dz_json_decoder* d, const char* d_bytes,
+                                    int64_t n_bytes) {
+    if (!d) return DZ_ERR;
+    DCHK(d, hipSetDevice(d->device));
+    d->n_rec = 0;
+    if (n_bytes <= 0) return DZ_OK;
+    int C = (int)std::min<int64_t>(512,
+                                   std::max<int64_t>(1, (n_bytes + 8191) / 8192));
+    int64_t chunk = (n_bytes + C - 1) / C;
+    /* tail byte decides whether bytes after the last newline are a record */
+    char tail = 0;
+    DCHK(d, hipMemcpy(&tail, d_bytes + n_bytes - 1, 1, hipMemcpyDeviceToHost));
+    dz::launch_json_count(d->stream, d_bytes, n_bytes, C, chunk,
+                          tail != '\n' ? 1 : 0, d->d_cnt, d->d_cnt + 512,
+                          d->d_cnt + 1024);
+    DCHK(d, hipMemcpyAsync(d->h_tot, d->d_cnt + 1024, 4,
+                           hipMemcpyDeviceToHost, d->stream));
+    DCHK(d, hipStreamSynchronize(d->stream));
+    const int64_t nrec = d->h_tot[0];
+    if (nrec == 0) return DZ_OK;
+    if (nrec > d->rec_cap || n_bytes > d->kdata_cap) {
+        DCHK(d, hipStreamSynchronize(d->stream));
+        int64_t rc = std::max(nrec, d->rec_cap);
+        int64_t kc = std::max(n_bytes, d->kdata_cap);
+        for (int i = 0; i < 2; i++) {
+            hipFree(d->d_recoff[i]); hipFree(d->o_ts[i]); hipFree(d->o_kbeg[i]);
+            hipFree(d->o_klen[i]); hipFree(d->o_val[i]); hipFree(d->o_koff[i]);
+            hipFree(d->o_kdata[i]);
+            DCHK(d, hipMalloc(&d->d_recoff[i], (size_t)rc * 8));
+            DCHK(d, hipMalloc(&d->o_ts[i], (size_t)rc * 8));
+            DCHK(d, hipMalloc(&d->o_kbeg[i], (size_t)rc * 8));
+            DCHK(d, hipMalloc(&d->o_klen[i], (size_t)rc * 4));
+            DCHK(d, hipMalloc(&d->o_val[i], (size_t)rc * 8));
+            DCHK(d, hipMalloc(&d->o_koff[i], ((size_t)rc + 1) * 4));
+            DCHK(d, hipMalloc(&d->o_kdata[i], (size_t)kc));
+        }
+        d->rec_cap = rc;
+        d->kdata_cap = kc;
+    }
+    const int64_t nb = (nrec + 4095) / 4096;
+    if (nb > d->bs_cap) {
+        hipFree(d->d_blocksum);
+        DCHK(d, hipMalloc(&d->d_blocksum, (size_t)nb * 4));
+        d->bs_cap = nb;
+    }
+    const int b = d->cur ^ 1;
+    dz::launch_json_parse(d->stream, d_bytes, n_bytes, C, chunk, d->jf,
+                          d->d_cnt + 512, d->d_cnt + 1024, d->d_recoff[b],
+                          nrec, d->o_ts[b], d->o_kbeg[b], d->o_klen[b],
+                          d->o_val[b], d->d_blocksum, d->d_cnt + 1025,
+                          d->o_koff[b], d->o_kdata[b], d->d_dbg);
+    DCHK(d, hipStreamSynchronize(d->stream));
+    uint32_t cells[4] = {0, 0, 0, 0};
+    DCHK(d, hipMemcpy(cells, d->d_dbg, 16, hipMemcpyDeviceToHost));
+    if (cells[2]) {
+        d->err = cells[2] == 2
+                     ? "numeric literal outside the exact parse subset "
+                       "(<=15 significant digits, |decimal exponent| <= 22)"
+                     : (cells[2] == 1 ? "malformed JSON record"
+                                      : "record missing a schema field");
+        return DZ_ERR;
+    }
+    d->cur = b;
+    d->n_rec = nrec;
+    return DZ_OK;
+}
+
+/* the LAST decode's columns (device pointers, valid until the second-next
+ * decode) — exactly the shape dz_window_op_push_device_utf8 consumes */
+extern "C" dz_status dz_json_decoder_batch(dz_json_decoder* d, int64_t* n_out,
+                                           const int64_t** d_ts,
+                                           const int32_t** d_key_offsets,
+                                           const char** d_key_data,
+                                           const double** d_vals) {
+    if (!d || !n_out) return DZ_ERR;
+    *n_out = d->n_rec;
+    if (d_ts) *d_ts = d->o_ts[d->cur];
+    if (d_key_offsets) *d_key_offsets = d->o_koff[d->cur];
+    if (d_key_data) *d_key_data = d->o_kdata[d->cur];
+    if (d_vals) *d_vals = d->o_val[d->cur];
+    return DZ_OK;
+}
+
+extern "C" dz_status dz_generate_json(int32_t device, uint64_t seed,
+                                      int64_t t0_ms, int64_t start_row,
+                                      int64_t n_rows, int64_t n_keys,
+                                      int64_t rows_per_ms, int32_t* d_lens,
+                                      const int64_t* d_offsets, char* d_data) {
+    if (hipSetDevice(device) != hipSuccess) {
+        g_err = "hipSetDevice failed";
+        return DZ_ERR;
+    }
+    dz::launch_gen_json(nullptr, seed, t0_ms, start_row, n_rows, n_keys,
+                        rows_per_ms, d_lens, d_offsets, d_data);
+    if (hipGetLastError() != hipSuccess) {
+        g_err = "dz_generate_json launch failed";
+        return DZ_ERR;
+    }
+    return DZ_OK;
+}

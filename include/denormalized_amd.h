@@ -261,6 +261,50 @@ dz_status dz_join_op_matches(dz_join_op* op, int64_t* n_out,
 int64_t dz_join_op_unmatched(dz_join_op* op);
 
 /* ------------------------------------------------------------------ */
+/* JSON ingest (from_topic's decode stage): the reference decodes Kafka
+ * payload bytes on the host with serde_json
+ * (crates/core/src/formats/decoders/json.rs:23-46, driven by
+ * kafka_stream_read.rs:165-296); here the decode runs ON DEVICE over
+ * newline-delimited records, producing (int64 ts, utf8 key column, f64
+ * value) device columns shaped for dz_window_op_push_device_utf8 — the
+ * on-wire-bytes -> windowed-aggregate pipeline stays in HBM end to end.
+ * Documented subset (anything else fails loudly): records are non-empty
+ * '\n'-separated JSON objects; the three schema fields are top-level
+ * (other fields, including nested objects/arrays, are skipped); no escape
+ * sequences inside schema field names or the key string; numeric literals
+ * within the exact Clinger fast path (<= 15 significant digits, |decimal
+ * exponent| <= 22 — equal to strtod bit-for-bit there). */
+
+typedef struct dz_json_decoder dz_json_decoder;
+
+dz_json_decoder* dz_json_decoder_create(int32_t device, const char* ts_field,
+                                        const char* key_field,
+                                        const char* val_field);
+void dz_json_decoder_destroy(dz_json_decoder* d);
+const char* dz_json_decoder_last_error(dz_json_decoder* d);
+
+/* Decode one device-resident byte buffer. Columns are retrieved with
+ * dz_json_decoder_batch and stay valid until the SECOND-next decode
+ * (double-buffered, covering a utf8 borrowed push across one step —
+ * provided batch sizes do not grow, which reallocates). */
+dz_status dz_json_decode(dz_json_decoder* d, const char* d_bytes,
+                         int64_t n_bytes);
+dz_status dz_json_decoder_batch(dz_json_decoder* d, int64_t* n_out,
+                                const int64_t** d_ts_ms,
+                                const int32_t** d_key_offsets,
+                                const char** d_key_data,
+                                const double** d_vals);
+
+/* Synthetic on-wire JSON generator (bench/test input): one newline-
+ * delimited record per row of the same seeded sensor stream, reading
+ * printed as fixed 6-decimal (inside the decoder's exact subset).
+ * Lens pass (host cumsums into int64 offsets) then fill pass. */
+dz_status dz_generate_json(int32_t device, uint64_t seed, int64_t t0_ms,
+                           int64_t start_row, int64_t n_rows, int64_t n_keys,
+                           int64_t rows_per_ms, int32_t* d_lens,
+                           const int64_t* d_offsets, char* d_data);
+
+/* ------------------------------------------------------------------ */
 /* Synthetic sensor stream generator, on device (bench/test input; spec in
  * DESIGN.md §Generator, bit-identical to oracle orc_gen; mirrors
  * examples/examples/emit_measurements.rs:30-67). Any output pointer may be
