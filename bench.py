@@ -66,6 +66,12 @@ def parse_args():
                         "join()+window(). Per-GPU shard; rows all match "
                         "(build side resident before the timed region).")
     p.add_argument("--trips", type=int, default=1_000_000)
+    p.add_argument("--ingest", action="store_true",
+                   help="on-wire pipeline: pre-generated newline-delimited "
+                        "JSON bytes (device-resident) -> device JSON decode "
+                        "-> device utf8 intern -> window()+filter(); value = "
+                        "decoded rows/s (the f4 ingest row; config reports "
+                        "bytes/row so bytes/s = value * bytes_per_row)")
     p.add_argument("--dist-backend", default=None,
                    help="torch.distributed backend override (default: nccl on GPU)")
     p.add_argument("--staged-push", action="store_true",
@@ -169,7 +175,9 @@ def main():
     # pre-generate the whole stream into HBM (inputs resident when the timed
     # region starts). Each rank owns a disjoint key shard (weak scaling):
     # same time distribution, rank-distinct keys (ids are rank-local).
-    utf8 = args.key_kind == "utf8" and not args.cfg5
+    if args.ingest and args.cfg5:
+        raise SystemExit("--ingest and --cfg5 are separate workloads")
+    utf8 = args.key_kind == "utf8" and not args.cfg5 and not args.ingest
     gseed = args.seed + 1000003 * rank  # rank-distinct draws per shard
     d_ts = dz.DeviceArray(device, total_rows * 8)
     d_kid = None if (utf8 or args.cfg5) else dz.DeviceArray(device,
@@ -198,6 +206,37 @@ def main():
         d_btr.free()
         d_bdr.free()
     d_offs = d_data = None
+    decoder = None
+    json_base = []
+    if args.ingest:
+        # pre-generate the on-wire JSON stream into HBM, one slice per step
+        d_lens = dz.DeviceArray(device, B * 4)
+        d_joffs = dz.DeviceArray(device, (B + 1) * 8)
+        total_bytes = 0
+        step_lens = []
+        for s_ in range(K + W):
+            dz.generate_json(device, gseed, 1_000_000, s_ * B, B, args.keys,
+                             args.rows_per_ms, d_lens=d_lens.ptr)
+            lens = d_lens.to_host(np.int32, B)
+            json_base.append(total_bytes)
+            step_lens.append(lens)
+            total_bytes += int(lens.sum())
+        d_data = dz.DeviceArray(device, total_bytes)
+        L = _lib.lib()
+        for s_ in range(K + W):
+            offs = np.zeros(B + 1, np.int64)
+            np.cumsum(step_lens[s_], out=offs[1:])
+            step_lens[s_] = int(offs[-1])  # slice byte length
+            d_joffs.from_host(offs)
+            dz.generate_json(device, gseed, 1_000_000, s_ * B, B, args.keys,
+                             args.rows_per_ms, d_offsets=d_joffs.ptr,
+                             d_data=ctypes.c_void_p(
+                                 d_data.ptr.value + json_base[s_]))
+        json_lens = step_lens
+        d_lens.free()
+        d_joffs.free()
+        decoder = dz.JsonDecoder(device=device)
+        json_bytes_total = total_bytes
     step_data_base = []
     if utf8:
         # utf8 key column, per-step Arrow slices: lens on device, host cumsum
@@ -233,13 +272,21 @@ def main():
 
     op = dz.WindowOp(length_ms=args.window_ms, slide_ms=args.slide_ms,
                      aggs=[("count", 0), ("min", 0), ("max", 0), ("avg", 0)],
-                     key_kind=_lib.KEY_UTF8 if utf8 else _lib.KEY_DENSE_INT64,
+                     key_kind=(_lib.KEY_UTF8 if (utf8 or args.ingest)
+                               else _lib.KEY_DENSE_INT64),
                      n_keys_hint=args.keys, device=device)
     if not args.no_filter:
         op.set_filter("max", ">", 113.0)
 
     def do_push(step):
         off = step * B
+        if args.ingest:
+            # on-wire bytes -> decode -> intern -> window, all on device
+            decoder.decode(ctypes.c_void_p(d_data.ptr.value + json_base[step]),
+                           json_lens[step])
+            nr, pts, pko, pkd, pv = decoder.batch()
+            op.push_device_utf8(nr, pts, pko, pkd, pv)
+            return
         if args.cfg5:
             # join -> window, all on device: probe the build table, feed the
             # matched (ts, driver kid, value) columns to the window op
@@ -392,7 +439,9 @@ def main():
             path["pmc_frac"] = pmc_bytes_per_step / step_s / HBM_PEAK
             path["amplification"] = pmc_bytes_per_step / (ALG_BYTES_PER_ROW * B)
         out = {
-            "metric": ("probe rows/sec through join()+window() on synthetic "
+            "metric": ("rows/sec from on-wire JSON bytes through "
+                       "decode()+window()+filter()" if args.ingest else
+                       "probe rows/sec through join()+window() on synthetic "
                        "rideshare stream" if args.cfg5 else
                        "rows/sec through window()+filter() on synthetic "
                        "sensor stream"),
@@ -408,7 +457,11 @@ def main():
             "dtype": "f64",
             "data": "synthetic",
             "config": {
-                "workload": ("cfg5 (per-GPU shard): join(trip_id, "
+                "workload": (f"f4 ingest: on-wire JSON "
+                             f"({json_bytes_total / (K + W) / B:.1f} B/row) "
+                             "-> device decode -> device intern -> window, "
+                             if args.ingest else
+                             "cfg5 (per-GPU shard): join(trip_id, "
                              f"{args.trips} trips)->window group-by driver, "
                              if args.cfg5 else
                              (("cfg2-utf8 (as written, device-interned "
@@ -458,6 +511,8 @@ def main():
     op.close()
     if join is not None:
         join.close()
+    if decoder is not None:
+        decoder.close()
     for a in (d_ts, d_kid, d_vals, d_offs, d_data, d_trips):
         if a is not None:
             a.free()
