@@ -2035,7 +2035,9 @@ void launch_json_parse(hipStream_t s, const char* d_buf, int64_t nbytes,
                        char* o_kdata, uint32_t* d_dbg) {
     hipLaunchKernelGGL(k_json_nl_emit, dim3(C), dim3(64), 0, s, d_buf, nbytes,
                        chunk, d_base, d_tot, d_recoff);
-    int pblocks = (int)std::min<int64_t>((nrec + BLOCK - 1) / BLOCK, 2048);
+    /* the per-record interpreter is latency/divergence-bound: give it a
+     * deep grid so each thread owns ~1 record */
+    int pblocks = (int)std::min<int64_t>((nrec + BLOCK - 1) / BLOCK, 32768);
     if (pblocks < 1) pblocks = 1;
     hipLaunchKernelGGL(k_json_parse, dim3(pblocks), dim3(BLOCK), 0, s, d_buf,
                        d_recoff, d_tot, nbytes, jf, o_ts, o_kbeg, o_klen,
