@@ -131,7 +131,8 @@ void launch_emission_slabread(hipStream_t stream, const uint64_t* slab_first,
                               uint8_t* oflags);
 void launch_emission_sort(hipStream_t stream, int64_t K, uint64_t* fkeys,
                           uint64_t* skeys, uint32_t* fiota, uint32_t* okid,
-                          uint32_t* counter2, uint32_t* rhist, uint32_t* roffs);
+                          uint32_t* counter2, uint32_t* rhist, uint32_t* roffs,
+                          uint64_t max_key /* host-known bound on `first` */);
 constexpr int EMIT_RCHUNK = 4096;
 constexpr int EMIT_RBINS = 2048;
 

@@ -1339,15 +1339,24 @@ void launch_emission_slabread(hipStream_t s, const uint64_t* slab_first,
 
 void launch_emission_sort(hipStream_t s, int64_t K, uint64_t* fkeys,
                           uint64_t* skeys, uint32_t* fiota, uint32_t* okid,
-                          uint32_t* counter2, uint32_t* rhist, uint32_t* roffs) {
+                          uint32_t* counter2, uint32_t* rhist, uint32_t* roffs,
+                          uint64_t max_key) {
     uint64_t* ekeys = fkeys;
     uint32_t* skid = fiota;
     uint32_t* counter = counter2;
     /* sort (first, compact-index) pairs: keys ekeys<->skeys, payload
-     * skid<->okid; RPASSES is even, so the sorted payload lands in skid.
+     * skid<->okid; the pass count covers the HOST-KNOWN key bound
+     * (first = batch_seq<<32 | rowidx, so any run under 2048 batches
+     * needs 4 passes, not the 66-bit worst case of 6), rounded up to
+     * EVEN so the sorted payload lands in skid.
      * (The device path only runs above the 64k-key host-emission cutoff,
      * so the multi-block form is always the right one.) */
     {
+        int passes = 0;
+        while (passes * RDIG < 64 && (max_key >> (passes * RDIG)) != 0)
+            passes++;
+        if (passes & 1) passes++;
+        if (passes < 2) passes = 2;
         int nblk = (int)((K + RCHUNK - 1) / RCHUNK);
         uint64_t* ka = ekeys;
         uint32_t* pa = skid;
@@ -1356,7 +1365,7 @@ void launch_emission_sort(hipStream_t s, int64_t K, uint64_t* fkeys,
         int bs = (nblk + RSEG - 1) / RSEG;
         uint32_t* psum = rhist + (int64_t)nblk * RBINS;  /* scratch tail */
         uint32_t* dbase = psum + (int64_t)RSEG * RBINS;
-        for (int p = 0; p < RPASSES; p++) {
+        for (int p = 0; p < passes; p++) {
             int shift = p * RDIG;
             hipLaunchKernelGGL(k_rhist, dim3(nblk), dim3(BLOCK), 0, s, ka,
                                counter, shift, rhist);

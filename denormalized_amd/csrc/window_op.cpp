@@ -1338,7 +1338,8 @@ static dz_status trigger_windows(dz_window_op* op) {
         if (dev_path && op->n_keys > 0) {
             dz::launch_emission_sort(cs, op->n_keys, d.fkeys,
                                      d.skeys, d.fiota, d.okid, d.counter + 1,
-                                     op->d_rhist[csi], op->d_roffs[csi]);
+                                     op->d_rhist[csi], op->d_roffs[csi],
+                                     (((uint64_t)op->batch_seq + 1) << 32));
             /* pack the final-order columns on device: the worker pulls one
              * contiguous span and builds with sequential copies */
             dz::launch_emission_permute(cs, op->n_keys, d.counter + 1,
