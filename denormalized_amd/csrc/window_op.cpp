@@ -1657,88 +1657,117 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
         return DZ_ERR;
     }
     if (ensure_scratch(op, C, nrec_max) != DZ_OK) return DZ_ERR;
-    if (sliding) {
-        timed(op, "hist", (double)n * 12, [&] {
-            dz::launch_hist(op->stream, d_kid, d_ts, n, chunk, C, wp,
-                            op->d_ghist[b], nullptr);
-        });
-        timed(op, "scan", (double)C * dz::NB * 12, [&] {
-            dz::launch_scan(op->stream, op->d_ghist[b], C, op->d_total[b],
-                            op->d_base[b], op->d_gofs[b]);
-        });
-    }
     /* supertile row budget: st_rows * expand staged records must fit the
      * ST_RECORDS LDS staging (a 64-row floor overflowed it for window/hop
      * ratios above 32 — found by the randomized deep matrix) */
     int32_t st_rows = (int32_t)std::max<int64_t>(1, dz::ST_RECORDS / expand);
-    timed(op, "scatter", (double)n * 24 + (double)nrec_max * 20, [&] {
-        dz::launch_scatter(op->stream, d_kid, d_ts, d_vals, d_valbm, n, chunk, C,
-                           st_rows, wp, op->d_gofs[b], op->d_grec,
-                           (uint32_t)op->rec_cap, op->d_dbg);
-    });
-    int64_t klocs = op->kcap >> dz::LOG_NB;
-    int64_t gtot = klocs * nw;
-    int64_t khigh = (klocs + 255) >> 8;
-    int64_t nb1 = khigh * nw;
-    if (gtot > dz::FOLD_GCAP && nb1 <= 256) {
-        /* two-level regime (large keyspaces, e.g. cfg3's 1M keys): split by
-         * (kloc>>8, widx), then the second pass splits each bin1 by kloc&255
-         * AND folds the staged bins in place (fused — the split-out records
-         * and the fold's re-read of them never touch HBM) */
-        if (!op->d_b1offs) {
-            CHK(op, hipMalloc(&op->d_b1offs, (size_t)dz::NB * 256 * 4));
-            CHK(op, hipMalloc(&op->d_b1lens, (size_t)dz::NB * 256 * 4));
+    const int64_t klocs = op->kcap >> dz::LOG_NB;
+    const int64_t khigh = (klocs + 255) >> 8;
+    /* Window-subrange self-split: when the batch spans so many windows that
+     * neither the direct fold (klocs*nw > GCAP) nor the two-level split
+     * (khigh*nw > 256) can cover them in one pass, process the batch in
+     * subranges of <= 256/khigh windows each — the partition re-reads the
+     * INPUT per subrange (36 B/row) instead of re-reading the expanded
+     * record set per (k,w) chunk, which measured 92% of cfg3's GPU time at
+     * 8-second batches (85 windows/batch). A tumbling subrange runs as
+     * sliding with slide == length: identical window membership PLUS the
+     * sliding path's range clamping (rows outside the subrange get
+     * multiplicity 0). */
+    int64_t wsub = nw;
+    if (klocs * nw > dz::FOLD_GCAP && khigh * nw > 256)
+        wsub = std::max<int64_t>(1, 256 / khigh);
+    const bool split = wsub < nw;
+    for (int64_t W0 = 0; W0 < nw; W0 += wsub) {
+        const int64_t nws = std::min<int64_t>(wsub, nw - W0);
+        dz::WinParams wps = wp;
+        wps.s0 = ws[W0];
+        wps.nw = (int32_t)nws;
+        if (split && !sliding) {
+            wps.is_sliding = 1;
+            wps.slide_ms = wp.len_ms;
         }
-        if (nrec_max > op->l2_cap) {
-            hipFree(op->d_grec2);
-            CHK(op, hipMalloc(&op->d_grec2, (size_t)nrec_max * 16));
-            op->l2_cap = nrec_max;
+        if (sliding || split) {
+            /* subrange (or sliding) reduction on the compute stream; the
+             * full-range tumbling histogram was fused into the ingest
+             * reduction and is already in d_gofs[b] */
+            timed(op, "hist", (double)n * 12, [&] {
+                dz::launch_hist(op->stream, d_kid, d_ts, n, chunk, C, wps,
+                                op->d_ghist[b], nullptr);
+            });
+            timed(op, "scan", (double)C * dz::NB * 12, [&] {
+                dz::launch_scan(op->stream, op->d_ghist[b], C, op->d_total[b],
+                                op->d_base[b], op->d_gofs[b]);
+            });
         }
-        dz::FoldChunk fc;
-        fc.w_lo = 0;
-        fc.w_hi = (int32_t)nw;
-        fc.k_lo = 0;
-        fc.k_hi = (int32_t)klocs;
-        fc.kcap = op->kcap;
-        fc.batch_seq = op->batch_seq;
-        fc.bin_stride = (int32_t)(nb1 * 256);
-        fc.tl_nw = (int32_t)nw;
-        timed(op, "regroup", (double)nrec_max * 40, [&] {
-            dz::launch_regroup_l1(op->stream, op->d_grec, op->d_base[b],
-                                  fc, op->d_b1offs, op->d_b1lens,
-                                  op->d_grec2);
+        timed(op, "scatter", (double)n * 24 + (double)nrec_max * 20, [&] {
+            dz::launch_scatter(op->stream, d_kid, d_ts, d_vals, d_valbm, n,
+                               chunk, C, st_rows, wps, op->d_gofs[b],
+                               op->d_grec, (uint32_t)op->rec_cap, op->d_dbg);
         });
-        timed(op, "regfold", (double)nrec_max * 24, [&] {
-            dz::launch_regroup_l2_fold(op->stream, op->d_grec2,
-                                       op->d_base[b], fc, (int)nb1,
-                                       op->d_b1offs, op->d_b1lens,
-                                       op->d_slotmap, op->s_cnt, op->s_min,
-                                       op->s_max, op->s_sum, op->s_first,
-                                       (int64_t)op->nslots * 5 * op->kcap,
-                                       op->d_dbg);
-        });
-    } else {
-        for (int64_t k_lo = 0; k_lo < klocs; k_lo += dz::FOLD_GCAP) {
-            int32_t nk = (int32_t)std::min<int64_t>(dz::FOLD_GCAP, klocs - k_lo);
-            int32_t wstep = dz::FOLD_GCAP / nk;
-            for (int64_t w_lo = 0; w_lo < nw; w_lo += wstep) {
-                dz::FoldChunk fc;
-                fc.w_lo = (int32_t)w_lo;
-                fc.w_hi = (int32_t)std::min<int64_t>(nw, w_lo + wstep);
-                fc.k_lo = (int32_t)k_lo;
-                fc.k_hi = (int32_t)(k_lo + nk);
-                fc.kcap = op->kcap;
-                fc.batch_seq = op->batch_seq;
-                fc.bin_stride = dz::FOLD_GCAP;
-                fc.tl_nw = 0;
-                timed(op, "regfold", (double)nrec_max * 24, [&] {
-                    dz::launch_regroup_fold(op->stream, op->d_grec,
-                                            op->d_base[b], fc, op->d_slotmap,
-                                            op->s_cnt, op->s_min, op->s_max,
-                                            op->s_sum, op->s_first,
-                                            (int64_t)op->nslots * 5 * op->kcap,
-                                            op->d_dbg);
-                });
+        const int64_t gtot = klocs * nws;
+        const int64_t nb1 = khigh * nws;
+        if (gtot > dz::FOLD_GCAP) {
+            /* two-level regime (large keyspaces, e.g. cfg3's 1M keys):
+             * split by (kloc>>8, widx), then the second pass splits each
+             * bin1 by kloc&255 AND folds the staged bins in place (fused —
+             * the split-out records and the fold's re-read of them never
+             * touch HBM). nb1 <= 256 by the subrange construction. */
+            if (!op->d_b1offs) {
+                CHK(op, hipMalloc(&op->d_b1offs, (size_t)dz::NB * 256 * 4));
+                CHK(op, hipMalloc(&op->d_b1lens, (size_t)dz::NB * 256 * 4));
+            }
+            if (nrec_max > op->l2_cap) {
+                hipFree(op->d_grec2);
+                CHK(op, hipMalloc(&op->d_grec2, (size_t)nrec_max * 16));
+                op->l2_cap = nrec_max;
+            }
+            dz::FoldChunk fc;
+            fc.w_lo = 0;
+            fc.w_hi = (int32_t)nws;
+            fc.k_lo = 0;
+            fc.k_hi = (int32_t)klocs;
+            fc.kcap = op->kcap;
+            fc.batch_seq = op->batch_seq;
+            fc.bin_stride = (int32_t)(nb1 * 256);
+            fc.tl_nw = (int32_t)nws;
+            timed(op, "regroup", (double)nrec_max * 40, [&] {
+                dz::launch_regroup_l1(op->stream, op->d_grec, op->d_base[b],
+                                      fc, op->d_b1offs, op->d_b1lens,
+                                      op->d_grec2);
+            });
+            timed(op, "regfold", (double)nrec_max * 24, [&] {
+                dz::launch_regroup_l2_fold(op->stream, op->d_grec2,
+                                           op->d_base[b], fc, (int)nb1,
+                                           op->d_b1offs, op->d_b1lens,
+                                           op->d_slotmap + W0, op->s_cnt,
+                                           op->s_min, op->s_max, op->s_sum,
+                                           op->s_first,
+                                           (int64_t)op->nslots * 5 * op->kcap,
+                                           op->d_dbg);
+            });
+        } else {
+            for (int64_t k_lo = 0; k_lo < klocs; k_lo += dz::FOLD_GCAP) {
+                int32_t nk = (int32_t)std::min<int64_t>(dz::FOLD_GCAP,
+                                                        klocs - k_lo);
+                int32_t wstep = dz::FOLD_GCAP / nk;
+                for (int64_t w_lo = 0; w_lo < nws; w_lo += wstep) {
+                    dz::FoldChunk fc;
+                    fc.w_lo = (int32_t)w_lo;
+                    fc.w_hi = (int32_t)std::min<int64_t>(nws, w_lo + wstep);
+                    fc.k_lo = (int32_t)k_lo;
+                    fc.k_hi = (int32_t)(k_lo + nk);
+                    fc.kcap = op->kcap;
+                    fc.batch_seq = op->batch_seq;
+                    fc.bin_stride = dz::FOLD_GCAP;
+                    fc.tl_nw = 0;
+                    timed(op, "regfold", (double)nrec_max * 24, [&] {
+                        dz::launch_regroup_fold(
+                            op->stream, op->d_grec, op->d_base[b], fc,
+                            op->d_slotmap + W0, op->s_cnt, op->s_min,
+                            op->s_max, op->s_sum, op->s_first,
+                            (int64_t)op->nslots * 5 * op->kcap, op->d_dbg);
+                    });
+                }
             }
         }
     }

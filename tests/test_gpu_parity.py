@@ -1175,3 +1175,44 @@ def test_nan_first_value_sticks():
     assert np.isnan(outs[0]["min"][0]) and np.isnan(outs[0]["max"][0])
     assert np.isnan(outs[0]["avg"][0])
     op.close()
+
+
+# ------------------------------------------- window-subrange self-split path
+
+def test_many_windows_sliding_subrange_split():
+    # a batch spanning >256 sliding windows (khigh*nw > 256) takes the
+    # window-subrange self-split; parity must be unchanged
+    rng = np.random.default_rng(4242)
+    n = 50_000
+    ts = (1_000_000 + np.arange(n) // 25).astype(np.int64)  # 2s span
+    k = rng.integers(0, 600, n)
+    v = rng.uniform(0, 115, n)
+    outs, exp = run_both(200, 5, [(ts, k, v)], n_keys_hint=600)
+    assert len(exp["key"]) > 0
+    assert_parity(outs, exp)
+
+
+def test_many_windows_tumbling_subrange_split():
+    # tumbling split: the subrange runs as sliding-with-slide==length
+    # (identical membership + clamping); bit-exact
+    rng = np.random.default_rng(4243)
+    n = 80_000
+    ts = (1_000_000 + np.arange(n) // 10).astype(np.int64)  # 8s span
+    k = rng.integers(0, 2000, n)
+    v = rng.uniform(0, 115, n)
+    outs, exp = run_both(20, 0, [(ts, k, v)], n_keys_hint=2000)
+    assert len(exp["key"]) > 400
+    assert_parity(outs, exp)
+
+
+def test_many_windows_split_multibatch_shuffled():
+    rng = np.random.default_rng(4244)
+    batches = []
+    for b in range(3):
+        n = 40_000
+        ts = (1_000_000 + b * 1500 + rng.integers(0, 2500, n)).astype(np.int64)
+        k = rng.integers(0, 900, n)
+        v = rng.uniform(-5, 115, n)
+        batches.append((ts, k, v))
+    outs, exp = run_both(150, 10, batches, n_keys_hint=900)
+    assert_parity(outs, exp)
