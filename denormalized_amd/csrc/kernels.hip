@@ -799,9 +799,13 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint4* rrec,
     };
 
     /* fold modes: thread t owns bin t — accumulator lives in registers,
-     * seeded from the persistent slab exactly like the standalone fold */
+     * seeded LAZILY from the persistent slab at the bin's first record and
+     * written back only if touched: untouched bins cost ZERO state traffic.
+     * (Eager seed/writeback of every bin measured ~1 KB/row of pure state
+     * traffic on cfg3's sliding 1M-key workload, where each launch covers
+     * many windows whose key slots are mostly absent from the batch.) */
     int64_t f_sidx = 0;
-    bool f_own = false;
+    bool f_own = false, f_loaded = false;
     uint64_t f_cnt = 0, f_fst = ~0ULL;
     double f_mn = 0.0, f_mx = 0.0, f_sm = 0.0;
     if (FOLD) {
@@ -823,15 +827,6 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint4* rrec,
             if (slot < 0 || f_sidx < 0 || f_sidx + 4 * fc.kcap >= slab_cells) {
                 dbg[1] = 1; /* bounds guard: disown, never corrupt */
                 f_own = false;
-            }
-        }
-        if (f_own) {
-            f_cnt = s_cnt[f_sidx];
-            f_fst = s_first[f_sidx];
-            if (f_cnt > 0) {
-                f_mn = s_min[f_sidx];
-                f_mx = s_max[f_sidx];
-                f_sm = s_sum[f_sidx];
             }
         }
     }
@@ -949,6 +944,16 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint4* rrec,
             const uint32_t seg0 = stoffs[g];
             const uint32_t seg1 = (g + 1 < GCAP) ? stoffs[g + 1] : s_total;
             if (f_own && seg1 > seg0) {
+                if (!f_loaded) { /* lazy seed at the bin's first record */
+                    f_loaded = true;
+                    f_cnt = s_cnt[f_sidx];
+                    f_fst = s_first[f_sidx];
+                    if (f_cnt > 0) {
+                        f_mn = s_min[f_sidx];
+                        f_mx = s_max[f_sidx];
+                        f_sm = s_sum[f_sidx];
+                    }
+                }
                 if (f_fst == ~0ULL)
                     f_fst = ((uint64_t)fc.batch_seq << 32) |
                             (s_rec[seg0].z & 0x7FFFFFFFu);
@@ -980,7 +985,7 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint4* rrec,
             __syncthreads();
         }
     }
-    if (FOLD && f_own) {
+    if (FOLD && f_own && f_loaded) { /* untouched bins write nothing */
         s_cnt[f_sidx] = f_cnt;
         s_first[f_sidx] = f_fst;
         s_min[f_sidx] = f_mn;
