@@ -133,6 +133,9 @@ void launch_emission_sort(hipStream_t stream, int64_t K, uint64_t* fkeys,
                           uint64_t* skeys, uint32_t* fiota, uint32_t* okid,
                           uint32_t* counter2, uint32_t* rhist, uint32_t* roffs,
                           uint64_t max_key /* host-known bound on `first` */);
+void launch_esort_small(hipStream_t stream, uint64_t* fkeys, uint64_t* skeys,
+                        uint32_t* fiota, uint32_t* okid, uint32_t* counter2,
+                        uint64_t max_key);
 constexpr int EMIT_RCHUNK = 4096;
 constexpr int EMIT_RBINS = 2048;
 

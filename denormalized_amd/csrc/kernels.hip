@@ -2158,3 +2158,111 @@ void launch_gen_json(hipStream_t s, uint64_t seed, int64_t t0,
 }
 
 } // namespace dz
+
+/* ------------------------------------------------------------------ */
+/* small-N emission sort: ONE launch replacing the ~15-launch          */
+/* multi-block radix chain when few groups pass the filter (cfg3's     */
+/* sliding closes pass ~1-2% of 1M keys). Single block, in-kernel      */
+/* passes, stable: all waves build the digit histogram, wave 0 places  */
+/* elements in order (ballot ranks + LDS cursors). Correct for ANY nt  */
+/* — just slow when large — so an adaptive host misprediction costs    */
+/* time, never correctness.                                            */
+/* ------------------------------------------------------------------ */
+
+namespace dz {
+
+__global__ __launch_bounds__(BLOCK) void k_esort_small(uint64_t* ka,
+        uint32_t* pa, uint64_t* kb, uint32_t* pb, const uint32_t* counter2,
+        int passes /* even */) {
+    __shared__ uint32_t hist[RBINS];
+    __shared__ uint32_t scanbuf[BLOCK];
+    const uint32_t nt = *counter2;
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    for (int p = 0; p < passes; p++) {
+        const int shift = p * RDIG;
+        for (int t = threadIdx.x; t < RBINS; t += BLOCK) hist[t] = 0;
+        __syncthreads();
+        for (uint32_t i = threadIdx.x; i < nt; i += BLOCK)
+            atomicAdd(&hist[(uint32_t)(ka[i] >> shift) & (RBINS - 1)], 1u);
+        __syncthreads();
+        { /* exclusive scan of 2048 bins: 8 per thread + block scan */
+            constexpr int PER = RBINS / BLOCK;
+            uint32_t loc[PER];
+            uint32_t s = 0;
+            for (int j = 0; j < PER; j++) {
+                loc[j] = s;
+                s += hist[threadIdx.x * PER + j];
+            }
+            scanbuf[threadIdx.x] = s;
+            __syncthreads();
+            for (int o = 1; o < BLOCK; o <<= 1) {
+                uint32_t v = (threadIdx.x >= (unsigned)o)
+                                 ? scanbuf[threadIdx.x - o] : 0;
+                __syncthreads();
+                scanbuf[threadIdx.x] += v;
+                __syncthreads();
+            }
+            const uint32_t pre = threadIdx.x ? scanbuf[threadIdx.x - 1] : 0;
+            for (int j = 0; j < PER; j++)
+                hist[threadIdx.x * PER + j] = pre + loc[j];
+            __syncthreads();
+        }
+        /* stable placement by wave 0 (ballot ranks + LDS cursors) */
+        if (wave == 0) {
+            for (uint32_t i0 = 0; i0 < nt; i0 += 64) {
+                const uint32_t i = i0 + lane;
+                const bool act = i < nt;
+                const uint64_t key = act ? ka[i] : ~0ULL;
+                const uint32_t d =
+                    act ? ((uint32_t)(key >> shift) & (RBINS - 1)) : 0xFFFFu;
+                uint64_t same = ~0ULL;
+                for (int b = 0; b < RDIG; b++) {
+                    uint64_t bb = __ballot((d >> b) & 1);
+                    same &= ((d >> b) & 1) ? bb : ~bb;
+                }
+                {
+                    uint64_t bb = __ballot(act);
+                    same &= act ? bb : ~bb;
+                }
+                const uint64_t below =
+                    (lane == 63) ? ~0ULL : ((1ULL << (lane + 1)) - 1);
+                const int rank = (int)__popcll(same & below) - 1;
+                const int leader = __ffsll((unsigned long long)same) - 1;
+                const uint32_t wtot = (uint32_t)__popcll(same);
+                uint32_t pos = 0;
+                {
+                    uint32_t pre = 0;
+                    if (lane == leader && act) {
+                        pre = hist[d];
+                        hist[d] = pre + wtot;
+                    }
+                    pre = (uint32_t)__shfl((int)pre, leader);
+                    pos = pre + (uint32_t)rank;
+                }
+                if (act) {
+                    kb[pos] = key;
+                    pb[pos] = pa[i];
+                }
+            }
+        }
+        __syncthreads();
+        uint64_t* tk = ka; ka = kb; kb = tk;
+        uint32_t* tp = pa; pa = pb; pb = tp;
+    }
+}
+
+void launch_esort_small(hipStream_t s, uint64_t* fkeys, uint64_t* skeys,
+                        uint32_t* fiota, uint32_t* okid, uint32_t* counter2,
+                        uint64_t max_key) {
+    int passes = 0;
+    while (passes * RDIG < 64 && (max_key >> (passes * RDIG)) != 0) passes++;
+    if (passes & 1) passes++;
+    if (passes < 2) passes = 2;
+    /* even passes => sorted (key, payload) land back in fkeys/fiota, the
+     * same contract as the multi-block chain */
+    hipLaunchKernelGGL(k_esort_small, dim3(1), dim3(BLOCK), 0, s, fkeys,
+                       fiota, skeys, okid, counter2, passes);
+}
+
+} // namespace dz

@@ -355,6 +355,8 @@ struct dz_window_op {
     uint64_t e_ticket_next = 0;      /* next ticket to assign (e_mtx) */
     uint64_t e_ticket_pop = 0;       /* next ticket poll may emit (out_mtx) */
     std::map<uint64_t, OutBuf> e_done; /* completed out-of-order (out_mtx) */
+    std::atomic<uint32_t> e_nt_hint{0}; /* last close's passer count:
+                                          * adaptive small-sort choice */
     std::atomic<uint64_t> e_build_ns{0};
     std::atomic<uint64_t> e_builds{0};
     std::atomic<uint64_t> e_zc_builds{0};   /* zero-copy vs materializing */
@@ -989,6 +991,7 @@ static void emit_worker_main(dz_window_op* op) {
             /* copy exactly nt2 packed rows (the filter already ran on
              * device, so this is the final output volume, not the keyspace) */
             uint32_t nt = op->e_pcnt[job.slab];
+            op->e_nt_hint.store(nt, std::memory_order_relaxed);
             if (nt > 0) {
                 dz_window_op::DevEmit& d = op->e_dev[job.slab];
                 /* the device already applied the sorted permutation and
@@ -1336,10 +1339,20 @@ static dz_status trigger_windows(dz_window_op* op) {
         int csi = (int)((ci - g0) % dz_window_op::E_CSTREAMS);
         hipStream_t cs = op->c_streams[csi];
         if (dev_path && op->n_keys > 0) {
-            dz::launch_emission_sort(cs, op->n_keys, d.fkeys,
-                                     d.skeys, d.fiota, d.okid, d.counter + 1,
-                                     op->d_rhist[csi], op->d_roffs[csi],
-                                     (((uint64_t)op->batch_seq + 1) << 32));
+            /* adaptive: when the last close passed few groups (filtered
+             * sliding closes pass ~1-2%), ONE single-block launch replaces
+             * the ~15-launch multi-block radix chain — the launch enqueue
+             * itself was the cfg3 push-thread wall. A misprediction is
+             * slow, never wrong (the small kernel handles any nt). */
+            const uint64_t maxk = ((uint64_t)op->batch_seq + 1) << 32;
+            if (op->e_nt_hint.load(std::memory_order_relaxed) <= 24000)
+                dz::launch_esort_small(cs, d.fkeys, d.skeys, d.fiota, d.okid,
+                                       d.counter + 1, maxk);
+            else
+                dz::launch_emission_sort(cs, op->n_keys, d.fkeys,
+                                         d.skeys, d.fiota, d.okid,
+                                         d.counter + 1, op->d_rhist[csi],
+                                         op->d_roffs[csi], maxk);
             /* pack the final-order columns on device: the worker pulls one
              * contiguous span and builds with sequential copies */
             dz::launch_emission_permute(cs, op->n_keys, d.counter + 1,
