@@ -1345,7 +1345,12 @@ static dz_status trigger_windows(dz_window_op* op) {
              * itself was the cfg3 push-thread wall. A misprediction is
              * slow, never wrong (the small kernel handles any nt). */
             const uint64_t maxk = ((uint64_t)op->batch_seq + 1) << 32;
-            if (op->e_nt_hint.load(std::memory_order_relaxed) <= 24000)
+            static const uint32_t small_max = [] {
+                const char* v = getenv("DZ_EMIT_SMALL_MAX");
+                return v ? (uint32_t)atoi(v) : 24000u;
+            }();
+            if (small_max &&
+                op->e_nt_hint.load(std::memory_order_relaxed) <= small_max)
                 dz::launch_esort_small(cs, d.fkeys, d.skeys, d.fiota, d.okid,
                                        d.counter + 1, maxk);
             else
