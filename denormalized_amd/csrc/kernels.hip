@@ -1083,9 +1083,13 @@ __global__ __launch_bounds__(BLOCK) void k_rhist(const uint64_t* keys,
         const uint32_t* counter, int shift, uint32_t* hist) {
     __shared__ uint32_t h[RBINS];
     const uint32_t nt = *counter;
+    const uint32_t lo = blockIdx.x * RCHUNK;
+    if (lo >= nt && blockIdx.x > 0) return; /* grids are sized for the full
+        keyspace; only ceil(nt/RCHUNK) blocks hold live elements — the rest
+        must not even write zeros (every downstream kernel clamps the same
+        way), or a filtered close pays full-keyspace scan traffic */
     for (int t = threadIdx.x; t < RBINS; t += BLOCK) h[t] = 0;
     __syncthreads();
-    const uint32_t lo = blockIdx.x * RCHUNK;
     const uint32_t hi = min(nt, lo + (uint32_t)RCHUNK);
     for (uint32_t i = lo + threadIdx.x; i < hi; i += BLOCK)
         atomicAdd(&h[(uint32_t)(keys[i] >> shift) & (RBINS - 1)], 1u);
@@ -1100,11 +1104,12 @@ __global__ __launch_bounds__(BLOCK) void k_rhist(const uint64_t* keys,
 constexpr int RSEG = 16;
 
 __global__ void k_rscan_a(const uint32_t* hist, int nblk, int bs,
-                          uint32_t* psum) {
+                          uint32_t* psum, const uint32_t* counter) {
     int d = blockIdx.x * blockDim.x + threadIdx.x;
     int seg = blockIdx.y;
     if (d >= RBINS) return;
-    int b0 = seg * bs, b1 = min(nblk, b0 + bs);
+    const int live = (int)((*counter + RCHUNK - 1) / RCHUNK);
+    int b0 = seg * bs, b1 = min(min(nblk, live < 1 ? 1 : live), b0 + bs);
     uint32_t s = 0;
     for (int b = b0; b < b1; b++) s += hist[(int64_t)b * RBINS + d];
     psum[(int64_t)seg * RBINS + d] = s;
@@ -1137,13 +1142,14 @@ __global__ __launch_bounds__(1024) void k_rscan_b(const uint32_t* psum,
 
 __global__ void k_rscan_c(const uint32_t* hist, const uint32_t* psum,
                           const uint32_t* dbase, int nblk, int bs,
-                          uint32_t* offs) {
+                          uint32_t* offs, const uint32_t* counter) {
     int d = blockIdx.x * blockDim.x + threadIdx.x;
     int seg = blockIdx.y;
     if (d >= RBINS) return;
+    const int live = (int)((*counter + RCHUNK - 1) / RCHUNK);
     uint32_t run = dbase[d];
     for (int g = 0; g < seg; g++) run += psum[(int64_t)g * RBINS + d];
-    int b0 = seg * bs, b1 = min(nblk, b0 + bs);
+    int b0 = seg * bs, b1 = min(min(nblk, live < 1 ? 1 : live), b0 + bs);
     for (int b = b0; b < b1; b++) {
         uint32_t t = hist[(int64_t)b * RBINS + d];
         offs[(int64_t)b * RBINS + d] = run;
@@ -1375,10 +1381,10 @@ void launch_emission_sort(hipStream_t s, int64_t K, uint64_t* fkeys,
             hipLaunchKernelGGL(k_rhist, dim3(nblk), dim3(BLOCK), 0, s, ka,
                                counter, shift, rhist);
             hipLaunchKernelGGL(k_rscan_a, dim3(RBINS / 256, RSEG), dim3(256), 0,
-                               s, rhist, nblk, bs, psum);
+                               s, rhist, nblk, bs, psum, counter);
             hipLaunchKernelGGL(k_rscan_b, dim3(1), dim3(1024), 0, s, psum, dbase);
             hipLaunchKernelGGL(k_rscan_c, dim3(RBINS / 256, RSEG), dim3(256), 0,
-                               s, rhist, psum, dbase, nblk, bs, roffs);
+                               s, rhist, psum, dbase, nblk, bs, roffs, counter);
             hipLaunchKernelGGL(k_rscatter, dim3(nblk), dim3(BLOCK), 0, s, ka, pa,
                                counter, shift, roffs, kb, pb);
             std::swap(ka, kb);
