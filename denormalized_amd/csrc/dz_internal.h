@@ -109,6 +109,8 @@ void launch_regroup_l2_fold(hipStream_t stream, const uint4* d_grec2,
                             uint64_t* s_first, int64_t slab_cells,
                             uint32_t* d_dbg);
 
+struct EGatherSlots { int32_t s[16]; };
+
 struct EmitFilter {
     int32_t on;      /* 0 = no filter */
     int32_t field;   /* 0 cnt, 1 min, 2 max, 3 sum, 4 avg */
@@ -133,6 +135,27 @@ void launch_emission_sort(hipStream_t stream, int64_t K, uint64_t* fkeys,
                           uint64_t* skeys, uint32_t* fiota, uint32_t* okid,
                           uint32_t* counter2, uint32_t* rhist, uint32_t* roffs,
                           uint64_t max_key /* host-known bound on `first` */);
+/* group-batched device emission (<=16 closes in one chain; see
+ * kernels.hip §GROUP-BATCHED): read phase (slot slabs release after it),
+ * then sort+pack */
+void launch_emission_group_read(hipStream_t stream, const uint64_t* s_base,
+                                int64_t stride_u64, const EGatherSlots& slots,
+                                int gcount, int64_t K, int64_t kcap,
+                                const EmitFilter& ef, int cshift,
+                                uint64_t* gkeys, uint32_t* gkid,
+                                uint32_t* giota, uint64_t* gcnt_col,
+                                double* gmin, double* gmax, double* gsum,
+                                double* gavg, uint8_t* gflags, uint32_t* gtot,
+                                uint32_t* gcnt);
+void launch_emission_group_sort(hipStream_t stream, int64_t elem_cap,
+                                int gcount, int cshift, uint64_t max_first,
+                                uint64_t* gkeys, uint64_t* gskeys,
+                                uint32_t* gkid, uint32_t* gokid,
+                                uint32_t* giota, uint64_t* gcnt_col,
+                                double* gmin, double* gmax, double* gsum,
+                                double* gavg, uint8_t* gflags, uint32_t* gtot,
+                                uint32_t* rhist, uint32_t* roffs, char* pout);
+
 void launch_esort_small(hipStream_t stream, uint64_t* fkeys, uint64_t* skeys,
                         uint32_t* fiota, uint32_t* okid, uint32_t* counter2,
                         uint64_t max_key);
@@ -145,7 +168,6 @@ void launch_reset_slots(hipStream_t stream, const int32_t* d_slots, int ns,
 /* host-path emission: pack up to 16 closing window slots' state slabs into
  * one contiguous staging area (one launch + ONE big D2H replaces a per-close
  * copy on the push thread; slot ids travel by value in the kernel args) */
-struct EGatherSlots { int32_t s[16]; };
 void launch_egather_slabs(hipStream_t stream, const uint64_t* s_base,
                           int64_t stride_u64, EGatherSlots slots, int gcount,
                           uint64_t* out);

@@ -2272,3 +2272,170 @@ void launch_esort_small(hipStream_t s, uint64_t* fkeys, uint64_t* skeys,
 }
 
 } // namespace dz
+
+/* ------------------------------------------------------------------ */
+/* GROUP-BATCHED device emission: one chain for a whole trigger group  */
+/* of <= 16 closing windows (the per-close chain cost ~17 host         */
+/* enqueues; cfg3's sliding steps close ~80 windows, and the push      */
+/* thread's enqueue serialization was the wall). Composite sort key =  */
+/* close_idx << cshift | first_seen_row, so ONE radix sort yields      */
+/* close-major, insertion-ordered rows for every close at once.        */
+/* ------------------------------------------------------------------ */
+
+namespace dz {
+
+/* fused compact+filter+gather over all closes of the group: grid.y = close
+ * index; element positions come from one global atomic counter (order is
+ * irrelevant pre-sort); per-close row counts accumulate in gcnt[] */
+__global__ __launch_bounds__(BLOCK) void k_egf(const uint64_t* s_base,
+        int64_t stride_u64, EGatherSlots slots, int64_t K, int64_t kcap,
+        EmitFilter ef, int cshift, uint64_t* gkeys, uint32_t* gkid,
+        uint32_t* giota, uint64_t* gcnt_col, double* gmin, double* gmax,
+        double* gsum, double* gavg, uint8_t* gflags, uint32_t* gtot,
+        uint32_t* gcnt) {
+    const int c = blockIdx.y;
+    const uint64_t* sl = s_base + (int64_t)slots.s[c] * stride_u64;
+    const uint64_t* f_cnt = sl;
+    const uint64_t* f_first = sl + kcap;
+    const double* f_min = (const double*)(sl + 2 * kcap);
+    const double* f_max = (const double*)(sl + 3 * kcap);
+    const double* f_sum = (const double*)(sl + 4 * kcap);
+    __shared__ uint32_t base;
+    __shared__ uint32_t wsum[WAVES_PER_BLOCK];
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t k0 = (int64_t)blockIdx.x * blockDim.x; k0 < K; k0 += stride) {
+        const int64_t k = k0 + threadIdx.x;
+        uint64_t f = (k < K) ? f_first[k] : ~0ULL;
+        bool pass = f != ~0ULL;
+        uint64_t cnt = 0;
+        if (pass) {
+            cnt = f_cnt[k];
+            if (ef.on) {
+                const bool valid = cnt > 0;
+                double v = 0.0;
+                bool fv = true;
+                switch (ef.field) {
+                    case 0: v = (double)cnt; break;
+                    case 1: v = valid ? f_min[k] : 0.0; fv = valid; break;
+                    case 2: v = valid ? f_max[k] : 0.0; fv = valid; break;
+                    case 3: v = valid ? f_sum[k] : 0.0; fv = valid; break;
+                    default:
+                        v = valid ? f_sum[k] / (double)cnt : 0.0;
+                        fv = valid;
+                        break;
+                }
+                if (!fv) {
+                    pass = false; /* NULL never passes a comparison filter */
+                } else {
+                    switch (ef.cmp) {
+                        case 0: pass = v < ef.lit; break;
+                        case 1: pass = v <= ef.lit; break;
+                        case 2: pass = v > ef.lit; break;
+                        case 3: pass = v >= ef.lit; break;
+                        case 4: pass = v == ef.lit; break;
+                        default: pass = v != ef.lit; break;
+                    }
+                }
+            }
+        }
+        const uint64_t m = __ballot(pass);
+        const uint64_t below = (lane == 63) ? ~0ULL : ((1ULL << (lane + 1)) - 1);
+        const uint32_t wrank = (uint32_t)__popcll(m & below) - (pass ? 1 : 0);
+        if (lane == 0) wsum[wave] = (uint32_t)__popcll(m);
+        __syncthreads();
+        uint32_t wbase = 0, tot = 0;
+        for (int w = 0; w < WAVES_PER_BLOCK; w++) {
+            if (w < wave) wbase += wsum[w];
+            tot += wsum[w];
+        }
+        if (threadIdx.x == 0) {
+            base = tot ? atomicAdd(gtot, tot) : 0u;
+            if (tot) atomicAdd(&gcnt[c], tot);
+        }
+        __syncthreads();
+        if (pass) {
+            const uint32_t p = base + wbase + wrank;
+            const bool valid = cnt > 0;
+            gkeys[p] = ((uint64_t)c << cshift) | f;
+            gkid[p] = (uint32_t)k;
+            giota[p] = p; /* sort payload: identity permutation */
+            gcnt_col[p] = cnt;
+            gmin[p] = valid ? f_min[k] : 0.0;
+            gmax[p] = valid ? f_max[k] : 0.0;
+            gsum[p] = valid ? f_sum[k] : 0.0;
+            gavg[p] = valid ? f_sum[k] / (double)cnt : 0.0;
+            gflags[p] = valid ? 1 : 0;
+        }
+        __syncthreads();
+    }
+}
+
+/* pack the sorted group into close-major 53 B/row columns (same layout as
+ * k_epermute but keyed off the composite-sorted permutation) */
+__global__ void k_egpermute(const uint32_t* gtot, const uint32_t* sidx,
+                            const uint32_t* gkid, const uint64_t* gcnt_col,
+                            const double* gmin, const double* gmax,
+                            const double* gsum, const double* gavg,
+                            const uint8_t* gflags, char* out) {
+    const uint32_t nt = *gtot;
+    int64_t* pkey = (int64_t*)out;
+    uint64_t* pcnt = (uint64_t*)(out + (size_t)nt * 8);
+    double* pmin = (double*)(out + (size_t)nt * 16);
+    double* pmax = (double*)(out + (size_t)nt * 24);
+    double* psum = (double*)(out + (size_t)nt * 32);
+    double* pavg = (double*)(out + (size_t)nt * 40);
+    uint32_t* pkid = (uint32_t*)(out + (size_t)nt * 48);
+    uint8_t* pfl = (uint8_t*)(out + (size_t)nt * 52);
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nt;
+         i += stride) {
+        const uint32_t j = sidx[i];
+        const uint32_t k = gkid[j];
+        pkey[i] = (int64_t)k;
+        pcnt[i] = gcnt_col[j];
+        pmin[i] = gmin[j];
+        pmax[i] = gmax[j];
+        psum[i] = gsum[j];
+        pavg[i] = gavg[j];
+        pkid[i] = k;
+        pfl[i] = gflags[j];
+    }
+}
+
+void launch_emission_group_read(hipStream_t s, const uint64_t* s_base,
+                                int64_t stride_u64, const EGatherSlots& slots,
+                                int gcount, int64_t K, int64_t kcap,
+                                const EmitFilter& ef, int cshift,
+                                uint64_t* gkeys, uint32_t* gkid,
+                                uint32_t* giota, uint64_t* gcnt_col,
+                                double* gmin, double* gmax, double* gsum,
+                                double* gavg, uint8_t* gflags, uint32_t* gtot,
+                                uint32_t* gcnt) {
+    int cblocks = (int)std::min<int64_t>((K + BLOCK - 1) / BLOCK, 1024);
+    hipLaunchKernelGGL(k_egf, dim3(cblocks, gcount), dim3(BLOCK), 0, s, s_base,
+                       stride_u64, slots, K, kcap, ef, cshift, gkeys, gkid,
+                       giota, gcnt_col, gmin, gmax, gsum, gavg, gflags, gtot,
+                       gcnt);
+}
+
+void launch_emission_group_sort(hipStream_t s, int64_t elem_cap, int gcount,
+                                int cshift, uint64_t max_first,
+                                uint64_t* gkeys, uint64_t* gskeys,
+                                uint32_t* gkid, uint32_t* gokid,
+                                uint32_t* giota, uint64_t* gcnt_col,
+                                double* gmin, double* gmax, double* gsum,
+                                double* gavg, uint8_t* gflags, uint32_t* gtot,
+                                uint32_t* rhist, uint32_t* roffs, char* pout) {
+    /* one multi-block radix chain over the whole group's elements; the
+     * key bound covers close_idx << cshift */
+    uint64_t max_key = ((uint64_t)gcount << cshift) | max_first;
+    launch_emission_sort(s, elem_cap, gkeys, gskeys, giota, gokid, gtot,
+                         rhist, roffs, max_key);
+    int pb = (int)std::min<int64_t>((elem_cap + BLOCK - 1) / BLOCK, 2048);
+    hipLaunchKernelGGL(k_egpermute, dim3(pb), dim3(BLOCK), 0, s, gtot, giota,
+                       gkid, gcnt_col, gmin, gmax, gsum, gavg, gflags, pout);
+}
+
+} // namespace dz

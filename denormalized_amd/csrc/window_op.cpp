@@ -303,6 +303,12 @@ struct dz_window_op {
     uint32_t* d_rhist[E_CSTREAMS] = {}; /* radix scratch, one per c_stream */
     uint32_t* d_roffs[E_CSTREAMS] = {};
     int64_t e_slab_kcap = 0;
+    struct GroupCtl { /* sibling coordination for a group-batched close */
+        std::atomic<int> claimed{0};
+        std::atomic<bool> ready{false};
+        uint32_t offs[17] = {};
+        uint32_t gtot = 0;
+    };
     struct EmitJob {
         hipEvent_t ev;
         int slab;
@@ -316,6 +322,8 @@ struct dz_window_op {
         std::shared_ptr<std::atomic<int>> grp_left;
         int gbuf = -1;  /* pinned group buffer index, or -1 (own slab) */
         int goff = 0;   /* this close's slab offset within the group buffer */
+        bool grouped = false;      /* group-batched device emission */
+        std::shared_ptr<GroupCtl> ctl;
     };
     /* pinned group buffers for batched host-path emission (gather kernel
      * packs up to EGROUP slots contiguously on device, one D2H lands here) */
@@ -356,7 +364,30 @@ struct dz_window_op {
     uint64_t e_ticket_pop = 0;       /* next ticket poll may emit (out_mtx) */
     std::map<uint64_t, OutBuf> e_done; /* completed out-of-order (out_mtx) */
     std::atomic<uint32_t> e_nt_hint{0}; /* last close's passer count:
-                                          * adaptive small-sort choice */
+                                          * adaptive emission-path choice */
+    /* GROUP-BATCHED device emission: one chain serves a whole trigger
+     * group of closes (the per-close chain's ~25 host enqueues were the
+     * push-thread wall on many-small-close workloads — cfg3 sliding).
+     * Worst-case sized: gcount is capped at E_GELEMS/kcap closes, so the
+     * fused gather can never overflow. */
+    static constexpr int64_t E_GELEMS = 8 << 20;
+    struct GDev {
+        char* base = nullptr;
+        uint64_t* gkeys; uint64_t* gskeys;
+        uint32_t* gkid; uint32_t* gokid; uint32_t* giota;
+        uint64_t* gcnt_col;
+        double* gmin; double* gmax; double* gsum; double* gavg;
+        uint8_t* gflags;
+        uint32_t* ctr;    /* [0]=total, [1..16]=per-close counts */
+        uint32_t* rhist; uint32_t* roffs;
+        char* pout;       /* packed 53 B/elem, close-major sorted */
+    };
+    static constexpr int E_GD = 2;
+    GDev e_gd[E_GD];
+    char* e_gpin[E_GD] = {};      /* pinned packed staging */
+    uint32_t* e_gpcnt = nullptr;  /* pinned, E_GD x 32 counters */
+    std::vector<int> e_gdfree;    /* guarded by e_mtx */
+    bool e_gd_ready = false;
     std::atomic<uint64_t> e_build_ns{0};
     std::atomic<uint64_t> e_builds{0};
     std::atomic<uint64_t> e_zc_builds{0};   /* zero-copy vs materializing */
@@ -396,6 +427,9 @@ static void emit_worker_main(dz_window_op* op);
 static dz_status ensure_emission(dz_window_op* op);
 static dz_status process_pending(dz_window_op* op);
 static dz_status intern_sync_mirror(dz_window_op* op);
+static void build_emission_slice(dz_window_op* op, int64_t wstart,
+                                 int64_t wend, uint32_t total, uint32_t off,
+                                 uint32_t n, const char* p, OutBuf* out);
 
 static hipEvent_t get_event(dz_window_op* op) {
     if (!op->ev_pool.empty()) {
@@ -630,6 +664,11 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
         if (g) hipHostFree(g);
     for (auto& d : op->e_dev)
         if (d.base) hipFree(d.base);
+    for (auto& g : op->e_gd)
+        if (g.base) hipFree(g.base);
+    for (auto& g : op->e_gpin)
+        if (g) hipHostFree(g);
+    if (op->e_gpcnt) hipHostFree(op->e_gpcnt);
     if (op->e_pcnt) hipHostFree(op->e_pcnt);
     for (int s = 0; s < dz_window_op::E_CSTREAMS; s++) {
         hipFree(op->d_rhist[s]);
@@ -987,7 +1026,32 @@ static void emit_worker_main(dz_window_op* op) {
         event_spin_relaxed(job.ev); /* emission chain complete */
         auto t0 = std::chrono::steady_clock::now();
         OutBuf ob = take_outbuf(op);
-        if (job.device) {
+        if (job.grouped) {
+            /* group-batched close: the first sibling to claim pulls the
+             * whole packed span D2H and publishes the per-close offsets;
+             * every sibling then builds its own slice */
+            auto& ctl = *job.ctl;
+            if (ctl.claimed.exchange(1) == 0) {
+                const uint32_t* c = op->e_gpcnt + 32 * job.gbuf;
+                uint32_t gtot = c[0];
+                ctl.offs[0] = 0;
+                for (int i = 0; i < 16; i++)
+                    ctl.offs[i + 1] = ctl.offs[i] + c[1 + i];
+                if (gtot)
+                    hipMemcpy(op->e_gpin[job.gbuf], op->e_gd[job.gbuf].pout,
+                              (size_t)gtot * 53, hipMemcpyDeviceToHost);
+                ctl.gtot = gtot;
+                ctl.ready.store(true, std::memory_order_release);
+            } else {
+                while (!ctl.ready.load(std::memory_order_acquire))
+                    std::this_thread::sleep_for(std::chrono::microseconds(20));
+            }
+            const uint32_t off = ctl.offs[job.goff];
+            const uint32_t cnt = ctl.offs[job.goff + 1] - off;
+            op->e_nt_hint.store(cnt, std::memory_order_relaxed);
+            build_emission_slice(op, job.wstart, job.wend, ctl.gtot, off, cnt,
+                                 op->e_gpin[job.gbuf], &ob);
+        } else if (job.device) {
             /* copy exactly nt2 packed rows (the filter already ran on
              * device, so this is the final output volume, not the keyspace) */
             uint32_t nt = op->e_pcnt[job.slab];
@@ -1057,7 +1121,8 @@ static void emit_worker_main(dz_window_op* op) {
              * sibling is past its spin (i.e. fully built) */
             if (!job.grp_left || job.grp_left->fetch_sub(1) == 1) {
                 op->e_ev_pool.push_back(job.ev);
-                if (job.gbuf >= 0) op->e_gfree.push_back(job.gbuf);
+                if (job.grouped) op->e_gdfree.push_back(job.gbuf);
+                else if (job.gbuf >= 0) op->e_gfree.push_back(job.gbuf);
             }
             op->e_inflight--;
         }
@@ -1138,6 +1203,101 @@ static dz_status ensure_emission(dz_window_op* op) {
     return DZ_OK;
 }
 
+/* allocate the group-batched emission scratch (one-time) */
+static dz_status ensure_gd(dz_window_op* op) {
+    if (op->e_gd_ready) return DZ_OK;
+    constexpr int64_t E = dz_window_op::E_GELEMS;
+    const int64_t nblk = (E + dz::EMIT_RCHUNK - 1) / dz::EMIT_RCHUNK;
+    for (int i = 0; i < dz_window_op::E_GD; i++) {
+        auto& g = op->e_gd[i];
+        size_t bytes = (size_t)E * (8 + 8 + 4 + 4 + 4 + 8 + 32 + 1 + 53) +
+                       (size_t)(nblk + 17 + nblk) * dz::EMIT_RBINS * 4 + 256;
+        CHK(op, hipMalloc(&g.base, bytes));
+        char* p = g.base;
+        g.gkeys = (uint64_t*)p; p += E * 8;
+        g.gskeys = (uint64_t*)p; p += E * 8;
+        g.gcnt_col = (uint64_t*)p; p += E * 8;
+        g.gmin = (double*)p; p += E * 8;
+        g.gmax = (double*)p; p += E * 8;
+        g.gsum = (double*)p; p += E * 8;
+        g.gavg = (double*)p; p += E * 8;
+        g.gkid = (uint32_t*)p; p += E * 4;
+        g.gokid = (uint32_t*)p; p += E * 4;
+        g.giota = (uint32_t*)p; p += E * 4;
+        g.rhist = (uint32_t*)p; p += (size_t)(nblk + 17) * dz::EMIT_RBINS * 4;
+        g.roffs = (uint32_t*)p; p += (size_t)nblk * dz::EMIT_RBINS * 4;
+        g.gflags = (uint8_t*)p; p += E;
+        g.ctr = (uint32_t*)p; p += 128;
+        g.pout = p;
+        CHK(op, hipHostMalloc((void**)&op->e_gpin[i], (size_t)E * 53));
+    }
+    CHK(op, hipHostMalloc((void**)&op->e_gpcnt,
+                          (size_t)dz_window_op::E_GD * 32 * 4));
+    {
+        std::lock_guard<std::mutex> lk(op->e_mtx);
+        for (int i = 0; i < dz_window_op::E_GD; i++) op->e_gdfree.push_back(i);
+    }
+    op->e_gd_ready = true;
+    return DZ_OK;
+}
+
+/* Build one close's slice of a group-batched packed span (close-major,
+ * insertion-ordered): total = the group's element count (column-section
+ * stride), [off, off+n) = this close's rows. Materialising build — the
+ * group path runs in the many-SMALL-closes regime by construction. */
+static void build_emission_slice(dz_window_op* op, int64_t wstart,
+                                 int64_t wend, uint32_t total, uint32_t off,
+                                 uint32_t n, const char* p, OutBuf* out) {
+    const int64_t* pkey = (const int64_t*)p + off;
+    const uint64_t* pcnt = (const uint64_t*)(p + (size_t)total * 8) + off;
+    const double* pmin = (const double*)(p + (size_t)total * 16) + off;
+    const double* pmax = (const double*)(p + (size_t)total * 24) + off;
+    const double* psum = (const double*)(p + (size_t)total * 32) + off;
+    const double* pavg = (const double*)(p + (size_t)total * 40) + off;
+    const uint32_t* pkid = (const uint32_t*)(p + (size_t)total * 48) + off;
+    const uint8_t* pfl = (const uint8_t*)(p + (size_t)total * 52) + off;
+    size_t na = op->aggs.size();
+    OutBuf& ob = *out;
+    ob.agg_i64.resize(na);
+    ob.agg_f64.resize(na);
+    if (op->no_group) {
+    } else if (op->key_kind == DZ_KEY_UTF8) {
+        ob.key_offsets.resize(n + 1);
+        ob.key_offsets[0] = 0;
+        size_t totalb = 0;
+        for (size_t i = 0; i < n; i++) totalb += op->dict_strs[pkid[i]].size();
+        ob.key_data.resize(totalb);
+        size_t pos = 0;
+        for (size_t i = 0; i < n; i++) {
+            const std::string& str = op->dict_strs[pkid[i]];
+            memcpy(ob.key_data.data() + pos, str.data(), str.size());
+            pos += str.size();
+            ob.key_offsets[i + 1] = (int32_t)pos;
+        }
+    } else if (op->key_kind == DZ_KEY_INT64) {
+        ob.key_i64.resize(n);
+        for (size_t i = 0; i < n; i++) ob.key_i64[i] = op->dict_vals[pkid[i]];
+    } else {
+        ob.key_i64.assign(pkey, pkey + n);
+    }
+    for (size_t a = 0; a < na; a++) {
+        switch (op->aggs[a].op) {
+            case DZ_AGG_COUNT:
+                ob.agg_i64[a].assign((const int64_t*)pcnt,
+                                     (const int64_t*)pcnt + n);
+                break;
+            case DZ_AGG_MIN: ob.agg_f64[a].assign(pmin, pmin + n); break;
+            case DZ_AGG_MAX: ob.agg_f64[a].assign(pmax, pmax + n); break;
+            case DZ_AGG_SUM: ob.agg_f64[a].assign(psum, psum + n); break;
+            case DZ_AGG_AVG: ob.agg_f64[a].assign(pavg, pavg + n); break;
+        }
+    }
+    ob.agg_valid.assign(pfl, pfl + n);
+    ob.wstart.assign(n, wstart);
+    ob.wend.assign(n, wend);
+    ob.view.n_rows = (int64_t)n;
+}
+
 /* trigger_windows (grouped_window_agg_stream.rs:220-253): closed windows are
  * copied D2H asynchronously and built by the worker thread off the push
  * critical path; dz_window_op_drain/finish wait for completion. */
@@ -1202,8 +1362,106 @@ static dz_status trigger_windows(dz_window_op* op) {
     struct Pending { int slab; hipEvent_t ev; };
     constexpr size_t EGROUP = dz_window_op::E_POOL / 2;
     const bool dev_path_g = op->n_keys > 65536;
-    for (size_t g0 = 0; g0 < closed.size(); g0 += EGROUP) {
-    const size_t g1 = std::min(closed.size(), g0 + EGROUP);
+    /* group-batched device path: a whole trigger group of SMALL closes in
+     * ONE chain (~25 host enqueues total instead of per close — the
+     * push-thread enqueue serialization was the cfg3 sliding wall). Large
+     * closes (hint above the threshold) keep the per-close chain with its
+     * zero-copy large-output path. Worst-case safe: gstep is capped so
+     * gcount * kcap <= E_GELEMS — the fused gather cannot overflow. */
+    const bool grouped_ok = dev_path_g && op->kcap > 0 &&
+        op->kcap <= dz_window_op::E_GELEMS &&
+        op->e_nt_hint.load(std::memory_order_relaxed) <= 65536 &&
+        op->batch_seq < (1u << 19);
+    size_t g0 = 0;
+    while (g0 < closed.size()) {
+    size_t gstep = EGROUP;
+    if (grouped_ok)
+        gstep = std::min<size_t>(gstep, std::max<size_t>(1,
+            (size_t)(dz_window_op::E_GELEMS / op->kcap)));
+    const size_t g1 = std::min(closed.size(), g0 + gstep);
+    if (grouped_ok) {
+        if (ensure_gd(op) != DZ_OK) return DZ_ERR;
+        int gdidx;
+        hipEvent_t gev;
+        {
+            HostTimer htw(op, "h_emit_slabwait");
+            std::unique_lock<std::mutex> lk(op->e_mtx);
+            op->e_cv.wait(lk, [&] { return !op->e_gdfree.empty(); });
+            gdidx = op->e_gdfree.back();
+            op->e_gdfree.pop_back();
+            if (!op->e_ev_pool.empty()) {
+                gev = op->e_ev_pool.back();
+                op->e_ev_pool.pop_back();
+            } else {
+                hipEventCreate(&gev);
+            }
+        }
+        auto& gd = op->e_gd[gdidx];
+        const int gcount = (int)(g1 - g0);
+        hipStream_t cs = op->c_streams[gdidx % dz_window_op::E_CSTREAMS];
+        dz::EGatherSlots gs;
+        hipEvent_t fr = op->e_frontier[op->e_frontier_idx];
+        if (!fr) {
+            hipEventCreateWithFlags(&fr, hipEventDisableTiming);
+            op->e_frontier[op->e_frontier_idx] = fr;
+        }
+        op->e_frontier_idx =
+            (op->e_frontier_idx + 1) % dz_window_op::E_FRONTIERS;
+        for (size_t ci = g0; ci < g1; ci++) {
+            gs.s[ci - g0] = closed[ci].slot;
+            op->free_slots.push_back({closed[ci].slot, fr, false});
+        }
+        const uint64_t max_first = (((uint64_t)op->batch_seq + 1) << 32);
+        const int cshift = 64 - __builtin_clzll(max_first);
+        {
+            HostTimer htg(op, "h_trig_gather");
+            CHK(op, hipMemsetAsync(gd.ctr, 0, 128, cs));
+            dz::launch_emission_group_read(cs, op->s_base, (int64_t)stride, gs,
+                                           gcount, op->n_keys, op->kcap, ef,
+                                           cshift, gd.gkeys, gd.gkid,
+                                           gd.giota, gd.gcnt_col, gd.gmin,
+                                           gd.gmax, gd.gsum, gd.gavg,
+                                           gd.gflags, gd.ctr, gd.ctr + 1);
+            CHK(op, hipEventRecord(fr, cs)); /* slabs fully read */
+            dz::launch_emission_group_sort(cs, dz_window_op::E_GELEMS, gcount,
+                                           cshift, max_first, gd.gkeys,
+                                           gd.gskeys, gd.gkid, gd.gokid,
+                                           gd.giota, gd.gcnt_col, gd.gmin,
+                                           gd.gmax, gd.gsum, gd.gavg,
+                                           gd.gflags, gd.ctr, gd.rhist,
+                                           gd.roffs, gd.pout);
+            CHK(op, hipMemcpyAsync(op->e_gpcnt + 32 * gdidx, gd.ctr, 128,
+                                   hipMemcpyDeviceToHost, cs));
+            CHK(op, hipEventRecord(gev, cs));
+        }
+        auto left = std::make_shared<std::atomic<int>>(gcount);
+        auto ctl = std::make_shared<dz_window_op::GroupCtl>();
+        {
+            HostTimer htj(op, "h_trig_jobs");
+            std::lock_guard<std::mutex> lk(op->e_mtx);
+            for (size_t ci = g0; ci < g1; ci++) {
+                dz_window_op::EmitJob j;
+                j.ev = gev;
+                j.slab = -1;
+                j.wstart = closed[ci].start;
+                j.wend = closed[ci].end;
+                j.n_keys = op->n_keys;
+                j.kcap = op->kcap;
+                j.device = true;
+                j.grouped = true;
+                j.gbuf = gdidx;
+                j.goff = (int)(ci - g0);
+                j.grp_left = left;
+                j.ctl = ctl;
+                j.ticket = op->e_ticket_next++;
+                op->e_jobs.push_back(std::move(j));
+                op->e_inflight++;
+            }
+        }
+        op->e_cv.notify_all();
+        g0 = g1;
+        continue;
+    }
     if (!dev_path_g && op->e_gbufs[0]) {
         /* host-built closes (small keyspaces): pack the whole group's slabs
          * with one gather launch + ONE D2H into a pinned group buffer behind
@@ -1272,6 +1530,7 @@ static dz_status trigger_windows(dz_window_op* op) {
             }
         }
         op->e_cv.notify_all();
+        g0 = g1;
         continue;
     }
     std::vector<Pending> pend;
@@ -1388,6 +1647,7 @@ static dz_status trigger_windows(dz_window_op* op) {
         }
     }
     op->e_cv.notify_all(); /* one wakeup per group, not per close */
+    g0 = g1;
     } /* group loop */
     return DZ_OK;
 }
