@@ -382,7 +382,8 @@ struct dz_window_op {
         uint32_t* rhist; uint32_t* roffs;
         char* pout;       /* packed 53 B/elem, close-major sorted */
     };
-    static constexpr int E_GD = 2;
+    static constexpr int E_GD = 4; /* groups in flight (one per
+                                    * emission stream) */
     GDev e_gd[E_GD];
     char* e_gpin[E_GD] = {};      /* pinned packed staging */
     uint32_t* e_gpcnt = nullptr;  /* pinned, E_GD x 32 counters */
