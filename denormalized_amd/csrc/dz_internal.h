@@ -48,7 +48,7 @@ void launch_gen(hipStream_t stream, uint64_t seed, int64_t t0, int64_t start_row
 
 /* device utf8 intern (GroupValues::intern analog at device rate) */
 void launch_intern(hipStream_t stream, const int32_t* d_offs, const char* d_data,
-                   int64_t n, uint64_t* tab_fp, uint32_t* tab_id,
+                   int64_t n, uint4* tab /* {fp, id, len|off<<6} slots */,
                    uint32_t* tab_row, uint32_t p_mask, uint32_t* id_off,
                    uint32_t* id_len, char* pool, uint32_t* ctrs,
                    uint32_t id_cap, uint32_t pool_cap, int32_t* out_kid,

@@ -137,9 +137,17 @@ __device__ __forceinline__ bool bytes_eq(const char* a, const char* b,
     return true;
 }
 
+/* table slot: 16 B {fp lo, fp hi, id, len|off<<6} — ONE load serves the
+ * probe AND the steady-state resolve (separate fp/id arrays cost a second
+ * dependent access per row). lenoff packs len<=63 and pool off<2^26;
+ * larger keys resolve through the id_off/id_len fallback arrays. */
+__device__ __forceinline__ uint64_t slot_fp(const uint4& v) {
+    return (uint64_t)v.x | ((uint64_t)v.y << 32);
+}
+
 __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
-        const char* data, int64_t n, uint64_t* tab_fp, uint32_t* tab_row,
-        const uint32_t* tab_id, const uint32_t* id_off, const uint32_t* id_len,
+        const char* data, int64_t n, uint4* tab, uint32_t* tab_row,
+        const uint32_t* id_off, const uint32_t* id_len,
         const char* pool, uint32_t p_mask, int32_t* out_kid, uint32_t* dbg) {
     /* phase 1: every row probes; exactly one row CASes each new
      * fingerprint in, recording itself as the claiming row. No lane ever
@@ -157,12 +165,11 @@ __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
         uint32_t slot = (uint32_t)fp & p_mask;
         int32_t out;
         for (uint32_t probes = 0;; slot = (slot + 1) & p_mask) {
-            uint64_t got = tab_fp[slot]; /* read first: the steady state
-                                          * (key already present) must not
-                                          * pay an atomic per row */
+            const uint4 v = tab[slot]; /* one 16 B load: fp + id + lenoff */
+            uint64_t got = slot_fp(v);
             if (got == 0) {
                 got = (uint64_t)atomicCAS(
-                    (unsigned long long*)&tab_fp[slot], 0ULL,
+                    (unsigned long long*)&tab[slot], 0ULL,
                     (unsigned long long)fp);
                 if (got == 0) {
                     tab_row[slot] = (uint32_t)i; /* claimed: I define bytes */
@@ -171,16 +178,23 @@ __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
                 }
             }
             if (got == fp) { /* same key (fp64 exact) */
-                const uint32_t cand = tab_id[slot];
-                if (cand != ~0u &&
-                    id_len[cand] == (uint32_t)len &&
-                    bytes_eq(pool + id_off[cand], data + o0, len)) {
-                    out = ~(int32_t)cand; /* resolved inline */
+                const uint32_t cand = v.z;
+                if (cand != ~0u) {
+                    const uint32_t lo = v.w;
+                    const uint32_t clen = lo & 63u;
+                    const bool fits = clen != 63u;
+                    const uint32_t co = fits ? (lo >> 6) : id_off[cand];
+                    const uint32_t cl = fits ? clen : id_len[cand];
+                    if (cl == (uint32_t)len &&
+                        bytes_eq(pool + co, data + o0, len)) {
+                        out = ~(int32_t)cand; /* resolved inline */
+                        break;
+                    }
+                    /* fp64 collision with a different key: probe on */
                 } else {
-                    out = (int32_t)slot; /* fresh this batch (or fp64
-                                          * collision): lookup decides */
+                    out = (int32_t)slot; /* fresh this batch: lookup decides */
+                    break;
                 }
-                break;
             }
             if (++probes > p_mask) {
                 dbg[3] = 4; /* table full */
@@ -192,8 +206,8 @@ __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
     }
 }
 
-__global__ __launch_bounds__(BLOCK) void k_intern_assign(uint64_t* tab_fp,
-        uint32_t* tab_row, uint32_t* tab_id, uint32_t p_count,
+__global__ __launch_bounds__(BLOCK) void k_intern_assign(uint4* tab,
+        uint32_t* tab_row, uint32_t p_count,
         const int32_t* offs, const char* data, uint32_t* id_off,
         uint32_t* id_len, char* pool, uint32_t* ctrs, uint32_t id_cap,
         uint32_t pool_cap, uint32_t* dbg) {
@@ -203,7 +217,8 @@ __global__ __launch_bounds__(BLOCK) void k_intern_assign(uint64_t* tab_fp,
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t sl = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
          sl < p_count; sl += stride) {
-        if (tab_fp[sl] == 0 || tab_id[sl] != ~0u) continue;
+        uint4 v = tab[sl];
+        if (slot_fp(v) == 0 || v.z != ~0u) continue;
         const uint32_t r = tab_row[sl];
         const int32_t o0 = offs[r];
         const int32_t len = offs[r + 1] - o0;
@@ -216,12 +231,17 @@ __global__ __launch_bounds__(BLOCK) void k_intern_assign(uint64_t* tab_fp,
         for (int32_t j = 0; j < len; j++) pool[po + j] = data[o0 + j];
         id_off[nid] = po;
         id_len[nid] = (uint32_t)len;
-        tab_id[sl] = nid;
+        v.z = nid;
+        /* lenoff fast path: len<=62 and off<2^26 (63 in the len field =
+         * "use the fallback arrays") */
+        v.w = (len <= 62 && po < (1u << 26)) ? ((po << 6) | (uint32_t)len)
+                                             : 0xFFFFFFFFu | 63u;
+        tab[sl] = v;
     }
 }
 
 __global__ __launch_bounds__(BLOCK) void k_intern_lookup(const int32_t* offs,
-        const char* data, int64_t n, const uint32_t* tab_id,
+        const char* data, int64_t n, const uint4* tab,
         const uint32_t* id_off, const uint32_t* id_len, const char* pool,
         int32_t* out_kid, uint32_t* dbg) {
     /* phase 3 (after assign): resolve each row's slot (stashed by the claim
@@ -237,7 +257,7 @@ __global__ __launch_bounds__(BLOCK) void k_intern_lookup(const int32_t* offs,
             continue;
         }
         const uint32_t slot = (uint32_t)v;
-        const uint32_t cand = tab_id[slot];
+        const uint32_t cand = tab[slot].z;
         int32_t id = 0;
         if (cand == ~0u) {
             dbg[3] = 5; /* capacity overflow left the slot unassigned */
@@ -254,23 +274,22 @@ __global__ __launch_bounds__(BLOCK) void k_intern_lookup(const int32_t* offs,
 }
 
 void launch_intern(hipStream_t s, const int32_t* d_offs, const char* d_data,
-                   int64_t n, uint64_t* tab_fp, uint32_t* tab_id,
-                   uint32_t* tab_row, uint32_t p_mask, uint32_t* id_off,
-                   uint32_t* id_len, char* pool, uint32_t* ctrs,
-                   uint32_t id_cap, uint32_t pool_cap, int32_t* out_kid,
-                   uint32_t* dbg) {
+                   int64_t n, uint4* tab, uint32_t* tab_row, uint32_t p_mask,
+                   uint32_t* id_off, uint32_t* id_len, char* pool,
+                   uint32_t* ctrs, uint32_t id_cap, uint32_t pool_cap,
+                   int32_t* out_kid, uint32_t* dbg) {
     int blocks = (int)std::min<int64_t>((n + BLOCK - 1) / BLOCK, 2048);
     if (blocks < 1) blocks = 1;
     const uint32_t P = p_mask + 1;
     int sblocks = (int)std::min<uint32_t>((P + BLOCK - 1) / BLOCK, 2048);
     hipLaunchKernelGGL(k_intern_claim, dim3(blocks), dim3(BLOCK), 0, s, d_offs,
-                       d_data, n, tab_fp, tab_row, tab_id, id_off, id_len,
+                       d_data, n, tab, tab_row, id_off, id_len,
                        pool, p_mask, out_kid, dbg);
     hipLaunchKernelGGL(k_intern_assign, dim3(sblocks), dim3(BLOCK), 0, s,
-                       tab_fp, tab_row, tab_id, P, d_offs, d_data, id_off,
+                       tab, tab_row, P, d_offs, d_data, id_off,
                        id_len, pool, ctrs, id_cap, pool_cap, dbg);
     hipLaunchKernelGGL(k_intern_lookup, dim3(blocks), dim3(BLOCK), 0, s,
-                       d_offs, d_data, n, tab_id, id_off,
+                       d_offs, d_data, n, tab, id_off,
                        id_len, pool, out_kid, dbg);
 }
 

@@ -225,8 +225,7 @@ struct dz_window_op {
 
     /* device utf8 intern (GroupValues::intern at device rate). Fixed-capacity
      * from n_keys_hint at create; capacity overflow flags d_dbg[3]. */
-    uint64_t* d_itab_fp = nullptr;  /* open-address fingerprint table */
-    uint32_t* d_itab_id = nullptr;
+    uint4* d_itab = nullptr;        /* 16 B slots {fp, id, len|off<<6} */
     uint32_t* d_itab_row = nullptr; /* claiming row, valid within one claim */
     uint32_t i_pmask = 0;
     uint32_t* d_ioff = nullptr;     /* per-id {pool offset, byte length} */
@@ -686,7 +685,7 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_scalars[0]); hipFree(op->d_scalars[1]);
     hipFree(op->d_slotmap); hipFree(op->d_zero_kid);
     hipFree(op->d_resetlist); hipFree(op->d_dbg);
-    hipFree(op->d_itab_fp); hipFree(op->d_itab_id); hipFree(op->d_itab_row);
+    hipFree(op->d_itab); hipFree(op->d_itab_row);
     hipFree(op->d_ioff); hipFree(op->d_ilen);
     hipFree(op->d_ipool); hipFree(op->d_ictrs);
     hipFree(op->d_ikid[0]); hipFree(op->d_ikid[1]);
@@ -2065,7 +2064,7 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
         op->watermark = mn;
         op->has_wm = true;
     }
-    if (op->d_itab_fp && intern_sync_mirror(op) != DZ_OK)
+    if (op->d_itab && intern_sync_mirror(op) != DZ_OK)
         return DZ_ERR; /* newly interned key bytes -> emission dictionary */
     return trigger_windows(op);
 }
@@ -2166,7 +2165,7 @@ extern "C" dz_status dz_window_op_push_device_borrowed(
 /* ------------------------------------------------------------------ */
 
 static dz_status ensure_intern(dz_window_op* op, int64_t n) {
-    if (!op->d_itab_fp) {
+    if (!op->d_itab) {
         uint64_t want = 4 * (uint64_t)std::max<int64_t>(op->kcap, 1);
         uint32_t P = 1u << 16;
         while (P < want && P < (1u << 26)) P <<= 1;
@@ -2174,15 +2173,17 @@ static dz_status ensure_intern(dz_window_op* op, int64_t n) {
         op->i_idcap = P / 2;
         op->i_poolcap = (uint32_t)std::min<uint64_t>((uint64_t)op->i_idcap * 64,
                                                      1u << 31);
-        CHK(op, hipMalloc(&op->d_itab_fp, (size_t)P * 8));
-        CHK(op, hipMalloc(&op->d_itab_id, (size_t)P * 4));
+        CHK(op, hipMalloc(&op->d_itab, (size_t)P * 16));
         CHK(op, hipMalloc(&op->d_itab_row, (size_t)P * 4));
         CHK(op, hipMalloc(&op->d_ioff, (size_t)op->i_idcap * 4));
         CHK(op, hipMalloc(&op->d_ilen, (size_t)op->i_idcap * 4));
         CHK(op, hipMalloc(&op->d_ipool, op->i_poolcap));
         CHK(op, hipMalloc(&op->d_ictrs, 8));
-        CHK(op, hipMemsetAsync(op->d_itab_fp, 0, (size_t)P * 8, op->stream));
-        CHK(op, hipMemsetAsync(op->d_itab_id, 0xFF, (size_t)P * 4, op->stream));
+        /* fp halves zero (empty), id halves ~0 (unassigned): zero the whole
+         * table then set .z/.w lanes via a strided 0xFF fill */
+        CHK(op, hipMemsetAsync(op->d_itab, 0, (size_t)P * 16, op->stream));
+        CHK(op, hipMemset2DAsync((char*)op->d_itab + 8, 16, 0xFF, 8, P,
+                                 op->stream));
         CHK(op, hipMemsetAsync(op->d_ictrs, 0, 8, op->stream));
         /* the intern runs on i_stream: make the init visible there */
         CHK(op, hipStreamSynchronize(op->stream));
@@ -2255,7 +2256,7 @@ extern "C" dz_status dz_window_op_push_device_utf8(dz_window_op* op,
         CHK(op, hipStreamWaitEvent(op->i_stream, op->ev_consumed[b], 0));
     timed_on(op, op->i_stream, "intern", (double)n_rows * 18, [&] {
         dz::launch_intern(op->i_stream, d_key_offsets, d_key_data, n_rows,
-                          op->d_itab_fp, op->d_itab_id, op->d_itab_row,
+                          op->d_itab, op->d_itab_row,
                           op->i_pmask, op->d_ioff, op->d_ilen, op->d_ipool,
                           op->d_ictrs, op->i_idcap, op->i_poolcap,
                           op->d_ikid[b], op->d_dbg);
@@ -2344,7 +2345,7 @@ extern "C" dz_status dz_window_op_push(dz_window_op* op, const dz_batch* batch) 
         const int32_t* offs = kc.offsets;
         const char* data = (const char*)kc.data;
         if (!offs || !data) { op->err = "utf8 key column needs offsets+data"; return DZ_ERR; }
-        if (op->d_itab_fp) {
+        if (op->d_itab) {
             op->err = "cannot mix host utf8 pushes and device utf8 pushes on "
                       "one operator (two dictionaries would assign "
                       "conflicting ids)";
