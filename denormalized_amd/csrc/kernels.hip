@@ -145,48 +145,6 @@ __device__ __forceinline__ uint64_t slot_fp(const uint4& v) {
     return (uint64_t)v.x | ((uint64_t)v.y << 32);
 }
 
-__device__ __forceinline__ int32_t intern_claim_one(const char* data,
-        const int32_t o0, const int32_t len, const uint64_t fp, uint32_t slot,
-        uint4 v, uint4* tab, uint32_t* tab_row, const uint32_t* id_off,
-        const uint32_t* id_len, const char* pool, uint32_t p_mask, int64_t i,
-        uint32_t* dbg) {
-    /* probe from an already-loaded first slot value */
-    for (uint32_t probes = 0;;) {
-        uint64_t got = slot_fp(v);
-        if (got == 0) {
-            got = (uint64_t)atomicCAS((unsigned long long*)&tab[slot], 0ULL,
-                                      (unsigned long long)fp);
-            if (got == 0) {
-                tab_row[slot] = (uint32_t)i; /* claimed: I define the bytes */
-                return (int32_t)slot;
-            }
-            v = tab[slot]; /* lost the race: re-read the full slot */
-            got = slot_fp(v);
-        }
-        if (got == fp) { /* same key (fp64 exact) */
-            const uint32_t cand = v.z;
-            if (cand != ~0u) {
-                const uint32_t lo = v.w;
-                const uint32_t clen = lo & 63u;
-                const bool fits = clen != 63u;
-                const uint32_t co = fits ? (lo >> 6) : id_off[cand];
-                const uint32_t cl = fits ? clen : id_len[cand];
-                if (cl == (uint32_t)len && bytes_eq(pool + co, data + o0, len))
-                    return ~(int32_t)cand; /* resolved inline */
-                /* fp64 collision with a different key: probe on */
-            } else {
-                return (int32_t)slot; /* fresh this batch: lookup decides */
-            }
-        }
-        if (++probes > p_mask) {
-            dbg[3] = 4; /* table full */
-            return (int32_t)slot;
-        }
-        slot = (slot + 1) & p_mask;
-        v = tab[slot];
-    }
-}
-
 __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
         const char* data, int64_t n, uint4* tab, uint32_t* tab_row,
         const uint32_t* id_off, const uint32_t* id_len,
@@ -196,35 +154,55 @@ __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
      * waits on another (a publish-wait design can cycle across waves).
      * STEADY STATE (slot already has an assigned id from an earlier batch):
      * byte-verify and resolve right here — out_kid = ~id — so the lookup
-     * phase touches key bytes only for rows of freshly claimed keys.
-     * Rows are processed in PAIRS per thread: both hashes and both first
-     * slot loads issue before either resolve consumes them — the
-     * single-row form serialized hash -> load -> verify per row and was
-     * latency-bound. */
-    int64_t stride = 2 * (int64_t)gridDim.x * blockDim.x;
-    for (int64_t i = 2 * (blockIdx.x * (int64_t)blockDim.x + threadIdx.x);
-         i < n; i += stride) {
-        const int64_t ia = i, ib = i + 1;
-        const bool hb = ib < n;
-        const int32_t o0a = offs[ia];
-        const int32_t lena = offs[ia + 1] - o0a;
-        const int32_t o0b = hb ? offs[ib] : o0a;
-        const int32_t lenb = hb ? offs[ib + 1] - o0b : 0;
-        uint64_t fpa = fnv1a64(data + o0a, lena);
-        uint64_t fpb = hb ? fnv1a64(data + o0b, lenb) : 0;
-        if (!fpa) fpa = 1;
-        if (!fpb) fpb = 1;
-        const uint32_t sa = (uint32_t)fpa & p_mask;
-        const uint32_t sb = (uint32_t)fpb & p_mask;
-        const uint4 va = tab[sa];
-        const uint4 vb = hb ? tab[sb] : va;
-        out_kid[ia] = intern_claim_one(data, o0a, lena, fpa, sa, va, tab,
-                                       tab_row, id_off, id_len, pool, p_mask,
-                                       ia, dbg);
-        if (hb)
-            out_kid[ib] = intern_claim_one(data, o0b, lenb, fpb, sb, vb, tab,
-                                           tab_row, id_off, id_len, pool,
-                                           p_mask, ib, dbg);
+     * phase touches key bytes only for rows of freshly claimed keys. */
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        const int32_t o0 = offs[i];
+        const int32_t len = offs[i + 1] - o0;
+        uint64_t fp = fnv1a64(data + o0, len);
+        if (!fp) fp = 1; /* 0 marks an empty slot */
+        uint32_t slot = (uint32_t)fp & p_mask;
+        int32_t out;
+        for (uint32_t probes = 0;; slot = (slot + 1) & p_mask) {
+            const uint4 v = tab[slot]; /* one 16 B load: fp + id + lenoff */
+            uint64_t got = slot_fp(v);
+            if (got == 0) {
+                got = (uint64_t)atomicCAS(
+                    (unsigned long long*)&tab[slot], 0ULL,
+                    (unsigned long long)fp);
+                if (got == 0) {
+                    tab_row[slot] = (uint32_t)i; /* claimed: I define bytes */
+                    out = (int32_t)slot;
+                    break;
+                }
+            }
+            if (got == fp) { /* same key (fp64 exact) */
+                const uint32_t cand = v.z;
+                if (cand != ~0u) {
+                    const uint32_t lo = v.w;
+                    const uint32_t clen = lo & 63u;
+                    const bool fits = clen != 63u;
+                    const uint32_t co = fits ? (lo >> 6) : id_off[cand];
+                    const uint32_t cl = fits ? clen : id_len[cand];
+                    if (cl == (uint32_t)len &&
+                        bytes_eq(pool + co, data + o0, len)) {
+                        out = ~(int32_t)cand; /* resolved inline */
+                        break;
+                    }
+                    /* fp64 collision with a different key: probe on */
+                } else {
+                    out = (int32_t)slot; /* fresh this batch: lookup decides */
+                    break;
+                }
+            }
+            if (++probes > p_mask) {
+                dbg[3] = 4; /* table full */
+                out = (int32_t)slot;
+                break;
+            }
+        }
+        out_kid[i] = out;
     }
 }
 
