@@ -165,8 +165,12 @@ __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
         uint32_t slot = (uint32_t)fp & p_mask;
         int32_t out;
         for (uint32_t probes = 0;; slot = (slot + 1) & p_mask) {
-            const uint4 v = tab[slot]; /* one 16 B load: fp + id + lenoff */
-            uint64_t got = slot_fp(v);
+            /* fp first as an aligned 8 B load (single-copy atomic vs a
+             * concurrent claim CAS — the fp halves of a 16 B vector load
+             * are not guaranteed tear-free); the id/lenoff follow-up hits
+             * the SAME cache line, so the steady state still pays one
+             * memory-system access */
+            uint64_t got = ((const uint64_t*)tab)[(size_t)slot * 2];
             if (got == 0) {
                 got = (uint64_t)atomicCAS(
                     (unsigned long long*)&tab[slot], 0ULL,
@@ -178,6 +182,7 @@ __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
                 }
             }
             if (got == fp) { /* same key (fp64 exact) */
+                const uint4 v = tab[slot]; /* L1-hot: same line as the fp */
                 const uint32_t cand = v.z;
                 if (cand != ~0u) {
                     const uint32_t lo = v.w;
