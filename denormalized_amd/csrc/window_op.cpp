@@ -2788,10 +2788,13 @@ static dz_status join_ensure_probe(dz_join_op* op, int64_t n) {
         JCHK(op, hipMalloc(&op->d_drvtmp, (size_t)need_drv * 4));
         op->drv_cap = need_drv;
     }
-    /* output must hold this probe's matches OR a full buffer re-probe */
+    /* output must hold this probe's matches OR a full buffer re-probe.
+     * GROWTH frees buffers a downstream window op may still be reading
+     * under the borrowed-push contract (its streams, not ours): take a
+     * full-device barrier before freeing. */
     int64_t need_o = std::max(n, op->u_n);
     if (need_o > op->o_cap) {
-        JCHK(op, hipStreamSynchronize(op->stream));
+        JCHK(op, hipDeviceSynchronize());
         for (int i = 0; i < 2; i++) {
             hipFree(op->o_ts[i]); hipFree(op->o_kid[i]); hipFree(op->o_val[i]);
             JCHK(op, hipMalloc(&op->o_ts[i], (size_t)need_o * 8));
@@ -3042,7 +3045,9 @@ extern "C" dz_status dz_json_decode(dz_json_decoder* d, const char* d_bytes,
     const int64_t nrec = d->h_tot[0];
     if (nrec == 0) return DZ_OK;
     if (nrec > d->rec_cap || n_bytes > d->kdata_cap) {
-        DCHK(d, hipStreamSynchronize(d->stream));
+        /* growth frees columns a downstream window op may still read
+         * under the borrowed-push contract: full-device barrier first */
+        DCHK(d, hipDeviceSynchronize());
         int64_t rc = std::max(nrec, d->rec_cap);
         int64_t kc = std::max(n_bytes, d->kdata_cap);
         for (int i = 0; i < 2; i++) {
