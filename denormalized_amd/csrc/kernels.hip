@@ -240,7 +240,8 @@ __global__ __launch_bounds__(BLOCK) void k_intern_assign(uint4* tab,
         /* lenoff fast path: len<=62 and off<2^26 (63 in the len field =
          * "use the fallback arrays") */
         v.w = (len <= 62 && po < (1u << 26)) ? ((po << 6) | (uint32_t)len)
-                                             : 0xFFFFFFFFu | 63u;
+                                             : 0xFFFFFFFFu; /* len field 63
+                                                => use the fallback arrays */
         tab[sl] = v;
     }
 }

@@ -2746,7 +2746,7 @@ extern "C" dz_join_op* dz_join_op_create(int32_t device, int64_t n_trips_hint) {
         hipMalloc(&op->d_dbg, 16) != hipSuccess ||
         hipHostMalloc((void**)&op->h_tot, 8) != hipSuccess) {
         g_err = "join op allocation failed";
-        delete op;
+        dz_join_op_destroy(op); /* frees whatever was allocated */
         return nullptr;
     }
     hipMemset(op->d_dbg, 0, 16);
@@ -2958,6 +2958,8 @@ struct dz_json_decoder {
     int64_t n_rec = 0;
 };
 
+extern "C" void dz_json_decoder_destroy(dz_json_decoder* d);
+
 extern "C" dz_json_decoder* dz_json_decoder_create(int32_t device,
                                                    const char* ts_field,
                                                    const char* key_field,
@@ -2988,7 +2990,7 @@ extern "C" dz_json_decoder* dz_json_decoder_create(int32_t device,
         hipMalloc(&d->d_dbg, 16) != hipSuccess ||
         hipHostMalloc((void**)&d->h_tot, 8) != hipSuccess) {
         g_err = "json decoder allocation failed";
-        delete d;
+        dz_json_decoder_destroy(d); /* frees whatever was allocated */
         return nullptr;
     }
     hipMemset(d->d_dbg, 0, 16);
