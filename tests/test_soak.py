@@ -174,12 +174,10 @@ def test_json_decoder_fuzz():
         rec = {"occurred_at_ms": int(rng.integers(0, 2**40)),
                "sensor_name": "k" * int(rng.integers(1, 30)) + str(i % 97),
                "reading": round(float(rng.uniform(-1e6, 1e6)), 6)}
-        extras = int(rng.integers(0, 3))
-        for e in range(extras):
-            rec[f"x{e}"] = rng.choice(
-                [True, False, None, 12, -3.5,
-                 {"nested": [1, 2, {"y": "z"}]}, "strv"]).item() \
-                if not isinstance(rng.choice([1, {"a": 1}]), dict) else {"a": [i]}
+        pool = [True, False, None, 12, -3.5, "strv",
+                {"nested": [1, 2, {"y": "z"}]}, [0, {"a": "b"}, []]]
+        for e in range(int(rng.integers(0, 3))):
+            rec[f"x{e}"] = pool[int(rng.integers(0, len(pool)))]
         items = list(rec.items())
         rng.shuffle(items)
         rows.append(dict(items))
