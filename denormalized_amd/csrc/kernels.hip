@@ -183,8 +183,7 @@ __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
                     (unsigned long long)fp);
                 if (got == 0) {
                     tab_row[slot] = (uint32_t)i; /* claimed: I define bytes */
-                    out = (int32_t)slot;
-                    break;
+                    break; /* out stays -1: fresh, fix-up resolves */
                 }
             }
             if (got == fp) { /* same key (fp64 exact) */
@@ -198,20 +197,25 @@ __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
                     const uint32_t cl = fits ? clen : id_len[cand];
                     if (cl == (uint32_t)len &&
                         bytes_eq(pool + co, data + o0, len)) {
-                        out = ~(int32_t)cand; /* resolved inline */
+                        out = (int32_t)cand; /* resolved: FINAL id */
                         break;
                     }
                     /* fp64 collision with a different key: probe on */
                 } else {
-                    out = (int32_t)slot; /* fresh this batch: lookup decides */
-                    break;
+                    break; /* claimed this batch, id pending: fresh */
                 }
             }
             if (++probes > p_mask) {
                 dbg[3] = 4; /* table full */
-                out = (int32_t)slot;
+                out = 0;
                 break;
             }
+        }
+        if (out < 0) { /* fresh: placeholder id 0 until the fix-up */
+            const uint32_t p = atomicAdd(fresh_ctr, 1u);
+            fresh_rows[p] = (uint32_t)i;
+            fresh_slots[p] = slot;
+            out = 0;
         }
         out_kid[i] = out;
     }
