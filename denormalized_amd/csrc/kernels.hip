@@ -183,7 +183,8 @@ __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
                     (unsigned long long)fp);
                 if (got == 0) {
                     tab_row[slot] = (uint32_t)i; /* claimed: I define bytes */
-                    break; /* out stays -1: fresh, fix-up resolves */
+                    out = (int32_t)slot;
+                    break;
                 }
             }
             if (got == fp) { /* same key (fp64 exact) */
@@ -197,25 +198,20 @@ __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
                     const uint32_t cl = fits ? clen : id_len[cand];
                     if (cl == (uint32_t)len &&
                         bytes_eq(pool + co, data + o0, len)) {
-                        out = (int32_t)cand; /* resolved: FINAL id */
+                        out = ~(int32_t)cand; /* resolved inline */
                         break;
                     }
                     /* fp64 collision with a different key: probe on */
                 } else {
-                    break; /* claimed this batch, id pending: fresh */
+                    out = (int32_t)slot; /* fresh this batch: lookup decides */
+                    break;
                 }
             }
             if (++probes > p_mask) {
                 dbg[3] = 4; /* table full */
-                out = 0;
+                out = (int32_t)slot;
                 break;
             }
-        }
-        if (out < 0) { /* fresh: placeholder id 0 until the fix-up */
-            const uint32_t p = atomicAdd(fresh_ctr, 1u);
-            fresh_rows[p] = (uint32_t)i;
-            fresh_slots[p] = slot;
-            out = 0;
         }
         out_kid[i] = out;
     }
