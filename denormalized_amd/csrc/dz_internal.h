@@ -52,8 +52,7 @@ void launch_intern(hipStream_t stream, const int32_t* d_offs, const char* d_data
                    uint32_t* tab_row, uint32_t p_mask, uint32_t* id_off,
                    uint32_t* id_len, char* pool, uint32_t* ctrs,
                    uint32_t id_cap, uint32_t pool_cap, int32_t* out_kid,
-                   uint32_t* fresh_rows, uint32_t* fresh_slots,
-                   uint32_t* fresh_ctr, uint32_t* d_dbg);
+                   uint32_t* d_dbg);
 
 /* synthetic utf8 key generator ("sensor_{k}"): lens pass and/or fill pass */
 void launch_gen_utf8(hipStream_t stream, uint64_t seed, int64_t start_row,

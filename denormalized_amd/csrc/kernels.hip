@@ -137,8 +137,6 @@ __device__ __forceinline__ bool bytes_eq(const char* a, const char* b,
     return true;
 }
 
-__global__ void k_zero1(uint32_t* p) { *p = 0; }
-
 /* table slot: 16 B {fp lo, fp hi, id, len|off<<6} — ONE load serves the
  * probe AND the steady-state resolve (separate fp/id arrays cost a second
  * dependent access per row). lenoff packs len<=63 and pool off<2^26;
@@ -150,17 +148,13 @@ __device__ __forceinline__ uint64_t slot_fp(const uint4& v) {
 __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
         const char* data, int64_t n, uint4* tab, uint32_t* tab_row,
         const uint32_t* id_off, const uint32_t* id_len,
-        const char* pool, uint32_t p_mask, int32_t* out_kid,
-        uint32_t* fresh_rows, uint32_t* fresh_slots, uint32_t* fresh_ctr,
-        uint32_t* dbg) {
+        const char* pool, uint32_t p_mask, int32_t* out_kid, uint32_t* dbg) {
     /* phase 1: every row probes; exactly one row CASes each new
      * fingerprint in, recording itself as the claiming row. No lane ever
      * waits on another (a publish-wait design can cycle across waves).
      * STEADY STATE (slot already has an assigned id from an earlier batch):
-     * byte-verify and resolve right here — out_kid gets the FINAL id. Rows
-     * whose key was claimed only this batch (id not yet assigned) go on a
-     * compact fresh list the fix-up phase resolves; in steady state the
-     * list is empty and the fix-up is a no-op — no per-row third pass. */
+     * byte-verify and resolve right here — out_kid = ~id — so the lookup
+     * phase touches key bytes only for rows of freshly claimed keys. */
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += stride) {
@@ -169,7 +163,7 @@ __global__ __launch_bounds__(BLOCK) void k_intern_claim(const int32_t* offs,
         uint64_t fp = fnv1a64(data + o0, len);
         if (!fp) fp = 1; /* 0 marks an empty slot */
         uint32_t slot = (uint32_t)fp & p_mask;
-        int32_t out = -1;
+        int32_t out;
         for (uint32_t probes = 0;; slot = (slot + 1) & p_mask) {
             /* fp first as an aligned 8 B load (single-copy atomic vs a
              * concurrent claim CAS — the fp halves of a 16 B vector load
@@ -253,21 +247,22 @@ __global__ __launch_bounds__(BLOCK) void k_intern_assign(uint4* tab,
 }
 
 __global__ __launch_bounds__(BLOCK) void k_intern_lookup(const int32_t* offs,
-        const char* data, const uint32_t* fresh_rows,
-        const uint32_t* fresh_slots, const uint32_t* fresh_ctr,
-        const uint4* tab, const uint32_t* id_off, const uint32_t* id_len,
-        const char* pool, int32_t* out_kid, uint32_t* dbg) {
-    /* phase 3 (after assign): fix up ONLY the fresh-listed rows — resolve
-     * their stashed slot to its now-assigned dense id, byte-verifying
-     * against the pool (distinct keys sharing a full 64-bit fingerprint
-     * that also collided in the claim are FLAGGED loudly). Steady state:
-     * the list is empty and this grid exits immediately. */
-    const uint32_t nf = *fresh_ctr;
+        const char* data, int64_t n, const uint4* tab,
+        const uint32_t* id_off, const uint32_t* id_len, const char* pool,
+        int32_t* out_kid, uint32_t* dbg) {
+    /* phase 3 (after assign): resolve each row's slot (stashed by the claim
+     * phase) to its dense id, byte-verifying against the pool — distinct
+     * keys sharing a full 64-bit fingerprint cannot be interned and are
+     * FLAGGED (results then fail loudly via the guard cells). */
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; j < nf;
-         j += stride) {
-        const uint32_t i = fresh_rows[j];
-        const uint32_t slot = fresh_slots[j];
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        const int32_t v = out_kid[i];
+        if (v < 0) { /* resolved inline by the claim phase */
+            out_kid[i] = ~v;
+            continue;
+        }
+        const uint32_t slot = (uint32_t)v;
         const uint32_t cand = tab[slot].z;
         int32_t id = 0;
         if (cand == ~0u) {
@@ -288,24 +283,20 @@ void launch_intern(hipStream_t s, const int32_t* d_offs, const char* d_data,
                    int64_t n, uint4* tab, uint32_t* tab_row, uint32_t p_mask,
                    uint32_t* id_off, uint32_t* id_len, char* pool,
                    uint32_t* ctrs, uint32_t id_cap, uint32_t pool_cap,
-                   int32_t* out_kid, uint32_t* fresh_rows,
-                   uint32_t* fresh_slots, uint32_t* fresh_ctr,
-                   uint32_t* dbg) {
+                   int32_t* out_kid, uint32_t* dbg) {
     int blocks = (int)std::min<int64_t>((n + BLOCK - 1) / BLOCK, 2048);
     if (blocks < 1) blocks = 1;
     const uint32_t P = p_mask + 1;
     int sblocks = (int)std::min<uint32_t>((P + BLOCK - 1) / BLOCK, 2048);
-    hipLaunchKernelGGL(k_zero1, dim3(1), dim3(1), 0, s, fresh_ctr);
     hipLaunchKernelGGL(k_intern_claim, dim3(blocks), dim3(BLOCK), 0, s, d_offs,
                        d_data, n, tab, tab_row, id_off, id_len,
-                       pool, p_mask, out_kid, fresh_rows, fresh_slots,
-                       fresh_ctr, dbg);
+                       pool, p_mask, out_kid, dbg);
     hipLaunchKernelGGL(k_intern_assign, dim3(sblocks), dim3(BLOCK), 0, s,
                        tab, tab_row, P, d_offs, d_data, id_off,
                        id_len, pool, ctrs, id_cap, pool_cap, dbg);
     hipLaunchKernelGGL(k_intern_lookup, dim3(blocks), dim3(BLOCK), 0, s,
-                       d_offs, d_data, fresh_rows, fresh_slots, fresh_ctr,
-                       tab, id_off, id_len, pool, out_kid, dbg);
+                       d_offs, d_data, n, tab, id_off,
+                       id_len, pool, out_kid, dbg);
 }
 
 /* synthetic utf8 key generator: "sensor_{k}" with k from the SAME splitmix

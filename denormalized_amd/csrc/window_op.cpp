@@ -235,9 +235,6 @@ struct dz_window_op {
     uint32_t i_idcap = 0, i_poolcap = 0;
     int32_t* d_ikid[2] = {};        /* interned dense ids, per pipeline buf */
     int64_t i_kid_cap[2] = {0, 0};
-    uint32_t* d_ifresh = nullptr;   /* fresh-row fixup list: rows + slots */
-    uint32_t* d_ifresh_ctr = nullptr;
-    int64_t i_fresh_cap = 0;
     int64_t mirror_keys = 0;        /* ids mirrored into dict_strs so far */
 
     /* input staging (host-batch path) */
@@ -692,7 +689,6 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     hipFree(op->d_ioff); hipFree(op->d_ilen);
     hipFree(op->d_ipool); hipFree(op->d_ictrs);
     hipFree(op->d_ikid[0]); hipFree(op->d_ikid[1]);
-    hipFree(op->d_ifresh); hipFree(op->d_ifresh_ctr);
     hipFree(op->d_ts); hipFree(op->d_kid); hipFree(op->d_vals); hipFree(op->d_valbm);
     for (int i = 0; i < 2; i++) {
         hipFree(op->d_sts[i]); hipFree(op->d_skid[i]); hipFree(op->d_svals[i]);
@@ -2199,13 +2195,6 @@ static dz_status ensure_intern(dz_window_op* op, int64_t n) {
         CHK(op, hipMalloc(&op->d_ikid[b], (size_t)n * 4));
         op->i_kid_cap[b] = n;
     }
-    if (n > op->i_fresh_cap) {
-        quiesce(op);
-        hipFree(op->d_ifresh);
-        CHK(op, hipMalloc(&op->d_ifresh, (size_t)n * 8));
-        if (!op->d_ifresh_ctr) CHK(op, hipMalloc(&op->d_ifresh_ctr, 4));
-        op->i_fresh_cap = n;
-    }
     return DZ_OK;
 }
 
@@ -2270,9 +2259,7 @@ extern "C" dz_status dz_window_op_push_device_utf8(dz_window_op* op,
                           op->d_itab, op->d_itab_row,
                           op->i_pmask, op->d_ioff, op->d_ilen, op->d_ipool,
                           op->d_ictrs, op->i_idcap, op->i_poolcap,
-                          op->d_ikid[b], op->d_ifresh,
-                          op->d_ifresh + op->i_fresh_cap, op->d_ifresh_ctr,
-                          op->d_dbg);
+                          op->d_ikid[b], op->d_dbg);
     });
     if (stage_core(op, n_rows, d_ts_ms, op->d_ikid[b], d_vals, nullptr,
                    /*keys_are_dense=*/true, /*deferred=*/true,
