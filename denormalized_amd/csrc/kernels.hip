@@ -1838,15 +1838,12 @@ __global__ __launch_bounds__(BLOCK) void k_json_parse(const char* buf,
                 sbuf[i - lo] = buf[i];
         }
         __syncthreads();
+        const char* base = use_lds ? sbuf - lo : buf;
         const int64_t r = r0 + threadIdx.x;
         if (r < nt) {
-        const int64_t ro = rec_off[r];
-        const int64_t eo = (r + 1 < nt) ? rec_off[r + 1] - 1 : nbytes;
-        /* in-bounds pointers only (a biased sbuf-minus-lo pointer is
-         * out-of-object arithmetic the optimizer may punish) */
-        const char* p = use_lds ? sbuf + (ro - lo) : buf + ro;
-        const char* e = use_lds ? sbuf + (eo - lo) : buf + eo;
-        const char* rec0 = p;
+        const char* p = base + rec_off[r];
+        const char* e = (r + 1 < nt) ? base + rec_off[r + 1] - 1
+                                     : base + nbytes;
         while (e > p && (*(e - 1) == '\n' || *(e - 1) == '\r')) e--;
         int64_t ts = 0;
         int64_t kbeg = -1;
@@ -1886,7 +1883,8 @@ __global__ __launch_bounds__(BLOCK) void k_json_parse(const char* buf,
                 }
                 if (bad || p >= e) { bad = true; break; }
                 if (is_key) {
-                    kbeg = ro + (int64_t)(s - rec0); /* global offset */
+                    kbeg = (int64_t)(s - base); /* global offset either way:
+                        base is buf, or sbuf biased by -lo */
                     klen = (int32_t)(p - s);
                     have_key = true;
                 } else if (is_ts || is_val) {
