@@ -1811,8 +1811,6 @@ __device__ __forceinline__ const char* j_ws(const char* p, const char* e) {
     return p;
 }
 
-constexpr int JP_LDS = 48 * 1024; /* per-block record staging */
-
 __global__ __launch_bounds__(BLOCK) void k_json_parse(const char* buf,
         const int64_t* rec_off, const uint32_t* tot, int64_t nbytes,
         JsonFields jf, int64_t* o_ts, int64_t* o_kbeg, int32_t* o_klen,
@@ -1820,30 +1818,12 @@ __global__ __launch_bounds__(BLOCK) void k_json_parse(const char* buf,
     const double p10[23] = {1e0, 1e1, 1e2, 1e3, 1e4, 1e5, 1e6, 1e7, 1e8, 1e9,
                             1e10, 1e11, 1e12, 1e13, 1e14, 1e15, 1e16, 1e17,
                             1e18, 1e19, 1e20, 1e21, 1e22};
-    /* per-block LDS staging of the block's CONTIGUOUS record byte range:
-     * 64 lanes each walking their own record as a byte interpreter thrash
-     * L1 with 64 divergent streams — staged once (coalesced) the
-     * per-iteration byte load is LDS-speed. Blocks whose range exceeds the
-     * staging (very long records) fall back to global reads. */
-    __shared__ char sbuf[JP_LDS];
     const uint32_t nt = tot[0];
-    for (int64_t r0 = (int64_t)blockIdx.x * BLOCK; r0 < nt;
-         r0 += (int64_t)gridDim.x * BLOCK) {
-        const int64_t r1 = i64min(nt, r0 + BLOCK);
-        const int64_t lo = rec_off[r0];
-        const int64_t hib = (r1 < nt) ? rec_off[r1] : nbytes;
-        const bool use_lds = hib - lo <= JP_LDS;
-        if (use_lds) {
-            for (int64_t i = lo + threadIdx.x; i < hib; i += BLOCK)
-                sbuf[i - lo] = buf[i];
-        }
-        __syncthreads();
-        const char* base = use_lds ? sbuf - lo : buf;
-        const int64_t r = r0 + threadIdx.x;
-        if (r < nt) {
-        const char* p = base + rec_off[r];
-        const char* e = (r + 1 < nt) ? base + rec_off[r + 1] - 1
-                                     : base + nbytes;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; r < nt;
+         r += stride) {
+        const char* p = buf + rec_off[r];
+        const char* e = (r + 1 < nt) ? buf + rec_off[r + 1] - 1 : buf + nbytes;
         while (e > p && (*(e - 1) == '\n' || *(e - 1) == '\r')) e--;
         int64_t ts = 0;
         int64_t kbeg = -1;
@@ -1883,8 +1863,7 @@ __global__ __launch_bounds__(BLOCK) void k_json_parse(const char* buf,
                 }
                 if (bad || p >= e) { bad = true; break; }
                 if (is_key) {
-                    kbeg = (int64_t)(s - base); /* global offset either way:
-                        base is buf, or sbuf biased by -lo */
+                    kbeg = (int64_t)(s - buf);
                     klen = (int32_t)(p - s);
                     have_key = true;
                 } else if (is_ts || is_val) {
@@ -1981,8 +1960,6 @@ __global__ __launch_bounds__(BLOCK) void k_json_parse(const char* buf,
         o_kbeg[r] = kbeg;
         o_klen[r] = klen;
         o_val[r] = val;
-        }
-        __syncthreads(); /* staging reused next block-round */
     }
 }
 
