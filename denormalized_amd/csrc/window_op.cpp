@@ -1368,9 +1368,13 @@ static dz_status trigger_windows(dz_window_op* op) {
      * closes (hint above the threshold) keep the per-close chain with its
      * zero-copy large-output path. Worst-case safe: gstep is capped so
      * gcount * kcap <= E_GELEMS — the fused gather cannot overflow. */
+    static const uint32_t group_max = [] {
+        const char* v = getenv("DZ_EMIT_GROUP_MAX");
+        return v ? (uint32_t)atoi(v) : 65536u;
+    }();
     const bool grouped_ok = dev_path_g && op->kcap > 0 &&
         op->kcap <= dz_window_op::E_GELEMS &&
-        op->e_nt_hint.load(std::memory_order_relaxed) <= 65536 &&
+        op->e_nt_hint.load(std::memory_order_relaxed) <= group_max &&
         op->batch_seq < (1u << 19);
     size_t g0 = 0;
     while (g0 < closed.size()) {
