@@ -64,5 +64,19 @@ def test_create_fails_loudly_without_gpu(built):
         WindowOp(length_ms=1000)
 
 
+def test_join_and_decoder_fail_loudly_without_gpu(built):
+    try:
+        import torch
+        if torch.cuda.is_available():
+            pytest.skip("GPU present")
+    except Exception:
+        pass
+    from denormalized_amd import JoinOp, JsonDecoder
+    with pytest.raises(RuntimeError, match="no HIP device|create failed"):
+        JoinOp(device=0)
+    with pytest.raises(RuntimeError, match="no HIP device|create failed"):
+        JsonDecoder(device=0)
+
+
 def test_version(built):
     assert b"gfx950" in built.dz_version()
