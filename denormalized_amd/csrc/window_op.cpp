@@ -116,15 +116,30 @@ struct KStatAcc {
     double last_bytes = 0.0;
 };
 
+struct GroupCtl {
+    /* sibling coordination for one group-batched emission chain */
+    std::atomic<int> claimed{0};
+    std::atomic<bool> ready{false};
+    uint32_t offs[17] = {};
+    uint32_t gtot = 0;
+    int span = -1;              /* pinned span index, or -1 (heap fallback) */
+    char* heap = nullptr;       /* liveness fallback when no span was free */
+    std::atomic<int> holders{0};   /* zero-copy slice OutBufs alive */
+    std::atomic<bool> all_built{false};
+};
+
 struct OutBuf {
     /* zero-copy device-path emission: the batch's aggregate columns (and
      * the dense key column) are served as views into the pinned packed
      * span; the OutBuf holds the slab until the consumer moves to the next
      * batch (poll pop releases it). hold_slab == -1 => fully materialized. */
     const char* packed = nullptr;
-    uint32_t packed_nt = 0;
+    uint32_t packed_nt = 0;   /* column-section stride of the packed span */
+    uint32_t packed_off = 0;  /* this batch's slice offset (group spans) */
     int hold_slab = -1;
     uint32_t slab_gen = 0;  /* pool generation the held slab belongs to */
+    std::shared_ptr<GroupCtl> hold_group; /* group span held until poll
+                                           * advances past this batch */
     std::vector<int64_t> key_i64;
     std::vector<int32_t> key_offsets;
     std::vector<char> key_data;
@@ -302,12 +317,6 @@ struct dz_window_op {
     uint32_t* d_rhist[E_CSTREAMS] = {}; /* radix scratch, one per c_stream */
     uint32_t* d_roffs[E_CSTREAMS] = {};
     int64_t e_slab_kcap = 0;
-    struct GroupCtl { /* sibling coordination for a group-batched close */
-        std::atomic<int> claimed{0};
-        std::atomic<bool> ready{false};
-        uint32_t offs[17] = {};
-        uint32_t gtot = 0;
-    };
     struct EmitJob {
         hipEvent_t ev;
         int slab;
@@ -384,7 +393,9 @@ struct dz_window_op {
     static constexpr int E_GD = 4; /* groups in flight (one per
                                     * emission stream) */
     GDev e_gd[E_GD];
-    char* e_gpin[E_GD] = {};      /* pinned packed staging */
+    char* e_gpin[E_GD] = {};      /* pinned packed span pool (free-listed;
+                                   * held by zero-copy slice consumers) */
+    std::vector<int> e_gspan_free; /* guarded by e_mtx */
     uint32_t* e_gpcnt = nullptr;  /* pinned, E_GD x 32 counters */
     std::vector<int> e_gdfree;    /* guarded by e_mtx */
     bool e_gd_ready = false;
@@ -430,6 +441,8 @@ static dz_status intern_sync_mirror(dz_window_op* op);
 static void build_emission_slice(dz_window_op* op, int64_t wstart,
                                  int64_t wend, uint32_t total, uint32_t off,
                                  uint32_t n, const char* p, OutBuf* out);
+static void build_emission_slice_views(dz_window_op* op, int64_t wstart,
+                                       int64_t wend, uint32_t n, OutBuf* out);
 
 static hipEvent_t get_event(dz_window_op* op) {
     if (!op->ev_pool.empty()) {
@@ -958,7 +971,9 @@ static OutBuf take_outbuf(dz_window_op* op) {
     op->ob_pool.pop_back();
     ob.packed = nullptr;
     ob.packed_nt = 0;
+    ob.packed_off = 0;
     ob.hold_slab = -1;
+    ob.hold_group.reset();
     ob.key_i64.clear();
     ob.key_offsets.clear();
     ob.key_data.clear();
@@ -1027,9 +1042,12 @@ static void emit_worker_main(dz_window_op* op) {
         auto t0 = std::chrono::steady_clock::now();
         OutBuf ob = take_outbuf(op);
         if (job.grouped) {
-            /* group-batched close: the first sibling to claim pulls the
-             * whole packed span D2H and publishes the per-close offsets;
-             * every sibling then builds its own slice */
+            /* group-batched close: the first sibling to claim grabs a
+             * pinned span (heap fallback keeps liveness if consumers lag),
+             * pulls the whole packed block D2H and publishes the per-close
+             * offsets; every sibling then builds its own slice — ZERO-COPY
+             * views into the span for large slices, the span held until
+             * every viewing batch has been consumed */
             auto& ctl = *job.ctl;
             if (ctl.claimed.exchange(1) == 0) {
                 const uint32_t* c = op->e_gpcnt + 32 * job.gbuf;
@@ -1037,9 +1055,24 @@ static void emit_worker_main(dz_window_op* op) {
                 ctl.offs[0] = 0;
                 for (int i = 0; i < 16; i++)
                     ctl.offs[i + 1] = ctl.offs[i] + c[1 + i];
-                if (gtot)
-                    hipMemcpy(op->e_gpin[job.gbuf], op->e_gd[job.gbuf].pout,
+                char* dst = nullptr;
+                if (gtot) {
+                    {
+                        std::lock_guard<std::mutex> lk(op->e_mtx);
+                        if (!op->e_gspan_free.empty()) {
+                            ctl.span = op->e_gspan_free.back();
+                            op->e_gspan_free.pop_back();
+                        }
+                    }
+                    if (ctl.span >= 0) {
+                        dst = op->e_gpin[ctl.span];
+                    } else {
+                        ctl.heap = (char*)malloc((size_t)gtot * 53);
+                        dst = ctl.heap;
+                    }
+                    hipMemcpy(dst, op->e_gd[job.gbuf].pout,
                               (size_t)gtot * 53, hipMemcpyDeviceToHost);
+                }
                 ctl.gtot = gtot;
                 ctl.ready.store(true, std::memory_order_release);
             } else {
@@ -1049,8 +1082,19 @@ static void emit_worker_main(dz_window_op* op) {
             const uint32_t off = ctl.offs[job.goff];
             const uint32_t cnt = ctl.offs[job.goff + 1] - off;
             op->e_nt_hint.store(cnt, std::memory_order_relaxed);
-            build_emission_slice(op, job.wstart, job.wend, ctl.gtot, off, cnt,
-                                 op->e_gpin[job.gbuf], &ob);
+            const char* src = ctl.span >= 0 ? op->e_gpin[ctl.span] : ctl.heap;
+            const bool zc = ctl.span >= 0 && cnt >= 16384;
+            if (zc) {
+                ctl.holders.fetch_add(1);
+                ob.packed = src;
+                ob.packed_nt = ctl.gtot;
+                ob.packed_off = off;
+                ob.hold_group = job.ctl;
+                build_emission_slice_views(op, job.wstart, job.wend, cnt, &ob);
+            } else {
+                build_emission_slice(op, job.wstart, job.wend, ctl.gtot, off,
+                                     cnt, src, &ob);
+            }
         } else if (job.device) {
             /* copy exactly nt2 packed rows (the filter already ran on
              * device, so this is the final output volume, not the keyspace) */
@@ -1121,8 +1165,21 @@ static void emit_worker_main(dz_window_op* op) {
              * sibling is past its spin (i.e. fully built) */
             if (!job.grp_left || job.grp_left->fetch_sub(1) == 1) {
                 op->e_ev_pool.push_back(job.ev);
-                if (job.grouped) op->e_gdfree.push_back(job.gbuf);
-                else if (job.gbuf >= 0) op->e_gfree.push_back(job.gbuf);
+                if (job.grouped) {
+                    op->e_gdfree.push_back(job.gbuf);
+                    auto& ctl = *job.ctl;
+                    ctl.all_built.store(true);
+                    if (ctl.holders.load() == 0) { /* no zero-copy viewers */
+                        if (ctl.span >= 0) {
+                            op->e_gspan_free.push_back(ctl.span);
+                            ctl.span = -1;
+                        }
+                        free(ctl.heap);
+                        ctl.heap = nullptr;
+                    }
+                } else if (job.gbuf >= 0) {
+                    op->e_gfree.push_back(job.gbuf);
+                }
             }
             op->e_inflight--;
         }
@@ -1235,7 +1292,10 @@ static dz_status ensure_gd(dz_window_op* op) {
                           (size_t)dz_window_op::E_GD * 32 * 4));
     {
         std::lock_guard<std::mutex> lk(op->e_mtx);
-        for (int i = 0; i < dz_window_op::E_GD; i++) op->e_gdfree.push_back(i);
+        for (int i = 0; i < dz_window_op::E_GD; i++) {
+            op->e_gdfree.push_back(i);
+            op->e_gspan_free.push_back(i);
+        }
     }
     op->e_gd_ready = true;
     return DZ_OK;
@@ -1293,6 +1353,42 @@ static void build_emission_slice(dz_window_op* op, int64_t wstart,
         }
     }
     ob.agg_valid.assign(pfl, pfl + n);
+    ob.wstart.assign(n, wstart);
+    ob.wend.assign(n, wend);
+    ob.view.n_rows = (int64_t)n;
+}
+
+/* Zero-copy slice build: the aggregate columns (and dense keys +
+ * validity) are served by poll() as views into the group span at
+ * packed_off; only dictionary-encoded keys materialize here. */
+static void build_emission_slice_views(dz_window_op* op, int64_t wstart,
+                                       int64_t wend, uint32_t n, OutBuf* out) {
+    OutBuf& ob = *out;
+    const char* p = ob.packed;
+    const uint32_t total = ob.packed_nt;
+    const uint32_t* pkid =
+        (const uint32_t*)(p + (size_t)total * 48) + ob.packed_off;
+    size_t na = op->aggs.size();
+    ob.agg_i64.resize(na);
+    ob.agg_f64.resize(na);
+    if (op->no_group) {
+    } else if (op->key_kind == DZ_KEY_UTF8) {
+        ob.key_offsets.resize(n + 1);
+        ob.key_offsets[0] = 0;
+        size_t totalb = 0;
+        for (size_t i = 0; i < n; i++) totalb += op->dict_strs[pkid[i]].size();
+        ob.key_data.resize(totalb);
+        size_t pos = 0;
+        for (size_t i = 0; i < n; i++) {
+            const std::string& str = op->dict_strs[pkid[i]];
+            memcpy(ob.key_data.data() + pos, str.data(), str.size());
+            pos += str.size();
+            ob.key_offsets[i + 1] = (int32_t)pos;
+        }
+    } else if (op->key_kind == DZ_KEY_INT64) {
+        ob.key_i64.resize(n);
+        for (size_t i = 0; i < n; i++) ob.key_i64[i] = op->dict_vals[pkid[i]];
+    } /* dense: the key column is a view (poll) */
     ob.wstart.assign(n, wstart);
     ob.wend.assign(n, wend);
     ob.view.n_rows = (int64_t)n;
@@ -1439,7 +1535,7 @@ static dz_status trigger_windows(dz_window_op* op) {
             CHK(op, hipEventRecord(gev, cs));
         }
         auto left = std::make_shared<std::atomic<int>>(gcount);
-        auto ctl = std::make_shared<dz_window_op::GroupCtl>();
+        auto ctl = std::make_shared<GroupCtl>();
         {
             HostTimer htj(op, "h_trig_jobs");
             std::lock_guard<std::mutex> lk(op->e_mtx);
@@ -2435,29 +2531,42 @@ extern "C" dz_status dz_window_op_poll(dz_window_op* op, const dz_out_batch** ou
     }
     int release = -1;
     uint32_t release_gen = 0;
+    std::shared_ptr<GroupCtl> release_group;
     {
         std::lock_guard<std::mutex> lk(op->out_mtx);
         if (op->outq.empty()) return DZ_OK;
         if (op->has_current) {
-            /* the previous batch's validity ends here: release its slab
-             * (zero-copy batches) and recycle the buffer husk */
+            /* the previous batch's validity ends here: release its slab /
+             * group span (zero-copy batches) and recycle the buffer husk */
             release = op->current.hold_slab;
             release_gen = op->current.slab_gen;
+            release_group = std::move(op->current.hold_group);
             op->current.hold_slab = -1;
             op->current.packed = nullptr;
+            op->current.hold_group.reset();
             if (op->ob_pool.size() < 16)
                 op->ob_pool.push_back(std::move(op->current));
         }
         op->current = std::move(op->outq.front());
         op->outq.pop_front();
     }
-    if (release >= 0) {
+    if (release >= 0 || release_group) {
         {
             std::lock_guard<std::mutex> lk(op->e_mtx);
             /* a hold that survived a pool regrow references graveyarded
              * memory: its slot number must not re-enter the new pool */
-            if (release_gen == op->e_slab_gen)
+            if (release >= 0 && release_gen == op->e_slab_gen)
                 op->e_free.push_back(release);
+            if (release_group &&
+                release_group->holders.fetch_sub(1) == 1 &&
+                release_group->all_built.load()) {
+                if (release_group->span >= 0) {
+                    op->e_gspan_free.push_back(release_group->span);
+                    release_group->span = -1;
+                }
+                free(release_group->heap);
+                release_group->heap = nullptr;
+            }
         }
         op->e_cv.notify_all();
     }
@@ -2466,28 +2575,31 @@ extern "C" dz_status dz_window_op_poll(dz_window_op* op, const dz_out_batch** ou
     ob.agg_ptrs.clear();
     if (ob.packed) {
         /* zero-copy batch: aggregate columns (+ dense keys + validity) are
-         * views into the pinned packed span the OutBuf holds */
+         * views into the pinned packed span the OutBuf holds; group-span
+         * slices carry their offset in packed_off (whole-slab batches have
+         * packed_off 0) */
         const char* p = ob.packed;
         const size_t ntp = ob.packed_nt;
+        const size_t po = ob.packed_off;
         for (size_t a = 0; a < op->aggs.size(); a++) {
             const void* ptr = nullptr;
             switch (op->aggs[a].op) {
-                case DZ_AGG_COUNT: ptr = p + ntp * 8; break;
-                case DZ_AGG_MIN: ptr = p + ntp * 16; break;
-                case DZ_AGG_MAX: ptr = p + ntp * 24; break;
-                case DZ_AGG_SUM: ptr = p + ntp * 32; break;
-                case DZ_AGG_AVG: ptr = p + ntp * 40; break;
+                case DZ_AGG_COUNT: ptr = p + ntp * 8 + po * 8; break;
+                case DZ_AGG_MIN: ptr = p + ntp * 16 + po * 8; break;
+                case DZ_AGG_MAX: ptr = p + ntp * 24 + po * 8; break;
+                case DZ_AGG_SUM: ptr = p + ntp * 32 + po * 8; break;
+                case DZ_AGG_AVG: ptr = p + ntp * 40 + po * 8; break;
             }
             ob.agg_ptrs.push_back(ptr);
         }
         bool dense_key = !op->no_group && op->key_kind == DZ_KEY_DENSE_INT64;
-        ob.view.key_i64 = dense_key ? (const int64_t*)p
+        ob.view.key_i64 = dense_key ? (const int64_t*)p + po
                                     : (ob.key_i64.empty() ? nullptr
                                                           : ob.key_i64.data());
         ob.view.key_offsets =
             ob.key_offsets.empty() ? nullptr : ob.key_offsets.data();
         ob.view.key_data = ob.key_data.empty() ? nullptr : ob.key_data.data();
-        ob.view.agg_valid = (const uint8_t*)(p + ntp * 52);
+        ob.view.agg_valid = (const uint8_t*)(p + ntp * 52 + po);
     } else {
         for (size_t a = 0; a < op->aggs.size(); a++) {
             if (op->aggs[a].op == DZ_AGG_COUNT)

@@ -1216,3 +1216,33 @@ def test_many_windows_split_multibatch_shuffled():
         batches.append((ts, k, v))
     outs, exp = run_both(150, 10, batches, n_keys_hint=900)
     assert_parity(outs, exp)
+
+
+def test_grouped_emission_zero_copy_slices():
+    # >64k keys with a cold passer hint takes the GROUP-BATCHED emission
+    # path; unfiltered closes pass ~all touched keys, so the slices exceed
+    # the zero-copy threshold and the consumer reads VIEWS into the shared
+    # group span (released across successive polls). Bit-exact, all rows.
+    rng = np.random.default_rng(515)
+    n = 600_000
+    nk = 100_000
+    ts = (1_000_000 + np.arange(n) // 200).astype(np.int64)  # 3s span
+    k = rng.integers(0, nk, n)
+    v = rng.uniform(0, 115, n)
+    op = make_op(1000, n_keys_hint=nk)
+    o = pyoracle.Oracle(1000, 0)
+    outs = []
+    step = 150_000
+    for lo in range(0, n, step):
+        sl = slice(lo, lo + step)
+        op.push(ts[sl], k[sl], v[sl])
+        o.push(ts[sl], k[sl], v[sl])
+        outs += op.poll_all()
+    op.finish()
+    o.finish()
+    outs += op.poll_all()
+    exp = o.fetch()
+    assert len(exp["key"]) > 100_000
+    assert_parity(outs, exp)
+    op.close()
+    o.close()
