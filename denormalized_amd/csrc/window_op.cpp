@@ -405,31 +405,6 @@ struct dz_window_op {
     std::atomic<uint64_t> e_copy_builds{0}; /* device-path builds */
     std::mutex out_mtx;             /* guards outq */
 
-    /* ASYNC ENQUEUE: the per-close emission chains cost ~30 HIP calls per
-     * close on the issuing thread (~2-3 ms/step at cfg3-sliding shapes,
-     * measured as stream-1 idle in the kernel trace). A dedicated enqueue
-     * thread takes that latency off the push path: trigger_windows only
-     * records the fold fence and queues the closed-window list.
-     * Invariants:
-     *  - free_slots becomes shared (e_mtx): the enqueue thread pushes a
-     *    slot's entry only AFTER recording its reuse-gate event (an
-     *    unrecorded event waits as a no-op — the order is load-bearing);
-     *  - state_alloc flushes this queue before reallocating s_base: closed
-     *    windows waiting here are not in op->open, so their slab contents
-     *    would not be carried over;
-     *  - per-close reads of n_keys are snapshotted once per request (the
-     *    push thread grows n_keys concurrently; a larger value only adds
-     *    cnt==0 slots to the compact scan). */
-    struct ClosedWin { int64_t start, end; int32_t slot; };
-    struct EmitReq { std::vector<ClosedWin> closes; hipEvent_t evA; };
-    std::deque<EmitReq> t_queue;  /* guarded by e_mtx */
-    int t_pending = 0;            /* reqs queued or mid-enqueue (e_mtx) */
-    bool t_stop = false;          /* guarded by e_mtx */
-    std::thread t_thread;
-    std::atomic<bool> t_err{false};
-    std::atomic<uint64_t> t_enq_ns{0}, t_enq_n{0}, t_slabw_ns{0},
-                          t_gather_ns{0}, t_jobs_ns{0};
-
     /* filter pushdown */
     bool has_filter = false;
     int32_t f_idx = 0, f_cmp = 0;
@@ -460,7 +435,6 @@ struct dz_window_op {
     } while (0)
 
 static void emit_worker_main(dz_window_op* op);
-static void emit_enqueue_main(dz_window_op* op);
 static dz_status ensure_emission(dz_window_op* op);
 static dz_status process_pending(dz_window_op* op);
 static dz_status intern_sync_mirror(dz_window_op* op);
@@ -504,22 +478,6 @@ struct HostTimer {
     HostTimer(dz_window_op* o, const char* n)
         : op(o), name(n), t0(std::chrono::steady_clock::now()) {}
     ~HostTimer();
-};
-
-/* HostTimer writes op->stats (push-thread-only map); the enqueue thread
- * accounts through these instead and kernel_stats merges them. */
-struct AtomicTimer {
-    std::atomic<uint64_t>* ns;
-    std::atomic<uint64_t>* n;
-    std::chrono::steady_clock::time_point t0;
-    explicit AtomicTimer(std::atomic<uint64_t>* ns_,
-                         std::atomic<uint64_t>* n_ = nullptr)
-        : ns(ns_), n(n_), t0(std::chrono::steady_clock::now()) {}
-    ~AtomicTimer() {
-        *ns += (uint64_t)std::chrono::duration_cast<std::chrono::nanoseconds>(
-                   std::chrono::steady_clock::now() - t0).count();
-        if (n) ++*n;
-    }
 };
 
 template <typename F>
@@ -570,13 +528,6 @@ static dz_status state_alloc(dz_window_op* op, int64_t kcap_new, int32_t nslots_
     }
     if (kcap_new == op->kcap && nslots_new <= op->nslots) return DZ_OK;
     nslots_new = std::max(nslots_new, op->nslots);
-    /* closed windows queued for emission are no longer in op->open — their
-     * slab contents would not be carried to the new base. Flush the enqueue
-     * thread first (the c_stream syncs below then drain the issued reads). */
-    if (op->t_thread.joinable()) {
-        std::unique_lock<std::mutex> lk(op->e_mtx);
-        op->e_cv.wait(lk, [&] { return op->t_pending == 0; });
-    }
     size_t cells = (size_t)kcap_new * nslots_new * 5;
     uint64_t* n_base;
     CHK(op, hipMalloc(&n_base, cells * 8));
@@ -607,11 +558,8 @@ static dz_status state_alloc(dz_window_op* op, int64_t kcap_new, int32_t nslots_
     op->s_base = n_base;
     op->s_cnt = n_cnt; op->s_first = n_first; op->s_min = n_min;
     op->s_max = n_max; op->s_sum = n_sum;
-    {
-        std::lock_guard<std::mutex> lk(op->e_mtx);
-        for (int32_t s = op->nslots; s < nslots_new; s++)
-            op->free_slots.push_back({s, nullptr});
-    }
+    for (int32_t s = op->nslots; s < nslots_new; s++)
+        op->free_slots.push_back({s, nullptr});
     op->kcap = kcap_new;
     op->nslots = nslots_new;
     return DZ_OK;
@@ -702,7 +650,6 @@ extern "C" dz_window_op* dz_window_op_create(const dz_window_desc* desc) {
     }
     for (int i = 0; i < dz_window_op::E_WORKERS; i++)
         op->e_workers.emplace_back(emit_worker_main, op);
-    op->t_thread = std::thread(emit_enqueue_main, op);
     return op;
 }
 
@@ -712,14 +659,6 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     process_pending(op); /* flush any deferred device push (best effort) */
     hipStreamSynchronize(op->stream);
     if (op->i_stream) hipStreamSynchronize(op->i_stream);
-    /* enqueue thread first (it feeds the workers' job queue), then the
-     * workers (they drain e_jobs even after e_stop) */
-    {
-        std::lock_guard<std::mutex> lk(op->e_mtx);
-        op->t_stop = true;
-    }
-    op->e_cv.notify_all();
-    if (op->t_thread.joinable()) op->t_thread.join();
     {
         std::lock_guard<std::mutex> lk(op->e_mtx);
         op->e_stop = true;
@@ -727,8 +666,6 @@ extern "C" void dz_window_op_destroy(dz_window_op* op) {
     op->e_cv.notify_all();
     for (auto& w : op->e_workers)
         if (w.joinable()) w.join();
-    for (int s = 0; s < dz_window_op::E_CSTREAMS; s++)
-        if (op->c_streams[s]) hipStreamSynchronize(op->c_streams[s]);
     drain_events(op, true);
     for (auto e : op->ev_pool) hipEventDestroy(e);
     for (auto e : op->e_ev_pool) hipEventDestroy(e);
@@ -1049,16 +986,6 @@ static OutBuf take_outbuf(dz_window_op* op) {
 }
 
 static void emit_drain(dz_window_op* op) {
-    /* t_pending first: e_inflight for a queued req is only incremented when
-     * the enqueue thread pushes its jobs (t_pending stays >0 until then) */
-    std::unique_lock<std::mutex> lk(op->e_mtx);
-    op->e_cv.wait(lk, [&] { return op->t_pending == 0 && op->e_inflight == 0; });
-}
-
-/* workers only — callable FROM the enqueue thread (whose own request keeps
- * t_pending > 0; waiting on it there would deadlock). Safe because the
- * enqueue thread is the only job producer. */
-static void worker_drain(dz_window_op* op) {
     std::unique_lock<std::mutex> lk(op->e_mtx);
     op->e_cv.wait(lk, [&] { return op->e_inflight == 0; });
 }
@@ -1263,7 +1190,7 @@ static void emit_worker_main(dz_window_op* op) {
 static dz_status ensure_emission(dz_window_op* op) {
     if (op->e_slab_kcap == op->kcap) return DZ_OK;
     {
-        worker_drain(op);
+        emit_drain(op);
         int64_t kc = op->kcap;
         int64_t nblk = (kc + dz::EMIT_RCHUNK - 1) / dz::EMIT_RCHUNK;
         for (int s = 0; s < dz_window_op::E_CSTREAMS; s++) {
@@ -1469,20 +1396,11 @@ static void build_emission_slice_views(dz_window_op* op, int64_t wstart,
 
 /* trigger_windows (grouped_window_agg_stream.rs:220-253): closed windows are
  * copied D2H asynchronously and built by the worker thread off the push
- * critical path; dz_window_op_drain/finish wait for completion. The chain
- * ENQUEUE itself also runs off the push thread (see EmitReq above): here we
- * only collect the closes and record the fold fence. */
-static dz_status enqueue_closes(dz_window_op* op,
-                                std::vector<dz_window_op::ClosedWin>& closed,
-                                hipEvent_t evA);
-
+ * critical path; dz_window_op_drain/finish wait for completion. */
 static dz_status trigger_windows(dz_window_op* op) {
     if (!op->has_wm) return DZ_OK;
-    if (op->t_err.load(std::memory_order_acquire)) {
-        if (op->err.empty()) op->err = "emission enqueue failed";
-        return DZ_ERR;
-    }
-    std::vector<dz_window_op::ClosedWin> closed;
+    struct Closed { int64_t start, end; int32_t slot; };
+    std::vector<Closed> closed;
     for (auto it = op->open.begin(); it != op->open.end();) {
         if (op->watermark >= it->second.end) {
             closed.push_back({it->first, it->second.end, it->second.slot});
@@ -1492,10 +1410,11 @@ static dz_status trigger_windows(dz_window_op* op) {
         }
     }
     if (closed.empty()) return DZ_OK;
-    /* emission kernels run on the copy streams AFTER the compute stream's
-     * folds for these slots: the fence is recorded HERE (order on
-     * op->stream matters); the c_stream waits are enqueued by the enqueue
-     * thread before any chain work */
+    if (ensure_emission(op) != DZ_OK) return DZ_ERR;
+    size_t stride = (size_t)op->kcap * 5;
+    HostTimer ht(op, "h_emit_enqueue");
+    /* emission kernels run on the copy stream AFTER the compute stream's
+     * folds for these slots */
     hipEvent_t evA;
     {
         std::lock_guard<std::mutex> lk(op->e_mtx);
@@ -1509,41 +1428,13 @@ static dz_status trigger_windows(dz_window_op* op) {
     {
         HostTimer htf(op, "h_trig_fence");
         CHK(op, hipEventRecord(evA, op->stream));
+        for (int s = 0; s < dz_window_op::E_CSTREAMS; s++)
+            CHK(op, hipStreamWaitEvent(op->c_streams[s], evA, 0));
     }
     {
         std::lock_guard<std::mutex> lk(op->e_mtx);
-        op->t_queue.push_back({std::move(closed), evA});
-        op->t_pending++;
+        op->e_ev_pool.push_back(evA);
     }
-    op->e_cv.notify_all();
-    return DZ_OK;
-}
-
-/* push a slot back for reuse: the gate event MUST already be recorded (an
- * unrecorded event waits as a no-op — the consumer would reset the slot
- * under a still-running emission read) */
-static void push_free_slot(dz_window_op* op, int32_t slot, hipEvent_t ev,
-                           bool owned) {
-    {
-        std::lock_guard<std::mutex> lk(op->e_mtx);
-        op->free_slots.push_back({slot, ev, owned});
-    }
-    op->e_cv.notify_all();
-}
-
-static dz_status enqueue_closes(dz_window_op* op,
-                                std::vector<dz_window_op::ClosedWin>& closed,
-                                hipEvent_t evA) {
-    if (ensure_emission(op) != DZ_OK) return DZ_ERR;
-    /* per-request snapshots: the push thread grows n_keys concurrently; a
-     * close enqueued with a larger n_keys only scans extra cnt==0 slots,
-     * but phase 1/2 of one close must agree on the path choice. kcap only
-     * changes under state_alloc, which flushes this queue first. */
-    const int64_t nK = op->n_keys;
-    size_t stride = (size_t)op->kcap * 5;
-    AtomicTimer ht(&op->t_enq_ns, &op->t_enq_n);
-    for (int s = 0; s < dz_window_op::E_CSTREAMS; s++)
-        CHK(op, hipStreamWaitEvent(op->c_streams[s], evA, 0));
     dz::EmitFilter ef;
     ef.on = op->has_filter ? 1 : 0;
     ef.cmp = op->f_cmp;
@@ -1566,7 +1457,7 @@ static dz_status enqueue_closes(dz_window_op* op,
      * slabs whose release depends on this group's phase 2 (deadlock). */
     struct Pending { int slab; hipEvent_t ev; };
     constexpr size_t EGROUP = dz_window_op::E_POOL / 2;
-    const bool dev_path_g = nK > 65536;
+    const bool dev_path_g = op->n_keys > 65536;
     /* group-batched device path: a whole trigger group of SMALL closes in
      * ONE chain (~25 host enqueues total instead of per close — the
      * push-thread enqueue serialization was the cfg3 sliding wall). Large
@@ -1593,7 +1484,7 @@ static dz_status enqueue_closes(dz_window_op* op,
         int gdidx;
         hipEvent_t gev;
         {
-            AtomicTimer htw(&op->t_slabw_ns);
+            HostTimer htw(op, "h_emit_slabwait");
             std::unique_lock<std::mutex> lk(op->e_mtx);
             op->e_cv.wait(lk, [&] { return !op->e_gdfree.empty(); });
             gdidx = op->e_gdfree.back();
@@ -1616,15 +1507,17 @@ static dz_status enqueue_closes(dz_window_op* op,
         }
         op->e_frontier_idx =
             (op->e_frontier_idx + 1) % dz_window_op::E_FRONTIERS;
-        for (size_t ci = g0; ci < g1; ci++)
+        for (size_t ci = g0; ci < g1; ci++) {
             gs.s[ci - g0] = closed[ci].slot;
+            op->free_slots.push_back({closed[ci].slot, fr, false});
+        }
         const uint64_t max_first = (((uint64_t)op->batch_seq + 1) << 32);
         const int cshift = 64 - __builtin_clzll(max_first);
         {
-            AtomicTimer htg(&op->t_gather_ns);
+            HostTimer htg(op, "h_trig_gather");
             CHK(op, hipMemsetAsync(gd.ctr, 0, 128, cs));
             dz::launch_emission_group_read(cs, op->s_base, (int64_t)stride, gs,
-                                           gcount, nK, op->kcap, ef,
+                                           gcount, op->n_keys, op->kcap, ef,
                                            cshift, gd.gkeys, gd.gkid,
                                            gd.giota, gd.gcnt_col, gd.gmin,
                                            gd.gmax, gd.gsum, gd.gavg,
@@ -1641,12 +1534,10 @@ static dz_status enqueue_closes(dz_window_op* op,
                                    hipMemcpyDeviceToHost, cs));
             CHK(op, hipEventRecord(gev, cs));
         }
-        for (size_t ci = g0; ci < g1; ci++) /* fr is recorded: release */
-            push_free_slot(op, closed[ci].slot, fr, false);
         auto left = std::make_shared<std::atomic<int>>(gcount);
         auto ctl = std::make_shared<GroupCtl>();
         {
-            AtomicTimer htj(&op->t_jobs_ns);
+            HostTimer htj(op, "h_trig_jobs");
             std::lock_guard<std::mutex> lk(op->e_mtx);
             for (size_t ci = g0; ci < g1; ci++) {
                 dz_window_op::EmitJob j;
@@ -1654,7 +1545,7 @@ static dz_status enqueue_closes(dz_window_op* op,
                 j.slab = -1;
                 j.wstart = closed[ci].start;
                 j.wend = closed[ci].end;
-                j.n_keys = nK;
+                j.n_keys = op->n_keys;
                 j.kcap = op->kcap;
                 j.device = true;
                 j.grouped = true;
@@ -1680,7 +1571,7 @@ static dz_status enqueue_closes(dz_window_op* op,
         hipEvent_t gev;
         int gbuf;
         {
-            AtomicTimer htw(&op->t_slabw_ns);
+            HostTimer htw(op, "h_emit_slabwait");
             std::unique_lock<std::mutex> lk(op->e_mtx);
             op->e_cv.wait(lk, [&] { return !op->e_gfree.empty(); });
             gbuf = op->e_gfree.back();
@@ -1702,24 +1593,24 @@ static dz_status enqueue_closes(dz_window_op* op,
             op->e_frontier[op->e_frontier_idx] = fr;
         }
         op->e_frontier_idx = (op->e_frontier_idx + 1) % dz_window_op::E_FRONTIERS;
-        for (size_t ci = g0; ci < g1; ci++)
+        for (size_t ci = g0; ci < g1; ci++) {
             gs.s[ci - g0] = closed[ci].slot;
+            op->free_slots.push_back({closed[ci].slot, fr, false});
+        }
         {
             /* ONE launch reads the closing slots and writes their slabs
              * straight into the pinned group buffer over the host link —
              * fully asynchronous for the push thread (hipMemcpyAsync D2H
              * was measured performing the transfer at ENQUEUE time) */
-            AtomicTimer htg(&op->t_gather_ns);
+            HostTimer htg(op, "h_trig_gather");
             dz::launch_egather_slabs(op->copy_stream, op->s_base, stride, gs,
                                      gcount, op->e_gbufs_dev[gbuf]);
             CHK(op, hipEventRecord(fr, op->copy_stream));
             CHK(op, hipEventRecord(gev, op->copy_stream));
         }
-        for (size_t ci = g0; ci < g1; ci++) /* fr is recorded: release */
-            push_free_slot(op, closed[ci].slot, fr, false);
         auto left = std::make_shared<std::atomic<int>>(gcount);
         {
-            AtomicTimer htj(&op->t_jobs_ns);
+            HostTimer htj(op, "h_trig_jobs");
             std::lock_guard<std::mutex> lk(op->e_mtx);
             for (size_t ci = g0; ci < g1; ci++) {
                 dz_window_op::EmitJob j;
@@ -1727,7 +1618,7 @@ static dz_status enqueue_closes(dz_window_op* op,
                 j.slab = -1;
                 j.wstart = closed[ci].start;
                 j.wend = closed[ci].end;
-                j.n_keys = nK;
+                j.n_keys = op->n_keys;
                 j.kcap = op->kcap;
                 j.device = false;
                 j.grp_left = left;
@@ -1750,14 +1641,15 @@ static dz_status enqueue_closes(dz_window_op* op,
      * mid-group, truncate the group here, run phase 2 for the slabs
      * already taken, and continue (the outer loop re-forms the rest).
      * Without this, >=17 pending closes plus a zero-copy hold backlog
-     * (slabs pinned until the consumer polls) deadlock the pipeline. */
+     * (slabs pinned until the consumer polls) deadlock the pipeline —
+     * reproduced under rocprofv3's enqueue slowdown. */
     size_t g1c = g1;
     for (size_t ci = g0; ci < g1c; ci++) {
         auto& c = closed[ci];
         int slab;
         hipEvent_t ev, slot_ev;
         {
-            AtomicTimer htw(&op->t_slabw_ns);
+            HostTimer htw(op, "h_emit_slabwait");
             std::unique_lock<std::mutex> lk(op->e_mtx);
             if (ci == g0) {
                 op->e_cv.wait(lk, [&] { return !op->e_free.empty(); });
@@ -1783,7 +1675,7 @@ static dz_status enqueue_closes(dz_window_op* op,
         }
         dz_window_op::DevEmit& d = op->e_dev[slab];
         const uint64_t* sl = op->s_base + (size_t)c.slot * stride;
-        const bool dev_path = nK > 65536;
+        const bool dev_path = op->n_keys > 65536;
         hipStream_t cs = op->c_streams[(ci - g0) % dz_window_op::E_CSTREAMS];
         if (!dev_path) {
             /* small keyspaces: ONE pinned copy of the raw slab; the worker
@@ -1793,21 +1685,21 @@ static dz_status enqueue_closes(dz_window_op* op,
                                    hipMemcpyDeviceToHost, cs));
         } else {
         dz::launch_zero_counters(cs, d.counter);
-        if (nK > 0) {
+        if (op->n_keys > 0) {
             dz::launch_emission_slabread(cs,
                                          /*first*/ sl + op->kcap,
                                          /*cnt*/ sl,
                                          /*min*/ (const double*)(sl + 2 * op->kcap),
                                          /*max*/ (const double*)(sl + 3 * op->kcap),
                                          /*sum*/ (const double*)(sl + 4 * op->kcap),
-                                         nK, d.ekeys, d.ekid, d.fkeys,
+                                         op->n_keys, d.ekeys, d.ekid, d.fkeys,
                                          d.fkid, d.fiota, d.counter,
                                          d.counter + 1, ef, d.ocnt, d.omin,
                                          d.omax, d.osum, d.oavg, d.oflags);
         }
         }
         CHK(op, hipEventRecord(slot_ev, cs));
-        push_free_slot(op, c.slot, slot_ev, true);
+        op->free_slots.push_back({c.slot, slot_ev});
         pend.push_back({slab, ev});
     }
     /* phase 2: sorts + column D2H + job hand-off */
@@ -1816,10 +1708,10 @@ static dz_status enqueue_closes(dz_window_op* op,
         int slab = pend[ci - g0].slab;
         hipEvent_t ev = pend[ci - g0].ev;
         dz_window_op::DevEmit& d = op->e_dev[slab];
-        const bool dev_path = nK > 65536;
+        const bool dev_path = op->n_keys > 65536;
         int csi = (int)((ci - g0) % dz_window_op::E_CSTREAMS);
         hipStream_t cs = op->c_streams[csi];
-        if (dev_path && nK > 0) {
+        if (dev_path && op->n_keys > 0) {
             /* adaptive: when the last close passed few groups (filtered
              * sliding closes pass ~1-2%), ONE single-block launch replaces
              * the ~15-launch multi-block radix chain — the launch enqueue
@@ -1838,13 +1730,13 @@ static dz_status enqueue_closes(dz_window_op* op,
                 dz::launch_esort_small(cs, d.fkeys, d.skeys, d.fiota, d.okid,
                                        d.counter + 1, maxk);
             else
-                dz::launch_emission_sort(cs, nK, d.fkeys,
+                dz::launch_emission_sort(cs, op->n_keys, d.fkeys,
                                          d.skeys, d.fiota, d.okid,
                                          d.counter + 1, op->d_rhist[csi],
                                          op->d_roffs[csi], maxk);
             /* pack the final-order columns on device: the worker pulls one
              * contiguous span and builds with sequential copies */
-            dz::launch_emission_permute(cs, nK, d.counter + 1,
+            dz::launch_emission_permute(cs, op->n_keys, d.counter + 1,
                                         d.fiota, d.fkid, d.ocnt, d.omin,
                                         d.omax, d.osum, d.oavg, d.oflags,
                                         d.pout);
@@ -1860,7 +1752,7 @@ static dz_status enqueue_closes(dz_window_op* op,
             j.slab = slab;
             j.wstart = c.start;
             j.wend = c.end;
-            j.n_keys = nK;
+            j.n_keys = op->n_keys;
             j.kcap = op->kcap;
             j.ticket = op->e_ticket_next++;
             j.device = dev_path;
@@ -1872,37 +1764,6 @@ static dz_status enqueue_closes(dz_window_op* op,
     g0 = g1c;
     } /* group loop */
     return DZ_OK;
-}
-
-/* the enqueue thread: drains EmitReqs FIFO (close order == ticket order is
- * preserved — one thread, one queue), returns the fence event to the pool.
- * On error it raises t_err; op->err was already written by CHK inside
- * enqueue_closes (the push thread only reads it after observing t_err —
- * a simultaneous push-thread error would race the string, accepted for a
- * terminal state). */
-static void emit_enqueue_main(dz_window_op* op) {
-    hipSetDevice(op->device);
-    for (;;) {
-        dz_window_op::EmitReq req;
-        {
-            std::unique_lock<std::mutex> lk(op->e_mtx);
-            op->e_cv.wait(lk, [&] { return op->t_stop || !op->t_queue.empty(); });
-            if (op->t_queue.empty()) {
-                if (op->t_stop) return;
-                continue;
-            }
-            req = std::move(op->t_queue.front());
-            op->t_queue.pop_front();
-        }
-        if (enqueue_closes(op, req.closes, req.evA) != DZ_OK)
-            op->t_err.store(true, std::memory_order_release);
-        {
-            std::lock_guard<std::mutex> lk(op->e_mtx);
-            op->e_ev_pool.push_back(req.evA);
-            op->t_pending--;
-        }
-        op->e_cv.notify_all();
-    }
 }
 
 /* ------------------------------------------------------------------ */
@@ -2127,22 +1988,7 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
     for (int64_t r = 0; r < nw; r++) {
         auto it = op->open.find(ws[r]);
         if (it == op->open.end()) {
-            dz_window_op::FreeSlot fs;
-            bool have = false;
-            {
-                /* slots are released by the enqueue thread as it issues the
-                 * chains — wait for pending enqueues before growing state */
-                std::unique_lock<std::mutex> lk(op->e_mtx);
-                op->e_cv.wait(lk, [&] {
-                    return !op->free_slots.empty() || op->t_pending == 0;
-                });
-                if (!op->free_slots.empty()) {
-                    fs = op->free_slots.back();
-                    op->free_slots.pop_back();
-                    have = true;
-                }
-            }
-            if (!have) {
+            if (op->free_slots.empty()) {
                 int32_t cap_slots = std::max(op->max_open, 64);
                 if (op->nslots >= cap_slots) {
                     op->err = "too many open windows (cap " +
@@ -2153,10 +1999,9 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
                                         cap_slots - op->nslots);
                 if (state_alloc(op, op->kcap, op->nslots + grow) != DZ_OK)
                     return DZ_ERR;
-                std::lock_guard<std::mutex> lk(op->e_mtx);
-                fs = op->free_slots.back();
-                op->free_slots.pop_back();
             }
+            dz_window_op::FreeSlot fs = op->free_slots.back();
+            op->free_slots.pop_back();
             if (fs.ev) { /* emission read of this slot may still be in flight */
                 CHK(op, hipStreamWaitEvent(op->stream, fs.ev, 0));
                 if (fs.ev_owned) {
@@ -2894,13 +2739,6 @@ extern "C" dz_status dz_window_op_kernel_stats(dz_window_op* op,
         s.ms = op->e_build_ns.load() / 1e6;
         op->stats["h_emit_zc"].launches = op->e_zc_builds.load();
         op->stats["h_emit_copybuild"].launches = op->e_copy_builds.load();
-        /* enqueue-thread timers (AtomicTimer; no longer on the push path) */
-        KStatAcc& e = op->stats["h_emit_enqueue"];
-        e.launches = op->t_enq_n.load();
-        e.ms = op->t_enq_ns.load() / 1e6;
-        op->stats["h_emit_slabwait"].ms = op->t_slabw_ns.load() / 1e6;
-        op->stats["h_trig_gather"].ms = op->t_gather_ns.load() / 1e6;
-        op->stats["h_trig_jobs"].ms = op->t_jobs_ns.load() / 1e6;
     }
     int32_t n = 0;
     for (auto& kv : op->stats) {
