@@ -1246,3 +1246,36 @@ def test_grouped_emission_zero_copy_slices():
     assert_parity(outs, exp)
     op.close()
     o.close()
+
+
+@pytest.mark.timeout(240)
+def test_finish_drain_many_closes_no_polls():
+    # LIVENESS regression shape for the mid-group slab-pool fix: dozens of
+    # device-path closes (>64k keys) drain at finish() with NO intermediate
+    # polls, so zero-copy holds ratchet the DevEmit pool down while phase 1
+    # of a 16-close group is still taking slabs. Pre-fix this could
+    # deadlock (reproduced under rocprofv3 slowdown); post-fix the group
+    # truncates at the empty-pool wait and drains. Bit-exact as always.
+    rng = np.random.default_rng(616)
+    n = 2_000_000
+    nk = 100_000
+    # 30 one-second windows, all closing at finish
+    ts = (1_000_000 + (np.arange(n) * 30_000 // n)).astype(np.int64)
+    k = rng.integers(0, nk, n)
+    v = rng.uniform(0, 115, n)
+    op = make_op(1000, n_keys_hint=nk)
+    o = pyoracle.Oracle(1000, 0)
+    step = 500_000
+    for lo in range(0, n, step):
+        sl = slice(lo, lo + step)
+        op.push(ts[sl], k[sl], v[sl])
+        o.push(ts[sl], k[sl], v[sl])
+        # deliberately no poll: the finish-time drain sees the backlog
+    op.finish()
+    o.finish()
+    outs = op.poll_all()
+    exp = o.fetch()
+    assert len(exp["key"]) > 500_000
+    assert_parity(outs, exp)
+    op.close()
+    o.close()
