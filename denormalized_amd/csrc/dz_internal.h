@@ -35,7 +35,11 @@ struct FoldChunk {
     int32_t w_lo, w_hi;  /* widx range [w_lo, w_hi) this launch folds */
     int32_t k_lo, k_hi;  /* kloc range [k_lo, k_hi) */
     int64_t kcap;        /* state key capacity (multiple of NB) */
-    uint32_t batch_seq;
+    uint64_t row_base;   /* global row id of this batch's row 0: first-seen
+                          * = row_base + rowidx, a monotone stream-lifetime
+                          * counter (same order as the old batch<<32|row
+                          * composite, but window-rebased sort keys stay
+                          * small forever — fixed radix pass count) */
     int32_t bin_stride;  /* binoffs/binlens row stride per bucket */
     int32_t tl_nw;       /* two-level mode: windows per batch (0 = off) */
 };
@@ -109,7 +113,11 @@ void launch_regroup_l2_fold(hipStream_t stream, const uint4* d_grec2,
                             uint64_t* s_first, int64_t slab_cells,
                             uint32_t* d_dbg);
 
-struct EGatherSlots { int32_t s[16]; };
+struct EGatherSlots {
+    int32_t s[16];
+    uint64_t base[16]; /* per-close window-open row base: the sort key is
+                        * first - base, bounded by the window's row span */
+};
 
 struct EmitFilter {
     int32_t on;      /* 0 = no filter */
@@ -125,7 +133,7 @@ struct EmitFilter {
 void launch_emission_slabread(hipStream_t stream, const uint64_t* slab_first,
                               const uint64_t* slab_cnt, const double* slab_min,
                               const double* slab_max, const double* slab_sum,
-                              int64_t K, uint64_t* ekeys, uint32_t* ekid,
+                              uint64_t base, int64_t K, uint64_t* ekeys, uint32_t* ekid,
                               uint64_t* fkeys, uint32_t* fkid, uint32_t* fiota,
                               uint32_t* counter, uint32_t* counter2,
                               const EmitFilter& ef, uint64_t* ocnt, double* omin,

@@ -980,8 +980,7 @@ __global__ __launch_bounds__(BLOCK) void k_regroup_t(const uint4* rrec,
                     }
                 }
                 if (f_fst == ~0ULL)
-                    f_fst = ((uint64_t)fc.batch_seq << 32) |
-                            (s_rec[seg0].z & 0x7FFFFFFFu);
+                    f_fst = fc.row_base + (s_rec[seg0].z & 0x7FFFFFFFu);
                 for (uint32_t r = seg0; r < seg1; r++) {
                     const uint4 rec = s_rec[r];
                     const double v = __longlong_as_double(
@@ -1067,8 +1066,8 @@ constexpr int RCHUNK = 4096;      /* elements per radix block */
 constexpr int RPASSES = 6;        /* 6*11 = 66 >= 64 bits */
 
 __global__ __launch_bounds__(BLOCK) void k_ecompact(const uint64_t* s_first,
-        int64_t K, uint64_t* ekeys, uint32_t* ekid, uint32_t* eiota,
-        uint32_t* counter) {
+        uint64_t fbase, int64_t K, uint64_t* ekeys, uint32_t* ekid,
+        uint32_t* eiota, uint32_t* counter) {
     /* one counter atomic per BLOCK (a returning per-thread atomic on one
      * word serialises); block-local order is irrelevant pre-sort */
     __shared__ uint32_t base;
@@ -1096,7 +1095,7 @@ __global__ __launch_bounds__(BLOCK) void k_ecompact(const uint64_t* s_first,
         __syncthreads();
         if (hit) {
             uint32_t p = base + wbase + wrank;
-            ekeys[p] = f;
+            ekeys[p] = f - fbase; /* window-rebased: small keys, 4 passes */
             ekid[p] = (uint32_t)k;
             eiota[p] = p;
         }
@@ -1352,7 +1351,7 @@ __global__ void k_egather(const uint32_t* fkid, const uint32_t* counter2,
 void launch_emission_slabread(hipStream_t s, const uint64_t* slab_first,
                               const uint64_t* slab_cnt, const double* slab_min,
                               const double* slab_max, const double* slab_sum,
-                              int64_t K, uint64_t* ekeys, uint32_t* ekid,
+                              uint64_t base, int64_t K, uint64_t* ekeys, uint32_t* ekid,
                               uint64_t* fkeys, uint32_t* fkid, uint32_t* fiota,
                               uint32_t* counter, uint32_t* counter2,
                               const EmitFilter& ef, uint64_t* ocnt, double* omin,
@@ -1363,7 +1362,7 @@ void launch_emission_slabread(hipStream_t s, const uint64_t* slab_first,
      * columns — all reads of the window slot's slab happen HERE, so the
      * slot is reusable right after these launches. */
     hipLaunchKernelGGL(k_ecompact, dim3(cblocks), dim3(BLOCK), 0, s, slab_first,
-                       K, ekeys, ekid, fiota /*scratch, rewritten below*/,
+                       base, K, ekeys, ekid, fiota /*scratch, rewritten below*/,
                        counter);
     hipLaunchKernelGGL(k_efilter, dim3(cblocks), dim3(BLOCK), 0, s, ekeys, ekid,
                        counter, slab_cnt, slab_min, slab_max, slab_sum, ef,
@@ -1382,8 +1381,9 @@ void launch_emission_sort(hipStream_t s, int64_t K, uint64_t* fkeys,
     uint32_t* counter = counter2;
     /* sort (first, compact-index) pairs: keys ekeys<->skeys, payload
      * skid<->okid; the pass count covers the HOST-KNOWN key bound
-     * (first = batch_seq<<32 | rowidx, so any run under 2048 batches
-     * needs 4 passes, not the 66-bit worst case of 6), rounded up to
+     * (first is window-rebased to the window's open-time global row
+     * counter, so the bound is the window's row span — 4 passes at any
+     * stream age, not the 66-bit worst case of 6), rounded up to
      * EVEN so the sorted payload lands in skid.
      * (The device path only runs above the 64k-key host-emission cutoff,
      * so the multi-block form is always the right one.) */
@@ -2383,7 +2383,7 @@ __global__ __launch_bounds__(BLOCK) void k_egf(const uint64_t* s_base,
         if (pass) {
             const uint32_t p = base + wbase + wrank;
             const bool valid = cnt > 0;
-            gkeys[p] = ((uint64_t)c << cshift) | f;
+            gkeys[p] = ((uint64_t)c << cshift) | (f - slots.base[c]);
             gkid[p] = (uint32_t)k;
             giota[p] = p; /* sort payload: identity permutation */
             gcnt_col[p] = cnt;

@@ -204,12 +204,15 @@ struct dz_window_op {
                                * reuse but never return it to the pool */
     };
     std::vector<FreeSlot> free_slots;
-    struct OpenWin { int64_t end; int32_t slot; };
+    struct OpenWin { int64_t end; int32_t slot;
+                     uint64_t row_base; /* rows_seen at open: emission sort
+                                         * keys are first - row_base */ };
     std::map<int64_t, OpenWin> open; /* by window start (BTreeMap order) */
 
     int64_t watermark = INT64_MIN;
     bool has_wm = false;
     uint32_t batch_seq = 0;
+    uint64_t rows_seen = 0;  /* stream-lifetime row counter (first-seen ids) */
 
     /* scratch (d_ghist/d_scalars double-buffered: the deferred device-push
      * pipeline runs batch N's reduction on i_stream while batch N-1's scan
@@ -1399,11 +1402,12 @@ static void build_emission_slice_views(dz_window_op* op, int64_t wstart,
  * critical path; dz_window_op_drain/finish wait for completion. */
 static dz_status trigger_windows(dz_window_op* op) {
     if (!op->has_wm) return DZ_OK;
-    struct Closed { int64_t start, end; int32_t slot; };
+    struct Closed { int64_t start, end; int32_t slot; uint64_t base; };
     std::vector<Closed> closed;
     for (auto it = op->open.begin(); it != op->open.end();) {
         if (op->watermark >= it->second.end) {
-            closed.push_back({it->first, it->second.end, it->second.slot});
+            closed.push_back({it->first, it->second.end, it->second.slot,
+                              it->second.row_base});
             it = op->open.erase(it);
         } else {
             ++it;
@@ -1507,11 +1511,17 @@ static dz_status trigger_windows(dz_window_op* op) {
         }
         op->e_frontier_idx =
             (op->e_frontier_idx + 1) % dz_window_op::E_FRONTIERS;
+        uint64_t max_first = 1;
         for (size_t ci = g0; ci < g1; ci++) {
             gs.s[ci - g0] = closed[ci].slot;
+            gs.base[ci - g0] = closed[ci].base;
+            max_first = std::max(max_first,
+                                 op->rows_seen - closed[ci].base);
             op->free_slots.push_back({closed[ci].slot, fr, false});
         }
-        const uint64_t max_first = (((uint64_t)op->batch_seq + 1) << 32);
+        /* composite sort key = close_idx << cshift | (first - base); the
+         * rebased first is bounded by the window's row span, so cshift
+         * stays ~24-34 bits at any stream age */
         const int cshift = 64 - __builtin_clzll(max_first);
         {
             HostTimer htg(op, "h_trig_gather");
@@ -1692,7 +1702,7 @@ static dz_status trigger_windows(dz_window_op* op) {
                                          /*min*/ (const double*)(sl + 2 * op->kcap),
                                          /*max*/ (const double*)(sl + 3 * op->kcap),
                                          /*sum*/ (const double*)(sl + 4 * op->kcap),
-                                         op->n_keys, d.ekeys, d.ekid, d.fkeys,
+                                         c.base, op->n_keys, d.ekeys, d.ekid, d.fkeys,
                                          d.fkid, d.fiota, d.counter,
                                          d.counter + 1, ef, d.ocnt, d.omin,
                                          d.omax, d.osum, d.oavg, d.oflags);
@@ -1717,7 +1727,10 @@ static dz_status trigger_windows(dz_window_op* op) {
              * the ~15-launch multi-block radix chain — the launch enqueue
              * itself was the cfg3 push-thread wall. A misprediction is
              * slow, never wrong (the small kernel handles any nt). */
-            const uint64_t maxk = ((uint64_t)op->batch_seq + 1) << 32;
+            /* sort keys are window-rebased (first - base): the bound is
+             * the window's row span, so the radix pass count stays at
+             * 3-4 regardless of stream age */
+            const uint64_t maxk = op->rows_seen - c.base;
             static const uint32_t small_max = [] {
                 const char* v = getenv("DZ_EMIT_SMALL_MAX");
                 /* default OFF: A/B on one box measured the single-block
@@ -2010,7 +2023,7 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
                 }
             }
             reset_list.push_back(fs.slot);
-            op->open[ws[r]] = {we[r], fs.slot};
+            op->open[ws[r]] = {we[r], fs.slot, op->rows_seen};
             slotmap[r] = fs.slot;
         } else {
             slotmap[r] = it->second.slot;
@@ -2122,7 +2135,7 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
             fc.k_lo = 0;
             fc.k_hi = (int32_t)klocs;
             fc.kcap = op->kcap;
-            fc.batch_seq = op->batch_seq;
+            fc.row_base = op->rows_seen;
             fc.bin_stride = (int32_t)(nb1 * 256);
             fc.tl_nw = (int32_t)nws;
             timed(op, "regroup", (double)nrec_max * 40, [&] {
@@ -2152,7 +2165,7 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
                     fc.k_lo = (int32_t)k_lo;
                     fc.k_hi = (int32_t)(k_lo + nk);
                     fc.kcap = op->kcap;
-                    fc.batch_seq = op->batch_seq;
+                    fc.row_base = op->rows_seen;
                     fc.bin_stride = dz::FOLD_GCAP;
                     fc.tl_nw = 0;
                     timed(op, "regfold", (double)nrec_max * 24, [&] {
@@ -2172,6 +2185,7 @@ static dz_status process_batch(dz_window_op* op, const dz_window_op::Pend& P) {
     CHK(op, hipEventRecord(op->ev_consumed[b], op->stream));
     op->consumed_valid[b] = true;
     op->batch_seq++;
+    op->rows_seen += (uint64_t)n;
 
     /* 4. watermark (running max of batch minimums, :255-266) + trigger */
     if (!op->has_wm || op->watermark <= mn) {
