@@ -52,6 +52,29 @@ def test_windows_for_range_matches_oracle(built):
         assert np.array_equal(ws[:n], ows) and np.array_equal(we[:n], owe)
 
 
+def test_windows_for_range_negative_and_epoch_straddle(built):
+    # pre-epoch and epoch-straddling ranges: the engine and the oracle both
+    # restate the reference's as_secs() truncation (toward zero), so the
+    # snap differs from floor for negative ts — pin that they AGREE there
+    ws = np.empty(65536, np.int64)
+    we = np.empty(65536, np.int64)
+    cases = [
+        (-10_000, -1, 1000, 0), (-10_000, 5_000, 1000, 0),
+        (-7_500, -2_200, 1500, 0), (-10_000, 2_000, 1000, 250),
+        (-5_000, 5_000, 2000, 500), (-999, 999, 500, 100),
+        (-60_001, -59_000, 1000, 0), (0, 0, 1000, 0),
+    ]
+    for mn, mx, len_ms, slide in cases:
+        n = built.dz_debug_windows_for_range(
+            mn, mx, len_ms, slide,
+            ws.ctypes.data_as(ctypes.c_void_p),
+            we.ctypes.data_as(ctypes.c_void_p), 65536)
+        ows, owe = pyoracle.windows_for_range(mn, mx, len_ms, slide)
+        assert n == len(ows), (mn, mx, len_ms, slide)
+        assert np.array_equal(ws[:n], ows) and np.array_equal(we[:n], owe), \
+            (mn, mx, len_ms, slide)
+
+
 def test_create_fails_loudly_without_gpu(built):
     try:
         import torch
