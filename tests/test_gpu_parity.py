@@ -1279,3 +1279,58 @@ def test_finish_drain_many_closes_no_polls():
     assert_parity(outs, exp)
     op.close()
     o.close()
+
+
+_CMP_FN = {"<": np.less, "<=": np.less_equal, ">": np.greater,
+           ">=": np.greater_equal, "==": np.equal, "!=": np.not_equal}
+
+
+def _filter_matrix_case(batches, n_keys_hint, field, cmp):
+    # oracle first (unfiltered) -> pick a literal that SPLITS the groups ->
+    # engine with the pushed-down filter -> expected = masked oracle
+    o = pyoracle.Oracle(1000, 0)
+    for ts, k, v in batches:
+        o.push(ts, k, v)
+    o.finish()
+    exp = o.fetch()
+    o.close()
+    vals = np.asarray(exp[field], np.float64)
+    lit = float(vals[0]) if cmp in ("==", "!=") else \
+        float(np.quantile(vals, 0.5))
+    keep = _CMP_FN[cmp](vals, lit)
+    assert 0 < keep.sum() < len(keep), (field, cmp, lit)
+    op = make_op(1000, n_keys_hint=n_keys_hint,
+                 aggs=(("count", 0), ("min", 0), ("max", 0), ("sum", 0),
+                       ("avg", 0)))
+    op.set_filter(field, cmp, lit)
+    outs = []
+    for ts, k, v in batches:
+        op.push(ts, k, v)
+        outs += op.poll_all()
+    op.finish()
+    outs += op.poll_all()
+    op.close()
+    assert np.array_equal(np.asarray(cat(outs, "key")), exp["key"][keep])
+    for f in ("count", "min", "max", "sum", "avg"):
+        assert np.array_equal(cat(outs, f), exp[f][keep]), f
+
+
+@pytest.mark.timeout(240)
+@pytest.mark.parametrize("field", ["count", "min", "max", "sum", "avg"])
+@pytest.mark.parametrize("cmp", ["<", "<=", ">", ">=", "==", "!="])
+def test_filter_pushdown_matrix_host_path(field, cmp):
+    # every comparator x aggregate-field combination of the pushed-down
+    # filter (streaming_window.rs filter fusion), host-emission path
+    _filter_matrix_case(gen_batches(41, 3, 30_000, 53, 25), 64, field, cmp)
+
+
+@pytest.mark.timeout(240)
+@pytest.mark.parametrize("field,cmp", [
+    ("count", "<="), ("min", ">"), ("max", "=="),
+    ("sum", "<"), ("avg", "!="),
+])
+def test_filter_pushdown_matrix_device_path(field, cmp):
+    # same matrix through the DEVICE emission path (>64k keys: k_efilter /
+    # k_egf run the comparison on-GPU before compaction)
+    _filter_matrix_case(gen_batches(43, 2, 200_000, 70_000, 100), 70_000,
+                        field, cmp)
